@@ -1,0 +1,125 @@
+"""Config loading / strategy codec tests (reference: tests/test_arguments.py)."""
+import json
+import os
+
+import pytest
+
+from hetu_galvatron_amd.config import (
+    GalvatronConfig, HybridParallelPlan, load_config, apply_overrides,
+    MODEL_PRESETS, even_pp_division,
+)
+
+
+def test_default_config():
+    cfg = GalvatronConfig()
+    assert cfg.model.model_name == "llama-3-8b"
+    assert cfg.parallel.mixed_precision == "bf16"
+    assert cfg.model.head_dim == 128
+    assert cfg.model.kv_heads == 8
+
+
+def test_load_with_overrides(tmp_path):
+    y = tmp_path / "cfg.yaml"
+    y.write_text("model:\n  model_name: llama2-7b\ntrain:\n  lr: 0.001\n")
+    cfg = load_config(str(y), ["train.train_iters=7", "parallel.pp_deg=2",
+                              "model.seq_length=2048"])
+    assert cfg.model.hidden_size == 4096
+    assert cfg.model.ffn_hidden_size == 11008  # from llama2-7b preset
+    assert cfg.train.lr == 0.001
+    assert cfg.train.train_iters == 7
+    assert cfg.parallel.pp_deg == 2
+    assert cfg.model.seq_length == 2048  # override beats preset
+
+
+def test_override_parsing():
+    out = apply_overrides({}, ["a.b=true", "a.c=1.5", "a.d=null", "a.e=[1,2]",
+                              "a.f=hello"])
+    assert out["a"] == {"b": True, "c": 1.5, "d": None, "e": [1, 2], "f": "hello"}
+
+
+def test_presets_valid():
+    for name in MODEL_PRESETS:
+        cfg = load_config(base={"model": {"model_name": name}})
+        assert cfg.model.hidden_size % cfg.model.num_attention_heads == 0, name
+
+
+def test_strategy_json_roundtrip(tmp_path):
+    plan = HybridParallelPlan.uniform(num_layers=8, world_size=8, pp=2, tp=2,
+                                      dp_type="zero2", checkpoint=True,
+                                      chunks=4, global_bsz=32)
+    plan.validate(8)
+    p = str(tmp_path / "plan.json")
+    plan.save(p)
+    loaded = HybridParallelPlan.load(p)
+    assert loaded.tp_sizes_enc == [2] * 8
+    assert loaded.pp_deg == 2
+    assert loaded.checkpoint_flags == [1] * 8
+    assert loaded.pp_division == [4, 4]
+    # json format is the reference-compatible comma-joined contract
+    raw = json.load(open(p))
+    assert raw["tp_sizes_enc"] == "2,2,2,2,2,2,2,2"
+    assert isinstance(raw["pp_deg"], int)
+
+
+def test_reference_style_plan_parses():
+    cfg = {
+        "pp_deg": 1,
+        "tp_sizes_enc": "1,1,1,1",
+        "tp_consecutive_flags": "1,1,1,1",
+        "dp_types_enc": "1,1,1,1",
+        "use_sp": "0,0,0,0",
+        "checkpoint": "1,1,0,0",
+        "global_bsz": 16,
+        "chunks": 1,
+        "pp_division": "4",
+        "pipeline_type": "pipedream_flush",
+        "default_dp_type": "zero2",
+        "vtp": 2,
+        "vsp": 1,
+        "embed_sdp": 1,
+    }
+    plan = HybridParallelPlan.from_config_dict(cfg)
+    plan.validate(8)
+    s0 = plan.layer(0, world_size=8)
+    assert s0.dp_type == "zero3" and s0.checkpoint
+    s3 = plan.layer(3, world_size=8)
+    assert s3.dp_type == "zero3" and not s3.checkpoint
+    vs = plan.vocab_strategy(8)
+    assert vs.sp == 2 and vs.tp == 1  # vsp=1 -> ulysses on vocab
+    assert vs.dp_type == "zero3"
+
+
+def test_per_layer_mixed_plan():
+    plan = HybridParallelPlan(
+        pp_deg=1,
+        tp_sizes_enc=[4, 2, 1, 1],
+        tp_consecutive_flags=[1, 1, 1, 1],
+        cp_sizes_enc=[1, 1, 2, 1],
+        dp_types_enc=[0, 0, 1, 1],
+        use_sp=[0, 1, 0, 0],
+        checkpoint_flags=[0, 0, 0, 1],
+        global_bsz=8, chunks=1, default_dp_type="ddp",
+    )
+    plan.validate(8)
+    l0 = plan.layer(0, 8)
+    assert (l0.tp, l0.dp) == (4, 2)
+    l1 = plan.layer(1, 8)
+    assert (l1.sp, l1.tp, l1.dp) == (2, 1, 4) and l1.use_ulysses
+    l2 = plan.layer(2, 8)
+    assert (l2.cp, l2.dp, l2.dp_type) == (2, 4, "zero3")
+    assert l2.sdp == 8
+
+
+def test_even_pp_division():
+    assert even_pp_division(32, 4) == [8, 8, 8, 8]
+    assert even_pp_division(10, 4) == [3, 3, 2, 2]
+
+
+def test_validation_errors():
+    plan = HybridParallelPlan.uniform(4, 8, tp=2)
+    with pytest.raises(ValueError):
+        plan.validate(7)  # tp=2 does not divide stage size 7
+    bad = HybridParallelPlan.uniform(4, 8, tp=2)
+    bad.cp_sizes_enc = [1, 1]
+    with pytest.raises(ValueError):
+        bad.validate(8)
