@@ -119,10 +119,15 @@ class LayerCommGroups:
     dp_group: CommGroup            # pure data-parallel group
     sdp_group: CommGroup           # dp*cp — the ZeRO sharding / grad-reduce domain
     tsp_cp_group: CommGroup        # tp_sp*cp — sequence-layout domain (redistribution)
+    stage_coords: Optional[List["StageCoord"]] = None  # in-stage rank -> coords
+    stage_base: int = 0            # first global rank of this pp stage
     # MoE (None for dense layers)
     ep_group: Optional[CommGroup] = None      # expert-parallel all-to-all
     edp_group: Optional[CommGroup] = None     # data-parallel of experts (grad reduce)
     etp_group: Optional[CommGroup] = None     # tp within experts
+
+    def coord_of(self, global_rank: int) -> "StageCoord":
+        return self.stage_coords[global_rank - self.stage_base]
 
     @property
     def seq_shard_degree(self) -> int:
@@ -136,7 +141,7 @@ class LayerCommGroups:
 
 def _groups_for_layout(
     world_size: int, pp_deg: int, tsp: int, cp: int, consecutive: bool,
-    cache: CommGroupCache, rank: int, ep: int = 1,
+    cache: CommGroupCache, rank: int, ep: int = 1, ulysses: bool = False,
 ) -> Dict[str, CommGroup]:
     """Create (collectively) all groups of one layout; return this rank's."""
     G = world_size // pp_deg
@@ -161,7 +166,14 @@ def _groups_for_layout(
     make(lambda c: (c.dp_idx, c.cp_idx), "tp")
     make(lambda c: (c.dp_idx, c.tp_idx), "cp")
     make(lambda c: (c.cp_idx, c.tp_idx), "dp")
-    make(lambda c: (c.tp_idx,), "sdp")
+    # ZeRO / grad-reduce domain = all ranks holding identical param copies:
+    # megatron-TP shards params over tp -> sdp = dp x cp; ulysses replicates
+    # them over sp too -> sdp = dp x cp x sp (reference: comm_groups.py:310,
+    # SDP group merges dp/sp/cp).
+    if ulysses:
+        make(lambda c: (), "sdp")
+    else:
+        make(lambda c: (c.tp_idx,), "sdp")
     make(lambda c: (c.dp_idx,), "tsp_cp")
 
     if ep > 1:
@@ -193,13 +205,16 @@ def gen_layer_comm_groups(
     for s in strategies:
         layout = _groups_for_layout(
             world_size, s.pp_deg, s.tp_sp, s.cp, s.tp_consecutive, cache, rank,
-            ep=s.ep,
+            ep=s.ep, ulysses=s.use_ulysses,
         )
+        G = world_size // s.pp_deg
         out.append(LayerCommGroups(
             strategy=s,
             tp_group=layout["tp"], sp_group=layout["tp"], cp_group=layout["cp"],
             dp_group=layout["dp"], sdp_group=layout["sdp"],
             tsp_cp_group=layout["tsp_cp"],
+            stage_coords=build_stage_coords(G, s.tp_sp, s.cp, s.tp_consecutive),
+            stage_base=(rank // G) * G,
             ep_group=layout.get("ep"), edp_group=layout.get("edp"),
             etp_group=layout.get("etp"),
         ))

@@ -1,0 +1,170 @@
+"""Autograd-wrapped ops dispatching native HIP (GPU) / torch reference (CPU).
+
+Every op here replaces an external CUDA dependency of the reference:
+  rms_norm / layer_norm  <- flash_attn.ops.rms_norm / dropout_layer_norm
+                            (reference: core/runtime/transformer/norm.py:3-30)
+  swiglu / bias-gelu     <- torch.jit fused kernels
+                            (reference: transformer/fused_kernels.py:143-226)
+  apply_rope             <- flash_attn rotary_emb ext
+                            (reference: transformer/fused_kernels.py:227-257)
+  flash_attention        <- flash_attn_varlen CUDA
+                            (reference: transformer/attention_impl.py:18-108)
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference_ops as ref
+from ._ext import get_ext, use_native
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        if use_native(x):
+            y, invrms = get_ext(False).rmsnorm_fwd(x, weight, eps)
+        else:
+            y, invrms = ref.rmsnorm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight, invrms)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, invrms = ctx.saved_tensors
+        if use_native(x):
+            dx, dw = get_ext(False).rmsnorm_bwd(dy.contiguous(), x, weight, invrms)
+        else:
+            dx, dw = ref.rmsnorm_bwd(dy, x, weight, invrms)
+        return dx, dw.to(weight.dtype), None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    return _RMSNorm.apply(x.contiguous(), weight, eps)
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        if use_native(x):
+            y, mean, invstd = get_ext(False).layernorm_fwd(x, weight, bias, eps)
+        else:
+            y, mean, invstd = ref.layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, invstd = ctx.saved_tensors
+        if use_native(x):
+            dx, dw, db = get_ext(False).layernorm_bwd(dy.contiguous(), x, weight, mean, invstd)
+        else:
+            dx, dw, db = ref.layernorm_bwd(dy, x, weight, mean, invstd)
+        return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-5) -> torch.Tensor:
+    return _LayerNorm.apply(x.contiguous(), weight, bias, eps)
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        if use_native(x):
+            return get_ext(False).swiglu_fwd(x)
+        return ref.swiglu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        if use_native(x):
+            return get_ext(False).swiglu_bwd(dy.contiguous(), x)
+        return ref.swiglu_bwd(dy, x)
+
+
+def swiglu(x: torch.Tensor) -> torch.Tensor:
+    """x[..., 2F] = [gate, up] -> silu(gate)*up."""
+    return _SwiGLU.apply(x.contiguous())
+
+
+class _RoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        ctx.save_for_backward(cos, sin)
+        if use_native(x):
+            return get_ext(False).rope_fwd(x, cos, sin, False)
+        return ref.rope_apply(x, cos, sin, conj=False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_native(dy):
+            return get_ext(False).rope_fwd(dy, cos, sin, True), None, None
+        return ref.rope_apply(dy, cos, sin, conj=True), None, None
+
+
+def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    """x: [s, b, h, d]; cos/sin: [s, d/2] (NEOX half-rotation)."""
+    return _RoPE.apply(x.contiguous(), cos, sin)
+
+
+class _FlashAttention(torch.autograd.Function):
+    """Flash attention with LSE output.
+
+    Layout [b, s, h, d] bf16/fp32; GQA supported (hq multiple of hkv);
+    causal masking bottom-right aligned.  Native path: CDNA4 MFMA kernel
+    (ops/csrc/flash_attn.hip).
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, softmax_scale):
+        scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        if use_native(q):
+            o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale)
+        else:
+            o, lse = ref.attention_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o, lse
+
+    @staticmethod
+    def backward(ctx, do, dlse):
+        q, k, v, o, lse = ctx.saved_tensors
+        do = do.contiguous()
+        if use_native(q):
+            dq, dk, dv = get_ext(False).flash_attn_bwd(
+                do, q, k, v, o, lse, ctx.causal, ctx.scale)
+        else:
+            dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q, k, v, causal: bool = True,
+                    softmax_scale: Optional[float] = None,
+                    return_lse: bool = False):
+    """q: [b,s,hq,d]; k,v: [b,s,hkv,d] -> o [b,s,hq,d] (+ lse [b,hq,s])."""
+    o, lse = _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                                   causal, softmax_scale)
+    return (o, lse) if return_lse else o
+
+
+def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None):
+    """No-autograd forward returning (o, lse) — building block for ring CP."""
+    scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if use_native(q):
+        return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
+                                             v.contiguous(), causal, scale)
+    return ref.attention_fwd(q, k, v, causal, scale)
+
+
+def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True, softmax_scale=None):
+    scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if use_native(q):
+        return get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                             causal, scale)
+    return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)

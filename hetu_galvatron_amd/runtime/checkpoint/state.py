@@ -1,0 +1,118 @@
+"""Canonical full-model state <-> per-rank TP-sharded block params.
+
+Reference: galvatron/core/runtime/checkpoint/llama_adapter.py:30-234 (QKV
+interleave, vocab shard slicing, column/row splits) — rebuilt around OUR
+canonical layout, which is simply the world_size=1 model's state_dict:
+
+  embedding.word_embeddings.weight           [V, h]
+  embedding.position_embeddings.weight       [P, h]        (gpt)
+  decoder.{i}.input_norm.{weight,bias}
+  decoder.{i}.attention.linear_qkv.{weight,bias}   KV-group-major interleave
+  decoder.{i}.attention.linear_proj.{weight,bias}
+  decoder.{i}.post_attn_norm.{weight,bias}
+  decoder.{i}.mlp.fc1.{weight,bias}          gated: [gate(F); up(F)] stacked
+  decoder.{i}.mlp.fc2.{weight,bias}
+  final_norm.norm.{weight,bias}
+  lm_head.lm_head.weight                     [V, h]
+
+Sharding rules per layer strategy (megatron tp=t, rank r; ulysses/cp/dp
+replicate):
+  vocab weights        : row slice V/t
+  linear_qkv.weight    : contiguous row slice = whole KV groups (canonical is
+                         group-major, so slice boundaries align)
+  fc1 (gated)          : [gate_r ; up_r] slices re-stacked per rank
+  column linears       : row slice;  row linears: column slice
+  norms / biases-of-row-linears / pos-emb: replicated
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+
+from ...config.schema import ModelArgs
+from ..models.builder import StageModel
+from ..tensor_parallel.mappings import group_rank, group_size
+
+
+def _block_prefix(block) -> Optional[str]:
+    if block.kind == "embedding":
+        return "embedding"
+    if block.kind == "decoder":
+        return f"decoder.{block.inner.layer_idx}"
+    if block.kind == "final_norm":
+        return "final_norm"
+    if block.kind == "lm_head":
+        return "lm_head"
+    return None
+
+
+def canonical_state_from_stage(sm: StageModel) -> Dict[str, torch.Tensor]:
+    """world_size==1 only: dump the canonical full state."""
+    assert sm.world_size == 1, "canonical dump requires a 1-rank model"
+    out: Dict[str, torch.Tensor] = {}
+    for blk in sm.blocks:
+        prefix = _block_prefix(blk)
+        for name, p in blk.inner.state_dict().items():
+            out[f"{prefix}.{name}"] = p.detach().clone()
+    return out
+
+
+def shard_for_rank(name: str, full: torch.Tensor, strategy, tp_rank: int,
+                   tp_size: int, margs: ModelArgs) -> torch.Tensor:
+    """Slice one canonical tensor for a megatron-TP rank."""
+    t, r = tp_size, tp_rank
+    if t == 1 or strategy.use_ulysses:
+        return full
+    base = name.split(".")[-2] + "." + name.split(".")[-1]
+    if "word_embeddings.weight" in name or "lm_head.weight" in name:
+        V = full.shape[0]
+        return full[r * V // t:(r + 1) * V // t]
+    if "linear_qkv." in name:
+        hkv = margs.kv_heads
+        rows_per_group = full.shape[0] // hkv
+        g0, g1 = r * hkv // t, (r + 1) * hkv // t
+        return full[g0 * rows_per_group:g1 * rows_per_group]
+    if "linear_proj.weight" in name or "fc2.weight" in name:
+        C = full.shape[1]
+        return full[:, r * C // t:(r + 1) * C // t]
+    if "fc1." in name:
+        gated = margs.hidden_act in ("silu", "swiglu", "geglu")
+        if gated:
+            F = full.shape[0] // 2
+            gate, up = full[:F], full[F:]
+            return torch.cat([gate[r * F // t:(r + 1) * F // t],
+                              up[r * F // t:(r + 1) * F // t]], dim=0)
+        R = full.shape[0]
+        return full[r * R // t:(r + 1) * R // t]
+    # norms, row-linear biases, position embeddings: replicated
+    return full
+
+
+def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
+                    margs: ModelArgs) -> None:
+    """Load a canonical full state into this rank's sharded blocks
+    (works for any world size / plan; reference: param_init_fn shard loading
+    parallel.py:87-97)."""
+    for blk in sm.blocks:
+        prefix = _block_prefix(blk)
+        s = blk.groups.strategy
+        tp_size = group_size(blk.groups.tp_group) if not s.use_ulysses else 1
+        tp_rank = group_rank(blk.groups.tp_group) if not s.use_ulysses else 0
+        if blk.flat is not None:
+            blk.flat.gather_params()
+        with torch.no_grad():
+            own = blk.inner.state_dict()
+            for name, p in own.items():
+                full_key = f"{prefix}.{name}"
+                if full_key not in state:
+                    continue
+                shard = shard_for_rank(full_key, state[full_key], s, tp_rank,
+                                       tp_size, margs)
+                if shard.shape != p.shape:
+                    raise ValueError(
+                        f"{full_key}: shard {tuple(shard.shape)} vs param "
+                        f"{tuple(p.shape)} (tp={tp_size} r={tp_rank})")
+                p.copy_(shard.to(p.dtype).to(p.device))
+        if blk.flat is not None:
+            blk.flat.refresh_from_params()

@@ -1,0 +1,65 @@
+"""Data pipeline: synthetic causal-LM batches (+ batch-context assembly).
+
+Reference: galvatron/core/runtime/dataloader.py:36-567 (FakeCausalLMDataset,
+get_batch, loss averaging).  The megatron mmap dataset stack hooks in via
+data.dataset="megatron" (see datasets/).
+
+Design note: every rank materializes the FULL global batch of token ids
+(ids are ~2 bytes/token of int64 -> negligible vs activations) and each
+layer module slices batch/sequence by its own layout — this is what makes
+per-layer dp/tp/cp degrees composable without a per-layer dataloader.
+"""
+from __future__ import annotations
+
+from typing import Dict, Iterator, Optional
+
+import torch
+
+from ..config import GalvatronConfig
+
+
+class SyntheticCausalLMDataset(torch.utils.data.Dataset):
+    """Deterministic synthetic token stream (same on every rank for a given
+    seed + index; reference: dataloader.py:36 FakeCausalLMDataset)."""
+
+    def __init__(self, vocab_size: int, seq_length: int, size: int = 1024,
+                 seed: int = 1234):
+        self.vocab_size = vocab_size
+        self.seq_length = seq_length
+        self.size = size
+        self.seed = seed
+
+    def __len__(self) -> int:
+        return self.size
+
+    def __getitem__(self, idx: int) -> torch.Tensor:
+        g = torch.Generator()
+        g.manual_seed(self.seed * 100003 + idx)
+        return torch.randint(0, self.vocab_size, (self.seq_length + 1,),
+                             generator=g)
+
+
+def build_batch_context(tokens: torch.Tensor, device) -> Dict:
+    """tokens: [B, S+1] -> ctx with input_ids/labels [B, S]."""
+    tokens = tokens.to(device)
+    return {
+        "input_ids": tokens[:, :-1].contiguous(),
+        "labels": tokens[:, 1:].contiguous(),
+        "batch_size": tokens.shape[0],
+        "seq_len": tokens.shape[1] - 1,
+    }
+
+
+def get_train_iterator(cfg: GalvatronConfig, device,
+                       global_batch: Optional[int] = None) -> Iterator[Dict]:
+    """Yield batch contexts of the global batch size, cycling the dataset
+    (reference: dataloader.py:462 get_train_valid_test_data_iterators)."""
+    B = global_batch or cfg.train.global_train_batch_size
+    ds = SyntheticCausalLMDataset(
+        cfg.model.vocab_size, cfg.model.seq_length,
+        size=max(cfg.data.synthetic_dataset_size, B), seed=cfg.train.seed)
+    idx = 0
+    while True:
+        batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])
+        idx = (idx + B) % len(ds)
+        yield build_batch_context(batch, device)

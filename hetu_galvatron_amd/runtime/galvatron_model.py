@@ -1,0 +1,77 @@
+"""Top-level hybrid-parallel model: StageModel + PipelineEngine + step API.
+
+Reference: galvatron/core/runtime/hybrid_parallel_model.py:50-107
+(GalvatronModel.forward_backward dispatching the schedules).
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..config import GalvatronConfig, HybridParallelPlan
+from .models.builder import StageModel, build_hybrid_parallel_model
+from .pipeline.engine import PipelineEngine, StepStats
+
+
+def resolve_plan(cfg: GalvatronConfig, world_size: int) -> HybridParallelPlan:
+    """Searched-plan JSON if given, else GLOBAL-mode uniform plan
+    (reference: hybrid_parallel_config.py:18 get_hybrid_parallel_configs_api)."""
+    p = cfg.parallel
+    if p.galvatron_config_path:
+        plan = HybridParallelPlan.load(p.galvatron_config_path)
+    else:
+        dp_type = "zero3" if p.sdp else p.default_dp_type
+        plan = HybridParallelPlan.uniform(
+            num_layers=cfg.model.num_hidden_layers, world_size=world_size,
+            pp=p.pp_deg, tp=p.global_tp_deg, cp=p.global_cp_deg,
+            use_sp=p.use_ulysses, dp_type=dp_type,
+            checkpoint=bool(p.global_checkpoint),
+            chunks=max(p.chunks, 1), global_bsz=cfg.train.global_train_batch_size,
+            pipeline_type=p.pipeline_type, vtp=p.vocab_tp,
+            vsp=bool(p.vocab_sp))
+    if plan.global_bsz != cfg.train.global_train_batch_size:
+        plan.global_bsz = cfg.train.global_train_batch_size
+    return plan
+
+
+class GalvatronModel(nn.Module):
+    """Hybrid-parallel causal LM ready for forward_backward steps."""
+
+    def __init__(self, cfg: GalvatronConfig,
+                 plan: Optional[HybridParallelPlan] = None,
+                 device: Optional[torch.device] = None):
+        super().__init__()
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        self.cfg = cfg
+        self.plan = plan or resolve_plan(cfg, world)
+        self.stage_model = build_hybrid_parallel_model(cfg, self.plan, device)
+        act_dtype = torch.bfloat16 if cfg.parallel.mixed_precision == "bf16" \
+            else torch.float32
+        self.engine = PipelineEngine(
+            self.stage_model, cfg.model.hidden_size,
+            pipeline_type=self.plan.pipeline_type,
+            overlap_grad_reduce=cfg.parallel.overlap_grad_reduce,
+            act_dtype=act_dtype)
+        self.chunks = max(self.plan.chunks, 1)
+
+    @property
+    def blocks(self):
+        return self.stage_model.blocks
+
+    def forward_backward(self, ctx: Dict, chunks: Optional[int] = None) -> StepStats:
+        return self.engine.forward_backward(ctx, chunks or self.chunks)
+
+    def global_loss(self, stats: StepStats, device=None) -> float:
+        """Token-weighted mean loss across all ranks (handles vtp row
+        duplication by weighting with per-rank token counts)."""
+        if not dist.is_initialized() or dist.get_world_size() == 1:
+            return stats.loss
+        dev = device or self.engine.device
+        t = torch.tensor([stats.loss_sum, stats.token_count],
+                         dtype=torch.float64,
+                         device=dev if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t)
+        return float(t[0] / max(t[1].item(), 1.0))

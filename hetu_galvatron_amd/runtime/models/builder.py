@@ -1,0 +1,207 @@
+"""Declarative model assembly: arch list -> per-stage LayerBlocks.
+
+Reference: galvatron/core/runtime/models/builder.py:42-207 (arch list,
+MODULE_REGISTRY), hybrid_parallel_model.py:107 (6-step assembly pipeline).
+Steps here: plan -> strategies -> comm groups (collective) -> stage modules
+-> LayerBlock wrap (relocation + ZeRO + activation ckpt).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.utils.checkpoint as torch_ckpt
+
+from ...config import GalvatronConfig, HybridParallelPlan, even_pp_division
+from ...config.strategy import LayerStrategy
+from ...core.comm_groups import (
+    CommGroup, CommGroupCache, LayerCommGroups, gen_embedding_group,
+    gen_layer_comm_groups, pp_neighbor_ranks, pp_stage_of_rank,
+)
+from ..redistribute import redistribute
+from ..zero import FlatParamBlock, _PostBackwardHook, _PreBackwardGather
+from .modules import (
+    GalvatronCausalLMHead, GalvatronDecoderLayer, GalvatronEmbedding,
+    GalvatronFinalNorm,
+)
+
+
+def build_causal_lm_arch(num_layers: int) -> List[str]:
+    """reference: builder.py:111 build_causal_lm_arch."""
+    return ["embedding"] + ["decoder"] * num_layers + ["final_norm", "lm_head"]
+
+
+class LayerBlock(nn.Module):
+    """One pipeline cell: input relocation + ZeRO-managed inner module +
+    optional activation checkpointing.
+    Reference: parallel.py:272 Module_with_relocation + wrap_modules_*."""
+
+    def __init__(self, inner: nn.Module, groups: LayerCommGroups, kind: str,
+                 checkpoint: bool = False,
+                 prev_groups: Optional[LayerCommGroups] = None):
+        super().__init__()
+        self.inner = inner
+        self.groups = groups
+        self.kind = kind  # embedding | decoder | final_norm | lm_head
+        self.checkpoint = checkpoint
+        self.prev_groups = prev_groups
+        self.flat: Optional[FlatParamBlock] = None
+
+    def setup_zero(self, mode: str, param_dtype: torch.dtype, device) -> None:
+        tp_group = None
+        if not self.groups.strategy.use_ulysses:
+            tp_group = self.groups.tp_group
+        self.flat = FlatParamBlock(self.inner, mode, self.groups.sdp_group,
+                                   tp_group=tp_group, param_dtype=param_dtype,
+                                   device=device)
+
+    def _inner_forward(self, x, ctx):
+        if self.kind == "embedding":
+            return self.inner(ctx)
+        return self.inner(x, ctx)
+
+    def forward(self, x: Optional[torch.Tensor], ctx: Dict) -> torch.Tensor:
+        if x is not None and self.prev_groups is not None:
+            x = redistribute(x, self.prev_groups, self.groups, ctx["batch_size"])
+        if self.flat is not None:
+            self.flat.gather_params()
+        if x is not None and self.flat is not None and x.requires_grad:
+            x = _PostBackwardHook.apply(self.flat, x)
+        if self.checkpoint and self.kind == "decoder" and torch.is_grad_enabled():
+            out = torch_ckpt.checkpoint(
+                lambda t: self._inner_forward(t, ctx), x, use_reentrant=False)
+        else:
+            out = self._inner_forward(x, ctx)
+        if self.flat is not None:
+            if out.requires_grad:
+                out = _PreBackwardGather.apply(self.flat, out)
+            self.flat.reshard_params()
+        return out
+
+    def finalize_backward(self) -> None:
+        """Idempotent cleanup after a full backward (embedding has no input
+        sentinel; zero3 reshard + auto-sync kick happen here)."""
+        if self.flat is not None:
+            self.flat.post_backward()
+
+
+@dataclass
+class StageModel:
+    """Everything one rank needs to run its pipeline stage."""
+
+    blocks: List[LayerBlock]
+    stage: int
+    pp_deg: int
+    world_size: int
+    rank: int
+    plan: HybridParallelPlan
+    embed_comm_group: Optional[CommGroup] = None
+    tied_embedding_pair: Optional[Tuple[LayerBlock, LayerBlock]] = None
+    cache: Optional[CommGroupCache] = None
+    # layouts at the stage boundaries (for p2p shapes + boundary relocation)
+    recv_layout: Optional[LayerCommGroups] = None   # prev stage's last layer
+    send_layout: Optional[LayerCommGroups] = None   # my last layer
+
+    @property
+    def is_first(self) -> bool:
+        return self.stage == 0
+
+    @property
+    def is_last(self) -> bool:
+        return self.stage == self.pp_deg - 1
+
+    def parameters(self):
+        for b in self.blocks:
+            yield from b.parameters()
+
+
+def layers_of_stage(division: List[int], stage: int) -> Tuple[int, int]:
+    start = sum(division[:stage])
+    return start, start + division[stage]
+
+
+def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
+                                device: Optional[torch.device] = None
+                                ) -> StageModel:
+    """Construct this rank's stage of the hybrid-parallel model.
+
+    Collective: every rank must call with the identical plan
+    (reference: construct_hybrid_parallel_model_api hybrid_parallel_model.py:107).
+    """
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    if device is None:
+        device = torch.device("cuda", torch.cuda.current_device()) \
+            if torch.cuda.is_available() else torch.device("cpu")
+    plan.validate(world)
+    margs = cfg.model
+    n_layers = plan.num_layers
+    pp = plan.pp_deg
+    division = plan.pp_division or even_pp_division(n_layers, pp)
+    my_stage = pp_stage_of_rank(rank, world, pp)
+    dtype = torch.bfloat16 if cfg.parallel.mixed_precision == "bf16" else torch.float32
+
+    strategies = [plan.layer(i, world) for i in range(n_layers)]
+    vocab_strat = plan.vocab_strategy(world)
+    all_strats = [vocab_strat] + strategies
+    groups_list, cache = gen_layer_comm_groups(all_strats, world, rank)
+    vg, layer_groups = groups_list[0], groups_list[1:]
+    embed_group = gen_embedding_group(world, pp, cache, rank)
+
+    lo, hi = layers_of_stage(division, my_stage)
+    blocks: List[LayerBlock] = []
+    prev: Optional[LayerCommGroups] = None
+    emb_block: Optional[LayerBlock] = None
+    head_block: Optional[LayerBlock] = None
+
+    torch.manual_seed(cfg.train.seed)  # deterministic per-rank module init
+    if my_stage == 0:
+        emb = GalvatronEmbedding(margs, vg, dtype=dtype)
+        emb_block = LayerBlock(emb, vg, "embedding")
+        blocks.append(emb_block)
+        prev = vg
+    else:
+        # incoming activation carries the PREVIOUS stage's last-layer layout
+        prev_strategy = strategies[lo - 1]
+        prev = gen_layer_comm_groups([prev_strategy], world, rank, cache)[0][0]
+
+    for i in range(lo, hi):
+        lg = layer_groups[i]
+        dec = GalvatronDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
+        blk = LayerBlock(dec, lg, "decoder",
+                         checkpoint=bool(plan.checkpoint_flags[i]),
+                         prev_groups=prev)
+        blocks.append(blk)
+        prev = lg
+
+    if my_stage == pp - 1:
+        fn = GalvatronFinalNorm(margs, vg, dtype=dtype)
+        blocks.append(LayerBlock(fn, vg, "final_norm", prev_groups=prev))
+        head = GalvatronCausalLMHead(margs, vg, dtype=dtype)
+        head_block = LayerBlock(head, vg, "lm_head", prev_groups=None)
+        if margs.tie_word_embeddings and not margs.untie_embeddings_and_output_weights:
+            if pp == 1 and emb_block is not None:
+                head.tie_to(emb_block.inner)
+        blocks.append(head_block)
+
+    for b in blocks:
+        b.inner.to(device)
+        mode = b.groups.strategy.dp_type
+        b.setup_zero(mode, dtype, device)
+
+    recv_layout = None
+    if my_stage > 0:
+        recv_layout = gen_layer_comm_groups([strategies[lo - 1]], world, rank,
+                                            cache)[0][0]
+    sm = StageModel(
+        blocks=blocks, stage=my_stage, pp_deg=pp, world_size=world, rank=rank,
+        plan=plan, embed_comm_group=embed_group, cache=cache,
+        recv_layout=recv_layout,
+        send_layout=layer_groups[hi - 1] if hi > lo else prev,
+        tied_embedding_pair=(emb_block, head_block)
+        if (emb_block is not None and head_block is not None) else None,
+    )
+    return sm

@@ -1,0 +1,136 @@
+"""Distributed optimizer over the ZeRO blocks' master partitions.
+
+Reference: galvatron/core/runtime/optimizer/utils.py:8-70 (apex FusedAdam ->
+here the HIP multi-tensor AdamW kernel ops/csrc/adamw.hip) and
+clip_grads.py:11-194 (grad-norm + clip; the MoE expert-grad rescale hooks in
+here once EP lands).
+
+Each FlatParamBlock owns a master fp32 partition + adam moments; the fused
+kernel updates master/m/v AND writes the bf16 model shard in one pass
+(saving one full HBM sweep per step on MI355X's 8 TB/s HBM3E).  Grad norm:
+every rank sums squares over elements it OWNS (each model element counted
+once world-wide), one scalar all-reduce, then a multi-tensor scale.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ...ops import reference_ops as ref
+from ...ops._ext import get_ext, native_available
+from ..zero import FlatParamBlock
+
+
+class GalvatronOptimizer:
+    def __init__(self, blocks: List[FlatParamBlock], lr: float = 1e-4,
+                 betas=(0.9, 0.95), eps: float = 1e-8,
+                 weight_decay: float = 0.01, clip_grad: float = 1.0,
+                 use_fused: bool = True):
+        self.blocks = [b for b in blocks if b is not None and b.total > 0]
+        self.lr = lr
+        self.betas = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.clip_grad = clip_grad
+        self.use_fused = use_fused
+        self.step_count = 0
+        self.last_grad_norm: Optional[float] = None
+
+    # -- grad norm + clip ---------------------------------------------------
+
+    def _global_grad_norm(self) -> float:
+        dev = self.blocks[0].device if self.blocks else torch.device("cpu")
+        total = torch.zeros((), dtype=torch.float32, device=dev)
+        for b in self.blocks:
+            total = total + b.grad_sumsq_owned()
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(total)
+        return float(total.sqrt().item())
+
+    def clip_gradients(self) -> float:
+        norm = self._global_grad_norm()
+        self.last_grad_norm = norm
+        if self.clip_grad and self.clip_grad > 0 and norm > self.clip_grad:
+            scale = self.clip_grad / (norm + 1e-6)
+            for b in self.blocks:
+                b.scale_grads(scale)
+        return norm
+
+    # -- step ---------------------------------------------------------------
+
+    def _block_tensors(self, b: FlatParamBlock):
+        master, grad = b.owned_master_and_grad()
+        if b.mode == "ddp":
+            bf16_out = b.flat_param
+        elif b.mode == "zero2":
+            bf16_out = b.flat_param[b._shard_slice()]
+        else:
+            bf16_out = b.param_shard
+        return master, grad, b.exp_avg, b.exp_avg_sq, bf16_out
+
+    def step(self) -> float:
+        norm = self.clip_gradients()
+        self.step_count += 1
+        use_native = (self.use_fused and native_available()
+                      and self.blocks and self.blocks[0].device.type == "cuda")
+        masters, grads, ms, vs, outs = [], [], [], [], []
+        for b in self.blocks:
+            t = self._block_tensors(b)
+            masters.append(t[0]); grads.append(t[1]); ms.append(t[2])
+            vs.append(t[3]); outs.append(t[4])
+        if use_native:
+            get_ext(False).fused_adamw(
+                masters, grads, ms, vs, outs, self.step_count, self.lr,
+                self.betas[0], self.betas[1], self.eps, self.weight_decay)
+        else:
+            ref.adamw_step(outs, grads, ms, vs, masters, self.step_count,
+                           self.lr, self.betas[0], self.betas[1], self.eps,
+                           self.weight_decay)
+        for b in self.blocks:
+            if b.mode == "zero2":
+                b.apply_master_to_params_post_step()
+        return norm
+
+    def zero_grad(self) -> None:
+        for b in self.blocks:
+            b.zero_grad()
+
+    # -- state --------------------------------------------------------------
+
+    def state_dict(self) -> dict:
+        return {
+            "step_count": self.step_count,
+            "blocks": [
+                {"master": b.master, "exp_avg": b.exp_avg,
+                 "exp_avg_sq": b.exp_avg_sq}
+                for b in self.blocks
+            ],
+        }
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.step_count = sd["step_count"]
+        for b, bs in zip(self.blocks, sd["blocks"]):
+            b.master.copy_(bs["master"].to(b.master.device))
+            b.exp_avg.copy_(bs["exp_avg"].to(b.master.device))
+            b.exp_avg_sq.copy_(bs["exp_avg_sq"].to(b.master.device))
+            b.apply_master_to_params()
+
+
+def get_optimizer_and_param_scheduler(stage_model, cfg):
+    """reference: optimizer/utils.py:44."""
+    from .scheduler import OptimizerParamScheduler
+
+    blocks = [blk.flat for blk in stage_model.blocks if blk.flat is not None]
+    t = cfg.train
+    opt = GalvatronOptimizer(
+        blocks, lr=t.lr, betas=(t.adam_beta1, t.adam_beta2), eps=t.adam_eps,
+        weight_decay=t.adam_weight_decay, clip_grad=t.clip_grad,
+        use_fused=t.use_fused_adam)
+    sched = OptimizerParamScheduler(
+        opt, max_lr=t.lr, min_lr=t.min_lr, warmup_steps=t.lr_warmup_iters,
+        decay_steps=t.lr_decay_iters or t.train_iters,
+        decay_style=t.lr_decay_style, wsd_decay_steps=t.lr_wsd_decay_iters)
+    return opt, sched

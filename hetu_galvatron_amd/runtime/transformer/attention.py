@@ -1,0 +1,99 @@
+"""Self-attention with per-layer TP / Megatron-SP / Ulysses / zigzag-CP.
+
+Reference: galvatron/core/runtime/transformer/attention.py:111-1017.
+Dispatch (attention.py:515-732 in the reference, CP-only gap fixed):
+    tp>1 (megatron)      : fused QKV CPL (SP-allgather inside) -> heads
+                           sharded; inner attention over full cp-local seq
+    ulysses sp>1         : heads full, seq sharded; DistributedAttention a2a
+    cp>1                 : inner attention = ZigzagRingAttention (composable
+                           with either of the above)
+GQA interleaved-group QKV layout: the fused weight is ordered by KV group
+[q*(hq/hkv), k, v] so ColumnParallelLinear's contiguous shard = whole groups
+(checkpoint adapters interleave accordingly; reference llama_adapter.py).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..tensor_parallel import (
+    ColumnParallelLinear, RowParallelLinear, divide, group_rank, group_size,
+)
+from .attention_impl import (
+    DistributedAttention, ZigzagRingAttention, local_attention,
+)
+from .rope import apply_rope_qk
+
+
+class SelfAttention(nn.Module):
+    def __init__(self, model_args, tp_group, sp_group, cp_group,
+                 use_ulysses: bool = False, sequence_parallel: bool = True,
+                 dtype=None):
+        super().__init__()
+        m = model_args
+        self.hidden_size = m.hidden_size
+        self.num_heads = m.num_attention_heads
+        self.num_kv_heads = m.kv_heads
+        self.head_dim = m.head_dim
+        self.use_ulysses = use_ulysses
+        self.tp_group = tp_group
+        self.sp_group = sp_group
+        self.cp_group = cp_group
+        tp = 1 if use_ulysses else group_size(tp_group)
+        self.tp = tp
+        assert self.num_heads % tp == 0, "tp must divide num heads"
+        assert self.num_kv_heads % tp == 0 or tp % self.num_kv_heads == 0, \
+            "tp must divide kv heads (kv replication unsupported yet)"
+        assert self.num_kv_heads % tp == 0, \
+            f"tp={tp} > kv_heads={self.num_kv_heads} needs KV replication"
+        self.num_groups_local = self.num_kv_heads // tp
+        self.q_per_group = self.num_heads // self.num_kv_heads
+        self.heads_local = self.num_heads // tp
+        qkv_out = (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
+        lin_group = tp_group if not use_ulysses else None
+        self.linear_qkv = ColumnParallelLinear(
+            self.hidden_size, qkv_out, lin_group,
+            bias=m.add_qkv_bias or m.add_bias_linear,
+            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
+        self.linear_proj = RowParallelLinear(
+            self.num_heads * self.head_dim, self.hidden_size, lin_group,
+            bias=m.add_bias_linear,
+            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
+        self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
+        cp = group_size(cp_group) if cp_group is not None else 1
+        inner = ZigzagRingAttention(cp_group) if cp > 1 else None
+        if use_ulysses and group_size(sp_group) > 1:
+            self.core_attention = DistributedAttention(
+                sp_group, inner_attention=inner)
+        elif inner is not None:
+            self.core_attention = inner
+        else:
+            self.core_attention = None  # plain local flash
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+                ) -> torch.Tensor:
+        """x: [s_local, b, h] SBH; cos/sin: tables matching the seq layout at
+        RoPE time (megatron-sp: cp-local full seq; ulysses: the local slice)."""
+        qkv = self.linear_qkv(x)  # [s, b, (hq_l + 2*hkv_l)*d]
+        s, b = qkv.shape[0], qkv.shape[1]
+        qkv = qkv.view(s, b, self.num_groups_local, self.q_per_group + 2,
+                       self.head_dim)
+        q = qkv[:, :, :, : self.q_per_group].reshape(s, b, -1, self.head_dim)
+        k = qkv[:, :, :, self.q_per_group].reshape(s, b, -1, self.head_dim)
+        v = qkv[:, :, :, self.q_per_group + 1].reshape(s, b, -1, self.head_dim)
+        q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin)
+        # [s,b,h,d] -> [b,s,h,d] for the flash kernels
+        q = q.permute(1, 0, 2, 3).contiguous()
+        k = k.permute(1, 0, 2, 3).contiguous()
+        v = v.permute(1, 0, 2, 3).contiguous()
+        if self.core_attention is not None:
+            o = self.core_attention(q, k, v, causal=True,
+                                    softmax_scale=self.softmax_scale)
+        else:
+            o = local_attention(q, k, v, causal=True,
+                                softmax_scale=self.softmax_scale)
+        o = o.permute(1, 0, 2, 3).reshape(s, b, -1)  # back to SBH
+        return self.linear_proj(o)

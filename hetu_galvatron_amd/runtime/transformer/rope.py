@@ -1,0 +1,87 @@
+"""Rotary position embeddings with CP/SP-aware frequency slicing.
+
+Reference: galvatron/core/runtime/transformer/rotary_pos_embedding.py:34-266
+and rope_utils.py:86-269.  cos/sin tables are HOST-precomputed (CDNA4
+elementwise guideline: no on-device trig in the hot path) and sliced per
+rank: Ulysses-SP ranks own a contiguous sequence slice; zigzag-CP ranks own
+chunks (i, 2cp-1-i) of 2cp chunks (rotary_pos_embedding.py:34
+get_pos_emb_on_this_cp_sp_rank_galvatron).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ...ops import apply_rope
+from ...ops.reference_ops import rope_freqs
+
+
+class RotaryEmbedding(torch.nn.Module):
+    """Caches cos/sin [seq, d/2] fp32 tables; returns per-rank slices."""
+
+    def __init__(self, head_dim: int, theta: float = 10000.0,
+                 scaling: Optional[float] = None, max_seq: int = 8192):
+        super().__init__()
+        self.head_dim = head_dim
+        self.theta = theta
+        self.scaling = scaling
+        self._cached: Tuple[int, Optional[torch.Tensor], Optional[torch.Tensor]] = (0, None, None)
+
+    def full_tables(self, seq_len: int, device) -> Tuple[torch.Tensor, torch.Tensor]:
+        cached_len, cos, sin = self._cached
+        if cos is None or cached_len < seq_len or cos.device != device:
+            cos, sin = rope_freqs(seq_len, self.head_dim, self.theta, device=device)
+            if self.scaling:
+                cos, sin = rope_freqs(int(seq_len), self.head_dim, self.theta,
+                                      device=device)
+            self._cached = (seq_len, cos, sin)
+        return cos[:seq_len], sin[:seq_len]
+
+    def get_for_rank(self, seq_len_global: int, device,
+                     sp_rank: int = 0, sp_size: int = 1,
+                     cp_rank: int = 0, cp_size: int = 1
+                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Slice of the tables covering this rank's local sequence shard.
+
+        Layout convention (shared with the dataloader's get_batch slicing):
+        the sequence is FIRST zigzag-split over cp ranks (rank's pair of
+        chunks), THEN that cp-local sequence is split into sp_size contiguous
+        slices (Ulysses / Megatron-SP) — so a Ulysses all-gather reassembles
+        exactly the cp rank's zigzag pair.
+        """
+        cos, sin = self.full_tables(seq_len_global, device)
+        if cp_size > 1:
+            cos = zigzag_slice(cos, cp_rank, cp_size)
+            sin = zigzag_slice(sin, cp_rank, cp_size)
+        if sp_size > 1:
+            sl = cos.shape[0] // sp_size
+            cos = cos[sp_rank * sl:(sp_rank + 1) * sl]
+            sin = sin[sp_rank * sl:(sp_rank + 1) * sl]
+        return cos, sin
+
+
+def zigzag_slice(x: torch.Tensor, cp_rank: int, cp_size: int) -> torch.Tensor:
+    """Take chunks (cp_rank, 2cp-1-cp_rank) of 2cp equal chunks along dim 0
+    (reference: redistribute.py:5-24 _zigzag_transformation)."""
+    chunks = x.chunk(2 * cp_size, dim=0)
+    return torch.cat([chunks[cp_rank], chunks[2 * cp_size - 1 - cp_rank]], dim=0)
+
+
+def zigzag_unslice_index(cp_size: int) -> list:
+    """Chunk order that reassembles a zigzag-sharded sequence: for gathered
+    [rank0(lo,hi), rank1(lo,hi), ...] -> natural chunk order."""
+    order = []
+    pos = {}
+    for r in range(cp_size):
+        pos[r] = 2 * r          # rank r's low chunk position in gathered list
+        pos[2 * cp_size - 1 - r] = 2 * r + 1
+    for c in range(2 * cp_size):
+        order.append(pos[c])
+    return order
+
+
+def apply_rope_qk(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
+                  sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """q,k: [s, b, h, d] SBH-heads layout."""
+    return apply_rope(q, cos, sin), apply_rope(k, cos, sin)

@@ -1,0 +1,378 @@
+"""Flat-parameter data-parallel engine: DDP / ZeRO-2 / ZeRO-3 per layer block.
+
+Replaces the reference's per-layer FSDP wrapping + monkey-patched internals
+(reference: core/runtime/parallel.py:100-411, pipeline/grad_reduce.py:48-235,
+pipeline/sp_grad_reduce.py:48-132) with an explicit engine designed for the
+per-layer-plan runtime (SURVEY hard part #2 recommends exactly this):
+
+  * each wrapped block owns ONE flat bf16 parameter buffer (module params are
+    views) + ONE persistent fp32 grad buffer; post-accumulate-grad hooks fold
+    param.grad into the fp32 buffer every microbatch (bf16-safe accumulation,
+    no FSDP hook patching).
+  * modes over the block's sdp group (dp x cp [x sp]):
+      ddp   - params + optimizer states replicated; grads all-reduced once
+      zero2 - params replicated; grads reduce-scattered; fp32 master +
+              optimizer states sharded; params all-gathered after step
+      zero3 - flat param sharded; storage-resize allgather before fwd/bwd,
+              freed after; grads reduce-scattered; master sharded
+  * grad sync is EXPLICIT (engine calls start/finish after the last
+    microbatch) and async -> overlaps with remaining backward compute on
+    RCCL's stream (the cost model's bct_dp_overlap term).
+  * params tagged `tp_replicated` (norms, RPL biases under Megatron-SP) get an
+    extra all-reduce over the layer's TP group (replacing the reference's
+    patched-FSDP SP-layernorm hook).
+  * reductions are SUMS: the loss is normalized by the GLOBAL token count, so
+    summing partial grads over every replication domain reproduces the
+    single-GPU gradient exactly (tested against a 1-process baseline).
+
+MI355X sizing: 288 GB HBM3E lets an 8B model keep fp32 master+moments
+resident at dp=1; the flat layout means reduce-scatter/all-gather are a few
+LARGE xGMI transfers instead of per-param traffic — xGMI rings are per-link
+bound, so fewer+bigger collectives is the right shape.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..core.comm_groups import CommGroup
+from .tensor_parallel.mappings import _is_gloo
+
+ALIGN = 128  # elements; xGMI/RCCL friendly alignment
+
+
+@dataclass
+class ParamSegment:
+    param: nn.Parameter
+    offset: int
+    numel: int
+    shape: torch.Size
+    tp_replicated: bool  # replicated over the layer's tp group (norms under SP)
+
+
+def _pad_to(n: int, m: int) -> int:
+    return (n + m - 1) // m * m
+
+
+class FlatParamBlock:
+    """One module's flat parameter/grad/optimizer-state management."""
+
+    def __init__(self, module: nn.Module, mode: str, sdp_group: CommGroup,
+                 tp_group: Optional[CommGroup] = None,
+                 param_dtype: torch.dtype = torch.bfloat16,
+                 device: Optional[torch.device] = None,
+                 owner_filter: bool = True):
+        assert mode in ("ddp", "zero2", "zero3")
+        self.module = module
+        self.mode = mode
+        self.sdp_group = sdp_group
+        self.tp_group = tp_group
+        self.ws = sdp_group.size
+        self.rank_in_group = sdp_group.index(
+            dist.get_rank()) if dist.is_initialized() and sdp_group.size > 1 else 0
+        if sdp_group.size > 1 and not dist.is_initialized():
+            self.rank_in_group = 0
+        self.param_dtype = param_dtype
+        self.segments: List[ParamSegment] = []
+        self._handles: List = []
+        self._gathered = False
+        self.auto_sync = False  # post-backward hook kicks async reduce
+        self._synced = False
+
+        params = []
+        for p in module.parameters():
+            if not p.requires_grad:
+                continue
+            if owner_filter and getattr(p, "_galvatron_owner", None) is not None:
+                continue  # tied param owned by another block
+            p._galvatron_owner = self
+            params.append(p)
+        if device is None:
+            device = params[0].device if params else torch.device("cpu")
+        self.device = device
+
+        off = 0
+        for p in params:
+            n = p.numel()
+            self.segments.append(ParamSegment(
+                p, off, n, p.shape,
+                tp_replicated=bool(getattr(p, "tp_replicated", False))))
+            off = _pad_to(off + n, ALIGN)
+        self.total = _pad_to(max(off, ALIGN), ALIGN * max(self.ws, 1))
+        self.shard_size = self.total // max(self.ws, 1)
+
+        # flat bf16 parameter buffer (full size; zero3 frees it between uses)
+        self.flat_param = torch.zeros(self.total, dtype=param_dtype, device=device)
+        with torch.no_grad():
+            for seg in self.segments:
+                self.flat_param[seg.offset:seg.offset + seg.numel].copy_(
+                    seg.param.data.flatten().to(param_dtype))
+        # fp32 master copy: full (ddp) or local shard (zero2/3)
+        if mode == "ddp":
+            self.master = self.flat_param.float()
+        else:
+            self.master = self.flat_param[self._shard_slice()].float()
+        if mode == "zero3":
+            # persistent bf16 shard; full flat storage freed when idle
+            self.param_shard = self.flat_param[self._shard_slice()].clone()
+        else:
+            self.param_shard = None
+        self._point_params_to_flat()
+        if mode == "zero3":
+            self._free_full()
+
+        # persistent fp32 grad accumulator (lazily allocated on first backward)
+        self.flat_grad: Optional[torch.Tensor] = None
+        self.grad_shard: Optional[torch.Tensor] = None  # zero2/3 post-reduce
+        for seg in self.segments:
+            seg.param.register_post_accumulate_grad_hook(self._make_acc_hook(seg))
+
+        # adam states on the master partition
+        self.exp_avg = torch.zeros_like(self.master)
+        self.exp_avg_sq = torch.zeros_like(self.master)
+        self.step_count = 0
+
+    # ------------------------------------------------------------------ utils
+
+    def _shard_slice(self) -> slice:
+        return slice(self.rank_in_group * self.shard_size,
+                     (self.rank_in_group + 1) * self.shard_size)
+
+    def _point_params_to_flat(self) -> None:
+        for seg in self.segments:
+            seg.param.data = self.flat_param[seg.offset:seg.offset + seg.numel] \
+                .view(seg.shape)
+
+    def _free_full(self) -> None:
+        self.flat_param.untyped_storage().resize_(0)
+        self._gathered = False
+
+    def _alloc_full(self) -> None:
+        self.flat_param.untyped_storage().resize_(
+            self.total * self.flat_param.element_size())
+
+    # ------------------------------------------------------------ zero3 moves
+
+    def gather_params(self) -> None:
+        """zero3: materialize the full flat param via all-gather of shards."""
+        if self.mode != "zero3" or self._gathered:
+            return
+        self._alloc_full()
+        with torch.no_grad():
+            if self.ws == 1:
+                self.flat_param.copy_(self.param_shard)
+            elif _is_gloo(self.sdp_group.group):
+                chunks = [torch.empty_like(self.param_shard) for _ in range(self.ws)]
+                dist.all_gather(chunks, self.param_shard, group=self.sdp_group.group)
+                self.flat_param.copy_(torch.cat(chunks))
+            else:
+                dist.all_gather_into_tensor(self.flat_param, self.param_shard,
+                                            group=self.sdp_group.group)
+        self._gathered = True
+
+    def reshard_params(self) -> None:
+        if self.mode == "zero3" and self._gathered:
+            self._free_full()
+
+    # -------------------------------------------------------------- gradients
+
+    def _ensure_grad_buffer(self) -> None:
+        if self.flat_grad is None:
+            self.flat_grad = torch.zeros(self.total, dtype=torch.float32,
+                                         device=self.device)
+
+    def _make_acc_hook(self, seg: ParamSegment):
+        def hook(param: nn.Parameter) -> None:
+            if param.grad is None:
+                return
+            self._ensure_grad_buffer()
+            self.flat_grad[seg.offset:seg.offset + seg.numel].add_(
+                param.grad.flatten().float())
+            param.grad = None
+        return hook
+
+    def post_backward(self) -> None:
+        """Called by the block wrapper when the block's input grad is ready
+        (all its param grads have accumulated)."""
+        self.reshard_params()
+        if self.auto_sync and not self._synced:
+            self.start_grad_sync()
+
+    def start_grad_sync(self) -> None:
+        """Async reduction of accumulated grads over the sdp group (+ tp
+        all-reduce for replicated segments)."""
+        if self._synced:
+            return
+        self._synced = True
+        self._ensure_grad_buffer()
+        g = self.flat_grad
+        # tp-replicated segments (SP norms): sum over the tp group first
+        if self.tp_group is not None and self.tp_group.size > 1:
+            for seg in self.segments:
+                if seg.tp_replicated:
+                    sl = g[seg.offset:seg.offset + seg.numel]
+                    if _is_gloo(self.tp_group.group):
+                        dist.all_reduce(sl, group=self.tp_group.group)
+                    else:
+                        self._handles.append(dist.all_reduce(
+                            sl, group=self.tp_group.group, async_op=True))
+        if self.ws == 1:
+            self.grad_shard = g if self.mode == "ddp" else g[self._shard_slice()]
+            return
+        if self.mode == "ddp":
+            if _is_gloo(self.sdp_group.group):
+                dist.all_reduce(g, group=self.sdp_group.group)
+            else:
+                self._handles.append(dist.all_reduce(
+                    g, group=self.sdp_group.group, async_op=True))
+            self.grad_shard = g
+        else:
+            self.grad_shard = torch.empty(self.shard_size, dtype=torch.float32,
+                                          device=self.device)
+            if _is_gloo(self.sdp_group.group):
+                dist.all_reduce(g, group=self.sdp_group.group)
+                self.grad_shard.copy_(g[self._shard_slice()])
+            else:
+                self._handles.append(dist.reduce_scatter_tensor(
+                    self.grad_shard, g, group=self.sdp_group.group,
+                    async_op=True))
+
+    def finish_grad_sync(self) -> None:
+        for h in self._handles:
+            h.wait()
+        self._handles = []
+
+    def zero_grad(self) -> None:
+        if self.flat_grad is not None:
+            self.flat_grad.zero_()
+        self.grad_shard = None
+        self._synced = False
+
+    # -------------------------------------------------------------- optimizer
+
+    def owned_master_and_grad(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """(fp32 master partition, matching fp32 reduced grad)."""
+        assert self.grad_shard is not None, "start/finish_grad_sync first"
+        if self.mode == "ddp":
+            return self.master, self.grad_shard
+        return self.master, self.grad_shard
+
+    def grad_sumsq_owned(self) -> torch.Tensor:
+        """Sum of squares of grads over elements OWNED by this rank (each
+        model element counted exactly once across the world):
+          * zero2/3: the reduced shard; ddp: the rank's virtual shard
+          * tp-replicated segments scaled by 1/tp (they repeat per tp rank)
+        """
+        assert self.grad_shard is not None
+        if self.mode == "ddp":
+            sl = self._shard_slice()
+            g = self.grad_shard[sl]
+            base = sl.start
+        else:
+            g = self.grad_shard
+            base = self.rank_in_group * self.shard_size
+        total = (g.float() ** 2).sum()
+        tp = self.tp_group.size if self.tp_group is not None else 1
+        if tp > 1:
+            for seg in self.segments:
+                if not seg.tp_replicated:
+                    continue
+                lo = max(seg.offset, base)
+                hi = min(seg.offset + seg.numel, base + g.numel())
+                if lo < hi:
+                    part = (g[lo - base:hi - base].float() ** 2).sum()
+                    total = total - part + part / tp
+        return total
+
+    def scale_grads(self, scale: float) -> None:
+        if self.grad_shard is not None:
+            self.grad_shard.mul_(scale)
+        if self.mode == "ddp" and self.flat_grad is not None \
+                and self.grad_shard is not self.flat_grad:
+            self.flat_grad.mul_(scale)
+
+    def refresh_from_params(self) -> None:
+        """Re-sync master/shard copies after params were overwritten in place
+        (checkpoint load).  For zero3, call with params gathered; reshards."""
+        with torch.no_grad():
+            if self.mode == "ddp":
+                self.master.copy_(self.flat_param.float())
+            elif self.mode == "zero2":
+                self.master.copy_(self.flat_param[self._shard_slice()].float())
+            else:
+                assert self._gathered, "gather_params() before refresh (zero3)"
+                self.param_shard.copy_(self.flat_param[self._shard_slice()])
+                self.master.copy_(self.param_shard.float())
+                self.reshard_params()
+
+    def apply_master_to_params_post_step(self) -> None:
+        """zero2 only: the fused adam kernel already wrote this rank's bf16
+        shard; all-gather the updated flat param across the sdp group."""
+        if self.mode != "zero2" or self.ws == 1:
+            return
+        with torch.no_grad():
+            shard = self.flat_param[self._shard_slice()].contiguous()
+            if _is_gloo(self.sdp_group.group):
+                chunks = [torch.empty_like(shard) for _ in range(self.ws)]
+                dist.all_gather(chunks, shard, group=self.sdp_group.group)
+                self.flat_param.copy_(torch.cat(chunks))
+            else:
+                dist.all_gather_into_tensor(self.flat_param, shard,
+                                            group=self.sdp_group.group)
+
+    def apply_master_to_params(self) -> None:
+        """Copy updated master back into bf16 params (+ allgather for zero2,
+        shard update for zero3)."""
+        with torch.no_grad():
+            if self.mode == "ddp":
+                self.flat_param.copy_(self.master.to(self.param_dtype))
+            elif self.mode == "zero2":
+                self.flat_param[self._shard_slice()].copy_(
+                    self.master.to(self.param_dtype))
+                if self.ws > 1:
+                    if _is_gloo(self.sdp_group.group):
+                        shard = self.flat_param[self._shard_slice()].contiguous()
+                        chunks = [torch.empty_like(shard) for _ in range(self.ws)]
+                        dist.all_gather(chunks, shard, group=self.sdp_group.group)
+                        self.flat_param.copy_(torch.cat(chunks))
+                    else:
+                        shard = self.flat_param[self._shard_slice()].contiguous()
+                        dist.all_gather_into_tensor(self.flat_param, shard,
+                                                    group=self.sdp_group.group)
+            else:  # zero3
+                self.param_shard.copy_(self.master.to(self.param_dtype))
+
+
+class _PreBackwardGather(torch.autograd.Function):
+    """Output sentinel: backward fires BEFORE the block's internal backward
+    -> re-gather zero3 params (replaces FSDP pre-backward unshard)."""
+
+    @staticmethod
+    def forward(ctx, block, x):
+        ctx.block = block
+        return x
+
+    @staticmethod
+    def backward(ctx, g):
+        ctx.block.gather_params()
+        return None, g
+
+
+class _PostBackwardHook(torch.autograd.Function):
+    """Input sentinel: backward fires AFTER the block's internal backward
+    -> reshard + optionally kick async grad reduction (overlap)."""
+
+    @staticmethod
+    def forward(ctx, block, x):
+        ctx.block = block
+        return x
+
+    @staticmethod
+    def backward(ctx, g):
+        ctx.block.post_backward()
+        return None, g
