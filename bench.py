@@ -1,0 +1,146 @@
+#!/usr/bin/env python3
+"""Flagship training-step benchmark (driver contract).
+
+`python bench.py --gpus N --steps K --warmup W` runs a Llama-3-8B training
+step (synthetic data, random-init weights, bf16) on N GPUs of one node.
+For N>1 the driver launches it under torch.distributed.run, one rank per
+GPU over RCCL; ranks read RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.
+
+Rank 0 prints exactly ONE JSON line with the whole-job tokens/sec
+(BASELINE.json metric: "tokens/sec (whole node), auto-searched plan,
+Llama-3-8B on 1/2/4/8 MI355X"). Weak scaling: per-GPU work fixed as N
+grows (global_batch = per-gpu-batch * N).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--model", type=str, default="llama-3-8b")
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--batch-per-gpu", type=int, default=8,
+                    help="sequences per GPU per step (weak scaling)")
+    ap.add_argument("--chunks", type=int, default=0,
+                    help="pipeline microbatches (0 = batch-per-gpu)")
+    ap.add_argument("--pp", type=int, default=1)
+    ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--dp-type", type=str, default="zero2",
+                    choices=["ddp", "zero2", "zero3"])
+    ap.add_argument("--checkpoint", action="store_true")
+    ap.add_argument("--plan", type=str, default="",
+                    help="searched-plan JSON path (overrides uniform knobs)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world, args.gpus) if world > 1 else args.gpus
+
+    use_gpu = torch.cuda.is_available()
+    if not use_gpu:
+        print("bench.py requires a GPU", file=sys.stderr)
+        sys.exit(1)
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import (
+        GalvatronModel, get_optimizer_and_param_scheduler, get_train_iterator)
+    from hetu_galvatron_amd.runtime.galvatron_model import resolve_plan
+
+    global_batch = args.batch_per_gpu * max(world, 1)
+    chunks = args.chunks or args.batch_per_gpu
+    cfg = load_config(base={
+        "model": {"model_name": args.model, "seq_length": args.seq_len},
+        "train": {"global_train_batch_size": global_batch,
+                  "train_iters": args.steps + args.warmup,
+                  "lr": 1e-4, "lr_decay_style": "constant",
+                  "distributed_backend": "nccl"},
+        "parallel": {"pp_deg": args.pp, "global_tp_deg": args.tp,
+                     "default_dp_type": args.dp_type,
+                     "global_checkpoint": 1 if args.checkpoint else 0,
+                     "chunks": chunks, "mixed_precision": "bf16",
+                     "galvatron_config_path": args.plan or None},
+    })
+
+    if world > 1:
+        initialize_galvatron(cfg, backend="nccl")
+    else:
+        torch.manual_seed(cfg.train.seed)
+
+    plan = resolve_plan(cfg, world)
+    model = GalvatronModel(cfg, plan, device=device)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, device)
+
+    def one_step():
+        opt.zero_grad()
+        ctx = next(it)
+        stats = model.forward_backward(ctx)
+        opt.step()
+        sched.step()
+        return stats
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks = whole-job wall time
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tokens = args.steps * global_batch * args.seq_len
+    value = tokens / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens/sec (whole node), auto-searched plan, "
+                      "Llama-3-8B on 1/2/4/8 MI355X",
+            "value": value,
+            "unit": "tokens/s",
+            "n_gpus": max(world, 1),
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.seq_len,
+                "parallelism": (f"pp{plan.pp_deg}-tp{args.tp}-"
+                                f"dp{max(world, 1) // (args.pp * args.tp)}-"
+                                f"{args.dp_type}"),
+            },
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,383 @@
+// Torch bindings for the gfx950 kernel suite (module: _galvatron_hip).
+//
+// HIP-native throughout (c10::hip stream API, no CUDA-compat shims).
+// Python-side dispatch contract: hetu_galvatron_amd/ops/functional.py,
+// runtime/tensor_parallel/cross_entropy.py, runtime/optimizer/optimizer.py.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPGuard.h>
+
+#include <hip/hip_runtime.h>
+
+#include <tuple>
+#include <vector>
+
+// ---- launcher prototypes (defined in the sibling .hip TUs) ----------------
+template <typename T>
+void rmsnorm_fwd_launch_t(const T*, const T*, T*, float*, long, int, float, hipStream_t);
+template <typename T>
+void rmsnorm_bwd_launch_t(const T*, const T*, const T*, const float*, T*, float*, long, int, hipStream_t);
+template <typename T>
+void layernorm_fwd_launch_t(const T*, const T*, const T*, T*, float*, float*, long, int, float, hipStream_t);
+template <typename T>
+void layernorm_bwd_launch_t(const T*, const T*, const T*, const float*, const float*, T*, float*, float*, long, int, hipStream_t);
+template <typename T>
+void swiglu_fwd_launch_t(const T*, T*, long, int, hipStream_t);
+template <typename T>
+void swiglu_bwd_launch_t(const T*, const T*, T*, long, int, hipStream_t);
+template <typename T>
+void rope_launch_t(const T*, T*, const float*, const float*, long, int, int, bool, hipStream_t);
+template <typename TG, typename TO>
+void adamw_launch_t(float*, const TG*, float*, float*, TO*, long, int, float, float, float, float, float, hipStream_t);
+template <typename T>
+void ce_max_launch_t(const T*, float*, long, long, hipStream_t);
+template <typename T>
+void ce_sum_target_launch_t(const T*, const long*, const float*, float*, float*, long, long, long, hipStream_t);
+template <typename T>
+void ce_bwd_launch_t(T*, const long*, const float*, const float*, const float*, long, long, long, hipStream_t);
+
+void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
+void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
+void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
+void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
+
+namespace {
+
+using at::Tensor;
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+#define CHECK_GPU(x) \
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
+
+const __bf16* bfp(const Tensor& t) {
+  return reinterpret_cast<const __bf16*>(t.data_ptr<at::BFloat16>());
+}
+__bf16* bfp_mut(Tensor& t) {
+  return reinterpret_cast<__bf16*>(t.data_ptr<at::BFloat16>());
+}
+
+bool is_bf16(const Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+// ---- norms ----------------------------------------------------------------
+std::tuple<Tensor, Tensor> rmsnorm_fwd(const Tensor& x, const Tensor& w,
+                                       double eps) {
+  CHECK_GPU(x);
+  const int H = x.size(-1);
+  const long n = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0 && H <= 16384, "rmsnorm: H must be %8, <=16k");
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({n}, x.options().dtype(at::kFloat));
+  if (is_bf16(x))
+    rmsnorm_fwd_launch_t<__bf16>(bfp(x), bfp(w), bfp_mut(y),
+                                 invrms.data_ptr<float>(), n, H, (float)eps,
+                                 cur_stream());
+  else
+    rmsnorm_fwd_launch_t<float>(x.data_ptr<float>(), w.data_ptr<float>(),
+                                y.data_ptr<float>(), invrms.data_ptr<float>(),
+                                n, H, (float)eps, cur_stream());
+  auto sizes = x.sizes().vec();
+  sizes.pop_back();
+  return {y, invrms.view(sizes)};
+}
+
+std::tuple<Tensor, Tensor> rmsnorm_bwd(const Tensor& dy, const Tensor& x,
+                                       const Tensor& w, const Tensor& invrms) {
+  CHECK_GPU(dy);
+  const int H = x.size(-1);
+  const long n = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto inv = invrms.contiguous();
+  if (is_bf16(x))
+    rmsnorm_bwd_launch_t<__bf16>(bfp(dy), bfp(x), bfp(w),
+                                 inv.data_ptr<float>(), bfp_mut(dx),
+                                 dw.data_ptr<float>(), n, H, cur_stream());
+  else
+    rmsnorm_bwd_launch_t<float>(dy.data_ptr<float>(), x.data_ptr<float>(),
+                                w.data_ptr<float>(), inv.data_ptr<float>(),
+                                dx.data_ptr<float>(), dw.data_ptr<float>(), n,
+                                H, cur_stream());
+  return {dx, dw};
+}
+
+std::tuple<Tensor, Tensor, Tensor> layernorm_fwd(const Tensor& x,
+                                                 const Tensor& w,
+                                                 const Tensor& b, double eps) {
+  CHECK_GPU(x);
+  const int H = x.size(-1);
+  const long n = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0 && H <= 16384, "layernorm: H must be %8, <=16k");
+  auto y = at::empty_like(x);
+  auto mean = at::empty({n}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty({n}, x.options().dtype(at::kFloat));
+  if (is_bf16(x))
+    layernorm_fwd_launch_t<__bf16>(bfp(x), bfp(w), bfp(b), bfp_mut(y),
+                                   mean.data_ptr<float>(),
+                                   invstd.data_ptr<float>(), n, H, (float)eps,
+                                   cur_stream());
+  else
+    layernorm_fwd_launch_t<float>(x.data_ptr<float>(), w.data_ptr<float>(),
+                                  b.data_ptr<float>(), y.data_ptr<float>(),
+                                  mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                                  n, H, (float)eps, cur_stream());
+  auto sizes = x.sizes().vec();
+  sizes.pop_back();
+  return {y, mean.view(sizes), invstd.view(sizes)};
+}
+
+std::tuple<Tensor, Tensor, Tensor> layernorm_bwd(const Tensor& dy,
+                                                 const Tensor& x,
+                                                 const Tensor& w,
+                                                 const Tensor& mean,
+                                                 const Tensor& invstd) {
+  CHECK_GPU(dy);
+  const int H = x.size(-1);
+  const long n = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto m = mean.contiguous(), r = invstd.contiguous();
+  if (is_bf16(x))
+    layernorm_bwd_launch_t<__bf16>(bfp(dy), bfp(x), bfp(w),
+                                   m.data_ptr<float>(), r.data_ptr<float>(),
+                                   bfp_mut(dx), dw.data_ptr<float>(),
+                                   db.data_ptr<float>(), n, H, cur_stream());
+  else
+    layernorm_bwd_launch_t<float>(dy.data_ptr<float>(), x.data_ptr<float>(),
+                                  w.data_ptr<float>(), m.data_ptr<float>(),
+                                  r.data_ptr<float>(), dx.data_ptr<float>(),
+                                  dw.data_ptr<float>(), db.data_ptr<float>(),
+                                  n, H, cur_stream());
+  return {dx, dw, db};
+}
+
+// ---- swiglu / rope --------------------------------------------------------
+Tensor swiglu_fwd(const Tensor& x) {
+  CHECK_GPU(x);
+  const int F2 = x.size(-1);
+  TORCH_CHECK(F2 % 16 == 0, "swiglu: last dim must be %16");
+  const int F = F2 / 2;
+  const long rows = x.numel() / F2;
+  auto sizes = x.sizes().vec();
+  sizes.back() = F;
+  auto y = at::empty(sizes, x.options());
+  if (is_bf16(x))
+    swiglu_fwd_launch_t<__bf16>(bfp(x), bfp_mut(y), rows, F, cur_stream());
+  else
+    swiglu_fwd_launch_t<float>(x.data_ptr<float>(), y.data_ptr<float>(),
+                               rows, F, cur_stream());
+  return y;
+}
+
+Tensor swiglu_bwd(const Tensor& dy, const Tensor& x) {
+  CHECK_GPU(dy);
+  const int F2 = x.size(-1);
+  const int F = F2 / 2;
+  const long rows = x.numel() / F2;
+  auto dx = at::empty_like(x);
+  if (is_bf16(x))
+    swiglu_bwd_launch_t<__bf16>(bfp(dy), bfp(x), bfp_mut(dx), rows, F,
+                                cur_stream());
+  else
+    swiglu_bwd_launch_t<float>(dy.data_ptr<float>(), x.data_ptr<float>(),
+                               dx.data_ptr<float>(), rows, F, cur_stream());
+  return dx;
+}
+
+Tensor rope_fwd(const Tensor& x, const Tensor& cos_t, const Tensor& sin_t,
+                bool conj) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.dim() == 4, "rope: x must be [s,b,h,d]");
+  const int d = x.size(3);
+  TORCH_CHECK(d % 16 == 0, "rope: head dim must be %16");
+  const long rows = x.numel() / d;
+  const int bh = x.size(1) * x.size(2);
+  auto c = cos_t.contiguous().to(at::kFloat);
+  auto s = sin_t.contiguous().to(at::kFloat);
+  auto y = at::empty_like(x);
+  if (is_bf16(x))
+    rope_launch_t<__bf16>(bfp(x), bfp_mut(y), c.data_ptr<float>(),
+                          s.data_ptr<float>(), rows, bh, d, conj,
+                          cur_stream());
+  else
+    rope_launch_t<float>(x.data_ptr<float>(), y.data_ptr<float>(),
+                         c.data_ptr<float>(), s.data_ptr<float>(),
+                         rows, bh, d, conj, cur_stream());
+  return y;
+}
+
+// ---- flash attention ------------------------------------------------------
+std::tuple<Tensor, Tensor> flash_attn_fwd(const Tensor& q, const Tensor& k,
+                                          const Tensor& v, bool causal,
+                                          double scale) {
+  CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v);
+  TORCH_CHECK(is_bf16(q), "flash_attn: bf16 only on the native path");
+  TORCH_CHECK(q.dim() == 4, "flash_attn: q must be [b,s,h,d]");
+  const int b = q.size(0), sq = q.size(1), hq = q.size(2), d = q.size(3);
+  const int skv = k.size(1), hkv = k.size(2);
+  TORCH_CHECK(d == 64 || d == 128, "flash_attn: head dim must be 64|128");
+  TORCH_CHECK(hq % hkv == 0, "flash_attn: GQA needs hq % hkv == 0");
+  auto o = at::empty_like(q);
+  auto lse = at::empty({b, hq, sq}, q.options().dtype(at::kFloat));
+  flash_fwd_launch(bfp(q), bfp(k), bfp(v), bfp_mut(o),
+                   lse.data_ptr<float>(), b, sq, skv, hq, hkv, d,
+                   (float)scale, causal, cur_stream());
+  return {o, lse};
+}
+
+std::tuple<Tensor, Tensor, Tensor> flash_attn_bwd(const Tensor& dout,
+                                                  const Tensor& q,
+                                                  const Tensor& k,
+                                                  const Tensor& v,
+                                                  const Tensor& o,
+                                                  const Tensor& lse,
+                                                  bool causal, double scale) {
+  CHECK_GPU(dout); CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v); CHECK_GPU(o);
+  const int b = q.size(0), sq = q.size(1), hq = q.size(2), d = q.size(3);
+  const int skv = k.size(1), hkv = k.size(2);
+  auto lsec = lse.contiguous();
+  auto di = at::empty({b, hq, sq}, q.options().dtype(at::kFloat));
+  attn_di_launch(bfp(dout), bfp(o), di.data_ptr<float>(), b, sq, hq, d,
+                 cur_stream());
+  auto dq = at::empty_like(q);
+  auto dk_exp = at::empty({b, skv, hq, d}, k.options());
+  auto dv_exp = at::empty({b, skv, hq, d}, v.options());
+  flash_bwd_launch(bfp(dout), bfp(q), bfp(k), bfp(v),
+                   lsec.data_ptr<float>(), di.data_ptr<float>(),
+                   bfp_mut(dq), bfp_mut(dk_exp), bfp_mut(dv_exp), b, sq, skv,
+                   hq, hkv, d, (float)scale, causal, cur_stream());
+  Tensor dk = dk_exp, dv = dv_exp;
+  if (hq != hkv) {
+    const int rep = hq / hkv;
+    dk = dk_exp.view({b, skv, hkv, rep, d}).sum(3);
+    dv = dv_exp.view({b, skv, hkv, rep, d}).sum(3);
+  }
+  return {dq, dk, dv};
+}
+
+Tensor mfma_probe(const Tensor& A, const Tensor& B, bool alt) {
+  CHECK_GPU(A); CHECK_GPU(B);
+  auto D = at::zeros({32, 32}, A.options().dtype(at::kFloat));
+  mfma_probe_launch(bfp(A), bfp(B), D.data_ptr<float>(), alt, cur_stream());
+  return D;
+}
+
+// ---- vocab-parallel CE ----------------------------------------------------
+Tensor ce_max(const Tensor& logits) {
+  CHECK_GPU(logits);
+  const long V = logits.size(-1);
+  const long n = logits.numel() / V;
+  auto out = at::empty({n}, logits.options().dtype(at::kFloat));
+  if (is_bf16(logits))
+    ce_max_launch_t<__bf16>(bfp(logits), out.data_ptr<float>(), n, V,
+                            cur_stream());
+  else
+    ce_max_launch_t<float>(logits.data_ptr<float>(),
+                           out.data_ptr<float>(), n, V, cur_stream());
+  return out;
+}
+
+std::tuple<Tensor, Tensor> ce_sum_target(const Tensor& logits,
+                                         const Tensor& target,
+                                         const Tensor& gmax,
+                                         long vocab_start) {
+  CHECK_GPU(logits);
+  const long V = logits.size(-1);
+  const long n = logits.numel() / V;
+  auto t = target.contiguous();
+  auto g = gmax.contiguous();
+  auto sumexp = at::empty({n}, logits.options().dtype(at::kFloat));
+  auto tlogit = at::empty({n}, logits.options().dtype(at::kFloat));
+  if (is_bf16(logits))
+    ce_sum_target_launch_t<__bf16>(bfp(logits), t.data_ptr<long>(),
+                                   g.data_ptr<float>(),
+                                   sumexp.data_ptr<float>(),
+                                   tlogit.data_ptr<float>(), n, V,
+                                   vocab_start, cur_stream());
+  else
+    ce_sum_target_launch_t<float>(logits.data_ptr<float>(),
+                                  t.data_ptr<long>(), g.data_ptr<float>(),
+                                  sumexp.data_ptr<float>(),
+                                  tlogit.data_ptr<float>(), n, V, vocab_start,
+                                  cur_stream());
+  return {sumexp, tlogit};
+}
+
+Tensor ce_bwd(Tensor logits, const Tensor& target, const Tensor& gmax,
+              const Tensor& sumexp, const Tensor& grad_out, long vocab_start) {
+  CHECK_GPU(logits);
+  const long V = logits.size(-1);
+  const long n = logits.numel() / V;
+  auto t = target.contiguous();
+  auto g = gmax.contiguous();
+  auto se = sumexp.contiguous();
+  auto go = grad_out.contiguous().to(at::kFloat);
+  if (is_bf16(logits))
+    ce_bwd_launch_t<__bf16>(bfp_mut(logits), t.data_ptr<long>(),
+                            g.data_ptr<float>(), se.data_ptr<float>(),
+                            go.data_ptr<float>(), n, V, vocab_start,
+                            cur_stream());
+  else
+    ce_bwd_launch_t<float>(logits.data_ptr<float>(), t.data_ptr<long>(),
+                           g.data_ptr<float>(), se.data_ptr<float>(),
+                           go.data_ptr<float>(), n, V, vocab_start,
+                           cur_stream());
+  return logits;
+}
+
+// ---- fused AdamW ----------------------------------------------------------
+void fused_adamw(std::vector<Tensor> masters, std::vector<Tensor> grads,
+                 std::vector<Tensor> ms, std::vector<Tensor> vs,
+                 std::vector<Tensor> outs, long step, double lr, double beta1,
+                 double beta2, double eps, double wd) {
+  auto st = cur_stream();
+  for (size_t i = 0; i < masters.size(); ++i) {
+    auto& m = masters[i];
+    const long n = m.numel();
+    if (n == 0) continue;
+    CHECK_GPU(m);
+    TORCH_CHECK(m.scalar_type() == at::kFloat, "adamw: master must be fp32");
+    float* mp = m.data_ptr<float>();
+    float* ma = ms[i].data_ptr<float>();
+    float* va = vs[i].data_ptr<float>();
+    const bool g_bf = is_bf16(grads[i]);
+    const bool o_bf = is_bf16(outs[i]);
+    if (g_bf && o_bf)
+      adamw_launch_t<__bf16, __bf16>(mp, bfp(grads[i]), ma, va,
+                                     bfp_mut(outs[i]), n, (int)step, lr,
+                                     beta1, beta2, eps, wd, st);
+    else if (g_bf && !o_bf)
+      adamw_launch_t<__bf16, float>(mp, bfp(grads[i]), ma, va,
+                                    outs[i].data_ptr<float>(), n, (int)step,
+                                    lr, beta1, beta2, eps, wd, st);
+    else if (!g_bf && o_bf)
+      adamw_launch_t<float, __bf16>(mp, grads[i].data_ptr<float>(), ma,
+                                    va, bfp_mut(outs[i]), n, (int)step, lr,
+                                    beta1, beta2, eps, wd, st);
+    else
+      adamw_launch_t<float, float>(mp, grads[i].data_ptr<float>(), ma,
+                                   va, outs[i].data_ptr<float>(), n,
+                                   (int)step, lr, beta1, beta2, eps, wd, st);
+  }
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("ce_max", &ce_max);
+  m.def("ce_sum_target", &ce_sum_target);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("fused_adamw", &fused_adamw);
+  m.attr("arch") = "gfx950";
+}

@@ -1,0 +1,35 @@
+"""In-tree build of the gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built .so lands next to the package sources
+(hetu_galvatron_amd/ops/_galvatron_hip.*.so) so it travels with repo
+snapshots to GPU boxes (it is git-ignored but NOT gpurun-ignored).
+"""
+import os
+from glob import glob
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+SRC = sorted(glob(os.path.join(HERE, "hetu_galvatron_amd/ops/csrc/*.hip")))
+
+setup(
+    name="hetu_galvatron_amd_ext",
+    version="0.1.0",
+    ext_modules=[
+        CUDAExtension(
+            name="hetu_galvatron_amd.ops._galvatron_hip",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

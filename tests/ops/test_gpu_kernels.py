@@ -1,0 +1,254 @@
+"""HIP kernel numerics vs the plain-PyTorch fp32 reference (reference_ops).
+
+All tests are @pytest.mark.gpu (MI355X box). Each kernel is compared
+against the same-op fp32 torch reference on random data; asymmetric
+operands everywhere so operand/output transposes cannot pass (guide
+rule 16).
+"""
+import math
+
+import pytest
+import torch
+
+from hetu_galvatron_amd.ops import reference_ops as ref
+
+pytestmark = pytest.mark.gpu
+
+
+def ext():
+    from hetu_galvatron_amd.ops._ext import get_ext
+    return get_ext(False)
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def assert_close(a, b, atol, rtol=2e-2, what=""):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    err = (a - b).abs().max().item()
+    denom = b.abs().max().item() + 1e-6
+    assert err <= atol + rtol * denom, \
+        f"{what}: max|err|={err:.4e} vs atol={atol} rtol*max={rtol * denom:.4e}"
+
+
+# ---------------------------------------------------------------------------
+# MFMA layout probe: the fragment lane-map assumed by every MFMA kernel
+# ---------------------------------------------------------------------------
+def test_mfma_layout():
+    torch.manual_seed(0)
+    A = torch.randn(32, 16, device=dev()).bfloat16()
+    B = torch.randn(16, 32, device=dev()).bfloat16()
+    want = (A.float() @ B.float()).cpu()
+    got_default = ext().mfma_probe(A, B, False).cpu()
+    got_alt = ext().mfma_probe(A, B, True).cpu()
+    err_d = (got_default - want).abs().max().item()
+    err_a = (got_alt - want).abs().max().item()
+    assert err_d < 0.1, (
+        f"default MFMA A/B lane map wrong (err {err_d:.3e}); "
+        f"alt map err {err_a:.3e} — if alt matches, switch mfma32_ab_k "
+        f"to the two-block-of-4 form in ops/csrc/common.h")
+
+
+# ---------------------------------------------------------------------------
+# norms
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+@pytest.mark.parametrize("shape", [(33, 4096), (4, 128, 1024), (7, 8192)])
+def test_rmsnorm(dtype, shape):
+    torch.manual_seed(1)
+    x = torch.randn(shape, device=dev(), dtype=dtype)
+    w = (torch.randn(shape[-1], device=dev(), dtype=dtype) * 0.1 + 1.0)
+    y, inv = ext().rmsnorm_fwd(x, w, 1e-5)
+    y_ref, inv_ref = ref.rmsnorm_fwd(x.float(), w.float(), 1e-5)
+    atol = 3e-2 if dtype == torch.bfloat16 else 1e-5
+    assert_close(y, y_ref, atol, what="rmsnorm y")
+    assert_close(inv, inv_ref, 1e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="rmsnorm invrms")
+
+    dy = torch.randn_like(x)
+    dx, dw = ext().rmsnorm_bwd(dy.contiguous(), x, w, inv)
+    dx_ref, dw_ref = ref.rmsnorm_bwd(dy.float(), x.float(), w.float(),
+                                     inv_ref.float())
+    assert_close(dx, dx_ref, 3e-2 if dtype == torch.bfloat16 else 1e-4,
+                 what="rmsnorm dx")
+    n_rows = x.numel() // shape[-1]
+    assert_close(dw, dw_ref, (3e-2 if dtype == torch.bfloat16 else 1e-3) *
+                 math.sqrt(n_rows), what="rmsnorm dw")
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_layernorm(dtype):
+    torch.manual_seed(2)
+    x = torch.randn(65, 2048, device=dev(), dtype=dtype)
+    w = (torch.randn(2048, device=dev(), dtype=dtype) * 0.1 + 1.0)
+    b = torch.randn(2048, device=dev(), dtype=dtype) * 0.1
+    y, mean, inv = ext().layernorm_fwd(x, w, b, 1e-5)
+    y_ref, mean_ref, inv_ref = ref.layernorm_fwd(x.float(), w.float(),
+                                                 b.float(), 1e-5)
+    atol = 3e-2 if dtype == torch.bfloat16 else 1e-5
+    assert_close(y, y_ref, atol, what="ln y")
+    assert_close(mean, mean_ref, 1e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="ln mean")
+
+    dy = torch.randn_like(x)
+    dx, dw, db = ext().layernorm_bwd(dy.contiguous(), x, w, mean, inv)
+    dx_ref, dw_ref, db_ref = ref.layernorm_bwd(dy.float(), x.float(),
+                                               w.float(), mean_ref.float(),
+                                               inv_ref.float())
+    assert_close(dx, dx_ref, 3e-2 if dtype == torch.bfloat16 else 1e-4,
+                 what="ln dx")
+    assert_close(dw, dw_ref, 0.3 if dtype == torch.bfloat16 else 1e-2,
+                 what="ln dw")
+    assert_close(db, db_ref, 0.3 if dtype == torch.bfloat16 else 1e-2,
+                 what="ln db")
+
+
+# ---------------------------------------------------------------------------
+# swiglu / rope
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_swiglu(dtype):
+    torch.manual_seed(3)
+    x = torch.randn(17, 512, device=dev(), dtype=dtype)
+    y = ext().swiglu_fwd(x)
+    y_ref = ref.swiglu_fwd(x.float())
+    assert_close(y, y_ref, 2e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="swiglu y")
+    dy = torch.randn_like(y)
+    dx = ext().swiglu_bwd(dy.contiguous(), x)
+    dx_ref = ref.swiglu_bwd(dy.float(), x.float())
+    assert_close(dx, dx_ref, 2e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="swiglu dx")
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_rope(dtype):
+    torch.manual_seed(4)
+    s, b, h, d = 128, 2, 4, 128
+    x = torch.randn(s, b, h, d, device=dev(), dtype=dtype)
+    cos, sin = ref.rope_freqs(s, d, device=dev())
+    y = ext().rope_fwd(x, cos, sin, False)
+    y_ref = ref.rope_apply(x.float(), cos, sin)
+    assert_close(y, y_ref, 2e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="rope y")
+    # conj is the inverse rotation
+    back = ext().rope_fwd(y, cos, sin, True)
+    assert_close(back, x.float(),
+                 5e-2 if dtype == torch.bfloat16 else 1e-5, what="rope inv")
+
+
+# ---------------------------------------------------------------------------
+# flash attention
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("d", [64, 128])
+@pytest.mark.parametrize("causal", [True, False])
+@pytest.mark.parametrize("hq,hkv", [(8, 8), (8, 2)])
+@pytest.mark.parametrize("sq", [256, 300])
+def test_flash_fwd(d, causal, hq, hkv, sq):
+    torch.manual_seed(5)
+    b = 2
+    q = torch.randn(b, sq, hq, d, device=dev()).bfloat16()
+    k = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    v = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, causal, scale)
+    o_ref, lse_ref = ref.attention_fwd(q.float(), k.float(), v.float(),
+                                       causal, scale)
+    assert_close(o, o_ref, 3e-2, what=f"flash o d{d} c{causal}")
+    assert_close(lse, lse_ref, 2e-2, what="flash lse")
+
+
+@pytest.mark.parametrize("d", [64, 128])
+@pytest.mark.parametrize("hq,hkv", [(4, 4), (4, 1)])
+def test_flash_bwd(d, hq, hkv):
+    torch.manual_seed(6)
+    b, sq = 2, 192
+    q = torch.randn(b, sq, hq, d, device=dev()).bfloat16()
+    k = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    v = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale)
+    do = torch.randn_like(o)
+    dq, dk, dv = ext().flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                      True, scale)
+    dq_ref, dk_ref, dv_ref = ref.attention_bwd(do.float(), q.float(),
+                                               k.float(), v.float(), None,
+                                               None, True, scale)
+    assert_close(dq, dq_ref, 6e-2, rtol=3e-2, what="flash dq")
+    assert_close(dk, dk_ref, 6e-2, rtol=3e-2, what="flash dk")
+    assert_close(dv, dv_ref, 6e-2, rtol=3e-2, what="flash dv")
+
+
+def test_flash_cross_lengths():
+    """sq != skv (ring-CP block form), bottom-right causal alignment."""
+    torch.manual_seed(7)
+    b, hq, d = 1, 4, 128
+    q = torch.randn(b, 128, hq, d, device=dev()).bfloat16()
+    k = torch.randn(b, 256, hq, d, device=dev()).bfloat16()
+    v = torch.randn(b, 256, hq, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale)
+    o_ref, lse_ref = ref.attention_fwd(q.float(), k.float(), v.float(),
+                                       True, scale)
+    assert_close(o, o_ref, 3e-2, what="flash cross o")
+    assert_close(lse, lse_ref, 2e-2, what="flash cross lse")
+
+
+# ---------------------------------------------------------------------------
+# cross entropy
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+@pytest.mark.parametrize("V", [1000, 16032])
+def test_vocab_ce(dtype, V):
+    torch.manual_seed(8)
+    n, vocab_start = 64, 2 * V
+    logits = torch.randn(n, V, device=dev(), dtype=dtype) * 4
+    # half the targets on-shard, half off
+    target = torch.randint(0, 4 * V, (n,), device=dev())
+    gmax = ext().ce_max(logits)
+    gmax_ref = logits.float().max(dim=-1).values
+    assert_close(gmax, gmax_ref, 1e-3, what="ce max")
+    sumexp, tlogit = ext().ce_sum_target(logits, target, gmax_ref.contiguous(),
+                                         vocab_start)
+    se_ref, tl_ref = ref.vocab_ce_fwd_local(logits.float(), target, gmax_ref,
+                                            vocab_start, vocab_start + V)
+    assert_close(sumexp, se_ref, 1e-2, what="ce sumexp")
+    assert_close(tlogit, tl_ref, 1e-3, what="ce tlogit")
+
+    gout = torch.randn(n, device=dev()).float()
+    dl_ref = ref.vocab_ce_bwd_local(logits.float(), target, gmax_ref, se_ref,
+                                    gout, vocab_start, vocab_start + V)
+    dl = ext().ce_bwd(logits.clone(), target, gmax_ref.contiguous(),
+                      se_ref.contiguous(), gout, vocab_start)
+    assert_close(dl, dl_ref, 2e-2 if dtype == torch.bfloat16 else 1e-5,
+                 what="ce dlogits")
+
+
+# ---------------------------------------------------------------------------
+# fused adamw
+# ---------------------------------------------------------------------------
+def test_fused_adamw():
+    torch.manual_seed(9)
+    sizes = [1000, 4097, 31]
+    masters = [torch.randn(s, device=dev()).float() for s in sizes]
+    grads = [torch.randn(s, device=dev()).bfloat16() for s in sizes]
+    ms = [torch.rand(s, device=dev()).float() * 0.1 for s in sizes]
+    vs = [torch.rand(s, device=dev()).float() * 0.01 for s in sizes]
+    outs = [torch.zeros(s, device=dev()).bfloat16() for s in sizes]
+
+    masters_ref = [t.clone() for t in masters]
+    ms_ref = [t.clone() for t in ms]
+    vs_ref = [t.clone() for t in vs]
+    outs_ref = [t.clone() for t in outs]
+
+    ext().fused_adamw(masters, grads, ms, vs, outs, 3, 1e-3, 0.9, 0.95,
+                      1e-8, 0.01)
+    ref.adamw_step(outs_ref, grads, ms_ref, vs_ref, masters_ref, 3, 1e-3,
+                   0.9, 0.95, 1e-8, 0.01)
+    for i in range(len(sizes)):
+        assert_close(masters[i], masters_ref[i], 1e-5, what=f"adam master {i}")
+        assert_close(ms[i], ms_ref[i], 1e-5, what=f"adam m {i}")
+        assert_close(vs[i], vs_ref[i], 1e-6, what=f"adam v {i}")
+        assert_close(outs[i], outs_ref[i], 1e-2, what=f"adam out {i}")
