@@ -46,7 +46,40 @@ __global__ void adamw_kernel(float* __restrict__ master,
   }
 }
 
+// Fused microbatch gradient accumulation: flat_fp32[off:off+n] += (f32)g.
+// Replaces the eager `flat_grad[slice].add_(grad.flatten().float())` pair
+// (bf16->f32 temp + f32 add measured ~600 ms/step on the 8B bench; this
+// single pass is ~80 GB/microbatch instead of ~144).
+template <typename TG>
+__global__ void grad_accum_kernel(float* __restrict__ flat,
+                                  const TG* __restrict__ g, long n) {
+  const long n8 = n & ~7L;
+  for (long i = (blockIdx.x * (long)blockDim.x + threadIdx.x) * 8; i < n8;
+       i += (long)gridDim.x * blockDim.x * 8) {
+    float gv[8], fv[8];
+    VecIO<TG>::load(gv, g + i);
+    VecIO<float>::load(fv, flat + i);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) fv[k] += gv[k];
+    VecIO<float>::store(flat + i, fv);
+  }
+  for (long i = n8 + blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    flat[i] += (float)g[i];
+}
+
 }  // namespace
+
+template <typename TG>
+void grad_accum_launch_t(float* flat, const TG* g, long n, hipStream_t st) {
+  int grid = galv_grid((n / 8 + 255) / 256);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL((grad_accum_kernel<TG>), dim3(grid), dim3(256), 0, st,
+                     flat, g, n);
+}
+
+template void grad_accum_launch_t<__bf16>(float*, const __bf16*, long, hipStream_t);
+template void grad_accum_launch_t<float>(float*, const float*, long, hipStream_t);
 
 template <typename TG, typename TO>
 void adamw_launch_t(float* master, const TG* grad, float* m, float* v,

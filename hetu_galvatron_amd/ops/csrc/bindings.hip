@@ -16,17 +16,20 @@
 template <typename T>
 void rmsnorm_fwd_launch_t(const T*, const T*, T*, float*, long, int, float, hipStream_t);
 template <typename T>
-void rmsnorm_bwd_launch_t(const T*, const T*, const T*, const float*, T*, float*, long, int, hipStream_t);
+void rmsnorm_bwd_launch_t(const T*, const T*, const T*, const float*, T*, float*, float*, long, int, hipStream_t);
 template <typename T>
 void layernorm_fwd_launch_t(const T*, const T*, const T*, T*, float*, float*, long, int, float, hipStream_t);
 template <typename T>
 void layernorm_bwd_launch_t(const T*, const T*, const T*, const float*, const float*, T*, float*, float*, long, int, hipStream_t);
+int norm_bwd_grid(long n);
 template <typename T>
 void swiglu_fwd_launch_t(const T*, T*, long, int, hipStream_t);
 template <typename T>
 void swiglu_bwd_launch_t(const T*, const T*, T*, long, int, hipStream_t);
 template <typename T>
 void rope_launch_t(const T*, T*, const float*, const float*, long, int, int, bool, hipStream_t);
+template <typename TG>
+void grad_accum_launch_t(float*, const TG*, long, hipStream_t);
 template <typename TG, typename TO>
 void adamw_launch_t(float*, const TG*, float*, float*, TO*, long, int, float, float, float, float, float, hipStream_t);
 template <typename T>
@@ -87,17 +90,20 @@ std::tuple<Tensor, Tensor> rmsnorm_bwd(const Tensor& dy, const Tensor& x,
   const int H = x.size(-1);
   const long n = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({H}, x.options().dtype(at::kFloat));
+  const int G = norm_bwd_grid(n);
+  auto dw_part = at::empty({G, H}, x.options().dtype(at::kFloat));
   auto inv = invrms.contiguous();
   if (is_bf16(x))
     rmsnorm_bwd_launch_t<__bf16>(bfp(dy), bfp(x), bfp(w),
                                  inv.data_ptr<float>(), bfp_mut(dx),
+                                 dw_part.data_ptr<float>(),
                                  dw.data_ptr<float>(), n, H, cur_stream());
   else
     rmsnorm_bwd_launch_t<float>(dy.data_ptr<float>(), x.data_ptr<float>(),
                                 w.data_ptr<float>(), inv.data_ptr<float>(),
-                                dx.data_ptr<float>(), dw.data_ptr<float>(), n,
-                                H, cur_stream());
+                                dx.data_ptr<float>(), dw_part.data_ptr<float>(),
+                                dw.data_ptr<float>(), n, H, cur_stream());
   return {dx, dw};
 }
 
@@ -135,21 +141,22 @@ std::tuple<Tensor, Tensor, Tensor> layernorm_bwd(const Tensor& dy,
   const int H = x.size(-1);
   const long n = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dwdb = at::empty({2, H}, x.options().dtype(at::kFloat));
+  const int G = norm_bwd_grid(n);
+  auto part = at::empty({2, G, H}, x.options().dtype(at::kFloat));
   auto m = mean.contiguous(), r = invstd.contiguous();
   if (is_bf16(x))
     layernorm_bwd_launch_t<__bf16>(bfp(dy), bfp(x), bfp(w),
                                    m.data_ptr<float>(), r.data_ptr<float>(),
-                                   bfp_mut(dx), dw.data_ptr<float>(),
-                                   db.data_ptr<float>(), n, H, cur_stream());
+                                   bfp_mut(dx), part.data_ptr<float>(),
+                                   dwdb.data_ptr<float>(), n, H, cur_stream());
   else
     layernorm_bwd_launch_t<float>(dy.data_ptr<float>(), x.data_ptr<float>(),
                                   w.data_ptr<float>(), m.data_ptr<float>(),
                                   r.data_ptr<float>(), dx.data_ptr<float>(),
-                                  dw.data_ptr<float>(), db.data_ptr<float>(),
-                                  n, H, cur_stream());
-  return {dx, dw, db};
+                                  part.data_ptr<float>(),
+                                  dwdb.data_ptr<float>(), n, H, cur_stream());
+  return {dx, dwdb[0], dwdb[1]};
 }
 
 // ---- swiglu / rope --------------------------------------------------------
@@ -326,6 +333,19 @@ Tensor ce_bwd(Tensor logits, const Tensor& target, const Tensor& gmax,
   return logits;
 }
 
+// ---- grad accumulation ----------------------------------------------------
+void grad_accum(Tensor flat, const Tensor& grad, long offset) {
+  CHECK_GPU(flat); CHECK_GPU(grad);
+  TORCH_CHECK(flat.scalar_type() == at::kFloat, "grad_accum: flat must be fp32");
+  const long n = grad.numel();
+  TORCH_CHECK(offset + n <= flat.numel(), "grad_accum: out of range");
+  float* fp = flat.data_ptr<float>() + offset;
+  if (is_bf16(grad))
+    grad_accum_launch_t<__bf16>(fp, bfp(grad), n, cur_stream());
+  else
+    grad_accum_launch_t<float>(fp, grad.data_ptr<float>(), n, cur_stream());
+}
+
 // ---- fused AdamW ----------------------------------------------------------
 void fused_adamw(std::vector<Tensor> masters, std::vector<Tensor> grads,
                  std::vector<Tensor> ms, std::vector<Tensor> vs,
@@ -379,5 +399,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_sum_target", &ce_sum_target);
   m.def("ce_bwd", &ce_bwd);
   m.def("fused_adamw", &fused_adamw);
+  m.def("grad_accum", &grad_accum);
   m.attr("arch") = "gfx950";
 }

@@ -252,3 +252,14 @@ def test_fused_adamw():
         assert_close(ms[i], ms_ref[i], 1e-5, what=f"adam m {i}")
         assert_close(vs[i], vs_ref[i], 1e-6, what=f"adam v {i}")
         assert_close(outs[i], outs_ref[i], 1e-2, what=f"adam out {i}")
+
+
+def test_grad_accum():
+    torch.manual_seed(10)
+    flat = torch.randn(10000, device=dev()).float()
+    base = flat.clone()
+    g = torch.randn(4097, device=dev()).bfloat16()
+    ext().grad_accum(flat, g, 123)
+    want = base.clone()
+    want[123:123 + 4097] += g.float()
+    assert_close(flat, want, 1e-5, what="grad_accum")
