@@ -8,8 +8,11 @@ Used with rocprofv3 PMC runs to drive the guide's diagnostic loop.
 """
 import argparse
 import math
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
