@@ -9,51 +9,66 @@
 // Layout: q [b, sq, hq, D], k/v [b, skv, hkv, D] bf16, GQA (hq % hkv == 0),
 // causal = bottom-right aligned. lse [b, hq, sq] fp32 (natural log).
 //
-// v1 structure (correctness-first, MFMA throughout):
-//   forward: 4-wave workgroup = 128 q rows (32/wave), KV tiles of 32 staged
-//   in LDS (K row-major padded, V transposed), online softmax with the
-//   D-layout row map of v_mfma_f32_32x32x16_bf16, P staged through a
-//   per-wave LDS tile to re-enter as the PV A-operand.
-//   backward: separate dq and dkv kernels (no atomics); dkv computes
-//   per-q-head expanded dK/dV, host sums over the GQA group.
-// The MFMA A/B lane maps assumed here are validated on-device by
-// galv_mfma_probe (tests/ops/test_gpu_kernels.py::test_mfma_layout).
+// v2 forward structure (guide §B 8-warp ladder, plain HIP):
+//   8 waves x 32 q rows (QB=256/workgroup), KV tiles of 64;
+//   swapped QK^T (S^T = mfma(K, Q^T)) so each lane owns one q column ->
+//   softmax fully in-register (per-lane m/l scalars, exp2 units);
+//   P redistributed to PV fragments by cvt_pk_bf16 + permlane32_swap
+//   (no LDS round-trip); PV computed as O^T = mfma(V^T, P^T) so the
+//   running rescale is a plain per-lane multiply;
+//   V staged TRANSPOSED straight from global (coalesced 2-byte column
+//   loads -> b128 LDS writes; the v1 scalar ds_write_b16 transpose was
+//   14-37% of wave cycles in SQ_LDS_BANK_CONFLICT);
+//   async-stage split (T14): next tile's global loads issue before the
+//   current tile's compute, LDS writes land after the barrier;
+//   per-wave causal tile skip.
+// Backward keeps the v1 two-kernel (dq + expanded dkv) structure with the
+// same column-load transpose staging.
 #include "common.h"
 
 namespace {
 
-constexpr int QB = 128;   // q rows per workgroup (4 waves x 32)
-constexpr int KVB = 32;   // kv tile
+constexpr int KVB = 32;   // kv tile of the BACKWARD kernels
+constexpr int QB = 128;   // q/key rows per workgroup (backward kernels)
 
 DEV_INLINE float neg_big() { return -1e30f; }
 
+DEV_INLINE unsigned pack_bf16(float lo, float hi) {
+  union { __hip_bfloat162 h2; unsigned u; } cv;
+  cv.h2 = __hip_bfloat162(__float2bfloat16(lo), __float2bfloat16(hi));
+  return cv.u;
+}
+
+typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+
 // ---------------------------------------------------------------------------
-// forward
+// forward (v2)
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(512, 2)
 void flash_fwd_kernel(const __bf16* __restrict__ q,
                       const __bf16* __restrict__ k,
                       const __bf16* __restrict__ v,
                       __bf16* __restrict__ o, float* __restrict__ lse,
                       int b, int sq, int skv, int hq, int hkv, float scale,
                       bool causal) {
-  constexpr int KROW = D + 8;        // padded K-tile row (bf16 elems)
-  constexpr int VTROW = KVB + 8;     // padded transposed-V row
-  constexpr int PROW = KVB + 8;      // per-wave P staging row
-  constexpr int NK = D / 16;         // K-slices per S tile
-  constexpr int ND = D / 32;         // 32-wide d tiles of O
+  constexpr int KB = 64;            // kv tile
+  constexpr int QBF = 256;          // q rows per workgroup (8 waves x 32)
+  constexpr int KROW = D + 8;       // padded K row (bf16 elems)
+  constexpr int VTROW = KB + 8;     // padded transposed-V row
+  constexpr int NK = D / 16;        // S^T K-slices
+  constexpr int ND = D / 32;        // 32-wide d tiles of O^T
+  constexpr int KPT = D / 64;       // staging packs per thread (512 thr)
 
-  __shared__ __align__(16) __bf16 smem[KVB * KROW + D * VTROW + 4 * KVB * PROW];
+  __shared__ __align__(16) __bf16 smem[KB * KROW + D * VTROW];
   __bf16* k_lds = smem;
-  __bf16* vt_lds = smem + KVB * KROW;
-  __bf16* p_lds = vt_lds + D * VTROW;
+  __bf16* vt_lds = smem + KB * KROW;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int col = lane & 31;   // MFMA col (key index / d index)
+  const int col = lane & 31;
   const int hi = lane >> 5;
+  const int wid = tid >> 6;
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -62,13 +77,14 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const int hk = h / (hq / hkv);
 
   const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base_k = ((long)bi * skv * hkv + hk) * D;
+  const long kv_base = ((long)bi * skv * hkv + hk) * D;
   const int q_stride = hq * D;
   const int kv_stride = hkv * D;
   const int off = skv - sq;
+  const int q0w = qblk * QBF + wid * 32;
+  const float sl2e = scale * 1.4426950408889634f;  // fold log2(e): exp2 units
 
-  const int q0w = qblk * QB + wid * KVB;  // this wave's first q row
-  // A-fragments of Q: lane holds Q[q0w+col][ks*16 + 8*hi + j]
+  // Q fragments (B-operand of S^T): lane holds Q[q0w+col][ks*16 + 8*hi + j]
   bf16x8 qf[NK];
   {
     const int qg = min(q0w + col, sq - 1);
@@ -78,111 +94,186 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
       qf[ks] = *reinterpret_cast<const bf16x8*>(qp + ks * 16);
   }
 
-  f32x16 ov[ND];
+  f32x16 ot[ND];
 #pragma unroll
-  for (int dt = 0; dt < ND; ++dt) ov[dt] = (f32x16)(0.f);
-  float m_run[16], l_run[16];
-#pragma unroll
-  for (int i = 0; i < 16; ++i) { m_run[i] = neg_big(); l_run[i] = 0.f; }
+  for (int dt = 0; dt < ND; ++dt) ot[dt] = (f32x16)(0.f);
+  float m_run = neg_big(), l_run = 0.f;
 
-  int kv_end = skv;
-  if (causal) kv_end = min(skv, qblk * QB + QB + off);
-  __bf16* my_p = p_lds + wid * KVB * PROW;
+  int kv_end = causal ? min(skv, qblk * QBF + QBF + off) : skv;
+  const int kv_last_w = causal ? min(skv, q0w + 32 + off) : skv;  // wave's own
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
-    // ---- stage K (row-major, padded) and V (transposed) ----
-    __syncthreads();
-    constexpr int PACKS = KVB * D / 8;  // 16B packs in the tile
+  // staged registers for the next tile
+  bf16x8 kst[KPT];
+  ushort8 vst[KPT];
+
+  auto stage_load = [&](int kv0) {
 #pragma unroll
-    for (int p = tid; p < PACKS; p += 256) {
-      const int row = p / (D / 8);
-      const int c8 = (p - row * (D / 8)) * 8;
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
       const int kg = kv0 + row;
-      bf16x8 kvv = (bf16x8)(__bf16(0.f));
+      kst[p] = (bf16x8)(__bf16(0.f));
       if (kg < skv)
-        kvv = *reinterpret_cast<const bf16x8*>(
-            k + kv_base_k + (long)kg * kv_stride + c8);
-      *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kvv;
-      bf16x8 vv = (bf16x8)(__bf16(0.f));
-      if (kg < skv)
-        vv = *reinterpret_cast<const bf16x8*>(
-            v + kv_base_k + (long)kg * kv_stride + c8);
-#pragma unroll
-      for (int i = 0; i < 8; ++i) vt_lds[(c8 + i) * VTROW + row] = vv[i];
+        kst[p] = *reinterpret_cast<const bf16x8*>(
+            k + kv_base + (long)kg * kv_stride + c8);
     }
-    __syncthreads();
+#pragma unroll
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      const int c = idx & (D - 1);
+      const int kc = (idx / D) * 8;
+      const unsigned short* vp = reinterpret_cast<const unsigned short*>(
+          v + kv_base + (long)(kv0 + kc) * kv_stride + c);
+      ushort8 vv = (ushort8)(0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (kv0 + kc + j < skv) vv[j] = vp[(long)j * kv_stride];
+      vst[p] = vv;
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kst[p];
+    }
+#pragma unroll
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      const int c = idx & (D - 1);
+      const int kc = (idx / D) * 8;
+      *reinterpret_cast<ushort8*>(vt_lds + c * VTROW + kc) = vst[p];
+    }
+  };
 
-    // ---- S = Q K^T ----
-    f32x16 s_acc = (f32x16)(0.f);
-#pragma unroll
-    for (int ks = 0; ks < NK; ++ks) {
-      bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-          k_lds + col * KROW + ks * 16 + 8 * hi);
-      s_acc = mfma32_bf16(qf[ks], kb, s_acc);
-    }
+  stage_load(0);
+  stage_write();
+  __syncthreads();
 
-    // ---- mask + online softmax ----
-    float sv[16];
-    const int colg = kv0 + col;
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const int rowg = q0w + mfma32_d_row(lane, i);
-      float x = s_acc[i] * scale;
-      if (colg >= skv || (causal && colg > rowg + off)) x = neg_big();
-      sv[i] = x;
-    }
-    float alpha[16];
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const float tm = half_wave_max(sv[i]);
-      const float mn = fmaxf(m_run[i], tm);
-      alpha[i] = (m_run[i] <= neg_big()) ? 0.f : __expf(m_run[i] - mn);
-      m_run[i] = mn;
-    }
-    float pv[16];
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      pv[i] = (m_run[i] <= neg_big()) ? 0.f : __expf(sv[i] - m_run[i]);
-      l_run[i] = l_run[i] * alpha[i] + half_wave_sum(pv[i]);
-      my_p[mfma32_d_row(lane, i) * PROW + col] = (__bf16)pv[i];
-    }
-#pragma unroll
-    for (int dt = 0; dt < ND; ++dt)
-#pragma unroll
-      for (int i = 0; i < 16; ++i) ov[dt][i] *= alpha[i];
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    const bool have_next = kv0 + KB < kv_end;
+    if (have_next) stage_load(kv0 + KB);  // async: lands at stage_write
 
-    // ---- O += P V ----  (P re-read as A-fragments from this wave's LDS)
+    if (kv0 < kv_last_w) {  // per-wave causal skip (wave-uniform)
+      // ---- S^T = K Q^T (two 32-key tiles) ----
+      f32x16 st0 = (f32x16)(0.f), st1 = (f32x16)(0.f);
 #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-          my_p + col * PROW + kh * 16 + 8 * hi);
-#pragma unroll
-      for (int dt = 0; dt < ND; ++dt) {
-        bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-            vt_lds + (dt * 32 + col) * VTROW + kh * 16 + 8 * hi);
-        ov[dt] = mfma32_bf16(pa, vb, ov[dt]);
+      for (int ks = 0; ks < NK; ++ks) {
+        bf16x8 k0 = *reinterpret_cast<const bf16x8*>(
+            k_lds + col * KROW + ks * 16 + 8 * hi);
+        bf16x8 k1 = *reinterpret_cast<const bf16x8*>(
+            k_lds + (col + 32) * KROW + ks * 16 + 8 * hi);
+        st0 = mfma32_bf16(k0, qf[ks], st0);
+        st1 = mfma32_bf16(k1, qf[ks], st1);
       }
+
+      // ---- scale (+ mask on straddle/tail tiles) ----
+      float pv[32];
+      const int qg = q0w + col;
+      const bool need_mask =
+          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
+      if (need_mask) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int key0 = kv0 + mfma32_d_row(lane, r);
+          float x0 = st0[r] * sl2e, x1 = st1[r] * sl2e;
+          if (key0 >= skv || (causal && key0 > qg + off)) x0 = neg_big();
+          if (key0 + 32 >= skv || (causal && key0 + 32 > qg + off))
+            x1 = neg_big();
+          pv[r] = x0;
+          pv[16 + r] = x1;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          pv[r] = st0[r] * sl2e;
+          pv[16 + r] = st1[r] * sl2e;
+        }
+      }
+
+      // ---- in-register online softmax (per-lane q column) ----
+      float mt = pv[0];
+#pragma unroll
+      for (int i = 1; i < 32; ++i) mt = fmaxf(mt, pv[i]);
+      mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+      const float mn = fmaxf(m_run, mt);
+      const float alpha = (m_run <= neg_big()) ? 0.f : exp2f(m_run - mn);
+      const bool dead = (mn <= neg_big());
+      float ps = 0.f;
+#pragma unroll
+      for (int i = 0; i < 32; ++i) {
+        pv[i] = dead ? 0.f : exp2f(pv[i] - mn);
+        ps += pv[i];
+      }
+      ps += __shfl_xor(ps, 32, 64);
+      l_run = l_run * alpha + ps;
+      m_run = mn;
+#pragma unroll
+      for (int dt = 0; dt < ND; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+
+      // ---- P -> bf16 PV fragments via cvt_pk + permlane32_swap ----
+      // frag[ks] holds P[q=lane][16*ks + 8*hi + j], j=0..7, as 4 u32 words
+      unsigned w[16];
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const float* pt = pv + 16 * t;
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+#pragma unroll
+          for (int g = 0; g < 2; ++g) {
+            // pairs (8i+2g, 8i+2g+1) with (8i+2g+4, 8i+2g+5)
+            unsigned a = pack_bf16(pt[8 * i + 2 * g], pt[8 * i + 2 * g + 1]);
+            unsigned bb = pack_bf16(pt[8 * i + 2 * g + 4],
+                                    pt[8 * i + 2 * g + 5]);
+            auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
+            // words of frag (2t + i): position 2g -> r2[0], 2g+1... see map
+            w[(2 * t + i) * 4 + g] = r2[0];      // keys (8hi + 2g, +1)
+            w[(2 * t + i) * 4 + 2 + g] = r2[1];  // keys (8hi + 4 + 2g, +1)
+          }
+        }
+      }
+
+      // ---- O^T += V^T P^T ----
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        union { unsigned u[4]; bf16x8 f; } pb;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) pb.u[g] = w[ks * 4 + g];
+#pragma unroll
+        for (int dt = 0; dt < ND; ++dt) {
+          bf16x8 va = *reinterpret_cast<const bf16x8*>(
+              vt_lds + (col + 32 * dt) * VTROW + ks * 16 + 8 * hi);
+          ot[dt] = mfma32_bf16(va, pb.f, ot[dt]);
+        }
+      }
+    }
+
+    __syncthreads();
+    if (have_next) {
+      stage_write();
+      __syncthreads();
     }
   }
 
-  // ---- epilogue ----
-  float inv_l[16];
+  // ---- epilogue: O^T[d][q=lane], per-lane stats ----
+  const int qg = q0w + col;
+  if (qg < sq) {
+    const float invl = l_run > 0.f ? 1.f / l_run : 0.f;
+    __bf16* orow = o + q_base + (long)qg * q_stride;
 #pragma unroll
-  for (int i = 0; i < 16; ++i)
-    inv_l[i] = l_run[i] > 0.f ? 1.f / l_run[i] : 0.f;
+    for (int dt = 0; dt < ND; ++dt)
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    const int rowg = q0w + mfma32_d_row(lane, i);
-    if (rowg < sq) {
-      __bf16* orow = o + q_base + (long)rowg * q_stride;
-#pragma unroll
-      for (int dt = 0; dt < ND; ++dt)
-        orow[dt * 32 + col] = (__bf16)(ov[dt][i] * inv_l[i]);
-      if (col == 0)
-        lse[((long)bi * hq + h) * sq + rowg] =
-            l_run[i] > 0.f ? m_run[i] + __logf(l_run[i])
-                           : -INFINITY;
-    }
+      for (int r = 0; r < 16; ++r)
+        orow[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)(ot[dt][r] * invl);
+    if (hi == 0)
+      lse[((long)bi * hq + h) * sq + qg] =
+          l_run > 0.f ? (m_run + log2f(l_run)) * 0.6931471805599453f
+                      : -INFINITY;
   }
 }
 
@@ -295,30 +386,46 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 
   int kv_end = skv;
   if (causal) kv_end = min(skv, qblk * QB + QB + off);
+  const int kv_last_w = causal ? min(skv, q0w + KVB + off) : skv;
   __bf16* my_ds = ds_lds + wid * KVB * PROW;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
     __syncthreads();
-    constexpr int PACKS = KVB * D / 8;
+    {
+      constexpr int PACKS = KVB * D / 8;
 #pragma unroll
-    for (int p = tid; p < PACKS; p += 256) {
-      const int row = p / (D / 8);
-      const int c8 = (p - row * (D / 8)) * 8;
-      const int kg = kv0 + row;
-      bf16x8 kvv = (bf16x8)(__bf16(0.f));
-      if (kg < skv)
-        kvv = *reinterpret_cast<const bf16x8*>(
-            k + kv_base + (long)kg * kv_stride + c8);
-      *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kvv;
+      for (int p = tid; p < PACKS; p += 256) {
+        const int row = p / (D / 8);
+        const int c8 = (p - row * (D / 8)) * 8;
+        const int kg = kv0 + row;
+        bf16x8 kvv = (bf16x8)(__bf16(0.f));
+        bf16x8 vv = (bf16x8)(__bf16(0.f));
+        if (kg < skv) {
+          kvv = *reinterpret_cast<const bf16x8*>(
+              k + kv_base + (long)kg * kv_stride + c8);
+          vv = *reinterpret_cast<const bf16x8*>(
+              v + kv_base + (long)kg * kv_stride + c8);
+        }
+        *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kvv;
+        *reinterpret_cast<bf16x8*>(v_lds + row * KROW + c8) = vv;
+      }
+      // K transposed: coalesced 2-byte column loads -> b128 LDS writes
+      constexpr int TPACKS = D * KVB / 8;
 #pragma unroll
-      for (int i = 0; i < 8; ++i) kt_lds[(c8 + i) * KTROW + row] = kvv[i];
-      bf16x8 vv = (bf16x8)(__bf16(0.f));
-      if (kg < skv)
-        vv = *reinterpret_cast<const bf16x8*>(
-            v + kv_base + (long)kg * kv_stride + c8);
-      *reinterpret_cast<bf16x8*>(v_lds + row * KROW + c8) = vv;
+      for (int p = tid; p < TPACKS; p += 256) {
+        const int c = p & (D - 1);
+        const int kc = (p / D) * 8;
+        const unsigned short* kp = reinterpret_cast<const unsigned short*>(
+            k + kv_base + (long)(kv0 + kc) * kv_stride + c);
+        ushort8 kv8 = (ushort8)(0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (kv0 + kc + j < skv) kv8[j] = kp[(long)j * kv_stride];
+        *reinterpret_cast<ushort8*>(kt_lds + c * KTROW + kc) = kv8;
+      }
     }
     __syncthreads();
+    if (kv0 >= kv_last_w) continue;  // wave-uniform causal skip
 
     // S and dP
     f32x16 s_acc = (f32x16)(0.f), dp_acc = (f32x16)(0.f);
@@ -446,38 +553,59 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   // first q row that can attend any key in this block (causal)
   int qstart = 0;
   if (causal) qstart = max(0, ((kvblk * QB - off) / KVB) * KVB);
+  // this wave's own first relevant q tile
+  const int qstart_w =
+      causal ? max(qstart, ((k0w - off) / KVB) * KVB) : qstart;
   __bf16* my_p = p_lds + wid * KVB * PROW;
 
   for (int qt0 = qstart; qt0 < sq; qt0 += KVB) {
     __syncthreads();
-    constexpr int PACKS = KVB * D / 8;
+    {
+      constexpr int PACKS = KVB * D / 8;
 #pragma unroll
-    for (int p = tid; p < PACKS; p += 256) {
-      const int row = p / (D / 8);
-      const int c8 = (p - row * (D / 8)) * 8;
-      const int qg = qt0 + row;
-      bf16x8 qv = (bf16x8)(__bf16(0.f));
-      bf16x8 dov = (bf16x8)(__bf16(0.f));
-      if (qg < sq) {
-        qv = *reinterpret_cast<const bf16x8*>(
-            q + q_base + (long)qg * q_stride + c8);
-        dov = *reinterpret_cast<const bf16x8*>(
-            dout + q_base + (long)qg * q_stride + c8);
+      for (int p = tid; p < PACKS; p += 256) {
+        const int row = p / (D / 8);
+        const int c8 = (p - row * (D / 8)) * 8;
+        const int qg = qt0 + row;
+        bf16x8 qv = (bf16x8)(__bf16(0.f));
+        bf16x8 dov = (bf16x8)(__bf16(0.f));
+        if (qg < sq) {
+          qv = *reinterpret_cast<const bf16x8*>(
+              q + q_base + (long)qg * q_stride + c8);
+          dov = *reinterpret_cast<const bf16x8*>(
+              dout + q_base + (long)qg * q_stride + c8);
+        }
+        *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qv;
+        *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dov;
       }
-      *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qv;
-      *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dov;
+      // transposed Q / dO via coalesced column loads
+      constexpr int TPACKS = D * KVB / 8;
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        qt_lds[(c8 + i) * TROW + row] = qv[i];
-        dot_lds[(c8 + i) * TROW + row] = dov[i];
+      for (int p = tid; p < TPACKS; p += 256) {
+        const int c = p & (D - 1);
+        const int qc = (p / D) * 8;
+        const unsigned short* qp = reinterpret_cast<const unsigned short*>(
+            q + q_base + (long)(qt0 + qc) * q_stride + c);
+        const unsigned short* dop = reinterpret_cast<const unsigned short*>(
+            dout + q_base + (long)(qt0 + qc) * q_stride + c);
+        ushort8 q8 = (ushort8)(0), d8 = (ushort8)(0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (qt0 + qc + j < sq) {
+            q8[j] = qp[(long)j * q_stride];
+            d8[j] = dop[(long)j * q_stride];
+          }
+        *reinterpret_cast<ushort8*>(qt_lds + c * TROW + qc) = q8;
+        *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = d8;
       }
-    }
-    if (tid < KVB) {
-      const int qg = qt0 + tid;
-      lse_lds[tid] = qg < sq ? lse[((long)bi * hq + h) * sq + qg] : INFINITY;
-      di_lds[tid] = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
+      if (tid < KVB) {
+        const int qg = qt0 + tid;
+        lse_lds[tid] = qg < sq ? lse[((long)bi * hq + h) * sq + qg] : INFINITY;
+        di_lds[tid] = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
+      }
     }
     __syncthreads();
+    if (qt0 < qstart_w) continue;  // wave-uniform causal skip
 
     // S^T = K Q^T  and  dP^T = V dO^T
     f32x16 st_acc = (f32x16)(0.f), dpt_acc = (f32x16)(0.f);
@@ -504,7 +632,6 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
       pt[i] = p;
       my_p[mfma32_d_row(lane, i) * PROW + col] = (__bf16)p;
     }
-    __syncthreads();  // none needed across waves for my_p; guards qt reuse
 
     // dV += P^T dO  (P^T via LDS A-frags, dO^T tile as B)
 #pragma unroll
@@ -585,8 +712,8 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                const __bf16* v, __bf16* o, float* lse, int b,
                                int sq, int skv, int hq, int hkv, float scale,
                                bool causal, hipStream_t st) {
-  dim3 grid((sq + QB - 1) / QB, b * hq);
-  hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(256), 0, st, q, k, v,
+  dim3 grid((sq + 255) / 256, b * hq);
+  hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k, v,
                      o, lse, b, sq, skv, hq, hkv, scale, causal);
 }
 

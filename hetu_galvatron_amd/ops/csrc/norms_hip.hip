@@ -121,21 +121,19 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
 }
 
 // column-sum of per-block partials: out[c] = sum_g part[g][c]
+// one column per thread; consecutive threads read consecutive columns of
+// each partial row (coalesced), grid-stride over columns
 __global__ void colsum_kernel(const float* __restrict__ part,
                               float* __restrict__ out, int G, int H,
                               int n_out) {
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x;
-       idx < (long)n_out * (H / 8); idx += (long)gridDim.x * blockDim.x) {
-    const int o = idx / (H / 8);
-    const int c = (idx - (long)o * (H / 8)) * 8;
-    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int g = 0; g < G; ++g) {
-      float v[8];
-      VecIO<float>::load(v, part + ((long)o * G + g) * H + c);
-#pragma unroll
-      for (int i = 0; i < 8; ++i) acc[i] += v[i];
-    }
-    VecIO<float>::store(out + (long)o * H + c, acc);
+       idx < (long)n_out * H; idx += (long)gridDim.x * blockDim.x) {
+    const int o = idx / H;
+    const int c = idx - (long)o * H;
+    float acc = 0.f;
+    const float* p = part + (long)o * G * H + c;
+    for (int g = 0; g < G; ++g) acc += p[(long)g * H];
+    out[idx] = acc;
   }
 }
 
@@ -287,7 +285,7 @@ void rmsnorm_fwd_launch_t(const T* x, const T* w, T* y, float* invrms,
 #undef RUN
 }
 
-int norm_bwd_grid(long n) { return galv_grid(n, 512); }
+int norm_bwd_grid(long n) { return galv_grid(n, 2048); }
 
 template <typename T>
 void rmsnorm_bwd_launch_t(const T* dy, const T* x, const T* w,
@@ -298,7 +296,7 @@ void rmsnorm_bwd_launch_t(const T* dy, const T* x, const T* w,
                                   dim3(256), 0, s, dy, x, w, invrms, dx, dw_part, n, H)
   DISPATCH_VPT(H, RUN);
 #undef RUN
-  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((H / 8 + 255) / 256)),
+  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((H + 255) / 256)),
                      dim3(256), 0, s, dw_part, dw, grid, H, 1);
 }
 
@@ -323,7 +321,7 @@ void layernorm_bwd_launch_t(const T* dy, const T* x, const T* w,
                                   dim3(256), 0, s, dy, x, w, mean, invstd, dx, dwdb_part, n, H)
   DISPATCH_VPT(H, RUN);
 #undef RUN
-  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((2 * H / 8 + 255) / 256)),
+  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((2 * H + 255) / 256)),
                      dim3(256), 0, s, dwdb_part, dwdb, grid, H, 2);
 }
 
