@@ -105,29 +105,47 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   bf16x8 kst[KPT];
   ushort8 vst[KPT];
 
+  // Guards are hoisted out of the load loops: per-element branches around
+  // staged loads make hipcc wait vmcnt(0) after EACH load (32 serialized
+  // memory round trips per tile, ~13 us/tile measured — guide §5 trap (c)).
+  // Fast path is branch-free; the tail tile loads clamped addresses and
+  // zeroes by value-select.
   auto stage_load = [&](int kv0) {
+    const bool full = (kv0 + KB <= skv);
 #pragma unroll
     for (int p = 0; p < KPT; ++p) {
       const int idx = tid + p * 512;
       const int row = idx / (D / 8);
       const int c8 = (idx - row * (D / 8)) * 8;
       const int kg = kv0 + row;
-      kst[p] = (bf16x8)(__bf16(0.f));
-      if (kg < skv)
+      if (full) {
         kst[p] = *reinterpret_cast<const bf16x8*>(
             k + kv_base + (long)kg * kv_stride + c8);
+      } else {
+        bf16x8 t = *reinterpret_cast<const bf16x8*>(
+            k + kv_base + (long)min(kg, skv - 1) * kv_stride + c8);
+        kst[p] = kg < skv ? t : (bf16x8)(__bf16(0.f));
+      }
     }
 #pragma unroll
     for (int p = 0; p < KPT; ++p) {
       const int idx = tid + p * 512;
       const int c = idx & (D - 1);
       const int kc = (idx / D) * 8;
-      const unsigned short* vp = reinterpret_cast<const unsigned short*>(
-          v + kv_base + (long)(kv0 + kc) * kv_stride + c);
-      ushort8 vv = (ushort8)(0);
+      const unsigned short* vb = reinterpret_cast<const unsigned short*>(
+          v + kv_base) + c;
+      ushort8 vv;
+      if (full) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        if (kv0 + kc + j < skv) vv[j] = vp[(long)j * kv_stride];
+        for (int j = 0; j < 8; ++j)
+          vv[j] = vb[(long)(kv0 + kc + j) * kv_stride];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned short t = vb[(long)min(kv0 + kc + j, skv - 1) * kv_stride];
+          vv[j] = kv0 + kc + j < skv ? t : (unsigned short)0;
+        }
+      }
       vst[p] = vv;
     }
   };
@@ -391,19 +409,20 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
     __syncthreads();
     {
+      const bool full = (kv0 + KVB <= skv);
       constexpr int PACKS = KVB * D / 8;
 #pragma unroll
       for (int p = tid; p < PACKS; p += 256) {
         const int row = p / (D / 8);
         const int c8 = (p - row * (D / 8)) * 8;
-        const int kg = kv0 + row;
-        bf16x8 kvv = (bf16x8)(__bf16(0.f));
-        bf16x8 vv = (bf16x8)(__bf16(0.f));
-        if (kg < skv) {
-          kvv = *reinterpret_cast<const bf16x8*>(
-              k + kv_base + (long)kg * kv_stride + c8);
-          vv = *reinterpret_cast<const bf16x8*>(
-              v + kv_base + (long)kg * kv_stride + c8);
+        const int kg = full ? kv0 + row : min(kv0 + row, skv - 1);
+        bf16x8 kvv = *reinterpret_cast<const bf16x8*>(
+            k + kv_base + (long)kg * kv_stride + c8);
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            v + kv_base + (long)kg * kv_stride + c8);
+        if (!full && kv0 + row >= skv) {
+          kvv = (bf16x8)(__bf16(0.f));
+          vv = (bf16x8)(__bf16(0.f));
         }
         *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kvv;
         *reinterpret_cast<bf16x8*>(v_lds + row * KROW + c8) = vv;
@@ -414,12 +433,21 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       for (int p = tid; p < TPACKS; p += 256) {
         const int c = p & (D - 1);
         const int kc = (p / D) * 8;
-        const unsigned short* kp = reinterpret_cast<const unsigned short*>(
-            k + kv_base + (long)(kv0 + kc) * kv_stride + c);
-        ushort8 kv8 = (ushort8)(0);
+        const unsigned short* kb = reinterpret_cast<const unsigned short*>(
+            k + kv_base) + c;
+        ushort8 kv8;
+        if (full) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (kv0 + kc + j < skv) kv8[j] = kp[(long)j * kv_stride];
+          for (int j = 0; j < 8; ++j)
+            kv8[j] = kb[(long)(kv0 + kc + j) * kv_stride];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            unsigned short t =
+                kb[(long)min(kv0 + kc + j, skv - 1) * kv_stride];
+            kv8[j] = kv0 + kc + j < skv ? t : (unsigned short)0;
+          }
+        }
         *reinterpret_cast<ushort8*>(kt_lds + c * KTROW + kc) = kv8;
       }
     }
@@ -560,19 +588,20 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   for (int qt0 = qstart; qt0 < sq; qt0 += KVB) {
     __syncthreads();
     {
+      const bool full = (qt0 + KVB <= sq);
       constexpr int PACKS = KVB * D / 8;
 #pragma unroll
       for (int p = tid; p < PACKS; p += 256) {
         const int row = p / (D / 8);
         const int c8 = (p - row * (D / 8)) * 8;
-        const int qg = qt0 + row;
-        bf16x8 qv = (bf16x8)(__bf16(0.f));
-        bf16x8 dov = (bf16x8)(__bf16(0.f));
-        if (qg < sq) {
-          qv = *reinterpret_cast<const bf16x8*>(
-              q + q_base + (long)qg * q_stride + c8);
-          dov = *reinterpret_cast<const bf16x8*>(
-              dout + q_base + (long)qg * q_stride + c8);
+        const int qg = full ? qt0 + row : min(qt0 + row, sq - 1);
+        bf16x8 qv = *reinterpret_cast<const bf16x8*>(
+            q + q_base + (long)qg * q_stride + c8);
+        bf16x8 dov = *reinterpret_cast<const bf16x8*>(
+            dout + q_base + (long)qg * q_stride + c8);
+        if (!full && qt0 + row >= sq) {
+          qv = (bf16x8)(__bf16(0.f));
+          dov = (bf16x8)(__bf16(0.f));
         }
         *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qv;
         *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dov;
@@ -583,17 +612,25 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
       for (int p = tid; p < TPACKS; p += 256) {
         const int c = p & (D - 1);
         const int qc = (p / D) * 8;
-        const unsigned short* qp = reinterpret_cast<const unsigned short*>(
-            q + q_base + (long)(qt0 + qc) * q_stride + c);
-        const unsigned short* dop = reinterpret_cast<const unsigned short*>(
-            dout + q_base + (long)(qt0 + qc) * q_stride + c);
-        ushort8 q8 = (ushort8)(0), d8 = (ushort8)(0);
+        const unsigned short* qb = reinterpret_cast<const unsigned short*>(
+            q + q_base) + c;
+        const unsigned short* dob = reinterpret_cast<const unsigned short*>(
+            dout + q_base) + c;
+        ushort8 q8, d8;
+        if (full) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (qt0 + qc + j < sq) {
-            q8[j] = qp[(long)j * q_stride];
-            d8[j] = dop[(long)j * q_stride];
+          for (int j = 0; j < 8; ++j) {
+            q8[j] = qb[(long)(qt0 + qc + j) * q_stride];
+            d8[j] = dob[(long)(qt0 + qc + j) * q_stride];
           }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const long rr = (long)min(qt0 + qc + j, sq - 1) * q_stride;
+            q8[j] = qt0 + qc + j < sq ? qb[rr] : (unsigned short)0;
+            d8[j] = qt0 + qc + j < sq ? dob[rr] : (unsigned short)0;
+          }
+        }
         *reinterpret_cast<ushort8*>(qt_lds + c * TROW + qc) = q8;
         *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = d8;
       }
