@@ -328,10 +328,14 @@ __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward dQ: grid over q blocks; loops kv tiles
+// backward dQ (v2): same swapped structure as the forward — 8 waves x 32 q
+// rows, per-lane lse/di scalars, dS^T packed to MFMA fragments with
+// cvt_pk + permlane32_swap (no dS LDS round-trip, no per-row stat arrays:
+// the v1 form spilled 34 dwords/lane to scratch).
+// dQ^T[d][q] += K^T dS accumulated in D-layout, epilogue like the forward.
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(512, 2)
 void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
                          const __bf16* __restrict__ k,
@@ -341,25 +345,24 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          __bf16* __restrict__ dq,
                          int b, int sq, int skv, int hq, int hkv,
                          float scale, bool causal) {
+  constexpr int KB = 32;         // kv tile
+  constexpr int QBF = 256;       // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;
-  constexpr int KTROW = KVB + 8;
-  constexpr int PROW = KVB + 8;
+  constexpr int KTROW = KB + 8;
   constexpr int NK = D / 16;
   constexpr int ND = D / 32;
+  constexpr int KPT = (KB * D / 8 + 511) / 512;  // row packs per thread
 
-  __shared__ __align__(16) __bf16
-      smem[KVB * KROW /*K*/ + D * KTROW /*KT*/ + KVB * KROW /*V*/ +
-           4 * KVB * PROW /*dS per wave*/];
+  __shared__ __align__(16) __bf16 smem[2 * KB * KROW + D * KTROW];
   __bf16* k_lds = smem;
-  __bf16* kt_lds = smem + KVB * KROW;
-  __bf16* v_lds = kt_lds + D * KTROW;
-  __bf16* ds_lds = v_lds + KVB * KROW;
+  __bf16* v_lds = smem + KB * KROW;
+  __bf16* kt_lds = v_lds + KB * KROW;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;
   const int col = lane & 31;
   const int hi = lane >> 5;
+  const int wid = tid >> 6;
 
   const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
@@ -372,9 +375,10 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const int q_stride = hq * D;
   const int kv_stride = hkv * D;
   const int off = skv - sq;
-  const int q0w = qblk * QB + wid * KVB;
+  const int q0w = qblk * QBF + wid * 32;
 
   bf16x8 qf[NK], dof[NK];
+  float lse_c, di_c;
   {
     const int qg = min(q0w + col, sq - 1);
     const __bf16* qp = q + q_base + (long)qg * q_stride + 8 * hi;
@@ -384,123 +388,170 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       qf[ks] = *reinterpret_cast<const bf16x8*>(qp + ks * 16);
       dof[ks] = *reinterpret_cast<const bf16x8*>(dop + ks * 16);
     }
-  }
-  float lse_r[16], di_r[16];
-#pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    const int rowg = q0w + mfma32_d_row(lane, i);
-    if (rowg < sq) {
-      lse_r[i] = lse[((long)bi * hq + h) * sq + rowg];
-      di_r[i] = di[((long)bi * hq + h) * sq + rowg];
-    } else {
-      lse_r[i] = INFINITY;  // exp(x - inf) = 0 -> junk rows contribute 0
-      di_r[i] = 0.f;
-    }
+    const float l0 = lse[((long)bi * hq + h) * sq + qg];
+    const float d0 = di[((long)bi * hq + h) * sq + qg];
+    const bool live = q0w + col < sq;
+    lse_c = live ? l0 : INFINITY;  // exp(x - inf) = 0 -> dead rows silent
+    di_c = live ? d0 : 0.f;
   }
 
   f32x16 dq_acc[ND];
 #pragma unroll
   for (int dt = 0; dt < ND; ++dt) dq_acc[dt] = (f32x16)(0.f);
 
-  int kv_end = skv;
-  if (causal) kv_end = min(skv, qblk * QB + QB + off);
-  const int kv_last_w = causal ? min(skv, q0w + KVB + off) : skv;
-  __bf16* my_ds = ds_lds + wid * KVB * PROW;
+  const int kv_end = causal ? min(skv, qblk * QBF + QBF + off) : skv;
+  const int kv_last_w = causal ? min(skv, q0w + 32 + off) : skv;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
-    __syncthreads();
-    {
-      const bool full = (kv0 + KVB <= skv);
-      constexpr int PACKS = KVB * D / 8;
+  bf16x8 kst[KPT], vst_r[KPT];
+  ushort8 ktst[KPT * 2];
+
+  auto stage_load = [&](int kv0) {
+    const bool full = (kv0 + KB <= skv);
 #pragma unroll
-      for (int p = tid; p < PACKS; p += 256) {
-        const int row = p / (D / 8);
-        const int c8 = (p - row * (D / 8)) * 8;
-        const int kg = full ? kv0 + row : min(kv0 + row, skv - 1);
-        bf16x8 kvv = *reinterpret_cast<const bf16x8*>(
-            k + kv_base + (long)kg * kv_stride + c8);
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            v + kv_base + (long)kg * kv_stride + c8);
-        if (!full && kv0 + row >= skv) {
-          kvv = (bf16x8)(__bf16(0.f));
-          vv = (bf16x8)(__bf16(0.f));
-        }
-        *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kvv;
-        *reinterpret_cast<bf16x8*>(v_lds + row * KROW + c8) = vv;
-      }
-      // K transposed: coalesced 2-byte column loads -> b128 LDS writes
-      constexpr int TPACKS = D * KVB / 8;
-#pragma unroll
-      for (int p = tid; p < TPACKS; p += 256) {
-        const int c = p & (D - 1);
-        const int kc = (p / D) * 8;
-        const unsigned short* kb = reinterpret_cast<const unsigned short*>(
-            k + kv_base) + c;
-        ushort8 kv8;
-        if (full) {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            kv8[j] = kb[(long)(kv0 + kc + j) * kv_stride];
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            unsigned short t =
-                kb[(long)min(kv0 + kc + j, skv - 1) * kv_stride];
-            kv8[j] = kv0 + kc + j < skv ? t : (unsigned short)0;
-          }
-        }
-        *reinterpret_cast<ushort8*>(kt_lds + c * KTROW + kc) = kv8;
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      if (idx >= KB * D / 8) break;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
+      const int kg = full ? kv0 + row : min(kv0 + row, skv - 1);
+      kst[p] = *reinterpret_cast<const bf16x8*>(
+          k + kv_base + (long)kg * kv_stride + c8);
+      vst_r[p] = *reinterpret_cast<const bf16x8*>(
+          v + kv_base + (long)kg * kv_stride + c8);
+      if (!full && kv0 + row >= skv) {
+        kst[p] = (bf16x8)(__bf16(0.f));
+        vst_r[p] = (bf16x8)(__bf16(0.f));
       }
     }
-    __syncthreads();
-    if (kv0 >= kv_last_w) continue;  // wave-uniform causal skip
-
-    // S and dP
-    f32x16 s_acc = (f32x16)(0.f), dp_acc = (f32x16)(0.f);
 #pragma unroll
-    for (int ks = 0; ks < NK; ++ks) {
-      bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-          k_lds + col * KROW + ks * 16 + 8 * hi);
-      s_acc = mfma32_bf16(qf[ks], kb, s_acc);
-      bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-          v_lds + col * KROW + ks * 16 + 8 * hi);
-      dp_acc = mfma32_bf16(dof[ks], vb, dp_acc);
+    for (int p = 0; p < KPT * 2; ++p) {
+      const int idx = tid + p * 512;
+      if (idx >= D * KB / 8) break;
+      const int c = idx & (D - 1);
+      const int kc = (idx / D) * 8;
+      const unsigned short* kb = reinterpret_cast<const unsigned short*>(
+          k + kv_base) + c;
+      ushort8 t;
+      if (full) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          t[j] = kb[(long)(kv0 + kc + j) * kv_stride];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned short x = kb[(long)min(kv0 + kc + j, skv - 1) * kv_stride];
+          t[j] = kv0 + kc + j < skv ? x : (unsigned short)0;
+        }
+      }
+      ktst[p] = t;
     }
-
-    const int colg = kv0 + col;
+  };
+  auto stage_write = [&]() {
 #pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const int rowg = q0w + mfma32_d_row(lane, i);
-      float p = 0.f;
-      if (colg < skv && !(causal && colg > rowg + off))
-        p = __expf(s_acc[i] * scale - lse_r[i]);
-      const float ds = p * (dp_acc[i] - di_r[i]) * scale;
-      my_ds[mfma32_d_row(lane, i) * PROW + col] = (__bf16)ds;
+    for (int p = 0; p < KPT; ++p) {
+      const int idx = tid + p * 512;
+      if (idx >= KB * D / 8) break;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(k_lds + row * KROW + c8) = kst[p];
+      *reinterpret_cast<bf16x8*>(v_lds + row * KROW + c8) = vst_r[p];
     }
+#pragma unroll
+    for (int p = 0; p < KPT * 2; ++p) {
+      const int idx = tid + p * 512;
+      if (idx >= D * KB / 8) break;
+      const int c = idx & (D - 1);
+      const int kc = (idx / D) * 8;
+      *reinterpret_cast<ushort8*>(kt_lds + c * KTROW + kc) = ktst[p];
+    }
+  };
 
-    // dQ += dS @ K   (dS via LDS as A-fragments; K^T via transposed tile)
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  const float sl2e = scale;  // bwd stays in natural-log units (lse is ln)
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    const bool have_next = kv0 + KB < kv_end;
+    if (have_next) stage_load(kv0 + KB);
+
+    if (kv0 < kv_last_w) {
+      // S^T = K Q^T ; dP^T = V dO^T
+      f32x16 st = (f32x16)(0.f), dp = (f32x16)(0.f);
 #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      bf16x8 da = *reinterpret_cast<const bf16x8*>(
-          my_ds + col * PROW + kh * 16 + 8 * hi);
-#pragma unroll
-      for (int dt = 0; dt < ND; ++dt) {
+      for (int ks = 0; ks < NK; ++ks) {
         bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-            kt_lds + (dt * 32 + col) * KTROW + kh * 16 + 8 * hi);
-        dq_acc[dt] = mfma32_bf16(da, kb, dq_acc[dt]);
+            k_lds + col * KROW + ks * 16 + 8 * hi);
+        st = mfma32_bf16(kb, qf[ks], st);
+        bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+            v_lds + col * KROW + ks * 16 + 8 * hi);
+        dp = mfma32_bf16(vb, dof[ks], dp);
       }
+
+      // dS^T = P^T (dP^T - Di), branchless mask
+      const int qg = q0w + col;
+      const bool need_mask =
+          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
+      float dsv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float e = __expf(st[r] * sl2e - lse_c);
+        float pr = e;
+        if (need_mask) {
+          const int key = kv0 + mfma32_d_row(lane, r);
+          const bool ok = key < skv && !(causal && key > qg + off);
+          pr = ok ? e : 0.f;
+        }
+        dsv[r] = pr * (dp[r] - di_c) * scale;
+      }
+
+      // pack dS^T to MFMA B-fragments (cvt_pk + permlane32_swap)
+      unsigned w[8];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+          unsigned a = pack_bf16(dsv[8 * i + 2 * g], dsv[8 * i + 2 * g + 1]);
+          unsigned bb = pack_bf16(dsv[8 * i + 2 * g + 4],
+                                  dsv[8 * i + 2 * g + 5]);
+          auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
+          w[i * 4 + g] = r2[0];
+          w[i * 4 + 2 + g] = r2[1];
+        }
+      }
+
+      // dQ^T += K^T dS
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+        union { unsigned u[4]; bf16x8 f; } db;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) db.u[g] = w[kh * 4 + g];
+#pragma unroll
+        for (int dt = 0; dt < ND; ++dt) {
+          bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+              kt_lds + (col + 32 * dt) * KTROW + kh * 16 + 8 * hi);
+          dq_acc[dt] = mfma32_bf16(ka, db.f, dq_acc[dt]);
+        }
+      }
+    }
+
+    __syncthreads();
+    if (have_next) {
+      stage_write();
+      __syncthreads();
     }
   }
 
+  // epilogue: dQ^T[d][q=lane]
+  const int qg = q0w + col;
+  if (qg < sq) {
+    __bf16* dqr = dq + q_base + (long)qg * q_stride;
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    const int rowg = q0w + mfma32_d_row(lane, i);
-    if (rowg < sq) {
-      __bf16* dqr = dq + q_base + (long)rowg * q_stride;
+    for (int dt = 0; dt < ND; ++dt)
 #pragma unroll
-      for (int dt = 0; dt < ND; ++dt)
-        dqr[dt * 32 + col] = (__bf16)dq_acc[dt][i];
-    }
+      for (int r = 0; r < 16; ++r)
+        dqr[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)dq_acc[dt][r];
   }
 }
 
@@ -657,15 +708,18 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
     }
 
     const int qg = qt0 + col;
-    const float lse_c = lse_lds[col];
+    const float lse_c = lse_lds[col];  // INF on padded q rows -> p = 0
     const float di_c = di_lds[col];
+    const bool need_mask = causal && (qt0 < k0w + KVB - off);
     float pt[16];
 #pragma unroll
     for (int i = 0; i < 16; ++i) {
-      const int keyg = k0w + mfma32_d_row(lane, i);
-      float p = 0.f;
-      if (qg < sq && keyg < skv && !(causal && keyg > qg + off))
-        p = __expf(st_acc[i] * scale - lse_c);
+      const float e = __expf(st_acc[i] * scale - lse_c);
+      float p = e;
+      if (need_mask) {
+        const int keyg = k0w + mfma32_d_row(lane, i);
+        p = (keyg < skv && keyg <= qg + off) ? e : 0.f;
+      }
       pt[i] = p;
       my_p[mfma32_d_row(lane, i) * PROW + col] = (__bf16)p;
     }
@@ -785,8 +839,8 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                __bf16* dk_exp, __bf16* dv_exp, int b, int sq,
                                int skv, int hq, int hkv, float scale,
                                bool causal, hipStream_t st) {
-  dim3 gq((sq + QB - 1) / QB, b * hq);
-  hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(256), 0, st, dout, q,
+  dim3 gq((sq + 255) / 256, b * hq);
+  hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout, q,
                      k, v, lse, di, dq, b, sq, skv, hq, hkv, scale, causal);
   dim3 gkv((skv + QB - 1) / QB, b * hq);
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(256), 0, st, dout,
