@@ -90,7 +90,7 @@ std::tuple<Tensor, Tensor> rmsnorm_bwd(const Tensor& dy, const Tensor& x,
   const int H = x.size(-1);
   const long n = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dw = at::empty({H}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
   const int G = norm_bwd_grid(n);
   auto dw_part = at::empty({G, H}, x.options().dtype(at::kFloat));
   auto inv = invrms.contiguous();
@@ -141,7 +141,7 @@ std::tuple<Tensor, Tensor, Tensor> layernorm_bwd(const Tensor& dy,
   const int H = x.size(-1);
   const long n = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dwdb = at::empty({2, H}, x.options().dtype(at::kFloat));
+  auto dwdb = at::zeros({2, H}, x.options().dtype(at::kFloat));
   const int G = norm_bwd_grid(n);
   auto part = at::empty({2, G, H}, x.options().dtype(at::kFloat));
   auto m = mean.contiguous(), r = invstd.contiguous();

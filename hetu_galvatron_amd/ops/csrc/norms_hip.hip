@@ -121,19 +121,25 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
 }
 
 // column-sum of per-block partials: out[c] = sum_g part[g][c]
-// one column per thread; consecutive threads read consecutive columns of
-// each partial row (coalesced), grid-stride over columns
+// grid.y parallelizes the g dimension (a single serial g-loop per column
+// measured 0.5 ms/call at G=2048); each slice commits one hardware fp32
+// atomic per element into the zero-initialized output.
 __global__ void colsum_kernel(const float* __restrict__ part,
                               float* __restrict__ out, int G, int H,
                               int n_out) {
+  const int g0 = blockIdx.y * ((G + gridDim.y - 1) / gridDim.y);
+  const int g1 = min(G, g0 + (int)((G + gridDim.y - 1) / gridDim.y));
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x;
        idx < (long)n_out * H; idx += (long)gridDim.x * blockDim.x) {
     const int o = idx / H;
     const int c = idx - (long)o * H;
     float acc = 0.f;
     const float* p = part + (long)o * G * H + c;
-    for (int g = 0; g < G; ++g) acc += p[(long)g * H];
-    out[idx] = acc;
+    for (int g = g0; g < g1; ++g) acc += p[(long)g * H];
+    if (gridDim.y == 1)
+      out[idx] = acc;
+    else
+      unsafeAtomicAdd(out + idx, acc);
   }
 }
 
@@ -296,8 +302,10 @@ void rmsnorm_bwd_launch_t(const T* dy, const T* x, const T* w,
                                   dim3(256), 0, s, dy, x, w, invrms, dx, dw_part, n, H)
   DISPATCH_VPT(H, RUN);
 #undef RUN
-  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((H + 255) / 256)),
-                     dim3(256), 0, s, dw_part, dw, grid, H, 1);
+  int gy = grid > 64 ? 32 : 1;
+  hipLaunchKernelGGL(colsum_kernel,
+                     dim3(galv_grid((H + 255) / 256), gy), dim3(256), 0, s,
+                     dw_part, dw, grid, H, 1);
 }
 
 template <typename T>
@@ -321,8 +329,10 @@ void layernorm_bwd_launch_t(const T* dy, const T* x, const T* w,
                                   dim3(256), 0, s, dy, x, w, mean, invstd, dx, dwdb_part, n, H)
   DISPATCH_VPT(H, RUN);
 #undef RUN
-  hipLaunchKernelGGL(colsum_kernel, dim3(galv_grid((2 * H + 255) / 256)),
-                     dim3(256), 0, s, dwdb_part, dwdb, grid, H, 2);
+  int gy = grid > 64 ? 32 : 1;
+  hipLaunchKernelGGL(colsum_kernel,
+                     dim3(galv_grid((2 * H + 255) / 256), gy), dim3(256), 0,
+                     s, dwdb_part, dwdb, grid, H, 2);
 }
 
 // explicit instantiations used by bindings.cpp
