@@ -1,0 +1,87 @@
+"""Strategy-space enumeration + filtering + inter-layer transition costs.
+
+Reference: galvatron/core/search_engine/search_engine.py:106-283
+(generate_strategy_list / filter_strategy_list) and
+dynamic_programming.py:161-210 (match_strategy redistribution penalty).
+"""
+from __future__ import annotations
+
+from dataclasses import replace
+from typing import List, Optional
+
+from ..config.schema import SearchArgs
+from ..config.strategy import LayerStrategy
+
+
+def _pow2_range(limit: int) -> List[int]:
+    out, v = [], 1
+    while v <= limit:
+        out.append(v)
+        v *= 2
+    return out
+
+
+def enumerate_strategies(world_size: int, args: SearchArgs,
+                         pp_deg: int) -> List[LayerStrategy]:
+    """All (tp|sp, dp_type, ckpt) combos for one pp degree.
+
+    Power-of-2 degrees; tp*dp = world/pp; ulysses-sp layers replace tp by
+    sp on the same degree (sp_space controls which are generated).
+    """
+    per_stage = world_size // pp_deg
+    out: List[LayerStrategy] = []
+    tp_degs = [t for t in _pow2_range(min(args.max_tp_deg, per_stage))]
+    if args.disable_tp:
+        tp_degs = [1]
+    for tsp in tp_degs:
+        dp = per_stage // tsp
+        if dp > 1 and args.disable_dp and tsp * pp_deg != world_size:
+            continue
+        modes = []
+        if tsp == 1:
+            modes = [("tp", tsp)]
+        else:
+            if args.sp_space in ("tp", "tp+sp"):
+                modes.append(("tp", tsp))
+            if args.sp_space in ("sp", "tp+sp") and not args.disable_sp:
+                modes.append(("sp", tsp))
+        for mode, deg in modes:
+            tp = deg if mode == "tp" else 1
+            sp = deg if mode == "sp" else 1
+            dp_types = ["ddp"]
+            if not args.disable_sdp and dp > 1:
+                dp_types.append("zero3")
+            for dpt in dp_types:
+                ckpts = [False] if args.disable_ckpt else [False, True]
+                for ck in ckpts:
+                    out.append(LayerStrategy(
+                        pp_deg=pp_deg, tp=tp, sp=sp, cp=1, dp=dp,
+                        dp_type="zero3" if dpt == "zero3" else "ddp",
+                        checkpoint=ck))
+    return out
+
+
+def strategy_key(s: LayerStrategy) -> str:
+    return (f"pp{s.pp_deg}_tp{s.tp}_sp{s.sp}_cp{s.cp}_dp{s.dp}_"
+            f"{s.dp_type}_ck{int(s.checkpoint)}")
+
+
+def transition_cost_mb(prev: LayerStrategy, cur: LayerStrategy,
+                       seq_len: int, hidden: int, local_bsz: float,
+                       mixed_precision: bool = True) -> float:
+    """Activation-redistribution penalty (MB moved) between two layouts.
+
+    Reference dynamic_programming.py:161-210 keys the penalty on layout
+    diffs (tp_sp/cp); here: if the (tp_sp, cp, use_sp) triple changes, the
+    boundary costs an allgather of the full activation on the old group +
+    a split — modelled as one activation's worth of bytes through the
+    slower of the two layouts.
+    """
+    if (prev.tp_sp, prev.cp, prev.use_ulysses) == \
+       (cur.tp_sp, cur.cp, cur.use_ulysses):
+        return 0.0
+    bytes_per = 2 if mixed_precision else 4
+    act_mb = local_bsz * seq_len * hidden * bytes_per / (1024 * 1024)
+    # allgather on the previous layout's group
+    g = max(prev.tp_sp * prev.cp, cur.tp_sp * cur.cp)
+    return act_mb * (g - 1) / g

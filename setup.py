@@ -13,7 +13,8 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import (  # noqa: E402
+    BuildExtension, CppExtension, CUDAExtension)
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 SRC = sorted(glob(os.path.join(HERE, "hetu_galvatron_amd/ops/csrc/*.hip")))
@@ -29,7 +30,12 @@ setup(
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950", "-std=c++17"],
             },
-        )
+        ),
+        CppExtension(
+            name="hetu_galvatron_amd._galvatron_dp_core",
+            sources=["hetu_galvatron_amd/csrc_cpu/dp_core.cpp"],
+            extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )
