@@ -1,0 +1,99 @@
+"""Training entry point (reference: galvatron/models/gpt/train_dist.py:76).
+
+  python -m torch.distributed.run --nnodes 1 --nproc-per-node N \\
+      --master-addr 127.0.0.1 -m hetu_galvatron_amd.cli.train \\
+      [cfg.yaml] [model.model_name=llama-3-8b train.train_iters=10 ...]
+
+Runs the hybrid-parallel train loop (GLOBAL-mode degrees or a searched
+plan via parallel.galvatron_config_path); with profile.profile=1 it writes
+computation/memory profiling JSONs and exits (the ModelProfiler's worker
+mode, reference model_profiler.py:215-420).
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+import torch
+import torch.distributed as dist
+
+
+def main(argv=None):
+    from ..config import load_config
+    from ..config.loader import config_from_cli
+    from ..core.initialize import initialize_galvatron
+    from ..profiler.runtime import RuntimeProfiler
+    from ..runtime import (GalvatronModel, get_optimizer_and_param_scheduler,
+                           get_train_iterator)
+
+    cfg = config_from_cli(argv)
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank) if use_gpu else torch.device("cpu")
+
+    if world > 1 or os.environ.get("RANK") is not None:
+        initialize_galvatron(cfg)
+    else:
+        torch.manual_seed(cfg.train.seed)
+    rank = dist.get_rank() if dist.is_initialized() else 0
+
+    model = GalvatronModel(cfg, device=device)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, device)
+    prof = RuntimeProfiler(enabled=use_gpu, device=device, rank=rank)
+
+    iters = cfg.train.train_iters
+    for i in range(iters):
+        prof.profile_memory("Before-Fwd")
+        prof.time_start()
+        opt.zero_grad()
+        ctx = next(it)
+        stats = model.forward_backward(ctx)
+        prof.profile_memory("After-Bwd")
+        norm = opt.step()
+        sched.step()
+        prof.profile_memory("After-step")
+        prof.time_end()
+        loss = model.global_loss(stats)
+        prof.log_iteration(loss, sched.get_lr(), norm,
+                           cfg.logging.log_interval)
+        if (cfg.ckpt.save and cfg.ckpt.save_interval
+                and (i + 1) % cfg.ckpt.save_interval == 0):
+            from ..runtime.checkpoint import save_distributed_checkpoint
+            save_distributed_checkpoint(model, opt, sched, cfg, i + 1)
+
+    # -- model-profiler worker mode ---------------------------------------
+    if cfg.profile.profile and rank == 0:
+        p = cfg.profile
+        L = cfg.model.num_hidden_layers
+        bsz = cfg.train.global_train_batch_size
+        seq = cfg.model.seq_length
+        key = f"layernum[{L}]_bsz{bsz}_seq{seq}"
+        os.makedirs(p.profile_dir, exist_ok=True)
+        prec = "bf16" if cfg.parallel.mixed_precision == "bf16" else "fp32"
+        name = cfg.model.model_name or "model"
+        if p.profile_type == "computation":
+            prof.save_time_profile(
+                os.path.join(p.profile_dir,
+                             f"computation_profiling_{prec}_{name}.json"),
+                key)
+        else:
+            pl = cfg.parallel
+            layout = (f"{pl.pp_deg}_{pl.global_tp_deg}_"
+                      f"{world // max(pl.pp_deg * pl.global_tp_deg, 1)}")
+            if pl.global_checkpoint:
+                layout += "_c"
+            prof.save_memory_profile(
+                os.path.join(p.profile_dir,
+                             f"memory_profiling_{prec}_{name}.json"),
+                f"{layout}/{key}_rank{rank}")
+    if dist.is_initialized():
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,75 @@
+"""Profiler CPU tests: post-processing math on fabricated raw profiles,
+hardware-bench script generation, runtime-profiler JSON plumbing."""
+import json
+import os
+
+import pytest
+
+from hetu_galvatron_amd.config import load_config
+from hetu_galvatron_amd.profiler.hardware import HardwareProfiler
+from hetu_galvatron_amd.profiler.model import ModelProfiler
+from hetu_galvatron_amd.search.engine import read_computation_profile
+
+
+def test_computation_differencing():
+    # fabricate: per-layer 4.4 ms/bsz + other 2.0 ms + 0.5 const
+    raw = {}
+    for ln in (2, 4):
+        for b in (1, 2, 4, 8):
+            raw[f"layernum[{ln}]_bsz{b}_seq4096"] = ln * (4.4 * b + 0.5) + 2.0 * b
+    m, c = read_computation_profile(raw, 4096)
+    assert abs(m - 4.4) < 1e-6 and abs(c - 0.5) < 1e-6
+
+
+def test_memory_postprocess(tmp_path):
+    cfg = load_config(base={"model": {"model_name": "tiny-llama"},
+                            "profile": {"profile_dir": str(tmp_path)}})
+    mp = ModelProfiler(cfg)
+    param, other_p = 840.0, 2100.0
+    act = {1: 520.0, 2: 260.0, 4: 130.0, 8: 65.0}
+    other_act = 600.0
+    bsz = 8
+    raw = {}
+    for ln in (1, 2):
+        for tp in (1, 2, 4, 8):
+            k = f"1_{tp}_{8 // tp}/layernum[{ln}]_bsz{bsz}_seq4096_rank0"
+            raw[k + "_ms"] = 4 * (ln * param / tp + other_p)
+            raw[k + "_act"] = ln * act[tp] * bsz + other_act * bsz
+            raw[k + "_act_peak"] = raw[k + "_act"] * 1.2
+        k = f"1_1_8_c/layernum[{ln}]_bsz{bsz}_seq4096_rank0"
+        raw[k + "_ms"] = 4 * (ln * param + other_p)
+        raw[k + "_act"] = ln * 34.0 * bsz + other_act * bsz
+        raw[k + "_act_peak"] = raw[k + "_act"]
+    parsed = mp.process_memory_data(raw, write=True)
+    lt = parsed["layertype_0"]
+    assert abs(lt["parameter_size"] - param) < 1e-6
+    assert abs(lt["tp_activation_per_bsz_dict"]["1"] - act[1]) < 1e-6
+    assert abs(lt["tp_activation_per_bsz_dict"]["4"] - act[4]) < 1e-6
+    assert abs(lt["tp_activation_per_bsz_dict"]["checkpoint"] - 34.0) < 1e-6
+    assert abs(parsed["other"]["parameter_size"] - other_p) < 1e-6
+    assert abs(parsed["other"]["tp_activation_per_bsz_dict"]["1"] - other_act) < 1e-6
+    assert os.path.exists(os.path.join(str(tmp_path),
+                                       "model_profile_bf16_tiny-llama.json"))
+
+
+def test_hardware_script_generation(tmp_path):
+    cfg = load_config(base={"model": {"model_name": "tiny-llama"}})
+    hp = HardwareProfiler(cfg)
+    paths = hp.generate_scripts(str(tmp_path))
+    assert len(paths) == 4
+    for p in paths:
+        body = open(p).read()
+        assert "torch.distributed.run" in body
+        assert "hardware_bench" in body
+        assert "--master-addr 127.0.0.1" in body
+
+
+def test_runtime_profiler_cpu_noop(tmp_path):
+    from hetu_galvatron_amd.profiler.runtime import RuntimeProfiler
+    rp = RuntimeProfiler(enabled=False)
+    rp.profile_memory("Before-Fwd")
+    rp.time_start()
+    assert rp.time_end() is None
+    rp.save_time_profile(os.path.join(tmp_path, "t.json"), "layernum[2]_bsz1_seq128")
+    d = json.load(open(os.path.join(tmp_path, "t.json")))
+    assert "layernum[2]_bsz1_seq128" in d
