@@ -1,0 +1,162 @@
+"""HuggingFace <-> canonical checkpoint layout conversion.
+
+Reference: galvatron/core/runtime/checkpoint/llama_adapter.py:30-163 /
+gpt_adapter.py:18-158 (QKV fusion with GQA interleave, gated-MLP stacking)
+and galvatron/tools/checkpoint_convert_h2g.py / _g2h.py.
+
+Canonical layout = this framework's world_size==1 state_dict (state.py).
+QKV fusion is KV-group-major: group g rows = [q_{g*rep..}, k_g, v_g]*d so
+a ColumnParallelLinear contiguous shard is whole groups.
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, Iterable, Optional
+
+import torch
+
+from ...config.schema import ModelArgs
+
+
+def fuse_qkv(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+             margs: ModelArgs) -> torch.Tensor:
+    """[hq*d, h], [hkv*d, h], [hkv*d, h] -> group-major fused [(hq+2hkv)*d, h].
+    Also fuses 1-D biases (last dim only)."""
+    d = margs.head_dim
+    hkv = margs.kv_heads
+    rep = margs.num_attention_heads // hkv
+    qg = q.reshape(hkv, rep * d, *q.shape[1:])
+    kg = k.reshape(hkv, d, *k.shape[1:])
+    vg = v.reshape(hkv, d, *v.shape[1:])
+    return torch.cat([qg, kg, vg], dim=1).reshape(-1, *q.shape[1:])
+
+
+def split_qkv(fused: torch.Tensor, margs: ModelArgs):
+    d = margs.head_dim
+    hkv = margs.kv_heads
+    rep = margs.num_attention_heads // hkv
+    g = fused.reshape(hkv, (rep + 2) * d, *fused.shape[1:])
+    q = g[:, :rep * d].reshape(-1, *fused.shape[1:])
+    k = g[:, rep * d:(rep + 1) * d].reshape(-1, *fused.shape[1:])
+    v = g[:, (rep + 1) * d:].reshape(-1, *fused.shape[1:])
+    return q, k, v
+
+
+def hf_llama_to_canonical(hf: Dict[str, torch.Tensor],
+                          margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    """HF LlamaForCausalLM keys -> canonical (reference h2g llama path)."""
+    out: Dict[str, torch.Tensor] = {}
+    out["embedding.word_embeddings.weight"] = hf["model.embed_tokens.weight"]
+    for i in range(margs.num_hidden_layers):
+        p = f"model.layers.{i}."
+        c = f"decoder.{i}."
+        out[c + "input_norm.weight"] = hf[p + "input_layernorm.weight"]
+        out[c + "post_attn_norm.weight"] = \
+            hf[p + "post_attention_layernorm.weight"]
+        out[c + "attention.linear_qkv.weight"] = fuse_qkv(
+            hf[p + "self_attn.q_proj.weight"],
+            hf[p + "self_attn.k_proj.weight"],
+            hf[p + "self_attn.v_proj.weight"], margs)
+        out[c + "attention.linear_proj.weight"] = \
+            hf[p + "self_attn.o_proj.weight"]
+        out[c + "mlp.fc1.weight"] = torch.cat(
+            [hf[p + "mlp.gate_proj.weight"], hf[p + "mlp.up_proj.weight"]], 0)
+        out[c + "mlp.fc2.weight"] = hf[p + "mlp.down_proj.weight"]
+    out["final_norm.norm.weight"] = hf["model.norm.weight"]
+    if "lm_head.weight" in hf:
+        out["lm_head.lm_head.weight"] = hf["lm_head.weight"]
+    else:  # tied embeddings
+        out["lm_head.lm_head.weight"] = hf["model.embed_tokens.weight"]
+    return out
+
+
+def canonical_to_hf_llama(can: Dict[str, torch.Tensor],
+                          margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    out: Dict[str, torch.Tensor] = {}
+    out["model.embed_tokens.weight"] = can["embedding.word_embeddings.weight"]
+    for i in range(margs.num_hidden_layers):
+        p = f"model.layers.{i}."
+        c = f"decoder.{i}."
+        out[p + "input_layernorm.weight"] = can[c + "input_norm.weight"]
+        out[p + "post_attention_layernorm.weight"] = \
+            can[c + "post_attn_norm.weight"]
+        q, k, v = split_qkv(can[c + "attention.linear_qkv.weight"], margs)
+        out[p + "self_attn.q_proj.weight"] = q
+        out[p + "self_attn.k_proj.weight"] = k
+        out[p + "self_attn.v_proj.weight"] = v
+        out[p + "self_attn.o_proj.weight"] = \
+            can[c + "attention.linear_proj.weight"]
+        fc1 = can[c + "mlp.fc1.weight"]
+        F = fc1.shape[0] // 2
+        out[p + "mlp.gate_proj.weight"] = fc1[:F]
+        out[p + "mlp.up_proj.weight"] = fc1[F:]
+        out[p + "mlp.down_proj.weight"] = can[c + "mlp.fc2.weight"]
+    out["model.norm.weight"] = can["final_norm.norm.weight"]
+    out["lm_head.weight"] = can["lm_head.lm_head.weight"]
+    return out
+
+
+def hf_gpt2_to_canonical(hf: Dict[str, torch.Tensor],
+                         margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    """HF GPT2LMHeadModel (Conv1D: stored transposed) -> canonical."""
+    out: Dict[str, torch.Tensor] = {}
+    out["embedding.word_embeddings.weight"] = hf["transformer.wte.weight"]
+    out["embedding.position_embeddings.weight"] = hf["transformer.wpe.weight"]
+    for i in range(margs.num_hidden_layers):
+        p = f"transformer.h.{i}."
+        c = f"decoder.{i}."
+        out[c + "input_norm.weight"] = hf[p + "ln_1.weight"]
+        out[c + "input_norm.bias"] = hf[p + "ln_1.bias"]
+        out[c + "post_attn_norm.weight"] = hf[p + "ln_2.weight"]
+        out[c + "post_attn_norm.bias"] = hf[p + "ln_2.bias"]
+        w = hf[p + "attn.c_attn.weight"].t().contiguous()  # [3h, h]
+        b = hf[p + "attn.c_attn.bias"]
+        h = margs.hidden_size
+        out[c + "attention.linear_qkv.weight"] = fuse_qkv(
+            w[:h], w[h:2 * h], w[2 * h:], margs)
+        out[c + "attention.linear_qkv.bias"] = fuse_qkv(
+            b[:h], b[h:2 * h], b[2 * h:], margs)
+        out[c + "attention.linear_proj.weight"] = \
+            hf[p + "attn.c_proj.weight"].t().contiguous()
+        out[c + "attention.linear_proj.bias"] = hf[p + "attn.c_proj.bias"]
+        out[c + "mlp.fc1.weight"] = hf[p + "mlp.c_fc.weight"].t().contiguous()
+        out[c + "mlp.fc1.bias"] = hf[p + "mlp.c_fc.bias"]
+        out[c + "mlp.fc2.weight"] = hf[p + "mlp.c_proj.weight"].t().contiguous()
+        out[c + "mlp.fc2.bias"] = hf[p + "mlp.c_proj.bias"]
+    out["final_norm.norm.weight"] = hf["transformer.ln_f.weight"]
+    out["final_norm.norm.bias"] = hf["transformer.ln_f.bias"]
+    out["lm_head.lm_head.weight"] = hf["transformer.wte.weight"]
+    return out
+
+
+def load_hf_checkpoint(path: str) -> Dict[str, torch.Tensor]:
+    """Load an HF checkpoint dir (safetensors shards or pytorch_model.bin)."""
+    state: Dict[str, torch.Tensor] = {}
+    st_files = sorted(f for f in os.listdir(path)
+                      if f.endswith(".safetensors"))
+    if st_files:
+        from safetensors.torch import load_file
+        for f in st_files:
+            state.update(load_file(os.path.join(path, f)))
+        return state
+    bins = sorted(f for f in os.listdir(path)
+                  if f.startswith("pytorch_model") and f.endswith(".bin"))
+    for f in bins:
+        state.update(torch.load(os.path.join(path, f), map_location="cpu",
+                                weights_only=True))
+    return state
+
+
+def save_hf_checkpoint(state: Dict[str, torch.Tensor], path: str) -> None:
+    from safetensors.torch import save_file
+    os.makedirs(path, exist_ok=True)
+    save_file({k: v.contiguous() for k, v in state.items()},
+              os.path.join(path, "model.safetensors"))
+
+
+def hf_to_canonical(hf: Dict[str, torch.Tensor], margs: ModelArgs
+                    ) -> Dict[str, torch.Tensor]:
+    if margs.model_type == "gpt":
+        return hf_gpt2_to_canonical(hf, margs)
+    return hf_llama_to_canonical(hf, margs)

@@ -1,0 +1,90 @@
+"""Checkpoint tests: HF<->canonical round trip + loss-equivalence, and
+distributed save/resume continuity (world 2, gloo)."""
+import os
+
+import pytest
+import torch
+
+from hetu_galvatron_amd.config import load_config
+from hetu_galvatron_amd.runtime import GalvatronModel
+from hetu_galvatron_amd.runtime.checkpoint import (
+    canonical_state_from_stage, hf_to_canonical, load_full_state)
+from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+    canonical_to_hf_llama, fuse_qkv, split_qkv)
+
+
+def tiny_cfg(**model_extra):
+    return load_config(base={
+        "model": dict({"model_name": "tiny-llama"}, **model_extra),
+        "train": {"global_train_batch_size": 4, "train_iters": 2,
+                  "lr": 1e-3, "lr_decay_style": "constant",
+                  "distributed_backend": "gloo"},
+    })
+
+
+def test_qkv_fuse_roundtrip():
+    cfg = tiny_cfg()
+    m = cfg.model
+    q = torch.randn(m.num_attention_heads * m.head_dim, m.hidden_size)
+    k = torch.randn(m.kv_heads * m.head_dim, m.hidden_size)
+    v = torch.randn(m.kv_heads * m.head_dim, m.hidden_size)
+    fused = fuse_qkv(q, k, v, m)
+    q2, k2, v2 = split_qkv(fused, m)
+    assert torch.equal(q, q2) and torch.equal(k, k2) and torch.equal(v, v2)
+
+
+def test_hf_llama_roundtrip_and_load():
+    cfg = tiny_cfg()
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    can = canonical_state_from_stage(model.stage_model)
+    hf = canonical_to_hf_llama(can, cfg.model)
+    assert "model.layers.0.self_attn.q_proj.weight" in hf
+    can2 = hf_to_canonical(hf, cfg.model)
+    for k in can:
+        assert torch.equal(can[k], can2[k]), k
+    # loading the round-tripped state reproduces the loss exactly
+    from hetu_galvatron_amd.runtime import get_train_iterator
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    ctx = next(it)
+    with torch.no_grad():
+        base = model.forward_backward.__self__  # engine; run fwd via step
+    s1 = model.forward_backward(ctx)
+    model2 = GalvatronModel(cfg)
+    load_full_state(model2.stage_model, can2, cfg.model)
+    s2 = model2.forward_backward(ctx)
+    assert abs(s1.loss - s2.loss) < 1e-5
+
+
+def test_distributed_save_resume_single(tmp_path):
+    """Save at iter k, keep training -> losses equal a run resumed from k."""
+    from hetu_galvatron_amd.runtime import (
+        get_optimizer_and_param_scheduler, get_train_iterator)
+    from hetu_galvatron_amd.runtime.checkpoint import (
+        load_distributed_checkpoint, save_distributed_checkpoint)
+
+    cfg = tiny_cfg()
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    batches = [next(it) for _ in range(6)]
+    for i in range(3):
+        opt.zero_grad(); model.forward_backward(batches[i]); opt.step(); sched.step()
+    save_distributed_checkpoint(model, opt, sched, cfg, 3, str(tmp_path))
+    cont = []
+    for i in range(3, 6):
+        opt.zero_grad(); st = model.forward_backward(batches[i]); opt.step(); sched.step()
+        cont.append(st.loss)
+
+    torch.manual_seed(123)  # different init; checkpoint must restore it all
+    model2 = GalvatronModel(cfg)
+    opt2, sched2 = get_optimizer_and_param_scheduler(model2.stage_model, cfg)
+    it2 = load_distributed_checkpoint(model2, opt2, sched2, cfg, str(tmp_path))
+    assert it2 == 3
+    resumed = []
+    for i in range(3, 6):
+        opt2.zero_grad(); st = model2.forward_backward(batches[i]); opt2.step(); sched2.step()
+        resumed.append(st.loss)
+    for a, b in zip(cont, resumed):
+        assert abs(a - b) < 1e-6, (cont, resumed)
