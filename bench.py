@@ -31,7 +31,7 @@ def main():
     ap.add_argument("--batch-per-gpu", type=int, default=8,
                     help="sequences per GPU per step (weak scaling)")
     ap.add_argument("--chunks", type=int, default=0,
-                    help="pipeline microbatches (0 = batch-per-gpu)")
+                    help="pipeline microbatches (0 = auto: max(2, 2*pp))")
     ap.add_argument("--pp", type=int, default=1)
     ap.add_argument("--tp", type=int, default=1)
     ap.add_argument("--dp-type", type=str, default="zero2",
@@ -60,7 +60,7 @@ def main():
     from hetu_galvatron_amd.runtime.galvatron_model import resolve_plan
 
     global_batch = args.batch_per_gpu * max(world, 1)
-    chunks = args.chunks or args.batch_per_gpu
+    chunks = args.chunks or max(2, 2 * args.pp)
     cfg = load_config(base={
         "model": {"model_name": args.model, "seq_length": args.seq_len},
         "train": {"global_train_batch_size": global_batch,
