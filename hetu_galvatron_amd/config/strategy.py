@@ -99,6 +99,7 @@ class HybridParallelPlan:
     vsp: int = 0            # vocab ulysses-sp flag
     vcp: int = 1            # vocab cp degree
     embed_sdp: int = 0      # 1 => zero3 for embedding/head
+    ep_deg: int = 1         # expert-parallel degree (uniform, MoE layers)
 
     @property
     def num_layers(self) -> int:
@@ -117,6 +118,7 @@ class HybridParallelPlan:
             pp_deg=self.pp_deg, tp=tp, sp=sp, cp=cp, dp=dp, dp_type=dp_type,
             checkpoint=bool(self.checkpoint_flags[i]),
             tp_consecutive=bool(self.tp_consecutive_flags[i]),
+            ep=self.ep_deg,
         )
 
     def vocab_strategy(self, world_size: int) -> LayerStrategy:
@@ -157,6 +159,7 @@ class HybridParallelPlan:
             vsp=int(cfg.get("vsp", 0)),
             vcp=int(cfg.get("vcp", 1)),
             embed_sdp=int(cfg.get("embed_sdp", 0)),
+            ep_deg=int(cfg.get("ep_deg", 1)),
         )
         return plan
 
@@ -177,6 +180,7 @@ class HybridParallelPlan:
             "default_dp_type": self.default_dp_type,
             "vtp": self.vtp,
             "vsp": self.vsp,
+            "ep_deg": self.ep_deg,
             "vcp": self.vcp,
             "embed_sdp": self.embed_sdp,
         }
@@ -196,7 +200,7 @@ class HybridParallelPlan:
                 cp: int = 1, use_sp: bool = False, dp_type: str = "ddp",
                 checkpoint: bool = False, chunks: int = 1, global_bsz: int = 8,
                 pipeline_type: str = "pipedream_flush", vtp: Optional[int] = None,
-                vsp: Optional[bool] = None) -> "HybridParallelPlan":
+                vsp: Optional[bool] = None, ep: int = 1) -> "HybridParallelPlan":
         """GLOBAL-mode plan: same strategy on every layer
         (reference: hybrid_parallel_config.py GLOBAL mode)."""
         assert world_size % (pp * tp * cp) == 0, \
@@ -215,7 +219,7 @@ class HybridParallelPlan:
             default_dp_type=dp_type if dp_type != "zero3" else "zero2",
             vtp=tp if vtp is None else vtp,
             vsp=int(use_sp if vsp is None else vsp),
-            vcp=1, embed_sdp=zero3,
+            vcp=1, embed_sdp=zero3, ep_deg=ep,
         )
 
     def validate(self, world_size: int) -> None:

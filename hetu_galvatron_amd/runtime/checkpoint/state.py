@@ -89,6 +89,28 @@ def shard_for_rank(name: str, full: torch.Tensor, strategy, tp_rank: int,
     return full
 
 
+def _expert_resolve(name: str, full_key: str, state: Dict[str, torch.Tensor],
+                    ep_rank: int, ep: int, num_experts: int):
+    """Map an EP-local expert param name to its canonical tensor slice.
+    Grouped weights (experts.w1/w2 [E, ...]): slice dim 0 by ep rank.
+    Sequential (experts.fc1.{e}.weight): renumber local->global index."""
+    if ".experts." not in name or ep <= 1:
+        return state.get(full_key)
+    e_local = num_experts // ep
+    parts = name.split(".")
+    i = parts.index("experts")
+    if parts[i + 1] in ("w1", "w2"):
+        full = state.get(full_key)
+        if full is None:
+            return None
+        return full[ep_rank * e_local:(ep_rank + 1) * e_local]
+    # sequential: experts.fc1.{e}.weight
+    g = int(parts[i + 2]) + ep_rank * e_local
+    parts[i + 2] = str(g)
+    key = full_key.rsplit(name, 1)[0] + ".".join(parts)
+    return state.get(key)
+
+
 def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
                     margs: ModelArgs) -> None:
     """Load a canonical full state into this rank's sharded blocks
@@ -99,12 +121,28 @@ def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
         s = blk.groups.strategy
         tp_size = group_size(blk.groups.tp_group) if not s.use_ulysses else 1
         tp_rank = group_rank(blk.groups.tp_group) if not s.use_ulysses else 0
+        epg = getattr(blk.groups, "ep_group", None)
+        ep = epg.size if epg is not None else 1
+        ep_rank = group_rank(epg) if epg is not None else 0
         if blk.flat is not None:
             blk.flat.gather_params()
+        if getattr(blk, "flat_expert", None) is not None:
+            blk.flat_expert.gather_params()
         with torch.no_grad():
             own = blk.inner.state_dict()
             for name, p in own.items():
                 full_key = f"{prefix}.{name}"
+                if ".experts." in name:
+                    full = _expert_resolve(name, full_key, state, ep_rank,
+                                           ep, margs.num_experts)
+                    if full is None:
+                        continue
+                    if full.shape != p.shape:
+                        raise ValueError(
+                            f"{full_key}: expert shard {tuple(full.shape)} "
+                            f"vs param {tuple(p.shape)} (ep={ep})")
+                    p.copy_(full.to(p.dtype).to(p.device))
+                    continue
                 if full_key not in state:
                     continue
                 shard = shard_for_rank(full_key, state[full_key], s, tp_rank,
@@ -116,3 +154,5 @@ def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
                 p.copy_(shard.to(p.dtype).to(p.device))
         if blk.flat is not None:
             blk.flat.refresh_from_params()
+        if getattr(blk, "flat_expert", None) is not None:
+            blk.flat_expert.refresh_from_params()

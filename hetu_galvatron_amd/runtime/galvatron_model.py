@@ -31,7 +31,7 @@ def resolve_plan(cfg: GalvatronConfig, world_size: int) -> HybridParallelPlan:
             checkpoint=bool(p.global_checkpoint),
             chunks=max(p.chunks, 1), global_bsz=cfg.train.global_train_batch_size,
             pipeline_type=p.pipeline_type, vtp=p.vocab_tp,
-            vsp=bool(p.vocab_sp))
+            vsp=bool(p.vocab_sp), ep=p.global_ep_deg)
     if plan.global_bsz != cfg.train.global_train_batch_size:
         plan.global_bsz = cfg.train.global_train_batch_size
     return plan

@@ -27,6 +27,7 @@ from .modules import (
     GalvatronCausalLMHead, GalvatronDecoderLayer, GalvatronEmbedding,
     GalvatronFinalNorm,
 )
+from .moe_modules import GalvatronMoEDecoderLayer
 
 
 def build_causal_lm_arch(num_layers: int) -> List[str]:
@@ -49,11 +50,19 @@ class LayerBlock(nn.Module):
         self.checkpoint = checkpoint
         self.prev_groups = prev_groups
         self.flat: Optional[FlatParamBlock] = None
+        self.flat_expert: Optional[FlatParamBlock] = None
 
     def setup_zero(self, mode: str, param_dtype: torch.dtype, device) -> None:
         tp_group = None
         if not self.groups.strategy.use_ulysses:
             tp_group = self.groups.tp_group
+        # MoE layers: expert params reduce over EDP (dp-of-experts), the rest
+        # over the sdp group (reference parallel.py MoE double-wrap)
+        if self.groups.ep_group is not None and self.groups.ep_group.size > 1:
+            self.flat_expert = FlatParamBlock(
+                self.inner, mode, self.groups.edp_group, tp_group=None,
+                param_dtype=param_dtype, device=device,
+                param_filter=lambda p: getattr(p, "expert_parallel", False))
         self.flat = FlatParamBlock(self.inner, mode, self.groups.sdp_group,
                                    tp_group=tp_group, param_dtype=param_dtype,
                                    device=device)
@@ -68,13 +77,21 @@ class LayerBlock(nn.Module):
             x = redistribute(x, self.prev_groups, self.groups, ctx["batch_size"])
         if self.flat is not None:
             self.flat.gather_params()
+        if self.flat_expert is not None:
+            self.flat_expert.gather_params()
         if x is not None and self.flat is not None and x.requires_grad:
             x = _PostBackwardHook.apply(self.flat, x)
+        if x is not None and self.flat_expert is not None and x.requires_grad:
+            x = _PostBackwardHook.apply(self.flat_expert, x)
         if self.checkpoint and self.kind == "decoder" and torch.is_grad_enabled():
             out = torch_ckpt.checkpoint(
                 lambda t: self._inner_forward(t, ctx), x, use_reentrant=False)
         else:
             out = self._inner_forward(x, ctx)
+        if self.flat_expert is not None:
+            if out.requires_grad:
+                out = _PreBackwardGather.apply(self.flat_expert, out)
+            self.flat_expert.reshard_params()
         if self.flat is not None:
             if out.requires_grad:
                 out = _PreBackwardGather.apply(self.flat, out)
@@ -86,6 +103,8 @@ class LayerBlock(nn.Module):
         sentinel; zero3 reshard + auto-sync kick happen here)."""
         if self.flat is not None:
             self.flat.post_backward()
+        if self.flat_expert is not None:
+            self.flat_expert.post_backward()
 
 
 @dataclass
@@ -168,9 +187,15 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         prev_strategy = strategies[lo - 1]
         prev = gen_layer_comm_groups([prev_strategy], world, rank, cache)[0][0]
 
+    is_moe = margs.model_type.startswith("moe") and margs.num_experts > 0
     for i in range(lo, hi):
         lg = layer_groups[i]
-        dec = GalvatronDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
+        if is_moe:
+            assert lg.strategy.tp == 1 or lg.strategy.use_ulysses, \
+                "MoE layers: expert-TP (etp) not supported yet; use ep/dp/sp"
+            dec = GalvatronMoEDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
+        else:
+            dec = GalvatronDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
         blk = LayerBlock(dec, lg, "decoder",
                          checkpoint=bool(plan.checkpoint_flags[i]),
                          prev_groups=prev)

@@ -1,0 +1,130 @@
+"""MoE layer modules (mixtral-style).
+
+Reference: galvatron/core/runtime/models/moe_modules.py:19-155
+(GalvatronMoEAttention / Router / MLP / DecoderLayer).  Here one module:
+pre-norm attention + pre-norm routed-MLP with residuals; the MoE MLP =
+TopKRouter -> dispatcher (alltoall|allgather over the layer's EP group)
+-> Grouped/Sequential experts -> optional shared expert; router aux loss
+injected through MoEAuxLossAutoScaler.
+
+v1 constraint: MoE layers run with tp==1 (ulysses/cp/dp/ep compose);
+expert-TP (etp) is future work — the builder asserts accordingly.
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ...config.schema import ModelArgs
+from ..moe.dispatcher import AllGatherDispatcher, AlltoAllDispatcher
+from ..moe.experts import GroupedMLP, SequentialMLP, SharedExpertMLP
+from ..moe.router import TopKRouter, attach_aux_loss
+from ..transformer import RotaryEmbedding
+from ..transformer.attention import SelfAttention
+from ..transformer.norm import build_norm
+from .modules import _my_rank, _tag_tp_replicated
+
+
+class GalvatronMoEMLP(nn.Module):
+    def __init__(self, margs: ModelArgs, groups, dtype=None):
+        super().__init__()
+        self.margs = margs
+        self.topk = margs.moe_router_topk
+        self.router = TopKRouter(margs, dtype=dtype)
+        ep_group = groups.ep_group
+        ep = ep_group.size if ep_group is not None else 1
+        assert margs.num_experts % max(ep, 1) == 0
+        if margs.moe_token_dispatcher_type == "allgather":
+            self.dispatcher = AllGatherDispatcher(ep_group, margs.num_experts)
+        else:
+            self.dispatcher = AlltoAllDispatcher(ep_group, margs.num_experts)
+        n_local = margs.num_experts // max(ep, 1)
+        ffn = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
+        gated = margs.hidden_act in ("silu", "swiglu", "geglu")
+        if margs.moe_grouped_gemm:
+            self.experts = GroupedMLP(n_local, margs.hidden_size, ffn,
+                                      dtype=dtype, gated=gated)
+        else:
+            self.experts = SequentialMLP(n_local, margs.hidden_size, ffn,
+                                         dtype=dtype, gated=gated)
+        if margs.moe_shared_expert_intermediate_size:
+            self.shared = SharedExpertMLP(
+                margs.hidden_size, margs.moe_shared_expert_intermediate_size,
+                dtype=dtype)
+        else:
+            self.shared = None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """x: [s_local, b, h] SBH."""
+        s, b, h = x.shape
+        flat = x.reshape(-1, h)
+        probs, idx, aux = self.router(flat)
+        expert_in, tokens_per_expert = self.dispatcher.dispatch(
+            flat, probs, idx)
+        expert_out = self.experts(expert_in, tokens_per_expert)
+        merged = self.dispatcher.combine(expert_out, flat.shape[0], self.topk)
+        if self.shared is not None:
+            merged = merged + self.shared(flat)
+        if self.training and aux.requires_grad:
+            merged = attach_aux_loss(merged, aux)
+        return merged.reshape(s, b, h)
+
+
+class GalvatronMoEDecoderLayer(nn.Module):
+    """Pre-norm attention + pre-norm routed MLP
+    (reference moe_modules.py:131)."""
+
+    def __init__(self, margs: ModelArgs, groups, layer_idx: int = 0,
+                 dtype=None):
+        super().__init__()
+        self.margs = margs
+        self.groups = groups
+        self.layer_idx = layer_idx
+        s = groups.strategy
+        self.strategy = s
+        seq_par = not s.use_ulysses
+        self.input_norm = build_norm(margs.normalization, margs.hidden_size,
+                                     margs.norm_epsilon, dtype)
+        self.post_attn_norm = build_norm(margs.normalization,
+                                         margs.hidden_size,
+                                         margs.norm_epsilon, dtype)
+        self.attention = SelfAttention(
+            margs, groups.tp_group, groups.sp_group, groups.cp_group,
+            use_ulysses=s.use_ulysses, sequence_parallel=seq_par, dtype=dtype)
+        self.mlp = GalvatronMoEMLP(margs, groups, dtype=dtype)
+        if margs.position_embedding_type == "rope":
+            self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta)
+        else:
+            self.rotary = None
+        self.dropout_p = margs.hidden_dropout
+
+    def _rope_tables(self, S: int, device):
+        if self.rotary is None:
+            return None, None
+        s = self.strategy
+        c = self.groups.coord_of(_my_rank())
+        if s.use_ulysses:
+            return self.rotary.get_for_rank(S, device, sp_rank=c.tp_idx,
+                                            sp_size=s.tp_sp, cp_rank=c.cp_idx,
+                                            cp_size=s.cp)
+        return self.rotary.get_for_rank(S, device, sp_rank=0, sp_size=1,
+                                        cp_rank=c.cp_idx, cp_size=s.cp)
+
+    def forward(self, hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
+        S = ctx["seq_len"]
+        cos, sin = self._rope_tables(S, hidden.device)
+        residual = hidden
+        x = self.input_norm(hidden)
+        x = self.attention(x, cos, sin)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        hidden = residual + x
+        residual = hidden
+        x = self.post_attn_norm(hidden)
+        x = self.mlp(x)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        return residual + x

@@ -123,7 +123,12 @@ def get_optimizer_and_param_scheduler(stage_model, cfg):
     """reference: optimizer/utils.py:44."""
     from .scheduler import OptimizerParamScheduler
 
-    blocks = [blk.flat for blk in stage_model.blocks if blk.flat is not None]
+    blocks = []
+    for blk in stage_model.blocks:
+        if getattr(blk, "flat_expert", None) is not None:
+            blocks.append(blk.flat_expert)
+        if blk.flat is not None:
+            blocks.append(blk.flat)
     t = cfg.train
     opt = GalvatronOptimizer(
         blocks, lr=t.lr, betas=(t.adam_beta1, t.adam_beta2), eps=t.adam_eps,

@@ -96,19 +96,22 @@ class PipelineEngine:
 
     def _set_auto_sync(self, flag: bool) -> None:
         for blk in self.sm.blocks:
-            if blk.flat is not None:
-                blk.flat.auto_sync = flag and self.overlap
+            for f in (blk.flat, getattr(blk, "flat_expert", None)):
+                if f is not None:
+                    f.auto_sync = flag and self.overlap
 
     def _finalize_grads(self) -> None:
         for blk in self.sm.blocks:
             blk.finalize_backward()
         self._sync_tied_embeddings_pre()
         for blk in self.sm.blocks:
-            if blk.flat is not None:
-                blk.flat.start_grad_sync()
+            for f in (blk.flat, getattr(blk, "flat_expert", None)):
+                if f is not None:
+                    f.start_grad_sync()
         for blk in self.sm.blocks:
-            if blk.flat is not None:
-                blk.flat.finish_grad_sync()
+            for f in (blk.flat, getattr(blk, "flat_expert", None)):
+                if f is not None:
+                    f.finish_grad_sync()
 
     def _sync_tied_embeddings_pre(self) -> None:
         """pp>1 tied embedding/lm-head: sum the tied segments' raw grads over

@@ -66,7 +66,8 @@ class FlatParamBlock:
                  tp_group: Optional[CommGroup] = None,
                  param_dtype: torch.dtype = torch.bfloat16,
                  device: Optional[torch.device] = None,
-                 owner_filter: bool = True):
+                 owner_filter: bool = True,
+                 param_filter=None):
         assert mode in ("ddp", "zero2", "zero3")
         self.module = module
         self.mode = mode
@@ -90,6 +91,8 @@ class FlatParamBlock:
                 continue
             if owner_filter and getattr(p, "_galvatron_owner", None) is not None:
                 continue  # tied param owned by another block
+            if param_filter is not None and not param_filter(p):
+                continue
             p._galvatron_owner = self
             params.append(p)
         if device is None:

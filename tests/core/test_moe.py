@@ -1,0 +1,141 @@
+"""MoE correctness: router invariants + EP distributed runs vs 1-process
+baseline (reference test style: tests/core/test_ep.py:245)."""
+import os
+
+import pytest
+import torch
+
+from hetu_galvatron_amd.config import HybridParallelPlan, load_config
+
+STEPS = 3
+TOL = 0.03
+
+BASE = {
+    "model": {"model_name": "tiny-moe"},
+    "train": {"global_train_batch_size": 4, "train_iters": STEPS, "lr": 1e-3,
+              "lr_decay_style": "constant", "distributed_backend": "gloo"},
+}
+
+
+def make_cfg(extra=None):
+    import copy
+    base = copy.deepcopy(BASE)
+    for k, v in (extra or {}).items():
+        base.setdefault(k, {}).update(v)
+    return load_config(base=base)
+
+
+def train_steps(model, cfg, steps=STEPS):
+    from hetu_galvatron_amd.runtime import (
+        get_optimizer_and_param_scheduler, get_train_iterator)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(model.cfg, torch.device("cpu"))
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        stats = model.forward_backward(next(it))
+        opt.step()
+        sched.step()
+        losses.append(model.global_loss(stats))
+    return losses
+
+
+def test_router_topk_and_aux():
+    cfg = make_cfg()
+    from hetu_galvatron_amd.runtime.moe.router import TopKRouter
+    torch.manual_seed(0)
+    r = TopKRouter(cfg.model)
+    r.train()
+    x = torch.randn(64, cfg.model.hidden_size)
+    probs, idx, aux = r(x)
+    assert probs.shape == (64, cfg.model.moe_router_topk)
+    assert torch.allclose(probs.float().sum(-1), torch.ones(64), atol=1e-3)
+    assert idx.max() < cfg.model.num_experts
+    assert float(aux) > 0  # load-balancing loss active in training
+
+
+def test_dispatcher_roundtrip_identity_experts():
+    """dispatch -> identity experts -> combine == prob-weighted passthrough."""
+    from hetu_galvatron_amd.runtime.moe.dispatcher import AlltoAllDispatcher
+    torch.manual_seed(1)
+    n, h, E, k = 32, 16, 4, 2
+    x = torch.randn(n, h)
+    probs = torch.softmax(torch.randn(n, k), dim=-1)
+    idx = torch.randint(0, E, (n, k))
+    d = AlltoAllDispatcher(None, E)
+
+    class G:  # fake group of size 1
+        size = 1
+    d.ep_group = None
+    inp, counts = d.dispatch(x, probs, idx)
+    assert int(counts.sum()) == n * k
+    out = d.combine(inp, n, k)
+    want = x * probs.sum(-1, keepdim=True)
+    assert torch.allclose(out, want, atol=1e-5)
+
+
+def _moe_worker(rank, world, plan_dict, state_path, cfg_extra):
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+
+    cfg = make_cfg(cfg_extra)
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.from_config_dict(plan_dict)
+    model = GalvatronModel(cfg, plan)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    return train_steps(model, cfg)
+
+
+_BASELINE = {}
+
+
+def get_baseline(tmp_dir="/tmp/galvatron_moe_test"):
+    if "v" not in _BASELINE:
+        from hetu_galvatron_amd.runtime import GalvatronModel
+        from hetu_galvatron_amd.runtime.checkpoint.state import (
+            canonical_state_from_stage)
+        os.makedirs(tmp_dir, exist_ok=True)
+        cfg = make_cfg()
+        torch.manual_seed(0)
+        model = GalvatronModel(cfg)
+        state = canonical_state_from_stage(model.stage_model)
+        path = os.path.join(tmp_dir, "tiny_moe_state.pt")
+        torch.save(state, path)
+        _BASELINE["v"] = (train_steps(model, cfg), path)
+    return _BASELINE["v"]
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("ep,dispatcher", [(2, "alltoall"), (2, "allgather"),
+                                           (1, "alltoall")])
+def test_moe_ep_vs_baseline(ep, dispatcher):
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=2, pp=1, tp=1, dp_type="ddp",
+        global_bsz=4, chunks=1, ep=ep)
+    res = run_distributed(
+        _moe_worker, world_size=2,
+        args=(plan.to_config_dict(), state_path,
+              {"model": {"moe_token_dispatcher_type": dispatcher}}))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: dist {a:.4f} vs base {b:.4f} " \
+                f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_moe_ep_zero3():
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=2, pp=1, tp=1, dp_type="zero3",
+        global_bsz=4, chunks=1, ep=2)
+    res = run_distributed(_moe_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for losses in res:
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, (losses, base_losses)
