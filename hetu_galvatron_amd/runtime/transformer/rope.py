@@ -83,5 +83,8 @@ def zigzag_unslice_index(cp_size: int) -> list:
 
 def apply_rope_qk(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
                   sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """q,k: [s, b, h, d] SBH-heads layout."""
+    """q,k: [s, b, h, d] SBH-heads layout. cos=None => no RoPE (learned
+    positions, gpt family)."""
+    if cos is None:
+        return q, k
     return apply_rope(q, cos, sin), apply_rope(k, cos, sin)
