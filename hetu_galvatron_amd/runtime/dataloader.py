@@ -55,9 +55,17 @@ def get_train_iterator(cfg: GalvatronConfig, device,
     """Yield batch contexts of the global batch size, cycling the dataset
     (reference: dataloader.py:462 get_train_valid_test_data_iterators)."""
     B = global_batch or cfg.train.global_train_batch_size
-    ds = SyntheticCausalLMDataset(
-        cfg.model.vocab_size, cfg.model.seq_length,
-        size=max(cfg.data.synthetic_dataset_size, B), seed=cfg.train.seed)
+    if cfg.data.dataset == "megatron" and cfg.data.data_path:
+        from .datasets import build_pretraining_dataset
+        ds = build_pretraining_dataset(
+            cfg.data.data_path, cfg.model.seq_length,
+            num_samples=max(B * max(cfg.train.train_iters, 1),
+                            cfg.data.synthetic_dataset_size),
+            seed=cfg.train.seed)
+    else:
+        ds = SyntheticCausalLMDataset(
+            cfg.model.vocab_size, cfg.model.seq_length,
+            size=max(cfg.data.synthetic_dataset_size, B), seed=cfg.train.seed)
     idx = 0
     while True:
         batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])

@@ -36,6 +36,11 @@ setup(
             sources=["hetu_galvatron_amd/csrc_cpu/dp_core.cpp"],
             extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
         ),
+        CppExtension(
+            name="hetu_galvatron_amd._galvatron_dataset_helpers",
+            sources=["hetu_galvatron_amd/csrc_cpu/dataset_helpers.cpp"],
+            extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )
