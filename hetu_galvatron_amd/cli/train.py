@@ -25,6 +25,9 @@ def main(argv=None):
     from ..profiler.runtime import RuntimeProfiler
     from ..runtime import (GalvatronModel, get_optimizer_and_param_scheduler,
                            get_train_iterator)
+    from ..runtime.rerun_state_machine import (
+        RerunDataIterator, initialize_rerun_state_machine)
+    from ..utils.logging import MetricsLogger
 
     cfg = config_from_cli(argv)
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -42,28 +45,43 @@ def main(argv=None):
 
     model = GalvatronModel(cfg, device=device)
     opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
-    it = get_train_iterator(cfg, device)
+    it = RerunDataIterator(get_train_iterator(cfg, device))
     prof = RuntimeProfiler(enabled=use_gpu, device=device, rank=rank)
+    rsm = initialize_rerun_state_machine(enabled=True)
+    mlog = MetricsLogger(cfg, rank=rank)
 
     iters = cfg.train.train_iters
     for i in range(iters):
         prof.profile_memory("Before-Fwd")
         prof.time_start()
-        opt.zero_grad()
-        ctx = next(it)
-        stats = model.forward_backward(ctx)
+        loss, norm = float("nan"), 0.0
+        while rsm.should_run_forward_backward(it):
+            opt.zero_grad()
+            ctx = next(it)
+            stats = model.forward_backward(ctx)
+            loss = model.global_loss(stats)
+            rsm.validate_result(loss)
         prof.profile_memory("After-Bwd")
         norm = opt.step()
         sched.step()
         prof.profile_memory("After-step")
-        prof.time_end()
-        loss = model.global_loss(stats)
+        ms = prof.time_end()
         prof.log_iteration(loss, sched.get_lr(), norm,
                            cfg.logging.log_interval)
+        mlog.log({"loss": loss, "lr": sched.get_lr(), "grad_norm": norm,
+                  "iter_ms": ms or 0.0}, i)
+        if rsm.request_checkpoint_and_exit:
+            if cfg.ckpt.save:
+                from ..runtime.checkpoint import save_distributed_checkpoint
+                save_distributed_checkpoint(model, opt, sched, cfg, i + 1)
+            print(f"[rerun] persistent fault at iter {i}: exiting "
+                  f"{rsm.exit_code}", file=sys.stderr)
+            sys.exit(rsm.exit_code)
         if (cfg.ckpt.save and cfg.ckpt.save_interval
                 and (i + 1) % cfg.ckpt.save_interval == 0):
             from ..runtime.checkpoint import save_distributed_checkpoint
             save_distributed_checkpoint(model, opt, sched, cfg, i + 1)
+    mlog.close()
 
     # -- model-profiler worker mode ---------------------------------------
     if cfg.profile.profile and rank == 0:
