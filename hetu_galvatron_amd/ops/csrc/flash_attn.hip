@@ -555,8 +555,16 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward dK/dV (expanded per q-head; host reduces over the GQA group).
-// grid over kv blocks; loops q tiles.
+// backward dK/dV (v3): expanded per q-head (host reduces over the GQA
+// group), 8 waves x 32 keys (256 keys/workgroup), async-staged q tiles.
+// Works in the S orientation (D-layout rows=q, cols=key) so BOTH the
+// dV^T and dK^T MFMAs take their P / dS operands from registers via the
+// cvt_pk+permlane32_swap conversion — the v1/v2 per-tile LDS round trips
+// (32 scalar ds_write_b16 x2, 23-37% bank-conflict cycles) are gone.
+//   S   = mfma(Q_frag[from q_lds rows], K^T_frag[kf regs])
+//   dP  = mfma(dO_frag[from do_lds rows], V^T_frag[vf regs])
+//   dV^T[d][key] = mfma(dO^T_frag[from dot_lds], P_frag[permlane])
+//   dK^T[d][key] = mfma(Q^T_frag[from qt_lds], dS_frag[permlane])
 // ---------------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(256, 1)
@@ -570,27 +578,31 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           __bf16* __restrict__ dv_exp,
                           int b, int sq, int skv, int hq, int hkv,
                           float scale, bool causal) {
+  constexpr int QT = 32;           // q tile
+  constexpr int KBW = 128;         // keys per workgroup (4 waves x 32);
+                                   // 1 wave/SIMD: the whole register file
+                                   // holds kf/vf + both accumulators +
+                                   // the async-staged tile
   constexpr int QROW = D + 8;
-  constexpr int TROW = KVB + 8;
-  constexpr int PROW = KVB + 8;
+  constexpr int TROW = QT + 8;
   constexpr int NK = D / 16;
   constexpr int ND = D / 32;
+  constexpr int QPT = (QT * D / 8 + 255) / 256;   // row packs per thread
 
   __shared__ __align__(16) __bf16
-      smem[KVB * QROW /*Q*/ + D * TROW /*QT*/ + KVB * QROW /*dO*/ +
-           D * TROW /*dOT*/ + 4 * KVB * PROW /*P/dS per wave*/];
+      smem[QT * QROW /*Q*/ + D * TROW /*QT*/ + QT * QROW /*dO*/ +
+           D * TROW /*dOT*/];
   __bf16* q_lds = smem;
-  __bf16* qt_lds = smem + KVB * QROW;
+  __bf16* qt_lds = smem + QT * QROW;
   __bf16* do_lds = qt_lds + D * TROW;
-  __bf16* dot_lds = do_lds + KVB * QROW;
-  __bf16* p_lds = dot_lds + D * TROW;
-  __shared__ float lse_lds[KVB];
-  __shared__ float di_lds[KVB];
+  __bf16* dot_lds = do_lds + QT * QROW;
+  __shared__ float lse_lds[QT];
+  __shared__ float di_lds[QT];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int col = lane & 31;   // q column in S^T
+  const int col = lane & 31;   // this wave's key index
   const int hi = lane >> 5;
 
   const int kvblk = blockIdx.x;
@@ -606,9 +618,9 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const int kv_stride = hkv * D;
   const int dkv_stride = hq * D;
   const int off = skv - sq;
-  const int k0w = kvblk * QB + wid * KVB;  // this wave's first key row
+  const int k0w = kvblk * KBW + wid * 32;  // this wave's first key row
 
-  // K/V A-fragments for this wave's 32 keys
+  // K/V fragments (B-operands of S / dP): lane holds K[key=k0w+col][...]
   bf16x8 kf[NK], vf[NK];
   {
     const int kg = min(k0w + col, skv - 1);
@@ -621,152 +633,220 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
     }
   }
 
-  f32x16 dk_acc[ND], dv_acc[ND];
+  f32x16 dkt[ND], dvt[ND];
 #pragma unroll
   for (int dt = 0; dt < ND; ++dt) {
-    dk_acc[dt] = (f32x16)(0.f);
-    dv_acc[dt] = (f32x16)(0.f);
+    dkt[dt] = (f32x16)(0.f);
+    dvt[dt] = (f32x16)(0.f);
   }
 
-  // first q row that can attend any key in this block (causal)
+  // first q tile that can attend any key of this block / this wave
   int qstart = 0;
-  if (causal) qstart = max(0, ((kvblk * QB - off) / KVB) * KVB);
-  // this wave's own first relevant q tile
-  const int qstart_w =
-      causal ? max(qstart, ((k0w - off) / KVB) * KVB) : qstart;
-  __bf16* my_p = p_lds + wid * KVB * PROW;
+  if (causal) qstart = max(0, ((kvblk * KBW - off) / QT) * QT);
+  const int qstart_w = causal ? max(qstart, ((k0w - off) / QT) * QT) : 0;
 
-  for (int qt0 = qstart; qt0 < sq; qt0 += KVB) {
-    __syncthreads();
-    {
-      const bool full = (qt0 + KVB <= sq);
-      constexpr int PACKS = KVB * D / 8;
+  // staged registers
+  bf16x8 qst[QPT], dost[QPT];
+  ushort8 qtst[QPT], dotst[QPT];
+  float lse_st, di_st;
+
+  auto stage_load = [&](int qt0) {
+    const bool full = (qt0 + QT <= sq);
 #pragma unroll
-      for (int p = tid; p < PACKS; p += 256) {
-        const int row = p / (D / 8);
-        const int c8 = (p - row * (D / 8)) * 8;
-        const int qg = full ? qt0 + row : min(qt0 + row, sq - 1);
-        bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-            q + q_base + (long)qg * q_stride + c8);
-        bf16x8 dov = *reinterpret_cast<const bf16x8*>(
-            dout + q_base + (long)qg * q_stride + c8);
-        if (!full && qt0 + row >= sq) {
-          qv = (bf16x8)(__bf16(0.f));
-          dov = (bf16x8)(__bf16(0.f));
+    for (int p = 0; p < QPT; ++p) {
+      const int idx = tid + p * 256;
+      if (idx >= QT * D / 8) break;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
+      const int qg = full ? qt0 + row : min(qt0 + row, sq - 1);
+      qst[p] = *reinterpret_cast<const bf16x8*>(
+          q + q_base + (long)qg * q_stride + c8);
+      dost[p] = *reinterpret_cast<const bf16x8*>(
+          dout + q_base + (long)qg * q_stride + c8);
+      if (!full && qt0 + row >= sq) {
+        qst[p] = (bf16x8)(__bf16(0.f));
+        dost[p] = (bf16x8)(__bf16(0.f));
+      }
+    }
+#pragma unroll
+    for (int p = 0; p < QPT; ++p) {
+      const int idx = tid + p * 256;
+      if (idx >= D * QT / 8) break;
+      const int c = idx & (D - 1);
+      const int qc = (idx / D) * 8;
+      const unsigned short* qb = reinterpret_cast<const unsigned short*>(
+          q + q_base) + c;
+      const unsigned short* dob = reinterpret_cast<const unsigned short*>(
+          dout + q_base) + c;
+      ushort8 q8, d8;
+      if (full) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          q8[j] = qb[(long)(qt0 + qc + j) * q_stride];
+          d8[j] = dob[(long)(qt0 + qc + j) * q_stride];
         }
-        *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qv;
-        *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dov;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const long rr = (long)min(qt0 + qc + j, sq - 1) * q_stride;
+          unsigned short a = qb[rr], bb2 = dob[rr];
+          q8[j] = qt0 + qc + j < sq ? a : (unsigned short)0;
+          d8[j] = qt0 + qc + j < sq ? bb2 : (unsigned short)0;
+        }
       }
-      // transposed Q / dO via coalesced column loads
-      constexpr int TPACKS = D * KVB / 8;
+      qtst[p] = q8;
+      dotst[p] = d8;
+    }
+    if (tid < QT) {
+      const int qg = qt0 + tid;
+      lse_st = qg < sq ? lse[((long)bi * hq + h) * sq + qg] : INFINITY;
+      di_st = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
+    }
+  };
+  auto stage_write = [&]() {
 #pragma unroll
-      for (int p = tid; p < TPACKS; p += 256) {
-        const int c = p & (D - 1);
-        const int qc = (p / D) * 8;
-        const unsigned short* qb = reinterpret_cast<const unsigned short*>(
-            q + q_base) + c;
-        const unsigned short* dob = reinterpret_cast<const unsigned short*>(
-            dout + q_base) + c;
-        ushort8 q8, d8;
-        if (full) {
+    for (int p = 0; p < QPT; ++p) {
+      const int idx = tid + p * 256;
+      if (idx >= QT * D / 8) break;
+      const int row = idx / (D / 8);
+      const int c8 = (idx - row * (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qst[p];
+      *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dost[p];
+    }
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            q8[j] = qb[(long)(qt0 + qc + j) * q_stride];
-            d8[j] = dob[(long)(qt0 + qc + j) * q_stride];
+    for (int p = 0; p < QPT; ++p) {
+      const int idx = tid + p * 256;
+      if (idx >= D * QT / 8) break;
+      const int c = idx & (D - 1);
+      const int qc = (idx / D) * 8;
+      *reinterpret_cast<ushort8*>(qt_lds + c * TROW + qc) = qtst[p];
+      *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = dotst[p];
+    }
+    if (tid < QT) {
+      lse_lds[tid] = lse_st;
+      di_lds[tid] = di_st;
+    }
+  };
+
+  stage_load(qstart);
+  stage_write();
+  __syncthreads();
+
+  for (int qt0 = qstart; qt0 < sq; qt0 += QT) {
+    const bool have_next = qt0 + QT < sq;
+    if (have_next) stage_load(qt0 + QT);
+
+    if (qt0 >= qstart_w) {  // wave-uniform causal skip
+      // S = Q K^T ; dP = dO V^T   (D-layout rows=q(crow), cols=key(lane))
+      f32x16 s_acc = (f32x16)(0.f), dp_acc = (f32x16)(0.f);
+#pragma unroll
+      for (int ks = 0; ks < NK; ++ks) {
+        bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+            q_lds + col * QROW + ks * 16 + 8 * hi);
+        s_acc = mfma32_bf16(qa, kf[ks], s_acc);
+        bf16x8 doa = *reinterpret_cast<const bf16x8*>(
+            do_lds + col * QROW + ks * 16 + 8 * hi);
+        dp_acc = mfma32_bf16(doa, vf[ks], dp_acc);
+      }
+      // NOTE the operand order: A = q/do frags have lane=q? No — see below.
+      // A-frag lane holds row l&31 of the Q tile; with A=Q (M=q) the
+      // D-layout is rows=q? D rows follow M=A's rows spread over regs and
+      // cols follow B's N=key per lane. Both operands here are read with
+      // lane=own-row, so: A=Q read q_lds row col -> A[q=l&31]; B=kf holds
+      // K[key=l&31] as B^T fragment... B-frag lane must hold B[kk][l&31]
+      // = K^T[kk][key] = K[key=l&31][kk] = kf ✓. So D = S[q? l&31 is BOTH
+      // A-row and B-col index per lane; the MFMA maps A rows over regs:
+      // D[row=crow -> q][col=lane -> key]... A-frag lane l holds
+      // A[l&31][kk] where l&31 indexes M rows -> correct A layout.
+
+      const int qg0 = qt0;
+      float pv[16], dsv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = mfma32_d_row(lane, r);
+        const float lse_r = lse_lds[qrow];
+        const float di_r = di_lds[qrow];
+        const float e = __expf(s_acc[r] * scale - lse_r);
+        float p = e;
+        if (causal) {
+          const int keyg = k0w + col;
+          const bool ok = keyg < skv && keyg <= qg0 + qrow + off;
+          p = ok ? e : 0.f;
+        } else if (k0w + col >= skv) {
+          p = 0.f;
+        }
+        pv[r] = p;
+        dsv[r] = p * (dp_acc[r] - di_r) * scale;
+      }
+
+      // convert P and dS (D-layout rows=q) to B-fragments (lane=key,
+      // in-lane=q) via cvt_pk + permlane32_swap
+      unsigned wp[8], wd[8];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+          {
+            unsigned a = pack_bf16(pv[8 * i + 2 * g], pv[8 * i + 2 * g + 1]);
+            unsigned bb = pack_bf16(pv[8 * i + 2 * g + 4],
+                                    pv[8 * i + 2 * g + 5]);
+            auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
+            wp[i * 4 + g] = r2[0];
+            wp[i * 4 + 2 + g] = r2[1];
           }
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const long rr = (long)min(qt0 + qc + j, sq - 1) * q_stride;
-            q8[j] = qt0 + qc + j < sq ? qb[rr] : (unsigned short)0;
-            d8[j] = qt0 + qc + j < sq ? dob[rr] : (unsigned short)0;
+          {
+            unsigned a = pack_bf16(dsv[8 * i + 2 * g],
+                                   dsv[8 * i + 2 * g + 1]);
+            unsigned bb = pack_bf16(dsv[8 * i + 2 * g + 4],
+                                    dsv[8 * i + 2 * g + 5]);
+            auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
+            wd[i * 4 + g] = r2[0];
+            wd[i * 4 + 2 + g] = r2[1];
           }
         }
-        *reinterpret_cast<ushort8*>(qt_lds + c * TROW + qc) = q8;
-        *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = d8;
       }
-      if (tid < KVB) {
-        const int qg = qt0 + tid;
-        lse_lds[tid] = qg < sq ? lse[((long)bi * hq + h) * sq + qg] : INFINITY;
-        di_lds[tid] = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
+
+      // dV^T += dO^T P ; dK^T += Q^T dS   (two K=16 slices of the q tile)
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+        union { unsigned u[4]; bf16x8 f; } pb, db;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          pb.u[g] = wp[kh * 4 + g];
+          db.u[g] = wd[kh * 4 + g];
+        }
+#pragma unroll
+        for (int dt = 0; dt < ND; ++dt) {
+          bf16x8 doa = *reinterpret_cast<const bf16x8*>(
+              dot_lds + (col + 32 * dt) * TROW + kh * 16 + 8 * hi);
+          dvt[dt] = mfma32_bf16(doa, pb.f, dvt[dt]);
+          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+              qt_lds + (col + 32 * dt) * TROW + kh * 16 + 8 * hi);
+          dkt[dt] = mfma32_bf16(qa, db.f, dkt[dt]);
+        }
       }
     }
+
     __syncthreads();
-    if (qt0 < qstart_w) continue;  // wave-uniform causal skip
-
-    // S^T = K Q^T  and  dP^T = V dO^T
-    f32x16 st_acc = (f32x16)(0.f), dpt_acc = (f32x16)(0.f);
-#pragma unroll
-    for (int ks = 0; ks < NK; ++ks) {
-      bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-          q_lds + col * QROW + ks * 16 + 8 * hi);
-      st_acc = mfma32_bf16(kf[ks], qb, st_acc);
-      bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-          do_lds + col * QROW + ks * 16 + 8 * hi);
-      dpt_acc = mfma32_bf16(vf[ks], dob, dpt_acc);
-    }
-
-    const int qg = qt0 + col;
-    const float lse_c = lse_lds[col];  // INF on padded q rows -> p = 0
-    const float di_c = di_lds[col];
-    const bool need_mask = causal && (qt0 < k0w + KVB - off);
-    float pt[16];
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const float e = __expf(st_acc[i] * scale - lse_c);
-      float p = e;
-      if (need_mask) {
-        const int keyg = k0w + mfma32_d_row(lane, i);
-        p = (keyg < skv && keyg <= qg + off) ? e : 0.f;
-      }
-      pt[i] = p;
-      my_p[mfma32_d_row(lane, i) * PROW + col] = (__bf16)p;
-    }
-
-    // dV += P^T dO  (P^T via LDS A-frags, dO^T tile as B)
-#pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-          my_p + col * PROW + kh * 16 + 8 * hi);
-#pragma unroll
-      for (int dt = 0; dt < ND; ++dt) {
-        bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-            dot_lds + (dt * 32 + col) * TROW + kh * 16 + 8 * hi);
-        dv_acc[dt] = mfma32_bf16(pa, dob, dv_acc[dt]);
-      }
-    }
-
-    // dS^T then dK += dS^T Q
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const float ds = pt[i] * (dpt_acc[i] - di_c) * scale;
-      my_p[mfma32_d_row(lane, i) * PROW + col] = (__bf16)ds;
-    }
-#pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      bf16x8 da = *reinterpret_cast<const bf16x8*>(
-          my_p + col * PROW + kh * 16 + 8 * hi);
-#pragma unroll
-      for (int dt = 0; dt < ND; ++dt) {
-        bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-            qt_lds + (dt * 32 + col) * TROW + kh * 16 + 8 * hi);
-        dk_acc[dt] = mfma32_bf16(da, qb, dk_acc[dt]);
-      }
+    if (have_next) {
+      stage_write();
+      __syncthreads();
     }
   }
 
+  // epilogue: dK^T/dV^T D-layout rows=d(crow), cols=key(lane)
+  const int keyg = k0w + col;
+  if (keyg < skv) {
+    __bf16* dkr = dk_exp + dkv_base + (long)keyg * dkv_stride;
+    __bf16* dvr = dv_exp + dkv_base + (long)keyg * dkv_stride;
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    const int keyg = k0w + mfma32_d_row(lane, i);
-    if (keyg < skv) {
-      __bf16* dkr = dk_exp + dkv_base + (long)keyg * dkv_stride;
-      __bf16* dvr = dv_exp + dkv_base + (long)keyg * dkv_stride;
+    for (int dt = 0; dt < ND; ++dt)
 #pragma unroll
-      for (int dt = 0; dt < ND; ++dt) {
-        dkr[dt * 32 + col] = (__bf16)dk_acc[dt][i];
-        dvr[dt * 32 + col] = (__bf16)dv_acc[dt][i];
+      for (int r = 0; r < 16; ++r) {
+        const int d0 = dt * 32 + mfma32_d_row(lane, r);
+        dkr[d0] = (__bf16)dkt[dt][r];
+        dvr[d0] = (__bf16)dvt[dt][r];
       }
-    }
   }
 }
 
@@ -841,7 +921,7 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
   dim3 gq((sq + 255) / 256, b * hq);
   hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout, q,
                      k, v, lse, di, dq, b, sq, skv, hq, hkv, scale, causal);
-  dim3 gkv((skv + QB - 1) / QB, b * hq);
+  dim3 gkv((skv + 127) / 128, b * hq);
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(256), 0, st, dout,
                      q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv, hq, hkv,
                      scale, causal);
