@@ -60,9 +60,10 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   constexpr int ND = D / 32;        // 32-wide d tiles of O^T
   constexpr int KPT = D / 64;       // staging packs per thread (512 thr)
 
-  __shared__ __align__(16) __bf16 smem[KB * KROW + D * VTROW];
-  __bf16* k_lds = smem;
-  __bf16* vt_lds = smem + KB * KROW;
+  // double-buffered tiles: ONE barrier per KV tile (PMC showed the
+  // 2-barrier synchronous form 38-44% parked in SQ_WAIT_ANY)
+  constexpr int BUFSZ = KB * KROW + D * VTROW;
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -150,7 +151,9 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
       vst[p] = vv;
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int buf) {
+    __bf16* k_lds = smem + buf * BUFSZ;
+    __bf16* vt_lds = k_lds + KB * KROW;
 #pragma unroll
     for (int p = 0; p < KPT; ++p) {
       const int idx = tid + p * 512;
@@ -168,11 +171,14 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   };
 
   stage_load(0);
-  stage_write();
+  stage_write(0);
   __syncthreads();
+  int cur = 0;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     const bool have_next = kv0 + KB < kv_end;
+    const __bf16* k_lds = smem + cur * BUFSZ;
+    const __bf16* vt_lds = k_lds + KB * KROW;
     if (have_next) stage_load(kv0 + KB);  // async: lands at stage_write
 
     if (kv0 < kv_last_w) {  // per-wave causal skip (wave-uniform)
@@ -271,11 +277,9 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
       }
     }
 
+    if (have_next) stage_write(cur ^ 1);
     __syncthreads();
-    if (have_next) {
-      stage_write();
-      __syncthreads();
-    }
+    cur ^= 1;
   }
 
   // ---- epilogue: O^T[d][q=lane], per-lane stats ----
@@ -353,10 +357,8 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   constexpr int ND = D / 32;
   constexpr int KPT = (KB * D / 8 + 511) / 512;  // row packs per thread
 
-  __shared__ __align__(16) __bf16 smem[2 * KB * KROW + D * KTROW];
-  __bf16* k_lds = smem;
-  __bf16* v_lds = smem + KB * KROW;
-  __bf16* kt_lds = v_lds + KB * KROW;
+  constexpr int BUFSZ = 2 * KB * KROW + D * KTROW;
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -446,7 +448,10 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       ktst[p] = t;
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int buf) {
+    __bf16* k_lds = smem + buf * BUFSZ;
+    __bf16* v_lds = k_lds + KB * KROW;
+    __bf16* kt_lds = v_lds + KB * KROW;
 #pragma unroll
     for (int p = 0; p < KPT; ++p) {
       const int idx = tid + p * 512;
@@ -467,13 +472,17 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   };
 
   stage_load(0);
-  stage_write();
+  stage_write(0);
   __syncthreads();
+  int cur = 0;
 
   const float sl2e = scale;  // bwd stays in natural-log units (lse is ln)
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     const bool have_next = kv0 + KB < kv_end;
+    const __bf16* k_lds = smem + cur * BUFSZ;
+    const __bf16* v_lds = k_lds + KB * KROW;
+    const __bf16* kt_lds = v_lds + KB * KROW;
     if (have_next) stage_load(kv0 + KB);
 
     if (kv0 < kv_last_w) {
@@ -536,11 +545,9 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       }
     }
 
+    if (have_next) stage_write(cur ^ 1);
     __syncthreads();
-    if (have_next) {
-      stage_write();
-      __syncthreads();
-    }
+    cur ^= 1;
   }
 
   // epilogue: dQ^T[d][q=lane]
@@ -590,15 +597,10 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   constexpr int ND = D / 32;
   constexpr int QPT = (QT * D / 8 + 255) / 256;   // row packs per thread
 
-  __shared__ __align__(16) __bf16
-      smem[QT * QROW /*Q*/ + D * TROW /*QT*/ + QT * QROW /*dO*/ +
-           D * TROW /*dOT*/];
-  __bf16* q_lds = smem;
-  __bf16* qt_lds = smem + QT * QROW;
-  __bf16* do_lds = qt_lds + D * TROW;
-  __bf16* dot_lds = do_lds + QT * QROW;
-  __shared__ float lse_lds[QT];
-  __shared__ float di_lds[QT];
+  constexpr int BUFSZ = 2 * QT * QROW + 2 * D * TROW;
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  __shared__ float lse_lds[2][QT];
+  __shared__ float di_lds[2][QT];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -704,7 +706,11 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
       di_st = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int buf) {
+    __bf16* q_lds = smem + buf * BUFSZ;
+    __bf16* qt_lds = q_lds + QT * QROW;
+    __bf16* do_lds = qt_lds + D * TROW;
+    __bf16* dot_lds = do_lds + QT * QROW;
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
       const int idx = tid + p * 256;
@@ -724,17 +730,22 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
       *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = dotst[p];
     }
     if (tid < QT) {
-      lse_lds[tid] = lse_st;
-      di_lds[tid] = di_st;
+      lse_lds[buf][tid] = lse_st;
+      di_lds[buf][tid] = di_st;
     }
   };
 
   stage_load(qstart);
-  stage_write();
+  stage_write(0);
   __syncthreads();
+  int cur = 0;
 
   for (int qt0 = qstart; qt0 < sq; qt0 += QT) {
     const bool have_next = qt0 + QT < sq;
+    const __bf16* q_lds = smem + cur * BUFSZ;
+    const __bf16* qt_lds = q_lds + QT * QROW;
+    const __bf16* do_lds = qt_lds + D * TROW;
+    const __bf16* dot_lds = do_lds + QT * QROW;
     if (have_next) stage_load(qt0 + QT);
 
     if (qt0 >= qstart_w) {  // wave-uniform causal skip
@@ -765,8 +776,8 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = mfma32_d_row(lane, r);
-        const float lse_r = lse_lds[qrow];
-        const float di_r = di_lds[qrow];
+        const float lse_r = lse_lds[cur][qrow];
+        const float di_r = di_lds[cur][qrow];
         const float e = __expf(s_acc[r] * scale - lse_r);
         float p = e;
         if (causal) {
@@ -828,11 +839,9 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
       }
     }
 
+    if (have_next) stage_write(cur ^ 1);
     __syncthreads();
-    if (have_next) {
-      stage_write();
-      __syncthreads();
-    }
+    cur ^= 1;
   }
 
   // epilogue: dK^T/dV^T D-layout rows=d(crow), cols=key(lane)
