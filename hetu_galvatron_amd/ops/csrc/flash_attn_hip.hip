@@ -71,7 +71,6 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const int hi = lane >> 5;
   const int wid = tid >> 6;
 
-  const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / hq;
   const int h = bh - bi * hq;
@@ -82,8 +81,16 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const int q_stride = hq * D;
   const int kv_stride = hkv * D;
   const int off = skv - sq;
-  const int q0w = qblk * QBF + wid * 32;
   const float sl2e = scale * 1.4426950408889634f;  // fold log2(e): exp2 units
+
+  // causal load balance: block x runs q blocks {x, nqb-1-x} — the pair's
+  // total KV-tile count is constant, so the makespan matches the average
+  // instead of 2x the deepest block
+  const int nqb = (sq + QBF - 1) / QBF;
+  for (int pass = 0; pass < 2; ++pass) {
+  const int qblk = pass ? nqb - 1 - (int)blockIdx.x : (int)blockIdx.x;
+  if (pass && (!causal || qblk <= (int)blockIdx.x)) break;
+  const int q0w = qblk * QBF + wid * 32;
 
   // Q fragments (B-operand of S^T): lane holds Q[q0w+col][ks*16 + 8*hi + j]
   bf16x8 qf[NK];
@@ -297,6 +304,7 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
           l_run > 0.f ? (m_run + log2f(l_run)) * 0.6931471805599453f
                       : -INFINITY;
   }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -366,7 +374,6 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const int hi = lane >> 5;
   const int wid = tid >> 6;
 
-  const int qblk = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / hq;
   const int h = bh - bi * hq;
@@ -377,6 +384,11 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const int q_stride = hq * D;
   const int kv_stride = hkv * D;
   const int off = skv - sq;
+
+  const int nqb = (sq + QBF - 1) / QBF;  // causal pair balance (see fwd)
+  for (int pass = 0; pass < 2; ++pass) {
+  const int qblk = pass ? nqb - 1 - (int)blockIdx.x : (int)blockIdx.x;
+  if (pass && (!causal || qblk <= (int)blockIdx.x)) break;
   const int q0w = qblk * QBF + wid * 32;
 
   bf16x8 qf[NK], dof[NK];
@@ -560,6 +572,7 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       for (int r = 0; r < 16; ++r)
         dqr[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)dq_acc[dt][r];
   }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -608,7 +621,6 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const int col = lane & 31;   // this wave's key index
   const int hi = lane >> 5;
 
-  const int kvblk = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / hq;
   const int h = bh - bi * hq;
@@ -621,6 +633,11 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const int kv_stride = hkv * D;
   const int dkv_stride = hq * D;
   const int off = skv - sq;
+
+  const int nkb = (skv + KBW - 1) / KBW;  // causal pair balance (see fwd)
+  for (int pass = 0; pass < 2; ++pass) {
+  const int kvblk = pass ? nkb - 1 - (int)blockIdx.x : (int)blockIdx.x;
+  if (pass && (!causal || kvblk <= (int)blockIdx.x)) break;
   const int k0w = kvblk * KBW + wid * 32;  // this wave's first key row
 
   // K/V fragments (B-operands of S / dP): lane holds K[key=k0w+col][...]
@@ -858,6 +875,7 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
         dvr[d0] = (__bf16)dvt[dt][r];
       }
   }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -892,7 +910,8 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                const __bf16* v, __bf16* o, float* lse, int b,
                                int sq, int skv, int hq, int hkv, float scale,
                                bool causal, hipStream_t st) {
-  dim3 grid((sq + 255) / 256, b * hq);
+  int nqb = (sq + 255) / 256;
+  dim3 grid(causal ? (nqb + 1) / 2 : nqb, b * hq);
   hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k, v,
                      o, lse, b, sq, skv, hq, hkv, scale, causal);
 }
@@ -928,10 +947,12 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                __bf16* dk_exp, __bf16* dv_exp, int b, int sq,
                                int skv, int hq, int hkv, float scale,
                                bool causal, hipStream_t st) {
-  dim3 gq((sq + 255) / 256, b * hq);
+  int nqb = (sq + 255) / 256;
+  dim3 gq(causal ? (nqb + 1) / 2 : nqb, b * hq);
   hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout, q,
                      k, v, lse, di, dq, b, sq, skv, hq, hkv, scale, causal);
-  dim3 gkv((skv + 127) / 128, b * hq);
+  int nkb = (skv + 127) / 128;
+  dim3 gkv(causal ? (nkb + 1) / 2 : nkb, b * hq);
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(256), 0, st, dout,
                      q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv, hq, hkv,
                      scale, causal);
