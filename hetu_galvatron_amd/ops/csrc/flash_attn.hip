@@ -44,13 +44,14 @@ typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
 // forward (v2)
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(512, 2)
-void flash_fwd_kernel(const __bf16* __restrict__ q,
-                      const __bf16* __restrict__ k,
-                      const __bf16* __restrict__ v,
-                      __bf16* __restrict__ o, float* __restrict__ lse,
-                      int b, int sq, int skv, int hq, int hkv, float scale,
-                      bool causal) {
+__device__ __attribute__((noinline))
+void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
+                     const __bf16* __restrict__ k,
+                     const __bf16* __restrict__ v,
+                     __bf16* __restrict__ o, float* __restrict__ lse,
+                     __bf16* smem_base, long q_base, long kv_base,
+                     long lse_base, int q_stride, int kv_stride, int off,
+                     int sq, int skv, float sl2e, bool causal) {
   constexpr int KB = 64;            // kv tile
   constexpr int QBF = 256;          // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;       // padded K row (bf16 elems)
@@ -59,36 +60,14 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   constexpr int ND = D / 32;        // 32-wide d tiles of O^T
   constexpr int KPT = D / 64;       // staging packs per thread (512 thr)
 
-  // double-buffered tiles: ONE barrier per KV tile (PMC showed the
-  // 2-barrier synchronous form 38-44% parked in SQ_WAIT_ANY)
   constexpr int BUFSZ = KB * KROW + D * VTROW;
-  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  __bf16* smem = smem_base;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int col = lane & 31;
   const int hi = lane >> 5;
   const int wid = tid >> 6;
-
-  const int bh = blockIdx.y;
-  const int bi = bh / hq;
-  const int h = bh - bi * hq;
-  const int hk = h / (hq / hkv);
-
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
-  const int q_stride = hq * D;
-  const int kv_stride = hkv * D;
-  const int off = skv - sq;
-  const float sl2e = scale * 1.4426950408889634f;  // fold log2(e): exp2 units
-
-  // causal load balance: block x runs q blocks {x, nqb-1-x} — the pair's
-  // total KV-tile count is constant, so the makespan matches the average
-  // instead of 2x the deepest block
-  const int nqb = (sq + QBF - 1) / QBF;
-  for (int pass = 0; pass < 2; ++pass) {
-  const int qblk = pass ? nqb - 1 - (int)blockIdx.x : (int)blockIdx.x;
-  if (pass && (!causal || qblk <= (int)blockIdx.x)) break;
   const int q0w = qblk * QBF + wid * 32;
 
   // Q fragments (B-operand of S^T): lane holds Q[q0w+col][ks*16 + 8*hi + j]
@@ -299,10 +278,43 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
       for (int r = 0; r < 16; ++r)
         orow[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)(ot[dt][r] * invl);
     if (hi == 0)
-      lse[((long)bi * hq + h) * sq + qg] =
+      lse[lse_base + qg] =
           l_run > 0.f ? (m_run + log2f(l_run)) * 0.6931471805599453f
                       : -INFINITY;
   }
+}
+
+// thin kernel: common indexing + complementary-pair causal load balance
+// (block x runs q blocks {x, nqb-1-x}: constant total KV tiles per block,
+// so the makespan matches the average instead of 2x the deepest block)
+template <int D>
+__global__ __launch_bounds__(512, 2)
+void flash_fwd_kernel(const __bf16* __restrict__ q,
+                      const __bf16* __restrict__ k,
+                      const __bf16* __restrict__ v,
+                      __bf16* __restrict__ o, float* __restrict__ lse,
+                      int b, int sq, int skv, int hq, int hkv, float scale,
+                      bool causal) {
+  constexpr int KB = 64, QBF = 256;
+  constexpr int BUFSZ = KB * (D + 8) + D * (KB + 8);
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  const int bh = blockIdx.y;
+  const int bi = bh / hq;
+  const int h = bh - bi * hq;
+  const int hk = h / (hq / hkv);
+  const long q_base = ((long)bi * sq * hq + h) * D;
+  const long kv_base = ((long)bi * skv * hkv + hk) * D;
+  const long lse_base = ((long)bi * hq + h) * sq;
+  const float sl2e = scale * 1.4426950408889634f;
+  const int off = skv - sq;
+  const int nqb = (sq + QBF - 1) / QBF;
+  flash_fwd_block<D>(blockIdx.x, q, k, v, o, lse, smem, q_base, kv_base,
+                     lse_base, hq * D, hkv * D, off, sq, skv, sl2e, causal);
+  const int qb2 = nqb - 1 - (int)blockIdx.x;
+  if (causal && qb2 > (int)blockIdx.x) {
+    __syncthreads();
+    flash_fwd_block<D>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
+                       lse_base, hq * D, hkv * D, off, sq, skv, sl2e, causal);
   }
 }
 
@@ -346,16 +358,17 @@ __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
 // dQ^T[d][q] += K^T dS accumulated in D-layout, epilogue like the forward.
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(512, 2)
-void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
-                         const __bf16* __restrict__ q,
-                         const __bf16* __restrict__ k,
-                         const __bf16* __restrict__ v,
-                         const float* __restrict__ lse,
-                         const float* __restrict__ di,
-                         __bf16* __restrict__ dq,
-                         int b, int sq, int skv, int hq, int hkv,
-                         float scale, bool causal) {
+__device__ __attribute__((noinline))
+void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
+                        const __bf16* __restrict__ q,
+                        const __bf16* __restrict__ k,
+                        const __bf16* __restrict__ v,
+                        const float* __restrict__ lse,
+                        const float* __restrict__ di,
+                        __bf16* __restrict__ dq, __bf16* smem_base,
+                        long q_base, long kv_base, long lse_base,
+                        int q_stride, int kv_stride, int off, int sq,
+                        int skv, float scale, bool causal) {
   constexpr int KB = 32;         // kv tile
   constexpr int QBF = 256;       // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;
@@ -365,29 +378,13 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   constexpr int KPT = (KB * D / 8 + 511) / 512;  // row packs per thread
 
   constexpr int BUFSZ = 2 * KB * KROW + D * KTROW;
-  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  __bf16* smem = smem_base;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int col = lane & 31;
   const int hi = lane >> 5;
   const int wid = tid >> 6;
-
-  const int bh = blockIdx.y;
-  const int bi = bh / hq;
-  const int h = bh - bi * hq;
-  const int hk = h / (hq / hkv);
-
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
-  const int q_stride = hq * D;
-  const int kv_stride = hkv * D;
-  const int off = skv - sq;
-
-  const int nqb = (sq + QBF - 1) / QBF;  // causal pair balance (see fwd)
-  for (int pass = 0; pass < 2; ++pass) {
-  const int qblk = pass ? nqb - 1 - (int)blockIdx.x : (int)blockIdx.x;
-  if (pass && (!causal || qblk <= (int)blockIdx.x)) break;
   const int q0w = qblk * QBF + wid * 32;
 
   bf16x8 qf[NK], dof[NK];
@@ -401,8 +398,8 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       qf[ks] = *reinterpret_cast<const bf16x8*>(qp + ks * 16);
       dof[ks] = *reinterpret_cast<const bf16x8*>(dop + ks * 16);
     }
-    const float l0 = lse[((long)bi * hq + h) * sq + qg];
-    const float d0 = di[((long)bi * hq + h) * sq + qg];
+    const float l0 = lse[lse_base + qg];
+    const float d0 = di[lse_base + qg];
     const bool live = q0w + col < sq;
     lse_c = live ? l0 : INFINITY;  // exp(x - inf) = 0 -> dead rows silent
     di_c = live ? d0 : 0.f;
@@ -571,6 +568,40 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
       for (int r = 0; r < 16; ++r)
         dqr[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)dq_acc[dt][r];
   }
+}
+
+template <int D>
+__global__ __launch_bounds__(512, 2)
+void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
+                         const __bf16* __restrict__ q,
+                         const __bf16* __restrict__ k,
+                         const __bf16* __restrict__ v,
+                         const float* __restrict__ lse,
+                         const float* __restrict__ di,
+                         __bf16* __restrict__ dq,
+                         int b, int sq, int skv, int hq, int hkv,
+                         float scale, bool causal) {
+  constexpr int KB = 32, QBF = 256;
+  constexpr int BUFSZ = 2 * KB * (D + 8) + D * (KB + 8);
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  const int bh = blockIdx.y;
+  const int bi = bh / hq;
+  const int h = bh - bi * hq;
+  const int hk = h / (hq / hkv);
+  const long q_base = ((long)bi * sq * hq + h) * D;
+  const long kv_base = ((long)bi * skv * hkv + hk) * D;
+  const long lse_base = ((long)bi * hq + h) * sq;
+  const int off = skv - sq;
+  const int nqb = (sq + QBF - 1) / QBF;
+  flash_bwd_dq_block<D>(blockIdx.x, dout, q, k, v, lse, di, dq, smem,
+                        q_base, kv_base, lse_base, hq * D, hkv * D, off,
+                        sq, skv, scale, causal);
+  const int qb2 = nqb - 1 - (int)blockIdx.x;
+  if (causal && qb2 > (int)blockIdx.x) {
+    __syncthreads();
+    flash_bwd_dq_block<D>(qb2, dout, q, k, v, lse, di, dq, smem, q_base,
+                          kv_base, lse_base, hq * D, hkv * D, off, sq, skv,
+                          scale, causal);
   }
 }
 
@@ -587,17 +618,19 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 //   dK^T[d][key] = mfma(Q^T_frag[from qt_lds], dS_frag[permlane])
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256, 1)
-void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
-                          const __bf16* __restrict__ q,
-                          const __bf16* __restrict__ k,
-                          const __bf16* __restrict__ v,
-                          const float* __restrict__ lse,
-                          const float* __restrict__ di,
-                          __bf16* __restrict__ dk_exp,   // [b, skv, hq, D]
-                          __bf16* __restrict__ dv_exp,
-                          int b, int sq, int skv, int hq, int hkv,
-                          float scale, bool causal) {
+__device__ __attribute__((noinline))
+void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
+                         const __bf16* __restrict__ q,
+                         const __bf16* __restrict__ k,
+                         const __bf16* __restrict__ v,
+                         const float* __restrict__ lse,
+                         const float* __restrict__ di,
+                         __bf16* __restrict__ dk_exp,
+                         __bf16* __restrict__ dv_exp, __bf16* smem_base,
+                         float* lsedi_base, long q_base, long kv_base,
+                         long dkv_base, long lse_base, int q_stride,
+                         int kv_stride, int dkv_stride, int off, int sq,
+                         int skv, float scale, bool causal) {
   constexpr int QT = 32;           // q tile
   constexpr int KBW = 128;         // keys per workgroup (4 waves x 32);
                                    // 1 wave/SIMD: the whole register file
@@ -610,33 +643,15 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   constexpr int QPT = (QT * D / 8 + 255) / 256;   // row packs per thread
 
   constexpr int BUFSZ = 2 * QT * QROW + 2 * D * TROW;
-  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
-  __shared__ float lse_lds[2][QT];
-  __shared__ float di_lds[2][QT];
+  __bf16* smem = smem_base;
+  float (*lse_lds)[QT] = reinterpret_cast<float (*)[QT]>(lsedi_base);
+  float (*di_lds)[QT] = reinterpret_cast<float (*)[QT]>(lsedi_base + 2 * QT);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int col = lane & 31;   // this wave's key index
   const int hi = lane >> 5;
-
-  const int bh = blockIdx.y;
-  const int bi = bh / hq;
-  const int h = bh - bi * hq;
-  const int hk = h / (hq / hkv);
-
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
-  const long dkv_base = ((long)bi * skv * hq + h) * D;
-  const int q_stride = hq * D;
-  const int kv_stride = hkv * D;
-  const int dkv_stride = hq * D;
-  const int off = skv - sq;
-
-  const int nkb = (skv + KBW - 1) / KBW;  // causal pair balance (see fwd)
-  for (int pass = 0; pass < 2; ++pass) {
-  const int kvblk = pass ? nkb - 1 - (int)blockIdx.x : (int)blockIdx.x;
-  if (pass && (!causal || kvblk <= (int)blockIdx.x)) break;
   const int k0w = kvblk * KBW + wid * 32;  // this wave's first key row
 
   // K/V fragments (B-operands of S / dP): lane holds K[key=k0w+col][...]
@@ -718,8 +733,8 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
     }
     if (tid < QT) {
       const int qg = qt0 + tid;
-      lse_st = qg < sq ? lse[((long)bi * hq + h) * sq + qg] : INFINITY;
-      di_st = qg < sq ? di[((long)bi * hq + h) * sq + qg] : 0.f;
+      lse_st = qg < sq ? lse[lse_base + qg] : INFINITY;
+      di_st = qg < sq ? di[lse_base + qg] : 0.f;
     }
   };
   auto stage_write = [&](int buf) {
@@ -874,6 +889,45 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
         dvr[d0] = (__bf16)dvt[dt][r];
       }
   }
+}
+
+template <int D>
+__global__ __launch_bounds__(256, 1)
+void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
+                          const __bf16* __restrict__ q,
+                          const __bf16* __restrict__ k,
+                          const __bf16* __restrict__ v,
+                          const float* __restrict__ lse,
+                          const float* __restrict__ di,
+                          __bf16* __restrict__ dk_exp,   // [b, skv, hq, D]
+                          __bf16* __restrict__ dv_exp,
+                          int b, int sq, int skv, int hq, int hkv,
+                          float scale, bool causal) {
+  constexpr int QT = 32, KBW = 128;
+  constexpr int BUFSZ = 2 * QT * (D + 8) + 2 * D * (QT + 8);
+  __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
+  __shared__ __align__(16) float lsedi[4 * QT];
+  const int bh = blockIdx.y;
+  const int bi = bh / hq;
+  const int h = bh - bi * hq;
+  const int hk = h / (hq / hkv);
+  const long q_base = ((long)bi * sq * hq + h) * D;
+  const long kv_base = ((long)bi * skv * hkv + hk) * D;
+  const long dkv_base = ((long)bi * skv * hq + h) * D;
+  const long lse_base = ((long)bi * hq + h) * sq;
+  const int off = skv - sq;
+  const int nkb = (skv + KBW - 1) / KBW;
+  flash_bwd_dkv_block<D>(blockIdx.x, dout, q, k, v, lse, di, dk_exp, dv_exp,
+                         smem, lsedi, q_base, kv_base, dkv_base, lse_base,
+                         hq * D, hkv * D, hq * D, off, sq, skv, scale,
+                         causal);
+  const int kb2 = nkb - 1 - (int)blockIdx.x;
+  if (causal && kb2 > (int)blockIdx.x) {
+    __syncthreads();
+    flash_bwd_dkv_block<D>(kb2, dout, q, k, v, lse, di, dk_exp, dv_exp,
+                           smem, lsedi, q_base, kv_base, dkv_base, lse_base,
+                           hq * D, hkv * D, hq * D, off, sq, skv, scale,
+                           causal);
   }
 }
 
