@@ -21,6 +21,47 @@ import torch
 import torch.distributed as dist
 
 
+def search_bench_plan(cfg, world, args):
+    """Run the strategy search for the bench config; returns a plan."""
+    from hetu_galvatron_amd.cli.search import default_mi355x_hardware
+    from hetu_galvatron_amd.search.costmodel import LayerProfile, OtherProfile
+    from hetu_galvatron_amd.search.engine import SearchEngine
+
+    m = cfg.model
+    # measured on MI355X (rocprofv3, profiles/bench_r01_kernel_stats_v2.md):
+    # ~7.3 ms fwd per decoder layer per local sample at seq 4096
+    per_tok_ms = 7.3 / 4096
+    lp = LayerProfile(
+        parameter_mb=(4 * m.hidden_size ** 2 +
+                      3 * m.hidden_size * m.ffn_hidden_size) * 4 / 1e6,
+        fct_linear=(per_tok_ms * m.seq_length, 0.05),
+        act_per_bsz_mb={str(t): m.seq_length * m.hidden_size * 2 * 18
+                        / 1e6 / t for t in (1, 2, 4, 8)}
+        | {"checkpoint": m.seq_length * m.hidden_size * 2 / 1e6},
+        seq_length=m.seq_length, hidden_size=m.hidden_size)
+    op = OtherProfile(
+        parameter_mb=m.vocab_size * m.hidden_size * 4 / 1e6,
+        act_per_bsz_mb={"1": m.seq_length * m.vocab_size * 2 / 1e6 / max(cfg.parallel.chunks, 1)},
+        fct_linear=(0.6, 0.05))
+    cfg.search.num_nodes = 1
+    cfg.search.num_gpus_per_node = max(world, 1)
+    cfg.search.memory_constraint = 270
+    cfg.search.settle_bsz = cfg.train.global_train_batch_size
+    cfg.search.settle_chunks = max(cfg.parallel.chunks, 1)
+    cfg.search.max_tp_deg = 1
+    cfg.search.disable_pp = 1
+    cfg.search.disable_sp = 1
+    cfg.search.max_pp_deg = 1
+    cfg.search.disable_vtp = 1
+    eng = SearchEngine(cfg, lp, op, default_mi355x_hardware(max(world, 1)))
+    best = eng.parallelism_optimization(None)
+    if best is None:
+        return None
+    plan = best.plan
+    plan.default_dp_type = "zero2"
+    return plan
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -37,8 +78,9 @@ def main():
     ap.add_argument("--dp-type", type=str, default="zero2",
                     choices=["ddp", "zero2", "zero3"])
     ap.add_argument("--checkpoint", action="store_true")
-    ap.add_argument("--plan", type=str, default="",
-                    help="searched-plan JSON path (overrides uniform knobs)")
+    ap.add_argument("--plan", type=str, default="auto",
+                    help="searched-plan JSON path, 'auto' (run the search "
+                         "engine), or 'none' (uniform knobs)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -71,7 +113,9 @@ def main():
                      "default_dp_type": args.dp_type,
                      "global_checkpoint": 1 if args.checkpoint else 0,
                      "chunks": chunks, "mixed_precision": "bf16",
-                     "galvatron_config_path": args.plan or None},
+                     "galvatron_config_path":
+                         args.plan if args.plan not in ("auto", "none", "")
+                         else None},
     })
 
     if world > 1:
@@ -79,7 +123,17 @@ def main():
     else:
         torch.manual_seed(cfg.train.seed)
 
-    plan = resolve_plan(cfg, world)
+    if args.plan == "auto" and args.pp == 1 and args.tp == 1:
+        # auto-searched plan (BASELINE metric): per-layer DP-type/ckpt search
+        # under the 288 GB budget, measured per-layer compute + analytic
+        # xGMI comm model. Round-1 space: dp-only degrees (the distributed
+        # paths validated on this hardware); tp/sp/pp degrees join the
+        # space as they get GPU-validated.
+        plan = search_bench_plan(cfg, world, args)
+        if plan is None:
+            plan = resolve_plan(cfg, world)
+    else:
+        plan = resolve_plan(cfg, world)
     model = GalvatronModel(cfg, plan, device=device)
     opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
     it = get_train_iterator(cfg, device)
