@@ -632,15 +632,17 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
                          int kv_stride, int dkv_stride, int off, int sq,
                          int skv, float scale, bool causal) {
   constexpr int QT = 32;           // q tile
-  constexpr int KBW = 128;         // keys per workgroup (4 waves x 32);
-                                   // 1 wave/SIMD: the whole register file
-                                   // holds kf/vf + both accumulators +
-                                   // the async-staged tile
+  constexpr int KBW = 256;         // keys per workgroup (8 waves x 32);
+                                   // K/V fragments re-read from L2 per q
+                                   // tile so 2 waves/SIMD fit (the 4-wave
+                                   // register-resident variant ran at
+                                   // 1 wave/SIMD and trailed the fwd
+                                   // kernel's efficiency 1.6x)
   constexpr int QROW = D + 8;
   constexpr int TROW = QT + 8;
   constexpr int NK = D / 16;
   constexpr int ND = D / 32;
-  constexpr int QPT = (QT * D / 8 + 255) / 256;   // row packs per thread
+  constexpr int QPT = (QT * D / 8 + 511) / 512;   // row packs per thread
 
   constexpr int BUFSZ = 2 * QT * QROW + 2 * D * TROW;
   __bf16* smem = smem_base;
@@ -654,18 +656,11 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
   const int hi = lane >> 5;
   const int k0w = kvblk * KBW + wid * 32;  // this wave's first key row
 
-  // K/V fragments (B-operands of S / dP): lane holds K[key=k0w+col][...]
-  bf16x8 kf[NK], vf[NK];
-  {
-    const int kg = min(k0w + col, skv - 1);
-    const __bf16* kp = k + kv_base + (long)kg * kv_stride + 8 * hi;
-    const __bf16* vp = v + kv_base + (long)kg * kv_stride + 8 * hi;
-#pragma unroll
-    for (int ks = 0; ks < NK; ++ks) {
-      kf[ks] = *reinterpret_cast<const bf16x8*>(kp + ks * 16);
-      vf[ks] = *reinterpret_cast<const bf16x8*>(vp + ks * 16);
-    }
-  }
+  // K/V fragment base pointers (B-operands of S / dP); the 16 frags are
+  // re-read per q tile (L2-resident) instead of held in 64 registers
+  const int kg_f = min(k0w + col, skv - 1);
+  const __bf16* kp_f = k + kv_base + (long)kg_f * kv_stride + 8 * hi;
+  const __bf16* vp_f = v + kv_base + (long)kg_f * kv_stride + 8 * hi;
 
   f32x16 dkt[ND], dvt[ND];
 #pragma unroll
@@ -688,7 +683,7 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
     const bool full = (qt0 + QT <= sq);
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
-      const int idx = tid + p * 256;
+      const int idx = tid + p * 512;
       if (idx >= QT * D / 8) break;
       const int row = idx / (D / 8);
       const int c8 = (idx - row * (D / 8)) * 8;
@@ -704,7 +699,7 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
     }
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
-      const int idx = tid + p * 256;
+      const int idx = tid + p * 512;
       if (idx >= D * QT / 8) break;
       const int c = idx & (D - 1);
       const int qc = (idx / D) * 8;
@@ -744,7 +739,7 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
     __bf16* dot_lds = do_lds + QT * QROW;
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
-      const int idx = tid + p * 256;
+      const int idx = tid + p * 512;
       if (idx >= QT * D / 8) break;
       const int row = idx / (D / 8);
       const int c8 = (idx - row * (D / 8)) * 8;
@@ -753,7 +748,7 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
     }
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
-      const int idx = tid + p * 256;
+      const int idx = tid + p * 512;
       if (idx >= D * QT / 8) break;
       const int c = idx & (D - 1);
       const int qc = (idx / D) * 8;
@@ -784,12 +779,14 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
       f32x16 s_acc = (f32x16)(0.f), dp_acc = (f32x16)(0.f);
 #pragma unroll
       for (int ks = 0; ks < NK; ++ks) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp_f + ks * 16);
         bf16x8 qa = *reinterpret_cast<const bf16x8*>(
             q_lds + col * QROW + ks * 16 + 8 * hi);
-        s_acc = mfma32_bf16(qa, kf[ks], s_acc);
+        s_acc = mfma32_bf16(qa, kf, s_acc);
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vp_f + ks * 16);
         bf16x8 doa = *reinterpret_cast<const bf16x8*>(
             do_lds + col * QROW + ks * 16 + 8 * hi);
-        dp_acc = mfma32_bf16(doa, vf[ks], dp_acc);
+        dp_acc = mfma32_bf16(doa, vf, dp_acc);
       }
       // NOTE the operand order: A = q/do frags have lane=q? No — see below.
       // A-frag lane holds row l&31 of the Q tile; with A=Q (M=q) the
@@ -892,7 +889,7 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
 }
 
 template <int D>
-__global__ __launch_bounds__(256, 1)
+__global__ __launch_bounds__(512, 2)
 void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           const __bf16* __restrict__ q,
                           const __bf16* __restrict__ k,
@@ -903,7 +900,7 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           __bf16* __restrict__ dv_exp,
                           int b, int sq, int skv, int hq, int hkv,
                           float scale, bool causal) {
-  constexpr int QT = 32, KBW = 128;
+  constexpr int QT = 32, KBW = 256;
   constexpr int BUFSZ = 2 * QT * (D + 8) + 2 * D * (QT + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
   __shared__ __align__(16) float lsedi[4 * QT];
@@ -1004,9 +1001,9 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
   dim3 gq(causal ? (nqb + 1) / 2 : nqb, b * hq);
   hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout, q,
                      k, v, lse, di, dq, b, sq, skv, hq, hkv, scale, causal);
-  int nkb = (skv + 127) / 128;
+  int nkb = (skv + 255) / 256;
   dim3 gkv(causal ? (nkb + 1) / 2 : nkb, b * hq);
-  hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(256), 0, st, dout,
+  hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st, dout,
                      q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv, hq, hkv,
                      scale, causal);
 }
