@@ -39,6 +39,8 @@ void ce_sum_target_launch_t(const T*, const long*, const float*, float*, float*,
 template <typename T>
 void ce_bwd_launch_t(T*, const long*, const float*, const float*, const float*, long, long, long, hipStream_t);
 
+void grouped_gemm_launch(const __bf16*, const __bf16*, __bf16*, const void*, int, int, int, long, bool, hipStream_t);
+void grouped_gemm_dw_launch(const __bf16*, const __bf16*, float*, const int*, int, int, int, hipStream_t);
 void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
@@ -346,6 +348,35 @@ void grad_accum(Tensor flat, const Tensor& grad, long offset) {
     grad_accum_launch_t<float>(fp, grad.data_ptr<float>(), n, cur_stream());
 }
 
+// ---- grouped GEMM (MoE experts) -------------------------------------------
+Tensor grouped_gemm(const Tensor& A, const Tensor& W, const Tensor& tiles,
+                    long n_out, bool trans_b) {
+  CHECK_GPU(A); CHECK_GPU(W); CHECK_GPU(tiles);
+  TORCH_CHECK(is_bf16(A) && is_bf16(W), "grouped_gemm: bf16 only");
+  const long M = A.size(0), K = A.size(1);
+  TORCH_CHECK(K % 32 == 0 && n_out % 128 == 0,
+              "grouped_gemm: K%32, N%128 required");
+  auto C = at::empty({M, n_out}, A.options());
+  if (M > 0)
+    grouped_gemm_launch(bfp(A), bfp(W), bfp_mut(C), tiles.data_ptr(),
+                        (int)tiles.size(0), (int)K, (int)n_out,
+                        W.size(1) * W.size(2), trans_b, cur_stream());
+  return C;
+}
+
+Tensor grouped_gemm_dw(const Tensor& A, const Tensor& dC,
+                       const Tensor& row_off, long E) {
+  CHECK_GPU(A); CHECK_GPU(dC); CHECK_GPU(row_off);
+  const long K = A.size(1), N = dC.size(1);
+  TORCH_CHECK(K % 128 == 0 && N % 128 == 0,
+              "grouped_gemm_dw: K%128 and N%128 required");
+  auto dW = at::empty({E, K, N}, A.options().dtype(at::kFloat));
+  grouped_gemm_dw_launch(bfp(A), bfp(dC), dW.data_ptr<float>(),
+                         row_off.data_ptr<int>(), (int)E, (int)K, (int)N,
+                         cur_stream());
+  return dW;
+}
+
 // ---- fused AdamW ----------------------------------------------------------
 void fused_adamw(std::vector<Tensor> masters, std::vector<Tensor> grads,
                  std::vector<Tensor> ms, std::vector<Tensor> vs,
@@ -400,5 +431,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("fused_adamw", &fused_adamw);
   m.def("grad_accum", &grad_accum);
+  m.def("grouped_gemm", &grouped_gemm);
+  m.def("grouped_gemm_dw", &grouped_gemm_dw);
   m.attr("arch") = "gfx950";
 }

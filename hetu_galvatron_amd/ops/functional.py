@@ -168,3 +168,60 @@ def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True, softmax_scale=Non
         return get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
                                              causal, scale)
     return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)
+
+
+def _gg_tiles(counts, device):
+    """Host-built [n_tiles, 3] int32 (expert, row0, rows) + row_off [E+1]."""
+    tiles = []
+    off = 0
+    row_off = [0]
+    for e, m in enumerate(counts):
+        m = int(m)
+        r = 0
+        while r < m:
+            tiles.append((e, off + r, min(128, m - r)))
+            r += 128
+        off += m
+        row_off.append(off)
+    td = torch.tensor(tiles if tiles else [(0, 0, 0)], dtype=torch.int32)
+    ro = torch.tensor(row_off, dtype=torch.int32)
+    return td.to(device, non_blocking=True), ro.to(device, non_blocking=True)
+
+
+class _GroupedGemm(torch.autograd.Function):
+    """C_e = A_e @ W[e] over expert-sorted rows (ops/csrc/grouped_gemm.hip).
+
+    Replaces the reference's external grouped_gemm CUDA dep
+    (moe/grouped_gemm_util.py). counts is a host list (the MoE dispatcher
+    already synchronized it for the all-to-all splits).
+    """
+
+    @staticmethod
+    def forward(ctx, a, w, counts):
+        tiles, row_off = _gg_tiles(counts, a.device)
+        c = get_ext(False).grouped_gemm(a, w, tiles, w.shape[2], False)
+        ctx.save_for_backward(a, w, tiles, row_off)
+        return c
+
+    @staticmethod
+    def backward(ctx, dc):
+        a, w, tiles, row_off = ctx.saved_tensors
+        dc = dc.contiguous()
+        ext = get_ext(False)
+        # dA = dC @ W^T ; dW = A^T dC
+        da = ext.grouped_gemm(dc, w, tiles, w.shape[1], True)
+        dw = ext.grouped_gemm_dw(a, dc, row_off, w.shape[0])
+        return da, dw.to(w.dtype), None
+
+
+def grouped_gemm(a: torch.Tensor, w: torch.Tensor, counts) -> torch.Tensor:
+    """a [M, K] bf16 expert-sorted; w [E, K, N]; counts: per-expert rows."""
+    return _GroupedGemm.apply(a.contiguous(), w, counts)
+
+
+def grouped_gemm_available(a, w) -> bool:
+    """fwd needs K%32, N%128; backward dW needs K%128 and N%128; dA needs
+    the reverse — so both dims of every W must be multiples of 128."""
+    return (a.is_cuda and a.dtype == torch.bfloat16
+            and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0
+            and use_native(a))

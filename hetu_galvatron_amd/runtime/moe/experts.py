@@ -19,6 +19,7 @@ import torch
 import torch.nn as nn
 
 from ...ops import swiglu
+from ...ops.functional import grouped_gemm, grouped_gemm_available
 
 
 def _mark_expert(p: nn.Parameter) -> nn.Parameter:
@@ -47,6 +48,11 @@ class GroupedMLP(nn.Module):
                 tokens_per_expert: torch.Tensor) -> torch.Tensor:
         """x [m, h] expert-sorted; tokens_per_expert [E_local]."""
         sizes = [int(v) for v in tokens_per_expert]
+        if grouped_gemm_available(x, self.w1) \
+                and grouped_gemm_available(x, self.w2) and x.shape[0] > 0:
+            h = grouped_gemm(x, self.w1, sizes)
+            h = swiglu(h) if self.gated else torch.nn.functional.gelu(h)
+            return grouped_gemm(h, self.w2, sizes)
         outs: List[torch.Tensor] = []
         start = 0
         for e, m in enumerate(sizes):

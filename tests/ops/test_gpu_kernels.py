@@ -263,3 +263,28 @@ def test_grad_accum():
     want = base.clone()
     want[123:123 + 4097] += g.float()
     assert_close(flat, want, 1e-5, what="grad_accum")
+
+
+@pytest.mark.parametrize("counts", [[100, 28, 0, 130], [128, 128, 128, 128]])
+def test_grouped_gemm(counts):
+    from hetu_galvatron_amd.ops.functional import grouped_gemm
+    torch.manual_seed(11)
+    E, K, N = 4, 256, 384
+    M = sum(counts)
+    a = (torch.randn(M, K, device=dev()) / 8).bfloat16().requires_grad_(True)
+    w = (torch.randn(E, K, N, device=dev()) / 8).bfloat16().requires_grad_(True)
+    c = grouped_gemm(a, w, counts)
+    # reference: per-expert mm in fp32
+    a32 = a.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    outs, s = [], 0
+    for e, m in enumerate(counts):
+        outs.append(a32[s:s + m] @ w32[e])
+        s += m
+    c_ref = torch.cat(outs)
+    assert_close(c, c_ref, 5e-2, what="grouped_gemm fwd")
+    dc = torch.randn_like(c)
+    c.backward(dc)
+    c_ref.backward(dc.float())
+    assert_close(a.grad, a32.grad, 6e-2, rtol=3e-2, what="grouped_gemm dA")
+    assert_close(w.grad, w32.grad, 6e-2, rtol=3e-2, what="grouped_gemm dW")
