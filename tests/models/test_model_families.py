@@ -16,9 +16,11 @@ def test_family_trains(name):
                   "lr": 5e-3, "lr_decay_style": "constant"},
     })
     torch.manual_seed(0)
-    m = GalvatronModel(cfg)
+    device = torch.device("cuda", 0) if torch.cuda.is_available() \
+        else torch.device("cpu")
+    m = GalvatronModel(cfg, device=device)
     opt, sched = get_optimizer_and_param_scheduler(m.stage_model, cfg)
-    it = get_train_iterator(cfg, torch.device("cpu"))
+    it = get_train_iterator(cfg, device)
     batch = next(it)  # overfit one batch: loss must drop
     losses = []
     for _ in range(8):
