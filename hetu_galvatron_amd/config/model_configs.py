@@ -91,6 +91,28 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         position_embedding_type="rope", rope_theta=1000000.0,
         num_experts=8, moe_router_topk=2,
     ),
+    # T5-3B (BASELINE config 4) — encoder-decoder; learned positions
+    # (T5's bucketized relative bias is a v2 flash-kernel item)
+    "t5-3b": dict(
+        model_type="t5", hidden_size=1024, num_hidden_layers=24,
+        num_decoder_layers=24, num_attention_heads=32,
+        num_key_value_heads=None, kv_channels=128, ffn_hidden_size=16384,
+        vocab_size=32128, max_position_embeddings=1024, seq_length=512,
+        encoder_seq_length=512, hidden_act="gelu",
+        normalization="layernorm", norm_epsilon=1e-6,
+        position_embedding_type="learned", add_bias_linear=False,
+        add_qkv_bias=False,
+    ),
+    "tiny-t5": dict(
+        model_type="t5", hidden_size=128, num_hidden_layers=2,
+        num_decoder_layers=2, num_attention_heads=2,
+        num_key_value_heads=None, kv_channels=64, ffn_hidden_size=256,
+        vocab_size=512, max_position_embeddings=256, seq_length=64,
+        encoder_seq_length=96, hidden_act="gelu",
+        normalization="layernorm", norm_epsilon=1e-6,
+        position_embedding_type="learned", add_bias_linear=False,
+        add_qkv_bias=False,
+    ),
     # tiny models for tests
     "tiny-llama": dict(
         model_type="llama", hidden_size=128, num_hidden_layers=2,

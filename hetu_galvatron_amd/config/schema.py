@@ -46,7 +46,7 @@ class ModelArgs(BaseModel):
     """Model architecture fields (reference: GalvatronModelArgs args_schema.py:51-177)."""
 
     model_name: str = "llama-3-8b"
-    model_type: str = "llama"  # llama | gpt | moe-llama (mixtral-style)
+    model_type: str = "llama"  # llama | gpt | moe-llama | t5 (enc-dec)
     hidden_size: int = 4096
     num_hidden_layers: int = 32
     num_attention_heads: int = 32
@@ -86,8 +86,14 @@ class ModelArgs(BaseModel):
     def kv_heads(self) -> int:
         return self.num_key_value_heads or self.num_attention_heads
 
+    num_decoder_layers: Optional[int] = None  # t5: decoder depth
+    encoder_seq_length: Optional[int] = None   # t5: encoder seq len
+    kv_channels: Optional[int] = None          # per-head dim override (t5 d_kv)
+
     @property
     def head_dim(self) -> int:
+        if self.kv_channels:
+            return self.kv_channels
         return self.hidden_size // self.num_attention_heads
 
 

@@ -76,6 +76,14 @@ def shard_for_rank(name: str, full: torch.Tensor, strategy, tp_rank: int,
     if "linear_proj.weight" in name or "fc2.weight" in name:
         C = full.shape[1]
         return full[:, r * C // t:(r + 1) * C // t]
+    if "linear_q." in name:  # cross-attention query: plain column-parallel
+        R = full.shape[0]
+        return full[r * R // t:(r + 1) * R // t]
+    if "linear_kv." in name:  # cross-attention fused [k; v]: slice halves
+        H = full.shape[0] // 2
+        kk, vv = full[:H], full[H:]
+        return torch.cat([kk[r * H // t:(r + 1) * H // t],
+                          vv[r * H // t:(r + 1) * H // t]], dim=0)
     if "fc1." in name:
         gated = margs.hidden_act in ("silu", "swiglu", "geglu")
         if gated:

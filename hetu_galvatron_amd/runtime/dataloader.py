@@ -50,6 +50,15 @@ def build_batch_context(tokens: torch.Tensor, device) -> Dict:
     }
 
 
+def build_enc_dec_batch_context(enc_ids: torch.Tensor,
+                                dec_tokens: torch.Tensor, device) -> Dict:
+    """t5: enc_ids [B, S_enc]; dec_tokens [B, S_dec+1]."""
+    ctx = build_batch_context(dec_tokens, device)
+    ctx["enc_input_ids"] = enc_ids.to(device).contiguous()
+    ctx["enc_seq_len"] = enc_ids.shape[1]
+    return ctx
+
+
 def get_train_iterator(cfg: GalvatronConfig, device,
                        global_batch: Optional[int] = None) -> Iterator[Dict]:
     """Yield batch contexts of the global batch size, cycling the dataset
@@ -66,6 +75,16 @@ def get_train_iterator(cfg: GalvatronConfig, device,
         ds = SyntheticCausalLMDataset(
             cfg.model.vocab_size, cfg.model.seq_length,
             size=max(cfg.data.synthetic_dataset_size, B), seed=cfg.train.seed)
+    if cfg.model.model_type == "t5":
+        s_enc = cfg.model.encoder_seq_length or cfg.model.seq_length
+        g = torch.Generator().manual_seed(cfg.train.seed + 77)
+        idx = 0
+        while True:
+            batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])
+            idx = (idx + B) % len(ds)
+            enc = torch.randint(0, cfg.model.vocab_size, (B, s_enc),
+                                generator=g)
+            yield build_enc_dec_batch_context(enc, batch, device)
     idx = 0
     while True:
         batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])

@@ -20,12 +20,15 @@ def resolve_plan(cfg: GalvatronConfig, world_size: int) -> HybridParallelPlan:
     """Searched-plan JSON if given, else GLOBAL-mode uniform plan
     (reference: hybrid_parallel_config.py:18 get_hybrid_parallel_configs_api)."""
     p = cfg.parallel
+    n_layers = cfg.model.num_hidden_layers
+    if cfg.model.model_type == "t5":
+        n_layers += cfg.model.num_decoder_layers or cfg.model.num_hidden_layers
     if p.galvatron_config_path:
         plan = HybridParallelPlan.load(p.galvatron_config_path)
     else:
         dp_type = "zero3" if p.sdp else p.default_dp_type
         plan = HybridParallelPlan.uniform(
-            num_layers=cfg.model.num_hidden_layers, world_size=world_size,
+            num_layers=n_layers, world_size=world_size,
             pp=p.pp_deg, tp=p.global_tp_deg, cp=p.global_cp_deg,
             use_sp=p.use_ulysses, dp_type=dp_type,
             checkpoint=bool(p.global_checkpoint),

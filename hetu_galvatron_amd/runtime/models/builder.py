@@ -27,12 +27,20 @@ from .modules import (
     GalvatronCausalLMHead, GalvatronDecoderLayer, GalvatronEmbedding,
     GalvatronFinalNorm,
 )
+from .encdec_modules import (GalvatronDecoderLayerX, GalvatronEncDecBridge,
+                             GalvatronEncoderLayer)
 from .moe_modules import GalvatronMoEDecoderLayer
 
 
 def build_causal_lm_arch(num_layers: int) -> List[str]:
     """reference: builder.py:111 build_causal_lm_arch."""
     return ["embedding"] + ["decoder"] * num_layers + ["final_norm", "lm_head"]
+
+
+def build_enc_dec_arch(num_enc: int, num_dec: int) -> List[str]:
+    """t5-style: plan layers map 1:1 to encoder then decoder layers."""
+    return (["embedding"] + ["encoder"] * num_enc + ["encdec_bridge"] +
+            ["decoder_x"] * num_dec + ["final_norm", "lm_head"])
 
 
 class LayerBlock(nn.Module):
@@ -83,7 +91,8 @@ class LayerBlock(nn.Module):
             x = _PostBackwardHook.apply(self.flat, x)
         if x is not None and self.flat_expert is not None and x.requires_grad:
             x = _PostBackwardHook.apply(self.flat_expert, x)
-        if self.checkpoint and self.kind == "decoder" and torch.is_grad_enabled():
+        if self.checkpoint and self.kind in ("decoder", "encoder") \
+                and torch.is_grad_enabled():
             out = torch_ckpt.checkpoint(
                 lambda t: self._inner_forward(t, ctx), x, use_reentrant=False)
         else:
@@ -178,7 +187,10 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
 
     torch.manual_seed(cfg.train.seed)  # deterministic per-rank module init
     if my_stage == 0:
-        emb = GalvatronEmbedding(margs, vg, dtype=dtype)
+        emb = GalvatronEmbedding(
+            margs, vg, dtype=dtype,
+            ids_key="enc_input_ids" if margs.model_type == "t5"
+            else "input_ids")
         emb_block = LayerBlock(emb, vg, "embedding")
         blocks.append(emb_block)
         prev = vg
@@ -188,15 +200,44 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         prev = gen_layer_comm_groups([prev_strategy], world, rank, cache)[0][0]
 
     is_moe = margs.model_type.startswith("moe") and margs.num_experts > 0
+    is_encdec = margs.model_type == "t5"
+    n_enc = margs.num_hidden_layers if is_encdec else 0
+    if is_encdec:
+        n_dec = margs.num_decoder_layers or margs.num_hidden_layers
+        assert n_layers == n_enc + n_dec, \
+            f"t5 plan must cover enc+dec layers ({n_enc}+{n_dec}), got {n_layers}" 
+    if is_encdec:
+        assert pp == 1, "t5 encoder-decoder: pp>1 is a v2 item"
+        dps = {plan.layer(i, world).dp for i in range(n_layers)}
+        assert len(dps) == 1, "t5: uniform dp degree required (memory " \
+            "replication domain); per-layer tp/zero/ckpt may still vary"
     for i in range(lo, hi):
         lg = layer_groups[i]
-        if is_moe:
+        if is_encdec:
+            if i < n_enc:
+                assert lg.strategy.cp == 1, "t5 encoder: cp is a v2 item"
+                dec = GalvatronEncoderLayer(margs, lg, layer_idx=i,
+                                            dtype=dtype)
+                kind = "encoder"
+            else:
+                dec = GalvatronDecoderLayerX(margs, lg, layer_idx=i,
+                                             dtype=dtype)
+                kind = "decoder"
+            if i == n_enc:  # first decoder layer: bridge precedes it
+                bridge = GalvatronEncDecBridge(margs, vg, dtype=dtype)
+                bblk = LayerBlock(bridge, vg, "encdec_bridge",
+                                  prev_groups=prev)
+                blocks.append(bblk)
+                prev = vg
+        elif is_moe:
             assert lg.strategy.tp == 1 or lg.strategy.use_ulysses, \
                 "MoE layers: expert-TP (etp) not supported yet; use ep/dp/sp"
             dec = GalvatronMoEDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
+            kind = "decoder"
         else:
             dec = GalvatronDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
-        blk = LayerBlock(dec, lg, "decoder",
+            kind = "decoder"
+        blk = LayerBlock(dec, lg, kind,
                          checkpoint=bool(plan.checkpoint_flags[i]),
                          prev_groups=prev)
         blocks.append(blk)

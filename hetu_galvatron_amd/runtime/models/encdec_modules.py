@@ -1,0 +1,180 @@
+"""Encoder-decoder (T5-style) layer modules.
+
+BASELINE config 4 ("T5-3B encoder-decoder auto-search") — beyond the
+reference's model zoo (gpt/llama/moe only).  Arch:
+
+  [embedding(enc ids)] + [encoder]*Ne + [encdec_bridge] +
+  [decoder_x]*Nd + [final_norm] + [lm_head]
+
+The bridge applies the encoder final norm, all-gathers the encoder output
+to the FULL sequence (replicated within the dp group: cross-attention
+then needs no per-layer redistribution however the decoder layers are
+laid out), stashes it in the batch context, and emits the decoder
+embedding.  v1 scope: pp=1, cross-attention under tp/dp/zero/ckpt
+(ulysses/cp on decoder layers are asserted off by the builder).
+
+Position encoding: learned absolute (T5's bucketized relative-position
+bias needs a bias input on the flash kernel — a v2 kernel item).
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ...config.schema import ModelArgs
+from ..redistribute import natural_rows
+from ..tensor_parallel import group_rank, group_size
+from ..tensor_parallel.mappings import gather_from_sequence_parallel_region
+from ..transformer import MLP, SelfAttention, build_norm
+from ..transformer.attention import CrossAttention
+from .modules import GalvatronEmbedding, _my_rank, _tag_tp_replicated
+
+
+class GalvatronEncoderLayer(nn.Module):
+    """Pre-norm bidirectional self-attention + MLP."""
+
+    def __init__(self, margs: ModelArgs, groups, layer_idx: int = 0,
+                 dtype=None):
+        super().__init__()
+        self.margs = margs
+        self.groups = groups
+        self.layer_idx = layer_idx
+        s = groups.strategy
+        self.strategy = s
+        seq_par = not s.use_ulysses
+        self.input_norm = build_norm(margs.normalization, margs.hidden_size,
+                                     margs.norm_epsilon, dtype)
+        self.post_attn_norm = build_norm(margs.normalization,
+                                         margs.hidden_size,
+                                         margs.norm_epsilon, dtype)
+        self.attention = SelfAttention(
+            margs, groups.tp_group, groups.sp_group, groups.cp_group,
+            use_ulysses=s.use_ulysses, sequence_parallel=seq_par,
+            dtype=dtype, causal=False)
+        self.mlp = MLP(margs.hidden_size, margs.ffn_hidden_size,
+                       None if s.use_ulysses else groups.tp_group,
+                       hidden_act="silu"
+                       if margs.hidden_act in ("silu", "swiglu")
+                       else margs.hidden_act,
+                       add_bias=margs.add_bias_linear,
+                       sequence_parallel=seq_par, dtype=dtype)
+        if seq_par and group_size(groups.tp_group) > 1:
+            _tag_tp_replicated(self.input_norm.weight,
+                               getattr(self.input_norm, "bias", None),
+                               self.post_attn_norm.weight,
+                               getattr(self.post_attn_norm, "bias", None),
+                               self.attention.linear_proj.bias,
+                               self.mlp.fc2.bias)
+        self.dropout_p = margs.hidden_dropout
+
+    def forward(self, hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
+        residual = hidden
+        x = self.input_norm(hidden)
+        x = self.attention(x, None, None)  # no RoPE (learned positions)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        hidden = residual + x
+        residual = hidden
+        x = self.post_attn_norm(hidden)
+        x = self.mlp(x)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        return residual + x
+
+
+class GalvatronEncDecBridge(nn.Module):
+    """Encoder final norm -> full-seq memory in ctx -> decoder embedding."""
+
+    def __init__(self, margs: ModelArgs, groups, dtype=None):
+        super().__init__()
+        self.margs = margs
+        self.groups = groups
+        self.strategy = groups.strategy
+        self.enc_final_norm = build_norm(margs.normalization,
+                                         margs.hidden_size,
+                                         margs.norm_epsilon, dtype)
+        self.dec_embedding = GalvatronEmbedding(margs, groups, dtype=dtype,
+                                                ids_key="input_ids")
+
+    def forward(self, enc_hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
+        mem = self.enc_final_norm(enc_hidden)
+        s = self.strategy
+        # encoder output arrives in this block's layout (seq / tsp*cp after
+        # LayerBlock redistribution); gather to the full sequence
+        group = self.groups.tsp_cp_group
+        if group is not None and group_size(group) > 1:
+            mem = gather_from_sequence_parallel_region(mem, group)
+        ctx["encoder_memory"] = mem            # [S_enc, b_loc, h]
+        return self.dec_embedding(ctx)
+
+
+class GalvatronDecoderLayerX(nn.Module):
+    """Pre-norm causal self-attention + cross-attention + MLP."""
+
+    def __init__(self, margs: ModelArgs, groups, layer_idx: int = 0,
+                 dtype=None):
+        super().__init__()
+        self.margs = margs
+        self.groups = groups
+        self.layer_idx = layer_idx
+        s = groups.strategy
+        self.strategy = s
+        assert not s.use_ulysses and s.cp == 1, \
+            "t5 decoder layers: ulysses/cp on cross-attention is v2"
+        seq_par = True
+        self.input_norm = build_norm(margs.normalization, margs.hidden_size,
+                                     margs.norm_epsilon, dtype)
+        self.cross_norm = build_norm(margs.normalization, margs.hidden_size,
+                                     margs.norm_epsilon, dtype)
+        self.post_attn_norm = build_norm(margs.normalization,
+                                         margs.hidden_size,
+                                         margs.norm_epsilon, dtype)
+        self.attention = SelfAttention(
+            margs, groups.tp_group, groups.sp_group, groups.cp_group,
+            use_ulysses=False, sequence_parallel=seq_par, dtype=dtype,
+            causal=True)
+        self.cross_attention = CrossAttention(
+            margs, groups.tp_group, sequence_parallel=seq_par, dtype=dtype)
+        self.mlp = MLP(margs.hidden_size, margs.ffn_hidden_size,
+                       groups.tp_group,
+                       hidden_act="silu"
+                       if margs.hidden_act in ("silu", "swiglu")
+                       else margs.hidden_act,
+                       add_bias=margs.add_bias_linear,
+                       sequence_parallel=seq_par, dtype=dtype)
+        if seq_par and group_size(groups.tp_group) > 1:
+            _tag_tp_replicated(self.input_norm.weight,
+                               getattr(self.input_norm, "bias", None),
+                               self.cross_norm.weight,
+                               getattr(self.cross_norm, "bias", None),
+                               self.post_attn_norm.weight,
+                               getattr(self.post_attn_norm, "bias", None),
+                               self.attention.linear_proj.bias,
+                               self.cross_attention.linear_proj.bias,
+                               self.mlp.fc2.bias)
+        self.dropout_p = margs.hidden_dropout
+
+    def forward(self, hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
+        residual = hidden
+        x = self.input_norm(hidden)
+        x = self.attention(x, None, None)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        hidden = residual + x
+
+        residual = hidden
+        x = self.cross_norm(hidden)
+        x = self.cross_attention(x, ctx["encoder_memory"])
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        hidden = residual + x
+
+        residual = hidden
+        x = self.post_attn_norm(hidden)
+        x = self.mlp(x)
+        if self.dropout_p > 0 and self.training:
+            x = F.dropout(x, self.dropout_p)
+        return residual + x

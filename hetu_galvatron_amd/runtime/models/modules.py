@@ -47,10 +47,12 @@ class GalvatronEmbedding(nn.Module):
     """Vocab-parallel embedding (+ optional learned positions) producing this
     rank's SBH activation shard (reference: modules.py:35-102)."""
 
-    def __init__(self, margs: ModelArgs, groups: LayerCommGroups, dtype=None):
+    def __init__(self, margs: ModelArgs, groups: LayerCommGroups, dtype=None,
+                 ids_key: str = "input_ids"):
         super().__init__()
         self.margs = margs
         self.groups = groups
+        self.ids_key = ids_key
         s = groups.strategy
         self.strategy = s
         if s.use_ulysses:
@@ -71,7 +73,7 @@ class GalvatronEmbedding(nn.Module):
         self.dropout_p = margs.hidden_dropout
 
     def forward(self, ctx: Dict) -> torch.Tensor:
-        ids = ctx["input_ids"]
+        ids = ctx[self.ids_key]
         B, S = ids.shape
         c = self.groups.coord_of(_my_rank())
         s = self.strategy

@@ -40,6 +40,9 @@ def chunk_batch(ctx: Dict, chunks: int) -> List[Dict]:
         sub = dict(ctx)
         sub["input_ids"] = ctx["input_ids"][m * b_mb:(m + 1) * b_mb]
         sub["labels"] = ctx["labels"][m * b_mb:(m + 1) * b_mb]
+        if "enc_input_ids" in ctx:
+            sub["enc_input_ids"] = \
+                ctx["enc_input_ids"][m * b_mb:(m + 1) * b_mb]
         sub["batch_size"] = b_mb
         out.append(sub)
     return out

@@ -31,8 +31,9 @@ from .rope import apply_rope_qk
 class SelfAttention(nn.Module):
     def __init__(self, model_args, tp_group, sp_group, cp_group,
                  use_ulysses: bool = False, sequence_parallel: bool = True,
-                 dtype=None):
+                 dtype=None, causal: bool = True):
         super().__init__()
+        self.causal = causal
         m = model_args
         self.hidden_size = m.hidden_size
         self.num_heads = m.num_attention_heads
@@ -90,10 +91,66 @@ class SelfAttention(nn.Module):
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()
         if self.core_attention is not None:
-            o = self.core_attention(q, k, v, causal=True,
+            o = self.core_attention(q, k, v, causal=self.causal,
                                     softmax_scale=self.softmax_scale)
         else:
-            o = local_attention(q, k, v, causal=True,
+            o = local_attention(q, k, v, causal=self.causal,
                                 softmax_scale=self.softmax_scale)
         o = o.permute(1, 0, 2, 3).reshape(s, b, -1)  # back to SBH
+        return self.linear_proj(o)
+
+
+class CrossAttention(nn.Module):
+    """Decoder->encoder cross attention (reference: attention.py:929
+    CrossAttention).  Queries come from the (possibly seq-sharded) decoder
+    hidden state; keys/values from the FULL-sequence encoder memory the
+    EncDecBridge replicated within the dp group, so no redistribution is
+    needed however the decoder layer is laid out (tp heads shard both q
+    and kv; cp/ulysses on cross-attention are a v2 item — the builder
+    asserts them off for t5).
+    """
+
+    def __init__(self, model_args, tp_group, sequence_parallel: bool = True,
+                 dtype=None):
+        super().__init__()
+        m = model_args
+        self.num_heads = m.num_attention_heads
+        self.num_kv_heads = m.kv_heads
+        self.head_dim = m.head_dim
+        self.tp_group = tp_group
+        tp = group_size(tp_group)
+        assert self.num_heads % tp == 0 and self.num_kv_heads % tp == 0
+        self.linear_q = ColumnParallelLinear(
+            m.hidden_size, self.num_heads * self.head_dim, tp_group,
+            bias=m.add_qkv_bias or m.add_bias_linear,
+            sequence_parallel=sequence_parallel, dtype=dtype)
+        # kv over the full-seq memory: no sequence-parallel gather
+        self.linear_kv = ColumnParallelLinear(
+            m.hidden_size, 2 * self.num_kv_heads * self.head_dim, tp_group,
+            bias=m.add_qkv_bias or m.add_bias_linear,
+            sequence_parallel=False, dtype=dtype)
+        self.linear_proj = RowParallelLinear(
+            self.num_heads * self.head_dim, m.hidden_size, tp_group,
+            bias=m.add_bias_linear, sequence_parallel=sequence_parallel,
+            dtype=dtype)
+        self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
+        tp = group_size(tp_group)
+        self.heads_local = self.num_heads // tp
+        self.kv_heads_local = self.num_kv_heads // tp
+
+    def forward(self, x: torch.Tensor, memory: torch.Tensor) -> torch.Tensor:
+        """x: [s_dec(_shard), b, h]; memory: [s_enc, b, h] full sequence."""
+        q = self.linear_q(x)            # [s_dec, b, hq_l*d]
+        kv = self.linear_kv(memory)     # [s_enc, b, 2*hkv_l*d]
+        s, bsz = q.shape[0], q.shape[1]
+        se = kv.shape[0]
+        q = q.view(s, bsz, self.heads_local, self.head_dim)
+        kv = kv.view(se, bsz, 2, self.kv_heads_local, self.head_dim)
+        k, v = kv[:, :, 0], kv[:, :, 1]
+        q = q.permute(1, 0, 2, 3).contiguous()
+        k = k.permute(1, 0, 2, 3).contiguous()
+        v = v.permute(1, 0, 2, 3).contiguous()
+        o = local_attention(q, k, v, causal=False,
+                            softmax_scale=self.softmax_scale)
+        o = o.permute(1, 0, 2, 3).reshape(s, bsz, -1)
         return self.linear_proj(o)
