@@ -37,5 +37,6 @@ def test_presets_resolve():
     for name in MODEL_PRESETS:
         cfg = load_config(base={"model": {"model_name": name}})
         assert cfg.model.hidden_size > 0
-        assert cfg.model.head_dim * cfg.model.num_attention_heads \
-            == cfg.model.hidden_size
+        if not cfg.model.kv_channels:
+            assert cfg.model.head_dim * cfg.model.num_attention_heads \
+                == cfg.model.hidden_size
