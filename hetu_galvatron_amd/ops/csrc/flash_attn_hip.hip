@@ -170,7 +170,6 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
     if (kv0 < kv_last_w) {  // per-wave causal skip (wave-uniform)
       // ---- S^T = K Q^T (two 32-key tiles) ----
       f32x16 st0 = (f32x16)(0.f), st1 = (f32x16)(0.f);
-      __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
 #pragma unroll
       for (int ks = 0; ks < NK; ++ks) {
         bf16x8 k0 = *reinterpret_cast<const bf16x8*>(
@@ -180,7 +179,6 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
         st0 = mfma32_bf16(k0, qf[ks], st0);
         st1 = mfma32_bf16(k1, qf[ks], st1);
       }
-      __builtin_amdgcn_s_setprio(0);
 
       // ---- scale (+ mask on straddle/tail tiles) ----
       float pv[32];
@@ -251,7 +249,6 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
       }
 
       // ---- O^T += V^T P^T ----
-      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
         union { unsigned u[4]; bf16x8 f; } pb;
@@ -264,7 +261,6 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
           ot[dt] = mfma32_bf16(va, pb.f, ot[dt]);
         }
       }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     if (have_next) stage_write(cur ^ 1);
