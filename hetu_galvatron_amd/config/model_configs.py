@@ -116,14 +116,16 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
     # tiny models for tests
     "tiny-llama": dict(
         model_type="llama", hidden_size=128, num_hidden_layers=2,
-        num_attention_heads=4, num_key_value_heads=2, ffn_hidden_size=256,
+        num_attention_heads=2, num_key_value_heads=2, kv_channels=64,
+        ffn_hidden_size=256,
         vocab_size=512, max_position_embeddings=256, seq_length=128,
         hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-5,
         position_embedding_type="rope", rope_theta=10000.0,
     ),
     "tiny-gpt": dict(
         model_type="gpt", hidden_size=128, num_hidden_layers=2,
-        num_attention_heads=4, num_key_value_heads=None, ffn_hidden_size=512,
+        num_attention_heads=2, num_key_value_heads=None, kv_channels=64,
+        ffn_hidden_size=512,
         vocab_size=512, max_position_embeddings=256, seq_length=128,
         hidden_act="gelu", normalization="layernorm", norm_epsilon=1e-5,
         position_embedding_type="learned", add_bias_linear=True,
@@ -132,7 +134,8 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
     ),
     "tiny-moe": dict(
         model_type="moe-llama", hidden_size=128, num_hidden_layers=2,
-        num_attention_heads=4, num_key_value_heads=2, ffn_hidden_size=256,
+        num_attention_heads=2, num_key_value_heads=2, kv_channels=64,
+        ffn_hidden_size=256,
         vocab_size=512, max_position_embeddings=256, seq_length=128,
         hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-5,
         position_embedding_type="rope", rope_theta=10000.0,

@@ -123,7 +123,7 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, softmax_scale):
         scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        if use_native(q) and q.dtype == torch.bfloat16:
+        if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale)
         else:
             o, lse = ref.attention_fwd(q, k, v, causal, scale)
@@ -136,7 +136,7 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do, dlse):
         q, k, v, o, lse = ctx.saved_tensors
         do = do.contiguous()
-        if use_native(q) and q.dtype == torch.bfloat16:
+        if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             dq, dk, dv = get_ext(False).flash_attn_bwd(
                 do, q, k, v, o, lse, ctx.causal, ctx.scale)
         else:
@@ -156,7 +156,7 @@ def flash_attention(q, k, v, causal: bool = True,
 def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None):
     """No-autograd forward returning (o, lse) — building block for ring CP."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if use_native(q) and q.dtype == torch.bfloat16:
+    if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
         return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
                                              v.contiguous(), causal, scale)
     return ref.attention_fwd(q, k, v, causal, scale)
@@ -164,7 +164,7 @@ def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None):
 
 def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True, softmax_scale=None):
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if use_native(q) and q.dtype == torch.bfloat16:
+    if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
         return get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
                                              causal, scale)
     return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)

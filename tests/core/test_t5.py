@@ -64,9 +64,11 @@ def test_t5_overfits_one_batch():
         GalvatronModel, get_optimizer_and_param_scheduler, get_train_iterator)
     cfg = make_cfg({"train": {"lr": 5e-3}})
     torch.manual_seed(0)
-    m = GalvatronModel(cfg)
+    device = torch.device("cuda", 0) if torch.cuda.is_available() \
+        else torch.device("cpu")
+    m = GalvatronModel(cfg, device=device)
     opt, sched = get_optimizer_and_param_scheduler(m.stage_model, cfg)
-    it = get_train_iterator(cfg, torch.device("cpu"))
+    it = get_train_iterator(cfg, device)
     batch = next(it)
     losses = []
     for _ in range(8):
