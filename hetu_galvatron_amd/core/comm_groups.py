@@ -26,7 +26,6 @@ from __future__ import annotations
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Tuple
 
-import torch
 import torch.distributed as dist
 
 from ..config.strategy import LayerStrategy

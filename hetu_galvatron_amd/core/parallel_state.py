@@ -7,7 +7,7 @@ the 288 GB HBM pool fragmentation-free under per-layer-varying shard sizes.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 import torch
 

@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import importlib
 import os
-from typing import Optional
 
 _EXT = None
 _TRIED = False

@@ -13,7 +13,7 @@ Every op here replaces an external CUDA dependency of the reference:
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -10,9 +10,8 @@ a ColumnParallelLinear contiguous shard is whole groups.
 """
 from __future__ import annotations
 
-import json
 import os
-from typing import Dict, Iterable, Optional
+from typing import Dict
 
 import torch
 

@@ -7,8 +7,7 @@ blended_megatron_dataset_*.py — sample index built by the C++ helper
 """
 from __future__ import annotations
 
-import hashlib
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 import numpy as np
 import torch

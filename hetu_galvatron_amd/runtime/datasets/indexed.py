@@ -9,12 +9,10 @@ indexed_dataset.py (mmap .bin/.idx document store).  Own, simpler format:
 """
 from __future__ import annotations
 
-import os
 import struct
 from typing import List, Sequence
 
 import numpy as np
-import torch
 
 MAGIC = b"GALVIDX1"
 DTYPES = {1: np.uint16, 2: np.int32, 3: np.int64}

@@ -12,7 +12,7 @@ import torch.distributed as dist
 import torch.nn as nn
 
 from ..config import GalvatronConfig, HybridParallelPlan
-from .models.builder import StageModel, build_hybrid_parallel_model
+from .models.builder import build_hybrid_parallel_model
 from .pipeline.engine import PipelineEngine, StepStats
 
 
@@ -27,12 +27,23 @@ def resolve_plan(cfg: GalvatronConfig, world_size: int) -> HybridParallelPlan:
         plan = HybridParallelPlan.load(p.galvatron_config_path)
     else:
         dp_type = "zero3" if p.sdp else p.default_dp_type
+        chunks = p.chunks
+        if chunks is None or chunks < 1:
+            # auto heuristic (reference: hybrid_parallel_config.py:359
+            # get_chunks): enough microbatches to fill the pipeline, at
+            # least 2 per-stage for grad-accum overlap, bounded by the
+            # per-dp-rank batch
+            dp = world_size // max(p.pp_deg * p.global_tp_deg
+                                   * p.global_cp_deg, 1)
+            per_dp = max(cfg.train.global_train_batch_size // max(dp, 1), 1)
+            chunks = min(max(2 * p.pp_deg, 2), per_dp) if p.pp_deg > 1 \
+                else min(2, per_dp)
         plan = HybridParallelPlan.uniform(
             num_layers=n_layers, world_size=world_size,
             pp=p.pp_deg, tp=p.global_tp_deg, cp=p.global_cp_deg,
             use_sp=p.use_ulysses, dp_type=dp_type,
             checkpoint=bool(p.global_checkpoint),
-            chunks=max(p.chunks, 1), global_bsz=cfg.train.global_train_batch_size,
+            chunks=chunks, global_bsz=cfg.train.global_train_batch_size,
             pipeline_type=p.pipeline_type, vtp=p.vocab_tp,
             vsp=bool(p.vocab_sp), ep=p.global_ep_deg)
     if plan.global_bsz != cfg.train.global_train_batch_size:

@@ -7,7 +7,7 @@ Steps here: plan -> strategies -> comm groups (collective) -> stage modules
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -16,10 +16,13 @@ import torch.nn as nn
 import torch.utils.checkpoint as torch_ckpt
 
 from ...config import GalvatronConfig, HybridParallelPlan, even_pp_division
-from ...config.strategy import LayerStrategy
 from ...core.comm_groups import (
-    CommGroup, CommGroupCache, LayerCommGroups, gen_embedding_group,
-    gen_layer_comm_groups, pp_neighbor_ranks, pp_stage_of_rank,
+    CommGroup,
+    CommGroupCache,
+    LayerCommGroups,
+    gen_embedding_group,
+    gen_layer_comm_groups,
+    pp_stage_of_rank,
 )
 from ..redistribute import redistribute
 from ..zero import FlatParamBlock, _PostBackwardHook, _PreBackwardGather

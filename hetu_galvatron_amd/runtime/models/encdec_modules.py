@@ -25,12 +25,11 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ...config.schema import ModelArgs
-from ..redistribute import natural_rows
-from ..tensor_parallel import group_rank, group_size
+from ..tensor_parallel import group_size
 from ..tensor_parallel.mappings import gather_from_sequence_parallel_region
 from ..transformer import MLP, SelfAttention, build_norm
 from ..transformer.attention import CrossAttention
-from .modules import GalvatronEmbedding, _my_rank, _tag_tp_replicated
+from .modules import GalvatronEmbedding, _tag_tp_replicated
 
 
 class GalvatronEncoderLayer(nn.Module):

@@ -11,8 +11,7 @@ seq_len S.
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 import torch.distributed as dist
@@ -20,9 +19,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ...config.schema import ModelArgs
-from ...config.strategy import LayerStrategy
 from ...core.comm_groups import LayerCommGroups
-from ..redistribute import natural_rows, redistribute
+from ..redistribute import natural_rows
 from ..tensor_parallel import (
     ColumnParallelLinear, VocabParallelEmbedding, group_rank, group_size,
     vocab_parallel_cross_entropy,

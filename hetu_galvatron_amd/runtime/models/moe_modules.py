@@ -25,7 +25,7 @@ from ..moe.router import TopKRouter, attach_aux_loss
 from ..transformer import RotaryEmbedding
 from ..transformer.attention import SelfAttention
 from ..transformer.norm import build_norm
-from .modules import _my_rank, _tag_tp_replicated
+from .modules import _my_rank
 
 
 class GalvatronMoEMLP(nn.Module):

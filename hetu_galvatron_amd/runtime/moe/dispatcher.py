@@ -11,12 +11,12 @@ per-expert sends (xGMI rings are per-link bound; fewer+bigger wins).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import torch
 import torch.distributed as dist
 
-from ..tensor_parallel.mappings import _is_gloo, group_rank, group_size
+from ..tensor_parallel.mappings import _is_gloo
 
 
 class _AllToAll(torch.autograd.Function):

@@ -13,7 +13,7 @@ their grads over the EDP group (reference parallel.py MoE double-wrap).
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn

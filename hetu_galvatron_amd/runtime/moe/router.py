@@ -7,7 +7,7 @@ capacity-factor dropping) and moe_utils.py:14-240.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn as nn

@@ -13,7 +13,6 @@ once world-wide), one scalar all-reduce, then a multi-tensor scale.
 """
 from __future__ import annotations
 
-import math
 from typing import List, Optional
 
 import torch

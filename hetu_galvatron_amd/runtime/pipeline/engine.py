@@ -14,9 +14,7 @@ from typing import Dict, List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
-from ...config import HybridParallelPlan
 from ..models.builder import StageModel
-from ..redistribute import natural_rows
 from . import p2p
 
 

@@ -9,7 +9,7 @@ time).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

@@ -12,7 +12,6 @@ activation memory and MI355X's 288 GB still appreciates not duplicating it.
 """
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -14,7 +14,6 @@ MI355X design notes:
 """
 from __future__ import annotations
 
-import math
 from typing import Callable, Optional
 
 import torch
@@ -23,12 +22,16 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .mappings import (
-    _gather_along_first_dim, _reduce_scatter_along_first_dim, _is_gloo,
-    all_reduce_sync, copy_to_tensor_model_parallel_region,
-    gather_from_tensor_model_parallel_region, group_rank, group_size,
+    _gather_along_first_dim,
+    _reduce_scatter_along_first_dim,
+    _is_gloo,
+    all_reduce_sync,
+    copy_to_tensor_model_parallel_region,
+    gather_from_tensor_model_parallel_region,
+    group_rank,
+    group_size,
     reduce_from_tensor_model_parallel_region,
     reduce_scatter_to_sequence_parallel_region,
-    scatter_to_sequence_parallel_region,
 )
 
 

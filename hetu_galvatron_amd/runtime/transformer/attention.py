@@ -14,13 +14,14 @@ GQA interleaved-group QKV layout: the fused weight is ordered by KV group
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn
 
 from ..tensor_parallel import (
-    ColumnParallelLinear, RowParallelLinear, divide, group_rank, group_size,
+    ColumnParallelLinear,
+    RowParallelLinear,
+    group_size,
 )
 from .attention_impl import (
     DistributedAttention, ZigzagRingAttention, local_attention,

@@ -32,9 +32,8 @@ bound, so fewer+bigger collectives is the right shape.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

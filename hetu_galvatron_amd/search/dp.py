@@ -12,7 +12,7 @@ buffers handed in from numpy while the back-trace stays in Python.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 

@@ -20,7 +20,6 @@ both simpler and stage-exact for 1F1B.
 from __future__ import annotations
 
 import json
-import math
 import os
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
@@ -29,11 +28,17 @@ import numpy as np
 
 from ..config import GalvatronConfig
 from ..config.strategy import HybridParallelPlan, LayerStrategy, write_json_config
-from .costmodel import (HardwareProfile, LayerProfile, OtherProfile,
-                        layer_memory_cost, layer_p2p_cost, layer_time_cost,
-                        pipeline_cost, stage_sums, zero3_ratio)
+from .costmodel import (
+    HardwareProfile,
+    LayerProfile,
+    OtherProfile,
+    layer_memory_cost,
+    layer_p2p_cost,
+    layer_time_cost,
+    pipeline_cost,
+)
 from .dp import backtrace, solve_layer_dp
-from .strategies import enumerate_strategies, strategy_key, transition_cost_mb
+from .strategies import enumerate_strategies, transition_cost_mb
 
 
 def fit_linear(xs: List[float], ys: List[float]) -> Tuple[float, float]:

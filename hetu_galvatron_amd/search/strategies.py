@@ -6,8 +6,7 @@ dynamic_programming.py:161-210 (match_strategy redistribution penalty).
 """
 from __future__ import annotations
 
-from dataclasses import replace
-from typing import List, Optional
+from typing import List
 
 from ..config.schema import SearchArgs
 from ..config.strategy import LayerStrategy
