@@ -168,6 +168,9 @@ def main():
 
     tokens = args.steps * global_batch * args.seq_len
     value = tokens / elapsed
+    if rank == 0 and torch.cuda.is_available():
+        peak = torch.cuda.max_memory_allocated() / (1 << 30)
+        print(f"[bench] peak GPU memory: {peak:.1f} GiB", file=sys.stderr)
     if rank == 0:
         print(json.dumps({
             "metric": "tokens/sec (whole node), auto-searched plan, "
