@@ -41,6 +41,16 @@ void ce_bwd_launch_t(T*, const long*, const float*, const float*, const float*, 
 
 void grouped_gemm_launch(const __bf16*, const __bf16*, __bf16*, const void*, int, int, int, long, bool, hipStream_t);
 void grouped_gemm_dw_launch(const __bf16*, const __bf16*, float*, const int*, int, int, int, hipStream_t);
+#define DECL_MOE(T) \
+  void moe_permute_launch_##T(const T*, T*, const long*, long, int, hipStream_t); \
+  void moe_permute_bwd_launch_##T(const T*, float*, const long*, long, int, hipStream_t); \
+  void moe_unpermute_launch_##T(const T*, const float*, const long*, T*, long, int, int, hipStream_t); \
+  void moe_unpermute_bwd_launch_##T(const T*, const T*, const float*, const long*, T*, float*, long, int, hipStream_t);
+typedef __bf16 bf16_t;
+typedef float f32_t;
+DECL_MOE(bf16_t)
+DECL_MOE(f32_t)
+
 void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
@@ -348,6 +358,90 @@ void grad_accum(Tensor flat, const Tensor& grad, long offset) {
     grad_accum_launch_t<float>(fp, grad.data_ptr<float>(), n, cur_stream());
 }
 
+// ---- MoE permute / unpermute ----------------------------------------------
+Tensor moe_permute(const Tensor& x, const Tensor& rows) {
+  CHECK_GPU(x); CHECK_GPU(rows);
+  const long m = rows.numel();
+  const int h = x.size(-1);
+  TORCH_CHECK(h % 8 == 0, "moe_permute: h % 8");
+  auto y = at::empty({m, (long)h}, x.options());
+  if (m == 0) return y;
+  if (is_bf16(x))
+    moe_permute_launch_bf16_t(bfp(x), bfp_mut(y), rows.data_ptr<long>(), m,
+                              h, cur_stream());
+  else
+    moe_permute_launch_f32_t(x.data_ptr<float>(), y.data_ptr<float>(),
+                             rows.data_ptr<long>(), m, h, cur_stream());
+  return y;
+}
+
+Tensor moe_permute_bwd(const Tensor& dy, const Tensor& rows, long n) {
+  CHECK_GPU(dy);
+  const long m = rows.numel();
+  const int h = dy.size(-1);
+  auto acc = at::zeros({n, (long)h}, dy.options().dtype(at::kFloat));
+  if (m > 0) {
+    if (is_bf16(dy))
+      moe_permute_bwd_launch_bf16_t(bfp(dy), acc.data_ptr<float>(),
+                                    rows.data_ptr<long>(), m, h,
+                                    cur_stream());
+    else
+      moe_permute_bwd_launch_f32_t(dy.data_ptr<float>(),
+                                   acc.data_ptr<float>(),
+                                   rows.data_ptr<long>(), m, h,
+                                   cur_stream());
+  }
+  return acc.to(dy.scalar_type());
+}
+
+Tensor moe_unpermute(const Tensor& back, const Tensor& probs,
+                     const Tensor& inv_order, long n, long k) {
+  CHECK_GPU(back);
+  const int h = back.size(-1);
+  auto out = at::empty({n, (long)h}, back.options());
+  if (n == 0) return out;
+  auto p = probs.contiguous().to(at::kFloat);
+  if (is_bf16(back))
+    moe_unpermute_launch_bf16_t(bfp(back), p.data_ptr<float>(),
+                                inv_order.data_ptr<long>(), bfp_mut(out), n,
+                                (int)k, h, cur_stream());
+  else
+    moe_unpermute_launch_f32_t(back.data_ptr<float>(), p.data_ptr<float>(),
+                               inv_order.data_ptr<long>(),
+                               out.data_ptr<float>(), n, (int)k, h,
+                               cur_stream());
+  return out;
+}
+
+std::tuple<Tensor, Tensor> moe_unpermute_bwd(const Tensor& dout,
+                                             const Tensor& back,
+                                             const Tensor& probs,
+                                             const Tensor& t_of) {
+  CHECK_GPU(dout); CHECK_GPU(back);
+  const long m = back.size(0);
+  const int h = back.size(-1);
+  auto dback = at::empty_like(back);
+  auto dprobs = at::zeros({m}, back.options().dtype(at::kFloat));
+  auto p = probs.contiguous().to(at::kFloat);
+  if (m > 0) {
+    if (is_bf16(back))
+      moe_unpermute_bwd_launch_bf16_t(bfp(dout), bfp(back),
+                                      p.data_ptr<float>(),
+                                      t_of.data_ptr<long>(), bfp_mut(dback),
+                                      dprobs.data_ptr<float>(), m, h,
+                                      cur_stream());
+    else
+      moe_unpermute_bwd_launch_f32_t(dout.data_ptr<float>(),
+                                     back.data_ptr<float>(),
+                                     p.data_ptr<float>(),
+                                     t_of.data_ptr<long>(),
+                                     dback.data_ptr<float>(),
+                                     dprobs.data_ptr<float>(), m, h,
+                                     cur_stream());
+  }
+  return {dback, dprobs};
+}
+
 // ---- grouped GEMM (MoE experts) -------------------------------------------
 Tensor grouped_gemm(const Tensor& A, const Tensor& W, const Tensor& tiles,
                     long n_out, bool trans_b) {
@@ -432,6 +526,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw);
   m.def("grad_accum", &grad_accum);
   m.def("grouped_gemm", &grouped_gemm);
+  m.def("moe_permute", &moe_permute);
+  m.def("moe_permute_bwd", &moe_permute_bwd);
+  m.def("moe_unpermute", &moe_unpermute);
+  m.def("moe_unpermute_bwd", &moe_unpermute_bwd);
   m.def("grouped_gemm_dw", &grouped_gemm_dw);
   m.attr("arch") = "gfx950";
 }

@@ -225,3 +225,59 @@ def grouped_gemm_available(a, w) -> bool:
     return (a.is_cuda and a.dtype == torch.bfloat16
             and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0
             and use_native(a))
+
+
+class _MoEPermute(torch.autograd.Function):
+    """Expert-sorted row gather (ops/csrc/moe_permute.hip); backward
+    scatter-adds repeated source rows through an fp32 accumulator."""
+
+    @staticmethod
+    def forward(ctx, x, rows):
+        ctx.save_for_backward(rows)
+        ctx.n = x.shape[0]
+        return get_ext(False).moe_permute(x, rows)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (rows,) = ctx.saved_tensors
+        return get_ext(False).moe_permute_bwd(dy.contiguous(), rows,
+                                              ctx.n), None
+
+
+def moe_permute(x: torch.Tensor, rows: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda and native_available() and x.shape[-1] % 8 == 0 \
+            and x.dtype in (torch.bfloat16, torch.float32):
+        return _MoEPermute.apply(x.contiguous(), rows.contiguous())
+    return x[rows]
+
+
+class _MoEUnpermute(torch.autograd.Function):
+    """Probability-weighted top-k merge back to token order."""
+
+    @staticmethod
+    def forward(ctx, back, probs, order, n, k):
+        inv = torch.argsort(order)
+        t_of = torch.div(order, k, rounding_mode="floor")
+        ctx.save_for_backward(back, probs, t_of)
+        return get_ext(False).moe_unpermute(back, probs, inv, n, k)
+
+    @staticmethod
+    def backward(ctx, dout):
+        back, probs, t_of = ctx.saved_tensors
+        dback, dprobs = get_ext(False).moe_unpermute_bwd(
+            dout.contiguous(), back, probs, t_of)
+        return dback, dprobs.to(probs.dtype), None, None, None
+
+
+def moe_unpermute(back: torch.Tensor, probs: torch.Tensor,
+                  order: torch.Tensor, n: int, k: int) -> torch.Tensor:
+    """back [m,h] expert-sorted outputs; probs [m] sorted routing weights;
+    order: argsort of the flat (token*k) expert assignment."""
+    if back.is_cuda and native_available() and back.shape[-1] % 8 == 0 \
+            and back.dtype in (torch.bfloat16, torch.float32):
+        return _MoEUnpermute.apply(back.contiguous(), probs.contiguous(),
+                                   order.contiguous(), n, k)
+    h = back.shape[-1]
+    out = back.new_zeros(n * k, h)
+    out[order] = back * probs.unsqueeze(-1).to(back.dtype)
+    return out.reshape(n, k, h).sum(1)

@@ -16,6 +16,7 @@ from typing import List
 import torch
 import torch.distributed as dist
 
+from ...ops.functional import moe_permute, moe_unpermute
 from ..tensor_parallel.mappings import _is_gloo
 
 
@@ -94,7 +95,7 @@ class AlltoAllDispatcher:
         self._order = order
         self._probs = probs.reshape(-1)[order]          # [n*k]
         rows = order // k                               # source token row
-        permuted = x[rows]                              # [n*k, h]
+        permuted = moe_permute(x, rows)                 # [n*k, h]
 
         counts = torch.bincount(flat_idx, minlength=self.num_experts)
         self._counts = counts
@@ -153,11 +154,8 @@ class AlltoAllDispatcher:
                               self._recv_splits)
         else:
             back = expert_out
-        # unpermute + weighted merge
-        h = back.shape[-1]
-        out = back.new_zeros(n_tokens * topk, h)
-        out[self._order] = back * self._probs.unsqueeze(-1).to(back.dtype)
-        return out.reshape(n_tokens, topk, h).sum(1)
+        # unpermute + weighted merge (HIP kernel on GPU)
+        return moe_unpermute(back, self._probs, self._order, n_tokens, topk)
 
 
 class AllGatherDispatcher:

@@ -288,3 +288,30 @@ def test_grouped_gemm(counts):
     c_ref.backward(dc.float())
     assert_close(a.grad, a32.grad, 6e-2, rtol=3e-2, what="grouped_gemm dA")
     assert_close(w.grad, w32.grad, 6e-2, rtol=3e-2, what="grouped_gemm dW")
+
+
+def test_moe_permute_unpermute():
+    from hetu_galvatron_amd.ops.functional import moe_permute, moe_unpermute
+    torch.manual_seed(12)
+    n, h, E, k = 64, 128, 4, 2
+    x = torch.randn(n, h, device=dev()).bfloat16().requires_grad_(True)
+    idx = torch.randint(0, E, (n, k), device=dev())
+    probs = torch.softmax(torch.randn(n, k, device=dev()), -1).reshape(-1)
+    flat = idx.reshape(-1)
+    order = torch.argsort(flat, stable=True)
+    rows = order // k
+    perm = moe_permute(x, rows)
+    assert torch.equal(perm.detach(), x.detach()[rows])
+    p_sorted = probs[order]
+    out = moe_unpermute(perm, p_sorted, order, n, k)
+    # reference (pure torch)
+    x2 = x.detach().clone().requires_grad_(True)
+    perm2 = x2[rows]
+    full = perm2.new_zeros(n * k, h)
+    full[order] = perm2 * p_sorted.unsqueeze(-1).to(perm2.dtype)
+    out_ref = full.reshape(n, k, h).sum(1)
+    assert_close(out, out_ref, 2e-2, what="moe unpermute fwd")
+    g = torch.randn_like(out)
+    out.backward(g)
+    out_ref.backward(g)
+    assert_close(x.grad, x2.grad, 3e-2, what="moe permute/unpermute grad")
