@@ -18,7 +18,7 @@ from typing import Optional
 import torch
 
 from . import reference_ops as ref
-from ._ext import get_ext, use_native
+from ._ext import get_ext, native_available, use_native
 
 
 class _RMSNorm(torch.autograd.Function):
