@@ -115,8 +115,10 @@ def read_hardware_profiles(hw_dir: str, nodes: int = 1, gpus: int = 8
                 d["popt"] = fit_linear([p[0] for p in pts], [p[1] for p in pts])
             elif pts:
                 d["popt"] = (0.0, pts[0][1])
-    ov = load("overlap_coefficient") or {"overlap_coe": 1.15}
-    hw.overlap_coe = float(ov.get("overlap_coe", 1.15))
+    ov_path = os.path.join(hw_dir, "overlap_coefficient.json")
+    if os.path.exists(ov_path):
+        with open(ov_path) as f:
+            hw.overlap_coe = float(json.load(f).get("overlap_coe", 1.15))
     return hw
 
 

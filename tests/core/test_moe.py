@@ -139,3 +139,19 @@ def test_moe_ep_zero3():
     for losses in res:
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_moe_world4_ep2_dp2():
+    """ep=2 x dp=2 on 4 ranks: experts split over half the sdp group,
+    expert grads reduced over edp."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=4, pp=1, tp=1, dp_type="ddp",
+        global_bsz=4, chunks=1, ep=2)
+    res = run_distributed(_moe_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)

@@ -187,3 +187,55 @@ def test_searched_plan_loads_into_runtime(tmp_path):
     assert plan.num_layers == 32
     s = plan.layer(0, world_size=8)
     assert s.degree_product() == 8
+
+
+def test_profiled_json_roundtrip(tmp_path):
+    """Fabricated profiler outputs (reference JSON schemas) -> load_profiles
+    -> search: the measured-profile path end to end."""
+    import json
+    hw_dir = tmp_path / "hw"
+    hw_dir.mkdir()
+    (hw_dir / "allreduce_bandwidth_1nodes_8gpus_per_node.json").write_text(
+        json.dumps({"allreduce_size_8_consec_1": 300.0,
+                    "allreduce_size_4_consec_1": 310.0,
+                    "allreduce_size_4_consec_0": 290.0,
+                    "allreduce_size_2_consec_1": 250.0,
+                    "allreduce_size_2_consec_0": 240.0}))
+    (hw_dir / "p2p_bandwidth_1nodes_8gpus_per_node.json").write_text(
+        json.dumps({"pp_size_2": 150.0, "pp_size_4": 140.0,
+                    "pp_size_8": 120.0}))
+    sp = {}
+    for n in (2, 4, 8):
+        for mb in (1, 4, 16, 64, 256):
+            sp[f"allreduce_size_{n}_{mb}MB_time"] = 0.05 + mb * 0.008
+            sp[f"all2all_size_{n}_{mb}MB_time"] = 0.05 + mb * 0.006
+    (hw_dir / "sp_time_1nodes_8gpus_per_node.json").write_text(json.dumps(sp))
+    (hw_dir / "overlap_coefficient.json").write_text(
+        json.dumps({"overlap_coe": 1.12}))
+
+    comp = {}
+    for ln in (1, 2):
+        for b in (1, 2, 4, 8):
+            comp[f"layernum[{ln}]_bsz{b}_seq4096"] = ln * (4.4 * b + 0.5) + b
+    comp_path = tmp_path / "computation_profiling_bf16_llama-3-8b.json"
+    comp_path.write_text(json.dumps(comp))
+    mem_path = tmp_path / "model_profile_bf16_llama-3-8b.json"
+    mem_path.write_text(json.dumps({
+        "layertype_0": {"parameter_size": 840.0,
+                        "tp_activation_per_bsz_dict":
+                            {"1": 520.0, "2": 270.0, "4": 140.0, "8": 75.0,
+                             "checkpoint": 34.0}},
+        "other": {"parameter_size": 2100.0,
+                  "tp_activation_per_bsz_dict": {"1": 600.0}}}))
+
+    cfg = load_config(base={"model": {"model_name": "llama-3-8b"},
+                            "search": {"settle_bsz": 64, "settle_chunks": 8,
+                                       "max_pp_deg": 2,
+                                       "memory_constraint": 240}})
+    eng = SearchEngine(cfg)
+    eng.load_profiles(str(comp_path), str(mem_path), str(hw_dir))
+    assert abs(eng.layer_profile.fct_linear[0] - 4.4) < 1e-6
+    assert abs(eng.hw.overlap_coe - 1.12) < 1e-9
+    assert 8 in eng.hw.allgather_latency
+    best = eng.parallelism_optimization(None)
+    assert best is not None and best.throughput > 0
