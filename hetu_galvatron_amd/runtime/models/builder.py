@@ -63,7 +63,8 @@ class LayerBlock(nn.Module):
         self.flat: Optional[FlatParamBlock] = None
         self.flat_expert: Optional[FlatParamBlock] = None
 
-    def setup_zero(self, mode: str, param_dtype: torch.dtype, device) -> None:
+    def setup_zero(self, mode: str, param_dtype: torch.dtype, device,
+                   reduce_in_fp32: bool = False) -> None:
         tp_group = None
         if not self.groups.strategy.use_ulysses:
             tp_group = self.groups.tp_group
@@ -73,10 +74,12 @@ class LayerBlock(nn.Module):
             self.flat_expert = FlatParamBlock(
                 self.inner, mode, self.groups.edp_group, tp_group=None,
                 param_dtype=param_dtype, device=device,
-                param_filter=lambda p: getattr(p, "expert_parallel", False))
+                param_filter=lambda p: getattr(p, "expert_parallel", False),
+                reduce_in_fp32=reduce_in_fp32)
         self.flat = FlatParamBlock(self.inner, mode, self.groups.sdp_group,
                                    tp_group=tp_group, param_dtype=param_dtype,
-                                   device=device)
+                                   device=device,
+                                   reduce_in_fp32=reduce_in_fp32)
 
     def _inner_forward(self, x, ctx):
         if self.kind == "embedding":
@@ -259,7 +262,8 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
     for b in blocks:
         b.inner.to(device)
         mode = b.groups.strategy.dp_type
-        b.setup_zero(mode, dtype, device)
+        b.setup_zero(mode, dtype, device,
+                     reduce_in_fp32=cfg.parallel.gradient_reduce_in_fp32)
 
     recv_layout = None
     if my_stage > 0:

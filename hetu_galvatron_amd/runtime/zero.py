@@ -66,7 +66,8 @@ class FlatParamBlock:
                  param_dtype: torch.dtype = torch.bfloat16,
                  device: Optional[torch.device] = None,
                  owner_filter: bool = True,
-                 param_filter=None):
+                 param_filter=None,
+                 reduce_in_fp32: bool = False):
         assert mode in ("ddp", "zero2", "zero3")
         self.module = module
         self.mode = mode
@@ -83,6 +84,8 @@ class FlatParamBlock:
         self._gathered = False
         self.auto_sync = False  # post-backward hook kicks async reduce
         self._synced = False
+        self.reduce_in_fp32 = reduce_in_fp32
+        self._post_sync = []  # deferred ops after the async handles drain
 
         params = []
         for p in module.parameters():
@@ -236,9 +239,20 @@ class FlatParamBlock:
         if self.ws == 1:
             self.grad_shard = g if self.mode == "ddp" else g[self._shard_slice()]
             return
+        # bf16-compressed reduction by default (matches the reference's FSDP
+        # bf16-reduce default, parallel.py:130; fp32 accumulation stays
+        # local): halves the xGMI bytes. gradient_reduce_in_fp32 opts out.
+        compress = (not self.reduce_in_fp32
+                    and self.param_dtype == torch.bfloat16
+                    and not _is_gloo(self.sdp_group.group))
         if self.mode == "ddp":
             if _is_gloo(self.sdp_group.group):
                 dist.all_reduce(g, group=self.sdp_group.group)
+            elif compress:
+                g16 = g.to(torch.bfloat16)
+                self._handles.append(dist.all_reduce(
+                    g16, group=self.sdp_group.group, async_op=True))
+                self._post_sync.append(lambda g=g, g16=g16: g.copy_(g16))
             else:
                 self._handles.append(dist.all_reduce(
                     g, group=self.sdp_group.group, async_op=True))
@@ -249,6 +263,14 @@ class FlatParamBlock:
             if _is_gloo(self.sdp_group.group):
                 dist.all_reduce(g, group=self.sdp_group.group)
                 self.grad_shard.copy_(g[self._shard_slice()])
+            elif compress:
+                g16 = g.to(torch.bfloat16)
+                s16 = torch.empty(self.shard_size, dtype=torch.bfloat16,
+                                  device=self.device)
+                self._handles.append(dist.reduce_scatter_tensor(
+                    s16, g16, group=self.sdp_group.group, async_op=True))
+                self._post_sync.append(
+                    lambda d=self.grad_shard, s=s16: d.copy_(s))
             else:
                 self._handles.append(dist.reduce_scatter_tensor(
                     self.grad_shard, g, group=self.sdp_group.group,
@@ -258,6 +280,9 @@ class FlatParamBlock:
         for h in self._handles:
             h.wait()
         self._handles = []
+        for op in self._post_sync:
+            op()
+        self._post_sync = []
 
     def zero_grad(self) -> None:
         if self.flat_grad is not None:
