@@ -55,6 +55,7 @@ void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, floa
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
+void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, int, int, int, int, int, int, float, hipStream_t);
 
 namespace {
 
@@ -273,6 +274,23 @@ std::tuple<Tensor, Tensor, Tensor> flash_attn_bwd(const Tensor& dout,
     dv = dv_exp.view({b, skv, hkv, rep, d}).sum(3);
   }
   return {dq, dk, dv};
+}
+
+// Single-token decode attention against a KV cache (serving path).
+// q: [b,hq,d]; k_cache/v_cache: [b,max_s,hkv,d]; attends to [0, cur_len).
+Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
+                   const Tensor& v_cache, int64_t cur_len, double scale) {
+  CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
+  TORCH_CHECK(is_bf16(q), "decode_attn: bf16 only");
+  TORCH_CHECK(q.dim() == 3 && k_cache.dim() == 4, "decode_attn shapes");
+  const int b = q.size(0), hq = q.size(1), d = q.size(2);
+  const int max_s = k_cache.size(1), hkv = k_cache.size(2);
+  TORCH_CHECK(d == 64 || d == 128, "decode_attn: head dim must be 64|128");
+  TORCH_CHECK(cur_len >= 1 && cur_len <= max_s, "decode_attn: bad cur_len");
+  auto o = at::empty_like(q);
+  decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o), b, hq,
+                     hkv, max_s, (int)cur_len, d, (float)scale, cur_stream());
+  return o;
 }
 
 Tensor mfma_probe(const Tensor& A, const Tensor& B, bool alt) {
@@ -519,6 +537,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("decode_attn", &decode_attn);
   m.def("mfma_probe", &mfma_probe);
   m.def("ce_max", &ce_max);
   m.def("ce_sum_target", &ce_sum_target);

@@ -136,3 +136,110 @@ template void swiglu_bwd_launch_t<__bf16>(const __bf16*, const __bf16*, __bf16*,
 template void swiglu_bwd_launch_t<float>(const float*, const float*, float*, long, int, hipStream_t);
 template void rope_launch_t<__bf16>(const __bf16*, __bf16*, const float*, const float*, long, int, int, bool, hipStream_t);
 template void rope_launch_t<float>(const float*, float*, const float*, const float*, long, int, int, bool, hipStream_t);
+
+// ---------------------------------------------------------------------------
+// Single-token decode attention (serving): o[b,hq,d] from a KV cache.
+// Reference role: nvidia_chunked_flash_attn flash-decode
+// (attention.py:398-514, optional dep).  Memory-bound KV read (guide
+// App. B "Attention decode"): one 256-thread block per (b, hq);
+// phase 1: lanes stripe cache positions, 16 B vectorized K dot + online
+// softmax, weights staged in LDS; phase 2: threads stripe d, stream V.
+// ---------------------------------------------------------------------------
+namespace {
+
+template <int D>
+__global__ void decode_attn_kernel(const __bf16* __restrict__ q,   // [b,hq,D]
+                                   const __bf16* __restrict__ kc,  // [b,S,hkv,D]
+                                   const __bf16* __restrict__ vc,
+                                   __bf16* __restrict__ o,         // [b,hq,D]
+                                   int b, int hq, int hkv, int max_s,
+                                   int cur_len, float scale) {
+  extern __shared__ __align__(16) float wsm[];  // [cur_len] weights
+  __shared__ float red[2][8];
+  const int bh = blockIdx.x;
+  const int bi = bh / hq;
+  const int h = bh - bi * hq;
+  const int hk = h / (hq / hkv);
+  const int tid = threadIdx.x;
+
+  // q into registers per thread chunk? phase 1 needs full q per lane: LDS
+  __shared__ __align__(16) float q_sh[D];
+  if (tid < D / 8) {
+    float v[8];
+    VecIO<__bf16>::load(v, q + ((long)bi * hq + h) * D + tid * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) q_sh[tid * 8 + i] = v[i];
+  }
+  __syncthreads();
+
+  const long kbase = ((long)bi * max_s * hkv + hk) * D;
+  const int kstride = hkv * D;
+
+  float m = -1e30f, l = 0.f;
+  for (int p = tid; p < cur_len; p += blockDim.x) {
+    const __bf16* kp = kc + kbase + (long)p * kstride;
+    float dot = 0.f;
+#pragma unroll
+    for (int c = 0; c < D; c += 8) {
+      float kv[8];
+      VecIO<__bf16>::load(kv, kp + c);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) dot += kv[i] * q_sh[c + i];
+    }
+    const float s = dot * scale;
+    wsm[p] = s;
+    if (s > m) {
+      l *= __expf(m - s);
+      m = s;
+    }
+    l += __expf(s - m);
+  }
+  // merge (m, l) across the block
+  const int lane = tid & 63, wid = tid >> 6;
+#pragma unroll
+  for (int sft = 32; sft >= 1; sft >>= 1) {
+    float mo = __shfl_xor(m, sft, 64);
+    float lo = __shfl_xor(l, sft, 64);
+    float mn = fmaxf(m, mo);
+    l = l * __expf(m - mn) + lo * __expf(mo - mn);
+    m = mn;
+  }
+  if (lane == 0) { red[0][wid] = m; red[1][wid] = l; }
+  __syncthreads();
+  float gm = -1e30f, gl = 0.f;
+  for (int w = 0; w < (int)(blockDim.x >> 6); ++w) {
+    float mn = fmaxf(gm, red[0][w]);
+    gl = gl * __expf(gm - mn) + red[1][w] * __expf(red[0][w] - mn);
+    gm = mn;
+  }
+  __syncthreads();
+  // normalize weights in LDS
+  for (int p = tid; p < cur_len; p += blockDim.x)
+    wsm[p] = __expf(wsm[p] - gm) / gl;
+  __syncthreads();
+
+  // phase 2: threads stripe d; stream V
+  const long vbase = kbase;
+  for (int d0 = tid; d0 < D; d0 += blockDim.x) {
+    float acc = 0.f;
+    for (int p = 0; p < cur_len; ++p)
+      acc += wsm[p] * (float)vc[vbase + (long)p * kstride + d0];
+    o[((long)bi * hq + h) * D + d0] = (__bf16)acc;
+  }
+}
+
+}  // namespace
+
+void decode_attn_launch(const __bf16* q, const __bf16* kc, const __bf16* vc,
+                        __bf16* o, int b, int hq, int hkv, int max_s,
+                        int cur_len, int d, float scale, hipStream_t st) {
+  const int shmem = cur_len * sizeof(float);
+  if (d == 64)
+    hipLaunchKernelGGL((decode_attn_kernel<64>), dim3(b * hq), dim3(256),
+                       shmem, st, q, kc, vc, o, b, hq, hkv, max_s, cur_len,
+                       scale);
+  else
+    hipLaunchKernelGGL((decode_attn_kernel<128>), dim3(b * hq), dim3(256),
+                       shmem, st, q, kc, vc, o, b, hq, hkv, max_s, cur_len,
+                       scale);
+}

@@ -281,3 +281,35 @@ def moe_unpermute(back: torch.Tensor, probs: torch.Tensor,
     out = back.new_zeros(n * k, h)
     out[order] = back * probs.unsqueeze(-1).to(back.dtype)
     return out.reshape(n, k, h).sum(1)
+
+
+@torch.no_grad()
+def decode_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                     v_cache: torch.Tensor, cur_len: int,
+                     softmax_scale=None) -> torch.Tensor:
+    """Single-token decode attention against a KV cache (serving path).
+
+    q: [b, hq, d]; k_cache/v_cache: [b, max_s, hkv, d]; attends to
+    positions [0, cur_len).  Native CDNA4 kernel on GPU bf16 (memory-bound
+    KV streaming, decode_attn_kernel in elementwise.hip); plain torch
+    reference elsewhere.  Reference role: the optional flash-decode path
+    (nvidia_chunked_flash_attn, attention.py:398-514).
+    """
+    scale = softmax_scale if softmax_scale is not None \
+        else q.shape[-1] ** -0.5
+    if q.is_cuda and native_available() and q.dtype == torch.bfloat16 \
+            and q.shape[-1] in (64, 128):
+        return get_ext().decode_attn(q.contiguous(), k_cache, v_cache,
+                                     int(cur_len), float(scale))
+    b, hq, d = q.shape
+    hkv = k_cache.shape[2]
+    k = k_cache[:, :cur_len].float()
+    v = v_cache[:, :cur_len].float()
+    if hq != hkv:
+        rep = hq // hkv
+        k = k.repeat_interleave(rep, dim=2)
+        v = v.repeat_interleave(rep, dim=2)
+    att = torch.einsum("bhd,bshd->bhs", q.float(), k) * scale
+    w = att.softmax(-1)
+    o = torch.einsum("bhs,bshd->bhd", w, v)
+    return o.to(q.dtype)

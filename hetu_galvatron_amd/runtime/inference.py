@@ -1,0 +1,196 @@
+"""Single-GPU serving path: KV-cache autoregressive generation.
+
+Reference role: the optional flash-decode inference dependency
+(nvidia_chunked_flash_attn, galvatron attention.py:398-514) — the
+reference treats decode as an optional attention backend; here it is a
+first-class engine over the same hybrid-parallel module weights.
+
+MI355X design: decode attention is one memory-bound HIP kernel
+(`decode_attn` in ops/csrc/elementwise.hip) streaming the bf16 KV cache
+at HBM rate; prefill reuses the MFMA flash kernel while writing the
+cache.  288 GB HBM3E comfortably holds an 8B model + tens of GB of KV
+cache on ONE GPU, so the v1 serving topology is world=1 (tp/pp/cp=1,
+any dp_type — at world 1 every flat-param mode keeps full params
+resident).  Multi-GPU serving (tp decode) is a documented v2 item.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+
+from ..ops import decode_attention, flash_attention_fwd_only
+
+__all__ = ["KVCache", "GalvatronGenerator"]
+
+
+class KVCache:
+    """Per-layer bf16 K/V ring of shape [b, max_seq, hkv, d]."""
+
+    def __init__(self, n_layers: int, b: int, max_seq: int, hkv: int, d: int,
+                 device, dtype=torch.bfloat16):
+        self.k = [torch.empty(b, max_seq, hkv, d, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.v = [torch.empty(b, max_seq, hkv, d, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.max_seq = max_seq
+        self.cur_len = 0
+
+    def write(self, layer: int, k: torch.Tensor, v: torch.Tensor) -> None:
+        """k, v: [b, s_new, hkv, d] for positions [cur_len, cur_len + s_new)."""
+        s = k.shape[1]
+        self.k[layer][:, self.cur_len:self.cur_len + s] = k
+        self.v[layer][:, self.cur_len:self.cur_len + s] = v
+
+    def advance(self, s: int) -> None:
+        self.cur_len += s
+        assert self.cur_len <= self.max_seq, "KV cache overflow"
+
+
+class GalvatronGenerator:
+    """Greedy / temperature sampling over a built GalvatronModel.
+
+    Drives the trained module weights directly (embedding -> decoder
+    layers -> final norm -> lm head), bypassing the training engine's
+    loss head.  Requires a world-size-1 model (serving topology v1).
+    """
+
+    def __init__(self, model, max_batch: int = 1, max_seq: int = 4096):
+        sm = model.stage_model
+        assert sm.world_size == 1 and sm.pp_deg == 1, \
+            "GalvatronGenerator v1 serves on one GPU (world=1)"
+        self.cfg = model.cfg
+        m = self.cfg.model
+        assert m.position_embedding_type == "rope", \
+            "generator v1 supports RoPE models (llama/mistral/qwen family)"
+        self.margs = m
+        self.blocks = sm.blocks
+        kinds = [b.kind for b in self.blocks]
+        assert kinds[0] == "embedding" and kinds[-1] == "lm_head"
+        self.layers = [b.inner for b in self.blocks
+                       if b.kind in ("decoder", "encoder")]
+        self.embedding = self.blocks[0].inner
+        self.final_norm = next(b.inner for b in self.blocks
+                               if b.kind == "final_norm")
+        self.lm_head = self.blocks[-1].inner
+        self.max_batch = max_batch
+        self.max_seq = max_seq
+        self.scale = 1.0 / (m.head_dim ** 0.5)
+        self._dev = next(self.embedding.parameters()).device
+
+    # -- one decoder layer, cache-aware ------------------------------------
+    def _split_qkv(self, attn, qkv: torch.Tensor):
+        """Interleaved-group [q*(hq/hkv), k, v] layout (attention.py)."""
+        s, b = qkv.shape[0], qkv.shape[1]
+        qkv = qkv.view(s, b, attn.num_groups_local, attn.q_per_group + 2,
+                       attn.head_dim)
+        q = qkv[:, :, :, : attn.q_per_group].reshape(s, b, -1, attn.head_dim)
+        k = qkv[:, :, :, attn.q_per_group].reshape(s, b, -1, attn.head_dim)
+        v = qkv[:, :, :, attn.q_per_group + 1].reshape(s, b, -1, attn.head_dim)
+        return q, k, v
+
+    def _layer_step(self, li: int, hidden: torch.Tensor, cache: KVCache,
+                    pos: int) -> torch.Tensor:
+        """hidden: [s_new, b, h] for absolute positions [pos, pos+s_new)."""
+        layer = self.layers[li]
+        attn = layer.attention
+        residual = hidden
+        x = layer.input_norm(hidden)
+        qkv = attn.linear_qkv(x)
+        q, k, v = self._split_qkv(attn, qkv)
+        if layer.rotary is not None:
+            cos, sin = layer.rotary.full_tables(pos + q.shape[0], hidden.device)
+            from .transformer.rope import apply_rope_qk
+            q, k = apply_rope_qk(q.contiguous(), k.contiguous(),
+                                 cos[pos:], sin[pos:])
+        # cache layout [b, s, hkv, d]
+        cache.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3))
+        s_new = q.shape[0]
+        if s_new == 1:
+            o = decode_attention(q[0], cache.k[li], cache.v[li],
+                                 pos + 1, softmax_scale=self.scale)
+            o = o.unsqueeze(0)  # [1, b, hq, d]
+        elif pos == 0:
+            qb = q.permute(1, 0, 2, 3).contiguous()
+            kb = k.permute(1, 0, 2, 3).contiguous()
+            vb = v.permute(1, 0, 2, 3).contiguous()
+            ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
+                                             softmax_scale=self.scale)
+            o = ob.permute(1, 0, 2, 3)
+        else:
+            # chunked prefill continuation: per-token decode (rare path)
+            outs = []
+            for t in range(s_new):
+                outs.append(decode_attention(q[t], cache.k[li], cache.v[li],
+                                             pos + t + 1,
+                                             softmax_scale=self.scale))
+            o = torch.stack(outs, dim=0)
+        o = o.reshape(s_new, o.shape[1], -1)
+        x = attn.linear_proj(o)
+        hidden = residual + x
+        residual = hidden
+        x = layer.post_attn_norm(hidden)
+        x = layer.mlp(x)
+        return residual + x
+
+    @torch.no_grad()
+    def _forward_tokens(self, tokens: torch.Tensor, cache: KVCache
+                        ) -> torch.Tensor:
+        """tokens: [b, s_new] at positions [cache.cur_len, ...); returns
+        last-position logits [b, V]."""
+        pos = cache.cur_len
+        h = self.embedding.word_embeddings(tokens)  # [s_new, b, h]
+        for li in range(len(self.layers)):
+            h = self._layer_step(li, h, cache, pos)
+        cache.advance(tokens.shape[1])
+        h = self.final_norm.norm(h[-1:])
+        logits = self.lm_head.lm_head(h)  # [1, b, V]
+        return logits[0].float()
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, top_k: int = 0,
+                 eos_id: Optional[int] = None) -> torch.Tensor:
+        """input_ids: [b, s_prompt] on the model device.  Returns
+        [b, s_prompt + n_generated] (greedy when temperature == 0)."""
+        b, sp = input_ids.shape
+        assert b <= self.max_batch and sp + max_new_tokens <= self.max_seq
+        m = self.margs
+        cache = KVCache(len(self.layers), b, self.max_seq, m.kv_heads,
+                        m.head_dim, self._dev,
+                        dtype=next(self.embedding.parameters()).dtype)
+        for blk in self.blocks:
+            if blk.flat is not None:
+                blk.flat.gather_params()
+        was_training = [l.training for l in self.layers]
+        for blk in self.blocks:
+            blk.inner.eval()
+        try:
+            logits = self._forward_tokens(input_ids, cache)
+            out: List[torch.Tensor] = [input_ids]
+            done = torch.zeros(b, dtype=torch.bool, device=self._dev)
+            for _ in range(max_new_tokens):
+                nxt = self._sample(logits, temperature, top_k)
+                if eos_id is not None:
+                    nxt = torch.where(done, torch.full_like(nxt, eos_id), nxt)
+                    done |= nxt.eq(eos_id)
+                out.append(nxt.unsqueeze(1))
+                if eos_id is not None and bool(done.all()):
+                    break
+                logits = self._forward_tokens(nxt.unsqueeze(1), cache)
+        finally:
+            for l, t in zip(self.layers, was_training):
+                l.train(t)
+        return torch.cat(out, dim=1)
+
+    @staticmethod
+    def _sample(logits: torch.Tensor, temperature: float, top_k: int
+                ) -> torch.Tensor:
+        if temperature <= 0.0:
+            return logits.argmax(-1)
+        logits = logits / temperature
+        if top_k > 0:
+            kth = logits.topk(top_k, dim=-1).values[..., -1, None]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        return torch.multinomial(probs, 1).squeeze(-1)

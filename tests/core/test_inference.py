@@ -1,0 +1,69 @@
+"""Serving path: KV-cache incremental decode == full-prefix recompute.
+
+Reference role parity: flash-decode inference (optional backend in the
+reference); here validated functionally on CPU via the torch fallback of
+decode_attention.
+"""
+import torch
+
+from hetu_galvatron_amd.config import load_config
+from hetu_galvatron_amd.ops import decode_attention
+from hetu_galvatron_amd.runtime import GalvatronModel
+from hetu_galvatron_amd.runtime.inference import GalvatronGenerator, KVCache
+
+
+def make_model(name="tiny-llama"):
+    cfg = load_config(base={
+        "model": {"model_name": name},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"},
+    })
+    torch.manual_seed(0)
+    return GalvatronModel(cfg)
+
+
+def test_decode_attention_matches_dense():
+    torch.manual_seed(1)
+    b, hq, hkv, d, S = 2, 4, 2, 64, 17
+    q = torch.randn(b, hq, d)
+    kc = torch.randn(b, 32, hkv, d)
+    vc = torch.randn(b, 32, hkv, d)
+    o = decode_attention(q, kc, vc, S)
+    # dense reference
+    k = kc[:, :S].repeat_interleave(hq // hkv, dim=2)
+    v = vc[:, :S].repeat_interleave(hq // hkv, dim=2)
+    att = torch.einsum("bhd,bshd->bhs", q, k) / d ** 0.5
+    want = torch.einsum("bhs,bshd->bhd", att.softmax(-1), v)
+    assert torch.allclose(o, want, atol=1e-5)
+
+
+def test_incremental_decode_matches_recompute():
+    model = make_model()
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    torch.manual_seed(2)
+    ids = torch.randint(0, model.cfg.model.vocab_size, (2, 9))
+    out = gen.generate(ids, max_new_tokens=5, temperature=0.0)
+    assert out.shape == (2, 14)
+    # recompute: at every step run the FULL prefix through a fresh cache
+    seq = ids
+    for _ in range(5):
+        cache = KVCache(len(gen.layers), 2, 64, model.cfg.model.kv_heads,
+                        model.cfg.model.head_dim, seq.device,
+                        dtype=torch.float32)
+        logits = gen._forward_tokens(seq, cache)
+        seq = torch.cat([seq, logits.argmax(-1, keepdim=True)], dim=1)
+    assert torch.equal(out, seq)
+
+
+def test_generate_eos_and_sampling():
+    model = make_model()
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    ids = torch.randint(0, model.cfg.model.vocab_size, (2, 4))
+    torch.manual_seed(3)
+    out = gen.generate(ids, max_new_tokens=4, temperature=0.8, top_k=5)
+    assert out.shape[1] <= 8 and out.shape[0] == 2
+    # eos stops generation early when every row has emitted it
+    out2 = gen.generate(ids, max_new_tokens=8, temperature=0.0,
+                        eos_id=int(out[0, -1]))
+    assert out2.shape[1] <= 12
