@@ -55,7 +55,7 @@ void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, floa
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
-void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, int, int, int, int, int, int, float, hipStream_t);
+void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, hipStream_t);
 
 namespace {
 
@@ -288,8 +288,22 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   TORCH_CHECK(d == 64 || d == 128, "decode_attn: head dim must be 64|128");
   TORCH_CHECK(cur_len >= 1 && cur_len <= max_s, "decode_attn: bad cur_len");
   auto o = at::empty_like(q);
-  decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o), b, hq,
-                     hkv, max_s, (int)cur_len, d, (float)scale, cur_stream());
+  // split-KV: fill >=512 workgroups (2/CU) when b*hq alone can't, but keep
+  // chunks >=128 positions so the merge stays cheap
+  int n_chunks = b * hq >= 256 ? 1 : std::max(1, 512 / (b * hq));
+  n_chunks = std::min<int>(n_chunks, (int)((cur_len + 127) / 128));
+  // LDS cap: the weight buffer is cur_len/chunk fp32 — keep chunks <= 8192
+  n_chunks = std::max<int>(n_chunks, (int)((cur_len + 8191) / 8192));
+  Tensor ws;
+  float* ws_ptr = nullptr;
+  if (n_chunks > 1) {
+    ws = at::empty({(long)b * hq * n_chunks * (d + 2)},
+                   q.options().dtype(at::kFloat));
+    ws_ptr = ws.data_ptr<float>();
+  }
+  decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o), ws_ptr,
+                     n_chunks, b, hq, hkv, max_s, (int)cur_len, d,
+                     (float)scale, cur_stream());
   return o;
 }
 
