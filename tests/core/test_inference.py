@@ -67,3 +67,13 @@ def test_generate_eos_and_sampling():
     out2 = gen.generate(ids, max_new_tokens=8, temperature=0.0,
                         eos_id=int(out[0, -1]))
     assert out2.shape[1] <= 12
+
+
+def test_generate_cli(capsys):
+    from hetu_galvatron_amd.cli.generate import main
+    out = main(["model.model_name=tiny-llama",
+                "parallel.mixed_precision=fp32",
+                "generate.max_new_tokens=3", "generate.prompt_ids=5,6,7"])
+    assert out.shape == (1, 6)
+    lines = capsys.readouterr().out.strip().splitlines()
+    assert lines[-1].startswith("5 6 7 ")
