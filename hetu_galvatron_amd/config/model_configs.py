@@ -100,7 +100,7 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         vocab_size=32128, max_position_embeddings=1024, seq_length=512,
         encoder_seq_length=512, hidden_act="gelu",
         normalization="layernorm", norm_epsilon=1e-6,
-        position_embedding_type="learned", add_bias_linear=False,
+        position_embedding_type="relative", add_bias_linear=False,
         add_qkv_bias=False,
     ),
     "tiny-t5": dict(
@@ -110,7 +110,7 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         vocab_size=512, max_position_embeddings=256, seq_length=64,
         encoder_seq_length=96, hidden_act="gelu",
         normalization="layernorm", norm_epsilon=1e-6,
-        position_embedding_type="learned", add_bias_linear=False,
+        position_embedding_type="relative", add_bias_linear=False,
         add_qkv_bias=False,
     ),
     # tiny models for tests

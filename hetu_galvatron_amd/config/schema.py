@@ -58,7 +58,9 @@ class ModelArgs(BaseModel):
     hidden_act: str = "silu"  # silu(swiglu) | gelu | geglu
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
     norm_epsilon: float = 1e-5
-    position_embedding_type: str = "rope"  # rope | learned
+    position_embedding_type: str = "rope"  # rope | learned | relative
+    relative_attention_num_buckets: int = 32    # t5 bucketized bias
+    relative_attention_max_distance: int = 128
     rope_theta: float = 500000.0
     rope_scaling: Optional[float] = None
     add_bias_linear: bool = False

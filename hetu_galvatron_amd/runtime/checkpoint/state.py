@@ -14,6 +14,11 @@ canonical layout, which is simply the world_size=1 model's state_dict:
   decoder.{i}.mlp.fc2.{weight,bias}
   final_norm.norm.{weight,bias}
   lm_head.lm_head.weight                     [V, h]
+  encoder.{i}.* / encdec_bridge.*            (t5: encoder stack, bridge's
+                                             enc_final_norm + dec_embedding;
+                                             decoder indices continue after
+                                             the encoder's)
+  *.rel_bias.weight                          [buckets, H] replicated
 
 Sharding rules per layer strategy (megatron tp=t, rank r; ulysses/cp/dp
 replicate):
@@ -40,11 +45,15 @@ def _block_prefix(block) -> Optional[str]:
         return "embedding"
     if block.kind == "decoder":
         return f"decoder.{block.inner.layer_idx}"
+    if block.kind == "encoder":
+        return f"encoder.{block.inner.layer_idx}"
+    if block.kind == "encdec_bridge":
+        return "encdec_bridge"
     if block.kind == "final_norm":
         return "final_norm"
     if block.kind == "lm_head":
         return "lm_head"
-    return None
+    raise ValueError(f"unknown block kind {block.kind!r}")
 
 
 def canonical_state_from_stage(sm: StageModel) -> Dict[str, torch.Tensor]:

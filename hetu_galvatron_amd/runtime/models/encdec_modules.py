@@ -29,7 +29,39 @@ from ..tensor_parallel import group_size
 from ..tensor_parallel.mappings import gather_from_sequence_parallel_region
 from ..transformer import MLP, SelfAttention, build_norm
 from ..transformer.attention import CrossAttention
-from .modules import GalvatronEmbedding, _tag_tp_replicated
+from ..transformer.relative_bias import RelativePositionBias
+from .modules import GalvatronEmbedding, _my_rank, _tag_tp_replicated
+
+
+def _build_rel_bias(margs: ModelArgs, groups, bidirectional: bool, dtype):
+    """T5 bucketized relative bias (per layer; see relative_bias.py for the
+    HF layer-0-shared layout note)."""
+    if margs.position_embedding_type != "relative":
+        return None
+    s = groups.strategy
+    assert not s.use_ulysses and s.cp == 1, \
+        "relative-position bias under ulysses/cp is a v2 item"
+    rb = RelativePositionBias(
+        margs.relative_attention_num_buckets,
+        margs.relative_attention_max_distance,
+        margs.num_attention_heads, bidirectional=bidirectional, dtype=dtype)
+    if group_size(groups.tp_group) > 1:
+        # full table on every tp rank; per-rank grads cover disjoint head
+        # columns, the tp grad all-reduce assembles the full gradient
+        _tag_tp_replicated(rb.weight)
+    return rb
+
+
+def _rank_bias(rel_bias, attention, groups, hidden):
+    """[h_local, S, S] bias for this tp rank; S = the attention-visible
+    sequence (megatron-SP input arrives seq-sharded S/tp)."""
+    if rel_bias is None:
+        return None
+    tp = group_size(groups.tp_group)
+    S = hidden.shape[0] * tp
+    c = groups.coord_of(_my_rank())
+    hl = attention.heads_local
+    return rel_bias(S, S, hidden.device, c.tp_idx * hl, (c.tp_idx + 1) * hl)
 
 
 class GalvatronEncoderLayer(nn.Module):
@@ -67,12 +99,15 @@ class GalvatronEncoderLayer(nn.Module):
                                getattr(self.post_attn_norm, "bias", None),
                                self.attention.linear_proj.bias,
                                self.mlp.fc2.bias)
+        self.rel_bias = _build_rel_bias(margs, groups, bidirectional=True,
+                                        dtype=dtype)
         self.dropout_p = margs.hidden_dropout
 
     def forward(self, hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
         residual = hidden
         x = self.input_norm(hidden)
-        x = self.attention(x, None, None)  # no RoPE (learned positions)
+        bias = _rank_bias(self.rel_bias, self.attention, self.groups, hidden)
+        x = self.attention(x, None, None, attn_bias=bias)
         if self.dropout_p > 0 and self.training:
             x = F.dropout(x, self.dropout_p)
         hidden = residual + x
@@ -154,12 +189,15 @@ class GalvatronDecoderLayerX(nn.Module):
                                self.attention.linear_proj.bias,
                                self.cross_attention.linear_proj.bias,
                                self.mlp.fc2.bias)
+        self.rel_bias = _build_rel_bias(margs, groups, bidirectional=False,
+                                        dtype=dtype)
         self.dropout_p = margs.hidden_dropout
 
     def forward(self, hidden: torch.Tensor, ctx: Dict) -> torch.Tensor:
         residual = hidden
         x = self.input_norm(hidden)
-        x = self.attention(x, None, None)
+        bias = _rank_bias(self.rel_bias, self.attention, self.groups, hidden)
+        x = self.attention(x, None, None, attn_bias=bias)
         if self.dropout_p > 0 and self.training:
             x = F.dropout(x, self.dropout_p)
         hidden = residual + x

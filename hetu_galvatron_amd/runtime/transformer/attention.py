@@ -75,10 +75,12 @@ class SelfAttention(nn.Module):
         else:
             self.core_attention = None  # plain local flash
 
-    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
-                ) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                attn_bias: torch.Tensor = None) -> torch.Tensor:
         """x: [s_local, b, h] SBH; cos/sin: tables matching the seq layout at
-        RoPE time (megatron-sp: cp-local full seq; ulysses: the local slice)."""
+        RoPE time (megatron-sp: cp-local full seq; ulysses: the local slice).
+        attn_bias: [h_local, sq, skv] additive scores bias (t5 relative
+        bias) — eager fp32 softmax path, local/megatron-tp layouts only."""
         qkv = self.linear_qkv(x)  # [s, b, (hq_l + 2*hkv_l)*d]
         s, b = qkv.shape[0], qkv.shape[1]
         qkv = qkv.view(s, b, self.num_groups_local, self.q_per_group + 2,
@@ -91,7 +93,12 @@ class SelfAttention(nn.Module):
         q = q.permute(1, 0, 2, 3).contiguous()
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()
-        if self.core_attention is not None:
+        if attn_bias is not None:
+            assert self.core_attention is None, \
+                "attn_bias (t5 relative bias) needs local/megatron-tp attention"
+            o = eager_bias_attention(q, k, v, attn_bias, self.causal,
+                                     self.softmax_scale)
+        elif self.core_attention is not None:
             o = self.core_attention(q, k, v, causal=self.causal,
                                     softmax_scale=self.softmax_scale)
         else:
@@ -99,6 +106,27 @@ class SelfAttention(nn.Module):
                                 softmax_scale=self.softmax_scale)
         o = o.permute(1, 0, 2, 3).reshape(s, b, -1)  # back to SBH
         return self.linear_proj(o)
+
+
+def eager_bias_attention(q, k, v, bias, causal: bool, scale: float
+                         ) -> torch.Tensor:
+    """fp32 softmax attention with an additive score bias [h, sq, skv]
+    (t5 relative-position bias; native flash bias input is a v2 kernel
+    item).  q/k/v: [b, s, h, d]."""
+    b, sq, h, d = q.shape
+    skv = k.shape[1]
+    if k.shape[2] != h:
+        rep = h // k.shape[2]
+        k = k.repeat_interleave(rep, dim=2)
+        v = v.repeat_interleave(rep, dim=2)
+    att = torch.einsum("bqhd,bkhd->bhqk", q.float(), k.float()) * scale
+    att = att + bias.unsqueeze(0).float()
+    if causal:
+        mask = torch.ones(sq, skv, dtype=torch.bool, device=q.device) \
+            .triu(diagonal=1 + skv - sq)
+        att = att.masked_fill(mask, float("-inf"))
+    o = torch.einsum("bhqk,bkhd->bqhd", att.softmax(-1), v.float())
+    return o.to(q.dtype)
 
 
 class CrossAttention(nn.Module):

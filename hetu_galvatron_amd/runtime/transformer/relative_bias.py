@@ -1,0 +1,68 @@
+"""T5 bucketized relative-position attention bias.
+
+Reference: the reference's T5 family relies on HF T5's
+relative_attention_bias (encoder bidirectional, decoder causal buckets);
+re-derived here from the T5 paper's bucketing scheme (log-spaced beyond
+num_buckets/2 exact offsets, capped at max_distance).
+
+Layout note vs HF T5: HF stores the bias table on layer 0 only and every
+layer shares it; here EVERY layer owns its table (a strict superset —
+per-layer tables compose cleanly with per-layer hybrid-parallel
+strategies and pipeline splits; an HF import replicates layer 0's table
+into each layer).  The weight is the FULL [num_buckets, num_heads] table
+on every tp rank; forward slices this rank's head columns, so per-rank
+grads are disjoint column slices and the tp-group grad all-reduce
+(`tp_replicated` tagging, zero.py) assembles the full gradient.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+
+def t5_relative_bucket(relative_position: torch.Tensor, bidirectional: bool,
+                       num_buckets: int, max_distance: int) -> torch.Tensor:
+    """relative_position = memory_pos - query_pos, any integer tensor."""
+    bucket = torch.zeros_like(relative_position)
+    if bidirectional:
+        num_buckets //= 2
+        bucket = bucket + (relative_position > 0).long() * num_buckets
+        n = relative_position.abs()
+    else:
+        n = (-relative_position).clamp(min=0)
+    max_exact = num_buckets // 2
+    is_small = n < max_exact
+    log_big = max_exact + (
+        torch.log(n.float().clamp(min=1) / max_exact)
+        / math.log(max_distance / max_exact) * (num_buckets - max_exact)
+    ).long()
+    log_big = log_big.clamp(max=num_buckets - 1)
+    return bucket + torch.where(is_small, n, log_big)
+
+
+class RelativePositionBias(nn.Module):
+    def __init__(self, num_buckets: int, max_distance: int, num_heads: int,
+                 bidirectional: bool, dtype=None):
+        super().__init__()
+        self.num_buckets = num_buckets
+        self.max_distance = max_distance
+        self.num_heads = num_heads
+        self.bidirectional = bidirectional
+        self.weight = nn.Parameter(torch.empty(
+            num_buckets, num_heads, **({"dtype": dtype} if dtype else {})))
+        nn.init.normal_(self.weight, 0.0, 0.02)
+
+    def forward(self, sq: int, skv: int, device,
+                head_start: int = 0, head_end: int | None = None
+                ) -> torch.Tensor:
+        """Bias [h_local, sq, skv] (fp32) for this rank's head slice."""
+        he = self.num_heads if head_end is None else head_end
+        qpos = torch.arange(sq, device=device)
+        kpos = torch.arange(skv, device=device)
+        rel = kpos[None, :] - qpos[:, None]
+        b = t5_relative_bucket(rel, self.bidirectional, self.num_buckets,
+                               self.max_distance)
+        bias = self.weight.float()[b]          # [sq, skv, H]
+        return bias[..., head_start:he].permute(2, 0, 1).contiguous()
