@@ -138,6 +138,10 @@ class StageModel:
     # layouts at the stage boundaries (for p2p shapes + boundary relocation)
     recv_layout: Optional[LayerCommGroups] = None   # prev stage's last layer
     send_layout: Optional[LayerCommGroups] = None   # my last layer
+    # t5 pipeline boundaries (see build_hybrid_parallel_model)
+    recv_is_encoder: bool = False
+    recv_carries_memory: bool = False
+    send_carries_memory: bool = False
 
     @property
     def is_first(self) -> bool:
@@ -213,8 +217,8 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         assert n_layers == n_enc + n_dec, \
             f"t5 plan must cover enc+dec layers ({n_enc}+{n_dec}), got {n_layers}" 
     if is_encdec:
-        assert pp == 1, "t5 encoder-decoder: pp>1 is a v2 item"
         dps = {plan.layer(i, world).dp for i in range(n_layers)}
+        dps.add(vocab_strat.dp)
         assert len(dps) == 1, "t5: uniform dp degree required (memory " \
             "replication domain); per-layer tp/zero/ckpt may still vary"
     for i in range(lo, hi):
@@ -276,5 +280,12 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         send_layout=layer_groups[hi - 1] if hi > lo else prev,
         tied_embedding_pair=(emb_block, head_block)
         if (emb_block is not None and head_block is not None) else None,
+        # t5 pp boundaries: a cut before layer n_enc carries an
+        # encoder-shaped activation; a cut after the bridge additionally
+        # carries the full-seq encoder memory (engine cats it onto the
+        # boundary rows; uniform dp keeps its batch layout rank-invariant)
+        recv_is_encoder=is_encdec and my_stage > 0 and lo - 1 < n_enc,
+        recv_carries_memory=is_encdec and my_stage > 0 and lo > n_enc,
+        send_carries_memory=is_encdec and my_stage < pp - 1 and hi > n_enc,
     )
     return sm

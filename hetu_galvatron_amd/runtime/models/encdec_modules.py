@@ -10,11 +10,13 @@ The bridge applies the encoder final norm, all-gathers the encoder output
 to the FULL sequence (replicated within the dp group: cross-attention
 then needs no per-layer redistribution however the decoder layers are
 laid out), stashes it in the batch context, and emits the decoder
-embedding.  v1 scope: pp=1, cross-attention under tp/dp/zero/ckpt
+embedding.  pp>1: the engine rides the memory along the p2p boundary
+(engine.py _fwd_step).  v1 scope: cross-attention under tp/dp/zero/ckpt/pp
 (ulysses/cp on decoder layers are asserted off by the builder).
 
-Position encoding: learned absolute (T5's bucketized relative-position
-bias needs a bias input on the flash kernel — a v2 kernel item).
+Position encoding: T5 bucketized relative-position bias
+(transformer/relative_bias.py) on an eager fp32-softmax attention path;
+a bias input on the native flash kernel is the remaining perf item.
 """
 from __future__ import annotations
 

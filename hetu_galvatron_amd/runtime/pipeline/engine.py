@@ -174,15 +174,34 @@ class PipelineEngine:
     # -- pp helpers
 
     def _recv_shape(self, ctx: Dict):
-        return boundary_shape(self.sm.recv_layout, ctx["batch_size"],
-                              ctx["seq_len"], self.hidden)
+        """Boundary rows at this stage's input.  t5: a cut inside the
+        encoder is encoder-seq shaped; a cut past the bridge appends the
+        full-seq encoder memory (rows concatenated by the sender)."""
+        seq = ctx["enc_input_ids"].shape[1] if self.sm.recv_is_encoder \
+            else ctx["seq_len"]
+        rows, b_loc, h = boundary_shape(self.sm.recv_layout,
+                                        ctx["batch_size"], seq, self.hidden)
+        if self.sm.recv_carries_memory:
+            rows += ctx["enc_input_ids"].shape[1]
+        return (rows, b_loc, h)
 
     def _fwd_step(self, ctx: Dict, stats: StepStats, recv_act):
+        feed = None
         if recv_act is not None:
             recv_act = recv_act.detach().requires_grad_(True)
-        out = self._forward_chunk(ctx, recv_act)
+            feed = recv_act
+            if self.sm.recv_carries_memory:
+                s_enc = ctx["enc_input_ids"].shape[1]
+                ctx["encoder_memory"] = recv_act[-s_enc:]
+                feed = recv_act[:-s_enc]
+        out = self._forward_chunk(ctx, feed)
         if self.sm.is_last:
             self._stat_update(stats, out)
+        elif self.sm.send_carries_memory:
+            # ride the memory along the boundary; autograd routes the
+            # receiver's grad back into both the stage output and the
+            # memory's producers (bridge or upstream boundary)
+            out = torch.cat([out, ctx["encoder_memory"]], dim=0)
         return recv_act, out
 
     def _bwd_step(self, inp, out, grad_out, ctx: Dict, chunks: int):

@@ -110,3 +110,25 @@ def test_t5_dist_vs_baseline(tp, dp_type):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("division,chunks", [([2, 2], 1), ([3, 1], 2),
+                                             ([1, 3], 1)])
+def test_t5_pp2_vs_baseline(division, chunks):
+    """pp=2 cuts: [2,2] boundary inside the encoder (encoder-shaped act),
+    [3,1] boundary past the bridge (decoder act + memory ride-along, 1F1B
+    with 2 chunks), [1,3] mid-encoder cut."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=2, pp=2, tp=1, dp_type="ddp",
+        global_bsz=4, chunks=chunks)
+    plan.pp_division = division
+    res = run_distributed(_t5_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
