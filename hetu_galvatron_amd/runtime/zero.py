@@ -242,13 +242,13 @@ class FlatParamBlock:
         # bf16-compressed reduction by default (matches the reference's FSDP
         # bf16-reduce default, parallel.py:130; fp32 accumulation stays
         # local): halves the xGMI bytes. gradient_reduce_in_fp32 opts out.
+        # The same path runs under gloo (torch>=2.10 gloo handles bf16
+        # collectives), so the world>1 CPU tests exercise the RCCL wire
+        # format bit-for-bit.
         compress = (not self.reduce_in_fp32
-                    and self.param_dtype == torch.bfloat16
-                    and not _is_gloo(self.sdp_group.group))
+                    and self.param_dtype == torch.bfloat16)
         if self.mode == "ddp":
-            if _is_gloo(self.sdp_group.group):
-                dist.all_reduce(g, group=self.sdp_group.group)
-            elif compress:
+            if compress:
                 g16 = g.to(torch.bfloat16)
                 self._handles.append(dist.all_reduce(
                     g16, group=self.sdp_group.group, async_op=True))
@@ -260,10 +260,7 @@ class FlatParamBlock:
         else:
             self.grad_shard = torch.empty(self.shard_size, dtype=torch.float32,
                                           device=self.device)
-            if _is_gloo(self.sdp_group.group):
-                dist.all_reduce(g, group=self.sdp_group.group)
-                self.grad_shard.copy_(g[self._shard_slice()])
-            elif compress:
+            if compress:
                 g16 = g.to(torch.bfloat16)
                 s16 = torch.empty(self.shard_size, dtype=torch.bfloat16,
                                   device=self.device)
