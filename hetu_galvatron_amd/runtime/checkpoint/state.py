@@ -106,26 +106,61 @@ def shard_for_rank(name: str, full: torch.Tensor, strategy, tp_rank: int,
     return full
 
 
+def _etp_ffn_slice(t: torch.Tensor, kind: str, tp_rank: int, tp_size: int,
+                   margs: ModelArgs):
+    """Slice an (ep-resolved) expert ffn weight for an etp rank.
+    kind: 'w1' [.., h, out1] / 'fc1' [out1, h] gated [gate(F); up(F)]:
+    each half sliced F/t; 'w2' [.., F, h] rows / 'fc2' [h, F] cols."""
+    if tp_size <= 1:
+        return t
+    gated = margs.hidden_act in ("silu", "swiglu", "geglu")
+    F = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
+    fl = F // tp_size
+    r = tp_rank
+    if kind == "w1":
+        if gated:
+            return torch.cat([t[..., r * fl:(r + 1) * fl],
+                              t[..., F + r * fl:F + (r + 1) * fl]], dim=-1)
+        return t[..., r * fl:(r + 1) * fl]
+    if kind == "w2":
+        return t[..., r * fl:(r + 1) * fl, :]
+    if kind == "fc1":
+        if gated:
+            return torch.cat([t[r * fl:(r + 1) * fl],
+                              t[F + r * fl:F + (r + 1) * fl]], dim=0)
+        return t[r * fl:(r + 1) * fl]
+    assert kind == "fc2"
+    return t[:, r * fl:(r + 1) * fl]
+
+
 def _expert_resolve(name: str, full_key: str, state: Dict[str, torch.Tensor],
-                    ep_rank: int, ep: int, num_experts: int):
-    """Map an EP-local expert param name to its canonical tensor slice.
-    Grouped weights (experts.w1/w2 [E, ...]): slice dim 0 by ep rank.
-    Sequential (experts.fc1.{e}.weight): renumber local->global index."""
-    if ".experts." not in name or ep <= 1:
+                    ep_rank: int, ep: int, num_experts: int,
+                    tp_rank: int = 0, tp_size: int = 1,
+                    margs: Optional[ModelArgs] = None):
+    """Map an EP/ETP-local expert param name to its canonical tensor slice.
+    Grouped weights (experts.w1/w2 [E, ...]): slice dim 0 by ep rank, then
+    ffn dim by etp rank.  Sequential (experts.fc1.{e}.weight): renumber
+    local->global index, then ffn slice."""
+    if ".experts." not in name:
         return state.get(full_key)
-    e_local = num_experts // ep
     parts = name.split(".")
     i = parts.index("experts")
+    e_local = num_experts // ep
     if parts[i + 1] in ("w1", "w2"):
         full = state.get(full_key)
         if full is None:
             return None
-        return full[ep_rank * e_local:(ep_rank + 1) * e_local]
+        if ep > 1:
+            full = full[ep_rank * e_local:(ep_rank + 1) * e_local]
+        return _etp_ffn_slice(full, parts[i + 1], tp_rank, tp_size, margs)
     # sequential: experts.fc1.{e}.weight
     g = int(parts[i + 2]) + ep_rank * e_local
     parts[i + 2] = str(g)
     key = full_key.rsplit(name, 1)[0] + ".".join(parts)
-    return state.get(key)
+    full = state.get(key)
+    if full is None:
+        return None
+    return _etp_ffn_slice(full, parts[i + 1], tp_rank, tp_size, margs)
 
 
 def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
@@ -151,7 +186,8 @@ def load_full_state(sm: StageModel, state: Dict[str, torch.Tensor],
                 full_key = f"{prefix}.{name}"
                 if ".experts." in name:
                     full = _expert_resolve(name, full_key, state, ep_rank,
-                                           ep, margs.num_experts)
+                                           ep, margs.num_experts,
+                                           tp_rank, tp_size, margs)
                     if full is None:
                         continue
                     if full.shape != p.shape:

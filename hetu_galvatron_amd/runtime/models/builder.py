@@ -240,8 +240,6 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
                 blocks.append(bblk)
                 prev = vg
         elif is_moe:
-            assert lg.strategy.tp == 1 or lg.strategy.use_ulysses, \
-                "MoE layers: expert-TP (etp) not supported yet; use ep/dp/sp"
             dec = GalvatronMoEDecoderLayer(margs, lg, layer_idx=i, dtype=dtype)
             kind = "decoder"
         else:

@@ -7,8 +7,14 @@ TopKRouter -> dispatcher (alltoall|allgather over the layer's EP group)
 -> Grouped/Sequential experts -> optional shared expert; router aux loss
 injected through MoEAuxLossAutoScaler.
 
-v1 constraint: MoE layers run with tp==1 (ulysses/cp/dp/ep compose);
-expert-TP (etp) is future work — the builder asserts accordingly.
+Expert tensor parallelism (etp == the layer's megatron tp, reference
+comm_groups.py:322-345): the seq-sharded activation is all-gathered over
+tp (every etp rank routes the identical full token set), experts hold
+ffn/etp column/row shards producing PARTIAL outputs, and the combined
+result is reduce-scattered back to the sequence shard — one gather +
+one reduce-scatter per MoE layer, exactly the dense CPL/RPL traffic.
+The shared expert stays on the local sequence shard (applied after the
+reduce-scatter; its replicated params are tp-summed via `tp_replicated`).
 """
 from __future__ import annotations
 
@@ -22,10 +28,13 @@ from ...config.schema import ModelArgs
 from ..moe.dispatcher import AllGatherDispatcher, AlltoAllDispatcher
 from ..moe.experts import GroupedMLP, SequentialMLP, SharedExpertMLP
 from ..moe.router import TopKRouter, attach_aux_loss
+from ..tensor_parallel.mappings import (
+    gather_from_sequence_parallel_region, group_size,
+    reduce_scatter_to_sequence_parallel_region)
 from ..transformer import RotaryEmbedding
 from ..transformer.attention import SelfAttention
 from ..transformer.norm import build_norm
-from .modules import _my_rank
+from .modules import _my_rank, _tag_tp_replicated
 
 
 class GalvatronMoEMLP(nn.Module):
@@ -44,21 +53,40 @@ class GalvatronMoEMLP(nn.Module):
         n_local = margs.num_experts // max(ep, 1)
         ffn = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
         gated = margs.hidden_act in ("silu", "swiglu", "geglu")
+        # expert-TP: ffn shard per etp rank, partial outputs reduced by the
+        # post-combine reduce-scatter
+        self.tp_group = None if groups.strategy.use_ulysses else groups.tp_group
+        self.etp = group_size(self.tp_group) if self.tp_group is not None else 1
+        assert ffn % self.etp == 0, f"etp={self.etp} must divide moe ffn {ffn}"
+        if self.etp > 1:
+            # routing-path router grads are PARTIAL per etp rank (combine
+            # weights partial expert outputs); the tp-group sum assembles
+            # them — aux is scaled 1/etp at attach so its full per-rank
+            # grad sums back to 1x
+            for prm in self.router.parameters():
+                _tag_tp_replicated(prm)
+        ffn_local = ffn // self.etp
         if margs.moe_grouped_gemm:
-            self.experts = GroupedMLP(n_local, margs.hidden_size, ffn,
+            self.experts = GroupedMLP(n_local, margs.hidden_size, ffn_local,
                                       dtype=dtype, gated=gated)
         else:
-            self.experts = SequentialMLP(n_local, margs.hidden_size, ffn,
+            self.experts = SequentialMLP(n_local, margs.hidden_size, ffn_local,
                                          dtype=dtype, gated=gated)
         if margs.moe_shared_expert_intermediate_size:
             self.shared = SharedExpertMLP(
                 margs.hidden_size, margs.moe_shared_expert_intermediate_size,
                 dtype=dtype)
+            if self.etp > 1:
+                for prm in self.shared.parameters():
+                    _tag_tp_replicated(prm)
         else:
             self.shared = None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        """x: [s_local, b, h] SBH."""
+        """x: [s_local, b, h] SBH (seq/tp shard under megatron-SP)."""
+        x_in = x
+        if self.etp > 1:
+            x = gather_from_sequence_parallel_region(x, self.tp_group)
         s, b, h = x.shape
         flat = x.reshape(-1, h)
         probs, idx, aux = self.router(flat)
@@ -66,11 +94,16 @@ class GalvatronMoEMLP(nn.Module):
             flat, probs, idx)
         expert_out = self.experts(expert_in, tokens_per_expert)
         merged = self.dispatcher.combine(expert_out, flat.shape[0], self.topk)
-        if self.shared is not None:
-            merged = merged + self.shared(flat)
         if self.training and aux.requires_grad:
-            merged = attach_aux_loss(merged, aux)
-        return merged.reshape(s, b, h)
+            merged = attach_aux_loss(merged, aux / self.etp)
+        out = merged.reshape(s, b, h)
+        if self.etp > 1:
+            # partial ffn outputs: the reduce-scatter both sums the etp
+            # partials and restores the sequence shard
+            out = reduce_scatter_to_sequence_parallel_region(out, self.tp_group)
+        if self.shared is not None:
+            out = out + self.shared(x_in.reshape(-1, h)).reshape(out.shape)
+        return out
 
 
 class GalvatronMoEDecoderLayer(nn.Module):
@@ -95,6 +128,12 @@ class GalvatronMoEDecoderLayer(nn.Module):
             margs, groups.tp_group, groups.sp_group, groups.cp_group,
             use_ulysses=s.use_ulysses, sequence_parallel=seq_par, dtype=dtype)
         self.mlp = GalvatronMoEMLP(margs, groups, dtype=dtype)
+        if seq_par and group_size(groups.tp_group) > 1:
+            _tag_tp_replicated(self.input_norm.weight,
+                               getattr(self.input_norm, "bias", None),
+                               self.post_attn_norm.weight,
+                               getattr(self.post_attn_norm, "bias", None),
+                               self.attention.linear_proj.bias)
         if margs.position_embedding_type == "rope":
             self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta)
         else:

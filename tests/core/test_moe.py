@@ -155,3 +155,23 @@ def test_moe_world4_ep2_dp2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("world,tp,ep", [(2, 2, 1), (4, 2, 2)])
+def test_moe_etp_vs_baseline(world, tp, ep):
+    """Expert tensor parallelism: megatron tp shards the expert ffn
+    (partial outputs reduce-scattered to the seq shard); tp=2 alone and
+    composed with ep=2 on 4 ranks."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=world, pp=1, tp=tp, dp_type="ddp",
+        global_bsz=4, chunks=1, ep=ep)
+    res = run_distributed(_moe_worker, world_size=world,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
