@@ -55,7 +55,7 @@ void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, floa
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
-void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, hipStream_t);
+void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, const int*, hipStream_t);
 
 namespace {
 
@@ -303,7 +303,33 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   }
   decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o), ws_ptr,
                      n_chunks, b, hq, hkv, max_s, (int)cur_len, d,
-                     (float)scale, cur_stream());
+                     (float)scale, nullptr, cur_stream());
+  return o;
+}
+
+// hipGraph-capturable variant: the attended length lives in device memory
+// (cur_len_dev int32 [1]); launch geometry + LDS are sized for max_len so
+// a captured graph replays correctly as the cache grows.
+Tensor decode_attn_graph(const Tensor& q, const Tensor& k_cache,
+                         const Tensor& v_cache, const Tensor& cur_len_dev,
+                         int64_t max_len, double scale) {
+  CHECK_GPU(q); CHECK_GPU(cur_len_dev);
+  TORCH_CHECK(is_bf16(q) && q.dim() == 3, "decode_attn_graph: bf16 [b,hq,d]");
+  TORCH_CHECK(cur_len_dev.scalar_type() == at::kInt, "cur_len_dev: int32");
+  const int b = q.size(0), hq = q.size(1), d = q.size(2);
+  const int max_s = k_cache.size(1), hkv = k_cache.size(2);
+  TORCH_CHECK(d == 64 || d == 128, "decode_attn_graph: head dim 64|128");
+  TORCH_CHECK(max_len >= 1 && max_len <= max_s, "bad max_len");
+  auto o = at::empty_like(q);
+  int n_chunks = b * hq >= 256 ? 1 : std::max(1, 512 / (b * hq));
+  n_chunks = std::min<int>(n_chunks, (int)((max_len + 127) / 128));
+  n_chunks = std::max<int>(n_chunks, (int)((max_len + 8191) / 8192));
+  auto ws = at::empty({(long)b * hq * n_chunks * (d + 2)},
+                      q.options().dtype(at::kFloat));
+  decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o),
+                     ws.data_ptr<float>(), n_chunks, b, hq, hkv, max_s,
+                     (int)max_len, d, (float)scale,
+                     cur_len_dev.data_ptr<int>(), cur_stream());
   return o;
 }
 
@@ -552,6 +578,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("decode_attn", &decode_attn);
+  m.def("decode_attn_graph", &decode_attn_graph);
   m.def("mfma_probe", &mfma_probe);
   m.def("ce_max", &ce_max);
   m.def("ce_sum_target", &ce_sum_target);

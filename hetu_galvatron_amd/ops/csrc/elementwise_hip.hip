@@ -254,9 +254,15 @@ __global__ void decode_attn_split_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ kc,
     const __bf16* __restrict__ vc,
     float* __restrict__ part,  // [b*hq, n_chunks, D+2]
-    int b, int hq, int hkv, int max_s, int cur_len, int chunk, float scale) {
+    int b, int hq, int hkv, int max_s, int cur_len, int chunk, float scale,
+    const int* __restrict__ cur_len_dev) {  // non-null: hipGraph mode, the
+                                            // length lives in device memory
   extern __shared__ __align__(16) float wsm[];  // [chunk] weights
   __shared__ float red[2][8];
+  if (cur_len_dev != nullptr) {
+    cur_len = *cur_len_dev;
+    chunk = (cur_len + gridDim.y - 1) / gridDim.y;
+  }
   const int bh = blockIdx.x;
   const int bi = bh / hq;
   const int h = bh - bi * hq;
@@ -371,8 +377,8 @@ __global__ void decode_attn_merge_kernel(const float* __restrict__ part,
 void decode_attn_launch(const __bf16* q, const __bf16* kc, const __bf16* vc,
                         __bf16* o, float* part_ws, int n_chunks, int b,
                         int hq, int hkv, int max_s, int cur_len, int d,
-                        float scale, hipStream_t st) {
-  if (n_chunks <= 1) {
+                        float scale, const int* cur_len_dev, hipStream_t st) {
+  if (n_chunks <= 1 && cur_len_dev == nullptr) {
     const int shmem = cur_len * sizeof(float);
     if (d == 64)
       hipLaunchKernelGGL((decode_attn_kernel<64>), dim3(b * hq), dim3(256),
@@ -384,19 +390,21 @@ void decode_attn_launch(const __bf16* q, const __bf16* kc, const __bf16* vc,
                          scale);
     return;
   }
+  // graph mode sizes the LDS weight buffer for the worst case (cur_len
+  // unknown at launch); chunk passed is then the max over the capture
   const int chunk = (cur_len + n_chunks - 1) / n_chunks;
   const int shmem = chunk * sizeof(float);
   dim3 grid(b * hq, n_chunks);
   if (d == 64) {
     hipLaunchKernelGGL((decode_attn_split_kernel<64>), grid, dim3(256),
                        shmem, st, q, kc, vc, part_ws, b, hq, hkv, max_s,
-                       cur_len, chunk, scale);
+                       cur_len, chunk, scale, cur_len_dev);
     hipLaunchKernelGGL((decode_attn_merge_kernel<64>), dim3(b * hq),
                        dim3(64), 0, st, part_ws, o, n_chunks);
   } else {
     hipLaunchKernelGGL((decode_attn_split_kernel<128>), grid, dim3(256),
                        shmem, st, q, kc, vc, part_ws, b, hq, hkv, max_s,
-                       cur_len, chunk, scale);
+                       cur_len, chunk, scale, cur_len_dev);
     hipLaunchKernelGGL((decode_attn_merge_kernel<128>), dim3(b * hq),
                        dim3(128), 0, st, part_ws, o, n_chunks);
   }

@@ -77,6 +77,8 @@ class GalvatronGenerator:
         self.max_seq = max_seq
         self.scale = 1.0 / (m.head_dim ** 0.5)
         self._dev = next(self.embedding.parameters()).device
+        self._graphs: dict = {}   # batch size -> captured decode step
+        self._kv: dict = {}       # batch size -> persistent KVCache
 
     # -- one decoder layer, cache-aware ------------------------------------
     def _split_qkv(self, attn, qkv: torch.Tensor):
@@ -182,6 +184,145 @@ class GalvatronGenerator:
             for l, t in zip(self.layers, was_training):
                 l.train(t)
         return torch.cat(out, dim=1)
+
+    # ------------------------------------------------------------ hipGraph
+    @torch.no_grad()
+    def generate_graphed(self, input_ids: torch.Tensor,
+                         max_new_tokens: int = 32,
+                         warmup_steps: int = 3) -> torch.Tensor:
+        """Greedy generation with the decode step captured in ONE hipGraph.
+
+        The step is fully device-driven — position index, KV-cache
+        index_copy, decode kernel length (decode_attn_graph reads it from
+        device memory), argmax and its feedback into the next step's
+        input all live inside the capture — so each new token is a single
+        graph replay with zero host work (the eager loop costs ~9 ms/token
+        in Python/launch overhead vs ~0.5 ms of kernel time).  Falls back
+        to the eager path if capture fails.
+        """
+        b, sp = input_ids.shape
+        assert sp + max_new_tokens <= self.max_seq
+        m = self.margs
+        dev = self._dev
+        # persistent per-batch-size cache: the captured graph holds device
+        # pointers into these tensors, so reuse across requests is what
+        # makes capture a one-time cost
+        cache = self._kv.get(b)
+        if cache is None:
+            cache = KVCache(len(self.layers), b, self.max_seq, m.kv_heads,
+                            m.head_dim, dev,
+                            dtype=next(self.embedding.parameters()).dtype)
+            self._kv[b] = cache
+        cache.cur_len = 0
+        for blk in self.blocks:
+            if blk.flat is not None:
+                blk.flat.gather_params()
+        was_training = [l.training for l in self.layers]
+        for blk in self.blocks:
+            blk.inner.eval()
+        try:
+            logits = self._forward_tokens(input_ids, cache)  # prefill
+            toks = [logits.argmax(-1)]
+            # eager head start: real decode steps double as hipBLASLt /
+            # allocator warmup for the capture
+            n_eager = min(warmup_steps, max_new_tokens - 1)
+            for _ in range(n_eager):
+                logits = self._forward_tokens(toks[-1].unsqueeze(1), cache)
+                toks.append(logits.argmax(-1))
+            n_graph = max_new_tokens - 1 - n_eager
+            if n_graph > 0:
+                out_g = self._run_graph_steps(toks[-1], cache, n_graph)
+                toks.extend(out_g)
+            return torch.cat([input_ids] +
+                             [t.unsqueeze(1) for t in toks[:max_new_tokens]],
+                             dim=1)
+        finally:
+            for l, t in zip(self.layers, was_training):
+                l.train(t)
+
+    def _graph_step(self, ids_buf, pos_i64, cur32, cache: KVCache):
+        """One decode step on static buffers (capture-safe)."""
+        from ..ops._ext import get_ext
+        from ..ops import apply_rope
+        ext = get_ext()
+        h = self.embedding.word_embeddings(ids_buf)  # [1, b, h]
+        for li, layer in enumerate(self.layers):
+            attn = layer.attention
+            residual = h
+            x = layer.input_norm(h)
+            qkv = attn.linear_qkv(x)
+            q, k, v = self._split_qkv(attn, qkv)
+            if layer.rotary is not None:
+                cos_t, sin_t = layer.rotary.full_tables(self.max_seq,
+                                                        h.device)
+                cos = cos_t.index_select(0, pos_i64)
+                sin = sin_t.index_select(0, pos_i64)
+                q = apply_rope(q.contiguous(), cos, sin)
+                k = apply_rope(k.contiguous(), cos, sin)
+            cache.k[li].index_copy_(1, pos_i64, k.permute(1, 0, 2, 3))
+            cache.v[li].index_copy_(1, pos_i64, v.permute(1, 0, 2, 3))
+            o = ext.decode_attn_graph(q[0].contiguous(), cache.k[li],
+                                      cache.v[li], cur32, self.max_seq,
+                                      self.scale)
+            o = o.reshape(1, o.shape[0], -1)
+            h = residual + attn.linear_proj(o)
+            residual = h
+            h = residual + layer.mlp(layer.post_attn_norm(h))
+        h = self.final_norm.norm(h)
+        logits = self.lm_head.lm_head(h)[0].float()
+        nxt = logits.argmax(-1)  # [b]
+        ids_buf.copy_(nxt.unsqueeze(1))
+        pos_i64.add_(1)
+        cur32.add_(1)
+        return nxt
+
+    def _run_graph_steps(self, first_ids: torch.Tensor, cache: KVCache,
+                         n_steps: int):
+        """Replay the captured decode step n_steps times.  The capture is
+        keyed by batch size and reused across generate_graphed calls (the
+        graph holds pointers into THIS generator's persistent KV cache, so
+        callers must pass the cache generate_graphed allocated)."""
+        dev = self._dev
+        b = first_ids.shape[0]
+        ent = self._graphs.get(b)
+        if ent is None:
+            ids_buf = first_ids.unsqueeze(1).clone()          # [b, 1]
+            pos_i64 = torch.tensor([cache.cur_len], dtype=torch.long,
+                                   device=dev)
+            cur32 = torch.tensor([cache.cur_len + 1], dtype=torch.int32,
+                                 device=dev)
+            out_buf = torch.empty(self.max_seq, b, dtype=first_ids.dtype,
+                                  device=dev)
+            step_i = torch.zeros(1, dtype=torch.long, device=dev)
+            for layer in self.layers:  # tables must exist before capture
+                if layer.rotary is not None:
+                    layer.rotary.full_tables(self.max_seq, dev)
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    nxt = self._graph_step(ids_buf, pos_i64, cur32, cache)
+                    out_buf.index_copy_(0, step_i, nxt.unsqueeze(0))
+                    step_i.add_(1)
+            except Exception:
+                # graph capture unavailable: eager fallback
+                out = []
+                ids = first_ids
+                for _ in range(n_steps):
+                    logits = self._forward_tokens(ids.unsqueeze(1), cache)
+                    ids = logits.argmax(-1)
+                    out.append(ids)
+                return out
+            ent = (g, ids_buf, pos_i64, cur32, out_buf, step_i)
+            self._graphs[b] = ent
+        g, ids_buf, pos_i64, cur32, out_buf, step_i = ent
+        ids_buf.copy_(first_ids.unsqueeze(1))
+        pos_i64.fill_(cache.cur_len)
+        cur32.fill_(cache.cur_len + 1)
+        step_i.zero_()
+        for _ in range(n_steps):
+            g.replay()
+        cache.cur_len = int(pos_i64.item())
+        return list(out_buf[:n_steps].unbind(0))
 
     @staticmethod
     def _sample(logits: torch.Tensor, temperature: float, top_k: int

@@ -77,3 +77,15 @@ def test_generate_cli(capsys):
     assert out.shape == (1, 6)
     lines = capsys.readouterr().out.strip().splitlines()
     assert lines[-1].startswith("5 6 7 ")
+
+
+def test_generate_graphed_cpu_fallback():
+    """On CPU the capture raises and generate_graphed must fall back to
+    the eager loop with identical greedy output."""
+    model = make_model()
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    torch.manual_seed(4)
+    ids = torch.randint(0, model.cfg.model.vocab_size, (2, 9))
+    out_e = gen.generate(ids, max_new_tokens=6, temperature=0.0)
+    out_g = gen.generate_graphed(ids, max_new_tokens=6, warmup_steps=2)
+    assert torch.equal(out_e, out_g)

@@ -364,3 +364,47 @@ def test_generate_gpu_native_decode():
         logits = gen._forward_tokens(out[:, :9 + i], cache)
         got = logits.gather(1, out[:, 9 + i:10 + i])
         assert (logits.max(-1, keepdim=True).values - got).max() < 0.05
+
+
+def test_decode_attn_graph_variant():
+    """Device-resident length (hipGraph mode) == host-length kernel."""
+    from hetu_galvatron_amd.ops._ext import get_ext
+    torch.manual_seed(3)
+    b, hq, hkv, d, max_s = 2, 32, 8, 128, 2048
+    q = torch.randn(b, hq, d, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(b, max_s, hkv, d, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(b, max_s, hkv, d, device="cuda", dtype=torch.bfloat16)
+    for cur in (1, 300, 2048):
+        cur32 = torch.tensor([cur], dtype=torch.int32, device="cuda")
+        og = get_ext().decode_attn_graph(q, kc, vc, cur32, max_s,
+                                         d ** -0.5)
+        k = kc[:, :cur].float().repeat_interleave(hq // hkv, dim=2)
+        v = vc[:, :cur].float().repeat_interleave(hq // hkv, dim=2)
+        att = torch.einsum("bhd,bshd->bhs", q.float(), k) * d ** -0.5
+        want = torch.einsum("bhs,bshd->bhd", att.softmax(-1), v)
+        assert (og.float() - want).abs().max() < 0.02, cur
+
+
+def test_generate_graphed_gpu():
+    """hipGraph-captured decode: every token a near-argmax of a
+    full-prefix recompute (falls back to eager if capture unavailable)."""
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.inference import (GalvatronGenerator,
+                                                      KVCache)
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1},
+    })
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    ids = torch.randint(0, cfg.model.vocab_size, (2, 9), device="cuda")
+    out = gen.generate_graphed(ids, max_new_tokens=8, warmup_steps=2)
+    assert out.shape == (2, 17)
+    for i in range(8):
+        cache = KVCache(len(gen.layers), 2, 64, cfg.model.kv_heads,
+                        cfg.model.head_dim, ids.device)
+        logits = gen._forward_tokens(out[:, :9 + i], cache)
+        got = logits.gather(1, out[:, 9 + i:10 + i])
+        assert (logits.max(-1, keepdim=True).values - got).max() < 0.05, i

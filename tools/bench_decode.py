@@ -73,6 +73,17 @@ for batch in (1, 8):
         "ms_per_token": round(dt / n_new * 1e3, 2),
         "decode_tokens_per_s": round(batch * n_new / dt, 1)}
     print(results[f"generate_8b_b{batch}"])
+    # hipGraph-captured decode step (one replay per token)
+    t0 = time.perf_counter()
+    outg = gen.generate_graphed(ids, max_new_tokens=n_new, warmup_steps=3)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    assert outg.shape == (batch, 512 + n_new)
+    results[f"generate_8b_b{batch}_hipgraph"] = {
+        "prompt": 512, "new_tokens": n_new,
+        "ms_per_token": round(dt / n_new * 1e3, 2),
+        "decode_tokens_per_s": round(batch * n_new / dt, 1)}
+    print(results[f"generate_8b_b{batch}_hipgraph"])
 
 os.makedirs("gpurun_out", exist_ok=True)
 json.dump(results, open("gpurun_out/decode_bench.json", "w"), indent=1)
