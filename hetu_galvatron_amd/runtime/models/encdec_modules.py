@@ -41,8 +41,7 @@ def _build_rel_bias(margs: ModelArgs, groups, bidirectional: bool, dtype):
     if margs.position_embedding_type != "relative":
         return None
     s = groups.strategy
-    assert not s.use_ulysses and s.cp == 1, \
-        "relative-position bias under ulysses/cp is a v2 item"
+    assert s.cp == 1, "relative-position bias under ring-CP is a v2 item"
     rb = RelativePositionBias(
         margs.relative_attention_num_buckets,
         margs.relative_attention_max_distance,
@@ -55,10 +54,17 @@ def _build_rel_bias(margs: ModelArgs, groups, bidirectional: bool, dtype):
 
 
 def _rank_bias(rel_bias, attention, groups, hidden):
-    """[h_local, S, S] bias for this tp rank; S = the attention-visible
-    sequence (megatron-SP input arrives seq-sharded S/tp)."""
+    """Bias for the heads of the q entering SelfAttention.forward.
+    megatron-SP: input arrives seq-sharded S/tp, heads pre-sharded ->
+    slice this tp rank's chunk of a full-seq bias.  ulysses: heads are
+    full before the a2a -> full-head bias (DistributedAttention slices
+    the post-a2a chunk itself)."""
     if rel_bias is None:
         return None
+    s = groups.strategy
+    if s.use_ulysses:
+        S = hidden.shape[0] * s.tp_sp
+        return rel_bias(S, S, hidden.device)
     tp = group_size(groups.tp_group)
     S = hidden.shape[0] * tp
     c = groups.coord_of(_my_rank())
@@ -158,9 +164,8 @@ class GalvatronDecoderLayerX(nn.Module):
         self.layer_idx = layer_idx
         s = groups.strategy
         self.strategy = s
-        assert not s.use_ulysses and s.cp == 1, \
-            "t5 decoder layers: ulysses/cp on cross-attention is v2"
-        seq_par = True
+        assert s.cp == 1, "t5 decoder layers: ring-CP is a v2 item"
+        seq_par = not s.use_ulysses
         self.input_norm = build_norm(margs.normalization, margs.hidden_size,
                                      margs.norm_epsilon, dtype)
         self.cross_norm = build_norm(margs.normalization, margs.hidden_size,
@@ -170,12 +175,13 @@ class GalvatronDecoderLayerX(nn.Module):
                                          margs.norm_epsilon, dtype)
         self.attention = SelfAttention(
             margs, groups.tp_group, groups.sp_group, groups.cp_group,
-            use_ulysses=False, sequence_parallel=seq_par, dtype=dtype,
+            use_ulysses=s.use_ulysses, sequence_parallel=seq_par, dtype=dtype,
             causal=True)
         self.cross_attention = CrossAttention(
-            margs, groups.tp_group, sequence_parallel=seq_par, dtype=dtype)
+            margs, groups.tp_group, sequence_parallel=seq_par, dtype=dtype,
+            sp_group=groups.sp_group, use_ulysses=s.use_ulysses)
         self.mlp = MLP(margs.hidden_size, margs.ffn_hidden_size,
-                       groups.tp_group,
+                       None if s.use_ulysses else groups.tp_group,
                        hidden_act="silu"
                        if margs.hidden_act in ("silu", "swiglu")
                        else margs.hidden_act,

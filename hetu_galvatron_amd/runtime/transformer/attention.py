@@ -24,7 +24,8 @@ from ..tensor_parallel import (
     group_size,
 )
 from .attention_impl import (
-    DistributedAttention, ZigzagRingAttention, local_attention,
+    DistributedAttention, ZigzagRingAttention, eager_bias_attention,
+    local_attention,
 )
 from .rope import apply_rope_qk
 
@@ -93,9 +94,16 @@ class SelfAttention(nn.Module):
         q = q.permute(1, 0, 2, 3).contiguous()
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()
-        if attn_bias is not None:
+        if attn_bias is not None and isinstance(self.core_attention,
+                                                DistributedAttention):
+            # ulysses: pass the full-head bias; the a2a re-shards heads and
+            # DistributedAttention slices the matching chunk
+            o = self.core_attention(q, k, v, causal=self.causal,
+                                    softmax_scale=self.softmax_scale,
+                                    attn_bias=attn_bias)
+        elif attn_bias is not None:
             assert self.core_attention is None, \
-                "attn_bias (t5 relative bias) needs local/megatron-tp attention"
+                "attn_bias (t5 relative bias) + ring-CP is a v2 item"
             o = eager_bias_attention(q, k, v, attn_bias, self.causal,
                                      self.softmax_scale)
         elif self.core_attention is not None:
@@ -108,67 +116,57 @@ class SelfAttention(nn.Module):
         return self.linear_proj(o)
 
 
-def eager_bias_attention(q, k, v, bias, causal: bool, scale: float
-                         ) -> torch.Tensor:
-    """fp32 softmax attention with an additive score bias [h, sq, skv]
-    (t5 relative-position bias; native flash bias input is a v2 kernel
-    item).  q/k/v: [b, s, h, d]."""
-    b, sq, h, d = q.shape
-    skv = k.shape[1]
-    if k.shape[2] != h:
-        rep = h // k.shape[2]
-        k = k.repeat_interleave(rep, dim=2)
-        v = v.repeat_interleave(rep, dim=2)
-    att = torch.einsum("bqhd,bkhd->bhqk", q.float(), k.float()) * scale
-    att = att + bias.unsqueeze(0).float()
-    if causal:
-        mask = torch.ones(sq, skv, dtype=torch.bool, device=q.device) \
-            .triu(diagonal=1 + skv - sq)
-        att = att.masked_fill(mask, float("-inf"))
-    o = torch.einsum("bhqk,bkhd->bqhd", att.softmax(-1), v.float())
-    return o.to(q.dtype)
-
-
 class CrossAttention(nn.Module):
     """Decoder->encoder cross attention (reference: attention.py:929
     CrossAttention).  Queries come from the (possibly seq-sharded) decoder
     hidden state; keys/values from the FULL-sequence encoder memory the
     EncDecBridge replicated within the dp group, so no redistribution is
-    needed however the decoder layer is laid out (tp heads shard both q
-    and kv; cp/ulysses on cross-attention are a v2 item — the builder
-    asserts them off for t5).
+    needed however the decoder layer is laid out.
+
+    megatron tp: heads shard both q and kv via the parallel linears.
+    ulysses sp: linears replicated; q a2a's decoder-seq->heads (like
+    DistributedAttention), kv computed on the full memory and sliced to
+    this rank's post-a2a head chunk (its weight-grad slices re-assemble
+    through the ulysses sdp all-reduce, which spans the sp group).
     """
 
     def __init__(self, model_args, tp_group, sequence_parallel: bool = True,
-                 dtype=None):
+                 dtype=None, sp_group=None, use_ulysses: bool = False):
         super().__init__()
         m = model_args
         self.num_heads = m.num_attention_heads
         self.num_kv_heads = m.kv_heads
         self.head_dim = m.head_dim
         self.tp_group = tp_group
-        tp = group_size(tp_group)
+        self.sp_group = sp_group
+        self.use_ulysses = use_ulysses
+        lin_group = None if use_ulysses else tp_group
+        seq_par = sequence_parallel and not use_ulysses
+        tp = 1 if use_ulysses else group_size(tp_group)
         assert self.num_heads % tp == 0 and self.num_kv_heads % tp == 0
+        if use_ulysses and sp_group is not None:
+            sp = group_size(sp_group)
+            assert self.num_heads % sp == 0, "sp must divide heads"
         self.linear_q = ColumnParallelLinear(
-            m.hidden_size, self.num_heads * self.head_dim, tp_group,
+            m.hidden_size, self.num_heads * self.head_dim, lin_group,
             bias=m.add_qkv_bias or m.add_bias_linear,
-            sequence_parallel=sequence_parallel, dtype=dtype)
+            sequence_parallel=seq_par, dtype=dtype)
         # kv over the full-seq memory: no sequence-parallel gather
         self.linear_kv = ColumnParallelLinear(
-            m.hidden_size, 2 * self.num_kv_heads * self.head_dim, tp_group,
+            m.hidden_size, 2 * self.num_kv_heads * self.head_dim, lin_group,
             bias=m.add_qkv_bias or m.add_bias_linear,
             sequence_parallel=False, dtype=dtype)
         self.linear_proj = RowParallelLinear(
-            self.num_heads * self.head_dim, m.hidden_size, tp_group,
-            bias=m.add_bias_linear, sequence_parallel=sequence_parallel,
+            self.num_heads * self.head_dim, m.hidden_size, lin_group,
+            bias=m.add_bias_linear, sequence_parallel=seq_par,
             dtype=dtype)
         self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
-        tp = group_size(tp_group)
         self.heads_local = self.num_heads // tp
         self.kv_heads_local = self.num_kv_heads // tp
 
     def forward(self, x: torch.Tensor, memory: torch.Tensor) -> torch.Tensor:
         """x: [s_dec(_shard), b, h]; memory: [s_enc, b, h] full sequence."""
+        from ..tensor_parallel.mappings import all_to_all, group_rank
         q = self.linear_q(x)            # [s_dec, b, hq_l*d]
         kv = self.linear_kv(memory)     # [s_enc, b, 2*hkv_l*d]
         s, bsz = q.shape[0], q.shape[1]
@@ -179,7 +177,23 @@ class CrossAttention(nn.Module):
         q = q.permute(1, 0, 2, 3).contiguous()
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()
+        sp = group_size(self.sp_group) if (self.use_ulysses and
+                                           self.sp_group is not None) else 1
+        if sp > 1:
+            hkv = k.shape[2]
+            if hkv < sp:
+                rep = sp // hkv
+                k = k.repeat_interleave(rep, dim=2)
+                v = v.repeat_interleave(rep, dim=2)
+            q = all_to_all(q, self.sp_group, scatter_dim=2, gather_dim=1)
+            # memory kv is already full-seq: take this rank's head chunk
+            r = group_rank(self.sp_group)
+            hl = k.shape[2] // sp
+            k = k[:, :, r * hl:(r + 1) * hl].contiguous()
+            v = v[:, :, r * hl:(r + 1) * hl].contiguous()
         o = local_attention(q, k, v, causal=False,
                             softmax_scale=self.softmax_scale)
+        if sp > 1:
+            o = all_to_all(o, self.sp_group, scatter_dim=1, gather_dim=2)
         o = o.permute(1, 0, 2, 3).reshape(s, bsz, -1)
         return self.linear_proj(o)

@@ -34,6 +34,27 @@ def local_attention(q, k, v, causal: bool = True,
 # Reference: attention_impl.py:139-405 (single_all_to_all, DistributedAttention)
 
 
+def eager_bias_attention(q, k, v, bias, causal: bool, scale: float
+                         ) -> torch.Tensor:
+    """fp32 softmax attention with an additive score bias [h, sq, skv]
+    (t5 relative-position bias; native flash bias input is a v2 kernel
+    item).  q/k/v: [b, s, h, d]."""
+    b, sq, h, d = q.shape
+    skv = k.shape[1]
+    if k.shape[2] != h:
+        rep = h // k.shape[2]
+        k = k.repeat_interleave(rep, dim=2)
+        v = v.repeat_interleave(rep, dim=2)
+    att = torch.einsum("bqhd,bkhd->bhqk", q.float(), k.float()) * scale
+    att = att + bias.unsqueeze(0).float()
+    if causal:
+        mask = torch.ones(sq, skv, dtype=torch.bool, device=q.device) \
+            .triu(diagonal=1 + skv - sq)
+        att = att.masked_fill(mask, float("-inf"))
+    o = torch.einsum("bhqk,bkhd->bqhd", att.softmax(-1), v.float())
+    return o.to(q.dtype)
+
+
 class DistributedAttention(torch.nn.Module):
     """a2a q,k,v (scatter heads, gather seq) -> inner attention -> a2a out.
 
@@ -47,8 +68,11 @@ class DistributedAttention(torch.nn.Module):
         self.sp_group = sp_group
         self.inner_attention = inner_attention
 
-    def forward(self, q, k, v, causal=True, softmax_scale=None):
-        # q: [b, s_local, hq, d]; k/v: [b, s_local, hkv, d]
+    def forward(self, q, k, v, causal=True, softmax_scale=None,
+                attn_bias=None):
+        # q: [b, s_local, hq, d]; k/v: [b, s_local, hkv, d].
+        # attn_bias [H_full, S, S] (t5 relative bias): sliced to this
+        # rank's post-a2a head chunk; incompatible with a ring inner.
         sp = group_size(self.sp_group)
         if sp > 1:
             hkv = k.shape[2]
@@ -61,7 +85,15 @@ class DistributedAttention(torch.nn.Module):
             q = all_to_all(q, self.sp_group, scatter_dim=2, gather_dim=1)
             k = all_to_all(k, self.sp_group, scatter_dim=2, gather_dim=1)
             v = all_to_all(v, self.sp_group, scatter_dim=2, gather_dim=1)
-        if self.inner_attention is not None:
+        if attn_bias is not None:
+            assert self.inner_attention is None, \
+                "relative bias + ring-CP inner attention is a v2 item"
+            r = group_rank(self.sp_group)
+            hl = q.shape[2]
+            o = eager_bias_attention(q, k, v,
+                                     attn_bias[r * hl:(r + 1) * hl],
+                                     causal, softmax_scale)
+        elif self.inner_attention is not None:
             o = self.inner_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
         else:
             o = local_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)

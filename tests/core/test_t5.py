@@ -132,3 +132,21 @@ def test_t5_pp2_vs_baseline(division, chunks):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_ulysses_vs_baseline():
+    """Ulysses SP on both stacks: encoder/decoder self-attention bias
+    sliced post-a2a, cross-attention q a2a + kv head-chunking."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=2, pp=1, tp=2, use_sp=True, dp_type="ddp",
+        global_bsz=4, chunks=1)
+    res = run_distributed(_t5_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
