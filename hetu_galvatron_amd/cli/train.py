@@ -45,13 +45,28 @@ def main(argv=None):
 
     model = GalvatronModel(cfg, device=device)
     opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
-    it = RerunDataIterator(get_train_iterator(cfg, device))
+    # batch ramp-up (reference num_microbatches_calculator): unit =
+    # dp * chunks so every ramped global batch chunks evenly
+    from ..runtime.optimizer.microbatches import build_batch_calculator
+    plan = model.plan
+    unit = plan.layer(0, max(world, 1)).dp * max(plan.chunks, 1)
+    calc = build_batch_calculator(cfg, dp=unit, micro_batch_size=1)
+    cur_gbs = calc.get()[0]
+    consumed = 0
+    it = RerunDataIterator(get_train_iterator(cfg, device,
+                                              global_batch=cur_gbs))
     prof = RuntimeProfiler(enabled=use_gpu, device=device, rank=rank)
     rsm = initialize_rerun_state_machine(enabled=True)
     mlog = MetricsLogger(cfg, rank=rank)
 
     iters = cfg.train.train_iters
     for i in range(iters):
+        calc.update(consumed)
+        if calc.get()[0] != cur_gbs:
+            cur_gbs = calc.get()[0]
+            it = RerunDataIterator(get_train_iterator(cfg, device,
+                                                      global_batch=cur_gbs))
+        consumed += cur_gbs
         prof.profile_memory("Before-Fwd")
         prof.time_start()
         loss, norm = float("nan"), 0.0

@@ -9,7 +9,7 @@ overrides (see loader.py); defaults sized for 288 GB HBM3E per GPU and an
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List, Optional, Union
 
 from pydantic import BaseModel, Field, model_validator
 
@@ -83,6 +83,10 @@ class ModelArgs(BaseModel):
     moe_ffn_hidden_size: Optional[int] = None
     moe_aux_loss_free: bool = False
     moe_router_bias_update_rate: float = 1e-3
+    moe_aux_loss_type: str = "aux_loss"  # aux_loss | seq_aux_loss
+    # DeepSeek-style group-limited (node-limited) routing
+    moe_router_num_groups: Optional[int] = None
+    moe_router_group_topk: Optional[int] = None
 
     @property
     def kv_heads(self) -> int:
@@ -103,6 +107,8 @@ class TrainArgs(BaseModel):
     """Reference: CommonTrainArgs (args_schema.py:195-270)."""
 
     global_train_batch_size: int = 8
+    # "start,increment,ramp_samples" (string or 3-int list)
+    rampup_batch_size: Optional[Union[str, List[int]]] = None
     epochs: int = 1
     train_iters: int = 20
     lr: float = 1e-4

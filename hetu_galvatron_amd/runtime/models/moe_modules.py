@@ -38,9 +38,11 @@ from .modules import _my_rank, _tag_tp_replicated
 
 
 class GalvatronMoEMLP(nn.Module):
-    def __init__(self, margs: ModelArgs, groups, dtype=None):
+    def __init__(self, margs: ModelArgs, groups, dtype=None,
+                 layer_idx: int = 0):
         super().__init__()
         self.margs = margs
+        self.layer_idx = layer_idx
         self.topk = margs.moe_router_topk
         self.router = TopKRouter(margs, dtype=dtype)
         ep_group = groups.ep_group
@@ -89,7 +91,11 @@ class GalvatronMoEMLP(nn.Module):
             x = gather_from_sequence_parallel_region(x, self.tp_group)
         s, b, h = x.shape
         flat = x.reshape(-1, h)
-        probs, idx, aux = self.router(flat)
+        probs, idx, aux = self.router(flat, seq_len=s)
+        if self.training:
+            from .. import moe  # noqa: F401  (package anchor)
+            from ..moe import tracker
+            tracker.save_aux_loss("moe_aux", self.layer_idx, aux)
         expert_in, tokens_per_expert = self.dispatcher.dispatch(
             flat, probs, idx)
         expert_out = self.experts(expert_in, tokens_per_expert)
@@ -127,7 +133,8 @@ class GalvatronMoEDecoderLayer(nn.Module):
         self.attention = SelfAttention(
             margs, groups.tp_group, groups.sp_group, groups.cp_group,
             use_ulysses=s.use_ulysses, sequence_parallel=seq_par, dtype=dtype)
-        self.mlp = GalvatronMoEMLP(margs, groups, dtype=dtype)
+        self.mlp = GalvatronMoEMLP(margs, groups, dtype=dtype,
+                                   layer_idx=layer_idx)
         if seq_par and group_size(groups.tp_group) > 1:
             _tag_tp_replicated(self.input_norm.weight,
                                getattr(self.input_norm, "bias", None),

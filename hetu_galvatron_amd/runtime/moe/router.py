@@ -58,6 +58,12 @@ class TopKRouter(nn.Module):
         self.aux_loss_free = margs.moe_aux_loss_free
         self.bias_update_rate = margs.moe_router_bias_update_rate
         self.capacity_factor = margs.moe_expert_capacity_factor
+        self.aux_loss_type = getattr(margs, "moe_aux_loss_type", "aux_loss")
+        self.num_groups = getattr(margs, "moe_router_num_groups", None)
+        self.group_topk = getattr(margs, "moe_router_group_topk", None)
+        if self.num_groups:
+            assert self.num_experts % self.num_groups == 0
+            assert self.group_topk and self.group_topk <= self.num_groups
         # fp32 router weight (routing numerics, reference router.py:70)
         self.weight = nn.Parameter(
             torch.empty(self.num_experts, margs.hidden_size,
@@ -70,9 +76,11 @@ class TopKRouter(nn.Module):
         else:
             self.expert_bias = None
 
-    def forward(self, x: torch.Tensor
+    def forward(self, x: torch.Tensor, seq_len: int = None
                 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-        """x: [n, h] -> (probs [n,k], indices [n,k] long, logits [n,E])."""
+        """x: [n, h] -> (probs [n,k], indices [n,k] long, logits [n,E]).
+        seq_len: rows per sequence in x (n = seq_len*b, seq-major), needed
+        only for moe_aux_loss_type=seq_aux_loss."""
         # routing math in fp32 (the flat-param engine may hold the
         # weight in bf16; cast per-call, grads flow through the cast)
         logits = F.linear(x.float(), self.weight.float())  # [n, E]
@@ -88,6 +96,19 @@ class TopKRouter(nn.Module):
         sel = scores
         if self.expert_bias is not None:
             sel = scores + self.expert_bias.unsqueeze(0)
+        if self.num_groups:
+            # group-limited (node-limited) routing, DeepSeek-style
+            # (reference moe_utils.py:372 group_limited_topk): score each
+            # expert group by its top-2 sum, keep the best group_topk
+            # groups, mask the rest before the expert topk
+            G = self.num_groups
+            gs = sel.view(n, G, E // G)
+            group_score = gs.topk(min(2, E // G), dim=-1).values.sum(-1)
+            keep = group_score.topk(self.group_topk, dim=-1).indices
+            gmask = torch.zeros(n, G, device=sel.device,
+                                dtype=torch.bool).scatter_(1, keep, True)
+            sel = gs.masked_fill(~gmask.unsqueeze(-1),
+                                 float("-inf")).view(n, E)
         top_vals, idx = torch.topk(sel, self.topk, dim=-1)
         gathered = torch.gather(scores, 1, idx)
 
@@ -99,17 +120,43 @@ class TopKRouter(nn.Module):
         else:
             probs = torch.softmax(gathered, dim=-1)
 
+        if self.capacity_factor:
+            # capacity dropping (reference moe_utils.py:147
+            # topk_softmax_with_capacity): tokens beyond
+            # ceil(n*topk/E * cf) per expert, in token order, get prob 0
+            # (they still travel through the dispatcher — pad-to-capacity
+            # buffers are a v2 item — but contribute nothing)
+            import math
+            cap = int(math.ceil(n * self.topk / E * self.capacity_factor))
+            with torch.no_grad():
+                flat_idx = idx.flatten()
+                onehot = F.one_hot(flat_idx, E)
+                rank_in_expert = onehot.cumsum(0).gather(
+                    1, flat_idx.unsqueeze(1)).squeeze(1)  # 1-based
+                keep_tok = (rank_in_expert <= cap).view(n, self.topk)
+            probs = probs * keep_tok.to(probs.dtype)
+
         aux = logits.new_zeros(())
         if self.aux_loss_coeff > 0 and not self.aux_loss_free \
                 and self.training:
-            # load-balancing loss: E * sum_e f_e * P_e
-            # (reference moe_utils.py:14)
             with torch.no_grad():
                 mask = torch.zeros_like(logits).scatter_(
                     1, idx, 1.0)
+            P_tok = torch.softmax(logits, dim=-1)
+            if self.aux_loss_type == "seq_aux_loss" and seq_len is not None \
+                    and n % seq_len == 0:
+                # per-sequence balance, averaged over the batch
+                # (reference moe_utils.py:62 sequence_load_balancing)
+                b = n // seq_len
+                f = mask.view(seq_len, b, E).mean(0) * E / self.topk
+                P = P_tok.view(seq_len, b, E).mean(0)
+                aux = aux + self.aux_loss_coeff * (f * P).sum(-1).mean()
+            else:
+                # load-balancing loss: E * sum_e f_e * P_e
+                # (reference moe_utils.py:14)
                 f = mask.mean(0) * E / self.topk
-            P = torch.softmax(logits, dim=-1).mean(0)
-            aux = aux + self.aux_loss_coeff * (f * P).sum()
+                P = P_tok.mean(0)
+                aux = aux + self.aux_loss_coeff * (f * P).sum()
         if self.z_loss_coeff > 0 and self.training:
             aux = aux + self.z_loss_coeff * \
                 torch.logsumexp(logits, dim=-1).square().mean()

@@ -175,3 +175,50 @@ def test_moe_etp_vs_baseline(world, tp, ep):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+def test_router_group_limited_topk():
+    """Experts must come only from the selected groups."""
+    cfg = make_cfg({"model": {"moe_router_num_groups": 2,
+                              "moe_router_group_topk": 1}})
+    from hetu_galvatron_amd.runtime.moe.router import TopKRouter
+    torch.manual_seed(0)
+    r = TopKRouter(cfg.model)
+    x = torch.randn(64, cfg.model.hidden_size)
+    probs, idx, _ = r(x)
+    E, G = cfg.model.num_experts, 2
+    per_g = E // G
+    groups = idx // per_g
+    # with group_topk=1 every token's experts are in ONE group
+    assert (groups == groups[:, :1]).all()
+
+
+def test_router_seq_aux_and_capacity():
+    from hetu_galvatron_amd.runtime.moe.router import TopKRouter
+    cfg = make_cfg({"model": {"moe_aux_loss_type": "seq_aux_loss"}})
+    torch.manual_seed(1)
+    r = TopKRouter(cfg.model)
+    r.train()
+    x = torch.randn(8 * 4, cfg.model.hidden_size)
+    _, _, aux_seq = r(x, seq_len=8)
+    assert float(aux_seq) > 0
+    # capacity: tight factor zeroes some probs
+    cfg2 = make_cfg({"model": {"moe_expert_capacity_factor": 0.5}})
+    r2 = TopKRouter(cfg2.model)
+    torch.manual_seed(2)
+    probs, idx, _ = r2(torch.randn(64, cfg2.model.hidden_size))
+    assert (probs == 0).any()          # overflow tokens dropped
+    assert (probs.sum(-1) > 0).any()   # within-capacity tokens kept
+
+
+def test_aux_loss_tracker():
+    from hetu_galvatron_amd.runtime.moe import tracker
+    tracker.clear()
+    tracker.save_aux_loss("load_balancing", 0, torch.tensor(0.5))
+    tracker.save_aux_loss("load_balancing", 0, torch.tensor(1.5))
+    tracker.save_aux_loss("z_loss", 1, 2.0)
+    got = tracker.reduce_and_get()
+    assert abs(got["load_balancing/layer_0"] - 1.0) < 1e-9
+    assert abs(got["z_loss/layer_1"] - 2.0) < 1e-9
+    tracker.clear()
+    assert tracker.reduce_and_get() == {}
