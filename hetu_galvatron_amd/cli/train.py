@@ -92,6 +92,13 @@ def main(argv=None):
             print(f"[rerun] persistent fault at iter {i}: exiting "
                   f"{rsm.exit_code}", file=sys.stderr)
             sys.exit(rsm.exit_code)
+        ci = cfg.train.check_weight_consistency_interval
+        if ci and (i + 1) % ci == 0:
+            from ..utils.consistency import check_param_consistency
+            bad = check_param_consistency(model.stage_model)
+            if bad:
+                raise RuntimeError(
+                    f"replicated params diverged at iter {i}: {bad[:8]}")
         if (cfg.ckpt.save and cfg.ckpt.save_interval
                 and (i + 1) % cfg.ckpt.save_interval == 0):
             from ..runtime.checkpoint import save_distributed_checkpoint

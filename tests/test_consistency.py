@@ -1,0 +1,41 @@
+"""Replicated-param consistency checker (reference test_mode realtime
+weight checks)."""
+import pytest
+import torch
+
+from hetu_galvatron_amd.config import HybridParallelPlan, load_config
+
+
+def _worker(rank, world, perturb):
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.utils.consistency import check_param_consistency
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "train": {"global_train_batch_size": 4, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.uniform(num_layers=2, world_size=2, pp=1,
+                                      tp=1, dp_type="ddp", global_bsz=4)
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg, plan)
+    clean = check_param_consistency(model.stage_model)
+    if perturb and rank == 1:
+        with torch.no_grad():
+            next(model.stage_model.blocks[1].inner.parameters()).add_(1.0)
+    dirty = check_param_consistency(model.stage_model)
+    return {"clean": clean, "dirty": dirty}
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("perturb", [False, True])
+def test_consistency_detects_divergence(perturb):
+    from tests.utils import run_distributed
+    res = run_distributed(_worker, world_size=2, args=(perturb,))
+    for r in res:
+        assert r["clean"] == []
+        if perturb:
+            assert r["dirty"], "divergence not detected"
+        else:
+            assert r["dirty"] == []
