@@ -25,6 +25,10 @@ void dynamic_programming_core(int layer_num, int max_mem, int strategy_num,
   const auto inter = inter_a.unchecked<3>();     // [L, S, S] (si -> s)
   auto f = f_a.mutable_unchecked<2>();           // [M, S]
   auto mark = mark_a.mutable_unchecked<3>();     // [L, M, S]
+  // accessors above hold raw buffers; the O(L*M*S^2) loops below never
+  // touch the Python API, so drop the GIL and let the engine's
+  // parallel_search thread pool overlap DP tasks
+  py::gil_scoped_release release;
   const double INF = std::numeric_limits<double>::infinity();
 
   std::vector<double> prev(max_mem * strategy_num, 0.0);

@@ -334,14 +334,25 @@ class SearchEngine:
         assert self.layer_profile is not None and self.hw is not None, \
             "load_profiles (or pass profiles) before searching"
         best: Optional[SearchResult] = None
-        for bsz in self._bsz_candidates():
-            for pp in self._pp_candidates():
-                for chunks in self._chunk_candidates(bsz, pp):
-                    r = self.search_task(bsz, chunks, pp)
-                    if r is not None:
-                        self.results.append(r)
-                        if best is None or r.throughput > best.throughput:
-                            best = r
+        tasks = [(bsz, chunks, pp)
+                 for bsz in self._bsz_candidates()
+                 for pp in self._pp_candidates()
+                 for chunks in self._chunk_candidates(bsz, pp)]
+        if getattr(self.args, "parallel_search", False) and len(tasks) > 1:
+            # the C++ DP core drops the GIL, so a thread pool overlaps the
+            # per-task O(L*M*S^2) solves (reference: thread-parallel task
+            # grid, search_engine.py:579-604)
+            from concurrent.futures import ThreadPoolExecutor
+            with ThreadPoolExecutor(max_workers=min(8, len(tasks))) as exe:
+                results = list(exe.map(lambda t: self.search_task(*t),
+                                       tasks))
+        else:
+            results = [self.search_task(*t) for t in tasks]
+        for r in results:
+            if r is not None:
+                self.results.append(r)
+                if best is None or r.throughput > best.throughput:
+                    best = r
         if best is not None:
             path = output_path or self.args.output_config_path
             if path:

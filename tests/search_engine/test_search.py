@@ -239,3 +239,14 @@ def test_profiled_json_roundtrip(tmp_path):
     assert 8 in eng.hw.allgather_latency
     best = eng.parallelism_optimization(None)
     assert best is not None and best.throughput > 0
+
+
+def test_parallel_search_matches_sequential(tmp_path):
+    """Thread-parallel task grid (GIL-free C++ DP core) == sequential."""
+    seq = make_engine(tmp_path).parallelism_optimization(None)
+    eng = make_engine(tmp_path)
+    eng.cfg.search.parallel_search = True
+    eng.args.parallel_search = True
+    par = eng.parallelism_optimization(None)
+    assert abs(par.throughput - seq.throughput) < 1e-9
+    assert par.pp_deg == seq.pp_deg and par.chunks == seq.chunks
