@@ -88,3 +88,55 @@ def test_distributed_save_resume_single(tmp_path):
         resumed.append(st.loss)
     for a, b in zip(cont, resumed):
         assert abs(a - b) < 1e-6, (cont, resumed)
+
+
+def _resume_worker(rank, world, tmp_dir):
+    import torch
+    from hetu_galvatron_amd.config import HybridParallelPlan
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import (
+        GalvatronModel, get_optimizer_and_param_scheduler, get_train_iterator)
+    from hetu_galvatron_amd.runtime.checkpoint import (
+        load_distributed_checkpoint, save_distributed_checkpoint)
+
+    cfg = tiny_cfg()
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.uniform(num_layers=2, world_size=2, pp=1,
+                                      tp=1, dp_type="zero2", global_bsz=4)
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg, plan)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    batches = [next(it) for _ in range(4)]
+    for i in range(2):
+        opt.zero_grad(); model.forward_backward(batches[i])
+        opt.step(); sched.step()
+    save_distributed_checkpoint(model, opt, sched, cfg, 2, tmp_dir)
+    cont = []
+    for i in range(2, 4):
+        opt.zero_grad(); st = model.forward_backward(batches[i])
+        opt.step(); sched.step()
+        cont.append(model.global_loss(st))
+
+    torch.manual_seed(999 + rank)
+    model2 = GalvatronModel(cfg, plan)
+    opt2, sched2 = get_optimizer_and_param_scheduler(model2.stage_model, cfg)
+    assert load_distributed_checkpoint(model2, opt2, sched2, cfg,
+                                       tmp_dir) == 2
+    resumed = []
+    for i in range(2, 4):
+        opt2.zero_grad(); st = model2.forward_backward(batches[i])
+        opt2.step(); sched2.step()
+        resumed.append(model2.global_loss(st))
+    return {"cont": cont, "resumed": resumed}
+
+
+@pytest.mark.distributed
+def test_distributed_save_resume_world2(tmp_path):
+    """Per-rank zero2 shards save/restore exactly under world 2."""
+    from tests.utils import run_distributed
+    res = run_distributed(_resume_worker, world_size=2,
+                          args=(str(tmp_path),))
+    for r in res:
+        for a, b in zip(r["cont"], r["resumed"]):
+            assert abs(a - b) < 1e-6, r
