@@ -153,21 +153,28 @@ def flash_attention(q, k, v, causal: bool = True,
     return (o, lse) if return_lse else o
 
 
-def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None):
-    """No-autograd forward returning (o, lse) — building block for ring CP."""
+def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None,
+                             bias=None):
+    """No-autograd forward returning (o, lse) — building block for ring CP.
+    bias [hq, sq, skv]: t5 relative bias — eager fp32 path (a bias input
+    on the native flash kernel is a v2 item)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+    if bias is None and use_native(q) and q.dtype == torch.bfloat16 \
+            and q.shape[-1] in (64, 128):
         return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
                                              v.contiguous(), causal, scale)
-    return ref.attention_fwd(q, k, v, causal, scale)
+    return ref.attention_fwd(q, k, v, causal, scale, bias)
 
 
-def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True, softmax_scale=None):
+def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True,
+                             softmax_scale=None, bias=None):
+    """Without bias: (dq, dk, dv); with bias also dbias (batch-summed)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+    if bias is None and use_native(q) and q.dtype == torch.bfloat16 \
+            and q.shape[-1] in (64, 128):
         return get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
                                              causal, scale)
-    return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)
+    return ref.attention_bwd(do, q, k, v, o, lse, causal, scale, bias)
 
 
 def _gg_tiles(counts, device):

@@ -124,10 +124,12 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
 
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                   causal: bool = True, softmax_scale: Optional[float] = None,
+                  bias: Optional[torch.Tensor] = None,
                   ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Reference attention WITH log-sum-exp output (needed for ring-CP merges).
 
     q: [b, s, hq, d]; k,v: [b, s, hkv, d] (GQA: hq % hkv == 0).
+    bias: optional [hq, sq, skv] additive scores bias (t5 relative bias).
     Returns o [b, s, hq, d], lse [b, hq, s] (natural log).
     """
     b, sq, hq, d = q.shape
@@ -141,6 +143,8 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     kf = kx.permute(0, 2, 1, 3).float()
     vf = vx.permute(0, 2, 1, 3).float()
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b,h,sq,skv]
+    if bias is not None:
+        scores = scores + bias.unsqueeze(0).float()
     if causal:
         # causal with bottom-right alignment when sq != skv
         i = torch.arange(sq, device=q.device)[:, None]
@@ -154,15 +158,66 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o.permute(0, 2, 1, 3).to(q.dtype), lse
 
 
-def attention_bwd(do, q, k, v, o, lse, causal=True, softmax_scale=None):
-    """Reference backward via autograd on the fp32 recompute."""
-    q32 = q.detach().float().requires_grad_(True)
-    k32 = k.detach().float().requires_grad_(True)
-    v32 = v.detach().float().requires_grad_(True)
-    with torch.enable_grad():
-        o2, _ = attention_fwd(q32, k32, v32, causal, softmax_scale)
-    gq, gk, gv = torch.autograd.grad(o2, (q32, k32, v32), do.float())
-    return gq.to(q.dtype), gk.to(k.dtype), gv.to(v.dtype)
+def attention_bwd(do, q, k, v, o, lse, causal=True, softmax_scale=None,
+                  bias=None):
+    """Reference backward, flash-style against the PASSED o/lse.
+
+    Using the caller's (global) lse/o matters for ring-CP: each per-block
+    backward must normalize with the GLOBAL softmax so the block gradients
+    are true partials that sum across ring steps (the native kernel does
+    exactly this; an autograd recompute would silently use block-LOCAL
+    softmax).  Falls back to autograd recompute when o/lse is None.
+    With a bias, also returns dbias [hq, sq, skv] (summed over batch).
+    """
+    if o is None or lse is None:
+        q32 = q.detach().float().requires_grad_(True)
+        k32 = k.detach().float().requires_grad_(True)
+        v32 = v.detach().float().requires_grad_(True)
+        b32 = bias.detach().float().requires_grad_(True) \
+            if bias is not None else None
+        with torch.enable_grad():
+            o2, _ = attention_fwd(q32, k32, v32, causal, softmax_scale, b32)
+        if b32 is None:
+            gq, gk, gv = torch.autograd.grad(o2, (q32, k32, v32), do.float())
+            return gq.to(q.dtype), gk.to(k.dtype), gv.to(v.dtype)
+        gq, gk, gv, gb = torch.autograd.grad(o2, (q32, k32, v32, b32),
+                                             do.float())
+        return gq.to(q.dtype), gk.to(k.dtype), gv.to(v.dtype), gb
+
+    b, sq, hq, d = q.shape
+    skv, hkv = k.shape[1], k.shape[2]
+    scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(d)
+    rep = hq // hkv
+    kx = k.repeat_interleave(rep, dim=2) if rep > 1 else k
+    vx = v.repeat_interleave(rep, dim=2) if rep > 1 else v
+    qf = q.permute(0, 2, 1, 3).float()
+    kf = kx.permute(0, 2, 1, 3).float()
+    vf = vx.permute(0, 2, 1, 3).float()
+    dof = do.permute(0, 2, 1, 3).float()
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b,h,sq,skv]
+    if bias is not None:
+        s = s + bias.unsqueeze(0).float()
+    if causal:
+        i = torch.arange(sq, device=q.device)[:, None]
+        j = torch.arange(skv, device=q.device)[None, :]
+        s = s.masked_fill(j > (i + (skv - sq)), float("-inf"))
+    p = torch.exp(s - lse.float().unsqueeze(-1))
+    p = torch.nan_to_num(p)
+    dv_h = torch.matmul(p.transpose(-1, -2), dof)       # [b,h,skv,d]
+    dp = torch.matmul(dof, vf.transpose(-1, -2))        # [b,h,sq,skv]
+    di = (do.float() * o.float()).sum(-1).permute(0, 2, 1)  # [b,h,sq]
+    dsr = p * (dp - di.unsqueeze(-1))                   # dL/ds_raw
+    dq = torch.matmul(dsr, kf) * scale
+    dk_h = torch.matmul(dsr.transpose(-1, -2), qf) * scale
+    if rep > 1:
+        dk_h = dk_h.view(b, hkv, rep, skv, d).sum(2)
+        dv_h = dv_h.view(b, hkv, rep, skv, d).sum(2)
+    gq = dq.permute(0, 2, 1, 3).to(q.dtype)
+    gk = dk_h.permute(0, 2, 1, 3).to(k.dtype)
+    gv = dv_h.permute(0, 2, 1, 3).to(v.dtype)
+    if bias is None:
+        return gq, gk, gv
+    return gq, gk, gv, dsr.sum(0)
 
 
 def vocab_ce_stats(logits: torch.Tensor, target: torch.Tensor,

@@ -41,6 +41,8 @@ def eager_bias_attention(q, k, v, bias, causal: bool, scale: float
     item).  q/k/v: [b, s, h, d]."""
     b, sq, h, d = q.shape
     skv = k.shape[1]
+    if scale is None:
+        scale = d ** -0.5
     if k.shape[2] != h:
         rep = h // k.shape[2]
         k = k.repeat_interleave(rep, dim=2)
@@ -160,19 +162,35 @@ def _merge_attn_out(o: Optional[torch.Tensor], lse: Optional[torch.Tensor],
     return o, new_lse
 
 
-def zigzag_ring_flash_attn_fwd(q, k, v, comm: RingComm, softmax_scale: float
+def _zigzag_cols(S: int, n: int, j: int, device) -> torch.Tensor:
+    """Natural col indices of rank j's zigzag kv chunks (j, 2n-1-j)."""
+    cs = S // (2 * n)
+    return torch.cat([torch.arange(j * cs, (j + 1) * cs, device=device),
+                      torch.arange((2 * n - 1 - j) * cs,
+                                   (2 * n - j) * cs, device=device)])
+
+
+def zigzag_ring_flash_attn_fwd(q, k, v, comm: RingComm, softmax_scale: float,
+                               causal: bool = True, bias=None
                                ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Forward ring pass over zigzag-sharded sequence.
 
-    Chunk math: rank r holds q/kv chunks (r, 2n-1-r) of 2n.  KV from rank j:
+    Chunk math (causal): rank r holds q/kv chunks (r, 2n-1-r) of 2n.  KV
+    from rank j:
       j == r : local causal attention
       j <  r : BOTH q chunks attend the FIRST kv half, un-masked
       j >  r : only the SECOND q half attends the FULL kv, un-masked
+    bias [h, s_local, S_global]: t5 relative bias rows for THIS rank's
+    packed zigzag q rows over the natural global kv axis; per-step col
+    slices follow the chunk math.  causal=False (t5 encoder): every step
+    attends rank j's full kv un-masked.
     (reference: attention_impl.py:564-653)
     """
     n, r = comm.size, comm.rank
     b, s_local, hq, d = q.shape
+    S = s_local * n
     half = s_local // 2
+    cs = S // (2 * n)
     o_acc: Optional[torch.Tensor] = None
     lse_acc: Optional[torch.Tensor] = None
     o2_acc: Optional[torch.Tensor] = None  # second-half-only accumulator
@@ -184,19 +202,32 @@ def zigzag_ring_flash_attn_fwd(q, k, v, comm: RingComm, softmax_scale: float
             v_nxt = comm.send_recv(v_cur)
             comm.commit()
         j = (r - step) % n
-        if step == 0:
+
+        def bcols(jj):
+            return None if bias is None else \
+                bias[:, :, _zigzag_cols(S, n, jj, q.device)]
+
+        if not causal:
+            o_i, lse_i = flash_attention_fwd_only(
+                q, k_cur, v_cur, causal=False, softmax_scale=softmax_scale,
+                bias=bcols(j))
+            o_acc, lse_acc = _merge_attn_out(o_acc, lse_acc, o_i, lse_i)
+        elif step == 0:
             o_i, lse_i = flash_attention_fwd_only(q, k_cur, v_cur, causal=True,
-                                                  softmax_scale=softmax_scale)
+                                                  softmax_scale=softmax_scale,
+                                                  bias=bcols(r))
             o_acc, lse_acc = _merge_attn_out(o_acc, lse_acc, o_i, lse_i)
         elif j < r:
+            bb = None if bias is None else bias[:, :, j * cs:(j + 1) * cs]
             o_i, lse_i = flash_attention_fwd_only(
                 q, k_cur[:, :half], v_cur[:, :half], causal=False,
-                softmax_scale=softmax_scale)
+                softmax_scale=softmax_scale, bias=bb)
             o_acc, lse_acc = _merge_attn_out(o_acc, lse_acc, o_i, lse_i)
         else:
+            bb = None if bias is None else bcols(j)[:, half:]
             o_i, lse_i = flash_attention_fwd_only(
                 q[:, half:], k_cur, v_cur, causal=False,
-                softmax_scale=softmax_scale)
+                softmax_scale=softmax_scale, bias=bb)
             o2_acc, lse2_acc = _merge_attn_out(o2_acc, lse2_acc, o_i, lse_i)
         if step + 1 < n:
             comm.wait()
@@ -211,14 +242,21 @@ def zigzag_ring_flash_attn_fwd(q, k, v, comm: RingComm, softmax_scale: float
 
 
 def zigzag_ring_flash_attn_bwd(do, q, k, v, o, lse, comm: RingComm,
-                               softmax_scale: float):
+                               softmax_scale: float, causal: bool = True,
+                               bias=None):
     """Backward ring pass: dq accumulates locally; (k, v, dk, dv) rotate
-    together n steps and arrive home fully accumulated
-    (reference: attention_impl.py:654-782)."""
+    together n steps and arrive home fully accumulated.  With a bias,
+    also accumulates dbias [h, s_local, S_global] on this rank's q rows
+    (the bias table's cross-cp grad sum rides the sdp reduction, which
+    spans cp) (reference: attention_impl.py:654-782)."""
     n, r = comm.size, comm.rank
     b, s_local, hq, d = q.shape
+    S = s_local * n
     half = s_local // 2
+    cs = S // (2 * n)
     dq = torch.zeros_like(q, dtype=torch.float32)
+    dbias = None if bias is None else torch.zeros_like(bias,
+                                                       dtype=torch.float32)
     k_cur, v_cur = k, v
     dk_cur = torch.zeros_like(k, dtype=torch.float32)
     dv_cur = torch.zeros_like(v, dtype=torch.float32)
@@ -228,27 +266,49 @@ def zigzag_ring_flash_attn_bwd(do, q, k, v, o, lse, comm: RingComm,
     q_half = q[:, half:].contiguous()
     for step in range(n):
         j = (r - step) % n
-        if step == 0:
-            dq_i, dk_i, dv_i = flash_attention_bwd_only(
+        cols = _zigzag_cols(S, n, j, q.device) if bias is not None else None
+        if not causal:
+            res = flash_attention_bwd_only(
+                do, q, k_cur, v_cur, o, lse, causal=False,
+                softmax_scale=softmax_scale,
+                bias=None if bias is None else bias[:, :, cols])
+            dq += res[0].float()
+            dk_cur += res[1].float()
+            dv_cur += res[2].float()
+            if bias is not None:
+                dbias[:, :, cols] += res[3]
+        elif step == 0:
+            res = flash_attention_bwd_only(
                 do, q, k_cur, v_cur, o, lse, causal=True,
-                softmax_scale=softmax_scale)
-            dq += dq_i.float()
-            dk_cur += dk_i.float()
-            dv_cur += dv_i.float()
+                softmax_scale=softmax_scale,
+                bias=None if bias is None else bias[:, :, cols])
+            dq += res[0].float()
+            dk_cur += res[1].float()
+            dv_cur += res[2].float()
+            if bias is not None:
+                dbias[:, :, cols] += res[3]
         elif j < r:
-            dq_i, dk_i, dv_i = flash_attention_bwd_only(
+            res = flash_attention_bwd_only(
                 do, q, k_cur[:, :half].contiguous(), v_cur[:, :half].contiguous(),
-                o, lse, causal=False, softmax_scale=softmax_scale)
-            dq += dq_i.float()
-            dk_cur[:, :half] += dk_i.float()
-            dv_cur[:, :half] += dv_i.float()
+                o, lse, causal=False, softmax_scale=softmax_scale,
+                bias=None if bias is None else
+                bias[:, :, j * cs:(j + 1) * cs])
+            dq += res[0].float()
+            dk_cur[:, :half] += res[1].float()
+            dv_cur[:, :half] += res[2].float()
+            if bias is not None:
+                dbias[:, :, j * cs:(j + 1) * cs] += res[3]
         else:
-            dq_i, dk_i, dv_i = flash_attention_bwd_only(
+            res = flash_attention_bwd_only(
                 do_half, q_half, k_cur, v_cur, o_half, lse_half, causal=False,
-                softmax_scale=softmax_scale)
-            dq[:, half:] += dq_i.float()
-            dk_cur += dk_i.float()
-            dv_cur += dv_i.float()
+                softmax_scale=softmax_scale,
+                bias=None if bias is None else bias[:, half:][:, :, cols])
+            dq[:, half:] += res[0].float()
+            dk_cur += res[1].float()
+            dv_cur += res[2].float()
+            if bias is not None:
+                db = dbias[:, half:]
+                db[:, :, cols] += res[3]
         # rotate kv together with accumulated dkv (n rotations total -> home)
         k_nxt = comm.send_recv(k_cur)
         v_nxt = comm.send_recv(v_cur)
@@ -257,30 +317,42 @@ def zigzag_ring_flash_attn_bwd(do, q, k, v, o, lse, comm: RingComm,
         comm.commit()
         comm.wait()
         k_cur, v_cur, dk_cur, dv_cur = k_nxt, v_nxt, dk_nxt, dv_nxt
-    return dq.to(q.dtype), dk_cur.to(k.dtype), dv_cur.to(v.dtype)
+    out = (dq.to(q.dtype), dk_cur.to(k.dtype), dv_cur.to(v.dtype))
+    return out + ((dbias,) if bias is not None else ())
 
 
 class ZigzagRingFlashAttnFunc(torch.autograd.Function):
     """reference: attention_impl.py:785 ZigZagRingFlashAttnFunc."""
 
     @staticmethod
-    def forward(ctx, q, k, v, cp_group, softmax_scale):
+    def forward(ctx, q, k, v, bias, cp_group, softmax_scale, causal):
         scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
         comm = RingComm(cp_group)
+        bias_c = bias.contiguous() if bias is not None else None
         o, lse = zigzag_ring_flash_attn_fwd(q.contiguous(), k.contiguous(),
-                                            v.contiguous(), comm, scale)
-        ctx.save_for_backward(q, k, v, o, lse)
+                                            v.contiguous(), comm, scale,
+                                            causal=causal, bias=bias_c)
+        ctx.save_for_backward(q, k, v, o, lse,
+                              *([] if bias is None else [bias_c]))
         ctx.cp_group = cp_group
         ctx.scale = scale
+        ctx.causal = causal
+        ctx.has_bias = bias is not None
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, o, lse = ctx.saved_tensors
+        if ctx.has_bias:
+            q, k, v, o, lse, bias = ctx.saved_tensors
+        else:
+            q, k, v, o, lse = ctx.saved_tensors
+            bias = None
         comm = RingComm(ctx.cp_group)
-        dq, dk, dv = zigzag_ring_flash_attn_bwd(
-            do.contiguous(), q, k, v, o, lse, comm, ctx.scale)
-        return dq, dk, dv, None, None
+        res = zigzag_ring_flash_attn_bwd(
+            do.contiguous(), q, k, v, o, lse, comm, ctx.scale,
+            causal=ctx.causal, bias=bias)
+        dbias = res[3].to(bias.dtype) if bias is not None else None
+        return res[0], res[1], res[2], dbias, None, None, None
 
 
 class ZigzagRingAttention(torch.nn.Module):
@@ -291,8 +363,15 @@ class ZigzagRingAttention(torch.nn.Module):
         super().__init__()
         self.cp_group = cp_group
 
-    def forward(self, q, k, v, causal=True, softmax_scale=None):
-        assert causal, "ring CP implements causal attention"
+    def forward(self, q, k, v, causal=True, softmax_scale=None,
+                attn_bias=None):
+        """attn_bias [h, s_local, S_global]: this rank's packed zigzag q
+        rows over the natural global kv axis (t5 relative bias)."""
         if group_size(self.cp_group) == 1:
+            if attn_bias is not None:
+                return eager_bias_attention(q, k, v, attn_bias, causal,
+                                            softmax_scale)
             return local_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
-        return ZigzagRingFlashAttnFunc.apply(q, k, v, self.cp_group, softmax_scale)
+        return ZigzagRingFlashAttnFunc.apply(q, k, v, attn_bias,
+                                             self.cp_group, softmax_scale,
+                                             causal)
