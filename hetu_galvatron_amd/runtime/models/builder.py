@@ -225,7 +225,6 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         lg = layer_groups[i]
         if is_encdec:
             if i < n_enc:
-                assert lg.strategy.cp == 1, "t5 encoder: cp is a v2 item"
                 dec = GalvatronEncoderLayer(margs, lg, layer_idx=i,
                                             dtype=dtype)
                 kind = "encoder"

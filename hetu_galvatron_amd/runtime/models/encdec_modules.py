@@ -27,6 +27,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ...config.schema import ModelArgs
+from ..redistribute import natural_rows
 from ..tensor_parallel import group_size
 from ..tensor_parallel.mappings import gather_from_sequence_parallel_region
 from ..transformer import MLP, SelfAttention, build_norm
@@ -40,8 +41,6 @@ def _build_rel_bias(margs: ModelArgs, groups, bidirectional: bool, dtype):
     HF layer-0-shared layout note)."""
     if margs.position_embedding_type != "relative":
         return None
-    s = groups.strategy
-    assert s.cp == 1, "relative-position bias under ring-CP is a v2 item"
     rb = RelativePositionBias(
         margs.relative_attention_num_buckets,
         margs.relative_attention_max_distance,
@@ -62,14 +61,22 @@ def _rank_bias(rel_bias, attention, groups, hidden):
     if rel_bias is None:
         return None
     s = groups.strategy
+    c = groups.coord_of(_my_rank())
     if s.use_ulysses:
+        assert s.cp == 1, "ulysses + ring-CP + relative bias is a v2 item"
         S = hidden.shape[0] * s.tp_sp
         return rel_bias(S, S, hidden.device)
     tp = group_size(groups.tp_group)
-    S = hidden.shape[0] * tp
-    c = groups.coord_of(_my_rank())
+    S = hidden.shape[0] * tp * s.cp
     hl = attention.heads_local
-    return rel_bias(S, S, hidden.device, c.tp_idx * hl, (c.tp_idx + 1) * hl)
+    full = rel_bias(S, S, hidden.device, c.tp_idx * hl, (c.tp_idx + 1) * hl)
+    if s.cp > 1:
+        # ring path: rows = this cp rank's packed zigzag q rows; the kv
+        # (column) axis stays natural/global (attention_impl slices it
+        # per ring step)
+        rows = natural_rows(S, s.cp, tp, c.cp_idx, c.tp_idx, hidden.device)
+        full = full[:, rows, :]
+    return full
 
 
 class GalvatronEncoderLayer(nn.Module):
@@ -149,6 +156,21 @@ class GalvatronEncDecBridge(nn.Module):
         group = self.groups.tsp_cp_group
         if group is not None and group_size(group) > 1:
             mem = gather_from_sequence_parallel_region(mem, group)
+            if s.cp > 1:
+                # gathered order is group-rank packed (each rank's zigzag
+                # pair); permute back to natural sequence order
+                import torch
+                S = mem.shape[0]
+                sl = S // group_size(group)
+                tsp = self.groups.seq_shard_degree // s.cp
+                perm = torch.empty(S, dtype=torch.long, device=mem.device)
+                for pi, gr in enumerate(group.ranks):
+                    c = self.groups.coord_of(gr)
+                    rows = natural_rows(S, s.cp, max(tsp, 1), c.cp_idx,
+                                        c.tp_idx, mem.device)
+                    perm[rows] = torch.arange(pi * sl, (pi + 1) * sl,
+                                              device=mem.device)
+                mem = mem[perm]
         ctx["encoder_memory"] = mem            # [S_enc, b_loc, h]
         return self.dec_embedding(ctx)
 
@@ -164,7 +186,8 @@ class GalvatronDecoderLayerX(nn.Module):
         self.layer_idx = layer_idx
         s = groups.strategy
         self.strategy = s
-        assert s.cp == 1, "t5 decoder layers: ring-CP is a v2 item"
+        assert s.cp == 1 or not s.use_ulysses, \
+            "t5: ulysses composed with ring-CP is a v2 item"
         seq_par = not s.use_ulysses
         self.input_norm = build_norm(margs.normalization, margs.hidden_size,
                                      margs.norm_epsilon, dtype)

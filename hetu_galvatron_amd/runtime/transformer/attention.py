@@ -94,16 +94,17 @@ class SelfAttention(nn.Module):
         q = q.permute(1, 0, 2, 3).contiguous()
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()
-        if attn_bias is not None and isinstance(self.core_attention,
-                                                DistributedAttention):
-            # ulysses: pass the full-head bias; the a2a re-shards heads and
-            # DistributedAttention slices the matching chunk
+        if attn_bias is not None and isinstance(
+                self.core_attention, (DistributedAttention,
+                                      ZigzagRingAttention)):
+            # ulysses: full-head bias, the a2a re-shards heads and
+            # DistributedAttention slices the matching chunk.
+            # ring-CP: bias rows are this rank's packed zigzag q rows over
+            # the natural global kv axis (the layer builds it that way)
             o = self.core_attention(q, k, v, causal=self.causal,
                                     softmax_scale=self.softmax_scale,
                                     attn_bias=attn_bias)
         elif attn_bias is not None:
-            assert self.core_attention is None, \
-                "attn_bias (t5 relative bias) + ring-CP is a v2 item"
             o = eager_bias_attention(q, k, v, attn_bias, self.causal,
                                      self.softmax_scale)
         elif self.core_attention is not None:

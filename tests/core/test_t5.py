@@ -150,3 +150,23 @@ def test_t5_ulysses_vs_baseline():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_cp2_vs_baseline():
+    """Ring-CP on both T5 stacks: bidirectional ring + bias on the
+    encoder, causal ring + bias on the decoder self-attention,
+    cross-attention over zigzag-sharded queries vs the full memory."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=2, pp=1, tp=1, cp=2, dp_type="ddp",
+        global_bsz=4, chunks=1)
+    plan.vcp = 2  # vocab layers share the cp split (uniform dp domain)
+    res = run_distributed(_t5_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
