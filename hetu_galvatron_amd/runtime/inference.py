@@ -285,6 +285,8 @@ class GalvatronGenerator:
         dev = self._dev
         b = first_ids.shape[0]
         ent = self._graphs.get(b)
+        if ent is not None and ent[4].shape[0] != self.max_seq:
+            ent = None  # max_seq changed: stale capture
         if ent is None:
             ids_buf = first_ids.unsqueeze(1).clone()          # [b, 1]
             pos_i64 = torch.tensor([cache.cur_len], dtype=torch.long,

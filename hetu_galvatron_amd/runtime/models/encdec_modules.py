@@ -71,10 +71,11 @@ def _rank_bias(rel_bias, attention, groups, hidden):
     hl = attention.heads_local
     full = rel_bias(S, S, hidden.device, c.tp_idx * hl, (c.tp_idx + 1) * hl)
     if s.cp > 1:
-        # ring path: rows = this cp rank's packed zigzag q rows; the kv
-        # (column) axis stays natural/global (attention_impl slices it
-        # per ring step)
-        rows = natural_rows(S, s.cp, tp, c.cp_idx, c.tp_idx, hidden.device)
+        # ring path: rows = this cp rank's packed zigzag q rows — the
+        # FULL pair (megatron-SP re-gathers the tp seq split before
+        # attention); the kv (column) axis stays natural/global
+        # (attention_impl slices it per ring step)
+        rows = natural_rows(S, s.cp, 1, c.cp_idx, 0, hidden.device)
         full = full[:, rows, :]
     return full
 

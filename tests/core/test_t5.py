@@ -170,3 +170,22 @@ def test_t5_cp2_vs_baseline():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_world4_tp2_cp2():
+    """tp=2 x cp=2 composed on T5: head-sliced bias rows over the full
+    cp-local zigzag pair (the SP allgather precedes attention)."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=4, pp=1, tp=2, cp=2, dp_type="ddp",
+        global_bsz=4, chunks=1, vtp=2)
+    plan.vcp = 2
+    res = run_distributed(_t5_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
