@@ -141,6 +141,32 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         position_embedding_type="rope", rope_theta=10000.0,
         num_experts=4, moe_router_topk=2,
     ),
+    # DeepSeek-V2-style MoE: shared expert + sigmoid scores + aux-free
+    # bias + group-limited (node-limited) routing
+    "deepseek-v2-lite": dict(
+        model_type="moe-llama", hidden_size=2048, num_hidden_layers=27,
+        num_attention_heads=16, num_key_value_heads=16, kv_channels=128,
+        ffn_hidden_size=10944, moe_ffn_hidden_size=1408,
+        vocab_size=102400, max_position_embeddings=4096, seq_length=4096,
+        hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-6,
+        position_embedding_type="rope", rope_theta=10000.0,
+        num_experts=64, moe_router_topk=6,
+        moe_router_score_function="sigmoid", moe_aux_loss_free=True,
+        moe_router_num_groups=8, moe_router_group_topk=3,
+        moe_shared_expert_intermediate_size=2816,
+    ),
+    "tiny-moe-shared": dict(
+        model_type="moe-llama", hidden_size=128, num_hidden_layers=2,
+        num_attention_heads=2, num_key_value_heads=2, kv_channels=64,
+        ffn_hidden_size=256,
+        vocab_size=512, max_position_embeddings=256, seq_length=128,
+        hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-5,
+        position_embedding_type="rope", rope_theta=10000.0,
+        num_experts=4, moe_router_topk=2,
+        moe_router_score_function="sigmoid",
+        moe_router_num_groups=2, moe_router_group_topk=1,
+        moe_shared_expert_intermediate_size=64,
+    ),
 }
 
 _HF_FIELD_MAP = {

@@ -96,6 +96,77 @@ def canonical_to_hf_llama(can: Dict[str, torch.Tensor],
     return out
 
 
+def hf_mixtral_to_canonical(hf: Dict[str, torch.Tensor],
+                            margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    """HF MixtralForCausalLM keys -> canonical (reference h2g mixtral
+    path, moe_adapter.py:37-349).  HF per-expert w1(gate)/w3(up)/w2(down)
+    [F,h]/[F,h]/[h,F] map to grouped w1 [E, h, 2F] ([gate;up] stacked on
+    the out dim, transposed) and w2 [E, F, h]."""
+    out: Dict[str, torch.Tensor] = {}
+    out["embedding.word_embeddings.weight"] = hf["model.embed_tokens.weight"]
+    E = margs.num_experts
+    for i in range(margs.num_hidden_layers):
+        p = f"model.layers.{i}."
+        c = f"decoder.{i}."
+        out[c + "input_norm.weight"] = hf[p + "input_layernorm.weight"]
+        out[c + "post_attn_norm.weight"] = \
+            hf[p + "post_attention_layernorm.weight"]
+        out[c + "attention.linear_qkv.weight"] = fuse_qkv(
+            hf[p + "self_attn.q_proj.weight"],
+            hf[p + "self_attn.k_proj.weight"],
+            hf[p + "self_attn.v_proj.weight"], margs)
+        out[c + "attention.linear_proj.weight"] = \
+            hf[p + "self_attn.o_proj.weight"]
+        out[c + "mlp.router.weight"] = \
+            hf[p + "block_sparse_moe.gate.weight"].float()
+        w1 = torch.stack([
+            torch.cat([hf[f"{p}block_sparse_moe.experts.{e}.w1.weight"],
+                       hf[f"{p}block_sparse_moe.experts.{e}.w3.weight"]],
+                      dim=0).t()
+            for e in range(E)])                       # [E, h, 2F]
+        w2 = torch.stack([
+            hf[f"{p}block_sparse_moe.experts.{e}.w2.weight"].t()
+            for e in range(E)])                       # [E, F, h]
+        out[c + "mlp.experts.w1"] = w1.contiguous()
+        out[c + "mlp.experts.w2"] = w2.contiguous()
+    out["final_norm.norm.weight"] = hf["model.norm.weight"]
+    out["lm_head.lm_head.weight"] = hf.get("lm_head.weight",
+                                           hf["model.embed_tokens.weight"])
+    return out
+
+
+def canonical_to_hf_mixtral(can: Dict[str, torch.Tensor],
+                            margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    out: Dict[str, torch.Tensor] = {}
+    out["model.embed_tokens.weight"] = can["embedding.word_embeddings.weight"]
+    F = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
+    for i in range(margs.num_hidden_layers):
+        p = f"model.layers.{i}."
+        c = f"decoder.{i}."
+        out[p + "input_layernorm.weight"] = can[c + "input_norm.weight"]
+        out[p + "post_attention_layernorm.weight"] = \
+            can[c + "post_attn_norm.weight"]
+        q, k, v = split_qkv(can[c + "attention.linear_qkv.weight"], margs)
+        out[p + "self_attn.q_proj.weight"] = q
+        out[p + "self_attn.k_proj.weight"] = k
+        out[p + "self_attn.v_proj.weight"] = v
+        out[p + "self_attn.o_proj.weight"] = \
+            can[c + "attention.linear_proj.weight"]
+        out[p + "block_sparse_moe.gate.weight"] = can[c + "mlp.router.weight"]
+        w1 = can[c + "mlp.experts.w1"]                # [E, h, 2F]
+        w2 = can[c + "mlp.experts.w2"]                # [E, F, h]
+        for e in range(w1.shape[0]):
+            out[f"{p}block_sparse_moe.experts.{e}.w1.weight"] = \
+                w1[e, :, :F].t().contiguous()
+            out[f"{p}block_sparse_moe.experts.{e}.w3.weight"] = \
+                w1[e, :, F:].t().contiguous()
+            out[f"{p}block_sparse_moe.experts.{e}.w2.weight"] = \
+                w2[e].t().contiguous()
+    out["model.norm.weight"] = can["final_norm.norm.weight"]
+    out["lm_head.weight"] = can["lm_head.lm_head.weight"]
+    return out
+
+
 def hf_gpt2_to_canonical(hf: Dict[str, torch.Tensor],
                          margs: ModelArgs) -> Dict[str, torch.Tensor]:
     """HF GPT2LMHeadModel (Conv1D: stored transposed) -> canonical."""
@@ -158,4 +229,6 @@ def hf_to_canonical(hf: Dict[str, torch.Tensor], margs: ModelArgs
                     ) -> Dict[str, torch.Tensor]:
     if margs.model_type == "gpt":
         return hf_gpt2_to_canonical(hf, margs)
+    if margs.model_type.startswith("moe") and margs.num_experts > 0:
+        return hf_mixtral_to_canonical(hf, margs)
     return hf_llama_to_canonical(hf, margs)

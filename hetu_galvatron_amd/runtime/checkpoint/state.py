@@ -73,6 +73,8 @@ def shard_for_rank(name: str, full: torch.Tensor, strategy, tp_rank: int,
     t, r = tp_size, tp_rank
     if t == 1 or strategy.use_ulysses:
         return full
+    if ".shared." in name or ".router." in name or "rel_bias" in name:
+        return full  # shared expert / router / relative bias: replicated
     base = name.split(".")[-2] + "." + name.split(".")[-1]
     if "word_embeddings.weight" in name or "lm_head.weight" in name:
         V = full.shape[0]
