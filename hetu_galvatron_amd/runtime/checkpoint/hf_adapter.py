@@ -225,10 +225,149 @@ def save_hf_checkpoint(state: Dict[str, torch.Tensor], path: str) -> None:
               os.path.join(path, "model.safetensors"))
 
 
+def hf_t5_to_canonical(hf: Dict[str, torch.Tensor],
+                       margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    """HF T5ForConditionalGeneration keys -> canonical (t5 family;
+    reference-equivalent role of the per-family adapters).
+
+    HF stores the relative_attention_bias table on block 0 only and
+    shares it; our per-layer tables each get a copy (relative_bias.py
+    layout note).  HF T5LayerNorm has no bias -> canonical layernorm
+    biases load as zeros.  Gated (v1.1 wi_0/wi_1) and non-gated (wi)
+    DenseReluDense both map onto fc1."""
+    out: Dict[str, torch.Tensor] = {}
+    emb = hf["shared.weight"]
+    out["embedding.word_embeddings.weight"] = emb
+    out["encdec_bridge.dec_embedding.word_embeddings.weight"] = emb
+    n_enc = margs.num_hidden_layers
+    n_dec = margs.num_decoder_layers or n_enc
+    h = margs.hidden_size
+
+    def norm(dst, src):
+        out[dst + ".weight"] = hf[src + ".weight"]
+        out[dst + ".bias"] = hf.get(src + ".bias",
+                                    torch.zeros_like(hf[src + ".weight"]))
+
+    def ffn(dst, src):
+        if src + ".DenseReluDense.wi_0.weight" in hf:  # gated v1.1
+            out[dst + ".fc1.weight"] = torch.cat(
+                [hf[src + ".DenseReluDense.wi_0.weight"],
+                 hf[src + ".DenseReluDense.wi_1.weight"]], dim=0)
+        else:
+            out[dst + ".fc1.weight"] = hf[src + ".DenseReluDense.wi.weight"]
+        out[dst + ".fc2.weight"] = hf[src + ".DenseReluDense.wo.weight"]
+
+    enc_bias = hf["encoder.block.0.layer.0.SelfAttention"
+                  ".relative_attention_bias.weight"]
+    dec_bias = hf["decoder.block.0.layer.0.SelfAttention"
+                  ".relative_attention_bias.weight"]
+    for i in range(n_enc):
+        p = f"encoder.block.{i}.layer"
+        c = f"encoder.{i}"
+        norm(c + ".input_norm", p + ".0.layer_norm")
+        norm(c + ".post_attn_norm", p + ".1.layer_norm")
+        a = p + ".0.SelfAttention"
+        out[c + ".attention.linear_qkv.weight"] = fuse_qkv(
+            hf[a + ".q.weight"], hf[a + ".k.weight"], hf[a + ".v.weight"],
+            margs)
+        out[c + ".attention.linear_proj.weight"] = hf[a + ".o.weight"]
+        out[c + ".rel_bias.weight"] = enc_bias.clone()
+        ffn(c + ".mlp", p + ".1")
+    norm("encdec_bridge.enc_final_norm", "encoder.final_layer_norm")
+    for j in range(n_dec):
+        p = f"decoder.block.{j}.layer"
+        c = f"decoder.{n_enc + j}"
+        norm(c + ".input_norm", p + ".0.layer_norm")
+        norm(c + ".cross_norm", p + ".1.layer_norm")
+        norm(c + ".post_attn_norm", p + ".2.layer_norm")
+        a = p + ".0.SelfAttention"
+        out[c + ".attention.linear_qkv.weight"] = fuse_qkv(
+            hf[a + ".q.weight"], hf[a + ".k.weight"], hf[a + ".v.weight"],
+            margs)
+        out[c + ".attention.linear_proj.weight"] = hf[a + ".o.weight"]
+        x = p + ".1.EncDecAttention"
+        out[c + ".cross_attention.linear_q.weight"] = hf[x + ".q.weight"]
+        out[c + ".cross_attention.linear_kv.weight"] = torch.cat(
+            [hf[x + ".k.weight"], hf[x + ".v.weight"]], dim=0)
+        out[c + ".cross_attention.linear_proj.weight"] = hf[x + ".o.weight"]
+        out[c + ".rel_bias.weight"] = dec_bias.clone()
+        ffn(c + ".mlp", p + ".2")
+    norm("final_norm.norm", "decoder.final_layer_norm")
+    out["lm_head.lm_head.weight"] = hf.get("lm_head.weight", emb)
+    return out
+
+
+def canonical_to_hf_t5(can: Dict[str, torch.Tensor],
+                       margs: ModelArgs) -> Dict[str, torch.Tensor]:
+    """Reverse mapping; per-layer bias tables collapse to HF's shared
+    block-0 table (layer 0's copy wins — exact when they are tied or an
+    HF import was the source)."""
+    out: Dict[str, torch.Tensor] = {}
+    out["shared.weight"] = can["embedding.word_embeddings.weight"]
+    n_enc = margs.num_hidden_layers
+    n_dec = margs.num_decoder_layers or n_enc
+    gated = margs.hidden_act in ("silu", "swiglu", "geglu")
+    for i in range(n_enc):
+        p = f"encoder.block.{i}.layer"
+        c = f"encoder.{i}"
+        out[p + ".0.layer_norm.weight"] = can[c + ".input_norm.weight"]
+        out[p + ".1.layer_norm.weight"] = can[c + ".post_attn_norm.weight"]
+        q, k, v = split_qkv(can[c + ".attention.linear_qkv.weight"], margs)
+        a = p + ".0.SelfAttention"
+        out[a + ".q.weight"], out[a + ".k.weight"], out[a + ".v.weight"] = \
+            q, k, v
+        out[a + ".o.weight"] = can[c + ".attention.linear_proj.weight"]
+        fc1 = can[c + ".mlp.fc1.weight"]
+        if gated:
+            F = fc1.shape[0] // 2
+            out[p + ".1.DenseReluDense.wi_0.weight"] = fc1[:F]
+            out[p + ".1.DenseReluDense.wi_1.weight"] = fc1[F:]
+        else:
+            out[p + ".1.DenseReluDense.wi.weight"] = fc1
+        out[p + ".1.DenseReluDense.wo.weight"] = can[c + ".mlp.fc2.weight"]
+    out["encoder.block.0.layer.0.SelfAttention"
+        ".relative_attention_bias.weight"] = can["encoder.0.rel_bias.weight"]
+    out["encoder.final_layer_norm.weight"] = \
+        can["encdec_bridge.enc_final_norm.weight"]
+    for j in range(n_dec):
+        p = f"decoder.block.{j}.layer"
+        c = f"decoder.{n_enc + j}"
+        out[p + ".0.layer_norm.weight"] = can[c + ".input_norm.weight"]
+        out[p + ".1.layer_norm.weight"] = can[c + ".cross_norm.weight"]
+        out[p + ".2.layer_norm.weight"] = can[c + ".post_attn_norm.weight"]
+        q, k, v = split_qkv(can[c + ".attention.linear_qkv.weight"], margs)
+        a = p + ".0.SelfAttention"
+        out[a + ".q.weight"], out[a + ".k.weight"], out[a + ".v.weight"] = \
+            q, k, v
+        out[a + ".o.weight"] = can[c + ".attention.linear_proj.weight"]
+        x = p + ".1.EncDecAttention"
+        out[x + ".q.weight"] = can[c + ".cross_attention.linear_q.weight"]
+        kv = can[c + ".cross_attention.linear_kv.weight"]
+        H = kv.shape[0] // 2
+        out[x + ".k.weight"], out[x + ".v.weight"] = kv[:H], kv[H:]
+        out[x + ".o.weight"] = can[c + ".cross_attention.linear_proj.weight"]
+        fc1 = can[c + ".mlp.fc1.weight"]
+        if gated:
+            F = fc1.shape[0] // 2
+            out[p + ".2.DenseReluDense.wi_0.weight"] = fc1[:F]
+            out[p + ".2.DenseReluDense.wi_1.weight"] = fc1[F:]
+        else:
+            out[p + ".2.DenseReluDense.wi.weight"] = fc1
+        out[p + ".2.DenseReluDense.wo.weight"] = can[c + ".mlp.fc2.weight"]
+    out["decoder.block.0.layer.0.SelfAttention"
+        ".relative_attention_bias.weight"] = \
+        can[f"decoder.{n_enc}.rel_bias.weight"]
+    out["decoder.final_layer_norm.weight"] = can["final_norm.norm.weight"]
+    out["lm_head.weight"] = can["lm_head.lm_head.weight"]
+    return out
+
+
 def hf_to_canonical(hf: Dict[str, torch.Tensor], margs: ModelArgs
                     ) -> Dict[str, torch.Tensor]:
     if margs.model_type == "gpt":
         return hf_gpt2_to_canonical(hf, margs)
+    if margs.model_type == "t5":
+        return hf_t5_to_canonical(hf, margs)
     if margs.model_type.startswith("moe") and margs.num_experts > 0:
         return hf_mixtral_to_canonical(hf, margs)
     return hf_llama_to_canonical(hf, margs)

@@ -173,3 +173,36 @@ def test_hf_mixtral_roundtrip_and_load():
     l1 = model.forward_backward(batch).loss
     l2 = model2.forward_backward(batch).loss
     assert abs(l1 - l2) < 1e-5
+
+
+def test_hf_t5_roundtrip_and_load():
+    """canonical -> HF-t5 layout -> canonical: loss equality (bias tables
+    collapse to HF's shared block-0 table, so start from an HF-imported
+    state where they are tied)."""
+    from hetu_galvatron_amd.runtime import get_train_iterator
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_t5, hf_t5_to_canonical)
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-t5"},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 4, "train_iters": 1,
+                  "distributed_backend": "gloo"},
+    })
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    can0 = canonical_state_from_stage(model.stage_model)
+    hf = canonical_to_hf_t5(can0, cfg.model)
+    can1 = hf_t5_to_canonical(hf, cfg.model)   # bias tables now tied
+    hf2 = canonical_to_hf_t5(can1, cfg.model)
+    for k in hf:
+        assert torch.equal(hf[k], hf2[k]), k
+    torch.manual_seed(9)
+    model2 = GalvatronModel(cfg)
+    load_full_state(model2.stage_model, can1, cfg.model)
+    load_full_state(model.stage_model, can1, cfg.model)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    batch = next(it)
+    l1 = model.forward_backward(batch).loss
+    l2 = model2.forward_backward(batch).loss
+    assert abs(l1 - l2) < 1e-5
