@@ -17,8 +17,8 @@ import torch
 def main(argv=None):
     from ..config.loader import load_config
     from ..runtime.checkpoint.hf_adapter import (
-        canonical_to_hf_llama, hf_to_canonical, load_hf_checkpoint,
-        save_hf_checkpoint)
+        canonical_to_hf_llama, canonical_to_hf_mixtral, canonical_to_hf_t5,
+        hf_to_canonical, load_hf_checkpoint, save_hf_checkpoint)
 
     args = list(sys.argv[1:] if argv is None else argv)
     mode = args.pop(0)
@@ -36,7 +36,13 @@ def main(argv=None):
         print(f"wrote canonical checkpoint: {ns.out} ({len(can)} tensors)")
     elif mode == "g2h":
         can = torch.load(ns.canonical, map_location="cpu", weights_only=True)
-        hf = canonical_to_hf_llama(can, cfg.model)
+        m = cfg.model
+        if m.model_type == "t5":
+            hf = canonical_to_hf_t5(can, m)
+        elif m.model_type.startswith("moe") and m.num_experts > 0:
+            hf = canonical_to_hf_mixtral(can, m)
+        else:
+            hf = canonical_to_hf_llama(can, m)
         save_hf_checkpoint(hf, ns.out_dir)
         print(f"wrote HF checkpoint dir: {ns.out_dir}")
     else:
