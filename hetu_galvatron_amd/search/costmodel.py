@@ -124,6 +124,18 @@ def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
         if select:
             tp_time = _msg_latency(select, msg_mb) * n_comm
 
+    # -- ring-CP kv exchange ----------------------------------------------
+    # zigzag ring: fwd rotates (k, v) cp-1 steps, bwd rotates (k, v, dk,
+    # dv) cp steps; per-step message = this rank's kv shard.  Priced at
+    # the 2-party p2p rate (dedicated xGMI links, overlappable with the
+    # attention MFMA work -> overlap_coe discount).
+    cp_time = 0.0
+    if s.cp > 1:
+        kv_mb = (lbsz * prof.seq_length / s.cp * prof.hidden_size *
+                 (2 if mixed_precision else 4) / (1024 * 1024))
+        p2p = hw.p2p_latency_per_mb.get(2, 0.0)
+        cp_time = kv_mb * p2p * (2 * (s.cp - 1) + 4 * s.cp) / hw.overlap_coe
+
     # -- combine with backward/DP overlap model ---------------------------
     if sdp > 1 and dc > 0:
         dp_overlap_time = dp_message_mb * dc_overlap
@@ -134,9 +146,10 @@ def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
         else:
             overlap_part = dp_overlap_time
             rest_part = bct - dp_overlap_time / hw.overlap_coe
-        result = fct + overlap_part + rest_part + tp_time + hw.extra_overhead
+        result = fct + overlap_part + rest_part + tp_time + cp_time \
+            + hw.extra_overhead
     else:
-        result = fct + bct + tp_time
+        result = fct + bct + tp_time + cp_time
 
     # -- ZeRO-3 forward allgather -----------------------------------------
     if s.dp_type == "zero3" and sdp > 1:

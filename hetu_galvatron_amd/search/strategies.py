@@ -32,31 +32,36 @@ def enumerate_strategies(world_size: int, args: SearchArgs,
     tp_degs = [t for t in _pow2_range(min(args.max_tp_deg, per_stage))]
     if args.disable_tp:
         tp_degs = [1]
+    cp_degs = [1] if args.disable_cp else \
+        _pow2_range(min(args.max_cp_deg, per_stage))
     for tsp in tp_degs:
-        dp = per_stage // tsp
-        if dp > 1 and args.disable_dp and tsp * pp_deg != world_size:
-            continue
-        modes = []
-        if tsp == 1:
-            modes = [("tp", tsp)]
-        else:
-            if args.sp_space in ("tp", "tp+sp"):
-                modes.append(("tp", tsp))
-            if args.sp_space in ("sp", "tp+sp") and not args.disable_sp:
-                modes.append(("sp", tsp))
-        for mode, deg in modes:
-            tp = deg if mode == "tp" else 1
-            sp = deg if mode == "sp" else 1
-            dp_types = ["ddp"]
-            if not args.disable_sdp and dp > 1:
-                dp_types.append("zero3")
-            for dpt in dp_types:
-                ckpts = [False] if args.disable_ckpt else [False, True]
-                for ck in ckpts:
-                    out.append(LayerStrategy(
-                        pp_deg=pp_deg, tp=tp, sp=sp, cp=1, dp=dp,
-                        dp_type="zero3" if dpt == "zero3" else "ddp",
-                        checkpoint=ck))
+        for cp in cp_degs:
+            if tsp * cp > per_stage:
+                continue
+            dp = per_stage // (tsp * cp)
+            if dp > 1 and args.disable_dp and tsp * cp * pp_deg != world_size:
+                continue
+            modes = []
+            if tsp == 1:
+                modes = [("tp", tsp)]
+            else:
+                if args.sp_space in ("tp", "tp+sp"):
+                    modes.append(("tp", tsp))
+                if args.sp_space in ("sp", "tp+sp") and not args.disable_sp:
+                    modes.append(("sp", tsp))
+            for mode, deg in modes:
+                tp = deg if mode == "tp" else 1
+                sp = deg if mode == "sp" else 1
+                dp_types = ["ddp"]
+                if not args.disable_sdp and dp * cp > 1:
+                    dp_types.append("zero3")
+                for dpt in dp_types:
+                    ckpts = [False] if args.disable_ckpt else [False, True]
+                    for ck in ckpts:
+                        out.append(LayerStrategy(
+                            pp_deg=pp_deg, tp=tp, sp=sp, cp=cp, dp=dp,
+                            dp_type="zero3" if dpt == "zero3" else "ddp",
+                            checkpoint=ck))
     return out
 
 

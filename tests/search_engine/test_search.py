@@ -250,3 +250,20 @@ def test_parallel_search_matches_sequential(tmp_path):
     par = eng.parallelism_optimization(None)
     assert abs(par.throughput - seq.throughput) < 1e-9
     assert par.pp_deg == seq.pp_deg and par.chunks == seq.chunks
+
+
+def test_enumerate_with_cp():
+    cfg = load_config(base={"model": {"model_name": "tiny-llama"},
+                            "search": {"disable_cp": 0, "max_cp_deg": 4}})
+    strats = enumerate_strategies(8, cfg.search, pp_deg=1)
+    assert all(s.degree_product() == 8 for s in strats)
+    assert {s.cp for s in strats} == {1, 2, 4}
+    assert any(s.cp == 2 and s.tp_sp == 2 for s in strats)
+
+
+def test_search_with_cp_space(tmp_path):
+    eng = make_engine(tmp_path, disable_cp=0, max_cp_deg=2)
+    best = eng.parallelism_optimization(None)
+    assert best is not None and best.throughput > 0
+    # cp plans must survive plan validation
+    best.plan.validate(8)
