@@ -77,6 +77,19 @@ def get_train_iterator(cfg: GalvatronConfig, device,
             size=max(cfg.data.synthetic_dataset_size, B), seed=cfg.train.seed)
     if cfg.model.model_type == "t5":
         s_enc = cfg.model.encoder_seq_length or cfg.model.seq_length
+        if cfg.data.dataset == "megatron" and cfg.data.data_path:
+            # span corruption over the token stream (t5 pretraining
+            # objective; datasets/t5_dataset.py)
+            from .datasets.t5_dataset import T5MaskedDataset
+            t5ds = T5MaskedDataset(ds, s_enc, cfg.model.seq_length,
+                                   cfg.model.vocab_size, seed=cfg.train.seed)
+            idx = 0
+            while True:
+                items = [t5ds[(idx + i) % len(t5ds)] for i in range(B)]
+                idx = (idx + B) % len(t5ds)
+                enc = torch.stack([it["enc_input_ids"] for it in items])
+                dec = torch.stack([it["dec_tokens"] for it in items])
+                yield build_enc_dec_batch_context(enc, dec, device)
         g = torch.Generator().manual_seed(cfg.train.seed + 77)
         idx = 0
         while True:

@@ -15,7 +15,7 @@ resident).  Multi-GPU serving (tp decode) is a documented v2 item.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 
