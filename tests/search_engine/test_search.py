@@ -267,3 +267,15 @@ def test_search_with_cp_space(tmp_path):
     assert best is not None and best.throughput > 0
     # cp plans must survive plan validation
     best.plan.validate(8)
+
+
+def test_example_plans_load():
+    """Checked-in example plans stay loadable + valid for 8 GPUs."""
+    import glob
+    from hetu_galvatron_amd.config import HybridParallelPlan
+    paths = glob.glob("examples/configs/galvatron_config_*.json")
+    assert len(paths) >= 3
+    for p in paths:
+        plan = HybridParallelPlan.load(p)
+        plan.validate(8)
+        assert plan.layer(0, 8).degree_product() == 8
