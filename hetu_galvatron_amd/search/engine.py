@@ -145,7 +145,13 @@ class SearchEngine:
         self.cfg = cfg
         self.args = cfg.search
         self.world = self.args.num_nodes * self.args.num_gpus_per_node
-        self.num_layers = cfg.model.num_hidden_layers
+        m = cfg.model
+        if m.model_type == "t5":
+            n_enc = m.num_hidden_layers
+            n_dec = m.num_decoder_layers or n_enc
+            self.num_layers = n_enc + n_dec
+        else:
+            self.num_layers = m.num_hidden_layers
         self.layer_profile = layer_profile
         self.other_profile = other_profile or OtherProfile()
         self.hw = hardware
@@ -153,18 +159,43 @@ class SearchEngine:
         self.mixed_precision = cfg.parallel.mixed_precision == "bf16"
         self.results: List[SearchResult] = []
 
+    @property
+    def layer_profiles(self) -> List[LayerProfile]:
+        """Per-layer-type profiles (reference: multi-layer-type DP,
+        dynamic_programming.py _build_dp_and_run_multi_layer_type).
+        `layer_profile` may be a single LayerProfile or a list."""
+        lp = self.layer_profile
+        return list(lp) if isinstance(lp, (list, tuple)) else [lp]
+
+    @property
+    def layer_types(self) -> List[int]:
+        """Layer index -> profile index.  t5: encoder layers type 0,
+        decoder layers type 1 when two profiles are given."""
+        m = self.cfg.model
+        n_types = len(self.layer_profiles)
+        if m.model_type == "t5" and n_types > 1:
+            n_enc = m.num_hidden_layers
+            return [0] * n_enc + [1] * (self.num_layers - n_enc)
+        return [0] * self.num_layers
+
     # -- profile loading ----------------------------------------------------
     def load_profiles(self, comp_path: str, mem_path: str, hw_dir: str) -> None:
         seq = self.cfg.model.seq_length
         fct = read_computation_profile(comp_path, seq)
         with open(mem_path) as f:
             mem = json.load(f)
-        lt = mem["layertype_0"]
-        self.layer_profile = LayerProfile(
-            parameter_mb=float(lt["parameter_size"]),
-            fct_linear=fct,
-            act_per_bsz_mb=lt["tp_activation_per_bsz_dict"],
-            seq_length=seq, hidden_size=self.cfg.model.hidden_size)
+        profiles = []
+        for t in range(8):
+            lt = mem.get(f"layertype_{t}")
+            if lt is None:
+                break
+            profiles.append(LayerProfile(
+                parameter_mb=float(lt["parameter_size"]),
+                fct_linear=tuple(lt.get("fct_linear", fct)),
+                act_per_bsz_mb=lt["tp_activation_per_bsz_dict"],
+                seq_length=seq, hidden_size=self.cfg.model.hidden_size))
+        assert profiles, "memory profile JSON missing layertype_0"
+        self.layer_profile = profiles[0] if len(profiles) == 1 else profiles
         other = mem.get("other", {})
         self.other_profile = OtherProfile(
             parameter_mb=float(other.get("parameter_size", 0.0)),
@@ -203,7 +234,9 @@ class SearchEngine:
     # -- core search --------------------------------------------------------
     def search_task(self, global_bsz: int, chunks: int, pp: int
                     ) -> Optional[SearchResult]:
-        lp, hw = self.layer_profile, self.hw
+        profiles, hw = self.layer_profiles, self.hw
+        lp = profiles[0]
+        ltypes = self.layer_types
         strategies = enumerate_strategies(self.world, self.args, pp)
         strategies = [s for s in strategies
                       if global_bsz % (s.dp * chunks) == 0]
@@ -230,34 +263,43 @@ class SearchEngine:
         other_time = {vtp: (m_o * (global_bsz / chunks) / vtp + c_o) *
                       (1 + hw.bct_fct_coe) for vtp in vtp_opts}
 
-        # per-layer intra cost (with and without grad sync) per strategy
-        intra_sync = np.zeros(S)
-        intra_nosync = np.zeros(S)
-        mem_per_strategy: Dict[int, np.ndarray] = {}
-        for si, s in enumerate(strategies):
-            intra_sync[si] = layer_time_cost(
-                s, lp, hw, global_bsz, chunks, self.world,
-                self.mixed_precision, no_gradient_sync=False)
-            intra_nosync[si] = layer_time_cost(
-                s, lp, hw, global_bsz, chunks, self.world,
-                self.mixed_precision, no_gradient_sync=True)
-        for stage in range(pp):
-            mem_per_strategy[stage] = np.array([
-                layer_memory_cost(s, lp, global_bsz, chunks, stage,
-                                  self.cfg.parallel.pipeline_type,
-                                  self.mixed_precision)["total"]
-                for s in strategies])
-
-        # inter-layer transition costs (layout change penalty, ms)
-        inter = np.zeros((S, S))
-        for a_i, sa in enumerate(strategies):
-            for b_i, sb in enumerate(strategies):
-                mb = transition_cost_mb(sa, sb, lp.seq_length, lp.hidden_size,
-                                        global_bsz / chunks / sb.dp,
-                                        self.mixed_precision)
-                lat = hw.allreduce_latency_per_mb.get(
-                    f"{max(sb.tp_sp * sb.cp, sa.tp_sp * sa.cp)}_1", 0.001)
-                inter[a_i, b_i] = mb * lat
+        # per-layer-TYPE intra cost (with and without grad sync) per
+        # strategy (reference: multi-layer-type DP)
+        intra_sync: List[np.ndarray] = []
+        intra_nosync: List[np.ndarray] = []
+        mem_per_strategy: List[Dict[int, np.ndarray]] = []
+        inter: List[np.ndarray] = []
+        for tlp in profiles:
+            isyn = np.zeros(S)
+            inos = np.zeros(S)
+            for si, s in enumerate(strategies):
+                isyn[si] = layer_time_cost(
+                    s, tlp, hw, global_bsz, chunks, self.world,
+                    self.mixed_precision, no_gradient_sync=False)
+                inos[si] = layer_time_cost(
+                    s, tlp, hw, global_bsz, chunks, self.world,
+                    self.mixed_precision, no_gradient_sync=True)
+            intra_sync.append(isyn)
+            intra_nosync.append(inos)
+            mem_per_strategy.append({
+                stage: np.array([
+                    layer_memory_cost(s, tlp, global_bsz, chunks, stage,
+                                      self.cfg.parallel.pipeline_type,
+                                      self.mixed_precision)["total"]
+                    for s in strategies])
+                for stage in range(pp)})
+            it = np.zeros((S, S))
+            for a_i, sa in enumerate(strategies):
+                for b_i, sb in enumerate(strategies):
+                    mb = transition_cost_mb(sa, sb, tlp.seq_length,
+                                            tlp.hidden_size,
+                                            global_bsz / chunks / sb.dp,
+                                            self.mixed_precision)
+                    lat = hw.allreduce_latency_per_mb.get(
+                        f"{max(sb.tp_sp * sb.cp, sa.tp_sp * sa.cp)}_1",
+                        0.001)
+                    it[a_i, b_i] = mb * lat
+            inter.append(it)
 
         best: Optional[SearchResult] = None
         for vtp in vtp_opts:
@@ -268,13 +310,16 @@ class SearchEngine:
             stage_nosync: List[float] = []
             stage_sync: List[float] = []
             feasible = True
+            lo = 0
             for stage in range(pp):
                 n_lay = division[stage]
-                v_units = np.ceil(mem_per_strategy[stage] /
-                                  self.mem_unit).astype(np.int32)
-                v_data = np.tile(v_units, (n_lay, 1))
-                intra_t = np.tile(intra_nosync, (n_lay, 1))
-                inter_t = np.tile(inter[None, :, :], (n_lay, 1, 1))
+                types = [ltypes[lo + l] for l in range(n_lay)]
+                lo += n_lay
+                v_data = np.stack([
+                    np.ceil(mem_per_strategy[t][stage] /
+                            self.mem_unit).astype(np.int32) for t in types])
+                intra_t = np.stack([intra_nosync[t] for t in types])
+                inter_t = np.stack([inter[t] for t in types])
                 inter_t[0] = 0.0
                 f, mark = solve_layer_dp(v_data, intra_t, inter_t,
                                          budget_units)
@@ -285,7 +330,8 @@ class SearchEngine:
                 stage_paths.append(path)
                 stage_nosync.append(cost)
                 stage_sync.append(cost + float(sum(
-                    intra_sync[p] - intra_nosync[p] for p in path)))
+                    intra_sync[t][p] - intra_nosync[t][p]
+                    for t, p in zip(types, path))))
             if not feasible:
                 continue
             # p2p term per stage boundary

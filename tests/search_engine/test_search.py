@@ -279,3 +279,31 @@ def test_example_plans_load():
         plan = HybridParallelPlan.load(p)
         plan.validate(8)
         assert plan.layer(0, 8).degree_product() == 8
+
+
+def test_multi_layertype_t5_search(tmp_path):
+    """t5: encoder/decoder layers carry distinct profiles through the DP
+    (reference multi-layer-type DP); decoder layers cost more -> the plan
+    covers enc+dec layer counts and stays valid."""
+    from hetu_galvatron_amd.search.costmodel import LayerProfile
+    base = {"model": {"model_name": "t5-3b"},
+            "search": {"num_nodes": 1, "num_gpus_per_node": 8,
+                       "memory_constraint": 240, "settle_bsz": 64,
+                       "settle_chunks": 8, "max_pp_deg": 2}}
+    cfg = load_config(base=base)
+    enc = mock_layer()
+    dec = LayerProfile(parameter_mb=1100.0, fct_linear=(6.0, 0.6),
+                       act_per_bsz_mb={"1": 640.0, "2": 330.0, "4": 170.0,
+                                       "8": 90.0, "checkpoint": 40.0},
+                       seq_length=512, hidden_size=1024)
+    op = OtherProfile(parameter_mb=130.0, act_per_bsz_mb={"1": 60.0},
+                      fct_linear=(0.3, 0.1))
+    eng = SearchEngine(cfg, [enc, dec], op, mock_hw())
+    n = cfg.model.num_hidden_layers + cfg.model.num_decoder_layers
+    assert eng.num_layers == n
+    assert eng.layer_types == [0] * cfg.model.num_hidden_layers + \
+        [1] * cfg.model.num_decoder_layers
+    best = eng.parallelism_optimization(None)
+    assert best is not None and best.throughput > 0
+    assert best.plan.num_layers == n
+    best.plan.validate(8)
