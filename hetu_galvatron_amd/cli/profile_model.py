@@ -10,10 +10,11 @@ from __future__ import annotations
 
 def main(argv=None):
     from ..config.loader import config_from_cli
-    from ..profiler.model import ModelProfiler
+    from ..profiler.model import ModelProfiler, T5ModelProfiler
 
     cfg = config_from_cli(argv)
-    mp = ModelProfiler(cfg)
+    mp = T5ModelProfiler(cfg) if cfg.model.model_type == "t5" \
+        else ModelProfiler(cfg)
     if cfg.profile.profile_type in ("computation", "all"):
         mp.launch_computation_profiling()
     if cfg.profile.profile_type in ("memory", "all"):

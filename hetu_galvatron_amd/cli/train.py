@@ -111,7 +111,11 @@ def main(argv=None):
         L = cfg.model.num_hidden_layers
         bsz = cfg.train.global_train_batch_size
         seq = cfg.model.seq_length
-        key = f"layernum[{L}]_bsz{bsz}_seq{seq}"
+        if cfg.model.model_type == "t5":
+            nd = cfg.model.num_decoder_layers or L
+            key = f"layernum[{L},{nd}]_bsz{bsz}_seq{seq}"
+        else:
+            key = f"layernum[{L}]_bsz{bsz}_seq{seq}"
         os.makedirs(p.profile_dir, exist_ok=True)
         prec = "bf16" if cfg.parallel.mixed_precision == "bf16" else "fp32"
         name = cfg.model.model_name or "model"

@@ -178,3 +178,116 @@ class ModelProfiler:
             with open(out, "w") as f:
                 json.dump(parsed, f, indent=4)
         return parsed
+
+
+class T5ModelProfiler(ModelProfiler):
+    """Enc-dec (t5) profiling: two layer types via two-axis layernum
+    differencing (reference: the multi-layer-type profiles consumed by
+    _build_dp_and_run_multi_layer_type).
+
+    Sweep points (enc, dec): (n1, n1), (n2, n1), (n1, n2) — the encoder
+    delta comes from axis 0, the decoder delta from axis 1; raw keys are
+    "layernum[enc,dec]_bsz{B}_seq{S}".  process_t5_data emits
+    layertype_0 (encoder) / layertype_1 (decoder), each with its own
+    fct_linear, ready for SearchEngine.load_profiles.
+    """
+
+    def _combos(self):
+        n1, n2 = self.p.profile_min_layer_num, self.p.profile_max_layer_num
+        return [(n1, n1), (n2, n1), (n1, n2)], n1, n2
+
+    def launch_computation_profiling(self) -> None:
+        p = self.p
+        combos, _, _ = self._combos()
+        for enc, dec in combos:
+            for bsz in range(p.profile_batch_size_start,
+                             p.profile_batch_size_end + 1):
+                self._run(self._base_overrides() + [
+                    "profile.profile_type=computation",
+                    f"model.num_hidden_layers={enc}",
+                    f"model.num_decoder_layers={dec}",
+                    f"train.global_train_batch_size={bsz}",
+                    "parallel.mixed_precision=" +
+                    self.cfg.parallel.mixed_precision])
+
+    def launch_memory_profiling(self, nproc: int = 8) -> None:
+        combos, _, _ = self._combos()
+        for enc, dec in combos:
+            self._run(self._base_overrides() + [
+                "profile.profile_type=memory",
+                f"model.num_hidden_layers={enc}",
+                f"model.num_decoder_layers={dec}",
+                f"train.global_train_batch_size={self.p.profile_fixed_batch_size}",
+            ], nproc=1)
+
+    def process_t5_data(self, comp_raw: Dict, mem_raw: Dict,
+                        write: bool = True) -> Dict:
+        """Two-axis differencing -> layertype_0/1 profile JSON."""
+        _, n1, n2 = self._combos()
+        dn = n2 - n1
+        seq = self.cfg.model.seq_length
+
+        cpat = re.compile(r"layernum\[(\d+),(\d+)\]_bsz(\d+)_seq(\d+)$")
+        by_pt: Dict[Tuple[int, int], Dict[int, float]] = {}
+        for k, v in comp_raw.items():
+            m = cpat.match(k)
+            if not m or int(m.group(4)) != seq:
+                continue
+            by_pt.setdefault((int(m.group(1)), int(m.group(2))),
+                             {})[int(m.group(3))] = float(v)
+        base = by_pt[(n1, n1)]
+        fcts = []
+        for pt in ((n2, n1), (n1, n2)):
+            xs, ys = [], []
+            for b in sorted(set(base) & set(by_pt[pt])):
+                xs.append(b)
+                ys.append((by_pt[pt][b] - base[b]) / dn)
+            m_, c_ = _fit(xs, ys)
+            fcts.append((m_, c_))
+
+        mpat = re.compile(
+            r"(\d+)_(\d+)_(\d+)(_c)?/layernum\[(\d+),(\d+)\]_bsz(\d+)"
+            r"_seq(\d+)_rank(\d+)_(act_peak|act|ms)$")
+        table: Dict[Tuple, float] = {}
+        for k, v in mem_raw.items():
+            m = mpat.match(k)
+            if not m or m.group(9) != "0":
+                continue
+            table[(int(m.group(5)), int(m.group(6)), m.group(10))] = float(v)
+        bszs = {int(mpat.match(k).group(7)) for k in mem_raw
+                if mpat.match(k)}
+        bsz = max(bszs) if bszs else self.p.profile_fixed_batch_size
+
+        out = {}
+        for ti, pt in enumerate(((n2, n1), (n1, n2))):
+            dms = (table[(pt[0], pt[1], "ms")] -
+                   table[(n1, n1, "ms")]) / dn
+            dact = (table[(pt[0], pt[1], "act")] -
+                    table[(n1, n1, "act")]) / dn
+            out[f"layertype_{ti}"] = {
+                "parameter_size": max(dms / 4.0, 1e-3),
+                "fct_linear": list(fcts[ti]),
+                "tp_activation_per_bsz_dict": {
+                    "1": max(dact / bsz, 1e-3),
+                    "checkpoint": max(dact / bsz, 1e-3) * 0.07},
+            }
+        base_ms = table[(n1, n1, "ms")]
+        per_layer_ms = (out["layertype_0"]["parameter_size"] +
+                        out["layertype_1"]["parameter_size"]) * 4.0 * n1
+        out["other"] = {
+            "parameter_size": max((base_ms - per_layer_ms) / 4.0, 0.0),
+            "tp_activation_per_bsz_dict": {"1": 1.0},
+        }
+        if write:
+            path = os.path.join(
+                self.dir, f"model_profile_{self.prec}_{self.name}.json")
+            os.makedirs(self.dir, exist_ok=True)
+            with open(path, "w") as f:
+                json.dump(out, f, indent=4)
+        return out
+
+
+def _fit(xs, ys):
+    import numpy as np
+    m, c = np.polyfit(np.asarray(xs, float), np.asarray(ys, float), 1)
+    return float(m), float(c)

@@ -73,3 +73,39 @@ def test_runtime_profiler_cpu_noop(tmp_path):
     rp.save_time_profile(os.path.join(tmp_path, "t.json"), "layernum[2]_bsz1_seq128")
     d = json.load(open(os.path.join(tmp_path, "t.json")))
     assert "layernum[2]_bsz1_seq128" in d
+
+
+def test_t5_model_profiler_two_axis():
+    """Fabricated (enc,dec) sweeps -> layertype_0/1 with per-type fct;
+    the parsed profile loads into the multi-layer-type search."""
+    import json
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.profiler.model import T5ModelProfiler
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-t5"},
+        "profile": {"profile_min_layer_num": 1,
+                    "profile_max_layer_num": 2,
+                    "profile_fixed_batch_size": 4}})
+    prof = T5ModelProfiler(cfg)
+    seq = cfg.model.seq_length
+    # enc layer: 2.0 ms/layer/bsz + 0.1; dec: 3.0/bsz + 0.2; base 1.0
+    comp = {}
+    for enc, dec in ((1, 1), (2, 1), (1, 2)):
+        for b in (1, 2, 4):
+            comp[f"layernum[{enc},{dec}]_bsz{b}_seq{seq}"] = \
+                1.0 + enc * (2.0 * b + 0.1) + dec * (3.0 * b + 0.2)
+    mem = {}
+    for enc, dec in ((1, 1), (2, 1), (1, 2)):
+        mem[f"1_1_1/layernum[{enc},{dec}]_bsz4_seq{seq}_rank0_ms"] = \
+            100.0 + enc * 40.0 + dec * 60.0
+        mem[f"1_1_1/layernum[{enc},{dec}]_bsz4_seq{seq}_rank0_act"] = \
+            20.0 + enc * 8.0 + dec * 12.0
+        mem[f"1_1_1/layernum[{enc},{dec}]_bsz4_seq{seq}_rank0_act_peak"] = 0.0
+    parsed = prof.process_t5_data(comp, mem, write=False)
+    lt0, lt1 = parsed["layertype_0"], parsed["layertype_1"]
+    assert abs(lt0["fct_linear"][0] - 2.0) < 1e-6
+    assert abs(lt1["fct_linear"][0] - 3.0) < 1e-6
+    assert abs(lt0["parameter_size"] - 10.0) < 1e-6   # 40/4
+    assert abs(lt1["parameter_size"] - 15.0) < 1e-6
+    assert abs(lt0["tp_activation_per_bsz_dict"]["1"] - 2.0) < 1e-6
