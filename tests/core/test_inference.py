@@ -89,3 +89,25 @@ def test_generate_graphed_cpu_fallback():
     out_e = gen.generate(ids, max_new_tokens=6, temperature=0.0)
     out_g = gen.generate_graphed(ids, max_new_tokens=6, warmup_steps=2)
     assert torch.equal(out_e, out_g)
+
+
+def test_generate_cli_with_hf_checkpoint(tmp_path):
+    """HF-dir load path: canonical -> HF safetensors dir -> generate CLI;
+    greedy tokens match the in-memory generator on the same weights."""
+    from hetu_galvatron_amd.cli.generate import main
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_llama, save_hf_checkpoint)
+    from hetu_galvatron_amd.runtime.checkpoint.state import (
+        canonical_state_from_stage)
+    model = make_model()
+    can = canonical_state_from_stage(model.stage_model)
+    hf_dir = tmp_path / "hf"
+    save_hf_checkpoint(canonical_to_hf_llama(can, model.cfg.model),
+                       str(hf_dir))
+    out = main(["model.model_name=tiny-llama",
+                "parallel.mixed_precision=fp32",
+                f"ckpt.load={hf_dir}",
+                "generate.max_new_tokens=4", "generate.prompt_ids=5,6,7"])
+    gen = GalvatronGenerator(model, max_batch=1, max_seq=64)
+    want = gen.generate(torch.tensor([[5, 6, 7]]), max_new_tokens=4)
+    assert torch.equal(out, want)
