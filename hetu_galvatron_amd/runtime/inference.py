@@ -120,13 +120,15 @@ class GalvatronGenerator:
                                              softmax_scale=self.scale)
             o = ob.permute(1, 0, 2, 3)
         else:
-            # chunked prefill continuation: per-token decode (rare path)
-            outs = []
-            for t in range(s_new):
-                outs.append(decode_attention(q[t], cache.k[li], cache.v[li],
-                                             pos + t + 1,
-                                             softmax_scale=self.scale))
-            o = torch.stack(outs, dim=0)
+            # chunked prefill continuation: one cross-length flash call —
+            # the new s_new queries attend the whole [0, pos+s_new) cache
+            # with bottom-right causal alignment (kernel supports sq<skv)
+            qb = q.permute(1, 0, 2, 3).contiguous()
+            kb = cache.k[li][:, :pos + s_new].contiguous()
+            vb = cache.v[li][:, :pos + s_new].contiguous()
+            ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
+                                             softmax_scale=self.scale)
+            o = ob.permute(1, 0, 2, 3)
         o = o.reshape(s_new, o.shape[1], -1)
         x = attn.linear_proj(o)
         hidden = residual + x

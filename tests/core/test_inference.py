@@ -111,3 +111,22 @@ def test_generate_cli_with_hf_checkpoint(tmp_path):
     gen = GalvatronGenerator(model, max_batch=1, max_seq=64)
     want = gen.generate(torch.tensor([[5, 6, 7]]), max_new_tokens=4)
     assert torch.equal(out, want)
+
+
+def test_chunked_prefill_matches_full():
+    """Feeding the prompt in two chunks == one-shot prefill (cross-length
+    bottom-right-causal attention against the cache)."""
+    model = make_model()
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    torch.manual_seed(6)
+    ids = torch.randint(0, model.cfg.model.vocab_size, (2, 12))
+    c1 = KVCache(len(gen.layers), 2, 64, model.cfg.model.kv_heads,
+                 model.cfg.model.head_dim, ids.device, dtype=torch.float32)
+    full = gen._forward_tokens(ids, c1)
+    c2 = KVCache(len(gen.layers), 2, 64, model.cfg.model.kv_heads,
+                 model.cfg.model.head_dim, ids.device, dtype=torch.float32)
+    gen._forward_tokens(ids[:, :5], c2)
+    chunked = gen._forward_tokens(ids[:, 5:], c2)
+    assert torch.allclose(full, chunked, atol=1e-4)
+    for li in range(len(gen.layers)):
+        assert torch.allclose(c1.k[li][:, :12], c2.k[li][:, :12], atol=1e-5)

@@ -408,3 +408,26 @@ def test_generate_graphed_gpu():
         logits = gen._forward_tokens(out[:, :9 + i], cache)
         got = logits.gather(1, out[:, 9 + i:10 + i])
         assert (logits.max(-1, keepdim=True).values - got).max() < 0.05, i
+
+
+def test_chunked_prefill_gpu():
+    """Chunked prefill == one-shot prefill on the native kernels (bf16)."""
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.inference import (GalvatronGenerator,
+                                                      KVCache)
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1},
+    })
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    ids = torch.randint(0, cfg.model.vocab_size, (2, 12), device="cuda")
+    mk = lambda: KVCache(len(gen.layers), 2, 64, cfg.model.kv_heads,
+                         cfg.model.head_dim, ids.device)
+    c1, c2 = mk(), mk()
+    full = gen._forward_tokens(ids, c1)
+    gen._forward_tokens(ids[:, :5], c2)
+    chunked = gen._forward_tokens(ids[:, 5:], c2)
+    assert (full - chunked).abs().max() < 0.1  # bf16 path tolerance
