@@ -117,18 +117,29 @@ def hf_mixtral_to_canonical(hf: Dict[str, torch.Tensor],
             hf[p + "self_attn.v_proj.weight"], margs)
         out[c + "attention.linear_proj.weight"] = \
             hf[p + "self_attn.o_proj.weight"]
-        out[c + "mlp.router.weight"] = \
-            hf[p + "block_sparse_moe.gate.weight"].float()
-        w1 = torch.stack([
-            torch.cat([hf[f"{p}block_sparse_moe.experts.{e}.w1.weight"],
-                       hf[f"{p}block_sparse_moe.experts.{e}.w3.weight"]],
-                      dim=0).t()
-            for e in range(E)])                       # [E, h, 2F]
-        w2 = torch.stack([
-            hf[f"{p}block_sparse_moe.experts.{e}.w2.weight"].t()
-            for e in range(E)])                       # [E, F, h]
-        out[c + "mlp.experts.w1"] = w1.contiguous()
-        out[c + "mlp.experts.w2"] = w2.contiguous()
+        # two HF layouts: legacy per-expert block_sparse_moe.experts.{e}
+        # (w1=gate, w3=up, w2=down) and the fused 3D layout
+        # (mlp.gate + mlp.experts.gate_up_proj/down_proj)
+        if p + "mlp.experts.gate_up_proj" in hf:
+            out[c + "mlp.router.weight"] = hf[p + "mlp.gate.weight"].float()
+            out[c + "mlp.experts.w1"] = \
+                hf[p + "mlp.experts.gate_up_proj"].transpose(1, 2) \
+                .contiguous()                          # [E, h, 2F]
+            out[c + "mlp.experts.w2"] = \
+                hf[p + "mlp.experts.down_proj"].transpose(1, 2).contiguous()
+        else:
+            out[c + "mlp.router.weight"] = \
+                hf[p + "block_sparse_moe.gate.weight"].float()
+            w1 = torch.stack([
+                torch.cat([hf[f"{p}block_sparse_moe.experts.{e}.w1.weight"],
+                           hf[f"{p}block_sparse_moe.experts.{e}.w3.weight"]],
+                          dim=0).t()
+                for e in range(E)])                   # [E, h, 2F]
+            w2 = torch.stack([
+                hf[f"{p}block_sparse_moe.experts.{e}.w2.weight"].t()
+                for e in range(E)])                   # [E, F, h]
+            out[c + "mlp.experts.w1"] = w1.contiguous()
+            out[c + "mlp.experts.w2"] = w2.contiguous()
     out["final_norm.norm.weight"] = hf["model.norm.weight"]
     out["lm_head.lm_head.weight"] = hf.get("lm_head.weight",
                                            hf["model.embed_tokens.weight"])
@@ -139,7 +150,6 @@ def canonical_to_hf_mixtral(can: Dict[str, torch.Tensor],
                             margs: ModelArgs) -> Dict[str, torch.Tensor]:
     out: Dict[str, torch.Tensor] = {}
     out["model.embed_tokens.weight"] = can["embedding.word_embeddings.weight"]
-    F = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
     for i in range(margs.num_hidden_layers):
         p = f"model.layers.{i}."
         c = f"decoder.{i}."
@@ -152,16 +162,12 @@ def canonical_to_hf_mixtral(can: Dict[str, torch.Tensor],
         out[p + "self_attn.v_proj.weight"] = v
         out[p + "self_attn.o_proj.weight"] = \
             can[c + "attention.linear_proj.weight"]
-        out[p + "block_sparse_moe.gate.weight"] = can[c + "mlp.router.weight"]
-        w1 = can[c + "mlp.experts.w1"]                # [E, h, 2F]
-        w2 = can[c + "mlp.experts.w2"]                # [E, F, h]
-        for e in range(w1.shape[0]):
-            out[f"{p}block_sparse_moe.experts.{e}.w1.weight"] = \
-                w1[e, :, :F].t().contiguous()
-            out[f"{p}block_sparse_moe.experts.{e}.w3.weight"] = \
-                w1[e, :, F:].t().contiguous()
-            out[f"{p}block_sparse_moe.experts.{e}.w2.weight"] = \
-                w2[e].t().contiguous()
+        # emit the fused 3D layout (what current transformers loads)
+        out[p + "mlp.gate.weight"] = can[c + "mlp.router.weight"]
+        out[p + "mlp.experts.gate_up_proj"] = \
+            can[c + "mlp.experts.w1"].transpose(1, 2).contiguous()
+        out[p + "mlp.experts.down_proj"] = \
+            can[c + "mlp.experts.w2"].transpose(1, 2).contiguous()
     out["model.norm.weight"] = can["final_norm.norm.weight"]
     out["lm_head.weight"] = can["lm_head.lm_head.weight"]
     return out

@@ -1,0 +1,144 @@
+"""Logit-level alignment vs HuggingFace transformers (the reference's
+test strategy: correctness vs HF baselines, SURVEY §4 / tests/core/test_tp.py).
+
+Random-init weights flow through our canonical->HF converters into the
+HF implementation; full-sequence logits must match to fp32 tolerance.
+This cross-validates RoPE conventions, GQA interleave, norm/activation
+math, router semantics, and the converters themselves.
+"""
+import pytest
+import torch
+
+from hetu_galvatron_amd.config import load_config
+from hetu_galvatron_amd.runtime import GalvatronModel
+from hetu_galvatron_amd.runtime.checkpoint.state import (
+    canonical_state_from_stage)
+
+TOL = 2e-5
+
+
+def build(model_name):
+    cfg = load_config(base={
+        "model": {"model_name": model_name},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    torch.manual_seed(0)
+    return cfg, GalvatronModel(cfg)
+
+
+def our_logits(model, ids):
+    """Full-sequence logits [b, s, V] via the serving engine's modules."""
+    from hetu_galvatron_amd.runtime.inference import (GalvatronGenerator,
+                                                      KVCache)
+    gen = GalvatronGenerator(model, max_batch=ids.shape[0], max_seq=64)
+    m = model.cfg.model
+    cache = KVCache(len(gen.layers), ids.shape[0], 64, m.kv_heads,
+                    m.head_dim, ids.device, dtype=torch.float32)
+    h = gen.embedding.word_embeddings(ids)
+    for li in range(len(gen.layers)):
+        h = gen._layer_step(li, h, cache, 0)
+    h = gen.final_norm.norm(h)
+    return gen.lm_head.lm_head(h).permute(1, 0, 2)
+
+
+def test_llama_logits_match_hf():
+    from transformers import LlamaConfig, LlamaForCausalLM
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_llama)
+    cfg, model = build("tiny-llama")
+    m = cfg.model
+    hf_sd = canonical_to_hf_llama(
+        canonical_state_from_stage(model.stage_model), m)
+    hf = LlamaForCausalLM(LlamaConfig(
+        vocab_size=m.vocab_size, hidden_size=m.hidden_size,
+        intermediate_size=m.ffn_hidden_size,
+        num_hidden_layers=m.num_hidden_layers,
+        num_attention_heads=m.num_attention_heads,
+        num_key_value_heads=m.kv_heads, head_dim=m.head_dim,
+        max_position_embeddings=m.max_position_embeddings,
+        rms_norm_eps=m.norm_epsilon, rope_theta=m.rope_theta,
+        attention_bias=False, tie_word_embeddings=False))
+    missing, unexpected = hf.load_state_dict(hf_sd, strict=False)
+    assert not missing and not unexpected
+    hf.eval()
+    ids = torch.randint(0, m.vocab_size, (2, 16))
+    with torch.no_grad():
+        want = hf(ids).logits
+    got = our_logits(model, ids)
+    assert (got - want).abs().max() < TOL
+
+
+def test_mixtral_logits_match_hf():
+    from transformers import MixtralConfig, MixtralForCausalLM
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_mixtral)
+    cfg, model = build("tiny-moe")
+    m = cfg.model
+    hf_sd = canonical_to_hf_mixtral(
+        canonical_state_from_stage(model.stage_model), m)
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=m.vocab_size, hidden_size=m.hidden_size,
+        intermediate_size=m.moe_ffn_hidden_size or m.ffn_hidden_size,
+        num_hidden_layers=m.num_hidden_layers,
+        num_attention_heads=m.num_attention_heads,
+        num_key_value_heads=m.kv_heads, head_dim=m.head_dim,
+        max_position_embeddings=m.max_position_embeddings,
+        rms_norm_eps=m.norm_epsilon, rope_theta=m.rope_theta,
+        num_local_experts=m.num_experts,
+        num_experts_per_tok=m.moe_router_topk,
+        router_aux_loss_coef=0.0, tie_word_embeddings=False))
+    missing, unexpected = hf.load_state_dict(hf_sd, strict=False)
+    assert not missing and not unexpected
+    hf.eval()
+    model.stage_model.blocks[1].inner.eval()  # no aux-loss attach
+    ids = torch.randint(0, m.vocab_size, (2, 16))
+    with torch.no_grad():
+        want = hf(ids).logits
+        ctx = {"input_ids": ids, "labels": ids.clone(),
+               "batch_size": 2, "seq_len": 16}
+        # forward through the training engine blocks, capture pre-head
+        hdn = None
+        x = None
+        for blk in model.stage_model.blocks:
+            if blk.kind == "lm_head":
+                hdn = x
+                break
+            x = blk(x, ctx)
+        logits = blk.inner.lm_head(hdn).permute(1, 0, 2)
+    assert (logits - want).abs().max() < TOL, \
+        (logits - want).abs().max()
+
+
+def test_gpt2_logits_match_hf():
+    from transformers import GPT2Config, GPT2LMHeadModel
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        hf_gpt2_to_canonical)
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    cfg, model = build("tiny-gpt")
+    m = cfg.model
+    hf = GPT2LMHeadModel(GPT2Config(
+        vocab_size=m.vocab_size, n_positions=m.max_position_embeddings,
+        n_embd=m.hidden_size, n_layer=m.num_hidden_layers,
+        n_head=m.num_attention_heads, n_inner=m.ffn_hidden_size,
+        layer_norm_epsilon=m.norm_epsilon,
+        activation_function="gelu_new" if False else "gelu",
+        resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0))
+    hf.eval()
+    can = hf_gpt2_to_canonical(hf.state_dict(), m)
+    load_full_state(model.stage_model, can, m)
+    for blk in model.stage_model.blocks:
+        blk.inner.eval()
+    ids = torch.randint(0, m.vocab_size, (2, 16))
+    with torch.no_grad():
+        want = hf(ids).logits
+        ctx = {"input_ids": ids, "labels": ids.clone(),
+               "batch_size": 2, "seq_len": 16}
+        x = None
+        for blk in model.stage_model.blocks:
+            if blk.kind == "lm_head":
+                break
+            x = blk(x, ctx)
+        logits = blk.inner.lm_head(x).permute(1, 0, 2)
+    assert (logits - want).abs().max() < 5e-4, \
+        (logits - want).abs().max()
