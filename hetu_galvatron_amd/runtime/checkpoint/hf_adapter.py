@@ -57,6 +57,11 @@ def hf_llama_to_canonical(hf: Dict[str, torch.Tensor],
             hf[p + "self_attn.q_proj.weight"],
             hf[p + "self_attn.k_proj.weight"],
             hf[p + "self_attn.v_proj.weight"], margs)
+        if p + "self_attn.q_proj.bias" in hf:  # qwen2-style qkv bias
+            out[c + "attention.linear_qkv.bias"] = fuse_qkv(
+                hf[p + "self_attn.q_proj.bias"],
+                hf[p + "self_attn.k_proj.bias"],
+                hf[p + "self_attn.v_proj.bias"], margs)
         out[c + "attention.linear_proj.weight"] = \
             hf[p + "self_attn.o_proj.weight"]
         out[c + "mlp.fc1.weight"] = torch.cat(
@@ -84,6 +89,12 @@ def canonical_to_hf_llama(can: Dict[str, torch.Tensor],
         out[p + "self_attn.q_proj.weight"] = q
         out[p + "self_attn.k_proj.weight"] = k
         out[p + "self_attn.v_proj.weight"] = v
+        if c + "attention.linear_qkv.bias" in can:  # qwen2-style qkv bias
+            qb, kb, vb = split_qkv(can[c + "attention.linear_qkv.bias"],
+                                   margs)
+            out[p + "self_attn.q_proj.bias"] = qb
+            out[p + "self_attn.k_proj.bias"] = kb
+            out[p + "self_attn.v_proj.bias"] = vb
         out[p + "self_attn.o_proj.weight"] = \
             can[c + "attention.linear_proj.weight"]
         fc1 = can[c + "mlp.fc1.weight"]
