@@ -307,3 +307,52 @@ def test_multi_layertype_t5_search(tmp_path):
     assert best is not None and best.throughput > 0
     assert best.plan.num_layers == n
     best.plan.validate(8)
+
+
+def test_two_node_mocked_search(tmp_path):
+    """2x8 GPUs via mocked bandwidth JSONs (multi-node is
+    exercised purely through the cost model): inter-node allreduce is ~5x
+    slower, so big-dp plans should prefer heavier sharding/ckpt under a
+    tight budget, and the search must emit a valid 16-GPU plan."""
+    import json
+    hw_dir = tmp_path / "hw2"
+    hw_dir.mkdir()
+    (hw_dir / "allreduce_bandwidth_2nodes_8gpus_per_node.json").write_text(
+        json.dumps({"allreduce_size_16_consec_1": 45.0,
+                    "allreduce_size_8_consec_1": 150.0,
+                    "allreduce_size_8_consec_0": 40.0,
+                    "allreduce_size_4_consec_1": 155.0,
+                    "allreduce_size_4_consec_0": 42.0,
+                    "allreduce_size_2_consec_1": 140.0,
+                    "allreduce_size_2_consec_0": 44.0}))
+    (hw_dir / "p2p_bandwidth_2nodes_8gpus_per_node.json").write_text(
+        json.dumps({"pp_size_2": 8.0, "pp_size_4": 80.0,
+                    "pp_size_8": 100.0, "pp_size_16": 90.0}))
+    sp = {}
+    for n in (2, 4, 8, 16):
+        for mb in (1, 4, 16, 64):
+            slow = 5.0 if n == 16 else 1.0
+            sp[f"allreduce_size_{n}_{mb}MB_time"] = (0.05 + mb * 0.008) * slow
+            sp[f"all2all_size_{n}_{mb}MB_time"] = (0.05 + mb * 0.006) * slow
+    (hw_dir / "sp_time_2nodes_8gpus_per_node.json").write_text(json.dumps(sp))
+    (hw_dir / "overlap_coefficient.json").write_text(
+        json.dumps({"overlap_coe": 1.15}))
+
+    from hetu_galvatron_amd.search.engine import read_hardware_profiles
+    hw = read_hardware_profiles(str(hw_dir), nodes=2, gpus=8)
+    assert "16_1" in hw.allreduce_latency_per_mb
+    assert 16 in hw.p2p_latency_per_mb
+
+    cfg = load_config(base={
+        "model": {"model_name": "llama-3-8b"},
+        "search": {"num_nodes": 2, "num_gpus_per_node": 8,
+                   "memory_constraint": 240, "settle_bsz": 128,
+                   "settle_chunks": 8, "max_pp_deg": 4,
+                   "max_tp_deg": 8}})
+    op = OtherProfile(parameter_mb=2100.0,
+                      act_per_bsz_mb={"1": 600.0}, fct_linear=(1.2, 0.2))
+    eng = SearchEngine(cfg, mock_layer(), op, hw)
+    best = eng.parallelism_optimization(None)
+    assert best is not None and best.throughput > 0
+    best.plan.validate(16)
+    assert best.plan.layer(0, 16).degree_product() == 16
