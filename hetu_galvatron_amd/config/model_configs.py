@@ -75,6 +75,7 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         vocab_size=32000, max_position_embeddings=8192, seq_length=4096,
         hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-5,
         position_embedding_type="rope", rope_theta=10000.0,
+        sliding_window=4096,
     ),
     "qwen2.5-7b": dict(
         model_type="llama", hidden_size=3584, num_hidden_layers=28,

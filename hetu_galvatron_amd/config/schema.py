@@ -63,6 +63,7 @@ class ModelArgs(BaseModel):
     relative_attention_max_distance: int = 128
     rope_theta: float = 500000.0
     rope_scaling: Optional[float] = None
+    sliding_window: Optional[int] = None  # mistral-style windowed attention
     add_bias_linear: bool = False
     add_qkv_bias: bool = False
     tie_word_embeddings: bool = False

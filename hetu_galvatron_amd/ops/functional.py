@@ -121,35 +121,46 @@ class _FlashAttention(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, softmax_scale):
+    def forward(ctx, q, k, v, causal, softmax_scale, window):
         scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        # an effective window (window < kv length) routes to the eager
+        # reference path (a window argument on the native kernel is a v2
+        # item); window >= kv length is a no-op -> native
+        eff_window = window if (window is not None and
+                                window < k.shape[1]) else None
+        if eff_window is None and use_native(q) \
+                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale)
         else:
-            o, lse = ref.attention_fwd(q, k, v, causal, scale)
+            o, lse = ref.attention_fwd(q, k, v, causal, scale,
+                                       window=eff_window)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.window = eff_window
         return o, lse
 
     @staticmethod
     def backward(ctx, do, dlse):
         q, k, v, o, lse = ctx.saved_tensors
         do = do.contiguous()
-        if use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        if ctx.window is None and use_native(q) \
+                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             dq, dk, dv = get_ext(False).flash_attn_bwd(
                 do, q, k, v, o, lse, ctx.causal, ctx.scale)
         else:
-            dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal, ctx.scale)
-        return dq, dk, dv, None, None
+            dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal,
+                                           ctx.scale, window=ctx.window)
+        return dq, dk, dv, None, None, None
 
 
 def flash_attention(q, k, v, causal: bool = True,
                     softmax_scale: Optional[float] = None,
-                    return_lse: bool = False):
-    """q: [b,s,hq,d]; k,v: [b,s,hkv,d] -> o [b,s,hq,d] (+ lse [b,hq,s])."""
+                    return_lse: bool = False, window: Optional[int] = None):
+    """q: [b,s,hq,d]; k,v: [b,s,hkv,d] -> o [b,s,hq,d] (+ lse [b,hq,s]).
+    window: mistral-style sliding window (causal only)."""
     o, lse = _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                                   causal, softmax_scale)
+                                   causal, softmax_scale, window)
     return (o, lse) if return_lse else o
 
 

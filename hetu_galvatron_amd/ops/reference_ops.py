@@ -125,6 +125,7 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                   causal: bool = True, softmax_scale: Optional[float] = None,
                   bias: Optional[torch.Tensor] = None,
+                  window: Optional[int] = None,
                   ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Reference attention WITH log-sum-exp output (needed for ring-CP merges).
 
@@ -150,6 +151,9 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         i = torch.arange(sq, device=q.device)[:, None]
         j = torch.arange(skv, device=q.device)[None, :]
         mask = j > (i + (skv - sq))
+        if window is not None:
+            # mistral sliding window: key positions below the window edge
+            mask = mask | (j <= (i + (skv - sq) - window))
         scores = scores.masked_fill(mask, float("-inf"))
     lse = torch.logsumexp(scores, dim=-1)  # [b,h,sq]
     p = torch.exp(scores - lse.unsqueeze(-1))
@@ -159,7 +163,7 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 def attention_bwd(do, q, k, v, o, lse, causal=True, softmax_scale=None,
-                  bias=None):
+                  bias=None, window=None):
     """Reference backward, flash-style against the PASSED o/lse.
 
     Using the caller's (global) lse/o matters for ring-CP: each per-block
@@ -176,7 +180,8 @@ def attention_bwd(do, q, k, v, o, lse, causal=True, softmax_scale=None,
         b32 = bias.detach().float().requires_grad_(True) \
             if bias is not None else None
         with torch.enable_grad():
-            o2, _ = attention_fwd(q32, k32, v32, causal, softmax_scale, b32)
+            o2, _ = attention_fwd(q32, k32, v32, causal, softmax_scale, b32,
+                                  window)
         if b32 is None:
             gq, gk, gv = torch.autograd.grad(o2, (q32, k32, v32), do.float())
             return gq.to(q.dtype), gk.to(k.dtype), gv.to(v.dtype)
@@ -200,7 +205,10 @@ def attention_bwd(do, q, k, v, o, lse, causal=True, softmax_scale=None,
     if causal:
         i = torch.arange(sq, device=q.device)[:, None]
         j = torch.arange(skv, device=q.device)[None, :]
-        s = s.masked_fill(j > (i + (skv - sq)), float("-inf"))
+        mask = j > (i + (skv - sq))
+        if window is not None:
+            mask = mask | (j <= (i + (skv - sq) - window))
+        s = s.masked_fill(mask, float("-inf"))
     p = torch.exp(s - lse.float().unsqueeze(-1))
     p = torch.nan_to_num(p)
     dv_h = torch.matmul(p.transpose(-1, -2), dof)       # [b,h,skv,d]

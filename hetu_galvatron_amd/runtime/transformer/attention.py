@@ -66,7 +66,10 @@ class SelfAttention(nn.Module):
             bias=m.add_bias_linear,
             sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
         self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
+        self.window = getattr(m, "sliding_window", None)
         cp = group_size(cp_group) if cp_group is not None else 1
+        if self.window is not None:
+            assert cp == 1, "sliding window + ring-CP is a v2 item"
         inner = ZigzagRingAttention(cp_group) if cp > 1 else None
         if use_ulysses and group_size(sp_group) > 1:
             self.core_attention = DistributedAttention(
@@ -108,11 +111,21 @@ class SelfAttention(nn.Module):
             o = eager_bias_attention(q, k, v, attn_bias, self.causal,
                                      self.softmax_scale)
         elif self.core_attention is not None:
-            o = self.core_attention(q, k, v, causal=self.causal,
-                                    softmax_scale=self.softmax_scale)
+            # ulysses a2a keeps the global seq intact: window semantics
+            # apply unchanged to the inner full-seq attention
+            if self.window is not None and isinstance(
+                    self.core_attention, DistributedAttention):
+                assert self.core_attention.inner_attention is None
+                o = self.core_attention(q, k, v, causal=self.causal,
+                                        softmax_scale=self.softmax_scale,
+                                        window=self.window)
+            else:
+                o = self.core_attention(q, k, v, causal=self.causal,
+                                        softmax_scale=self.softmax_scale)
         else:
             o = local_attention(q, k, v, causal=self.causal,
-                                softmax_scale=self.softmax_scale)
+                                softmax_scale=self.softmax_scale,
+                                window=self.window)
         o = o.permute(1, 0, 2, 3).reshape(s, b, -1)  # back to SBH
         return self.linear_proj(o)
 

@@ -24,9 +24,12 @@ from ..tensor_parallel.mappings import all_to_all, group_rank, group_size
 
 
 def local_attention(q, k, v, causal: bool = True,
-                    softmax_scale: Optional[float] = None) -> torch.Tensor:
-    """Plain local flash attention, autograd-capable. [b, s, h, d]."""
-    return flash_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
+                    softmax_scale: Optional[float] = None,
+                    window: Optional[int] = None) -> torch.Tensor:
+    """Plain local flash attention, autograd-capable. [b, s, h, d].
+    window: mistral sliding window (no-op when window >= kv length)."""
+    return flash_attention(q, k, v, causal=causal, softmax_scale=softmax_scale,
+                           window=window)
 
 
 # ---------------------------------------------------------------------------
@@ -71,7 +74,7 @@ class DistributedAttention(torch.nn.Module):
         self.inner_attention = inner_attention
 
     def forward(self, q, k, v, causal=True, softmax_scale=None,
-                attn_bias=None):
+                attn_bias=None, window=None):
         # q: [b, s_local, hq, d]; k/v: [b, s_local, hkv, d].
         # attn_bias [H_full, S, S] (t5 relative bias): sliced to this
         # rank's post-a2a head chunk; incompatible with a ring inner.
@@ -98,7 +101,8 @@ class DistributedAttention(torch.nn.Module):
         elif self.inner_attention is not None:
             o = self.inner_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
         else:
-            o = local_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
+            o = local_attention(q, k, v, causal=causal,
+                                softmax_scale=softmax_scale, window=window)
         if sp > 1:
             o = all_to_all(o, self.sp_group, scatter_dim=1, gather_dim=2)
         return o

@@ -170,3 +170,46 @@ def test_qwen2_logits_match_hf():
         want = hf(ids).logits
     got = our_logits(model, ids)
     assert (got - want).abs().max() < TOL, (got - want).abs().max()
+
+
+def test_mistral_sliding_window_logits_match_hf():
+    """Sliding-window attention (window < seq so the mask is active)
+    matches transformers' Mistral."""
+    from transformers import MistralConfig, MistralForCausalLM
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_llama)
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama", "sliding_window": 6},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    m = cfg.model
+    hf_sd = canonical_to_hf_llama(
+        canonical_state_from_stage(model.stage_model), m)
+    hf = MistralForCausalLM(MistralConfig(
+        vocab_size=m.vocab_size, hidden_size=m.hidden_size,
+        intermediate_size=m.ffn_hidden_size,
+        num_hidden_layers=m.num_hidden_layers,
+        num_attention_heads=m.num_attention_heads,
+        num_key_value_heads=m.kv_heads, head_dim=m.head_dim,
+        max_position_embeddings=m.max_position_embeddings,
+        rms_norm_eps=m.norm_epsilon, rope_theta=m.rope_theta,
+        sliding_window=6, tie_word_embeddings=False,
+        attn_implementation="eager"))
+    missing, unexpected = hf.load_state_dict(hf_sd, strict=False)
+    assert not missing and not unexpected
+    hf.eval()
+    ids = torch.randint(0, m.vocab_size, (2, 16))
+    with torch.no_grad():
+        want = hf(ids).logits
+        ctx = {"input_ids": ids, "labels": ids.clone(),
+               "batch_size": 2, "seq_len": 16}
+        x = None
+        for blk in model.stage_model.blocks:
+            if blk.kind == "lm_head":
+                break
+            x = blk(x, ctx)
+        got = blk.inner.lm_head(x).permute(1, 0, 2)
+    assert (got - want).abs().max() < TOL, (got - want).abs().max()
