@@ -75,6 +75,10 @@ class GalvatronGenerator:
         self.lm_head = self.blocks[-1].inner
         self.max_batch = max_batch
         self.max_seq = max_seq
+        if getattr(m, "sliding_window", None) is not None:
+            # decode attends the full cache; windowed eviction is a v2 item
+            assert max_seq <= m.sliding_window, \
+                "generator v1: max_seq must fit the sliding window"
         self.scale = 1.0 / (m.head_dim ** 0.5)
         self._dev = next(self.embedding.parameters()).device
         self._graphs: dict = {}   # batch size -> captured decode step
