@@ -277,3 +277,20 @@ def test_moe_shared_expert_etp2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_moe_world4_pp2_ep2():
+    """MoE composed with pipeline parallelism: 2 stages x (ep=2 within
+    each stage's sdp domain); expert grads reduce over edp while boundary
+    activations flow 1F1B."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=4, pp=2, tp=1, dp_type="ddp",
+        global_bsz=4, chunks=2, ep=2)
+    res = run_distributed(_moe_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)

@@ -189,3 +189,21 @@ def test_t5_world4_tp2_cp2():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_world4_pp2_tp2():
+    """t5 with pp=2 x tp=2: encoder-shaped boundary under megatron-SP
+    seq sharding + head-sliced relative bias on both stages."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=4, pp=2, tp=2, dp_type="ddp",
+        global_bsz=4, chunks=2, vtp=2)
+    res = run_distributed(_t5_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
