@@ -9,9 +9,10 @@ MI355X design: decode attention is one memory-bound HIP kernel
 (`decode_attn` in ops/csrc/elementwise.hip) streaming the bf16 KV cache
 at HBM rate; prefill reuses the MFMA flash kernel while writing the
 cache.  288 GB HBM3E comfortably holds an 8B model + tens of GB of KV
-cache on ONE GPU, so the v1 serving topology is world=1 (tp/pp/cp=1,
+cache on ONE GPU, so the primary serving topology is world=1 (tp/pp/cp=1,
 any dp_type — at world 1 every flat-param mode keeps full params
-resident).  Multi-GPU serving (tp decode) is a documented v2 item.
+resident).  GalvatronTPGenerator adds megatron-TP decode across GPUs
+(world == tp) for models beyond one GPU's memory.
 """
 from __future__ import annotations
 
@@ -21,7 +22,7 @@ import torch
 
 from ..ops import decode_attention, flash_attention_fwd_only
 
-__all__ = ["KVCache", "GalvatronGenerator"]
+__all__ = ["KVCache", "GalvatronGenerator", "GalvatronTPGenerator"]
 
 
 class KVCache:
@@ -85,7 +86,8 @@ class GalvatronGenerator:
         self._kv: dict = {}       # batch size -> persistent KVCache
 
     # -- one decoder layer, cache-aware ------------------------------------
-    def _split_qkv(self, attn, qkv: torch.Tensor):
+    @staticmethod
+    def _split_qkv(attn, qkv: torch.Tensor):
         """Interleaved-group [q*(hq/hkv), k, v] layout (attention.py)."""
         s, b = qkv.shape[0], qkv.shape[1]
         qkv = qkv.view(s, b, attn.num_groups_local, attn.q_per_group + 2,
@@ -343,3 +345,141 @@ class GalvatronGenerator:
             logits = logits.masked_fill(logits < kth, float("-inf"))
         probs = torch.softmax(logits, dim=-1)
         return torch.multinomial(probs, 1).squeeze(-1)
+
+
+class GalvatronTPGenerator:
+    """Megatron-TP decode across GPUs (serving topology v2 -> now v1.5):
+    world == tp, dp/pp/cp = 1, non-ulysses.
+
+    Runs the SHARDED module weights directly with replicated single-token
+    activations — column linears produce local heads / columns, row
+    linears produce partials reduced over tp, the vocab-sharded lm head
+    is gathered on the last dim before argmax — so decode needs exactly
+    TWO all-reduces + one head gather per token per layer-stack pass
+    (the same collective count as a Megatron training step's attention +
+    MLP row projections).  Greedy only; every rank computes identical
+    tokens (collectives keep them in lockstep).
+    """
+
+    def __init__(self, model, max_batch: int = 1, max_seq: int = 4096):
+        import torch.distributed as dist
+        sm = model.stage_model
+        self.world = sm.world_size
+        assert sm.pp_deg == 1, "tp decode: pp=1"
+        self.cfg = model.cfg
+        m = self.cfg.model
+        assert m.position_embedding_type == "rope"
+        self.margs = m
+        self.blocks = sm.blocks
+        self.layers = [b.inner for b in self.blocks if b.kind == "decoder"]
+        self.embedding = self.blocks[0].inner
+        self.final_norm = next(b.inner for b in self.blocks
+                               if b.kind == "final_norm")
+        self.lm_head = self.blocks[-1].inner
+        g0 = self.blocks[1].groups
+        s0 = g0.strategy
+        assert not s0.use_ulysses and s0.cp == 1 and s0.dp == 1 and \
+            s0.tp == self.world, \
+            "tp decode expects a pure megatron-tp plan (tp == world)"
+        self.tp_group = g0.tp_group.group
+        self.max_batch = max_batch
+        self.max_seq = max_seq
+        self.scale = 1.0 / (m.head_dim ** 0.5)
+        self._dev = next(self.embedding.parameters()).device
+
+    def _embed(self, ids: torch.Tensor) -> torch.Tensor:
+        """ids [b, s] -> replicated [s, b, h] (vocab-sharded table)."""
+        import torch.distributed as dist
+        import torch.nn.functional as F
+        we = self.embedding.word_embeddings
+        mask = (ids < we.vocab_start_index) | (ids >= we.vocab_end_index)
+        local = (ids - we.vocab_start_index).masked_fill(mask, 0)
+        h = F.embedding(local, we.weight).masked_fill(mask.unsqueeze(-1), 0.0)
+        dist.all_reduce(h, group=self.tp_group)
+        return h.transpose(0, 1).contiguous()  # [s, b, h]
+
+    def _layer_step(self, li: int, hidden, cache: KVCache, pos: int):
+        """hidden [s,b,h] replicated; local-head attention + tp-reduced
+        projections."""
+        import torch.distributed as dist
+        import torch.nn.functional as F
+        from ..ops import swiglu
+        layer = self.layers[li]
+        attn = layer.attention
+        residual = hidden
+        x = layer.input_norm(hidden)
+        qkv = F.linear(x, attn.linear_qkv.weight,
+                       getattr(attn.linear_qkv, "bias", None))
+        q, k, v = GalvatronGenerator._split_qkv(attn, qkv)
+        if layer.rotary is not None:
+            cos, sin = layer.rotary.full_tables(pos + q.shape[0], x.device)
+            from .transformer.rope import apply_rope_qk
+            q, k = apply_rope_qk(q.contiguous(), k.contiguous(),
+                                 cos[pos:], sin[pos:])
+        cache.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3))
+        s_new = q.shape[0]
+        if s_new == 1:
+            o = decode_attention(q[0], cache.k[li], cache.v[li], pos + 1,
+                                 softmax_scale=self.scale).unsqueeze(0)
+        else:
+            qb = q.permute(1, 0, 2, 3).contiguous()
+            kb = cache.k[li][:, :pos + s_new].contiguous()
+            vb = cache.v[li][:, :pos + s_new].contiguous()
+            ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
+                                             softmax_scale=self.scale)
+            o = ob.permute(1, 0, 2, 3)
+        o = o.reshape(s_new, o.shape[1], -1)
+        part = F.linear(o, attn.linear_proj.weight)  # row-parallel partial
+        dist.all_reduce(part, group=self.tp_group)
+        if attn.linear_proj.bias is not None:
+            part = part + attn.linear_proj.bias
+        hidden = residual + part
+        residual = hidden
+        x = layer.post_attn_norm(hidden)
+        h1 = F.linear(x, layer.mlp.fc1.weight,
+                      getattr(layer.mlp.fc1, "bias", None))
+        h1 = swiglu(h1)
+        h2 = F.linear(h1, layer.mlp.fc2.weight)
+        dist.all_reduce(h2, group=self.tp_group)
+        if layer.mlp.fc2.bias is not None:
+            h2 = h2 + layer.mlp.fc2.bias
+        return residual + h2
+
+    @torch.no_grad()
+    def _forward_tokens(self, tokens, cache: KVCache):
+        import torch.distributed as dist
+        pos = cache.cur_len
+        h = self._embed(tokens)
+        for li in range(len(self.layers)):
+            h = self._layer_step(li, h, cache, pos)
+        cache.advance(tokens.shape[1])
+        h = self.final_norm.norm(h[-1:])
+        import torch.nn.functional as F
+        local = F.linear(h, self.lm_head.lm_head.weight)  # [1,b,V/t]
+        parts = [torch.empty_like(local) for _ in range(self.world)]
+        dist.all_gather(parts, local.contiguous(), group=self.tp_group)
+        return torch.cat(parts, dim=-1)[0].float()  # [b, V]
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32
+                 ) -> torch.Tensor:
+        b, sp = input_ids.shape
+        assert sp + max_new_tokens <= self.max_seq
+        m = self.margs
+        tpd = self.world
+        cache = KVCache(len(self.layers), b, self.max_seq,
+                        max(m.kv_heads // tpd, 1), m.head_dim, self._dev,
+                        dtype=next(self.embedding.parameters()).dtype)
+        for blk in self.blocks:
+            if blk.flat is not None:
+                blk.flat.gather_params()
+        for blk in self.blocks:
+            blk.inner.eval()
+        logits = self._forward_tokens(input_ids, cache)
+        out = [input_ids]
+        for _ in range(max_new_tokens):
+            nxt = logits.argmax(-1)
+            out.append(nxt.unsqueeze(1))
+            if len(out) - 1 < max_new_tokens:
+                logits = self._forward_tokens(nxt.unsqueeze(1), cache)
+        return torch.cat(out, dim=1)
