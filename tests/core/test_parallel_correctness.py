@@ -202,3 +202,13 @@ def test_world4_mixed_ulysses_cp():
         checkpoint_flags=[0, 0], global_bsz=4, chunks=1,
         default_dp_type="ddp", vtp=1, vsp=0, vcp=1)
     run_case(4, plan)
+
+
+@pytest.mark.distributed
+def test_world4_pp2_zero3_ckpt():
+    """pp2 x dp2-zero3 x activation ckpt x 2 chunks — the deepest dense
+    composition (sharded params re-gathered inside checkpointed
+    recompute, 1F1B boundaries)."""
+    run_case(4, HybridParallelPlan.uniform(
+        N_LAYERS, 4, pp=2, tp=1, dp_type="zero3",
+        checkpoint=True, global_bsz=4, chunks=2))
