@@ -63,9 +63,14 @@ def _rank_bias(rel_bias, attention, groups, hidden):
     s = groups.strategy
     c = groups.coord_of(_my_rank())
     if s.use_ulysses:
-        assert s.cp == 1, "ulysses + ring-CP + relative bias is a v2 item"
-        S = hidden.shape[0] * s.tp_sp
-        return rel_bias(S, S, hidden.device)
+        # rows after the a2a = the full cp-local zigzag pair; heads are
+        # chunked inside DistributedAttention
+        S = hidden.shape[0] * s.tp_sp * s.cp
+        full = rel_bias(S, S, hidden.device)
+        if s.cp > 1:
+            rows = natural_rows(S, s.cp, 1, c.cp_idx, 0, hidden.device)
+            full = full[:, rows, :]
+        return full
     tp = group_size(groups.tp_group)
     S = hidden.shape[0] * tp * s.cp
     hl = attention.heads_local
@@ -187,8 +192,6 @@ class GalvatronDecoderLayerX(nn.Module):
         self.layer_idx = layer_idx
         s = groups.strategy
         self.strategy = s
-        assert s.cp == 1 or not s.use_ulysses, \
-            "t5: ulysses composed with ring-CP is a v2 item"
         seq_par = not s.use_ulysses
         self.input_norm = build_norm(margs.normalization, margs.hidden_size,
                                      margs.norm_epsilon, dtype)

@@ -91,13 +91,19 @@ class DistributedAttention(torch.nn.Module):
             k = all_to_all(k, self.sp_group, scatter_dim=2, gather_dim=1)
             v = all_to_all(v, self.sp_group, scatter_dim=2, gather_dim=1)
         if attn_bias is not None:
-            assert self.inner_attention is None, \
-                "relative bias + ring-CP inner attention is a v2 item"
+            # slice this rank's post-a2a head chunk; with a ring inner the
+            # rows are the cp-local zigzag pair and the ring handles the
+            # per-step kv column slices itself
             r = group_rank(self.sp_group)
             hl = q.shape[2]
-            o = eager_bias_attention(q, k, v,
-                                     attn_bias[r * hl:(r + 1) * hl],
-                                     causal, softmax_scale)
+            bias_l = attn_bias[r * hl:(r + 1) * hl]
+            if self.inner_attention is not None:
+                o = self.inner_attention(q, k, v, causal=causal,
+                                         softmax_scale=softmax_scale,
+                                         attn_bias=bias_l)
+            else:
+                o = eager_bias_attention(q, k, v, bias_l, causal,
+                                         softmax_scale)
         elif self.inner_attention is not None:
             o = self.inner_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
         else:

@@ -207,3 +207,22 @@ def test_t5_world4_pp2_tp2():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_world4_ulysses2_cp2():
+    """ulysses sp=2 composed with ring-CP cp=2 on T5: post-a2a head
+    chunks carry the bias into the ring inner attention."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=4, pp=1, tp=2, use_sp=True, cp=2,
+        dp_type="ddp", global_bsz=4, chunks=1)
+    plan.vcp = 2
+    res = run_distributed(_t5_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
