@@ -11,8 +11,9 @@ to the FULL sequence (replicated within the dp group: cross-attention
 then needs no per-layer redistribution however the decoder layers are
 laid out), stashes it in the batch context, and emits the decoder
 embedding.  pp>1: the engine rides the memory along the p2p boundary
-(engine.py _fwd_step).  v1 scope: cross-attention under tp/dp/zero/ckpt/pp
-(ulysses/cp on decoder layers are asserted off by the builder).
+(engine.py _fwd_step).  Scope: tp / ulysses / ring-CP / ulysses x CP /
+pp all compose on both stacks (relative bias follows each layout:
+tp head slices, post-a2a chunks, cp-local zigzag rows).
 
 Position encoding: T5 bucketized relative-position bias
 (transformer/relative_bias.py) on an eager fp32-softmax attention path;

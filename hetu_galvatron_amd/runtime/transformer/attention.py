@@ -83,8 +83,10 @@ class SelfAttention(nn.Module):
                 attn_bias: torch.Tensor = None) -> torch.Tensor:
         """x: [s_local, b, h] SBH; cos/sin: tables matching the seq layout at
         RoPE time (megatron-sp: cp-local full seq; ulysses: the local slice).
-        attn_bias: [h_local, sq, skv] additive scores bias (t5 relative
-        bias) — eager fp32 softmax path, local/megatron-tp layouts only."""
+        attn_bias: additive scores bias (t5 relative bias) — eager fp32
+        softmax path; megatron-tp gets [h_local, sq, skv], ulysses the
+        full-head table (sliced post-a2a), ring-CP the cp-local rows
+        over the global kv axis."""
         qkv = self.linear_qkv(x)  # [s, b, (hq_l + 2*hkv_l)*d]
         s, b = qkv.shape[0], qkv.shape[1]
         qkv = qkv.view(s, b, self.num_groups_local, self.q_per_group + 2,
