@@ -294,3 +294,32 @@ def test_moe_world4_pp2_ep2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_moe_etp2_sequential_experts():
+    """etp with SequentialMLP experts (per-expert nn.Linear fc1/fc2
+    sliced by the gated-halves rule)."""
+    from tests.utils import run_distributed
+    import copy
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import (
+        canonical_state_from_stage)
+    cfg = make_cfg({"model": {"moe_grouped_gemm": False}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    state = canonical_state_from_stage(model.stage_model)
+    path = "/tmp/galvatron_moe_test/tiny_moe_seq_state.pt"
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    torch.save(state, path)
+    base_losses = train_steps(model, cfg)
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=2, pp=1, tp=2, dp_type="ddp",
+        global_bsz=4, chunks=1)
+    res = run_distributed(
+        _moe_worker, world_size=2,
+        args=(plan.to_config_dict(), path,
+              {"model": {"moe_grouped_gemm": False}}))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
