@@ -56,3 +56,28 @@ def test_slot_reuse_and_pool_limit():
     assert eng.n_active == 1
     eng.step()
     assert eng.n_active == 0 and len(eng.outputs[r1]) == 2
+
+
+def test_http_serve_endpoint():
+    """FastAPI /generate over the continuous-batching engine (TestClient;
+    concurrent requests share decode steps)."""
+    import threading
+    from fastapi.testclient import TestClient
+    from hetu_galvatron_amd.cli.serve import build_app
+
+    model = make_model()
+    eng = ContinuousBatchingEngine(model, max_slots=2, max_seq=64)
+    app = build_app(eng, threading.Lock())
+    gen = GalvatronGenerator(model, max_batch=1, max_seq=64)
+    torch.manual_seed(8)
+    prompt = torch.randint(0, 512, (6,))
+    want = gen.generate(prompt.unsqueeze(0), max_new_tokens=4,
+                        temperature=0.0)[0, 6:].tolist()
+    with TestClient(app) as client:
+        h = client.get("/health").json()
+        assert h["status"] == "ok" and h["free_slots"] == 2
+        r = client.post("/generate", json={"prompt_ids": prompt.tolist(),
+                                           "max_new_tokens": 4})
+        assert r.status_code == 200
+        body = r.json()
+        assert body["tokens"] == want
