@@ -99,8 +99,9 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         num_decoder_layers=24, num_attention_heads=32,
         num_key_value_heads=None, kv_channels=128, ffn_hidden_size=16384,
         vocab_size=32128, max_position_embeddings=1024, seq_length=512,
-        encoder_seq_length=512, hidden_act="gelu",
-        normalization="layernorm", norm_epsilon=1e-6,
+        encoder_seq_length=512, hidden_act="geglu",
+        normalization="rmsnorm", norm_epsilon=1e-6,
+        attention_softmax_scale=1.0,
         position_embedding_type="relative", add_bias_linear=False,
         add_qkv_bias=False,
     ),
@@ -109,8 +110,9 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         num_decoder_layers=2, num_attention_heads=2,
         num_key_value_heads=None, kv_channels=64, ffn_hidden_size=256,
         vocab_size=512, max_position_embeddings=256, seq_length=64,
-        encoder_seq_length=96, hidden_act="gelu",
-        normalization="layernorm", norm_epsilon=1e-6,
+        encoder_seq_length=96, hidden_act="geglu",
+        normalization="rmsnorm", norm_epsilon=1e-6,
+        attention_softmax_scale=1.0,
         position_embedding_type="relative", add_bias_linear=False,
         add_qkv_bias=False,
     ),

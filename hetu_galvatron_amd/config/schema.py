@@ -64,6 +64,8 @@ class ModelArgs(BaseModel):
     rope_theta: float = 500000.0
     rope_scaling: Optional[float] = None
     sliding_window: Optional[int] = None  # mistral-style windowed attention
+    # None => 1/sqrt(head_dim); t5 uses 1.0 (unscaled scores)
+    attention_softmax_scale: Optional[float] = None
     add_bias_linear: bool = False
     add_qkv_bias: bool = False
     tie_word_embeddings: bool = False

@@ -65,7 +65,9 @@ class SelfAttention(nn.Module):
             self.num_heads * self.head_dim, self.hidden_size, lin_group,
             bias=m.add_bias_linear,
             sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
-        self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
+        self.softmax_scale = m.attention_softmax_scale \
+            if getattr(m, "attention_softmax_scale", None) is not None \
+            else 1.0 / math.sqrt(self.head_dim)
         self.window = getattr(m, "sliding_window", None)
         cp = group_size(cp_group) if cp_group is not None else 1
         if self.window is not None:
@@ -176,7 +178,9 @@ class CrossAttention(nn.Module):
             self.num_heads * self.head_dim, m.hidden_size, lin_group,
             bias=m.add_bias_linear, sequence_parallel=seq_par,
             dtype=dtype)
-        self.softmax_scale = 1.0 / math.sqrt(self.head_dim)
+        self.softmax_scale = m.attention_softmax_scale \
+            if getattr(m, "attention_softmax_scale", None) is not None \
+            else 1.0 / math.sqrt(self.head_dim)
         self.heads_local = self.num_heads // tp
         self.kv_heads_local = self.num_kv_heads // tp
 

@@ -213,3 +213,45 @@ def test_mistral_sliding_window_logits_match_hf():
             x = blk(x, ctx)
         got = blk.inner.lm_head(x).permute(1, 0, 2)
     assert (got - want).abs().max() < TOL, (got - want).abs().max()
+
+
+def test_t5_logits_match_hf():
+    """Full enc-dec logit alignment vs transformers' T5 (v1.1 semantics:
+    T5LayerNorm == rmsnorm, unscaled attention, gated-gelu, per-layer
+    bias tables tied to HF's shared block-0 table)."""
+    from transformers import T5Config, T5ForConditionalGeneration
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        hf_t5_to_canonical)
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    cfg, model = build("tiny-t5")
+    m = cfg.model
+    hf = T5ForConditionalGeneration(T5Config(
+        vocab_size=m.vocab_size, d_model=m.hidden_size, d_kv=m.head_dim,
+        d_ff=m.ffn_hidden_size, num_layers=m.num_hidden_layers,
+        num_decoder_layers=m.num_decoder_layers, num_heads=m.num_attention_heads,
+        relative_attention_num_buckets=m.relative_attention_num_buckets,
+        relative_attention_max_distance=m.relative_attention_max_distance,
+        layer_norm_epsilon=m.norm_epsilon, feed_forward_proj="gated-gelu",
+        dropout_rate=0.0, tie_word_embeddings=False, is_encoder_decoder=True))
+    hf.eval()
+    can = hf_t5_to_canonical(
+        {k: v for k, v in hf.state_dict().items()}, m)
+    load_full_state(model.stage_model, can, m)
+    for blk in model.stage_model.blocks:
+        blk.inner.eval()
+    torch.manual_seed(4)
+    enc_ids = torch.randint(0, m.vocab_size, (2, m.encoder_seq_length))
+    dec_ids = torch.randint(0, m.vocab_size, (2, m.seq_length))
+    with torch.no_grad():
+        want = hf(input_ids=enc_ids, decoder_input_ids=dec_ids).logits
+        ctx = {"enc_input_ids": enc_ids, "input_ids": dec_ids,
+               "labels": dec_ids.clone(), "batch_size": 2,
+               "seq_len": m.seq_length,
+               "enc_seq_len": m.encoder_seq_length}
+        x = None
+        for blk in model.stage_model.blocks:
+            if blk.kind == "lm_head":
+                break
+            x = blk(x, ctx)
+        got = blk.inner.lm_head(x).permute(1, 0, 2)
+    assert (got - want).abs().max() < 1e-4, (got - want).abs().max()
