@@ -55,7 +55,7 @@ class ModelArgs(BaseModel):
     vocab_size: int = 128256
     max_position_embeddings: int = 8192
     seq_length: int = 4096
-    hidden_act: str = "silu"  # silu(swiglu) | gelu | geglu
+    hidden_act: str = "silu"  # silu(swiglu) | gelu | geglu | relu
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
     norm_epsilon: float = 1e-5
     position_embedding_type: str = "rope"  # rope | learned | relative

@@ -70,7 +70,8 @@ class GalvatronMoEMLP(nn.Module):
         ffn_local = ffn // self.etp
         if margs.moe_grouped_gemm:
             self.experts = GroupedMLP(n_local, margs.hidden_size, ffn_local,
-                                      dtype=dtype, gated=gated)
+                                      dtype=dtype, gated=gated,
+                                      act=margs.hidden_act)
         else:
             self.experts = SequentialMLP(n_local, margs.hidden_size, ffn_local,
                                          dtype=dtype, gated=gated)

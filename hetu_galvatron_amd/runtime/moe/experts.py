@@ -31,10 +31,12 @@ class GroupedMLP(nn.Module):
     """num_local_experts gated MLPs over an expert-sorted token buffer."""
 
     def __init__(self, num_local_experts: int, hidden_size: int,
-                 ffn_hidden: int, dtype=None, gated: bool = True):
+                 ffn_hidden: int, dtype=None, gated: bool = True,
+                 act: str = "silu"):
         super().__init__()
         self.num_local_experts = num_local_experts
         self.gated = gated
+        self.act = act
         out1 = 2 * ffn_hidden if gated else ffn_hidden
         kw = {"dtype": dtype} if dtype else {}
         self.w1 = _mark_expert(nn.Parameter(
@@ -44,6 +46,16 @@ class GroupedMLP(nn.Module):
         nn.init.normal_(self.w1, 0.0, 0.02)
         nn.init.normal_(self.w2, 0.0, 0.02)
 
+    def _act(self, h: torch.Tensor) -> torch.Tensor:
+        import torch.nn.functional as F
+        if not self.gated:
+            return F.relu(h) if self.act == "relu" \
+                else F.gelu(h, approximate="tanh")
+        if self.act == "geglu":
+            gate, up = h.chunk(2, dim=-1)
+            return F.gelu(gate, approximate="tanh") * up
+        return swiglu(h)
+
     def forward(self, x: torch.Tensor,
                 tokens_per_expert: torch.Tensor) -> torch.Tensor:
         """x [m, h] expert-sorted; tokens_per_expert [E_local]."""
@@ -51,7 +63,7 @@ class GroupedMLP(nn.Module):
         if grouped_gemm_available(x, self.w1) \
                 and grouped_gemm_available(x, self.w2) and x.shape[0] > 0:
             h = grouped_gemm(x, self.w1, sizes)
-            h = swiglu(h) if self.gated else torch.nn.functional.gelu(h)
+            h = self._act(h)
             return grouped_gemm(h, self.w2, sizes)
         outs: List[torch.Tensor] = []
         start = 0
@@ -61,7 +73,7 @@ class GroupedMLP(nn.Module):
             if m == 0:
                 continue  # unused experts: flat-grad segment stays zero
             h = xe @ self.w1[e]
-            h = swiglu(h) if self.gated else torch.nn.functional.gelu(h)
+            h = self._act(h)
             outs.append(h @ self.w2[e])
         return torch.cat(outs) if outs else x[:0]
 

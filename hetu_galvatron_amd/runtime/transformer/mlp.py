@@ -39,6 +39,8 @@ class MLP(nn.Module):
             h = F.gelu(gate, approximate="tanh") * up
         elif self.hidden_act == "gelu":
             h = F.gelu(h, approximate="tanh")
+        elif self.hidden_act == "relu":
+            h = F.relu(h)
         else:
             raise ValueError(f"unknown activation {self.hidden_act}")
         return self.fc2(h)
