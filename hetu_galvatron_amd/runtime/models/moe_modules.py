@@ -74,7 +74,8 @@ class GalvatronMoEMLP(nn.Module):
                                       act=margs.hidden_act)
         else:
             self.experts = SequentialMLP(n_local, margs.hidden_size, ffn_local,
-                                         dtype=dtype, gated=gated)
+                                         dtype=dtype, gated=gated,
+                                         act=margs.hidden_act)
         if margs.moe_shared_expert_intermediate_size:
             self.shared = SharedExpertMLP(
                 margs.hidden_size, margs.moe_shared_expert_intermediate_size,

@@ -82,9 +82,11 @@ class SequentialMLP(nn.Module):
     """Per-expert nn.Linear modules (reference moe/mlp.py:128)."""
 
     def __init__(self, num_local_experts: int, hidden_size: int,
-                 ffn_hidden: int, dtype=None, gated: bool = True):
+                 ffn_hidden: int, dtype=None, gated: bool = True,
+                 act: str = "silu"):
         super().__init__()
         self.gated = gated
+        self.act = act
         out1 = 2 * ffn_hidden if gated else ffn_hidden
         kw = {"dtype": dtype} if dtype else {}
         self.fc1 = nn.ModuleList([
@@ -106,7 +108,7 @@ class SequentialMLP(nn.Module):
             if m == 0:
                 continue
             h = self.fc1[e](xe)
-            h = swiglu(h) if self.gated else torch.nn.functional.gelu(h)
+            h = GroupedMLP._act(self, h)
             outs.append(self.fc2[e](h))
         return torch.cat(outs) if outs else x[:0]
 
