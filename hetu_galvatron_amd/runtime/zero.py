@@ -226,16 +226,24 @@ class FlatParamBlock:
         self._synced = True
         self._ensure_grad_buffer()
         g = self.flat_grad
-        # tp-replicated segments (SP norms): sum over the tp group first
+        # tp-replicated segments (SP norms): sum over the tp group first.
+        # MUST complete before the sdp reduction (or the bf16 compress
+        # snapshot) below: the tp and sdp collectives run on different
+        # communicators with no mutual ordering, and both touch slices of
+        # the same flat tensor — so the tp reduces stay async only among
+        # themselves and are drained here.
         if self.tp_group is not None and self.tp_group.size > 1:
+            tp_handles = []
             for seg in self.segments:
                 if seg.tp_replicated:
                     sl = g[seg.offset:seg.offset + seg.numel]
                     if _is_gloo(self.tp_group.group):
                         dist.all_reduce(sl, group=self.tp_group.group)
                     else:
-                        self._handles.append(dist.all_reduce(
+                        tp_handles.append(dist.all_reduce(
                             sl, group=self.tp_group.group, async_op=True))
+            for h in tp_handles:
+                h.wait()
         if self.ws == 1:
             self.grad_shard = g if self.mode == "ddp" else g[self._shard_slice()]
             return
