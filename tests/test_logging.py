@@ -1,0 +1,29 @@
+"""MetricsLogger sinks (JSONL always; tb/wandb degrade gracefully)."""
+import json
+
+from hetu_galvatron_amd.config import load_config
+from hetu_galvatron_amd.utils.logging import MetricsLogger
+
+
+def test_jsonl_sink(tmp_path):
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "logging": {"tensorboard_dir": str(tmp_path)}})
+    ml = MetricsLogger(cfg, rank=0)
+    ml.log({"loss": 1.5, "lr": 1e-4}, step=0)
+    ml.log({"loss": 1.2, "lr": 1e-4}, step=1)
+    ml.close()
+    lines = [json.loads(l) for l in
+             open(tmp_path / "metrics.jsonl").read().splitlines()]
+    assert [l["step"] for l in lines] == [0, 1]
+    assert abs(lines[1]["loss"] - 1.2) < 1e-9
+
+
+def test_nonzero_rank_disabled(tmp_path):
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "logging": {"tensorboard_dir": str(tmp_path)}})
+    ml = MetricsLogger(cfg, rank=1)
+    ml.log({"loss": 1.0}, 0)
+    ml.close()
+    assert not (tmp_path / "metrics.jsonl").exists()
