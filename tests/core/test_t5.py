@@ -226,3 +226,21 @@ def test_t5_world4_ulysses2_cp2():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_t5_pp2_ckpt_memory_boundary():
+    """Activation checkpointing on all t5 layers with the memory-carrying
+    pp cut ([3,1]): the encoder memory captured in the non-reentrant ckpt
+    closure must re-materialize correctly."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=4, world_size=2, pp=2, tp=1, dp_type="ddp",
+        checkpoint=True, global_bsz=4, chunks=2)
+    plan.pp_division = [3, 1]
+    res = run_distributed(_t5_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
