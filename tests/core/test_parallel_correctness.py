@@ -212,3 +212,12 @@ def test_world4_pp2_zero3_ckpt():
     run_case(4, HybridParallelPlan.uniform(
         N_LAYERS, 4, pp=2, tp=1, dp_type="zero3",
         checkpoint=True, global_bsz=4, chunks=2))
+
+
+@pytest.mark.distributed
+def test_world4_pp2_cp2():
+    """pp2 x cp2: zigzag-sharded boundary activations + ring attention
+    across pipeline stages."""
+    run_case(4, HybridParallelPlan.uniform(
+        N_LAYERS, 4, pp=2, cp=2, dp_type="ddp", global_bsz=4, chunks=2,
+        vtp=1))
