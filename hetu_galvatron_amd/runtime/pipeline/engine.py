@@ -186,6 +186,10 @@ class PipelineEngine:
         return (rows, b_loc, h)
 
     def _fwd_step(self, ctx: Dict, stats: StepStats, recv_act):
+        with torch.profiler.record_function("galvatron::fwd_chunk"):
+            return self._fwd_step_inner(ctx, stats, recv_act)
+
+    def _fwd_step_inner(self, ctx: Dict, stats: StepStats, recv_act):
         feed = None
         if recv_act is not None:
             recv_act = recv_act.detach().requires_grad_(True)
@@ -205,6 +209,10 @@ class PipelineEngine:
         return recv_act, out
 
     def _bwd_step(self, inp, out, grad_out, ctx: Dict, chunks: int):
+        with torch.profiler.record_function("galvatron::bwd_chunk"):
+            return self._bwd_step_inner(inp, out, grad_out, ctx, chunks)
+
+    def _bwd_step_inner(self, inp, out, grad_out, ctx: Dict, chunks: int):
         if self.sm.is_last:
             loss = self._loss_of(out, ctx, chunks)
             loss.backward()

@@ -223,6 +223,14 @@ class FlatParamBlock:
         all-reduce for replicated segments)."""
         if self._synced:
             return
+        _rf = torch.profiler.record_function("galvatron::grad_sync_start")
+        _rf.__enter__()
+        try:
+            self._start_grad_sync_impl()
+        finally:
+            _rf.__exit__(None, None, None)
+
+    def _start_grad_sync_impl(self) -> None:
         self._synced = True
         self._ensure_grad_buffer()
         g = self.flat_grad
