@@ -109,3 +109,24 @@ def test_t5_model_profiler_two_axis():
     assert abs(lt0["parameter_size"] - 10.0) < 1e-6   # 40/4
     assert abs(lt1["parameter_size"] - 15.0) < 1e-6
     assert abs(lt0["tp_activation_per_bsz_dict"]["1"] - 2.0) < 1e-6
+
+
+def test_record_function_scopes():
+    """galvatron:: tracing scopes appear in a torch.profiler capture."""
+    import torch
+    from torch.profiler import ProfilerActivity, profile
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    with profile(activities=[ProfilerActivity.CPU]) as prof:
+        model.forward_backward(next(it))
+    names = {e.key for e in prof.key_averages()}
+    assert any("galvatron::grad_sync_start" in n for n in names), names
