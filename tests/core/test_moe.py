@@ -51,7 +51,7 @@ def test_router_topk_and_aux():
     assert probs.shape == (64, cfg.model.moe_router_topk)
     assert torch.allclose(probs.float().sum(-1), torch.ones(64), atol=1e-3)
     assert idx.max() < cfg.model.num_experts
-    assert float(aux) > 0  # load-balancing loss active in training
+    assert float(aux.detach()) > 0  # load-balancing loss active in training
 
 
 def test_dispatcher_roundtrip_identity_experts():
