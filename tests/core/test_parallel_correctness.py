@@ -221,3 +221,50 @@ def test_world4_pp2_cp2():
     run_case(4, HybridParallelPlan.uniform(
         N_LAYERS, 4, pp=2, cp=2, dp_type="ddp", global_bsz=4, chunks=2,
         vtp=1))
+
+
+def get_baseline_with(cfg_extra):
+    """1-process baseline under extra model config (e.g. sliding window)."""
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import (
+        canonical_state_from_stage)
+    cfg = make_cfg(cfg_extra)
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    state = canonical_state_from_stage(model.stage_model)
+    path = "/tmp/galvatron_test/tiny_llama_win_state.pt"
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    torch.save(state, path)
+    return train_steps(model, cfg), path
+
+
+@pytest.mark.distributed
+def test_ulysses_sp2_sliding_window():
+    """Sliding window under ulysses (window semantics apply to the
+    post-a2a full-seq inner attention)."""
+    from tests.utils import run_distributed
+    cfg_extra = {"model": {"sliding_window": 48}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, tp=2, use_sp=True,
+                                      dp_type="ddp", global_bsz=4)
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_tp2_sliding_window():
+    from tests.utils import run_distributed
+    cfg_extra = {"model": {"sliding_window": 48}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, tp=2, dp_type="ddp",
+                                      global_bsz=4)
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
