@@ -88,7 +88,8 @@ def main(argv=None):
         if rsm.request_checkpoint_and_exit:
             if cfg.ckpt.save:
                 from ..runtime.checkpoint import save_distributed_checkpoint
-                save_distributed_checkpoint(model, opt, sched, cfg, i + 1)
+                save_distributed_checkpoint(model, opt, sched, cfg, i + 1,
+                                            rerun_state_machine=rsm)
             print(f"[rerun] persistent fault at iter {i}: exiting "
                   f"{rsm.exit_code}", file=sys.stderr)
             sys.exit(rsm.exit_code)
@@ -102,7 +103,8 @@ def main(argv=None):
         if (cfg.ckpt.save and cfg.ckpt.save_interval
                 and (i + 1) % cfg.ckpt.save_interval == 0):
             from ..runtime.checkpoint import save_distributed_checkpoint
-            save_distributed_checkpoint(model, opt, sched, cfg, i + 1)
+            save_distributed_checkpoint(model, opt, sched, cfg, i + 1,
+                                        rerun_state_machine=rsm)
     mlog.close()
 
     # -- model-profiler worker mode ---------------------------------------

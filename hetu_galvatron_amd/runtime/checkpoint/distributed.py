@@ -32,7 +32,8 @@ def _rank_world():
 
 
 def save_distributed_checkpoint(model, opt, sched, cfg, iteration: int,
-                                save_dir: Optional[str] = None) -> str:
+                                save_dir: Optional[str] = None,
+                                rerun_state_machine=None) -> str:
     save_dir = save_dir or cfg.ckpt.save
     rank, world = _rank_world()
     it_dir = os.path.join(save_dir, f"iter_{iteration:07d}")
@@ -58,6 +59,10 @@ def save_distributed_checkpoint(model, opt, sched, cfg, iteration: int,
                 "cuda": (torch.cuda.get_rng_state()
                          if torch.cuda.is_available() else None)},
         "world_size": world,
+        # fault-attribution state travels with the run (reference
+        # rerun_state_machine.py:871-902 persists into checkpoints)
+        "rerun_state_machine": (rerun_state_machine.state_dict()
+                                if rerun_state_machine is not None else None),
     }
     torch.save(payload, os.path.join(it_dir, f"rank_{rank:05d}.pt"))
     if rank == 0:
@@ -81,7 +86,8 @@ def latest_iteration(load_dir: str) -> Optional[int]:
 
 def load_distributed_checkpoint(model, opt, sched, cfg,
                                 load_dir: Optional[str] = None,
-                                iteration: Optional[int] = None) -> int:
+                                iteration: Optional[int] = None,
+                                rerun_state_machine=None) -> int:
     load_dir = load_dir or cfg.ckpt.load
     if iteration is None:
         iteration = latest_iteration(load_dir)
@@ -118,6 +124,9 @@ def load_distributed_checkpoint(model, opt, sched, cfg,
         opt.step_count = payload["optimizer"]["step_count"]
     if sched is not None and payload["scheduler"] is not None:
         sched.load_state_dict(payload["scheduler"])
+    if rerun_state_machine is not None \
+            and payload.get("rerun_state_machine") is not None:
+        rerun_state_machine.load_state_dict(payload["rerun_state_machine"])
     if payload["rng"]["torch"] is not None:
         torch.set_rng_state(payload["rng"]["torch"])
     if payload["rng"]["cuda"] is not None and torch.cuda.is_available():

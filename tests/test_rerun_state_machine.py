@@ -84,3 +84,32 @@ def test_disabled_passthrough():
     it = iter(range(5))
     calls = drive(rsm, it, [float("nan")])
     assert calls == 1
+
+
+def test_rerun_state_persists_in_checkpoint(tmp_path):
+    """rsm state rides the distributed checkpoint (reference persists it
+    at :871-902)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import (
+        GalvatronModel, get_optimizer_and_param_scheduler)
+    from hetu_galvatron_amd.runtime.checkpoint import (
+        load_distributed_checkpoint, save_distributed_checkpoint)
+    from hetu_galvatron_amd.runtime.rerun_state_machine import (
+        initialize_rerun_state_machine)
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    rsm = initialize_rerun_state_machine(enabled=True)
+    rsm.skipped = [3, 7]  # fabricated attribution history
+    save_distributed_checkpoint(model, opt, sched, cfg, 1, str(tmp_path),
+                                rerun_state_machine=rsm)
+    rsm2 = initialize_rerun_state_machine(enabled=True)
+    load_distributed_checkpoint(model, opt, sched, cfg, str(tmp_path),
+                                rerun_state_machine=rsm2)
+    assert rsm2.state_dict() == rsm.state_dict()
