@@ -83,8 +83,14 @@ def main(argv=None):
         ms = prof.time_end()
         prof.log_iteration(loss, sched.get_lr(), norm,
                            cfg.logging.log_interval)
-        mlog.log({"loss": loss, "lr": sched.get_lr(), "grad_norm": norm,
-                  "iter_ms": ms or 0.0}, i)
+        metrics = {"loss": loss, "lr": sched.get_lr(), "grad_norm": norm,
+                   "iter_ms": ms or 0.0}
+        if cfg.model.num_experts:
+            from ..runtime.moe import tracker as moe_tracker
+            metrics.update({f"moe/{k}": v
+                            for k, v in moe_tracker.reduce_and_get().items()})
+            moe_tracker.clear()
+        mlog.log(metrics, i)
         if rsm.request_checkpoint_and_exit:
             if cfg.ckpt.save:
                 from ..runtime.checkpoint import save_distributed_checkpoint
