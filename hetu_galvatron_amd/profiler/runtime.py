@@ -126,10 +126,11 @@ class RuntimeProfiler:
     def log_iteration(self, loss: float, lr: float, grad_norm: float,
                       interval: int = 1) -> None:
         if self.rank == 0 and self.iteration % max(interval, 1) == 0:
-            ms = self.iter_times_ms[-1] if self.iter_times_ms else float("nan")
+            ms = self.iter_times_ms[-1] if self.iter_times_ms else None
+            ms_s = f"{ms:.1f} ms" if ms is not None else "- ms"
             print(f"iter {self.iteration:5d} | loss {loss:.4f} | "
                   f"lr {lr:.3e} | grad-norm {grad_norm:.3f} | "
-                  f"{ms:.1f} ms", flush=True)
+                  f"{ms_s}", flush=True)
 
 
 def get_runtime_profiler(cfg, device=None) -> RuntimeProfiler:
