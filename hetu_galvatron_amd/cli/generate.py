@@ -26,7 +26,7 @@ def main(argv=None):
     # generate.* keys are CLI-local (not part of the schema)
     gen_args = {"max_new_tokens": 32, "temperature": 0.0, "top_k": 0,
                 "prompt_ids": "", "batch": 1, "prompt_len": 8,
-                "max_seq": 4096, "eos_id": -1}
+                "max_seq": 4096, "eos_id": -1, "use_graph": 0}
     rest = []
     for a in argv:
         if a.startswith("generate.") and "=" in a:
@@ -65,10 +65,15 @@ def main(argv=None):
                             device=device)
     gen = GalvatronGenerator(model, max_batch=ids.shape[0],
                              max_seq=gen_args["max_seq"])
-    out = gen.generate(
-        ids, max_new_tokens=gen_args["max_new_tokens"],
-        temperature=gen_args["temperature"], top_k=gen_args["top_k"],
-        eos_id=None if gen_args["eos_id"] < 0 else gen_args["eos_id"])
+    if gen_args["use_graph"]:
+        # hipGraph-captured decode step (greedy); falls back to eager off-GPU
+        out = gen.generate_graphed(
+            ids, max_new_tokens=gen_args["max_new_tokens"])
+    else:
+        out = gen.generate(
+            ids, max_new_tokens=gen_args["max_new_tokens"],
+            temperature=gen_args["temperature"], top_k=gen_args["top_k"],
+            eos_id=None if gen_args["eos_id"] < 0 else gen_args["eos_id"])
     for row in out.tolist():
         print(" ".join(str(t) for t in row))
     return out

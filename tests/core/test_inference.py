@@ -177,3 +177,15 @@ def test_tp_decode_matches_single(tmp_path):
     res = run_distributed(_tp_gen_worker, world_size=2,
                           args=(path, ids.tolist(), want))
     assert all(res)
+
+
+def test_generate_cli_graph_flag():
+    from hetu_galvatron_amd.cli.generate import main
+    out = main(["model.model_name=tiny-llama",
+                "parallel.mixed_precision=fp32",
+                "generate.max_new_tokens=3", "generate.prompt_ids=5,6,7",
+                "generate.use_graph=1"])
+    want = main(["model.model_name=tiny-llama",
+                 "parallel.mixed_precision=fp32",
+                 "generate.max_new_tokens=3", "generate.prompt_ids=5,6,7"])
+    assert out.tolist() == want.tolist()
