@@ -17,7 +17,9 @@ from torch.utils.cpp_extension import (  # noqa: E402
     BuildExtension, CppExtension, CUDAExtension)
 
 HERE = os.path.dirname(os.path.abspath(__file__))
-SRC = sorted(glob(os.path.join(HERE, "hetu_galvatron_amd/ops/csrc/*.hip")))
+SRC = sorted(f for f in glob(os.path.join(HERE,
+             "hetu_galvatron_amd/ops/csrc/*.hip"))
+             if not f.endswith("_hip.hip"))  # torch hipify artifacts
 
 setup(
     name="hetu_galvatron_amd_ext",
