@@ -27,3 +27,17 @@ def test_nonzero_rank_disabled(tmp_path):
     ml.log({"loss": 1.0}, 0)
     ml.close()
     assert not (tmp_path / "metrics.jsonl").exists()
+
+
+def test_global_memory_buffer_reuse():
+    import torch
+    from hetu_galvatron_amd.core.parallel_state import GlobalMemoryBuffer
+    b = GlobalMemoryBuffer()
+    t1 = b.get_tensor((4, 8), torch.float32, "ws")
+    ptr1 = t1.data_ptr()
+    t2 = b.get_tensor((2, 8), torch.float32, "ws")   # smaller: same storage
+    assert t2.data_ptr() == ptr1 and t2.shape == (2, 8)
+    t3 = b.get_tensor((16, 8), torch.float32, "ws")  # larger: grows
+    assert t3.shape == (16, 8)
+    t4 = b.get_tensor((4, 4), torch.float16, "other")
+    assert t4.dtype == torch.float16
