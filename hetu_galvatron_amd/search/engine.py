@@ -243,7 +243,6 @@ class SearchEngine:
         if not strategies:
             return None
         S = len(strategies)
-        division = self._even_division(pp)
 
         budget_mb = self.args.memory_constraint * 1024.0
         budget_mb -= 2048.0  # allocator/runtime context reserve
@@ -302,7 +301,8 @@ class SearchEngine:
             inter.append(it)
 
         best: Optional[SearchResult] = None
-        for vtp in vtp_opts:
+        for division in self._division_candidates(pp):
+          for vtp in vtp_opts:
             budget_units = int((budget_mb - other_mem[vtp]) // self.mem_unit)
             if budget_units <= 0:
                 continue
@@ -351,6 +351,37 @@ class SearchEngine:
                 best = SearchResult(thr, plan, t_ms, global_bsz, chunks, pp,
                                     vtp)
         return best
+
+    def _division_candidates(self, pp: int) -> List[List[int]]:
+        """Even split + memory-balanced variants that relieve the edge
+        stages (stage 0 carries the embedding's states/activations, the
+        last stage the lm head) by shifting one layer inward
+        (reference: memory-balanced pp division, search_engine.py:954)."""
+        even = self._even_division(pp)
+        if pp <= 1 or self.num_layers <= pp:
+            return [even]
+        cands = [even]
+
+        def shift(frm: int, to: int, base: List[int]) -> None:
+            d = list(base)
+            if d[frm] > 1:
+                d[frm] -= 1
+                d[to] += 1
+                if d not in cands:
+                    cands.append(d)
+
+        shift(0, 1, even)
+        shift(pp - 1, pp - 2, even)
+        if pp > 2:
+            d2 = list(even)
+            if d2[0] > 1 and d2[-1] > 1:
+                d2[0] -= 1
+                d2[-1] -= 1
+                d2[1] += 1
+                d2[-2] += 1
+                if d2 not in cands:
+                    cands.append(d2)
+        return cands
 
     def _even_division(self, pp: int) -> List[int]:
         base = self.num_layers // pp

@@ -356,3 +356,17 @@ def test_two_node_mocked_search(tmp_path):
     assert best is not None and best.throughput > 0
     best.plan.validate(16)
     assert best.plan.layer(0, 16).degree_product() == 16
+
+
+def test_division_candidates():
+    cfg = load_config(base={"model": {"model_name": "llama-3-8b"},
+                            "search": {"num_gpus_per_node": 8}})
+    eng = SearchEngine(cfg)
+    cands = eng._division_candidates(4)   # 32 layers / pp4
+    assert [8, 8, 8, 8] in cands
+    assert [7, 9, 8, 8] in cands          # relieve the embedding stage
+    assert [8, 8, 9, 7] in cands          # relieve the head stage
+    assert [7, 9, 9, 7] in cands
+    assert eng._division_candidates(1) == [[32]]
+    for d in cands:
+        assert sum(d) == 32 and len(d) == 4
