@@ -370,3 +370,13 @@ def test_division_candidates():
     assert eng._division_candidates(1) == [[32]]
     for d in cands:
         assert sum(d) == 32 and len(d) == 4
+
+
+def test_plan_report_tool():
+    import glob
+    import subprocess
+    import sys
+    p = sorted(glob.glob("examples/configs/galvatron_config_llama3-8b*270GB*"))[0]
+    out = subprocess.run([sys.executable, "tools/plan_report.py", p],
+                         capture_output=True, text=True, check=True).stdout
+    assert "dp8" in out and "pp=1" in out
