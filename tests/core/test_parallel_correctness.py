@@ -268,3 +268,21 @@ def test_tp2_sliding_window():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+@pytest.mark.slow
+def test_world8_dp8_zero2():
+    """The round-end scale-bench topology (dp8 zero2, bf16-compressed
+    reduction) at world 8 over gloo."""
+    from tests.utils import run_distributed
+    cfg_extra = {"train": {"global_train_batch_size": 8}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 8, dp_type="zero2",
+                                      global_bsz=8)
+    res = run_distributed(_dist_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
