@@ -232,7 +232,8 @@ def get_baseline_with(cfg_extra):
     torch.manual_seed(0)
     model = GalvatronModel(cfg)
     state = canonical_state_from_stage(model.stage_model)
-    path = "/tmp/galvatron_test/tiny_llama_win_state.pt"
+    tag = abs(hash(str(sorted(str(cfg_extra))))) % 10**8
+    path = f"/tmp/galvatron_test/tiny_llama_state_{tag}.pt"
     os.makedirs(os.path.dirname(path), exist_ok=True)
     torch.save(state, path)
     return train_steps(model, cfg), path
