@@ -317,6 +317,11 @@ class FlatParamBlock:
         model element counted exactly once across the world):
           * zero2/3: the reduced shard; ddp: the rank's virtual shard
           * tp-replicated segments scaled by 1/tp (they repeat per tp rank)
+        Known approximation: params replicated across tp WITHOUT the
+        tp_replicated tag (the MoE router under etp — identical grads on
+        every tp rank by construction) are counted tp times; they are a
+        vanishing fraction of the norm and the count is identical on all
+        ranks, so clipping stays globally consistent.
         """
         assert self.grad_shard is not None
         if self.mode == "ddp":
