@@ -166,7 +166,9 @@ class ContinuousBatchingEngine:
         return out
 
     def release(self, rid: int) -> None:
-        slot = self.slot_of.pop(rid)
+        slot = self.slot_of.pop(rid, None)
+        if slot is None:
+            return  # idempotent: already released
         self.rid_of.pop(slot, None)
         self.budgets.pop(slot, None)
         self.last_tok.pop(slot, None)

@@ -81,3 +81,12 @@ def test_http_serve_endpoint():
         assert r.status_code == 200
         body = r.json()
         assert body["tokens"] == want
+
+
+def test_release_idempotent():
+    model = make_model()
+    eng = ContinuousBatchingEngine(model, max_slots=1, max_seq=64)
+    rid = eng.add_request(torch.randint(0, 512, (3,)), 2)
+    eng.release(rid)
+    eng.release(rid)  # no-op
+    assert eng.n_active == 0 and len(eng.free) == 1
