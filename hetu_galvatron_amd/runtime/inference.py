@@ -461,8 +461,11 @@ class GalvatronTPGenerator:
         return torch.cat(parts, dim=-1)[0].float()  # [b, V]
 
     @torch.no_grad()
-    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32
-                 ) -> torch.Tensor:
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, seed: int = 0) -> torch.Tensor:
+        """Greedy, or temperature sampling with a seed-shared generator —
+        every tp rank draws the identical sample from the identical
+        gathered logits, keeping ranks in lockstep."""
         b, sp = input_ids.shape
         assert sp + max_new_tokens <= self.max_seq
         m = self.margs
@@ -475,10 +478,17 @@ class GalvatronTPGenerator:
                 blk.flat.gather_params()
         for blk in self.blocks:
             blk.inner.eval()
+        gen = None
+        if temperature > 0:
+            gen = torch.Generator(device="cpu").manual_seed(seed)
         logits = self._forward_tokens(input_ids, cache)
         out = [input_ids]
         for _ in range(max_new_tokens):
-            nxt = logits.argmax(-1)
+            if temperature > 0:
+                probs = torch.softmax(logits.cpu() / temperature, dim=-1)
+                nxt = torch.multinomial(probs, 1, generator=gen)                     .squeeze(-1).to(logits.device)
+            else:
+                nxt = logits.argmax(-1)
             out.append(nxt.unsqueeze(1))
             if len(out) - 1 < max_new_tokens:
                 logits = self._forward_tokens(nxt.unsqueeze(1), cache)

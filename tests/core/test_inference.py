@@ -189,3 +189,45 @@ def test_generate_cli_graph_flag():
                  "parallel.mixed_precision=fp32",
                  "generate.max_new_tokens=3", "generate.prompt_ids=5,6,7"])
     assert out.tolist() == want.tolist()
+
+
+def _tp_sample_worker(rank, world, state_path):
+    import torch
+    from hetu_galvatron_amd.config import HybridParallelPlan
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    from hetu_galvatron_amd.runtime.inference import GalvatronTPGenerator
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.uniform(num_layers=2, world_size=2, pp=1,
+                                      tp=2, dp_type="ddp", global_bsz=2,
+                                      vtp=2)
+    model = GalvatronModel(cfg, plan)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    gen = GalvatronTPGenerator(model, max_batch=1, max_seq=64)
+    torch.manual_seed(100 + rank)  # rank-divergent ambient RNG on purpose
+    out = gen.generate(torch.tensor([[5, 6, 7]]), max_new_tokens=5,
+                       temperature=0.9, seed=42)
+    return out.tolist()
+
+
+@pytest.mark.distributed
+def test_tp_decode_sampling_lockstep(tmp_path):
+    """Temperature sampling stays identical across tp ranks (seed-shared
+    generator), even with divergent ambient RNG state."""
+    from hetu_galvatron_amd.runtime.checkpoint.state import (
+        canonical_state_from_stage)
+    from tests.utils import run_distributed
+    model = make_model()
+    state = canonical_state_from_stage(model.stage_model)
+    path = str(tmp_path / "state.pt")
+    torch.save(state, path)
+    res = run_distributed(_tp_sample_worker, world_size=2, args=(path,))
+    assert res[0] == res[1]
