@@ -54,6 +54,7 @@ class ContinuousBatchingEngine:
         self.slot_of: Dict[int, int] = {}       # request id -> slot
         self.rid_of: Dict[int, int] = {}        # slot -> request id
         self.outputs: Dict[int, List[int]] = {} # request id -> tokens
+        self.samplers: Dict[int, tuple] = {}    # slot -> (temp, generator)
         self._next_rid = 0
         for blk in self.gen.blocks:
             if blk.flat is not None:
@@ -61,11 +62,12 @@ class ContinuousBatchingEngine:
             blk.inner.eval()
 
     # -- request lifecycle --------------------------------------------------
-    def add_request(self, prompt_ids: torch.Tensor, max_new_tokens: int
-                    ) -> int:
+    def add_request(self, prompt_ids: torch.Tensor, max_new_tokens: int,
+                    temperature: float = 0.0, seed: int = 0) -> int:
         """prompt_ids: [s]; prefills a slot, producing the request's
         FIRST generated token immediately.  Returns a request id whose
-        tokens accumulate in `self.outputs[rid]` (kept after release)."""
+        tokens accumulate in `self.outputs[rid]` (kept after release).
+        temperature > 0 samples with a per-request seeded generator."""
         assert self.free, "no free slots"
         assert prompt_ids.dim() == 1 and max_new_tokens >= 1
         slot = self.free.pop()
@@ -76,7 +78,13 @@ class ContinuousBatchingEngine:
         # single-slot prefill on the slot's contiguous cache row
         view = _SlotCache(self.cache, slot)
         logits = self._prefill(prompt_ids.unsqueeze(0), view)
-        first = int(logits.argmax(-1))
+        if temperature > 0:
+            g = torch.Generator(device="cpu").manual_seed(seed)
+            self.samplers[slot] = (temperature, g)
+            first = self._sample(logits[0], slot)
+        else:
+            self.samplers.pop(slot, None)
+            first = int(logits.argmax(-1))
         self.outputs[rid] = [first]
         self.slot_of[rid] = slot
         self.rid_of[slot] = rid
@@ -151,10 +159,11 @@ class ContinuousBatchingEngine:
             h = residual + layer.mlp(layer.post_attn_norm(h))
         h = gen.final_norm.norm(h)
         logits = gen.lm_head.lm_head(h)[0].float()          # [n, V]
-        toks = logits.argmax(-1)
+        greedy = logits.argmax(-1)
         out: Dict[int, int] = {}
         for i, s in enumerate(active):
-            t = int(toks[i])
+            t = self._sample(logits[i], s) if s in self.samplers \
+                else int(greedy[i])
             rid = self.rid_of[s]
             out[rid] = t
             self.outputs[rid].append(t)
@@ -165,10 +174,16 @@ class ContinuousBatchingEngine:
                 self.release(rid)
         return out
 
+    def _sample(self, logits_row: torch.Tensor, slot: int) -> int:
+        temp, g = self.samplers[slot]
+        probs = torch.softmax(logits_row.float().cpu() / temp, dim=-1)
+        return int(torch.multinomial(probs, 1, generator=g))
+
     def release(self, rid: int) -> None:
         slot = self.slot_of.pop(rid, None)
         if slot is None:
             return  # idempotent: already released
+        self.samplers.pop(slot, None)
         self.rid_of.pop(slot, None)
         self.budgets.pop(slot, None)
         self.last_tok.pop(slot, None)

@@ -90,3 +90,22 @@ def test_release_idempotent():
     eng.release(rid)
     eng.release(rid)  # no-op
     assert eng.n_active == 0 and len(eng.free) == 1
+
+
+def test_per_request_sampling_deterministic():
+    """Per-request seeded sampling: same seed -> same tokens, regardless
+    of what else shares the batch."""
+    model = make_model()
+    torch.manual_seed(11)
+    prompt = torch.randint(0, 512, (5,))
+    eng1 = ContinuousBatchingEngine(model, max_slots=2, max_seq=64)
+    r = eng1.add_request(prompt, 4, temperature=0.8, seed=7)
+    while eng1.n_active:
+        eng1.step()
+    alone = eng1.outputs[r]
+    eng2 = ContinuousBatchingEngine(model, max_slots=2, max_seq=64)
+    r1 = eng2.add_request(prompt, 4, temperature=0.8, seed=7)
+    r2 = eng2.add_request(torch.randint(0, 512, (9,)), 6)  # greedy neighbor
+    while eng2.n_active:
+        eng2.step()
+    assert eng2.outputs[r1] == alone
