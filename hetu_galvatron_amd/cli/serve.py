@@ -49,10 +49,13 @@ def build_app(engine, lock: threading.Lock):
         ids = torch.tensor(payload["prompt_ids"], dtype=torch.long,
                            device=engine.gen._dev)
         max_new = int(payload.get("max_new_tokens", 32))
+        temperature = float(payload.get("temperature", 0.0))
+        seed = int(payload.get("seed", 0))
         with lock:
             if not engine.free:
                 raise HTTPException(503, "no free slots")
-            rid = engine.add_request(ids, max_new)
+            rid = engine.add_request(ids, max_new, temperature=temperature,
+                                     seed=seed)
             ev = threading.Event()
             if rid not in engine.slot_of:   # finished at prefill
                 ev.set()

@@ -109,3 +109,20 @@ def test_per_request_sampling_deterministic():
     while eng2.n_active:
         eng2.step()
     assert eng2.outputs[r1] == alone
+
+
+def test_http_sampling_params():
+    import threading
+    from fastapi.testclient import TestClient
+    from hetu_galvatron_amd.cli.serve import build_app
+    model = make_model()
+    eng = ContinuousBatchingEngine(model, max_slots=1, max_seq=64)
+    app = build_app(eng, threading.Lock())
+    with TestClient(app) as client:
+        b1 = client.post("/generate", json={
+            "prompt_ids": [3, 4, 5], "max_new_tokens": 4,
+            "temperature": 0.9, "seed": 11}).json()
+        b2 = client.post("/generate", json={
+            "prompt_ids": [3, 4, 5], "max_new_tokens": 4,
+            "temperature": 0.9, "seed": 11}).json()
+        assert b1["tokens"] == b2["tokens"]
