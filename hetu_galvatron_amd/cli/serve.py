@@ -54,8 +54,11 @@ def build_app(engine, lock: threading.Lock):
         with lock:
             if not engine.free:
                 raise HTTPException(503, "no free slots")
+            eos = payload.get("eos_id")
             rid = engine.add_request(ids, max_new, temperature=temperature,
-                                     seed=seed)
+                                     seed=seed,
+                                     eos_id=int(eos) if eos is not None
+                                     else None)
             ev = threading.Event()
             if rid not in engine.slot_of:   # finished at prefill
                 ev.set()

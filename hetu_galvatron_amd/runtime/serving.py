@@ -55,6 +55,7 @@ class ContinuousBatchingEngine:
         self.rid_of: Dict[int, int] = {}        # slot -> request id
         self.outputs: Dict[int, List[int]] = {} # request id -> tokens
         self.samplers: Dict[int, tuple] = {}    # slot -> (temp, generator)
+        self.eos: Dict[int, int] = {}           # slot -> eos id (or None)
         self._next_rid = 0
         for blk in self.gen.blocks:
             if blk.flat is not None:
@@ -63,7 +64,8 @@ class ContinuousBatchingEngine:
 
     # -- request lifecycle --------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor, max_new_tokens: int,
-                    temperature: float = 0.0, seed: int = 0) -> int:
+                    temperature: float = 0.0, seed: int = 0,
+                    eos_id: int = None) -> int:
         """prompt_ids: [s]; prefills a slot, producing the request's
         FIRST generated token immediately.  Returns a request id whose
         tokens accumulate in `self.outputs[rid]` (kept after release).
@@ -90,8 +92,10 @@ class ContinuousBatchingEngine:
         self.rid_of[slot] = rid
         self.lengths[slot] = sp
         self.last_tok[slot] = first
+        self.eos[slot] = eos_id
         self.budgets[slot] = max_new_tokens - 1
-        if self.budgets[slot] == 0:
+        if self.budgets[slot] == 0 or (eos_id is not None
+                                       and first == eos_id):
             self.release(rid)
         return rid
 
@@ -170,7 +174,8 @@ class ContinuousBatchingEngine:
             self.lengths[s] += 1
             self.last_tok[s] = t
             self.budgets[s] -= 1
-            if self.budgets[s] == 0:
+            if self.budgets[s] == 0 or (self.eos.get(s) is not None
+                                        and t == self.eos[s]):
                 self.release(rid)
         return out
 
@@ -184,6 +189,7 @@ class ContinuousBatchingEngine:
         if slot is None:
             return  # idempotent: already released
         self.samplers.pop(slot, None)
+        self.eos.pop(slot, None)
         self.rid_of.pop(slot, None)
         self.budgets.pop(slot, None)
         self.last_tok.pop(slot, None)

@@ -126,3 +126,23 @@ def test_http_sampling_params():
             "prompt_ids": [3, 4, 5], "max_new_tokens": 4,
             "temperature": 0.9, "seed": 11}).json()
         assert b1["tokens"] == b2["tokens"]
+
+
+def test_eos_early_release():
+    model = make_model()
+    eng = ContinuousBatchingEngine(model, max_slots=1, max_seq=64)
+    torch.manual_seed(13)
+    prompt = torch.randint(0, 512, (5,))
+    rid0 = eng.add_request(prompt, 8)
+    while eng.n_active:
+        eng.step()
+    toks = eng.outputs[rid0]
+    # re-run with eos set to the 3rd generated token: must stop there
+    eos = toks[2]
+    rid = eng.add_request(prompt, 8, eos_id=eos)
+    steps = 0
+    while eng.n_active:
+        eng.step()
+        steps += 1
+        assert steps < 10
+    assert eng.outputs[rid] == toks[:3]
