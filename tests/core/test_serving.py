@@ -137,7 +137,8 @@ def test_eos_early_release():
     while eng.n_active:
         eng.step()
     toks = eng.outputs[rid0]
-    # re-run with eos set to the 3rd generated token: must stop there
+    # re-run with eos set to the 3rd generated token: generation must
+    # stop at the FIRST occurrence of that token
     eos = toks[2]
     rid = eng.add_request(prompt, 8, eos_id=eos)
     steps = 0
@@ -145,4 +146,4 @@ def test_eos_early_release():
         eng.step()
         steps += 1
         assert steps < 10
-    assert eng.outputs[rid] == toks[:3]
+    assert eng.outputs[rid] == toks[:toks.index(eos) + 1]
