@@ -323,3 +323,19 @@ def test_moe_etp2_sequential_experts():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_moe_world4_ulysses2_ep2():
+    """ulysses sp=2 composed with ep=2: each sp rank routes its sequence
+    shard through its own ep plane (ep groups keyed by sp index)."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=4, pp=1, tp=2, use_sp=True,
+        dp_type="ddp", global_bsz=4, chunks=1, ep=2)
+    res = run_distributed(_moe_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
