@@ -86,3 +86,13 @@ def test_train_cli_with_rampup(tmp_path):
           "train.train_iters=6", "train.lr=1e-4",
           "train.lr_decay_style=constant",
           "train.distributed_backend=gloo"])
+
+
+def test_train_cli_t5(tmp_path):
+    """t5 through the train CLI (enc-dec batches, relative bias)."""
+    from hetu_galvatron_amd.cli.train import main
+    main(["model.model_name=tiny-t5",
+          "parallel.mixed_precision=fp32",
+          "train.global_train_batch_size=2", "train.train_iters=2",
+          "train.lr=1e-4", "train.lr_decay_style=constant",
+          "train.distributed_backend=gloo"])
