@@ -130,3 +130,22 @@ def test_record_function_scopes():
         model.forward_backward(next(it))
     names = {e.key for e in prof.key_averages()}
     assert any("galvatron::grad_sync_start" in n for n in names), names
+
+
+def test_train_cli_profile_mode_t5_keys(tmp_path):
+    """profile.profile=1 computation run writes t5 'layernum[enc,dec]'
+    keys (consumed by T5ModelProfiler.process_t5_data)."""
+    import json
+    import os
+    from hetu_galvatron_amd.cli.train import main
+    main(["model.model_name=tiny-t5",
+          "parallel.mixed_precision=fp32",
+          "model.num_hidden_layers=1", "model.num_decoder_layers=1",
+          "train.global_train_batch_size=2", "train.train_iters=2",
+          "train.lr=1e-4", "train.lr_decay_style=constant",
+          "train.distributed_backend=gloo",
+          "profile.profile=1", "profile.profile_type=computation",
+          f"profile.profile_dir={tmp_path}"])
+    path = os.path.join(tmp_path, "computation_profiling_fp32_tiny-t5.json")
+    d = json.load(open(path))
+    assert any(k.startswith("layernum[1,1]_bsz2_seq") for k in d), d
