@@ -206,3 +206,34 @@ def test_hf_t5_roundtrip_and_load():
     l1 = model.forward_backward(batch).loss
     l2 = model2.forward_backward(batch).loss
     assert abs(l1 - l2) < 1e-5
+
+
+def test_convert_checkpoint_cli_roundtrip(tmp_path):
+    """h2g -> g2h through the CLI; hf dir round-trips bit-exact."""
+    import subprocess
+    import sys
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_llama, load_hf_checkpoint, save_hf_checkpoint)
+
+    cfg = tiny_cfg()
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    can = canonical_state_from_stage(model.stage_model)
+    hf_dir = tmp_path / "hf_in"
+    save_hf_checkpoint(canonical_to_hf_llama(can, cfg.model), str(hf_dir))
+
+    can_path = tmp_path / "canonical.pt"
+    out_dir = tmp_path / "hf_out"
+    base = [sys.executable, "-m", "hetu_galvatron_amd.cli.convert_checkpoint"]
+    ov = ["model.model_name=tiny-llama", "parallel.mixed_precision=fp32"]
+    subprocess.run(base + ["h2g", "--hf-dir", str(hf_dir),
+                           "--out", str(can_path)] + ov, check=True,
+                   capture_output=True, text=True, timeout=300)
+    subprocess.run(base + ["g2h", "--canonical", str(can_path),
+                           "--out-dir", str(out_dir)] + ov, check=True,
+                   capture_output=True, text=True, timeout=300)
+    a = load_hf_checkpoint(str(hf_dir))
+    b = load_hf_checkpoint(str(out_dir))
+    assert set(a) == set(b)
+    for k in a:
+        assert torch.equal(a[k], b[k]), k
