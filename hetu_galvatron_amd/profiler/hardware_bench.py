@@ -126,22 +126,47 @@ def bench_p2p(args, rank, world) -> Dict[str, float]:
 
 
 def bench_sp_time(args, rank, world) -> Dict[str, float]:
-    """allreduce + all2all latency vs message size (reference sp_time_*)."""
+    """allreduce + all2all latency vs message size, PER SUB-GROUP SIZE
+    (the cost model needs latency curves for every tp/sp degree —
+    reference sp_time_* sweeps sizes, not just the full world)."""
     out = {}
     dev = _dev()
-    mb = args.start_mb
-    group = dist.group.WORLD
-    while mb <= args.end_mb:
-        numel = mb * 1024 * 1024 // 2
-        buf = torch.randn(numel, dtype=torch.bfloat16, device=dev)
-        t = _time_op(lambda: dist.all_reduce(buf, group=group),
-                     args.warmup_iters, args.measure_iters)
-        out[f"allreduce_size_{world}_{mb}MB_time"] = round(t * 1000, 5)
-        obuf = torch.empty_like(buf)
-        t = _time_op(lambda: dist.all_to_all_single(obuf, buf, group=group),
-                     args.warmup_iters, args.measure_iters)
-        out[f"all2all_size_{world}_{mb}MB_time"] = round(t * 1000, 5)
-        mb *= 2
+    sizes = [s for s in (2, 4, 8, 16) if s <= world] or [world]
+    if world not in sizes and world > 1:
+        sizes.append(world)
+    for size in sizes:
+        groups = _groups_for(world, size, True)
+        my = None
+        for ranks in groups:
+            g = dist.new_group(ranks)
+            if rank in ranks:
+                my = g
+        dist.barrier()
+        mb = args.start_mb
+        while mb <= args.end_mb:
+            numel = mb * 1024 * 1024 // 2
+            buf = torch.randn(numel, dtype=torch.bfloat16, device=dev)
+            obuf = torch.empty_like(buf)
+            t_ar = _time_op(lambda: dist.all_reduce(buf, group=my),
+                            args.warmup_iters, args.measure_iters) \
+                if my is not None else 0.0
+            try:
+                t_a2a = _time_op(
+                    lambda: dist.all_to_all_single(obuf, buf, group=my),
+                    args.warmup_iters, args.measure_iters) \
+                    if my is not None else 0.0
+            except RuntimeError:
+                t_a2a = 0.0  # gloo: no all_to_all_single
+            tt = torch.tensor([t_ar, t_a2a], dtype=torch.float64,
+                              device=dev if dist.get_backend() == "nccl"
+                              else "cpu")
+            dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+            out[f"allreduce_size_{size}_{mb}MB_time"] = \
+                round(float(tt[0]) * 1000, 5)
+            if float(tt[1]) > 0:
+                out[f"all2all_size_{size}_{mb}MB_time"] = \
+                    round(float(tt[1]) * 1000, 5)
+            mb *= 2
     return out
 
 

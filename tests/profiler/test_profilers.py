@@ -149,3 +149,25 @@ def test_train_cli_profile_mode_t5_keys(tmp_path):
     path = os.path.join(tmp_path, "computation_profiling_fp32_tiny-t5.json")
     d = json.load(open(path))
     assert any(k.startswith("layernum[1,1]_bsz2_seq") for k in d), d
+
+
+def _sp_time_worker(rank, world):
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+    import argparse
+    from hetu_galvatron_amd.profiler.hardware_bench import bench_sp_time
+    args = argparse.Namespace(start_mb=1, end_mb=2, warmup_iters=1,
+                              measure_iters=2)
+    out = bench_sp_time(args, rank, world)
+    dist.barrier()
+    return out
+
+
+def test_sp_time_worker_per_size_keys():
+    """sp_time sweeps per sub-group size (cost-model input curves)."""
+    from tests.utils import run_distributed
+    res = run_distributed(_sp_time_worker, world_size=2)
+    keys = set(res[0])
+    assert "allreduce_size_2_1MB_time" in keys
+    assert "allreduce_size_2_2MB_time" in keys
