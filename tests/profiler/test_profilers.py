@@ -171,3 +171,30 @@ def test_sp_time_worker_per_size_keys():
     keys = set(res[0])
     assert "allreduce_size_2_1MB_time" in keys
     assert "allreduce_size_2_2MB_time" in keys
+
+
+def _hw_worker(rank, world, op):
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+    import argparse
+    from hetu_galvatron_amd.profiler import hardware_bench as hb
+    args = argparse.Namespace(message_mb=1, warmup_iters=1, measure_iters=2,
+                              start_mb=1, end_mb=1)
+    out = {"allreduce": hb.bench_allreduce,
+           "p2p": hb.bench_p2p}[op](args, rank, world)
+    dist.barrier()
+    return out
+
+
+def test_allreduce_worker_keys():
+    from tests.utils import run_distributed
+    res = run_distributed(_hw_worker, world_size=2, args=("allreduce",))
+    assert "allreduce_size_2_consec_1" in res[0]
+    assert res[0]["allreduce_size_2_consec_1"] > 0
+
+
+def test_p2p_worker_keys():
+    from tests.utils import run_distributed
+    res = run_distributed(_hw_worker, world_size=2, args=("p2p",))
+    assert any(k.startswith("pp_size_") for k in res[0])
