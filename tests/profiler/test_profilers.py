@@ -198,3 +198,20 @@ def test_p2p_worker_keys():
     from tests.utils import run_distributed
     res = run_distributed(_hw_worker, world_size=2, args=("p2p",))
     assert any(k.startswith("pp_size_") for k in res[0])
+
+
+def _overlap_worker(rank, world):
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+    import argparse
+    from hetu_galvatron_amd.profiler.hardware_bench import bench_overlap
+    out = bench_overlap(argparse.Namespace(), rank, world)
+    dist.barrier()
+    return out
+
+
+def test_overlap_worker_cpu_default():
+    from tests.utils import run_distributed
+    res = run_distributed(_overlap_worker, world_size=2)
+    assert res[0]["overlap_coe"] >= 1.0
