@@ -355,3 +355,18 @@ def test_moe_world4_cp2_ep2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_moe_world4_pp2_etp2():
+    """pipeline x expert-TP: tp=2 within each of 2 stages."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=4, pp=2, tp=2, dp_type="ddp",
+        global_bsz=4, chunks=2, vtp=2)
+    res = run_distributed(_moe_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for losses in res:
+        for a, b in zip(losses, base_losses):
+            assert abs(a - b) < TOL, (losses, base_losses)
