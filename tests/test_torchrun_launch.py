@@ -1,9 +1,16 @@
 """End-to-end torchrun launch of the train CLI (the exact launcher the
 driver uses for bench.py): 2 ranks, gloo, tiny model."""
+import socket
 import subprocess
 import sys
 
 import pytest
+
+
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 @pytest.mark.distributed
@@ -11,7 +18,7 @@ import pytest
 def test_torchrun_train_cli(tmp_path):
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes", "1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29531",
+           "--master-port", str(_free_port()),
            "-m", "hetu_galvatron_amd.cli.train",
            "model.model_name=tiny-llama",
            "parallel.mixed_precision=fp32",
