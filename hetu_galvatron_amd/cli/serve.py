@@ -66,7 +66,7 @@ def build_app(engine, lock: threading.Lock):
                 done[rid] = ev
         ev.wait()
         done.pop(rid, None)
-        return {"request_id": rid, "tokens": engine.outputs[rid]}
+        return {"request_id": rid, "tokens": engine.collect(rid)}
 
     return app
 

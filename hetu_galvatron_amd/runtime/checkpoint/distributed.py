@@ -40,19 +40,26 @@ def save_distributed_checkpoint(model, opt, sched, cfg, iteration: int,
     os.makedirs(it_dir, exist_ok=True)
 
     blocks = {}
+    expert_blocks = {}
     for i, blk in enumerate(model.stage_model.blocks):
-        if blk.flat is None or blk.flat.total == 0:
-            continue
-        f = blk.flat
-        blocks[i] = {
-            "master": f.master.detach().cpu(),
-            "exp_avg": f.exp_avg.detach().cpu(),
-            "exp_avg_sq": f.exp_avg_sq.detach().cpu(),
-            "mode": f.mode,
-        }
+        # both flat-param partitions travel: dense (blk.flat) and, for MoE
+        # layers under ep>1, the expert partition on the EDP group
+        # (blk.flat_expert) — its fp32 masters + Adam moments are stepped by
+        # optimizer.py:127 and must resume exactly like the dense ones
+        for f, store in ((blk.flat, blocks),
+                         (getattr(blk, "flat_expert", None), expert_blocks)):
+            if f is None or f.total == 0:
+                continue
+            store[i] = {
+                "master": f.master.detach().cpu(),
+                "exp_avg": f.exp_avg.detach().cpu(),
+                "exp_avg_sq": f.exp_avg_sq.detach().cpu(),
+                "mode": f.mode,
+            }
     payload = {
         "iteration": iteration,
         "blocks": blocks,
+        "expert_blocks": expert_blocks,
         "optimizer": {"step_count": opt.step_count},
         "scheduler": sched.state_dict() if sched is not None else None,
         "rng": {"torch": torch.get_rng_state(),
@@ -98,8 +105,8 @@ def load_distributed_checkpoint(model, opt, sched, cfg,
     with open(os.path.join(it_dir, "hybrid_parallel_config.json")) as f:
         saved_plan = json.load(f)
     cur_plan = model.plan.to_config_dict()
-    for key in ("pp_deg", "tp_sizes_enc", "cp_sizes_enc", "dp_types_enc",
-                "use_sp", "vtp"):
+    for key in ("pp_deg", "pp_division", "tp_sizes_enc", "cp_sizes_enc",
+                "dp_types_enc", "use_sp", "checkpoint_flags", "vtp"):
         assert str(saved_plan.get(key)) == str(cur_plan.get(key)), \
             (f"checkpoint plan mismatch on {key}: saved "
              f"{saved_plan.get(key)} vs current {cur_plan.get(key)} — "
@@ -112,14 +119,16 @@ def load_distributed_checkpoint(model, opt, sched, cfg,
         f"world size changed: {payload['world_size']} -> {world}"
     with torch.no_grad():
         for i, blk in enumerate(model.stage_model.blocks):
-            if blk.flat is None or i not in payload["blocks"]:
-                continue
-            st = payload["blocks"][i]
-            f = blk.flat
-            f.master.copy_(st["master"].to(f.master.device))
-            f.exp_avg.copy_(st["exp_avg"].to(f.exp_avg.device))
-            f.exp_avg_sq.copy_(st["exp_avg_sq"].to(f.exp_avg_sq.device))
-            f.apply_master_to_params()
+            for f, store in ((blk.flat, payload["blocks"]),
+                             (getattr(blk, "flat_expert", None),
+                              payload.get("expert_blocks", {}))):
+                if f is None or i not in store:
+                    continue
+                st = store[i]
+                f.master.copy_(st["master"].to(f.master.device))
+                f.exp_avg.copy_(st["exp_avg"].to(f.exp_avg.device))
+                f.exp_avg_sq.copy_(st["exp_avg_sq"].to(f.exp_avg_sq.device))
+                f.apply_master_to_params()
     if opt is not None:
         opt.step_count = payload["optimizer"]["step_count"]
     if sched is not None and payload["scheduler"] is not None:

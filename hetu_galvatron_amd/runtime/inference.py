@@ -384,6 +384,11 @@ class GalvatronTPGenerator:
         self.tp_group = g0.tp_group.group
         self.max_batch = max_batch
         self.max_seq = max_seq
+        if getattr(m, "sliding_window", None) is not None:
+            # same guard as GalvatronGenerator: decode attends the full
+            # cache, so a window shorter than max_seq would silently diverge
+            assert max_seq <= m.sliding_window, \
+                "generator v1: max_seq must fit the sliding window"
         self.scale = 1.0 / (m.head_dim ** 0.5)
         self._dev = next(self.embedding.parameters()).device
 

@@ -196,6 +196,11 @@ class ContinuousBatchingEngine:
         self.lengths[slot] = 0
         self.free.append(slot)
 
+    def collect(self, rid: int) -> List[int]:
+        """Return-and-delete a request's output tokens: the long-running
+        server path, so `outputs` does not grow without bound."""
+        return self.outputs.pop(rid, [])
+
     @property
     def n_active(self) -> int:
         return len(self.budgets)

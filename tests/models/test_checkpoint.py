@@ -142,6 +142,70 @@ def test_distributed_save_resume_world2(tmp_path):
             assert abs(a - b) < 1e-6, r
 
 
+def _moe_resume_worker(rank, world, tmp_dir):
+    import torch
+    from hetu_galvatron_amd.config import HybridParallelPlan
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import (
+        GalvatronModel, get_optimizer_and_param_scheduler, get_train_iterator)
+    from hetu_galvatron_amd.runtime.checkpoint import (
+        load_distributed_checkpoint, save_distributed_checkpoint)
+    from hetu_galvatron_amd.config import load_config
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-moe"},
+        "train": {"global_train_batch_size": 4, "train_iters": 4,
+                  "lr": 1e-3, "lr_decay_style": "constant",
+                  "distributed_backend": "gloo"},
+    })
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.uniform(num_layers=2, world_size=4, pp=1,
+                                      tp=1, dp_type="zero2", global_bsz=4,
+                                      ep=2)
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg, plan)
+    assert any(getattr(b, "flat_expert", None) is not None
+               for b in model.stage_model.blocks), \
+        "test setup: expected an expert flat block under ep=2"
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    batches = [next(it) for _ in range(4)]
+    for i in range(2):
+        opt.zero_grad(); model.forward_backward(batches[i])
+        opt.step(); sched.step()
+    save_distributed_checkpoint(model, opt, sched, cfg, 2, tmp_dir)
+    cont = []
+    for i in range(2, 4):
+        opt.zero_grad(); st = model.forward_backward(batches[i])
+        opt.step(); sched.step()
+        cont.append(model.global_loss(st))
+
+    torch.manual_seed(999 + rank)
+    model2 = GalvatronModel(cfg, plan)
+    opt2, sched2 = get_optimizer_and_param_scheduler(model2.stage_model, cfg)
+    assert load_distributed_checkpoint(model2, opt2, sched2, cfg,
+                                       tmp_dir) == 2
+    resumed = []
+    for i in range(2, 4):
+        opt2.zero_grad(); st = model2.forward_backward(batches[i])
+        opt2.step(); sched2.step()
+        resumed.append(model2.global_loss(st))
+    return {"cont": cont, "resumed": resumed}
+
+
+@pytest.mark.distributed
+def test_moe_distributed_save_resume_world4_ep2(tmp_path):
+    """MoE ep=2 x dp=2 save/resume: the expert flat blocks' masters + Adam
+    moments (blk.flat_expert) must round-trip exactly, not just the dense
+    ones (advisor finding: expert state was silently reinitialized)."""
+    from tests.utils import run_distributed
+    res = run_distributed(_moe_resume_worker, world_size=4,
+                          args=(str(tmp_path),))
+    for r in res:
+        for a, b in zip(r["cont"], r["resumed"]):
+            assert abs(a - b) < 1e-6, r
+
+
 def test_hf_mixtral_roundtrip_and_load():
     """Fabricated HF-mixtral-layout state -> canonical -> model; loss
     matches a model loaded from the direct canonical dump; g2h
