@@ -21,45 +21,127 @@ import torch
 import torch.distributed as dist
 
 
-def search_bench_plan(cfg, world, args):
-    """Run the strategy search for the bench config; returns a plan."""
-    from hetu_galvatron_amd.cli.search import default_mi355x_hardware
+PROFILE_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "profiles")
+
+
+def bench_layer_profiles(cfg, args):
+    """LayerProfile/OtherProfile from the MEASURED model profiles committed
+    under profiles/ (cli/profile_model output on MI355X); analytic
+    fallback only if the files are absent (fresh checkout)."""
     from hetu_galvatron_amd.search.costmodel import LayerProfile, OtherProfile
-    from hetu_galvatron_amd.search.engine import SearchEngine
+    from hetu_galvatron_amd.search.engine import read_computation_profile
 
     m = cfg.model
-    # measured on MI355X (rocprofv3, profiles/bench_r01_kernel_stats_v2.md):
-    # ~7.3 ms fwd per decoder layer per local sample at seq 4096
-    per_tok_ms = 7.3 / 4096
-    lp = LayerProfile(
-        parameter_mb=(4 * m.hidden_size ** 2 +
-                      3 * m.hidden_size * m.ffn_hidden_size) * 4 / 1e6,
-        fct_linear=(per_tok_ms * m.seq_length, 0.05),
-        act_per_bsz_mb={str(t): m.seq_length * m.hidden_size * 2 * 18
-                        / 1e6 / t for t in (1, 2, 4, 8)}
-        | {"checkpoint": m.seq_length * m.hidden_size * 2 / 1e6},
-        seq_length=m.seq_length, hidden_size=m.hidden_size)
-    op = OtherProfile(
-        parameter_mb=m.vocab_size * m.hidden_size * 4 / 1e6,
-        act_per_bsz_mb={"1": m.seq_length * m.vocab_size * 2 / 1e6 / max(cfg.parallel.chunks, 1)},
-        fct_linear=(0.6, 0.05))
+    comp_path = os.path.join(
+        PROFILE_DIR, f"computation_profiling_bf16_{args.model}.json")
+    mem_path = os.path.join(
+        PROFILE_DIR, f"model_profile_bf16_{args.model}.json")
+
+    other_fct = None
+    if os.path.exists(comp_path):
+        from hetu_galvatron_amd.search.engine import (
+            read_other_computation_profile)
+        fct = read_computation_profile(comp_path, m.seq_length)
+        other_fct = read_other_computation_profile(comp_path, m.seq_length)
+    else:
+        # analytic fallback: ~7.3 ms fwd per decoder layer per local sample
+        # at seq 4096 (rocprofv3, profiles/bench_r01_kernel_stats_v2.md)
+        print(f"[bench] {comp_path} missing; analytic fct fallback",
+              file=sys.stderr)
+        fct = (7.3 / 4096 * m.seq_length, 0.05)
+
+    if os.path.exists(mem_path):
+        with open(mem_path) as f:
+            mem = json.load(f)
+        lt = mem["layertype_0"]
+        other = mem.get("other", {})
+        lp = LayerProfile(
+            parameter_mb=float(lt["parameter_size"]),
+            fct_linear=fct,
+            act_per_bsz_mb=lt["tp_activation_per_bsz_dict"],
+            seq_length=m.seq_length, hidden_size=m.hidden_size)
+        op = OtherProfile(
+            parameter_mb=float(other.get("parameter_size", 0.0)),
+            act_per_bsz_mb=other.get("tp_activation_per_bsz_dict", {"1": 0.0}),
+            fct_linear=tuple(other.get("fct_linear",
+                                       other_fct or (0.6, 0.05))))
+    else:
+        print(f"[bench] {mem_path} missing; analytic memory fallback",
+              file=sys.stderr)
+        lp = LayerProfile(
+            parameter_mb=(4 * m.hidden_size ** 2 +
+                          3 * m.hidden_size * m.ffn_hidden_size) * 4 / 1e6,
+            fct_linear=fct,
+            act_per_bsz_mb={str(t): m.seq_length * m.hidden_size * 2 * 18
+                            / 1e6 / t for t in (1, 2, 4, 8)}
+            | {"checkpoint": m.seq_length * m.hidden_size * 2 / 1e6},
+            seq_length=m.seq_length, hidden_size=m.hidden_size)
+        op = OtherProfile(
+            parameter_mb=m.vocab_size * m.hidden_size * 4 / 1e6,
+            act_per_bsz_mb={"1": m.seq_length * m.vocab_size * 2 / 1e6
+                            / max(cfg.parallel.chunks, 1)},
+            fct_linear=(0.6, 0.05))
+    return lp, op
+
+
+def search_bench_plan(cfg, world, args):
+    """Run the FULL strategy search for the bench config; returns a plan.
+
+    The space is unclamped: per-layer TP (Megatron-SP) / Ulysses-SP /
+    DP / ZeRO-3 / activation-ckpt, pp in {1..world}, vocab-tp variants —
+    the same task grid the reference searches (search_engine.py:520-575).
+    Compute/memory inputs are measured profiles (profiles/); the xGMI
+    collective model is measured hardware_configs when present, analytic
+    otherwise (1-GPU leases cannot measure 8-GPU collectives).
+    """
+    from hetu_galvatron_amd.cli.search import default_mi355x_hardware
+    from hetu_galvatron_amd.search.engine import (SearchEngine,
+                                                  read_hardware_profiles)
+
+    lp, op = bench_layer_profiles(cfg, args)
+    world = max(world, 1)
+    hw_dir = os.path.join(PROFILE_DIR, "hardware_configs")
+    hw_file = os.path.join(
+        hw_dir, f"allreduce_bandwidth_1nodes_{world}gpus_per_node.json")
+    if os.path.exists(hw_file):
+        hw = read_hardware_profiles(hw_dir, 1, world)
+    else:
+        hw = default_mi355x_hardware(world)
     cfg.search.num_nodes = 1
-    cfg.search.num_gpus_per_node = max(world, 1)
+    cfg.search.num_gpus_per_node = world
     cfg.search.memory_constraint = 270
     cfg.search.settle_bsz = cfg.train.global_train_batch_size
-    cfg.search.settle_chunks = max(cfg.parallel.chunks, 1)
-    cfg.search.max_tp_deg = 1
-    cfg.search.disable_pp = 1
-    cfg.search.disable_sp = 1
-    cfg.search.max_pp_deg = 1
-    cfg.search.disable_vtp = 1
-    eng = SearchEngine(cfg, lp, op, default_mi355x_hardware(max(world, 1)))
+    if args.chunks:
+        cfg.search.settle_chunks = args.chunks
+    eng = SearchEngine(cfg, lp, op, hw)
     best = eng.parallelism_optimization(None)
     if best is None:
         return None
-    plan = best.plan
-    plan.default_dp_type = "zero2"
-    return plan
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] searched plan: pp{best.plan.pp_deg} "
+              f"tp{best.plan.tp_sizes_enc[:4]}... "
+              f"sp{best.plan.use_sp[:4]}... "
+              f"zero3-flags{best.plan.dp_types_enc[:4]}... "
+              f"ckpt{best.plan.checkpoint_flags[:4]}... "
+              f"chunks={best.plan.chunks} vtp={best.plan.vtp} "
+              f"predicted {best.throughput:.2f} samples/s", file=sys.stderr)
+    return best.plan
+
+
+def plan_desc(plan, world: int) -> str:
+    """Compact human-readable summary of the executed per-layer plan."""
+    n = plan.num_layers
+    tps = sorted(set(plan.tp_sizes_enc))
+    tp_s = str(tps[0]) if len(tps) == 1 else f"{tps[0]}-{tps[-1]}"
+    per_stage = world // plan.pp_deg
+    dps = sorted({per_stage // (t * c) for t, c in
+                  zip(plan.tp_sizes_enc, plan.cp_sizes_enc)})
+    dp_s = str(dps[0]) if len(dps) == 1 else f"{dps[0]}-{dps[-1]}"
+    return (f"searched:pp{plan.pp_deg}-tp{tp_s}-dp{dp_s}-"
+            f"{plan.default_dp_type}"
+            f"-sp{sum(plan.use_sp)}/{n}-zero3:{sum(plan.dp_types_enc)}/{n}"
+            f"-ckpt:{sum(plan.checkpoint_flags)}/{n}-chunks{plan.chunks}")
 
 
 def main():
@@ -124,11 +206,9 @@ def main():
         torch.manual_seed(cfg.train.seed)
 
     if args.plan == "auto" and args.pp == 1 and args.tp == 1:
-        # auto-searched plan (BASELINE metric): per-layer DP-type/ckpt search
-        # under the 288 GB budget, measured per-layer compute + analytic
-        # xGMI comm model. Round-1 space: dp-only degrees (the distributed
-        # paths validated on this hardware); tp/sp/pp degrees join the
-        # space as they get GPU-validated.
+        # auto-searched plan (BASELINE metric): full per-layer
+        # TP/SP/DP/ZeRO-3/ckpt x pp x vtp space under the 288 GB budget,
+        # measured per-layer compute/memory profiles (profiles/)
         plan = search_bench_plan(cfg, world, args)
         if plan is None:
             plan = resolve_plan(cfg, world)
@@ -190,9 +270,7 @@ def main():
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": args.seq_len,
-                "parallelism": (f"pp{plan.pp_deg}-tp{args.tp}-"
-                                f"dp{max(world, 1) // (args.pp * args.tp)}-"
-                                f"{args.dp_type}"),
+                "parallelism": plan_desc(plan, max(world, 1)),
             },
         }))
     if world > 1:

@@ -18,7 +18,9 @@ def main(argv=None):
     if cfg.profile.profile_type in ("computation", "all"):
         mp.launch_computation_profiling()
     if cfg.profile.profile_type in ("memory", "all"):
-        mp.launch_memory_profiling()
+        import torch
+        nproc = max(min(torch.cuda.device_count(), 8), 1)
+        mp.launch_memory_profiling(nproc=nproc)
         mp.process_memory_data()
     print("[profile_model] done")
 

@@ -56,6 +56,8 @@ def main(argv=None):
     it = RerunDataIterator(get_train_iterator(cfg, device,
                                               global_batch=cur_gbs))
     prof = RuntimeProfiler(enabled=use_gpu, device=device, rank=rank)
+    if cfg.profile.profile:
+        model.engine.profiler = prof  # fwd-only timing for computation JSON
     rsm = initialize_rerun_state_machine(enabled=True)
     mlog = MetricsLogger(cfg, rank=rank)
 

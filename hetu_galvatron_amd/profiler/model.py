@@ -52,13 +52,17 @@ class ModelProfiler:
         """layernum x bsz sweep, 1 GPU (reference :343-420)."""
         p = self.p
         for layernum in (p.profile_min_layer_num, p.profile_max_layer_num):
-            for bsz in range(p.profile_batch_size_start,
-                             p.profile_batch_size_end + 1):
+            bsz = p.profile_batch_size_start
+            while bsz <= p.profile_batch_size_end:
+                # doubling sweep (reference uses halving 1024->1): the fct
+                # fit is linear in bsz, so power-of-2 points suffice and
+                # keep the GPU sweep short
                 self._run(self._base_overrides() + [
                     "profile.profile_type=computation",
                     f"model.num_hidden_layers={layernum}",
                     f"train.global_train_batch_size={bsz}",
                     "parallel.chunks=1"])
+                bsz *= 2
 
     def launch_memory_profiling(self, nproc: int = 8) -> None:
         """pp1 x tp x {ckpt} (+ pp sweeps) layouts (reference :231-343)."""

@@ -29,8 +29,10 @@ class RuntimeProfiler:
         self.iteration = 0
         self.mem_snaps: Dict[str, Dict[str, float]] = {}
         self.iter_times_ms: List[float] = []
+        self.fwd_times_ms: List[float] = []
         self._ev_start: Optional[torch.cuda.Event] = None
         self._ev_end: Optional[torch.cuda.Event] = None
+        self._fwd_pairs: List = []
 
     # -- memory ------------------------------------------------------------
     def snap(self, tag: str) -> None:
@@ -79,6 +81,21 @@ class RuntimeProfiler:
         self._ev_end = torch.cuda.Event(enable_timing=True)
         self._ev_start.record()
 
+    def fwd_start(self) -> None:
+        """Bracket one forward chunk (engine calls these around each
+        microbatch forward; reference profile_forward methodology — the
+        computation profile's fct is FORWARD time, bct derives from it)."""
+        if not self.enabled:
+            return
+        a = torch.cuda.Event(enable_timing=True)
+        b = torch.cuda.Event(enable_timing=True)
+        a.record()
+        self._fwd_pairs.append((a, b))
+
+    def fwd_end(self) -> None:
+        if self.enabled and self._fwd_pairs:
+            self._fwd_pairs[-1][1].record()
+
     def time_end(self) -> Optional[float]:
         if not self.enabled or self._ev_start is None:
             self.iteration += 1
@@ -86,6 +103,11 @@ class RuntimeProfiler:
         self._ev_end.record()
         torch.cuda.synchronize(self.device)
         ms = self._ev_start.elapsed_time(self._ev_end)
+        if self._fwd_pairs:
+            fwd = sum(a.elapsed_time(b) for a, b in self._fwd_pairs)
+            self._fwd_pairs = []
+            if self.iteration >= self.warmup:
+                self.fwd_times_ms.append(fwd)
         if self.iteration >= self.warmup:
             self.iter_times_ms.append(ms)
         self.iteration += 1
@@ -97,14 +119,23 @@ class RuntimeProfiler:
         return sum(self.iter_times_ms) / len(self.iter_times_ms)
 
     # -- persistence -------------------------------------------------------
+    def avg_fwd_ms(self) -> float:
+        if not self.fwd_times_ms:
+            return 0.0
+        return sum(self.fwd_times_ms) / len(self.fwd_times_ms)
+
     def save_time_profile(self, path: str, key: str) -> None:
-        """Append {key: avg fwd(+bwd) ms} into a computation_profiling JSON
-        (reference key format 'layernum[N]_bsz{b}_seq{s}')."""
+        """Append {key: avg FORWARD ms} into a computation_profiling JSON
+        (reference key format 'layernum[N]_bsz{b}_seq{s}'; that value is
+        the search's fct).  The full iteration time rides along under an
+        'iter_'-prefixed key (prefix, not suffix, so the search's
+        layernum regex never matches it)."""
         data = {}
         if os.path.exists(path):
             with open(path) as f:
                 data = json.load(f)
-        data[key] = self.avg_iter_ms()
+        data[key] = self.avg_fwd_ms() or self.avg_iter_ms()
+        data[f"iter_{key}"] = self.avg_iter_ms()
         os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
         with open(path, "w") as f:
             json.dump(data, f, indent=4)

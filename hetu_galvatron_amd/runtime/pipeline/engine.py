@@ -68,6 +68,8 @@ class PipelineEngine:
         self.pipeline_type = pipeline_type
         self.overlap = overlap_grad_reduce
         self.act_dtype = act_dtype
+        self.profiler = None  # RuntimeProfiler: fwd_start/fwd_end brackets
+        self._after_fwd_snapped = False
         self.device = next(stage_model.parameters()).device \
             if any(True for _ in stage_model.parameters()) else torch.device("cpu")
         world = stage_model.world_size
@@ -80,9 +82,13 @@ class PipelineEngine:
     # ------------------------------------------------------------------ steps
 
     def _forward_chunk(self, ctx: Dict, recv_act: Optional[torch.Tensor]):
+        if self.profiler is not None:
+            self.profiler.fwd_start()
         h = recv_act
         for blk in self.sm.blocks:
             h = blk(h, ctx)
+        if self.profiler is not None:
+            self.profiler.fwd_end()
         return h
 
     def _loss_of(self, per_token: torch.Tensor, ctx: Dict, chunks: int):
@@ -151,6 +157,7 @@ class PipelineEngine:
             m["global_batch"] = ctx["global_batch"]
         stats = StepStats()
         self._set_auto_sync(False)
+        self._after_fwd_snapped = False
         if self.sm.pp_deg == 1:
             self._no_pipeline(mb, stats)
         elif self.pipeline_type == "gpipe":
@@ -169,6 +176,7 @@ class PipelineEngine:
             per_token = self._forward_chunk(ctx, None)
             self._stat_update(stats, per_token)
             loss = self._loss_of(per_token, ctx, n)
+            self._snap_after_fwd()
             loss.backward()
 
     # -- pp helpers
@@ -208,7 +216,16 @@ class PipelineEngine:
             out = torch.cat([out, ctx["encoder_memory"]], dim=0)
         return recv_act, out
 
+    def _snap_after_fwd(self) -> None:
+        """Record the After-Fwd memory point once per step, just before the
+        first backward — the moment stored activations peak (the memory
+        profile's activation_mb = After-Fwd minus Before-Fwd)."""
+        if self.profiler is not None and not self._after_fwd_snapped:
+            self.profiler.profile_memory("After-Fwd")
+            self._after_fwd_snapped = True
+
     def _bwd_step(self, inp, out, grad_out, ctx: Dict, chunks: int):
+        self._snap_after_fwd()
         with torch.profiler.record_function("galvatron::bwd_chunk"):
             return self._bwd_step_inner(inp, out, grad_out, ctx, chunks)
 

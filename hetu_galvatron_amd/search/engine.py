@@ -75,6 +75,34 @@ def read_computation_profile(path_or_dict, seq_len: int) -> Tuple[float, float]:
     return fit_linear(xs, ys)
 
 
+def read_other_computation_profile(path_or_dict, seq_len: int
+                                   ) -> Tuple[float, float]:
+    """Embedding+head+loss ("other") fwd ms from the SAME layernum sweep:
+    t_other(b) = t[N1](b) - N1 * t_layer(b) — the whole-model time minus
+    the extrapolated decoder-layer share (includes fixed launch overhead,
+    which honestly belongs to the non-repeating part)."""
+    d = path_or_dict
+    if isinstance(d, str):
+        with open(d) as f:
+            d = json.load(f)
+    import re
+    by_ln: Dict[int, Dict[int, float]] = {}
+    for k, v in d.items():
+        m = re.match(r"layernum\[(\d+)\]_bsz(\d+)_seq(\d+)", k)
+        if not m or int(m.group(3)) != seq_len:
+            continue
+        by_ln.setdefault(int(m.group(1)), {})[int(m.group(2))] = float(v)
+    lns = sorted(by_ln)
+    assert len(lns) >= 2, f"need >=2 layernum sweeps, got {lns}"
+    n1, n2 = lns[0], lns[-1]
+    xs, ys = [], []
+    for b in sorted(set(by_ln[n1]) & set(by_ln[n2])):
+        t_layer = (by_ln[n2][b] - by_ln[n1][b]) / (n2 - n1)
+        xs.append(b)
+        ys.append(max(by_ln[n1][b] - n1 * t_layer, 0.0))
+    return fit_linear(xs, ys)
+
+
 def read_hardware_profiles(hw_dir: str, nodes: int = 1, gpus: int = 8
                            ) -> HardwareProfile:
     """Reference hardware_configs/*.json -> HardwareProfile."""
@@ -200,7 +228,9 @@ class SearchEngine:
         self.other_profile = OtherProfile(
             parameter_mb=float(other.get("parameter_size", 0.0)),
             act_per_bsz_mb=other.get("tp_activation_per_bsz_dict", {}),
-            fct_linear=tuple(other.get("fct_linear", (0.0, 0.0))))
+            fct_linear=tuple(other.get(
+                "fct_linear",
+                read_other_computation_profile(comp_path, seq))))
         self.hw = read_hardware_profiles(hw_dir, self.args.num_nodes,
                                          self.args.num_gpus_per_node)
 
@@ -259,8 +289,27 @@ class SearchEngine:
             act = float(act) * global_bsz / chunks / max(self.world // pp // vtp, 1)
             other_mem[vtp] = states + act
         m_o, c_o = self.other_profile.fct_linear
-        other_time = {vtp: (m_o * (global_bsz / chunks) / vtp + c_o) *
-                      (1 + hw.bct_fct_coe) for vtp in vtp_opts}
+        other_time = {}
+        for vtp in vtp_opts:
+            t = (m_o * (global_bsz / chunks) / vtp + c_o) * \
+                (1 + hw.bct_fct_coe)
+            if vtp > 1:
+                # vocab-tp is not free: the embedding output is allreduced
+                # (masked-sum) over the vtp group, the head input is
+                # (all)gathered from / re-split to the decoder layout at two
+                # boundaries, and the collectives mirror in backward.  Price
+                # fwd+bwd as 2 x (1 allreduce + 2 allgather-class moves) of
+                # the per-dp-replica boundary activation.
+                from .costmodel import _msg_latency
+                dp_v = max((self.world // pp) // vtp, 1)
+                msg_mb = ((global_bsz / chunks / dp_v) * lp.seq_length *
+                          lp.hidden_size *
+                          (2 if self.mixed_precision else 4) / (1024 * 1024))
+                ar = hw.allreduce_latency_per_mb.get(f"{vtp}_1", 0.0)
+                sel = hw.allgather_latency.get(vtp)
+                ag = _msg_latency(sel, msg_mb) if sel else msg_mb * ar / 2
+                t += 2 * (msg_mb * ar + 2 * ag)
+            other_time[vtp] = t
 
         # per-layer-TYPE intra cost (with and without grad sync) per
         # strategy (reference: multi-layer-type DP)
