@@ -59,8 +59,12 @@ class RuntimeProfiler:
 
     def memory_summary(self) -> Dict[str, float]:
         """model-states / activation split from the 4-point snapshots
-        (reference post_profile_memory:134-194)."""
+        (reference post_profile_memory:134-194).  Uses the LAST iteration
+        that actually recorded a Before-Fwd snap (after the training loop
+        self.iteration already points one past the end)."""
         it = self.iteration
+        while it > 0 and f"iter{it}_Before-Fwd" not in self.mem_snaps:
+            it -= 1
         g = lambda tag, k="allocated_mb": self.mem_snaps.get(
             f"iter{it}_{tag}", {}).get(k, 0.0)
         before = g("Before-Fwd")
