@@ -72,7 +72,10 @@ def read_computation_profile(path_or_dict, seq_len: int) -> Tuple[float, float]:
     for b in sorted(set(by_ln[n1]) & set(by_ln[n2])):
         xs.append(b)
         ys.append((by_ln[n2][b] - by_ln[n1][b]) / (n2 - n1))
-    return fit_linear(xs, ys)
+    m, c = fit_linear(xs, ys)
+    # clamp the intercept: a small negative fit residue would make the DP
+    # chunk term (chunks * c) reward infinite microbatching
+    return (m, max(c, 0.0))
 
 
 def read_other_computation_profile(path_or_dict, seq_len: int
@@ -100,7 +103,8 @@ def read_other_computation_profile(path_or_dict, seq_len: int
         t_layer = (by_ln[n2][b] - by_ln[n1][b]) / (n2 - n1)
         xs.append(b)
         ys.append(max(by_ln[n1][b] - n1 * t_layer, 0.0))
-    return fit_linear(xs, ys)
+    m, c = fit_linear(xs, ys)
+    return (m, max(c, 0.0))
 
 
 def read_hardware_profiles(hw_dir: str, nodes: int = 1, gpus: int = 8
@@ -291,8 +295,11 @@ class SearchEngine:
         m_o, c_o = self.other_profile.fct_linear
         other_time = {}
         for vtp in vtp_opts:
-            t = (m_o * (global_bsz / chunks) / vtp + c_o) * \
-                (1 + hw.bct_fct_coe)
+            # vocab rows are data-parallel over the stage's remaining ranks
+            # regardless of vtp (dp_v = stage_ranks/vtp): per-rank samples
+            # do NOT shrink with vtp — only memory and comm layout change
+            per_rank_samples = (global_bsz / chunks) / max(self.world // pp, 1)
+            t = (m_o * per_rank_samples + c_o) * (1 + hw.bct_fct_coe)
             if vtp > 1:
                 # vocab-tp is not free: the embedding output is allreduced
                 # (masked-sum) over the vtp group, the head input is
