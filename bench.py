@@ -163,6 +163,9 @@ def main():
     ap.add_argument("--plan", type=str, default="auto",
                     help="searched-plan JSON path, 'auto' (run the search "
                          "engine), or 'none' (uniform knobs)")
+    ap.add_argument("--tune-gemms", type=str, default=None,
+                    help="TunableOp tuning during warmup; CSV written here "
+                         "at exit (then commit as profiles/tunableop_gfx950.csv)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -176,6 +179,27 @@ def main():
         sys.exit(1)
     torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
+
+    # hipBLASLt/rocBLAS algorithm selection tuned offline on MI355X for the
+    # bench GEMM shapes (tools/tune_gemms.py -> profiles/tunableop_gfx950.csv):
+    # TunableOp in read-only mode replays the tuned picks, no runtime sweep
+    tuned = os.path.join(PROFILE_DIR, "tunableop_gfx950.csv")
+    if args.tune_gemms:
+        torch.cuda.tunable.enable(True)
+        torch.cuda.tunable.tuning_enable(True)
+        torch.cuda.tunable.set_filename(args.tune_gemms,
+                                        insert_device_ordinal=False)
+        torch.cuda.tunable.set_max_tuning_duration(10)
+        torch.cuda.tunable.set_max_tuning_iterations(10)
+        if os.path.exists(tuned):
+            torch.cuda.tunable.read_file(tuned)  # warm-start from committed
+    elif os.path.exists(tuned) and os.environ.get("GALVATRON_NO_TUNABLEOP") != "1":
+        torch.cuda.tunable.enable(True)
+        torch.cuda.tunable.tuning_enable(False)
+        torch.cuda.tunable.set_filename(tuned, insert_device_ordinal=False)
+        if not torch.cuda.tunable.read_file(tuned):
+            print(f"[bench] tunableop read_file failed: {tuned}",
+                  file=sys.stderr)
 
     from hetu_galvatron_amd.config import load_config
     from hetu_galvatron_amd.core.initialize import initialize_galvatron
@@ -228,6 +252,10 @@ def main():
 
     for _ in range(args.warmup):
         one_step()
+    if args.tune_gemms:
+        # freeze the tuned selections: the timed region below measures the
+        # replayed picks, never an in-flight sweep
+        torch.cuda.tunable.tuning_enable(False)
 
     if world > 1:
         dist.barrier()
