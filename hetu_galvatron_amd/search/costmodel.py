@@ -75,14 +75,30 @@ def _msg_latency(select: Dict, mb: float) -> float:
     return m * mb + c
 
 
+def ddp_ratio(mixed_precision: bool) -> float:
+    """Model-states bytes relative to 16 B/param, for THIS runtime's flat
+    engine (zero.py): bf16 param 2 + fp32 grad accumulator 4 (FULL — grads
+    accumulate locally across microbatches before reduction) + fp32 master
+    4 + Adam moments 8."""
+    return 18 / 16 if mixed_precision else 1.0
+
+
 def zero2_ratio(d: int, mixed_precision: bool) -> float:
-    if mixed_precision:
-        return 7 / 8 * (1 / d + 0.003) + 1 / 8
-    return 3 / 4 * (1 / d + 0.003) + 1 / 4
+    """bf16 param full 2 + fp32 grad full 4 + reduced grad shard 4/d +
+    (master 4 + moments 8)/d."""
+    if not mixed_precision:
+        return 3 / 4 * (1 / d + 0.003) + 1 / 4
+    if d <= 1:
+        return ddp_ratio(mixed_precision)
+    return 6 / 16 + (1 / d + 0.003)
 
 
 def zero3_ratio(d: int, mixed_precision: bool) -> float:
-    return 1 / d + 0.003
+    """bf16 shard 2/d (+ transient gathered block, amortized into the pad)
+    + fp32 grad full 4 + grad shard 4/d + masters/moments 12/d."""
+    if not mixed_precision:
+        return 1 / d + 0.003
+    return 4 / 16 + (18 / 16) * (1 / d + 0.003)
 
 
 def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
@@ -98,15 +114,29 @@ def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
         bct += fct
 
     # -- DP gradient sync --------------------------------------------------
+    # unit = one half-collective (reduce-scatter OR all-gather) of the
+    # layer's wire-dtype grads/params over the sdp group.  Mode split
+    # (matches zero.py's schedule):
+    #   ddp   - allreduce during backward: 2 units overlappable
+    #   zero2 - RS during backward (1 unit overlappable) + param AG after
+    #           the optimizer step (1 unit exposed)
+    #   zero3 - bwd param re-gather + grad RS (2 units overlappable) +
+    #           fwd param AG (1 unit exposed; forward has no overlap)
     sdp = s.sdp
-    dp_message_mb = 2 * (sdp - 1) / max(sdp, 1) * (prof.parameter_mb / s.tp_sp)
+    unit_mb = (sdp - 1) / max(sdp, 1) * (prof.parameter_mb / s.tp_sp)
     if mixed_precision:
-        dp_message_mb /= 2
+        unit_mb /= 2
     if no_gradient_sync:
-        dp_message_mb = 0.0
+        unit_mb = 0.0
     key = f"{sdp}_0" if s.tp > 1 else f"{sdp}_1"
     dc = hw.allreduce_latency_per_mb.get(key, 0.0) if sdp > 1 else 0.0
     dc_overlap = dc * hw.overlap_coe
+    if s.dp_type == "zero3":
+        dp_message_mb, exposed_mb = 2 * unit_mb, unit_mb
+    elif s.dp_type == "zero2":
+        dp_message_mb, exposed_mb = unit_mb, unit_mb
+    else:
+        dp_message_mb, exposed_mb = 2 * unit_mb, 0.0
 
     # -- TP / SP collectives ----------------------------------------------
     tp_time = 0.0
@@ -151,9 +181,9 @@ def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
     else:
         result = fct + bct + tp_time + cp_time
 
-    # -- ZeRO-3 forward allgather -----------------------------------------
-    if s.dp_type == "zero3" and sdp > 1:
-        result += 0.5 * dp_message_mb * dc
+    # -- exposed (non-overlappable) DP traffic ----------------------------
+    if sdp > 1:
+        result += exposed_mb * dc
 
     return result * hw.costmodel_coe  # ms
 
@@ -192,13 +222,20 @@ def layer_memory_cost(s: LayerStrategy, prof: LayerProfile,
         states *= zero3_ratio(s.sdp, mixed_precision)
     elif s.dp_type == "zero2":
         states *= zero2_ratio(s.sdp, mixed_precision)
+    else:
+        states *= ddp_ratio(mixed_precision)
 
     if s.checkpoint:
         act = prof.act_per_bsz_mb["checkpoint"] * cum_lbsz / s.cp
         act /= s.tp_sp
     else:
         d = prof.act_per_bsz_mb
-        act = d.get(str(s.tp_sp), d.get(s.tp_sp)) * cum_lbsz / s.cp
+        per_bsz = d.get(str(s.tp_sp), d.get(s.tp_sp))
+        if per_bsz is None:
+            # profile measured at tp=1 only (1-GPU lease): activations
+            # shard ~1/tp_sp under Megatron-SP / Ulysses
+            per_bsz = d.get("1", d.get(1)) / s.tp_sp
+        act = per_bsz * cum_lbsz / s.cp
     return {"parameter": param_mb, "model_states": states, "activation": act,
             "total": states + act}
 

@@ -271,7 +271,9 @@ class SearchEngine:
         profiles, hw = self.layer_profiles, self.hw
         lp = profiles[0]
         ltypes = self.layer_types
-        strategies = enumerate_strategies(self.world, self.args, pp)
+        strategies = enumerate_strategies(
+            self.world, self.args, pp,
+            default_dp_type=self.cfg.parallel.default_dp_type)
         strategies = [s for s in strategies
                       if global_bsz % (s.dp * chunks) == 0]
         if not strategies:

@@ -21,7 +21,8 @@ def _pow2_range(limit: int) -> List[int]:
 
 
 def enumerate_strategies(world_size: int, args: SearchArgs,
-                         pp_deg: int) -> List[LayerStrategy]:
+                         pp_deg: int,
+                         default_dp_type: str = "ddp") -> List[LayerStrategy]:
     """All (tp|sp, dp_type, ckpt) combos for one pp degree.
 
     Power-of-2 degrees; tp*dp = world/pp; ulysses-sp layers replace tp by
@@ -52,7 +53,9 @@ def enumerate_strategies(world_size: int, args: SearchArgs,
             for mode, deg in modes:
                 tp = deg if mode == "tp" else 1
                 sp = deg if mode == "sp" else 1
-                dp_types = ["ddp"]
+                # the "0" leg of dp_types_enc executes as the runtime's
+                # default_dp_type — price what will actually run
+                dp_types = [default_dp_type if dp * cp > 1 else "ddp"]
                 if not args.disable_sdp and dp * cp > 1:
                     dp_types.append("zero3")
                 for dpt in dp_types:
@@ -60,8 +63,7 @@ def enumerate_strategies(world_size: int, args: SearchArgs,
                     for ck in ckpts:
                         out.append(LayerStrategy(
                             pp_deg=pp_deg, tp=tp, sp=sp, cp=cp, dp=dp,
-                            dp_type="zero3" if dpt == "zero3" else "ddp",
-                            checkpoint=ck))
+                            dp_type=dpt, checkpoint=ck))
     return out
 
 

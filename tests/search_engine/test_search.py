@@ -111,7 +111,10 @@ def test_memory_cost_shards_and_ckpt():
     ck = layer_memory_cost(LayerStrategy(tp=1, dp=8, checkpoint=True), lp,
                            64, 8, 0)
     tp2 = layer_memory_cost(LayerStrategy(tp=2, dp=4), lp, 64, 8, 0)
-    assert z3["model_states"] < base["model_states"] / 4
+    # zero3 shards params/masters/moments 1/d but keeps the full fp32 grad
+    # accumulator (zero.py flat_grad): floor is 4/18 of ddp states, not 1/d
+    assert z3["model_states"] < base["model_states"] / 2
+    assert z3["model_states"] > base["model_states"] * 4 / 18
     assert ck["activation"] < base["activation"]
     assert abs(tp2["parameter"] - base["parameter"] / 2) < 1e-6
 
