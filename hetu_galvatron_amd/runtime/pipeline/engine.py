@@ -27,22 +27,31 @@ def boundary_shape(layout, batch_size: int, seq_len: int, hidden: int):
     return (rows, b_loc, hidden)
 
 
-def chunk_batch(ctx: Dict, chunks: int) -> List[Dict]:
+def chunk_batch(ctx: Dict, chunks: int, dp: int = 1) -> List[Dict]:
     """Split the global batch context into microbatch contexts
-    (reference: pipeline.py:327 chunk_batch)."""
+    (reference: pipeline.py:327 chunk_batch + :275 remainder shapes).
+
+    Non-divisible batches are allowed: each microbatch's size is a
+    multiple of `dp` (the data-parallel split inside the model), with the
+    first `rem` microbatches one dp-row larger.  Per-microbatch p2p shape
+    negotiation downstream keys off ctx["batch_size"], so remainder
+    microbatches flow through GPipe/1F1B unchanged."""
     B = ctx["batch_size"]
-    assert B % chunks == 0, f"global batch {B} not divisible by chunks {chunks}"
-    b_mb = B // chunks
-    out = []
-    for m in range(chunks):
+    assert B % dp == 0, f"global batch {B} not divisible by dp {dp}"
+    rows = B // dp
+    chunks = min(chunks, rows)
+    base, rem = divmod(rows, chunks)
+    sizes = [(base + (1 if m < rem else 0)) * dp for m in range(chunks)]
+    out, lo = [], 0
+    for b_mb in sizes:
         sub = dict(ctx)
-        sub["input_ids"] = ctx["input_ids"][m * b_mb:(m + 1) * b_mb]
-        sub["labels"] = ctx["labels"][m * b_mb:(m + 1) * b_mb]
+        sub["input_ids"] = ctx["input_ids"][lo:lo + b_mb]
+        sub["labels"] = ctx["labels"][lo:lo + b_mb]
         if "enc_input_ids" in ctx:
-            sub["enc_input_ids"] = \
-                ctx["enc_input_ids"][m * b_mb:(m + 1) * b_mb]
+            sub["enc_input_ids"] = ctx["enc_input_ids"][lo:lo + b_mb]
         sub["batch_size"] = b_mb
         out.append(sub)
+        lo += b_mb
     return out
 
 
@@ -152,7 +161,11 @@ class PipelineEngine:
     def forward_backward(self, ctx: Dict, chunks: int) -> StepStats:
         ctx = dict(ctx)
         ctx["global_batch"] = ctx["batch_size"]
-        mb = chunk_batch(ctx, chunks)
+        # every layer's dp must divide each microbatch (degrees are powers
+        # of two, so the max is the lcm)
+        dp = max((blk.groups.strategy.dp for blk in self.sm.blocks
+                  if getattr(blk, "groups", None) is not None), default=1)
+        mb = chunk_batch(ctx, chunks, dp)
         for m in mb:
             m["global_batch"] = ctx["global_batch"]
         stats = StepStats()

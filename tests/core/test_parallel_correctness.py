@@ -167,6 +167,40 @@ def test_pp2_1f1b():
 
 
 @pytest.mark.distributed
+def test_pp2_1f1b_uneven_chunks():
+    """chunks=3 with 4 rows -> microbatches [2,1,1]: the remainder shape
+    negotiation (reference pipeline.py:275) across 1F1B."""
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, pp=2, chunks=3,
+                                      global_bsz=4,
+                                      pipeline_type="pipedream_flush")
+    run_case(2, plan)
+
+
+@pytest.mark.distributed
+def test_pp2_gpipe_uneven_chunks():
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, pp=2, chunks=3,
+                                      global_bsz=4, pipeline_type="gpipe")
+    run_case(2, plan)
+
+
+def test_uneven_chunks_no_pipeline():
+    """1-process path: global batch 4, chunks=3 -> [2,1,1] must reproduce
+    the chunks=1 loss exactly (loss normalized by global tokens)."""
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    base_losses, state_path = get_baseline()
+    cfg = make_cfg()
+    plan = HybridParallelPlan.uniform(N_LAYERS, 1, chunks=3, global_bsz=4)
+    model = GalvatronModel(cfg, plan)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    losses = train_steps(model, cfg)
+    for a, b in zip(losses, base_losses):
+        # fp32 grad accumulation order differs across microbatch splits
+        assert abs(a - b) < 1e-4, (losses, base_losses)
+
+
+@pytest.mark.distributed
 def test_mixed_per_layer_tp_dp():
     """layer0 tp2, layer1 dp2(zero3) -> exercises redistribution."""
     plan = HybridParallelPlan(
