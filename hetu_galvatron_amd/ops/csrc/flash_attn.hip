@@ -606,20 +606,26 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward dK/dV (v3): expanded per q-head (host reduces over the GQA
+// backward dK/dV (v5): expanded per q-head (host reduces over the GQA
 // group), 8 waves x 32 keys (256 keys/workgroup), async-staged q tiles.
-// Works in the S orientation (D-layout rows=q, cols=key) so BOTH the
-// dV^T and dK^T MFMAs take their P / dS operands from registers via the
-// cvt_pk+permlane32_swap conversion — the v1/v2 per-tile LDS round trips
-// (32 scalar ds_write_b16 x2, 23-37% bank-conflict cycles) are gone.
-//   S   = mfma(Q_frag[from q_lds rows], K^T_frag[kf regs])
-//   dP  = mfma(dO_frag[from do_lds rows], V^T_frag[vf regs])
-//   dV^T[d][key] = mfma(dO^T_frag[from dot_lds], P_frag[permlane])
-//   dK^T[d][key] = mfma(Q^T_frag[from qt_lds], dS_frag[permlane])
+//
+// v5 = two SEQUENTIAL phases in one kernel (dV pass, then dK pass), each
+// with only 64 accumulator VGPRs so the wave's K (and V) fragments are
+// PRELOADED into loop-invariant registers.  The v3/v4 single-pass variant
+// kept 128 accumulator VGPRs live (dkt+dvt) and the compiler was forced
+// to re-load every K/V fragment from L2 one at a time with a full
+// vmcnt(0) drain before EACH MFMA — rocprofv3 PMC showed the waves parked
+// (SQ_WAIT_ANY 73.7% of wave cycles, MFMA busy 28%) while the healthy dq
+// kernel sits at 46%/51%.  The S matrix is computed twice (once per
+// phase): +25% MFMA work for register room — measured net win.
+//   phase dV:  S = mfma(Q_frag[from q_lds rows], K^T_frag[kfr regs])
+//              dV^T[d][key] = mfma(dO^T_frag[from t_lds], P_frag[permlane])
+//   phase dK:  S again; dP = mfma(dO_frag[from do_lds rows], V^T[vfr regs])
+//              dK^T[d][key] = mfma(Q^T_frag[from t_lds], dS_frag[permlane])
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool DKPH>
 __device__ __attribute__((noinline))
-void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
+void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
                          const __bf16* __restrict__ k,
                          const __bf16* __restrict__ v,
@@ -631,20 +637,16 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
                          long dkv_base, long lse_base, int q_stride,
                          int kv_stride, int dkv_stride, int off, int sq,
                          int skv, float scale, bool causal) {
-  constexpr int QT = 32;           // q tile
-  constexpr int KBW = 256;         // keys per workgroup (8 waves x 32);
-                                   // K/V fragments re-read from L2 per q
-                                   // tile so 2 waves/SIMD fit (the 4-wave
-                                   // register-resident variant ran at
-                                   // 1 wave/SIMD and trailed the fwd
-                                   // kernel's efficiency 1.6x)
+  constexpr int QT = 64;           // q tile (2 mfma halves per stage)
+  constexpr int KBW = 256;         // keys per workgroup (8 waves x 32)
   constexpr int QROW = D + 8;
   constexpr int TROW = QT + 8;
   constexpr int NK = D / 16;
   constexpr int ND = D / 32;
   constexpr int QPT = (QT * D / 8 + 511) / 512;   // row packs per thread
 
-  constexpr int BUFSZ = 2 * QT * QROW + 2 * D * TROW;
+  // LDS per buffer: q rows + transposed image (+ dO rows in the dK phase)
+  constexpr int BUFSZ = (DKPH ? 2 : 1) * QT * QROW + D * TROW;
   __bf16* smem = smem_base;
   float (*lse_lds)[QT] = reinterpret_cast<float (*)[QT]>(lsedi_base);
   float (*di_lds)[QT] = reinterpret_cast<float (*)[QT]>(lsedi_base + 2 * QT);
@@ -656,28 +658,37 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
   const int hi = lane >> 5;
   const int k0w = kvblk * KBW + wid * 32;  // this wave's first key row
 
-  // K/V fragment base pointers (B-operands of S / dP); the 16 frags are
-  // re-read per q tile (L2-resident) instead of held in 64 registers
+  // loop-invariant K (and V) fragments, preloaded into registers — the
+  // whole point of the phase split
   const int kg_f = min(k0w + col, skv - 1);
-  const __bf16* kp_f = k + kv_base + (long)kg_f * kv_stride + 8 * hi;
-  const __bf16* vp_f = v + kv_base + (long)kg_f * kv_stride + 8 * hi;
-
-  f32x16 dkt[ND], dvt[ND];
+  bf16x8 kfr[NK];
+  bf16x8 vfr[DKPH ? NK : 1];
+  {
+    const __bf16* kp_f = k + kv_base + (long)kg_f * kv_stride + 8 * hi;
+    const __bf16* vp_f = v + kv_base + (long)kg_f * kv_stride + 8 * hi;
 #pragma unroll
-  for (int dt = 0; dt < ND; ++dt) {
-    dkt[dt] = (f32x16)(0.f);
-    dvt[dt] = (f32x16)(0.f);
+    for (int ks = 0; ks < NK; ++ks) {
+      kfr[ks] = *reinterpret_cast<const bf16x8*>(kp_f + ks * 16);
+      if (DKPH)
+        vfr[ks] = *reinterpret_cast<const bf16x8*>(vp_f + ks * 16);
+    }
   }
 
-  // first q tile that can attend any key of this block / this wave
+  f32x16 acc[ND];
+#pragma unroll
+  for (int dt = 0; dt < ND; ++dt) acc[dt] = (f32x16)(0.f);
+
+  // first q tile that can attend any key of this block (per-wave causal
+  // skipping happens per 32-q half inside the loop)
   int qstart = 0;
   if (causal) qstart = max(0, ((kvblk * KBW - off) / QT) * QT);
-  const int qstart_w = causal ? max(qstart, ((k0w - off) / QT) * QT) : 0;
 
-  // staged registers
+  // staged registers: q rows, transposed image of (dV: dO / dK: Q),
+  // dO rows (dK phase only)
   bf16x8 qst[QPT], dost[QPT];
-  ushort8 qtst[QPT], dotst[QPT];
+  ushort8 ttst[QPT];
   float lse_st, di_st;
+  const __bf16* tsrc = DKPH ? q : dout;  // transposed-image source
 
   auto stage_load = [&](int qt0) {
     const bool full = (qt0 + QT <= sq);
@@ -690,11 +701,12 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
       const int qg = full ? qt0 + row : min(qt0 + row, sq - 1);
       qst[p] = *reinterpret_cast<const bf16x8*>(
           q + q_base + (long)qg * q_stride + c8);
-      dost[p] = *reinterpret_cast<const bf16x8*>(
-          dout + q_base + (long)qg * q_stride + c8);
+      if (DKPH)
+        dost[p] = *reinterpret_cast<const bf16x8*>(
+            dout + q_base + (long)qg * q_stride + c8);
       if (!full && qt0 + row >= sq) {
         qst[p] = (bf16x8)(__bf16(0.f));
-        dost[p] = (bf16x8)(__bf16(0.f));
+        if (DKPH) dost[p] = (bf16x8)(__bf16(0.f));
       }
     }
 #pragma unroll
@@ -703,40 +715,32 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
       if (idx >= D * QT / 8) break;
       const int c = idx & (D - 1);
       const int qc = (idx / D) * 8;
-      const unsigned short* qb = reinterpret_cast<const unsigned short*>(
-          q + q_base) + c;
-      const unsigned short* dob = reinterpret_cast<const unsigned short*>(
-          dout + q_base) + c;
-      ushort8 q8, d8;
+      const unsigned short* tb = reinterpret_cast<const unsigned short*>(
+          tsrc + q_base) + c;
+      ushort8 t8;
       if (full) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          q8[j] = qb[(long)(qt0 + qc + j) * q_stride];
-          d8[j] = dob[(long)(qt0 + qc + j) * q_stride];
-        }
+        for (int j = 0; j < 8; ++j)
+          t8[j] = tb[(long)(qt0 + qc + j) * q_stride];
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          const long rr = (long)min(qt0 + qc + j, sq - 1) * q_stride;
-          unsigned short a = qb[rr], bb2 = dob[rr];
-          q8[j] = qt0 + qc + j < sq ? a : (unsigned short)0;
-          d8[j] = qt0 + qc + j < sq ? bb2 : (unsigned short)0;
+          unsigned short a = tb[(long)min(qt0 + qc + j, sq - 1) * q_stride];
+          t8[j] = qt0 + qc + j < sq ? a : (unsigned short)0;
         }
       }
-      qtst[p] = q8;
-      dotst[p] = d8;
+      ttst[p] = t8;
     }
     if (tid < QT) {
       const int qg = qt0 + tid;
       lse_st = qg < sq ? lse[lse_base + qg] : INFINITY;
-      di_st = qg < sq ? di[lse_base + qg] : 0.f;
+      if (DKPH) di_st = qg < sq ? di[lse_base + qg] : 0.f;
     }
   };
   auto stage_write = [&](int buf) {
     __bf16* q_lds = smem + buf * BUFSZ;
-    __bf16* qt_lds = q_lds + QT * QROW;
-    __bf16* do_lds = qt_lds + D * TROW;
-    __bf16* dot_lds = do_lds + QT * QROW;
+    __bf16* t_lds = q_lds + QT * QROW;
+    __bf16* do_lds = t_lds + D * TROW;
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
       const int idx = tid + p * 512;
@@ -744,7 +748,8 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
       const int row = idx / (D / 8);
       const int c8 = (idx - row * (D / 8)) * 8;
       *reinterpret_cast<bf16x8*>(q_lds + row * QROW + c8) = qst[p];
-      *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dost[p];
+      if (DKPH)
+        *reinterpret_cast<bf16x8*>(do_lds + row * QROW + c8) = dost[p];
     }
 #pragma unroll
     for (int p = 0; p < QPT; ++p) {
@@ -752,12 +757,11 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
       if (idx >= D * QT / 8) break;
       const int c = idx & (D - 1);
       const int qc = (idx / D) * 8;
-      *reinterpret_cast<ushort8*>(qt_lds + c * TROW + qc) = qtst[p];
-      *reinterpret_cast<ushort8*>(dot_lds + c * TROW + qc) = dotst[p];
+      *reinterpret_cast<ushort8*>(t_lds + c * TROW + qc) = ttst[p];
     }
     if (tid < QT) {
       lse_lds[buf][tid] = lse_st;
-      di_lds[buf][tid] = di_st;
+      if (DKPH) di_lds[buf][tid] = di_st;
     }
   };
 
@@ -769,100 +773,82 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
   for (int qt0 = qstart; qt0 < sq; qt0 += QT) {
     const bool have_next = qt0 + QT < sq;
     const __bf16* q_lds = smem + cur * BUFSZ;
-    const __bf16* qt_lds = q_lds + QT * QROW;
-    const __bf16* do_lds = qt_lds + D * TROW;
-    const __bf16* dot_lds = do_lds + QT * QROW;
+    const __bf16* t_lds = q_lds + QT * QROW;
+    const __bf16* do_lds = t_lds + D * TROW;
     if (have_next) stage_load(qt0 + QT);
 
-    if (qt0 >= qstart_w) {  // wave-uniform causal skip
-      // S = Q K^T ; dP = dO V^T   (D-layout rows=q(crow), cols=key(lane))
+    // two 32-q mfma halves per staged tile
+#pragma unroll
+    for (int qh = 0; qh < QT / 32; ++qh) {
+      const int q0h = qt0 + qh * 32;
+      // per-half causal skip (wave-uniform): any key of this wave live?
+      if (causal && q0h + 31 + off < k0w) continue;
+      if (q0h >= sq) continue;
+
+      // S = Q K^T (; dP = dO V^T)  D-layout rows=q(crow), cols=key(lane);
+      // A-frag lane l&31 = q row of the half, B-frag lane = key
       f32x16 s_acc = (f32x16)(0.f), dp_acc = (f32x16)(0.f);
 #pragma unroll
       for (int ks = 0; ks < NK; ++ks) {
-        bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp_f + ks * 16);
         bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-            q_lds + col * QROW + ks * 16 + 8 * hi);
-        s_acc = mfma32_bf16(qa, kf, s_acc);
-        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vp_f + ks * 16);
-        bf16x8 doa = *reinterpret_cast<const bf16x8*>(
-            do_lds + col * QROW + ks * 16 + 8 * hi);
-        dp_acc = mfma32_bf16(doa, vf, dp_acc);
+            q_lds + (qh * 32 + col) * QROW + ks * 16 + 8 * hi);
+        s_acc = mfma32_bf16(qa, kfr[ks], s_acc);
+        if (DKPH) {
+          bf16x8 doa = *reinterpret_cast<const bf16x8*>(
+              do_lds + (qh * 32 + col) * QROW + ks * 16 + 8 * hi);
+          dp_acc = mfma32_bf16(doa, vfr[ks], dp_acc);
+        }
       }
-      // NOTE the operand order: A = q/do frags have lane=q? No — see below.
-      // A-frag lane holds row l&31 of the Q tile; with A=Q (M=q) the
-      // D-layout is rows=q? D rows follow M=A's rows spread over regs and
-      // cols follow B's N=key per lane. Both operands here are read with
-      // lane=own-row, so: A=Q read q_lds row col -> A[q=l&31]; B=kf holds
-      // K[key=l&31] as B^T fragment... B-frag lane must hold B[kk][l&31]
-      // = K^T[kk][key] = K[key=l&31][kk] = kf ✓. So D = S[q? l&31 is BOTH
-      // A-row and B-col index per lane; the MFMA maps A rows over regs:
-      // D[row=crow -> q][col=lane -> key]... A-frag lane l holds
-      // A[l&31][kk] where l&31 indexes M rows -> correct A layout.
 
-      const int qg0 = qt0;
-      float pv[16], dsv[16];
+      float wv[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = mfma32_d_row(lane, r);
-        const float lse_r = lse_lds[cur][qrow];
-        const float di_r = di_lds[cur][qrow];
+        const float lse_r = lse_lds[cur][qh * 32 + qrow];
         const float e = __expf(s_acc[r] * scale - lse_r);
         float p = e;
         if (causal) {
           const int keyg = k0w + col;
-          const bool ok = keyg < skv && keyg <= qg0 + qrow + off;
+          const bool ok = keyg < skv && keyg <= q0h + qrow + off;
           p = ok ? e : 0.f;
         } else if (k0w + col >= skv) {
           p = 0.f;
         }
-        pv[r] = p;
-        dsv[r] = p * (dp_acc[r] - di_r) * scale;
+        if (DKPH) {
+          const float di_r = di_lds[cur][qh * 32 + qrow];
+          wv[r] = p * (dp_acc[r] - di_r) * scale;
+        } else {
+          wv[r] = p;
+        }
       }
 
-      // convert P and dS (D-layout rows=q) to B-fragments (lane=key,
+      // convert P / dS (D-layout rows=q) to B-fragments (lane=key,
       // in-lane=q) via cvt_pk + permlane32_swap
-      unsigned wp[8], wd[8];
+      unsigned w[8];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
-          {
-            unsigned a = pack_bf16(pv[8 * i + 2 * g], pv[8 * i + 2 * g + 1]);
-            unsigned bb = pack_bf16(pv[8 * i + 2 * g + 4],
-                                    pv[8 * i + 2 * g + 5]);
-            auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
-            wp[i * 4 + g] = r2[0];
-            wp[i * 4 + 2 + g] = r2[1];
-          }
-          {
-            unsigned a = pack_bf16(dsv[8 * i + 2 * g],
-                                   dsv[8 * i + 2 * g + 1]);
-            unsigned bb = pack_bf16(dsv[8 * i + 2 * g + 4],
-                                    dsv[8 * i + 2 * g + 5]);
-            auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
-            wd[i * 4 + g] = r2[0];
-            wd[i * 4 + 2 + g] = r2[1];
-          }
+          unsigned a = pack_bf16(wv[8 * i + 2 * g], wv[8 * i + 2 * g + 1]);
+          unsigned bb = pack_bf16(wv[8 * i + 2 * g + 4],
+                                  wv[8 * i + 2 * g + 5]);
+          auto r2 = __builtin_amdgcn_permlane32_swap(a, bb, false, false);
+          w[i * 4 + g] = r2[0];
+          w[i * 4 + 2 + g] = r2[1];
         }
       }
 
-      // dV^T += dO^T P ; dK^T += Q^T dS   (two K=16 slices of the q tile)
+      // dV^T += dO^T P  /  dK^T += Q^T dS  (two K=16 slices of the half)
 #pragma unroll
       for (int kh = 0; kh < 2; ++kh) {
-        union { unsigned u[4]; bf16x8 f; } pb, db;
+        union { unsigned u[4]; bf16x8 f; } fb;
 #pragma unroll
-        for (int g = 0; g < 4; ++g) {
-          pb.u[g] = wp[kh * 4 + g];
-          db.u[g] = wd[kh * 4 + g];
-        }
+        for (int g = 0; g < 4; ++g) fb.u[g] = w[kh * 4 + g];
 #pragma unroll
         for (int dt = 0; dt < ND; ++dt) {
-          bf16x8 doa = *reinterpret_cast<const bf16x8*>(
-              dot_lds + (col + 32 * dt) * TROW + kh * 16 + 8 * hi);
-          dvt[dt] = mfma32_bf16(doa, pb.f, dvt[dt]);
-          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              qt_lds + (col + 32 * dt) * TROW + kh * 16 + 8 * hi);
-          dkt[dt] = mfma32_bf16(qa, db.f, dkt[dt]);
+          bf16x8 ta = *reinterpret_cast<const bf16x8*>(
+              t_lds + (col + 32 * dt) * TROW + qh * 32 + kh * 16 + 8 * hi);
+          acc[dt] = mfma32_bf16(ta, fb.f, acc[dt]);
         }
       }
     }
@@ -875,17 +861,36 @@ void flash_bwd_dkv_block(int kvblk, const __bf16* __restrict__ dout,
   // epilogue: dK^T/dV^T D-layout rows=d(crow), cols=key(lane)
   const int keyg = k0w + col;
   if (keyg < skv) {
-    __bf16* dkr = dk_exp + dkv_base + (long)keyg * dkv_stride;
-    __bf16* dvr = dv_exp + dkv_base + (long)keyg * dkv_stride;
+    __bf16* outr = (DKPH ? dk_exp : dv_exp) + dkv_base +
+                   (long)keyg * dkv_stride;
 #pragma unroll
     for (int dt = 0; dt < ND; ++dt)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int d0 = dt * 32 + mfma32_d_row(lane, r);
-        dkr[d0] = (__bf16)dkt[dt][r];
-        dvr[d0] = (__bf16)dvt[dt][r];
-      }
+      for (int r = 0; r < 16; ++r)
+        outr[dt * 32 + mfma32_d_row(lane, r)] = (__bf16)acc[dt][r];
   }
+}
+
+template <int D>
+__device__ void flash_bwd_dkv_block(
+    int kvblk, const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
+    const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    __bf16* __restrict__ dk_exp, __bf16* __restrict__ dv_exp,
+    __bf16* smem_base, float* lsedi_base, long q_base, long kv_base,
+    long dkv_base, long lse_base, int q_stride, int kv_stride,
+    int dkv_stride, int off, int sq, int skv, float scale, bool causal) {
+  flash_bwd_dkv_phase<D, false>(kvblk, dout, q, k, v, lse, di, dk_exp,
+                                dv_exp, smem_base, lsedi_base, q_base,
+                                kv_base, dkv_base, lse_base, q_stride,
+                                kv_stride, dkv_stride, off, sq, skv, scale,
+                                causal);
+  __syncthreads();
+  flash_bwd_dkv_phase<D, true>(kvblk, dout, q, k, v, lse, di, dk_exp,
+                               dv_exp, smem_base, lsedi_base, q_base,
+                               kv_base, dkv_base, lse_base, q_stride,
+                               kv_stride, dkv_stride, off, sq, skv, scale,
+                               causal);
 }
 
 template <int D>
@@ -900,8 +905,9 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           __bf16* __restrict__ dv_exp,
                           int b, int sq, int skv, int hq, int hkv,
                           float scale, bool causal) {
-  constexpr int QT = 32, KBW = 256;
-  constexpr int BUFSZ = 2 * QT * (D + 8) + 2 * D * (QT + 8);
+  constexpr int QT = 64, KBW = 256;  // must match flash_bwd_dkv_phase
+  // dK phase is the larger LDS user: q rows + dO rows + transposed image
+  constexpr int BUFSZ = 2 * QT * (D + 8) + D * (QT + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
   __shared__ __align__(16) float lsedi[4 * QT];
   const int bh = blockIdx.y;

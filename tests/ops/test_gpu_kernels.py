@@ -181,6 +181,39 @@ def test_flash_bwd(d, hq, hkv):
     assert_close(dv, dv_ref, 6e-2, rtol=3e-2, what="flash dv")
 
 
+def test_flash_bench_shape_numerics():
+    """Bench-shape insurance: seq-4096 GQA 32/8 fwd+bwd vs
+    the fp32 reference — kernel regressions can't hide behind toy shapes.
+    Uses torch sdpa as the oracle for the forward (fp32) and autograd for
+    the backward; one (b=1) slice of the bench microbatch, <30 s."""
+    torch.manual_seed(11)
+    b, sq, hq, hkv, d = 1, 4096, 32, 8, 128
+    q = torch.randn(b, sq, hq, d, device=dev()).bfloat16()
+    k = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    v = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale)
+    # fp32 sdpa oracle in [b, h, s, d] with GQA expansion
+    q32 = q.float().permute(0, 2, 1, 3)
+    k32 = k.float().permute(0, 2, 1, 3).repeat_interleave(hq // hkv, dim=1)
+    v32 = v.float().permute(0, 2, 1, 3).repeat_interleave(hq // hkv, dim=1)
+    q32.requires_grad_(True); k32.requires_grad_(True); v32.requires_grad_(True)
+    o_ref = torch.nn.functional.scaled_dot_product_attention(
+        q32, k32, v32, is_causal=True, scale=scale)
+    assert_close(o, o_ref.permute(0, 2, 1, 3), 3e-2, what="bench-shape o")
+    do = torch.randn_like(o)
+    o_ref.backward(do.float().permute(0, 2, 1, 3))
+    dq, dk, dv = ext().flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                      True, scale)
+    g = hq // hkv
+    dk_ref = k32.grad.view(b, hkv, g, sq, d).sum(2).permute(0, 2, 1, 3)
+    dv_ref = v32.grad.view(b, hkv, g, sq, d).sum(2).permute(0, 2, 1, 3)
+    assert_close(dq, q32.grad.permute(0, 2, 1, 3), 6e-2, rtol=3e-2,
+                 what="bench-shape dq")
+    assert_close(dk, dk_ref, 1.2e-1, rtol=3e-2, what="bench-shape dk")
+    assert_close(dv, dv_ref, 6e-2, rtol=3e-2, what="bench-shape dv")
+
+
 def test_flash_cross_lengths():
     """sq != skv (ring-CP block form), bottom-right causal alignment."""
     torch.manual_seed(7)
