@@ -147,15 +147,19 @@ class RingComm:
         return recv
 
     def commit(self):
-        # even ranks send first to avoid gloo pairwise deadlock; nccl handles
+        """Launch the batched ops; returns the request list so callers can
+        keep several batches in flight (kv rotation overlapping dkv)."""
         self._reqs = dist.batch_isend_irecv(self._ops)
         self._ops = []
+        return self._reqs
 
-    def wait(self):
-        if self._reqs:
-            for r in self._reqs:
+    def wait(self, reqs=None):
+        reqs = self._reqs if reqs is None else reqs
+        if reqs:
+            for r in reqs:
                 r.wait()
-        self._reqs = None
+        if reqs is self._reqs:
+            self._reqs = None
 
 
 def _merge_attn_out(o: Optional[torch.Tensor], lse: Optional[torch.Tensor],
@@ -274,59 +278,83 @@ def zigzag_ring_flash_attn_bwd(do, q, k, v, o, lse, comm: RingComm,
     do_half = do[:, half:].contiguous()
     lse_half = lse[:, :, half:].contiguous()
     q_half = q[:, half:].contiguous()
+    # comm/compute overlap (the forward already pre-posts its sends): the
+    # next step's (k, v) rotation is in flight during this step's flash
+    # kernels, and the accumulated (dk, dv) rotation posted after step s
+    # completes during step s+1's compute — nothing blocks inline except
+    # the final homecoming rotation.
+    kv_reqs = dkv_reqs = None
+    k_nxt = v_nxt = dk_nxt = dv_nxt = None
+    if n > 1:
+        k_nxt = comm.send_recv(k_cur)
+        v_nxt = comm.send_recv(v_cur)
+        kv_reqs = comm.commit()
     for step in range(n):
         j = (r - step) % n
         cols = _zigzag_cols(S, n, j, q.device) if bias is not None else None
+        # -- compute this step's flash backward (kernels launch async) ----
         if not causal:
+            mode = "full"
             res = flash_attention_bwd_only(
                 do, q, k_cur, v_cur, o, lse, causal=False,
                 softmax_scale=softmax_scale,
                 bias=None if bias is None else bias[:, :, cols])
-            dq += res[0].float()
-            dk_cur += res[1].float()
-            dv_cur += res[2].float()
-            if bias is not None:
-                dbias[:, :, cols] += res[3]
         elif step == 0:
+            mode = "full"
             res = flash_attention_bwd_only(
                 do, q, k_cur, v_cur, o, lse, causal=True,
                 softmax_scale=softmax_scale,
                 bias=None if bias is None else bias[:, :, cols])
-            dq += res[0].float()
-            dk_cur += res[1].float()
-            dv_cur += res[2].float()
-            if bias is not None:
-                dbias[:, :, cols] += res[3]
         elif j < r:
+            mode = "kv_half"
             res = flash_attention_bwd_only(
                 do, q, k_cur[:, :half].contiguous(), v_cur[:, :half].contiguous(),
                 o, lse, causal=False, softmax_scale=softmax_scale,
                 bias=None if bias is None else
                 bias[:, :, j * cs:(j + 1) * cs])
+        else:
+            mode = "q_half"
+            res = flash_attention_bwd_only(
+                do_half, q_half, k_cur, v_cur, o_half, lse_half, causal=False,
+                softmax_scale=softmax_scale,
+                bias=None if bias is None else bias[:, half:][:, :, cols])
+        # -- the dkv posted after step-1 completed during the kernels above
+        if step > 0:
+            comm.wait(dkv_reqs)
+            dk_cur, dv_cur = dk_nxt, dv_nxt
+        # -- accumulate -----------------------------------------------------
+        if mode == "full":
+            dq += res[0].float()
+            dk_cur += res[1].float()
+            dv_cur += res[2].float()
+            if bias is not None:
+                dbias[:, :, cols] += res[3]
+        elif mode == "kv_half":
             dq += res[0].float()
             dk_cur[:, :half] += res[1].float()
             dv_cur[:, :half] += res[2].float()
             if bias is not None:
                 dbias[:, :, j * cs:(j + 1) * cs] += res[3]
         else:
-            res = flash_attention_bwd_only(
-                do_half, q_half, k_cur, v_cur, o_half, lse_half, causal=False,
-                softmax_scale=softmax_scale,
-                bias=None if bias is None else bias[:, half:][:, :, cols])
             dq[:, half:] += res[0].float()
             dk_cur += res[1].float()
             dv_cur += res[2].float()
             if bias is not None:
                 db = dbias[:, half:]
                 db[:, :, cols] += res[3]
-        # rotate kv together with accumulated dkv (n rotations total -> home)
-        k_nxt = comm.send_recv(k_cur)
-        v_nxt = comm.send_recv(v_cur)
+        # -- rotate: accumulated dkv now; kv for the step after next --------
         dk_nxt = comm.send_recv(dk_cur)
         dv_nxt = comm.send_recv(dv_cur)
-        comm.commit()
-        comm.wait()
-        k_cur, v_cur, dk_cur, dv_cur = k_nxt, v_nxt, dk_nxt, dv_nxt
+        dkv_reqs = comm.commit()
+        if step < n - 1:
+            comm.wait(kv_reqs)
+            k_cur, v_cur = k_nxt, v_nxt
+            if step < n - 2:
+                k_nxt = comm.send_recv(k_cur)
+                v_nxt = comm.send_recv(v_cur)
+                kv_reqs = comm.commit()
+    comm.wait(dkv_reqs)  # final homecoming rotation
+    dk_cur, dv_cur = dk_nxt, dv_nxt
     out = (dq.to(q.dtype), dk_cur.to(k.dtype), dv_cur.to(v.dtype))
     return out + ((dbias,) if bias is not None else ())
 
