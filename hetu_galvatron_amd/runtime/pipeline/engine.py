@@ -79,6 +79,7 @@ class PipelineEngine:
         self.act_dtype = act_dtype
         self.profiler = None  # RuntimeProfiler: fwd_start/fwd_end brackets
         self._after_fwd_snapped = False
+        self._pending_sends = []  # [(reqs, tensor)] in-flight async p2p
         self.device = next(stage_model.parameters()).device \
             if any(True for _ in stage_model.parameters()) else torch.device("cpu")
         world = stage_model.world_size
@@ -87,6 +88,15 @@ class PipelineEngine:
         self.prev_rank = (stage_model.stage - 1) * G + idx if stage_model.stage > 0 else None
         self.next_rank = (stage_model.stage + 1) * G + idx \
             if stage_model.stage < stage_model.pp_deg - 1 else None
+
+    def _send_async(self, t, dst) -> None:
+        self._pending_sends.append(p2p.send_tensor_async(t, dst))
+
+    def _drain_sends(self) -> None:
+        for reqs, _t in self._pending_sends:
+            for r in reqs:
+                r.wait()
+        self._pending_sends = []
 
     # ------------------------------------------------------------------ steps
 
@@ -177,6 +187,7 @@ class PipelineEngine:
             self._gpipe(mb, stats)
         else:
             self._pipedream_flush(mb, stats)
+        self._drain_sends()
         self._finalize_grads()
         return stats
 
@@ -261,7 +272,7 @@ class PipelineEngine:
                                        self.prev_rank, self.device)
             inp, out = self._fwd_step(ctx, stats, recv)
             if not self.sm.is_last:
-                p2p.send_tensor(out, self.next_rank)
+                self._send_async(out, self.next_rank)
             saved.append((inp, out, ctx))
         for m, (inp, out, ctx) in enumerate(saved):
             if m == n - 1:
@@ -272,7 +283,7 @@ class PipelineEngine:
                                            self.next_rank, self.device)
             din = self._bwd_step(inp, out, grad_out, ctx, n)
             if not self.sm.is_first and din is not None:
-                p2p.send_tensor(din, self.prev_rank)
+                self._send_async(din, self.prev_rank)
             saved[m] = None
 
     def _pipedream_flush(self, mb: List[Dict], stats: StepStats) -> None:
@@ -293,7 +304,7 @@ class PipelineEngine:
                                        self.prev_rank, self.device)
             inp, out = self._fwd_step(ctx, stats, recv)
             if not self.sm.is_last:
-                p2p.send_tensor(out, self.next_rank)
+                self._send_async(out, self.next_rank)
             saved.append((inp, out, ctx))
             fwd_i += 1
         # ---- steady 1F1B
@@ -310,16 +321,18 @@ class PipelineEngine:
             grad_out = None
             b_inp, b_out, b_ctx = saved[bwd_i]
             if not self.sm.is_last:
-                grad_out = p2p.send_recv(out, self.next_rank,
-                                         tuple(b_out.shape), b_out.dtype,
-                                         self.next_rank, self.device)
+                grad_out, sreqs, stens = p2p.send_recv_async(
+                    out, self.next_rank, tuple(b_out.shape), b_out.dtype,
+                    self.next_rank, self.device)
+                if sreqs:
+                    self._pending_sends.append((sreqs, stens))
             if bwd_i == n - 1:
                 self._set_auto_sync(True)
             din = self._bwd_step(b_inp, b_out, grad_out, b_ctx, n)
             saved[bwd_i] = None
             bwd_i += 1
             if not self.sm.is_first and din is not None:
-                p2p.send_tensor(din, self.prev_rank)
+                self._send_async(din, self.prev_rank)
         # ---- cooldown backwards
         while bwd_i < n:
             b_inp, b_out, b_ctx = saved[bwd_i]
@@ -333,4 +346,4 @@ class PipelineEngine:
             saved[bwd_i] = None
             bwd_i += 1
             if not self.sm.is_first and din is not None:
-                p2p.send_tensor(din, self.prev_rank)
+                self._send_async(din, self.prev_rank)
