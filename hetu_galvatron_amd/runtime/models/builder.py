@@ -62,6 +62,9 @@ class LayerBlock(nn.Module):
         self.prev_groups = prev_groups
         self.flat: Optional[FlatParamBlock] = None
         self.flat_expert: Optional[FlatParamBlock] = None
+        # neighbor links for zero3 prefetch (set by build_hybrid_parallel_model)
+        self.next_blk: Optional["LayerBlock"] = None
+        self.prev_blk: Optional["LayerBlock"] = None
 
     def setup_zero(self, mode: str, param_dtype: torch.dtype, device,
                    reduce_in_fp32: bool = False) -> None:
@@ -93,6 +96,13 @@ class LayerBlock(nn.Module):
             self.flat.gather_params()
         if self.flat_expert is not None:
             self.flat_expert.gather_params()
+        # zero3 prefetch: kick the NEXT block's param all-gather so it
+        # overlaps this block's forward compute
+        if self.next_blk is not None:
+            if self.next_blk.flat is not None:
+                self.next_blk.flat.prefetch_gather()
+            if self.next_blk.flat_expert is not None:
+                self.next_blk.flat_expert.prefetch_gather()
         if x is not None and self.flat is not None and x.requires_grad:
             x = _PostBackwardHook.apply(self.flat, x)
         if x is not None and self.flat_expert is not None and x.requires_grad:
@@ -103,13 +113,18 @@ class LayerBlock(nn.Module):
                 lambda t: self._inner_forward(t, ctx), x, use_reentrant=False)
         else:
             out = self._inner_forward(x, ctx)
+        prev_flat = self.prev_blk.flat if self.prev_blk is not None else None
+        prev_exp = self.prev_blk.flat_expert if self.prev_blk is not None \
+            else None
         if self.flat_expert is not None:
             if out.requires_grad:
-                out = _PreBackwardGather.apply(self.flat_expert, out)
+                out = _PreBackwardGather.apply(self.flat_expert, out, prev_exp)
             self.flat_expert.reshard_params()
         if self.flat is not None:
             if out.requires_grad:
-                out = _PreBackwardGather.apply(self.flat, out)
+                # while this block's backward runs, prefetch the previous
+                # block (the next one needed in backward order)
+                out = _PreBackwardGather.apply(self.flat, out, prev_flat)
             self.flat.reshard_params()
         return out
 
@@ -265,6 +280,11 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         mode = b.groups.strategy.dp_type
         b.setup_zero(mode, dtype, device,
                      reduce_in_fp32=cfg.parallel.gradient_reduce_in_fp32)
+    for i, b in enumerate(blocks):  # zero3 prefetch neighbor links
+        # plain-dict assignment: LayerBlock is an nn.Module and a normal
+        # setattr would register the neighbor as a submodule (a cycle)
+        b.__dict__["prev_blk"] = blocks[i - 1] if i > 0 else None
+        b.__dict__["next_blk"] = blocks[i + 1] if i + 1 < len(blocks) else None
 
     recv_layout = None
     if my_stage > 0:
