@@ -12,7 +12,7 @@ from typing import List, Sequence
 import numpy as np
 import torch
 
-from .indexed import IndexedDataset
+from .indexed import IndexedDataset, load_indexed_dataset
 
 
 def _helpers():
@@ -77,13 +77,13 @@ def build_pretraining_dataset(data_paths: List[str], seq_length: int,
     """data_paths: ["w1", "prefix1", "w2", "prefix2", ...] or ["prefix"]
     (reference megatron data_path convention)."""
     if len(data_paths) == 1:
-        return GPTDataset(IndexedDataset(data_paths[0]), seq_length,
+        return GPTDataset(load_indexed_dataset(data_paths[0]), seq_length,
                           num_samples, seed)
     assert len(data_paths) % 2 == 0, \
         "data_path must be 'prefix' or 'w1 prefix1 w2 prefix2 ...'"
     weights = [float(data_paths[i]) for i in range(0, len(data_paths), 2)]
     prefixes = [data_paths[i] for i in range(1, len(data_paths), 2)]
     per = [max(int(num_samples * w / sum(weights)) + 1, 1) for w in weights]
-    dsets = [GPTDataset(IndexedDataset(p), seq_length, n, seed + i)
+    dsets = [GPTDataset(load_indexed_dataset(p), seq_length, n, seed + i)
              for i, (p, n) in enumerate(zip(prefixes, per))]
     return BlendedDataset(dsets, weights, num_samples)
