@@ -43,7 +43,7 @@ typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
 // ---------------------------------------------------------------------------
 // forward (v2)
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool BIASED = false>
 __device__ __attribute__((noinline))
 void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
                      const __bf16* __restrict__ k,
@@ -51,7 +51,9 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
                      __bf16* __restrict__ o, float* __restrict__ lse,
                      __bf16* smem_base, long q_base, long kv_base,
                      long lse_base, int q_stride, int kv_stride, int off,
-                     int sq, int skv, float sl2e, bool causal) {
+                     int sq, int skv, float sl2e, bool causal,
+                     const __bf16* __restrict__ bias = nullptr,
+                     long bias_base = 0) {
   constexpr int KB = 64;            // kv tile
   constexpr int QBF = 256;          // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;       // padded K row (bf16 elems)
@@ -179,16 +181,24 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
         st1 = mfma32_bf16(k1, qf[ks], st1);
       }
 
-      // ---- scale (+ mask on straddle/tail tiles) ----
+      // ---- scale (+ additive bias, + mask on straddle/tail tiles) ----
       float pv[32];
       const int qg = q0w + col;
+      const __bf16* brow = nullptr;
+      if (BIASED)
+        brow = bias + bias_base + (long)min(qg, sq - 1) * skv;
       const bool need_mask =
           (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
+      constexpr float LOG2E = 1.4426950408889634f;
       if (need_mask) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int key0 = kv0 + mfma32_d_row(lane, r);
           float x0 = st0[r] * sl2e, x1 = st1[r] * sl2e;
+          if (BIASED) {
+            x0 += LOG2E * (float)brow[min(key0, skv - 1)];
+            x1 += LOG2E * (float)brow[min(key0 + 32, skv - 1)];
+          }
           if (key0 >= skv || (causal && key0 > qg + off)) x0 = neg_big();
           if (key0 + 32 >= skv || (causal && key0 + 32 > qg + off))
             x1 = neg_big();
@@ -198,8 +208,14 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          pv[r] = st0[r] * sl2e;
-          pv[16 + r] = st1[r] * sl2e;
+          float b0 = 0.f, b1 = 0.f;
+          if (BIASED) {
+            const int key0 = kv0 + mfma32_d_row(lane, r);
+            b0 = LOG2E * (float)brow[key0];
+            b1 = LOG2E * (float)brow[key0 + 32];
+          }
+          pv[r] = st0[r] * sl2e + b0;
+          pv[16 + r] = st1[r] * sl2e + b1;
         }
       }
 
@@ -287,14 +303,14 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
 // thin kernel: common indexing + complementary-pair causal load balance
 // (block x runs q blocks {x, nqb-1-x}: constant total KV tiles per block,
 // so the makespan matches the average instead of 2x the deepest block)
-template <int D>
+template <int D, bool BIASED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_fwd_kernel(const __bf16* __restrict__ q,
                       const __bf16* __restrict__ k,
                       const __bf16* __restrict__ v,
                       __bf16* __restrict__ o, float* __restrict__ lse,
                       int b, int sq, int skv, int hq, int hkv, float scale,
-                      bool causal) {
+                      bool causal, const __bf16* __restrict__ bias = nullptr) {
   constexpr int KB = 64, QBF = 256;
   constexpr int BUFSZ = KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -305,16 +321,19 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const long q_base = ((long)bi * sq * hq + h) * D;
   const long kv_base = ((long)bi * skv * hkv + hk) * D;
   const long lse_base = ((long)bi * hq + h) * sq;
+  const long bias_base = (long)h * sq * skv;  // bias [hq, sq, skv]
   const float sl2e = scale * 1.4426950408889634f;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
-  flash_fwd_block<D>(blockIdx.x, q, k, v, o, lse, smem, q_base, kv_base,
-                     lse_base, hq * D, hkv * D, off, sq, skv, sl2e, causal);
+  flash_fwd_block<D, BIASED>(blockIdx.x, q, k, v, o, lse, smem, q_base,
+                             kv_base, lse_base, hq * D, hkv * D, off, sq,
+                             skv, sl2e, causal, bias, bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && qb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_fwd_block<D>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
-                       lse_base, hq * D, hkv * D, off, sq, skv, sl2e, causal);
+    flash_fwd_block<D, BIASED>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
+                               lse_base, hq * D, hkv * D, off, sq, skv,
+                               sl2e, causal, bias, bias_base);
   }
 }
 
@@ -357,7 +376,7 @@ __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
 // the v1 form spilled 34 dwords/lane to scratch).
 // dQ^T[d][q] += K^T dS accumulated in D-layout, epilogue like the forward.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool BIASED = false>
 __device__ __attribute__((noinline))
 void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
                         const __bf16* __restrict__ q,
@@ -368,7 +387,9 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
                         __bf16* __restrict__ dq, __bf16* smem_base,
                         long q_base, long kv_base, long lse_base,
                         int q_stride, int kv_stride, int off, int sq,
-                        int skv, float scale, bool causal) {
+                        int skv, float scale, bool causal,
+                        const __bf16* __restrict__ bias = nullptr,
+                        long bias_base = 0) {
   constexpr int KB = 32;         // kv tile
   constexpr int QBF = 256;       // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;
@@ -508,12 +529,20 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
 
       // dS^T = P^T (dP^T - Di), branchless mask
       const int qg = q0w + col;
+      const __bf16* brow = nullptr;
+      if (BIASED)
+        brow = bias + bias_base + (long)min(qg, sq - 1) * skv;
       const bool need_mask =
           (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
       float dsv[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float e = __expf(st[r] * sl2e - lse_c);
+        float sc = st[r] * sl2e;
+        if (BIASED) {
+          const int key = kv0 + mfma32_d_row(lane, r);
+          sc += (float)brow[min(key, skv - 1)];
+        }
+        const float e = __expf(sc - lse_c);
         float pr = e;
         if (need_mask) {
           const int key = kv0 + mfma32_d_row(lane, r);
@@ -570,7 +599,7 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
   }
 }
 
-template <int D>
+template <int D, bool BIASED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
@@ -580,7 +609,8 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          const float* __restrict__ di,
                          __bf16* __restrict__ dq,
                          int b, int sq, int skv, int hq, int hkv,
-                         float scale, bool causal) {
+                         float scale, bool causal,
+                         const __bf16* __restrict__ bias = nullptr) {
   constexpr int KB = 32, QBF = 256;
   constexpr int BUFSZ = 2 * KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -591,17 +621,20 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const long q_base = ((long)bi * sq * hq + h) * D;
   const long kv_base = ((long)bi * skv * hkv + hk) * D;
   const long lse_base = ((long)bi * hq + h) * sq;
+  const long bias_base = (long)h * sq * skv;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
-  flash_bwd_dq_block<D>(blockIdx.x, dout, q, k, v, lse, di, dq, smem,
-                        q_base, kv_base, lse_base, hq * D, hkv * D, off,
-                        sq, skv, scale, causal);
+  flash_bwd_dq_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dq,
+                                smem, q_base, kv_base, lse_base, hq * D,
+                                hkv * D, off, sq, skv, scale, causal, bias,
+                                bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && qb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_bwd_dq_block<D>(qb2, dout, q, k, v, lse, di, dq, smem, q_base,
-                          kv_base, lse_base, hq * D, hkv * D, off, sq, skv,
-                          scale, causal);
+    flash_bwd_dq_block<D, BIASED>(qb2, dout, q, k, v, lse, di, dq, smem,
+                                  q_base, kv_base, lse_base, hq * D,
+                                  hkv * D, off, sq, skv, scale, causal,
+                                  bias, bias_base);
   }
 }
 
@@ -623,7 +656,7 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 //   phase dK:  S again; dP = mfma(dO_frag[from do_lds rows], V^T[vfr regs])
 //              dK^T[d][key] = mfma(Q^T_frag[from t_lds], dS_frag[permlane])
 // ---------------------------------------------------------------------------
-template <int D, bool DKPH>
+template <int D, bool DKPH, bool BIASED = false>
 __device__ __attribute__((noinline))
 void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
@@ -636,7 +669,10 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
                          float* lsedi_base, long q_base, long kv_base,
                          long dkv_base, long lse_base, int q_stride,
                          int kv_stride, int dkv_stride, int off, int sq,
-                         int skv, float scale, bool causal) {
+                         int skv, float scale, bool causal,
+                         const __bf16* __restrict__ bias = nullptr,
+                         float* __restrict__ dbias = nullptr,
+                         long bias_base = 0) {
   constexpr int QT = 64;           // q tile (2 mfma halves per stage)
   constexpr int KBW = 256;         // keys per workgroup (8 waves x 32)
   constexpr int QROW = D + 8;
@@ -805,7 +841,13 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
       for (int r = 0; r < 16; ++r) {
         const int qrow = mfma32_d_row(lane, r);
         const float lse_r = lse_lds[cur][qh * 32 + qrow];
-        const float e = __expf(s_acc[r] * scale - lse_r);
+        float sc = s_acc[r] * scale;
+        if (BIASED) {
+          const long bq = (long)min(q0h + qrow, sq - 1);
+          sc += (float)bias[bias_base + bq * skv +
+                            min(k0w + col, skv - 1)];
+        }
+        const float e = __expf(sc - lse_r);
         float p = e;
         if (causal) {
           const int keyg = k0w + col;
@@ -816,7 +858,15 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
         }
         if (DKPH) {
           const float di_r = di_lds[cur][qh * 32 + qrow];
-          wv[r] = p * (dp_acc[r] - di_r) * scale;
+          const float t = p * (dp_acc[r] - di_r);
+          if (BIASED) {
+            // d(bias) = dS (un-scaled); batch contributions accumulate
+            // via fp32 atomics into [hq, sq, skv]
+            const int bq = q0h + qrow, bk = k0w + col;
+            if (bq < sq && bk < skv)
+              atomicAdd(dbias + bias_base + (long)bq * skv + bk, t);
+          }
+          wv[r] = t * scale;
         } else {
           wv[r] = p;
         }
@@ -871,7 +921,7 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
   }
 }
 
-template <int D>
+template <int D, bool BIASED = false>
 __device__ void flash_bwd_dkv_block(
     int kvblk, const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
@@ -879,21 +929,21 @@ __device__ void flash_bwd_dkv_block(
     __bf16* __restrict__ dk_exp, __bf16* __restrict__ dv_exp,
     __bf16* smem_base, float* lsedi_base, long q_base, long kv_base,
     long dkv_base, long lse_base, int q_stride, int kv_stride,
-    int dkv_stride, int off, int sq, int skv, float scale, bool causal) {
-  flash_bwd_dkv_phase<D, false>(kvblk, dout, q, k, v, lse, di, dk_exp,
-                                dv_exp, smem_base, lsedi_base, q_base,
-                                kv_base, dkv_base, lse_base, q_stride,
-                                kv_stride, dkv_stride, off, sq, skv, scale,
-                                causal);
+    int dkv_stride, int off, int sq, int skv, float scale, bool causal,
+    const __bf16* bias = nullptr, float* dbias = nullptr,
+    long bias_base = 0) {
+  flash_bwd_dkv_phase<D, false, BIASED>(
+      kvblk, dout, q, k, v, lse, di, dk_exp, dv_exp, smem_base, lsedi_base,
+      q_base, kv_base, dkv_base, lse_base, q_stride, kv_stride, dkv_stride,
+      off, sq, skv, scale, causal, bias, dbias, bias_base);
   __syncthreads();
-  flash_bwd_dkv_phase<D, true>(kvblk, dout, q, k, v, lse, di, dk_exp,
-                               dv_exp, smem_base, lsedi_base, q_base,
-                               kv_base, dkv_base, lse_base, q_stride,
-                               kv_stride, dkv_stride, off, sq, skv, scale,
-                               causal);
+  flash_bwd_dkv_phase<D, true, BIASED>(
+      kvblk, dout, q, k, v, lse, di, dk_exp, dv_exp, smem_base, lsedi_base,
+      q_base, kv_base, dkv_base, lse_base, q_stride, kv_stride, dkv_stride,
+      off, sq, skv, scale, causal, bias, dbias, bias_base);
 }
 
-template <int D>
+template <int D, bool BIASED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           const __bf16* __restrict__ q,
@@ -904,7 +954,9 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           __bf16* __restrict__ dk_exp,   // [b, skv, hq, D]
                           __bf16* __restrict__ dv_exp,
                           int b, int sq, int skv, int hq, int hkv,
-                          float scale, bool causal) {
+                          float scale, bool causal,
+                          const __bf16* __restrict__ bias = nullptr,
+                          float* __restrict__ dbias = nullptr) {
   constexpr int QT = 64, KBW = 256;  // must match flash_bwd_dkv_phase
   // dK phase is the larger LDS user: q rows + dO rows + transposed image
   constexpr int BUFSZ = 2 * QT * (D + 8) + D * (QT + 8);
@@ -920,17 +972,20 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const long lse_base = ((long)bi * hq + h) * sq;
   const int off = skv - sq;
   const int nkb = (skv + KBW - 1) / KBW;
-  flash_bwd_dkv_block<D>(blockIdx.x, dout, q, k, v, lse, di, dk_exp, dv_exp,
-                         smem, lsedi, q_base, kv_base, dkv_base, lse_base,
-                         hq * D, hkv * D, hq * D, off, sq, skv, scale,
-                         causal);
+  const long bias_base = (long)h * sq * skv;
+  flash_bwd_dkv_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dk_exp,
+                                 dv_exp, smem, lsedi, q_base, kv_base,
+                                 dkv_base, lse_base, hq * D, hkv * D,
+                                 hq * D, off, sq, skv, scale, causal, bias,
+                                 dbias, bias_base);
   const int kb2 = nkb - 1 - (int)blockIdx.x;
   if (causal && kb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_bwd_dkv_block<D>(kb2, dout, q, k, v, lse, di, dk_exp, dv_exp,
-                           smem, lsedi, q_base, kv_base, dkv_base, lse_base,
-                           hq * D, hkv * D, hq * D, off, sq, skv, scale,
-                           causal);
+    flash_bwd_dkv_block<D, BIASED>(kb2, dout, q, k, v, lse, di, dk_exp,
+                                   dv_exp, smem, lsedi, q_base, kv_base,
+                                   dkv_base, lse_base, hq * D, hkv * D,
+                                   hq * D, off, sq, skv, scale, causal,
+                                   bias, dbias, bias_base);
   }
 }
 
@@ -965,23 +1020,29 @@ template <int D>
 static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                const __bf16* v, __bf16* o, float* lse, int b,
                                int sq, int skv, int hq, int hkv, float scale,
-                               bool causal, hipStream_t st) {
+                               bool causal, const __bf16* bias,
+                               hipStream_t st) {
   int nqb = (sq + 255) / 256;
   dim3 grid(causal ? (nqb + 1) / 2 : nqb, b * hq);
-  hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k, v,
-                     o, lse, b, sq, skv, hq, hkv, scale, causal);
+  if (bias != nullptr)
+    hipLaunchKernelGGL((flash_fwd_kernel<D, true>), grid, dim3(512), 0, st,
+                       q, k, v, o, lse, b, sq, skv, hq, hkv, scale, causal,
+                       bias);
+  else
+    hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k,
+                       v, o, lse, b, sq, skv, hq, hkv, scale, causal);
 }
 
 void flash_fwd_launch(const __bf16* q, const __bf16* k, const __bf16* v,
                       __bf16* o, float* lse, int b, int sq, int skv, int hq,
                       int hkv, int d, float scale, bool causal,
-                      hipStream_t st) {
+                      hipStream_t st, const __bf16* bias) {
   if (d == 64)
     flash_fwd_launch_d<64>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                           causal, st);
+                           causal, bias, st);
   else
     flash_fwd_launch_d<128>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                            causal, st);
+                            causal, bias, st);
 }
 
 void attn_di_launch(const __bf16* dout, const __bf16* o, float* di, int b,
@@ -1002,29 +1063,42 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                const float* lse, const float* di, __bf16* dq,
                                __bf16* dk_exp, __bf16* dv_exp, int b, int sq,
                                int skv, int hq, int hkv, float scale,
-                               bool causal, hipStream_t st) {
+                               bool causal, const __bf16* bias, float* dbias,
+                               hipStream_t st) {
   int nqb = (sq + 255) / 256;
   dim3 gq(causal ? (nqb + 1) / 2 : nqb, b * hq);
-  hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout, q,
-                     k, v, lse, di, dq, b, sq, skv, hq, hkv, scale, causal);
   int nkb = (skv + 255) / 256;
   dim3 gkv(causal ? (nkb + 1) / 2 : nkb, b * hq);
-  hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st, dout,
-                     q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv, hq, hkv,
-                     scale, causal);
+  if (bias != nullptr) {
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<D, true>), gq, dim3(512), 0, st,
+                       dout, q, k, v, lse, di, dq, b, sq, skv, hq, hkv,
+                       scale, causal, bias);
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, true>), gkv, dim3(512), 0,
+                       st, dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq,
+                       skv, hq, hkv, scale, causal, bias, dbias);
+  } else {
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout,
+                       q, k, v, lse, di, dq, b, sq, skv, hq, hkv, scale,
+                       causal);
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st,
+                       dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv,
+                       hq, hkv, scale, causal);
+  }
 }
 
 void flash_bwd_launch(const __bf16* dout, const __bf16* q, const __bf16* k,
                       const __bf16* v, const float* lse, const float* di,
                       __bf16* dq, __bf16* dk_exp, __bf16* dv_exp, int b,
                       int sq, int skv, int hq, int hkv, int d, float scale,
-                      bool causal, hipStream_t st) {
+                      bool causal, hipStream_t st, const __bf16* bias,
+                      float* dbias) {
   if (d == 64)
     flash_bwd_launch_d<64>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b, sq,
-                           skv, hq, hkv, scale, causal, st);
+                           skv, hq, hkv, scale, causal, bias, dbias, st);
   else
     flash_bwd_launch_d<128>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b,
-                            sq, skv, hq, hkv, scale, causal, st);
+                            sq, skv, hq, hkv, scale, causal, bias, dbias,
+                            st);
 }
 
 void mfma_probe_launch(const __bf16* A, const __bf16* B, float* Dst, bool alt,

@@ -164,27 +164,63 @@ def flash_attention(q, k, v, causal: bool = True,
     return (o, lse) if return_lse else o
 
 
+class _FlashBiasAttention(torch.autograd.Function):
+    """Biased flash attention (t5 relative bias [hq, sq, skv]) with grads
+    for q/k/v AND the bias (dbias batch-summed, flows back into the bias
+    table's construction graph).  Native CDNA4 path when available."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, causal, scale):
+        o, lse = flash_attention_fwd_only(q, k, v, causal, scale, bias=bias)
+        ctx.save_for_backward(q, k, v, o, lse, bias)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse, bias = ctx.saved_tensors
+        dq, dk, dv, dbias = flash_attention_bwd_only(
+            do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale,
+            bias=bias)
+        return dq, dk, dv, dbias.to(bias.dtype), None, None
+
+
+def flash_bias_attention(q, k, v, bias, causal: bool = True,
+                         softmax_scale: Optional[float] = None):
+    """q: [b,s,hq,d]; bias [hq, sq, skv] additive pre-softmax scores."""
+    scale = softmax_scale if softmax_scale is not None \
+        else 1.0 / math.sqrt(q.shape[-1])
+    return _FlashBiasAttention.apply(q.contiguous(), k.contiguous(),
+                                     v.contiguous(), bias, causal, scale)
+
+
 def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None,
                              bias=None):
     """No-autograd forward returning (o, lse) — building block for ring CP.
-    bias [hq, sq, skv]: t5 relative bias — eager fp32 path (a bias input
-    on the native flash kernel is a v2 item)."""
+    bias [hq, sq, skv] (t5 relative bias): native CDNA4 kernel path
+    (bf16 bias added to the pre-softmax scores in-register)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if bias is None and use_native(q) and q.dtype == torch.bfloat16 \
+    if use_native(q) and q.dtype == torch.bfloat16 \
             and q.shape[-1] in (64, 128):
+        bias_n = None if bias is None else bias.to(torch.bfloat16).contiguous()
         return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
-                                             v.contiguous(), causal, scale)
+                                             v.contiguous(), causal, scale,
+                                             bias_n)
     return ref.attention_fwd(q, k, v, causal, scale, bias)
 
 
 def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True,
                              softmax_scale=None, bias=None):
-    """Without bias: (dq, dk, dv); with bias also dbias (batch-summed)."""
+    """Without bias: (dq, dk, dv); with bias also dbias (batch-summed,
+    fp32 — the native kernel accumulates it atomically in the dK phase)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    if bias is None and use_native(q) and q.dtype == torch.bfloat16 \
+    if use_native(q) and q.dtype == torch.bfloat16 \
             and q.shape[-1] in (64, 128):
-        return get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
-                                             causal, scale)
+        bias_n = None if bias is None else bias.to(torch.bfloat16).contiguous()
+        out = get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                            causal, scale, bias_n)
+        return tuple(out)
     return ref.attention_bwd(do, q, k, v, o, lse, causal, scale, bias)
 
 

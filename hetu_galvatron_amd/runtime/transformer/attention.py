@@ -25,7 +25,7 @@ from ..tensor_parallel import (
 )
 from .attention_impl import (
     DistributedAttention, ZigzagRingAttention, eager_bias_attention,
-    local_attention,
+    flash_bias_attention, local_attention,
 )
 from .rope import apply_rope_qk
 
@@ -112,7 +112,7 @@ class SelfAttention(nn.Module):
                                     softmax_scale=self.softmax_scale,
                                     attn_bias=attn_bias)
         elif attn_bias is not None:
-            o = eager_bias_attention(q, k, v, attn_bias, self.causal,
+            o = flash_bias_attention(q, k, v, attn_bias, self.causal,
                                      self.softmax_scale)
         elif self.core_attention is not None:
             # ulysses a2a keeps the global seq intact: window semantics

@@ -19,7 +19,8 @@ from typing import Callable, List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
-from ...ops import flash_attention, flash_attention_bwd_only, flash_attention_fwd_only
+from ...ops import (flash_attention, flash_attention_bwd_only,
+                    flash_attention_fwd_only, flash_bias_attention)
 from ..tensor_parallel.mappings import all_to_all, group_rank, group_size
 
 
@@ -102,7 +103,7 @@ class DistributedAttention(torch.nn.Module):
                                          softmax_scale=softmax_scale,
                                          attn_bias=bias_l)
             else:
-                o = eager_bias_attention(q, k, v, bias_l, causal,
+                o = flash_bias_attention(q, k, v, bias_l, causal,
                                          softmax_scale)
         elif self.inner_attention is not None:
             o = self.inner_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
@@ -407,7 +408,7 @@ class ZigzagRingAttention(torch.nn.Module):
         rows over the natural global kv axis (t5 relative bias)."""
         if group_size(self.cp_group) == 1:
             if attn_bias is not None:
-                return eager_bias_attention(q, k, v, attn_bias, causal,
+                return flash_bias_attention(q, k, v, attn_bias, causal,
                                             softmax_scale)
             return local_attention(q, k, v, causal=causal, softmax_scale=softmax_scale)
         return ZigzagRingFlashAttnFunc.apply(q, k, v, attn_bias,

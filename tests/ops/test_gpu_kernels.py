@@ -214,6 +214,58 @@ def test_flash_bench_shape_numerics():
     assert_close(dv, dv_ref, 6e-2, rtol=3e-2, what="bench-shape dv")
 
 
+@pytest.mark.parametrize("causal", [True, False])
+@pytest.mark.parametrize("sq,skv", [(256, 256), (128, 256)])
+def test_flash_bias_fwd_bwd(causal, sq, skv):
+    """Biased flash (t5 relative bias): o/lse/dq/dk/dv/dbias vs the fp32
+    reference (the round-2 native-bias kernel item)."""
+    torch.manual_seed(12)
+    b, h, d = 2, 4, 64
+    q = torch.randn(b, sq, h, d, device=dev()).bfloat16()
+    k = torch.randn(b, skv, h, d, device=dev()).bfloat16()
+    v = torch.randn(b, skv, h, d, device=dev()).bfloat16()
+    bias = (torch.randn(h, sq, skv, device=dev()) * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, causal, scale, bias)
+    o_ref, lse_ref = ref.attention_fwd(q.float(), k.float(), v.float(),
+                                       causal, scale, bias.float())
+    assert_close(o, o_ref, 3e-2, what=f"bias o c{causal}")
+    assert_close(lse, lse_ref, 2e-2, what="bias lse")
+    do = torch.randn_like(o)
+    dq, dk, dv, dbias = ext().flash_attn_bwd(do.contiguous(), q, k, v, o,
+                                             lse, causal, scale, bias)
+    dq_r, dk_r, dv_r, db_r = ref.attention_bwd(
+        do.float(), q.float(), k.float(), v.float(), None, None, causal,
+        scale, bias.float())
+    assert_close(dq, dq_r, 6e-2, rtol=3e-2, what="bias dq")
+    assert_close(dk, dk_r, 6e-2, rtol=3e-2, what="bias dk")
+    assert_close(dv, dv_r, 6e-2, rtol=3e-2, what="bias dv")
+    assert_close(dbias, db_r, 6e-2, rtol=3e-2, what="dbias")
+
+
+def test_flash_bias_gqa_d128():
+    torch.manual_seed(13)
+    b, sq, hq, hkv, d = 1, 192, 4, 2, 128
+    q = torch.randn(b, sq, hq, d, device=dev()).bfloat16()
+    k = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    v = torch.randn(b, sq, hkv, d, device=dev()).bfloat16()
+    bias = (torch.randn(hq, sq, sq, device=dev()) * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale, bias)
+    o_ref, _ = ref.attention_fwd(q.float(), k.float(), v.float(), True,
+                                 scale, bias.float())
+    assert_close(o, o_ref, 3e-2, what="bias gqa o")
+    do = torch.randn_like(o)
+    dq, dk, dv, dbias = ext().flash_attn_bwd(do.contiguous(), q, k, v, o,
+                                             lse, True, scale, bias)
+    dq_r, dk_r, dv_r, db_r = ref.attention_bwd(
+        do.float(), q.float(), k.float(), v.float(), None, None, True,
+        scale, bias.float())
+    assert_close(dq, dq_r, 6e-2, rtol=3e-2, what="bias gqa dq")
+    assert_close(dk, dk_r, 6e-2, rtol=3e-2, what="bias gqa dk")
+    assert_close(dbias, db_r, 6e-2, rtol=3e-2, what="bias gqa dbias")
+
+
 def test_flash_cross_lengths():
     """sq != skv (ring-CP block form), bottom-right causal alignment."""
     torch.manual_seed(7)
