@@ -39,6 +39,24 @@ DEV_INLINE unsigned pack_bf16(float lo, float hi) {
 }
 
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// 16 bias values for one 32-key D-layout column: row(r) = 8*(r>>2) +
+// 4*hi + (r&3) — each r-quad is 4 contiguous bf16, so 4 8-byte loads
+// cover the column.  base must be 4-element aligned (callers pass
+// kv0 + 4*hi with kv0 % 32 == 0); the clamp keeps the tail in bounds
+// (masked lanes discard the values).
+DEV_INLINE void load_bias16(const __bf16* brow, int base, int limit,
+                            float* out /*16*/) {
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    int idx = base + 8 * g;
+    idx = min(idx, max((limit - 4) & ~3, 0));
+    const bf16x4 b4 = *reinterpret_cast<const bf16x4*>(brow + idx);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[4 * g + j] = (float)b4[j];
+  }
+}
 
 // ---------------------------------------------------------------------------
 // forward (v2)
@@ -190,14 +208,19 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
       const bool need_mask =
           (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
       constexpr float LOG2E = 1.4426950408889634f;
+      float bv[32];
+      if (BIASED) {
+        load_bias16(brow, kv0 + 4 * hi, skv, bv);
+        load_bias16(brow, kv0 + 32 + 4 * hi, skv, bv + 16);
+      }
       if (need_mask) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int key0 = kv0 + mfma32_d_row(lane, r);
           float x0 = st0[r] * sl2e, x1 = st1[r] * sl2e;
           if (BIASED) {
-            x0 += LOG2E * (float)brow[min(key0, skv - 1)];
-            x1 += LOG2E * (float)brow[min(key0 + 32, skv - 1)];
+            x0 += LOG2E * bv[r];
+            x1 += LOG2E * bv[16 + r];
           }
           if (key0 >= skv || (causal && key0 > qg + off)) x0 = neg_big();
           if (key0 + 32 >= skv || (causal && key0 + 32 > qg + off))
@@ -208,14 +231,8 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float b0 = 0.f, b1 = 0.f;
-          if (BIASED) {
-            const int key0 = kv0 + mfma32_d_row(lane, r);
-            b0 = LOG2E * (float)brow[key0];
-            b1 = LOG2E * (float)brow[key0 + 32];
-          }
-          pv[r] = st0[r] * sl2e + b0;
-          pv[16 + r] = st1[r] * sl2e + b1;
+          pv[r] = st0[r] * sl2e + (BIASED ? LOG2E * bv[r] : 0.f);
+          pv[16 + r] = st1[r] * sl2e + (BIASED ? LOG2E * bv[16 + r] : 0.f);
         }
       }
 
@@ -535,13 +552,13 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
       const bool need_mask =
           (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
       float dsv[16];
+      float bv[16];
+      if (BIASED)
+        load_bias16(brow, kv0 + 4 * hi, skv, bv);
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         float sc = st[r] * sl2e;
-        if (BIASED) {
-          const int key = kv0 + mfma32_d_row(lane, r);
-          sc += (float)brow[min(key, skv - 1)];
-        }
+        if (BIASED) sc += bv[r];
         const float e = __expf(sc - lse_c);
         float pr = e;
         if (need_mask) {

@@ -202,7 +202,8 @@ def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None,
     (bf16 bias added to the pre-softmax scores in-register)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if use_native(q) and q.dtype == torch.bfloat16 \
-            and q.shape[-1] in (64, 128):
+            and q.shape[-1] in (64, 128) \
+            and (bias is None or bias.shape[-1] % 4 == 0):
         bias_n = None if bias is None else bias.to(torch.bfloat16).contiguous()
         return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
                                              v.contiguous(), causal, scale,
@@ -216,7 +217,8 @@ def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True,
     fp32 — the native kernel accumulates it atomically in the dK phase)."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if use_native(q) and q.dtype == torch.bfloat16 \
-            and q.shape[-1] in (64, 128):
+            and q.shape[-1] in (64, 128) \
+            and (bias is None or bias.shape[-1] % 4 == 0):
         bias_n = None if bias is None else bias.to(torch.bfloat16).contiguous()
         out = get_ext(False).flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
                                             causal, scale, bias_n)
