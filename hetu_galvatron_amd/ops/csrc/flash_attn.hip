@@ -327,7 +327,8 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
                       const __bf16* __restrict__ v,
                       __bf16* __restrict__ o, float* __restrict__ lse,
                       int b, int sq, int skv, int hq, int hkv, float scale,
-                      bool causal, const __bf16* __restrict__ bias = nullptr) {
+                      bool causal, const __bf16* __restrict__ bias = nullptr,
+                      bool paired = true) {
   constexpr int KB = 64, QBF = 256;
   constexpr int BUFSZ = KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -346,7 +347,7 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
                              kv_base, lse_base, hq * D, hkv * D, off, sq,
                              skv, sl2e, causal, bias, bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
-  if (causal && qb2 > (int)blockIdx.x) {
+  if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_fwd_block<D, BIASED>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
                                lse_base, hq * D, hkv * D, off, sq, skv,
@@ -627,7 +628,8 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          __bf16* __restrict__ dq,
                          int b, int sq, int skv, int hq, int hkv,
                          float scale, bool causal,
-                         const __bf16* __restrict__ bias = nullptr) {
+                         const __bf16* __restrict__ bias = nullptr,
+                         bool paired = true) {
   constexpr int KB = 32, QBF = 256;
   constexpr int BUFSZ = 2 * KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -646,7 +648,7 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                                 hkv * D, off, sq, skv, scale, causal, bias,
                                 bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
-  if (causal && qb2 > (int)blockIdx.x) {
+  if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_bwd_dq_block<D, BIASED>(qb2, dout, q, k, v, lse, di, dq, smem,
                                   q_base, kv_base, lse_base, hq * D,
@@ -973,7 +975,8 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           int b, int sq, int skv, int hq, int hkv,
                           float scale, bool causal,
                           const __bf16* __restrict__ bias = nullptr,
-                          float* __restrict__ dbias = nullptr) {
+                          float* __restrict__ dbias = nullptr,
+                          bool paired = true) {
   constexpr int QT = 64, KBW = 256;  // must match flash_bwd_dkv_phase
   // dK phase is the larger LDS user: q rows + dO rows + transposed image
   constexpr int BUFSZ = 2 * QT * (D + 8) + D * (QT + 8);
@@ -996,7 +999,7 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                                  hq * D, off, sq, skv, scale, causal, bias,
                                  dbias, bias_base);
   const int kb2 = nkb - 1 - (int)blockIdx.x;
-  if (causal && kb2 > (int)blockIdx.x) {
+  if (causal && paired && kb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_bwd_dkv_block<D, BIASED>(kb2, dout, q, k, v, lse, di, dk_exp,
                                    dv_exp, smem, lsedi, q_base, kv_base,
@@ -1040,14 +1043,18 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                bool causal, const __bf16* bias,
                                hipStream_t st) {
   int nqb = (sq + 255) / 256;
-  dim3 grid(causal ? (nqb + 1) / 2 : nqb, b * hq);
+  // complementary-pair causal scheduling halves the grid; skip it when
+  // the halved grid underfills the 256-CU chip (small-seq shapes)
+  const bool paired = causal && ((nqb + 1) / 2) * (long)b * hq >= 256;
+  dim3 grid((causal && paired) ? (nqb + 1) / 2 : nqb, b * hq);
   if (bias != nullptr)
     hipLaunchKernelGGL((flash_fwd_kernel<D, true>), grid, dim3(512), 0, st,
                        q, k, v, o, lse, b, sq, skv, hq, hkv, scale, causal,
-                       bias);
+                       bias, paired);
   else
     hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k,
-                       v, o, lse, b, sq, skv, hq, hkv, scale, causal);
+                       v, o, lse, b, sq, skv, hq, hkv, scale, causal,
+                       nullptr, paired);
 }
 
 void flash_fwd_launch(const __bf16* q, const __bf16* k, const __bf16* v,
@@ -1083,23 +1090,25 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                bool causal, const __bf16* bias, float* dbias,
                                hipStream_t st) {
   int nqb = (sq + 255) / 256;
-  dim3 gq(causal ? (nqb + 1) / 2 : nqb, b * hq);
+  const bool pq = causal && ((nqb + 1) / 2) * (long)b * hq >= 256;
+  dim3 gq((causal && pq) ? (nqb + 1) / 2 : nqb, b * hq);
   int nkb = (skv + 255) / 256;
-  dim3 gkv(causal ? (nkb + 1) / 2 : nkb, b * hq);
+  const bool pkv = causal && ((nkb + 1) / 2) * (long)b * hq >= 256;
+  dim3 gkv((causal && pkv) ? (nkb + 1) / 2 : nkb, b * hq);
   if (bias != nullptr) {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D, true>), gq, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dq, b, sq, skv, hq, hkv,
-                       scale, causal, bias);
+                       scale, causal, bias, pq);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, true>), gkv, dim3(512), 0,
                        st, dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq,
-                       skv, hq, hkv, scale, causal, bias, dbias);
+                       skv, hq, hkv, scale, causal, bias, dbias, pkv);
   } else {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout,
                        q, k, v, lse, di, dq, b, sq, skv, hq, hkv, scale,
-                       causal);
+                       causal, nullptr, pq);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv,
-                       hq, hkv, scale, causal);
+                       hq, hkv, scale, causal, nullptr, nullptr, pkv);
   }
 }
 
