@@ -14,14 +14,14 @@ __global__ void adamw_kernel(float* __restrict__ master,
                              float* __restrict__ m, float* __restrict__ v,
                              TO* __restrict__ out, long n, float lr,
                              float beta1, float beta2, float eps, float wd,
-                             float bc1, float bc2) {
+                             float bc1, float bc2, float gscale) {
   const long n4 = n & ~3L;
   for (long i = (blockIdx.x * (long)blockDim.x + threadIdx.x) * 4; i < n4;
        i += (long)gridDim.x * blockDim.x * 4) {
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
       const long j = i + k;
-      const float g = (float)grad[j];
+      const float g = (float)grad[j] * gscale;
       float mv = m[j] = beta1 * m[j] + (1.f - beta1) * g;
       float vv = v[j] = beta2 * v[j] + (1.f - beta2) * g * g;
       const float denom = sqrtf(vv / bc2) + eps;
@@ -34,7 +34,7 @@ __global__ void adamw_kernel(float* __restrict__ master,
   // tail
   for (long j = n4 + blockIdx.x * (long)blockDim.x + threadIdx.x; j < n;
        j += (long)gridDim.x * blockDim.x) {
-    const float g = (float)grad[j];
+    const float g = (float)grad[j] * gscale;
     float mv = m[j] = beta1 * m[j] + (1.f - beta1) * g;
     float vv = v[j] = beta2 * v[j] + (1.f - beta2) * g * g;
     const float denom = sqrtf(vv / bc2) + eps;
@@ -67,7 +67,49 @@ __global__ void grad_accum_kernel(float* __restrict__ flat,
     flat[i] += (float)g[i];
 }
 
+// Multi-tensor sum-of-squares for the global grad norm: one launch over
+// ALL flat grad shards (replaces the per-block eager pow+reduce chain —
+// ~105 launch pairs per step on the 8B bench).  chunks: [n,3] int64
+// (ptr, offset, count); partial[] accumulated with fp32 atomics.
+__global__ void multi_sumsq_kernel(const long* __restrict__ chunks,
+                                   int n_chunks, float* __restrict__ out) {
+  float acc = 0.f;
+  for (int c = blockIdx.x; c < n_chunks; c += gridDim.x) {
+    const float* p = reinterpret_cast<const float*>(chunks[3 * c]) +
+                     chunks[3 * c + 1];
+    const long cnt = chunks[3 * c + 2];
+    const long cnt4 = cnt & ~3L;
+    for (long i = threadIdx.x * 4L; i < cnt4; i += (long)blockDim.x * 4) {
+      const f32x4 g4 = *reinterpret_cast<const f32x4*>(p + i);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) acc += g4[k] * g4[k];
+    }
+    for (long i = cnt4 + threadIdx.x; i < cnt; i += blockDim.x) {
+      const float g = p[i];
+      acc += g * g;
+    }
+  }
+  acc = wave_sum(acc);
+  __shared__ float warp_acc[8];
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) warp_acc[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += warp_acc[w];
+    atomicAdd(out, t);
+  }
+}
+
 }  // namespace
+
+void multi_sumsq_launch(const long* chunks, int n_chunks, float* out,
+                        hipStream_t st) {
+  int grid = galv_grid(n_chunks * 4);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL((multi_sumsq_kernel), dim3(grid), dim3(512), 0, st,
+                     chunks, n_chunks, out);
+}
 
 template <typename TG>
 void grad_accum_launch_t(float* flat, const TG* g, long n, hipStream_t st) {
@@ -83,17 +125,18 @@ template void grad_accum_launch_t<float>(float*, const float*, long, hipStream_t
 template <typename TG, typename TO>
 void adamw_launch_t(float* master, const TG* grad, float* m, float* v,
                     TO* out, long n, int step, float lr, float beta1,
-                    float beta2, float eps, float wd, hipStream_t st) {
+                    float beta2, float eps, float wd, float gscale,
+                    hipStream_t st) {
   const float bc1 = 1.f - powf(beta1, (float)step);
   const float bc2 = 1.f - powf(beta2, (float)step);
   int grid = galv_grid((n / 4 + 255) / 256);
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL((adamw_kernel<TG, TO>), dim3(grid), dim3(256), 0, st,
                      master, grad, m, v, out, n, lr, beta1, beta2, eps, wd,
-                     bc1, bc2);
+                     bc1, bc2, gscale);
 }
 
-template void adamw_launch_t<float, float>(float*, const float*, float*, float*, float*, long, int, float, float, float, float, float, hipStream_t);
-template void adamw_launch_t<float, __bf16>(float*, const float*, float*, float*, __bf16*, long, int, float, float, float, float, float, hipStream_t);
-template void adamw_launch_t<__bf16, float>(float*, const __bf16*, float*, float*, float*, long, int, float, float, float, float, float, hipStream_t);
-template void adamw_launch_t<__bf16, __bf16>(float*, const __bf16*, float*, float*, __bf16*, long, int, float, float, float, float, float, hipStream_t);
+template void adamw_launch_t<float, float>(float*, const float*, float*, float*, float*, long, int, float, float, float, float, float, float, hipStream_t);
+template void adamw_launch_t<float, __bf16>(float*, const float*, float*, float*, __bf16*, long, int, float, float, float, float, float, float, hipStream_t);
+template void adamw_launch_t<__bf16, float>(float*, const __bf16*, float*, float*, float*, long, int, float, float, float, float, float, float, hipStream_t);
+template void adamw_launch_t<__bf16, __bf16>(float*, const __bf16*, float*, float*, __bf16*, long, int, float, float, float, float, float, float, hipStream_t);

@@ -31,7 +31,7 @@ void rope_launch_t(const T*, T*, const float*, const float*, long, int, int, boo
 template <typename TG>
 void grad_accum_launch_t(float*, const TG*, long, hipStream_t);
 template <typename TG, typename TO>
-void adamw_launch_t(float*, const TG*, float*, float*, TO*, long, int, float, float, float, float, float, hipStream_t);
+void adamw_launch_t(float*, const TG*, float*, float*, TO*, long, int, float, float, float, float, float, float, hipStream_t);
 template <typename T>
 void ce_max_launch_t(const T*, float*, long, long, hipStream_t);
 template <typename T>
@@ -54,6 +54,7 @@ DECL_MOE(f32_t)
 void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr);
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
 void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, float* dbias = nullptr);
+void multi_sumsq_launch(const long*, int, float*, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
 void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, const int*, hipStream_t);
 
@@ -555,7 +556,7 @@ Tensor grouped_gemm_dw(const Tensor& A, const Tensor& dC,
 void fused_adamw(std::vector<Tensor> masters, std::vector<Tensor> grads,
                  std::vector<Tensor> ms, std::vector<Tensor> vs,
                  std::vector<Tensor> outs, long step, double lr, double beta1,
-                 double beta2, double eps, double wd) {
+                 double beta2, double eps, double wd, double gscale = 1.0) {
   auto st = cur_stream();
   for (size_t i = 0; i < masters.size(); ++i) {
     auto& m = masters[i];
@@ -571,23 +572,50 @@ void fused_adamw(std::vector<Tensor> masters, std::vector<Tensor> grads,
     if (g_bf && o_bf)
       adamw_launch_t<__bf16, __bf16>(mp, bfp(grads[i]), ma, va,
                                      bfp_mut(outs[i]), n, (int)step, lr,
-                                     beta1, beta2, eps, wd, st);
+                                     beta1, beta2, eps, wd, (float)gscale, st);
     else if (g_bf && !o_bf)
       adamw_launch_t<__bf16, float>(mp, bfp(grads[i]), ma, va,
                                     outs[i].data_ptr<float>(), n, (int)step,
-                                    lr, beta1, beta2, eps, wd, st);
+                                    lr, beta1, beta2, eps, wd, (float)gscale, st);
     else if (!g_bf && o_bf)
       adamw_launch_t<float, __bf16>(mp, grads[i].data_ptr<float>(), ma,
                                     va, bfp_mut(outs[i]), n, (int)step, lr,
-                                    beta1, beta2, eps, wd, st);
+                                    beta1, beta2, eps, wd, (float)gscale, st);
     else
       adamw_launch_t<float, float>(mp, grads[i].data_ptr<float>(), ma,
                                    va, outs[i].data_ptr<float>(), n,
-                                   (int)step, lr, beta1, beta2, eps, wd, st);
+                                   (int)step, lr, beta1, beta2, eps, wd, (float)gscale, st);
   }
 }
 
 }  // namespace
+
+// ---- multi-tensor grad-norm sum of squares --------------------------------
+Tensor multi_sumsq(std::vector<Tensor> grads) {
+  TORCH_CHECK(!grads.empty(), "multi_sumsq: empty list");
+  auto dev = grads[0].device();
+  constexpr long CHUNK = 1 << 20;
+  std::vector<long> meta;
+  for (auto& g : grads) {
+    CHECK_GPU(g);
+    TORCH_CHECK(g.scalar_type() == at::kFloat && g.is_contiguous(),
+                "multi_sumsq: fp32 contiguous only");
+    const long n = g.numel();
+    const long base = (long)(intptr_t)g.data_ptr<float>();
+    for (long off = 0; off < n; off += CHUNK) {
+      meta.push_back(base);
+      meta.push_back(off);
+      meta.push_back(std::min(CHUNK, n - off));
+    }
+  }
+  auto meta_t = at::from_blob(meta.data(), {(long)meta.size()},
+                              at::TensorOptions().dtype(at::kLong))
+                    .to(dev, /*non_blocking=*/false);
+  auto out = at::zeros({1}, grads[0].options().dtype(at::kFloat));
+  multi_sumsq_launch(meta_t.data_ptr<long>(), (int)(meta.size() / 3),
+                     out.data_ptr<float>(), cur_stream());
+  return out;
+}
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -605,7 +633,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_max", &ce_max);
   m.def("ce_sum_target", &ce_sum_target);
   m.def("ce_bwd", &ce_bwd);
-  m.def("fused_adamw", &fused_adamw);
+  m.def("fused_adamw", &fused_adamw, py::arg("masters"), py::arg("grads"),
+        py::arg("ms"), py::arg("vs"), py::arg("outs"), py::arg("step"),
+        py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("wd"), py::arg("gscale") = 1.0);
+  m.def("multi_sumsq", &multi_sumsq);
   m.def("grad_accum", &grad_accum);
   m.def("grouped_gemm", &grouped_gemm);
   m.def("moe_permute", &moe_permute);

@@ -42,9 +42,22 @@ class GalvatronOptimizer:
 
     def _global_grad_norm(self) -> float:
         dev = self.blocks[0].device if self.blocks else torch.device("cpu")
-        total = torch.zeros((), dtype=torch.float32, device=dev)
-        for b in self.blocks:
-            total = total + b.grad_sumsq_owned()
+        if (native_available() and dev.type == "cuda" and self.blocks
+                and all(b.grad_shard is not None for b in self.blocks)):
+            # one fused multi-tensor Σg² launch over every owned shard
+            # (the eager per-block pow+reduce chain was ~2 launches per
+            # block per step), plus tiny eager corrections for
+            # tp-replicated segments
+            views = [b.owned_grad_view() for b in self.blocks]
+            total = get_ext(False).multi_sumsq(views)[0]
+            for b in self.blocks:
+                c = b.sumsq_tp_correction()
+                if c is not None:
+                    total = total + c
+        else:
+            total = torch.zeros((), dtype=torch.float32, device=dev)
+            for b in self.blocks:
+                total = total + b.grad_sumsq_owned()
         if dist.is_initialized() and dist.get_world_size() > 1:
             dist.all_reduce(total)
         return float(total.sqrt().item())
@@ -71,10 +84,19 @@ class GalvatronOptimizer:
         return master, grad, b.exp_avg, b.exp_avg_sq, bf16_out
 
     def step(self) -> float:
-        norm = self.clip_gradients()
-        self.step_count += 1
         use_native = (self.use_fused and native_available()
                       and self.blocks and self.blocks[0].device.type == "cuda")
+        norm = self._global_grad_norm()
+        self.last_grad_norm = norm
+        scale = 1.0
+        if self.clip_grad and self.clip_grad > 0 and norm > self.clip_grad:
+            scale = self.clip_grad / (norm + 1e-6)
+            if not use_native:
+                # fallback path applies the clip as a separate pass; the
+                # native path folds it into the adam kernel (gscale)
+                for b in self.blocks:
+                    b.scale_grads(scale)
+        self.step_count += 1
         masters, grads, ms, vs, outs = [], [], [], [], []
         for b in self.blocks:
             t = self._block_tensors(b)
@@ -83,7 +105,8 @@ class GalvatronOptimizer:
         if use_native:
             get_ext(False).fused_adamw(
                 masters, grads, ms, vs, outs, self.step_count, self.lr,
-                self.betas[0], self.betas[1], self.eps, self.weight_decay)
+                self.betas[0], self.betas[1], self.eps, self.weight_decay,
+                scale)
         else:
             ref.adamw_step(outs, grads, ms, vs, masters, self.step_count,
                            self.lr, self.betas[0], self.betas[1], self.eps,

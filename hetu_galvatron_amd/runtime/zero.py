@@ -347,6 +347,38 @@ class FlatParamBlock:
             return self.master, self.grad_shard
         return self.master, self.grad_shard
 
+    def owned_grad_view(self) -> torch.Tensor:
+        """This rank's OWNED fp32 reduced-grad elements (1-D contiguous)."""
+        assert self.grad_shard is not None, "start/finish_grad_sync first"
+        if self.mode == "ddp":
+            return self.grad_shard[self._shard_slice()]
+        return self.grad_shard
+
+    def _owned_base(self) -> int:
+        if self.mode == "ddp":
+            return self._shard_slice().start
+        return self.rank_in_group * self.shard_size
+
+    def sumsq_tp_correction(self) -> Optional[torch.Tensor]:
+        """Correction making Σg² count tp-replicated segments (SP norms)
+        once per model element: -part + part/tp; None when no-op."""
+        tp = self.tp_group.size if self.tp_group is not None else 1
+        if tp <= 1 or not any(s.tp_replicated for s in self.segments):
+            return None
+        g = self.owned_grad_view()
+        base = self._owned_base()
+        corr = None
+        for seg in self.segments:
+            if not seg.tp_replicated:
+                continue
+            lo = max(seg.offset, base)
+            hi = min(seg.offset + seg.numel, base + g.numel())
+            if lo < hi:
+                part = (g[lo - base:hi - base].float() ** 2).sum()
+                c = part / tp - part
+                corr = c if corr is None else corr + c
+        return corr
+
     def grad_sumsq_owned(self) -> torch.Tensor:
         """Sum of squares of grads over elements OWNED by this rank (each
         model element counted exactly once across the world):
@@ -358,26 +390,10 @@ class FlatParamBlock:
         vanishing fraction of the norm and the count is identical on all
         ranks, so clipping stays globally consistent.
         """
-        assert self.grad_shard is not None
-        if self.mode == "ddp":
-            sl = self._shard_slice()
-            g = self.grad_shard[sl]
-            base = sl.start
-        else:
-            g = self.grad_shard
-            base = self.rank_in_group * self.shard_size
+        g = self.owned_grad_view()
         total = (g.float() ** 2).sum()
-        tp = self.tp_group.size if self.tp_group is not None else 1
-        if tp > 1:
-            for seg in self.segments:
-                if not seg.tp_replicated:
-                    continue
-                lo = max(seg.offset, base)
-                hi = min(seg.offset + seg.numel, base + g.numel())
-                if lo < hi:
-                    part = (g[lo - base:hi - base].float() ** 2).sum()
-                    total = total - part + part / tp
-        return total
+        corr = self.sumsq_tp_correction()
+        return total if corr is None else total + corr
 
     def scale_grads(self, scale: float) -> None:
         if self.grad_shard is not None:

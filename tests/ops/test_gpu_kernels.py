@@ -339,6 +339,37 @@ def test_fused_adamw():
         assert_close(outs[i], outs_ref[i], 1e-2, what=f"adam out {i}")
 
 
+def test_fused_adamw_gscale():
+    """gscale folds the clip factor into the kernel: equals pre-scaling
+    the grads then running with gscale=1."""
+    torch.manual_seed(14)
+    n = 5000
+    master = torch.randn(n, device=dev()).float()
+    g = torch.randn(n, device=dev()).bfloat16()
+    m = torch.rand(n, device=dev()).float() * 0.1
+    v = torch.rand(n, device=dev()).float() * 0.01
+    out = torch.zeros(n, device=dev()).bfloat16()
+    m2, v2 = m.clone(), v.clone()
+    master2, out2 = master.clone(), out.clone()
+    s = 0.37
+    ext().fused_adamw([master], [g], [m], [v], [out], 2, 1e-3, 0.9, 0.95,
+                      1e-8, 0.01, s)
+    ext().fused_adamw([master2], [(g.float() * s).bfloat16()], [m2], [v2],
+                      [out2], 2, 1e-3, 0.9, 0.95, 1e-8, 0.01, 1.0)
+    # bf16 re-quantization of the pre-scaled grads is the only difference
+    assert_close(master, master2, 2e-3, what="gscale master")
+    assert_close(m, m2, 2e-3, what="gscale m")
+
+
+def test_multi_sumsq():
+    torch.manual_seed(15)
+    xs = [torch.randn(n, device=dev()).float()
+          for n in (17, 4096, 1 << 21, 123457)]
+    got = ext().multi_sumsq(xs)[0]
+    want = sum((x.double() ** 2).sum() for x in xs)
+    assert abs(float(got) - float(want)) / float(want) < 1e-5, (got, want)
+
+
 def test_grad_accum():
     torch.manual_seed(10)
     flat = torch.randn(10000, device=dev()).float()
