@@ -14,9 +14,9 @@
 
 // ---- launcher prototypes (defined in the sibling .hip TUs) ----------------
 template <typename T>
-void rmsnorm_fwd_launch_t(const T*, const T*, T*, float*, long, int, float, hipStream_t);
+void rmsnorm_fwd_launch_t(const T*, const T*, T*, float*, long, int, float, hipStream_t, const T* res = nullptr, T* sum_out = nullptr);
 template <typename T>
-void rmsnorm_bwd_launch_t(const T*, const T*, const T*, const float*, T*, float*, float*, long, int, hipStream_t);
+void rmsnorm_bwd_launch_t(const T*, const T*, const T*, const float*, T*, float*, float*, long, int, hipStream_t, const T* dsum_in = nullptr);
 template <typename T>
 void layernorm_fwd_launch_t(const T*, const T*, const T*, T*, float*, float*, long, int, float, hipStream_t);
 template <typename T>
@@ -118,6 +118,51 @@ std::tuple<Tensor, Tensor> rmsnorm_bwd(const Tensor& dy, const Tensor& x,
                                 w.data_ptr<float>(), inv.data_ptr<float>(),
                                 dx.data_ptr<float>(), dw_part.data_ptr<float>(),
                                 dw.data_ptr<float>(), n, H, cur_stream());
+  return {dx, dw};
+}
+
+// fused residual-add + rmsnorm: y = rmsnorm(x + res) * w, also returns
+// the sum (the downstream residual stream) and invrms
+std::tuple<Tensor, Tensor, Tensor> add_rmsnorm_fwd(const Tensor& x,
+                                                   const Tensor& res,
+                                                   const Tensor& w,
+                                                   double eps) {
+  CHECK_GPU(x); CHECK_GPU(res);
+  TORCH_CHECK(is_bf16(x) && is_bf16(res), "add_rmsnorm: bf16 only");
+  const int H = x.size(-1);
+  const long n = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0 && H <= 16384, "rmsnorm: H must be %8, <=16k");
+  auto y = at::empty_like(x);
+  auto sum = at::empty_like(x);
+  auto invrms = at::empty({n}, x.options().dtype(at::kFloat));
+  rmsnorm_fwd_launch_t<__bf16>(bfp(x), bfp(w), bfp_mut(y),
+                               invrms.data_ptr<float>(), n, H, (float)eps,
+                               cur_stream(), bfp(res), bfp_mut(sum));
+  auto sizes = x.sizes().vec();
+  sizes.pop_back();
+  return {y, sum, invrms.view(sizes)};
+}
+
+// backward with the residual-stream grad folded in:
+// dx = rmsnorm_bwd(dy wrt sum) + dsum  (dx == dres)
+std::tuple<Tensor, Tensor> add_rmsnorm_bwd(const Tensor& dy,
+                                           const Tensor& dsum,
+                                           const Tensor& sum,
+                                           const Tensor& w,
+                                           const Tensor& invrms) {
+  CHECK_GPU(dy);
+  const int H = sum.size(-1);
+  const long n = sum.numel() / H;
+  auto dx = at::empty_like(sum);
+  auto dw = at::zeros({H}, sum.options().dtype(at::kFloat));
+  const int G = norm_bwd_grid(n);
+  auto dw_part = at::empty({G, H}, sum.options().dtype(at::kFloat));
+  auto inv = invrms.contiguous();
+  rmsnorm_bwd_launch_t<__bf16>(bfp(dy), bfp(sum), bfp(w),
+                               inv.data_ptr<float>(), bfp_mut(dx),
+                               dw_part.data_ptr<float>(),
+                               dw.data_ptr<float>(), n, H, cur_stream(),
+                               bfp(dsum));
   return {dx, dw};
 }
 
@@ -620,6 +665,8 @@ Tensor multi_sumsq(std::vector<Tensor> grads) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
+  m.def("add_rmsnorm_bwd", &add_rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);

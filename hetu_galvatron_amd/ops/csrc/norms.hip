@@ -16,12 +16,17 @@
 
 namespace {
 
+// res/sum_out (optional): fused residual-add — normalizes (x + res) and
+// writes the sum for the downstream residual stream (the reference's
+// DropoutAddLayerNorm fusion, norm.py:3-30)
 template <typename T, int VPT>
 __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x,
                                    const T* __restrict__ w,
                                    T* __restrict__ y,
                                    float* __restrict__ invrms,
-                                   long n, int H, float eps) {
+                                   long n, int H, float eps,
+                                   const T* __restrict__ res = nullptr,
+                                   T* __restrict__ sum_out = nullptr) {
   __shared__ float red[4];
   const int tid = threadIdx.x;
   for (long row = blockIdx.x; row < n; row += gridDim.x) {
@@ -34,6 +39,13 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x,
       int c = (tid + p * 256) * 8;
       if (c < H) {
         VecIO<T>::load(xv[p], xr + c);
+        if (res != nullptr) {
+          float rv[8];
+          VecIO<T>::load(rv, res + row * (long)H + c);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) xv[p][i] += rv[i];
+          VecIO<T>::store(sum_out + row * (long)H + c, xv[p]);
+        }
 #pragma unroll
         for (int i = 0; i < 8; ++i) ss += xv[p][i] * xv[p][i];
       }
@@ -61,7 +73,8 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
                                    const float* __restrict__ invrms,
                                    T* __restrict__ dx,
                                    float* __restrict__ dw_part,  // [grid][H]
-                                   long n, int H) {
+                                   long n, int H,
+                                   const T* __restrict__ dsum_in = nullptr) {
   __shared__ float red[4];
   const int tid = threadIdx.x;
   float dw_acc[VPT][8];
@@ -106,6 +119,12 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
           const float xhat = xv[p][i] * inv;
           out[i] = inv * (dyv[p][i] * wv[p][i] - xhat * c1);
           dw_acc[p][i] += dyv[p][i] * xhat;
+        }
+        if (dsum_in != nullptr) {
+          float dsv[8];
+          VecIO<T>::load(dsv, dsum_in + row * (long)H + c);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) out[i] += dsv[i];
         }
         VecIO<T>::store(dxr + c, out);
       }
@@ -282,10 +301,12 @@ struct NormDispatch {};
 
 template <typename T>
 void rmsnorm_fwd_launch_t(const T* x, const T* w, T* y, float* invrms,
-                          long n, int H, float eps, hipStream_t s) {
+                          long n, int H, float eps, hipStream_t s,
+                          const T* res, T* sum_out) {
   int grid = galv_grid(n);
 #define RUN(V) hipLaunchKernelGGL((rmsnorm_fwd_kernel<T, V>), dim3(grid), \
-                                  dim3(256), 0, s, x, w, y, invrms, n, H, eps)
+                                  dim3(256), 0, s, x, w, y, invrms, n, H, \
+                                  eps, res, sum_out)
   DISPATCH_VPT(H, RUN);
 #undef RUN
 }
@@ -295,10 +316,12 @@ int norm_bwd_grid(long n) { return galv_grid(n, 2048); }
 template <typename T>
 void rmsnorm_bwd_launch_t(const T* dy, const T* x, const T* w,
                           const float* invrms, T* dx, float* dw_part,
-                          float* dw, long n, int H, hipStream_t s) {
+                          float* dw, long n, int H, hipStream_t s,
+                          const T* dsum_in) {
   int grid = norm_bwd_grid(n);
 #define RUN(V) hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, V>), dim3(grid), \
-                                  dim3(256), 0, s, dy, x, w, invrms, dx, dw_part, n, H)
+                                  dim3(256), 0, s, dy, x, w, invrms, dx, \
+                                  dw_part, n, H, dsum_in)
   DISPATCH_VPT(H, RUN);
 #undef RUN
   int gy = grid > 64 ? 32 : 1;
@@ -335,10 +358,10 @@ void layernorm_bwd_launch_t(const T* dy, const T* x, const T* w,
 }
 
 // explicit instantiations used by bindings.cpp
-template void rmsnorm_fwd_launch_t<__bf16>(const __bf16*, const __bf16*, __bf16*, float*, long, int, float, hipStream_t);
-template void rmsnorm_fwd_launch_t<float>(const float*, const float*, float*, float*, long, int, float, hipStream_t);
-template void rmsnorm_bwd_launch_t<__bf16>(const __bf16*, const __bf16*, const __bf16*, const float*, __bf16*, float*, float*, long, int, hipStream_t);
-template void rmsnorm_bwd_launch_t<float>(const float*, const float*, const float*, const float*, float*, float*, float*, long, int, hipStream_t);
+template void rmsnorm_fwd_launch_t<__bf16>(const __bf16*, const __bf16*, __bf16*, float*, long, int, float, hipStream_t, const __bf16*, __bf16*);
+template void rmsnorm_fwd_launch_t<float>(const float*, const float*, float*, float*, long, int, float, hipStream_t, const float*, float*);
+template void rmsnorm_bwd_launch_t<__bf16>(const __bf16*, const __bf16*, const __bf16*, const float*, __bf16*, float*, float*, long, int, hipStream_t, const __bf16*);
+template void rmsnorm_bwd_launch_t<float>(const float*, const float*, const float*, const float*, float*, float*, float*, long, int, hipStream_t, const float*);
 template void layernorm_fwd_launch_t<__bf16>(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, float*, long, int, float, hipStream_t);
 template void layernorm_fwd_launch_t<float>(const float*, const float*, const float*, float*, float*, float*, long, int, float, hipStream_t);
 template void layernorm_bwd_launch_t<__bf16>(const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, float*, float*, long, int, hipStream_t);

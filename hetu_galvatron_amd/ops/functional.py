@@ -45,6 +45,32 @@ def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.
     return _RMSNorm.apply(x.contiguous(), weight, eps)
 
 
+class _AddRMSNorm(torch.autograd.Function):
+    """Fused residual-add + RMSNorm: (y, sum) = (rmsnorm(x+res)*w, x+res).
+    One kernel each way — removes the standalone residual-add in forward
+    AND the grad-join add in backward (the reference fuses the same way
+    via flash-attn's DropoutAddLayerNorm, norm.py:3-30)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, eps):
+        y, s, invrms = get_ext(False).add_rmsnorm_fwd(x, res, weight, eps)
+        ctx.save_for_backward(s, weight, invrms)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, dsum):
+        s, weight, invrms = ctx.saved_tensors
+        dx, dw = get_ext(False).add_rmsnorm_bwd(
+            dy.contiguous(), dsum.contiguous(), s, weight, invrms)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def fused_add_rms_norm(x, res, weight, eps: float = 1e-5):
+    """Returns (normed, sum).  Native bf16 path only; callers fall back to
+    `rms_norm(x + res)` elsewhere."""
+    return _AddRMSNorm.apply(x.contiguous(), res.contiguous(), weight, eps)
+
+
 class _LayerNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):

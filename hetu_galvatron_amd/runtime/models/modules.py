@@ -157,9 +157,15 @@ class GalvatronDecoderLayer(nn.Module):
         x = self.attention(x, cos, sin)
         if self.dropout_p > 0 and self.training:
             x = F.dropout(x, self.dropout_p)
-        hidden = residual + x
-        residual = hidden
-        x = self.post_attn_norm(hidden)
+        if (self.dropout_p == 0 or not self.training) \
+                and hasattr(self.post_attn_norm, "forward_fused_add") \
+                and x.shape == residual.shape:
+            # fused residual-add + norm (one kernel each way)
+            x, residual = self.post_attn_norm.forward_fused_add(x, residual)
+        else:
+            hidden = residual + x
+            residual = hidden
+            x = self.post_attn_norm(hidden)
         x = self.mlp(x)
         if self.dropout_p > 0 and self.training:
             x = F.dropout(x, self.dropout_p)

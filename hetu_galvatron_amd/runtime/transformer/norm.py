@@ -27,6 +27,16 @@ class RMSNorm(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return rms_norm(x, self.weight, self.eps)
 
+    def forward_fused_add(self, x, res):
+        """(normed, sum=x+res) in one kernel each way on the native path;
+        eager equivalent elsewhere."""
+        from ...ops._ext import use_native
+        if use_native(x) and x.dtype == torch.bfloat16:
+            from ...ops import fused_add_rms_norm
+            return fused_add_rms_norm(x, res, self.weight, self.eps)
+        s = x + res
+        return rms_norm(s, self.weight, self.eps), s
+
 
 class LayerNorm(nn.Module):
     def __init__(self, hidden_size: int, eps: float = 1e-5, dtype=None):

@@ -339,6 +339,34 @@ def test_fused_adamw():
         assert_close(outs[i], outs_ref[i], 1e-2, what=f"adam out {i}")
 
 
+def test_add_rmsnorm_fused():
+    """Fused residual-add + rmsnorm vs the unfused composition, fwd+bwd
+    (incl. the dsum fold in backward)."""
+    torch.manual_seed(16)
+    n, H = 1024, 4096
+    x = torch.randn(n, H, device=dev()).bfloat16().requires_grad_(True)
+    r = torch.randn(n, H, device=dev()).bfloat16().requires_grad_(True)
+    w = torch.randn(H, device=dev()).bfloat16().requires_grad_(True)
+    from hetu_galvatron_amd.ops import fused_add_rms_norm, rms_norm
+    y, s = fused_add_rms_norm(x, r, w, 1e-5)
+    x2 = x.detach().requires_grad_(True)
+    r2 = r.detach().requires_grad_(True)
+    w2 = w.detach().requires_grad_(True)
+    s2 = x2 + r2
+    y2 = rms_norm(s2, w2, 1e-5)
+    assert_close(y, y2, 3e-2, what="addnorm y")
+    assert_close(s, s2, 1e-6, what="addnorm sum")
+    dy = torch.randn_like(y)
+    ds = torch.randn_like(s)
+    (y.float() * dy.float()).sum().backward(retain_graph=True)
+    (s.float() * ds.float()).sum().backward()
+    (y2.float() * dy.float()).sum().backward(retain_graph=True)
+    (s2.float() * ds.float()).sum().backward()
+    assert_close(x.grad, x2.grad, 3e-2, what="addnorm dx")
+    assert_close(r.grad, r2.grad, 3e-2, what="addnorm dres")
+    assert_close(w.grad, w2.grad, 1e-1, rtol=3e-2, what="addnorm dw")
+
+
 def test_fused_adamw_gscale():
     """gscale folds the clip factor into the kernel: equals pre-scaling
     the grads then running with gscale=1."""
