@@ -51,9 +51,9 @@ typedef float f32_t;
 DECL_MOE(bf16_t)
 DECL_MOE(f32_t)
 
-void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr);
-void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t);
-void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, float* dbias = nullptr);
+void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, bool sbhd = false);
+void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t, bool sbhd = false);
+void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, float* dbias = nullptr, bool sbhd = false);
 void multi_sumsq_launch(const long*, int, float*, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
 void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, const int*, hipStream_t);
@@ -276,12 +276,16 @@ Tensor rope_fwd(const Tensor& x, const Tensor& cos_t, const Tensor& sin_t,
 // ---- flash attention ------------------------------------------------------
 std::tuple<Tensor, Tensor> flash_attn_fwd(
     const Tensor& q, const Tensor& k, const Tensor& v, bool causal,
-    double scale, const c10::optional<Tensor>& bias = c10::nullopt) {
+    double scale, const c10::optional<Tensor>& bias = c10::nullopt,
+    bool sbhd = false) {
   CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v);
   TORCH_CHECK(is_bf16(q), "flash_attn: bf16 only on the native path");
-  TORCH_CHECK(q.dim() == 4, "flash_attn: q must be [b,s,h,d]");
-  const int b = q.size(0), sq = q.size(1), hq = q.size(2), d = q.size(3);
-  const int skv = k.size(1), hkv = k.size(2);
+  TORCH_CHECK(q.dim() == 4, "flash_attn: q must be [b,s,h,d] (or [s,b,h,d] with sbhd)");
+  const int b = sbhd ? q.size(1) : q.size(0);
+  const int sq = sbhd ? q.size(0) : q.size(1);
+  const int hq = q.size(2), d = q.size(3);
+  const int skv = sbhd ? k.size(0) : k.size(1);
+  const int hkv = k.size(2);
   TORCH_CHECK(d == 64 || d == 128, "flash_attn: head dim must be 64|128");
   TORCH_CHECK(hq % hkv == 0, "flash_attn: GQA needs hq % hkv == 0");
   const __bf16* bp = nullptr;
@@ -298,24 +302,29 @@ std::tuple<Tensor, Tensor> flash_attn_fwd(
   auto lse = at::empty({b, hq, sq}, q.options().dtype(at::kFloat));
   flash_fwd_launch(bfp(q), bfp(k), bfp(v), bfp_mut(o),
                    lse.data_ptr<float>(), b, sq, skv, hq, hkv, d,
-                   (float)scale, causal, cur_stream(), bp);
+                   (float)scale, causal, cur_stream(), bp, sbhd);
   return {o, lse};
 }
 
 std::vector<Tensor> flash_attn_bwd(
     const Tensor& dout, const Tensor& q, const Tensor& k, const Tensor& v,
     const Tensor& o, const Tensor& lse, bool causal, double scale,
-    const c10::optional<Tensor>& bias = c10::nullopt) {
+    const c10::optional<Tensor>& bias = c10::nullopt, bool sbhd = false) {
   CHECK_GPU(dout); CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v); CHECK_GPU(o);
-  const int b = q.size(0), sq = q.size(1), hq = q.size(2), d = q.size(3);
-  const int skv = k.size(1), hkv = k.size(2);
+  const int b = sbhd ? q.size(1) : q.size(0);
+  const int sq = sbhd ? q.size(0) : q.size(1);
+  const int hq = q.size(2), d = q.size(3);
+  const int skv = sbhd ? k.size(0) : k.size(1);
+  const int hkv = k.size(2);
   auto lsec = lse.contiguous();
   auto di = at::empty({b, hq, sq}, q.options().dtype(at::kFloat));
   attn_di_launch(bfp(dout), bfp(o), di.data_ptr<float>(), b, sq, hq, d,
-                 cur_stream());
+                 cur_stream(), sbhd);
   auto dq = at::empty_like(q);
-  auto dk_exp = at::empty({b, skv, hq, d}, k.options());
-  auto dv_exp = at::empty({b, skv, hq, d}, v.options());
+  auto dk_exp = sbhd ? at::empty({skv, b, hq, d}, k.options())
+                     : at::empty({b, skv, hq, d}, k.options());
+  auto dv_exp = sbhd ? at::empty({skv, b, hq, d}, v.options())
+                     : at::empty({b, skv, hq, d}, v.options());
   const __bf16* bp = nullptr;
   float* dbp = nullptr;
   Tensor bias_c, dbias;
@@ -332,12 +341,18 @@ std::vector<Tensor> flash_attn_bwd(
   flash_bwd_launch(bfp(dout), bfp(q), bfp(k), bfp(v),
                    lsec.data_ptr<float>(), di.data_ptr<float>(),
                    bfp_mut(dq), bfp_mut(dk_exp), bfp_mut(dv_exp), b, sq, skv,
-                   hq, hkv, d, (float)scale, causal, cur_stream(), bp, dbp);
+                   hq, hkv, d, (float)scale, causal, cur_stream(), bp, dbp,
+                   sbhd);
   Tensor dk = dk_exp, dv = dv_exp;
   if (hq != hkv) {
     const int rep = hq / hkv;
-    dk = dk_exp.view({b, skv, hkv, rep, d}).sum(3);
-    dv = dv_exp.view({b, skv, hkv, rep, d}).sum(3);
+    if (sbhd) {
+      dk = dk_exp.view({skv, b, hkv, rep, d}).sum(3);
+      dv = dv_exp.view({skv, b, hkv, rep, d}).sum(3);
+    } else {
+      dk = dk_exp.view({b, skv, hkv, rep, d}).sum(3);
+      dv = dv_exp.view({b, skv, hkv, rep, d}).sum(3);
+    }
   }
   if (bias.has_value())
     return {dq, dk, dv, dbias};
@@ -672,8 +687,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rope_fwd", &rope_fwd);
-  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none());
-  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none());
+  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false);
+  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false);
   m.def("decode_attn", &decode_attn);
   m.def("decode_attn_graph", &decode_attn_graph);
   m.def("mfma_probe", &mfma_probe);

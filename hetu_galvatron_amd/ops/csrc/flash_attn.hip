@@ -328,7 +328,7 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
                       __bf16* __restrict__ o, float* __restrict__ lse,
                       int b, int sq, int skv, int hq, int hkv, float scale,
                       bool causal, const __bf16* __restrict__ bias = nullptr,
-                      bool paired = true) {
+                      bool paired = true, bool sbhd = false) {
   constexpr int KB = 64, QBF = 256;
   constexpr int BUFSZ = KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -336,21 +336,27 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const int bi = bh / hq;
   const int h = bh - bi * hq;
   const int hk = h / (hq / hkv);
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
+  // sbhd: tensors are [s, b, h, d] (the runtime's native activation
+  // layout) — only the per-(b,h) base and the s-stride change
+  const long q_base = sbhd ? ((long)bi * hq + h) * D
+                           : ((long)bi * sq * hq + h) * D;
+  const long kv_base = sbhd ? ((long)bi * hkv + hk) * D
+                            : ((long)bi * skv * hkv + hk) * D;
+  const int q_str = (sbhd ? b * hq : hq) * D;
+  const int kv_str = (sbhd ? b * hkv : hkv) * D;
   const long lse_base = ((long)bi * hq + h) * sq;
   const long bias_base = (long)h * sq * skv;  // bias [hq, sq, skv]
   const float sl2e = scale * 1.4426950408889634f;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
   flash_fwd_block<D, BIASED>(blockIdx.x, q, k, v, o, lse, smem, q_base,
-                             kv_base, lse_base, hq * D, hkv * D, off, sq,
+                             kv_base, lse_base, q_str, kv_str, off, sq,
                              skv, sl2e, causal, bias, bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_fwd_block<D, BIASED>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
-                               lse_base, hq * D, hkv * D, off, sq, skv,
+                               lse_base, q_str, kv_str, off, sq, skv,
                                sl2e, causal, bias, bias_base);
   }
 }
@@ -361,8 +367,10 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
 template <int D>
 __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
                                const __bf16* __restrict__ o,
-                               float* __restrict__ di, int b, int sq, int hq) {
-  const long rows = (long)b * sq * hq;  // row index = (bi*sq + s)*hq + h
+                               float* __restrict__ di, int b, int sq, int hq,
+                               bool sbhd = false) {
+  // row index = (bi*sq + s)*hq + h (bshd) or (s*b + bi)*hq + h (sbhd)
+  const long rows = (long)b * sq * hq;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   constexpr int EPL = D / 64;  // elems per lane
@@ -378,10 +386,18 @@ __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
     acc = wave_sum(acc);
     if (lane == 0) {
       // di layout [b, hq, sq]
-      const long bi = row / ((long)sq * hq);
-      const long rem = row - bi * sq * hq;
-      const long s = rem / hq;
-      const long h = rem - s * hq;
+      long bi, s, h;
+      if (sbhd) {
+        s = row / ((long)b * hq);
+        const long rem = row - s * b * hq;
+        bi = rem / hq;
+        h = rem - bi * hq;
+      } else {
+        bi = row / ((long)sq * hq);
+        const long rem = row - bi * sq * hq;
+        s = rem / hq;
+        h = rem - s * hq;
+      }
       di[(bi * hq + h) * sq + s] = acc;
     }
   }
@@ -629,7 +645,7 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          int b, int sq, int skv, int hq, int hkv,
                          float scale, bool causal,
                          const __bf16* __restrict__ bias = nullptr,
-                         bool paired = true) {
+                         bool paired = true, bool sbhd = false) {
   constexpr int KB = 32, QBF = 256;
   constexpr int BUFSZ = 2 * KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -637,22 +653,26 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const int bi = bh / hq;
   const int h = bh - bi * hq;
   const int hk = h / (hq / hkv);
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
+  const long q_base = sbhd ? ((long)bi * hq + h) * D
+                           : ((long)bi * sq * hq + h) * D;
+  const long kv_base = sbhd ? ((long)bi * hkv + hk) * D
+                            : ((long)bi * skv * hkv + hk) * D;
+  const int q_str = (sbhd ? b * hq : hq) * D;
+  const int kv_str = (sbhd ? b * hkv : hkv) * D;
   const long lse_base = ((long)bi * hq + h) * sq;
   const long bias_base = (long)h * sq * skv;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
   flash_bwd_dq_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dq,
-                                smem, q_base, kv_base, lse_base, hq * D,
-                                hkv * D, off, sq, skv, scale, causal, bias,
+                                smem, q_base, kv_base, lse_base, q_str,
+                                kv_str, off, sq, skv, scale, causal, bias,
                                 bias_base);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_bwd_dq_block<D, BIASED>(qb2, dout, q, k, v, lse, di, dq, smem,
-                                  q_base, kv_base, lse_base, hq * D,
-                                  hkv * D, off, sq, skv, scale, causal,
+                                  q_base, kv_base, lse_base, q_str,
+                                  kv_str, off, sq, skv, scale, causal,
                                   bias, bias_base);
   }
 }
@@ -976,7 +996,7 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           float scale, bool causal,
                           const __bf16* __restrict__ bias = nullptr,
                           float* __restrict__ dbias = nullptr,
-                          bool paired = true) {
+                          bool paired = true, bool sbhd = false) {
   constexpr int QT = 64, KBW = 256;  // must match flash_bwd_dkv_phase
   // dK phase is the larger LDS user: q rows + dO rows + transposed image
   constexpr int BUFSZ = 2 * QT * (D + 8) + D * (QT + 8);
@@ -986,25 +1006,32 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const int bi = bh / hq;
   const int h = bh - bi * hq;
   const int hk = h / (hq / hkv);
-  const long q_base = ((long)bi * sq * hq + h) * D;
-  const long kv_base = ((long)bi * skv * hkv + hk) * D;
-  const long dkv_base = ((long)bi * skv * hq + h) * D;
+  const long q_base = sbhd ? ((long)bi * hq + h) * D
+                           : ((long)bi * sq * hq + h) * D;
+  const long kv_base = sbhd ? ((long)bi * hkv + hk) * D
+                            : ((long)bi * skv * hkv + hk) * D;
+  // dk_exp/dv_exp are [b, skv, hq, D] (bshd) or [skv, b, hq, D] (sbhd)
+  const long dkv_base = sbhd ? ((long)bi * hq + h) * D
+                             : ((long)bi * skv * hq + h) * D;
+  const int q_str = (sbhd ? b * hq : hq) * D;
+  const int kv_str = (sbhd ? b * hkv : hkv) * D;
+  const int dkv_str = (sbhd ? b * hq : hq) * D;
   const long lse_base = ((long)bi * hq + h) * sq;
   const int off = skv - sq;
   const int nkb = (skv + KBW - 1) / KBW;
   const long bias_base = (long)h * sq * skv;
   flash_bwd_dkv_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dk_exp,
                                  dv_exp, smem, lsedi, q_base, kv_base,
-                                 dkv_base, lse_base, hq * D, hkv * D,
-                                 hq * D, off, sq, skv, scale, causal, bias,
+                                 dkv_base, lse_base, q_str, kv_str,
+                                 dkv_str, off, sq, skv, scale, causal, bias,
                                  dbias, bias_base);
   const int kb2 = nkb - 1 - (int)blockIdx.x;
   if (causal && paired && kb2 > (int)blockIdx.x) {
     __syncthreads();
     flash_bwd_dkv_block<D, BIASED>(kb2, dout, q, k, v, lse, di, dk_exp,
                                    dv_exp, smem, lsedi, q_base, kv_base,
-                                   dkv_base, lse_base, hq * D, hkv * D,
-                                   hq * D, off, sq, skv, scale, causal,
+                                   dkv_base, lse_base, q_str, kv_str,
+                                   dkv_str, off, sq, skv, scale, causal,
                                    bias, dbias, bias_base);
   }
 }
@@ -1041,7 +1068,7 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                const __bf16* v, __bf16* o, float* lse, int b,
                                int sq, int skv, int hq, int hkv, float scale,
                                bool causal, const __bf16* bias,
-                               hipStream_t st) {
+                               bool sbhd, hipStream_t st) {
   int nqb = (sq + 255) / 256;
   // complementary-pair causal scheduling halves the grid; skip it when
   // the halved grid underfills the 256-CU chip (small-seq shapes)
@@ -1050,35 +1077,36 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
   if (bias != nullptr)
     hipLaunchKernelGGL((flash_fwd_kernel<D, true>), grid, dim3(512), 0, st,
                        q, k, v, o, lse, b, sq, skv, hq, hkv, scale, causal,
-                       bias, paired);
+                       bias, paired, sbhd);
   else
     hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k,
                        v, o, lse, b, sq, skv, hq, hkv, scale, causal,
-                       nullptr, paired);
+                       nullptr, paired, sbhd);
 }
 
 void flash_fwd_launch(const __bf16* q, const __bf16* k, const __bf16* v,
                       __bf16* o, float* lse, int b, int sq, int skv, int hq,
                       int hkv, int d, float scale, bool causal,
-                      hipStream_t st, const __bf16* bias) {
+                      hipStream_t st, const __bf16* bias, bool sbhd) {
   if (d == 64)
     flash_fwd_launch_d<64>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                           causal, bias, st);
+                           causal, bias, sbhd, st);
   else
     flash_fwd_launch_d<128>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                            causal, bias, st);
+                            causal, bias, sbhd, st);
 }
 
 void attn_di_launch(const __bf16* dout, const __bf16* o, float* di, int b,
-                    int sq, int hq, int d, hipStream_t st) {
+                    int sq, int hq, int d, hipStream_t st,
+                    bool sbhd = false) {
   long rows = (long)b * sq * hq;
   int grid = galv_grid((rows + 3) / 4);
   if (d == 64)
     hipLaunchKernelGGL((attn_di_kernel<64>), dim3(grid), dim3(256), 0, st,
-                       dout, o, di, b, sq, hq);
+                       dout, o, di, b, sq, hq, sbhd);
   else
     hipLaunchKernelGGL((attn_di_kernel<128>), dim3(grid), dim3(256), 0, st,
-                       dout, o, di, b, sq, hq);
+                       dout, o, di, b, sq, hq, sbhd);
 }
 
 template <int D>
@@ -1088,7 +1116,7 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                __bf16* dk_exp, __bf16* dv_exp, int b, int sq,
                                int skv, int hq, int hkv, float scale,
                                bool causal, const __bf16* bias, float* dbias,
-                               hipStream_t st) {
+                               bool sbhd, hipStream_t st) {
   int nqb = (sq + 255) / 256;
   const bool pq = causal && ((nqb + 1) / 2) * (long)b * hq >= 256;
   dim3 gq((causal && pq) ? (nqb + 1) / 2 : nqb, b * hq);
@@ -1098,17 +1126,17 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
   if (bias != nullptr) {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D, true>), gq, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dq, b, sq, skv, hq, hkv,
-                       scale, causal, bias, pq);
+                       scale, causal, bias, pq, sbhd);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, true>), gkv, dim3(512), 0,
                        st, dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq,
-                       skv, hq, hkv, scale, causal, bias, dbias, pkv);
+                       skv, hq, hkv, scale, causal, bias, dbias, pkv, sbhd);
   } else {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout,
                        q, k, v, lse, di, dq, b, sq, skv, hq, hkv, scale,
-                       causal, nullptr, pq);
+                       causal, nullptr, pq, sbhd);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv,
-                       hq, hkv, scale, causal, nullptr, nullptr, pkv);
+                       hq, hkv, scale, causal, nullptr, nullptr, pkv, sbhd);
   }
 }
 
@@ -1117,14 +1145,15 @@ void flash_bwd_launch(const __bf16* dout, const __bf16* q, const __bf16* k,
                       __bf16* dq, __bf16* dk_exp, __bf16* dv_exp, int b,
                       int sq, int skv, int hq, int hkv, int d, float scale,
                       bool causal, hipStream_t st, const __bf16* bias,
-                      float* dbias) {
+                      float* dbias, bool sbhd) {
   if (d == 64)
     flash_bwd_launch_d<64>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b, sq,
-                           skv, hq, hkv, scale, causal, bias, dbias, st);
+                           skv, hq, hkv, scale, causal, bias, dbias, sbhd,
+                           st);
   else
     flash_bwd_launch_d<128>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b,
                             sq, skv, hq, hkv, scale, causal, bias, dbias,
-                            st);
+                            sbhd, st);
 }
 
 void mfma_probe_launch(const __bf16* A, const __bf16* B, float* Dst, bool alt,

@@ -147,16 +147,23 @@ class _FlashAttention(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, softmax_scale, window):
+    def forward(ctx, q, k, v, causal, softmax_scale, window, sbhd):
         scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        skv = k.shape[0] if sbhd else k.shape[1]
         # an effective window (window < kv length) routes to the eager
         # reference path (a window argument on the native kernel is a v2
         # item); window >= kv length is a no-op -> native
         eff_window = window if (window is not None and
-                                window < k.shape[1]) else None
+                                window < skv) else None
         if eff_window is None and use_native(q) \
                 and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
-            o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale)
+            o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale,
+                                                   None, sbhd)
+        elif sbhd:
+            ob, lse = ref.attention_fwd(
+                q.permute(1, 0, 2, 3), k.permute(1, 0, 2, 3),
+                v.permute(1, 0, 2, 3), causal, scale, window=eff_window)
+            o = ob.permute(1, 0, 2, 3).contiguous()
         else:
             o, lse = ref.attention_fwd(q, k, v, causal, scale,
                                        window=eff_window)
@@ -164,6 +171,7 @@ class _FlashAttention(torch.autograd.Function):
         ctx.causal = causal
         ctx.scale = scale
         ctx.window = eff_window
+        ctx.sbhd = sbhd
         return o, lse
 
     @staticmethod
@@ -173,20 +181,32 @@ class _FlashAttention(torch.autograd.Function):
         if ctx.window is None and use_native(q) \
                 and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             dq, dk, dv = get_ext(False).flash_attn_bwd(
-                do, q, k, v, o, lse, ctx.causal, ctx.scale)
+                do, q, k, v, o, lse, ctx.causal, ctx.scale, None, ctx.sbhd)
+        elif ctx.sbhd:
+            dq, dk, dv = ref.attention_bwd(
+                do.permute(1, 0, 2, 3), q.permute(1, 0, 2, 3),
+                k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3),
+                o.permute(1, 0, 2, 3), lse, ctx.causal, ctx.scale,
+                window=ctx.window)
+            dq = dq.permute(1, 0, 2, 3).contiguous()
+            dk = dk.permute(1, 0, 2, 3).contiguous()
+            dv = dv.permute(1, 0, 2, 3).contiguous()
         else:
             dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal,
                                            ctx.scale, window=ctx.window)
-        return dq, dk, dv, None, None, None
+        return dq, dk, dv, None, None, None, None
 
 
 def flash_attention(q, k, v, causal: bool = True,
                     softmax_scale: Optional[float] = None,
-                    return_lse: bool = False, window: Optional[int] = None):
+                    return_lse: bool = False, window: Optional[int] = None,
+                    sbhd: bool = False):
     """q: [b,s,hq,d]; k,v: [b,s,hkv,d] -> o [b,s,hq,d] (+ lse [b,hq,s]).
+    sbhd=True: tensors are [s,b,h,d] (the runtime's native activation
+    layout) — saves the permute+contiguous copies around the kernel.
     window: mistral-style sliding window (causal only)."""
     o, lse = _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                                   causal, softmax_scale, window)
+                                   causal, softmax_scale, window, sbhd)
     return (o, lse) if return_lse else o
 
 

@@ -97,7 +97,14 @@ class SelfAttention(nn.Module):
         k = qkv[:, :, :, self.q_per_group].reshape(s, b, -1, self.head_dim)
         v = qkv[:, :, :, self.q_per_group + 1].reshape(s, b, -1, self.head_dim)
         q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin)
-        # [s,b,h,d] -> [b,s,h,d] for the flash kernels
+        if self.core_attention is None and attn_bias is None:
+            # plain local flash takes the native [s,b,h,d] layout directly
+            # (sbhd kernels) — no permute+contiguous copies either way
+            o = local_attention(q, k, v.contiguous(), causal=self.causal,
+                                softmax_scale=self.softmax_scale,
+                                window=self.window, sbhd=True)
+            return self.linear_proj(o.reshape(s, b, -1))
+        # [s,b,h,d] -> [b,s,h,d] for the distributed attention paths
         q = q.permute(1, 0, 2, 3).contiguous()
         k = k.permute(1, 0, 2, 3).contiguous()
         v = v.permute(1, 0, 2, 3).contiguous()

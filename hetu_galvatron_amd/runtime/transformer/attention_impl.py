@@ -26,11 +26,14 @@ from ..tensor_parallel.mappings import all_to_all, group_rank, group_size
 
 def local_attention(q, k, v, causal: bool = True,
                     softmax_scale: Optional[float] = None,
-                    window: Optional[int] = None) -> torch.Tensor:
-    """Plain local flash attention, autograd-capable. [b, s, h, d].
-    window: mistral sliding window (no-op when window >= kv length)."""
+                    window: Optional[int] = None,
+                    sbhd: bool = False) -> torch.Tensor:
+    """Plain local flash attention, autograd-capable. [b, s, h, d]
+    (or [s, b, h, d] with sbhd=True — the runtime's native layout, no
+    permute copies).  window: mistral sliding window (no-op when
+    window >= kv length)."""
     return flash_attention(q, k, v, causal=causal, softmax_scale=softmax_scale,
-                           window=window)
+                           window=window, sbhd=sbhd)
 
 
 # ---------------------------------------------------------------------------

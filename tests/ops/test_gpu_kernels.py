@@ -181,6 +181,33 @@ def test_flash_bwd(d, hq, hkv):
     assert_close(dv, dv_ref, 6e-2, rtol=3e-2, what="flash dv")
 
 
+def test_flash_sbhd_layout():
+    """sbhd=True ([s,b,h,d], the runtime's native layout) must match the
+    bshd path bit-for-bit on the same data, fwd + bwd."""
+    torch.manual_seed(17)
+    b, sq, hq, hkv, d = 3, 512, 8, 2, 128
+    q = torch.randn(sq, b, hq, d, device=dev()).bfloat16()
+    k = torch.randn(sq, b, hkv, d, device=dev()).bfloat16()
+    v = torch.randn(sq, b, hkv, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale, None, True)
+    qb = q.permute(1, 0, 2, 3).contiguous()
+    kb = k.permute(1, 0, 2, 3).contiguous()
+    vb = v.permute(1, 0, 2, 3).contiguous()
+    ob, lseb = ext().flash_attn_fwd(qb, kb, vb, True, scale)
+    assert torch.equal(o.permute(1, 0, 2, 3), ob), "sbhd fwd o mismatch"
+    assert torch.equal(lse, lseb), "sbhd lse mismatch"
+    do = torch.randn_like(o)
+    dq, dk, dv = ext().flash_attn_bwd(do, q, k, v, o, lse, True, scale,
+                                      None, True)
+    dqb, dkb, dvb = ext().flash_attn_bwd(
+        do.permute(1, 0, 2, 3).contiguous(), qb, kb, vb, ob, lseb, True,
+        scale)
+    assert torch.equal(dq.permute(1, 0, 2, 3), dqb), "sbhd dq"
+    assert torch.equal(dk.permute(1, 0, 2, 3), dkb), "sbhd dk"
+    assert torch.equal(dv.permute(1, 0, 2, 3), dvb), "sbhd dv"
+
+
 def test_flash_bench_shape_numerics():
     """Bench-shape insurance: seq-4096 GQA 32/8 fwd+bwd vs
     the fp32 reference — kernel regressions can't hide behind toy shapes.
