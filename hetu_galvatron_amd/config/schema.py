@@ -40,6 +40,10 @@ class ParallelArgs(BaseModel):
     galvatron_config_path: Optional[str] = None  # searched per-layer plan JSON
     gradient_reduce_in_fp32: bool = False
     overlap_grad_reduce: bool = True
+    # reduce-scatter grads every microbatch and accumulate in the shard
+    # domain (fp32): costs chunks x reduce traffic, saves the 4 B/param
+    # full-size fp32 accumulator — required for 70B-class zero3 fits
+    reduce_grads_each_microbatch: bool = False
 
 
 class ModelArgs(BaseModel):

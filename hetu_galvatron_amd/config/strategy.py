@@ -72,8 +72,11 @@ class LayerStrategy:
 
     @property
     def sdp(self) -> int:
-        """Group size for ZeRO sharding: dp*cp (sp/tp excluded)."""
-        return self.dp * self.cp
+        """Group size for ZeRO sharding: dp*cp, and for ulysses layers also
+        sp (params are replicated over sp, so the ZeRO domain merges it —
+        reference comm_groups.py:310; runtime core/comm_groups.py builds
+        the sdp group the same way)."""
+        return self.dp * self.cp * (self.sp if self.sp > 1 else 1)
 
     def degree_product(self) -> int:
         return self.pp_deg * self.tp_sp * self.cp * self.dp

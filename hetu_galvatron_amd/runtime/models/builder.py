@@ -67,7 +67,8 @@ class LayerBlock(nn.Module):
         self.prev_blk: Optional["LayerBlock"] = None
 
     def setup_zero(self, mode: str, param_dtype: torch.dtype, device,
-                   reduce_in_fp32: bool = False) -> None:
+                   reduce_in_fp32: bool = False,
+                   reduce_each_microbatch: bool = False) -> None:
         tp_group = None
         if not self.groups.strategy.use_ulysses:
             tp_group = self.groups.tp_group
@@ -78,11 +79,13 @@ class LayerBlock(nn.Module):
                 self.inner, mode, self.groups.edp_group, tp_group=None,
                 param_dtype=param_dtype, device=device,
                 param_filter=lambda p: getattr(p, "expert_parallel", False),
-                reduce_in_fp32=reduce_in_fp32)
+                reduce_in_fp32=reduce_in_fp32,
+                reduce_each_microbatch=reduce_each_microbatch)
         self.flat = FlatParamBlock(self.inner, mode, self.groups.sdp_group,
                                    tp_group=tp_group, param_dtype=param_dtype,
                                    device=device,
-                                   reduce_in_fp32=reduce_in_fp32)
+                                   reduce_in_fp32=reduce_in_fp32,
+                                   reduce_each_microbatch=reduce_each_microbatch)
 
     def _inner_forward(self, x, ctx):
         if self.kind == "embedding":
@@ -279,7 +282,17 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
         b.inner.to(device)
         mode = b.groups.strategy.dp_type
         b.setup_zero(mode, dtype, device,
-                     reduce_in_fp32=cfg.parallel.gradient_reduce_in_fp32)
+                     reduce_in_fp32=cfg.parallel.gradient_reduce_in_fp32,
+                     reduce_each_microbatch=
+                     cfg.parallel.reduce_grads_each_microbatch)
+    if emb_block is not None and head_block is not None:
+        # pp-tied embedding/lm-head: the tied grads all-reduce over the
+        # embedding group BEFORE the sdp reduction — incompatible with
+        # per-microbatch shard-domain accumulation, so those two blocks
+        # keep the full accumulator
+        for tb in (emb_block, head_block):
+            if tb.flat is not None:
+                tb.flat.reduce_each_microbatch = False
     for i, b in enumerate(blocks):  # zero3 prefetch neighbor links
         # plain-dict assignment: LayerBlock is an nn.Module and a normal
         # setattr would register the neighbor as a submodule (a cycle)

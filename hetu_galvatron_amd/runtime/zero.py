@@ -67,7 +67,8 @@ class FlatParamBlock:
                  device: Optional[torch.device] = None,
                  owner_filter: bool = True,
                  param_filter=None,
-                 reduce_in_fp32: bool = False):
+                 reduce_in_fp32: bool = False,
+                 reduce_each_microbatch: bool = False):
         assert mode in ("ddp", "zero2", "zero3")
         self.module = module
         self.mode = mode
@@ -85,6 +86,15 @@ class FlatParamBlock:
         self.auto_sync = False  # post-backward hook kicks async reduce
         self._synced = False
         self._gather_pending = None  # (handle, gloo chunks) of a prefetch
+        # per-microbatch reduction: grads accumulate in the SHARD domain
+        # (fp32 grad_accum_shard) instead of a full-size fp32 buffer —
+        # trades chunks x reduce-scatter traffic for 4 B/param of memory,
+        # which is what makes zero3 feasible for 70B-class models under
+        # 288 GB (the full accumulator alone would be 280 GB)
+        self.reduce_each_microbatch = (reduce_each_microbatch
+                                       and mode != "ddp")
+        self.grad_accum_shard: Optional[torch.Tensor] = None
+        self._mb_pending = None  # (handle, wire tensor) of in-flight mb RS
         self.reduce_in_fp32 = reduce_in_fp32
         self._post_sync = []  # deferred ops after the async handles drain
 
@@ -250,8 +260,68 @@ class FlatParamBlock:
         """Called by the block wrapper when the block's input grad is ready
         (all its param grads have accumulated)."""
         self.reshard_params()
-        if self.auto_sync and not self._synced:
+        if self.reduce_each_microbatch and self.ws > 1:
+            self._reduce_microbatch()
+        elif self.auto_sync and not self._synced:
             self.start_grad_sync()
+
+    # ---- per-microbatch shard-domain accumulation ----
+
+    def _tp_reduce_replicated(self, g: torch.Tensor) -> None:
+        """SP-replicated segments (norms): sum over the tp group before the
+        sdp reduction (see _start_grad_sync_impl for the ordering note)."""
+        if self.tp_group is None or self.tp_group.size <= 1:
+            return
+        handles = []
+        for seg in self.segments:
+            if seg.tp_replicated:
+                sl = g[seg.offset:seg.offset + seg.numel]
+                if _is_gloo(self.tp_group.group):
+                    dist.all_reduce(sl, group=self.tp_group.group)
+                else:
+                    handles.append(dist.all_reduce(
+                        sl, group=self.tp_group.group, async_op=True))
+        for h in handles:
+            h.wait()
+
+    def _finish_mb(self) -> None:
+        if self._mb_pending is None:
+            return
+        h, wire = self._mb_pending
+        self._mb_pending = None
+        if h is not None:
+            h.wait()
+        if self.grad_accum_shard is None:
+            self.grad_accum_shard = torch.zeros(
+                self.shard_size, dtype=torch.float32, device=self.device)
+        self.grad_accum_shard += wire.float()
+
+    def _reduce_microbatch(self) -> None:
+        """Reduce-scatter THIS microbatch's grads into the shard domain and
+        release the full fp32 accumulator; the async handle drains at the
+        start of the next microbatch's reduce (overlapping the next
+        forward/backward compute)."""
+        if self.flat_grad is None:
+            return
+        self._finish_mb()
+        g = self.flat_grad
+        self._tp_reduce_replicated(g)
+        compress = (not self.reduce_in_fp32
+                    and self.param_dtype == torch.bfloat16)
+        wire = g.to(torch.bfloat16) if compress else g
+        shard = torch.empty(self.shard_size,
+                            dtype=wire.dtype, device=self.device)
+        if _is_gloo(self.sdp_group.group):
+            dist.reduce_scatter_tensor(shard, wire,
+                                       group=self.sdp_group.group)
+            self._mb_pending = (None, shard)
+        else:
+            h = dist.reduce_scatter_tensor(shard, wire,
+                                           group=self.sdp_group.group,
+                                           async_op=True)
+            self._mb_pending = (h, shard)
+        # drop the full-size fp32 buffer (next microbatch reallocates)
+        self.flat_grad = None
 
     def start_grad_sync(self) -> None:
         """Async reduction of accumulated grads over the sdp group (+ tp
@@ -267,6 +337,16 @@ class FlatParamBlock:
 
     def _start_grad_sync_impl(self) -> None:
         self._synced = True
+        if self.reduce_each_microbatch and self.ws > 1:
+            # catch grads not yet reduced (e.g. a block whose sentinel
+            # didn't fire this microbatch), drain, expose the accumulation
+            self._reduce_microbatch()
+            self._finish_mb()
+            if self.grad_accum_shard is None:
+                self.grad_accum_shard = torch.zeros(
+                    self.shard_size, dtype=torch.float32, device=self.device)
+            self.grad_shard = self.grad_accum_shard
+            return
         self._ensure_grad_buffer()
         g = self.flat_grad
         # tp-replicated segments (SP norms): sum over the tp group first.
@@ -335,6 +415,9 @@ class FlatParamBlock:
     def zero_grad(self) -> None:
         if self.flat_grad is not None:
             self.flat_grad.zero_()
+        if self.grad_accum_shard is not None:
+            self.grad_accum_shard.zero_()
+        self._mb_pending = None
         self.grad_shard = None
         self._synced = False
 

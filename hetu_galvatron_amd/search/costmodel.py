@@ -83,21 +83,31 @@ def ddp_ratio(mixed_precision: bool) -> float:
     return 18 / 16 if mixed_precision else 1.0
 
 
-def zero2_ratio(d: int, mixed_precision: bool) -> float:
+def zero2_ratio(d: int, mixed_precision: bool,
+                grad_accum_in_shard: bool = False) -> float:
     """bf16 param full 2 + fp32 grad full 4 + reduced grad shard 4/d +
-    (master 4 + moments 8)/d."""
+    (master 4 + moments 8)/d.  grad_accum_in_shard
+    (parallel.reduce_grads_each_microbatch): the full fp32 accumulator is
+    replaced by per-microbatch reduce-scatters into the shard."""
     if not mixed_precision:
         return 3 / 4 * (1 / d + 0.003) + 1 / 4
     if d <= 1:
         return ddp_ratio(mixed_precision)
+    if grad_accum_in_shard:
+        return 2 / 16 + (1 / d + 0.003)
     return 6 / 16 + (1 / d + 0.003)
 
 
-def zero3_ratio(d: int, mixed_precision: bool) -> float:
+def zero3_ratio(d: int, mixed_precision: bool,
+                grad_accum_in_shard: bool = False) -> float:
     """bf16 shard 2/d (+ transient gathered block, amortized into the pad)
-    + fp32 grad full 4 + grad shard 4/d + masters/moments 12/d."""
+    + fp32 grad full 4 + grad shard 4/d + masters/moments 12/d; with
+    grad_accum_in_shard the full fp32 term drops (one in-flight block's
+    wire buffer amortized into the pad)."""
     if not mixed_precision:
         return 1 / d + 0.003
+    if grad_accum_in_shard:
+        return (18 / 16) * (1 / d + 0.003) + 0.02
     return 4 / 16 + (18 / 16) * (1 / d + 0.003)
 
 
@@ -123,7 +133,7 @@ def layer_time_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
     #   zero3 - bwd param re-gather + grad RS (2 units overlappable) +
     #           fwd param AG (1 unit exposed; forward has no overlap)
     sdp = s.sdp
-    unit_mb = (sdp - 1) / max(sdp, 1) * (prof.parameter_mb / s.tp_sp)
+    unit_mb = (sdp - 1) / max(sdp, 1) * (prof.parameter_mb / max(s.tp, 1))
     if mixed_precision:
         unit_mb /= 2
     if no_gradient_sync:
@@ -204,7 +214,8 @@ def layer_p2p_cost(s: LayerStrategy, prof: LayerProfile, hw: HardwareProfile,
 def layer_memory_cost(s: LayerStrategy, prof: LayerProfile,
                       global_bsz: int, chunks: int, stage_idx: int,
                       pipeline_type: str = "pipedream_flush",
-                      mixed_precision: bool = True) -> Dict[str, float]:
+                      mixed_precision: bool = True,
+                      grad_accum_in_shard: bool = False) -> Dict[str, float]:
     """One layer's memory (MB): model states + activation.
     Reference layer_cost.py:215-328."""
     lbsz = global_bsz / chunks / s.dp
@@ -216,12 +227,14 @@ def layer_memory_cost(s: LayerStrategy, prof: LayerProfile,
         cumulative = chunks
     cum_lbsz = cumulative * lbsz
 
-    param_mb = prof.parameter_mb / s.tp_sp
+    # params shard over megatron-tp only; ulysses keeps them whole and the
+    # ZeRO ratio (over sdp = dp*cp*sp) does the sharding
+    param_mb = prof.parameter_mb / max(s.tp, 1)
     states = 4 * param_mb
     if s.dp_type == "zero3":
-        states *= zero3_ratio(s.sdp, mixed_precision)
+        states *= zero3_ratio(s.sdp, mixed_precision, grad_accum_in_shard)
     elif s.dp_type == "zero2":
-        states *= zero2_ratio(s.sdp, mixed_precision)
+        states *= zero2_ratio(s.sdp, mixed_precision, grad_accum_in_shard)
     else:
         states *= ddp_ratio(mixed_precision)
 

@@ -340,9 +340,12 @@ class SearchEngine:
             intra_nosync.append(inos)
             mem_per_strategy.append({
                 stage: np.array([
-                    layer_memory_cost(s, tlp, global_bsz, chunks, stage,
-                                      self.cfg.parallel.pipeline_type,
-                                      self.mixed_precision)["total"]
+                    layer_memory_cost(
+                        s, tlp, global_bsz, chunks, stage,
+                        self.cfg.parallel.pipeline_type,
+                        self.mixed_precision,
+                        self.cfg.parallel.reduce_grads_each_microbatch,
+                    )["total"]
                     for s in strategies])
                 for stage in range(pp)})
             it = np.zeros((S, S))

@@ -55,8 +55,11 @@ def enumerate_strategies(world_size: int, args: SearchArgs,
                 sp = deg if mode == "sp" else 1
                 # the "0" leg of dp_types_enc executes as the runtime's
                 # default_dp_type — price what will actually run
-                dp_types = [default_dp_type if dp * cp > 1 else "ddp"]
-                if not args.disable_sdp and dp * cp > 1:
+                # the ZeRO domain for ulysses layers includes sp (params
+                # replicated over sp; sdp = dp*cp*sp)
+                sdp_deg = dp * cp * (sp if sp > 1 else 1)
+                dp_types = [default_dp_type if sdp_deg > 1 else "ddp"]
+                if not args.disable_sdp and sdp_deg > 1:
                     dp_types.append("zero3")
                 for dpt in dp_types:
                     ckpts = [False] if args.disable_ckpt else [False, True]

@@ -183,6 +183,18 @@ def test_pp2_gpipe_uneven_chunks():
     run_case(2, plan)
 
 
+@pytest.mark.distributed
+@pytest.mark.parametrize("dp_type", ["zero2", "zero3"])
+def test_dp2_reduce_each_microbatch(dp_type):
+    """Per-microbatch shard-domain grad accumulation (the 70B memory mode:
+    no full fp32 accumulator) must train identically to the default
+    accumulate-then-reduce path."""
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, dp_type=dp_type,
+                                      global_bsz=4, chunks=2)
+    run_case(2, plan,
+             {"parallel": {"reduce_grads_each_microbatch": True}})
+
+
 def test_uneven_chunks_no_pipeline():
     """1-process path: global batch 4, chunks=3 -> [2,1,1] must reproduce
     the chunks=1 loss exactly (loss normalized by global tokens)."""
