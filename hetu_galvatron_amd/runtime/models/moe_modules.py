@@ -98,6 +98,12 @@ class GalvatronMoEMLP(nn.Module):
             from .. import moe  # noqa: F401  (package anchor)
             from ..moe import tracker
             tracker.save_aux_loss("moe_aux", self.layer_idx, aux)
+        # shared expert FIRST: its GEMMs enqueue on the compute stream
+        # before the dispatch all-to-all's stream dependency, so on RCCL
+        # they overlap the a2a (the reference overlaps the same way via
+        # SharedExpertMLP + the flex dispatcher, moe/mlp.py:215)
+        shared_out = (self.shared(x_in.reshape(-1, x_in.shape[-1]))
+                      if self.shared is not None else None)
         expert_in, tokens_per_expert = self.dispatcher.dispatch(
             flat, probs, idx)
         expert_out = self.experts(expert_in, tokens_per_expert)
@@ -109,8 +115,8 @@ class GalvatronMoEMLP(nn.Module):
             # partial ffn outputs: the reduce-scatter both sums the etp
             # partials and restores the sequence shard
             out = reduce_scatter_to_sequence_parallel_region(out, self.tp_group)
-        if self.shared is not None:
-            out = out + self.shared(x_in.reshape(-1, h)).reshape(out.shape)
+        if shared_out is not None:
+            out = out + shared_out.reshape(out.shape)
         return out
 
 
