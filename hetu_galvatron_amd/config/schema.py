@@ -36,7 +36,11 @@ class ParallelArgs(BaseModel):
     vocab_tp: int = 1
     vocab_sp: int = 0
     vocab_cp: int = 1
-    mixed_precision: str = "bf16"  # fp32 | bf16
+    mixed_precision: str = "bf16"  # fp32 | bf16 | fp16
+    # fp16 dynamic loss scaling (reference: Megatron-style scaler)
+    loss_scale_init: float = 65536.0
+    loss_scale_window: int = 1000
+    min_loss_scale: float = 1.0
     galvatron_config_path: Optional[str] = None  # searched per-layer plan JSON
     gradient_reduce_in_fp32: bool = False
     overlap_grad_reduce: bool = True
@@ -259,7 +263,7 @@ class GalvatronConfig(BaseModel):
             raise ValueError(f"bad default_dp_type {self.parallel.default_dp_type}")
         if self.parallel.pipeline_type not in ("gpipe", "pipedream_flush"):
             raise ValueError(f"bad pipeline_type {self.parallel.pipeline_type}")
-        if self.parallel.mixed_precision not in ("fp32", "bf16"):
+        if self.parallel.mixed_precision not in ("fp32", "bf16", "fp16"):
             raise ValueError(f"bad mixed_precision {self.parallel.mixed_precision}")
         if self.model.hidden_size % self.model.num_attention_heads != 0:
             raise ValueError("hidden_size must be divisible by num_attention_heads")

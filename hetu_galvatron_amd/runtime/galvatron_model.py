@@ -62,8 +62,9 @@ class GalvatronModel(nn.Module):
         self.cfg = cfg
         self.plan = plan or resolve_plan(cfg, world)
         self.stage_model = build_hybrid_parallel_model(cfg, self.plan, device)
-        act_dtype = torch.bfloat16 if cfg.parallel.mixed_precision == "bf16" \
-            else torch.float32
+        act_dtype = {"bf16": torch.bfloat16,
+                     "fp16": torch.float16}.get(
+            cfg.parallel.mixed_precision, torch.float32)
         self.engine = PipelineEngine(
             self.stage_model, cfg.model.hidden_size,
             pipeline_type=self.plan.pipeline_type,
@@ -76,6 +77,9 @@ class GalvatronModel(nn.Module):
         return self.stage_model.blocks
 
     def forward_backward(self, ctx: Dict, chunks: Optional[int] = None) -> StepStats:
+        scaler = getattr(self.stage_model, "loss_scaler", None)
+        if scaler is not None:
+            self.engine.loss_scale = scaler.scale
         return self.engine.forward_backward(ctx, chunks or self.chunks)
 
     def global_loss(self, stats: StepStats, device=None) -> float:

@@ -198,7 +198,8 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
     pp = plan.pp_deg
     division = plan.pp_division or even_pp_division(n_layers, pp)
     my_stage = pp_stage_of_rank(rank, world, pp)
-    dtype = torch.bfloat16 if cfg.parallel.mixed_precision == "bf16" else torch.float32
+    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16}.get(
+        cfg.parallel.mixed_precision, torch.float32)
 
     strategies = [plan.layer(i, world) for i in range(n_layers)]
     vocab_strat = plan.vocab_strategy(world)

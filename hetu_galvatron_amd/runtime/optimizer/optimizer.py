@@ -23,12 +23,47 @@ from ...ops._ext import get_ext, native_available
 from ..zero import FlatParamBlock
 
 
+class LossScaler:
+    """Megatron-style dynamic loss scaling for fp16 (reference:
+    optimizer mixed-precision wrapper).  bf16 needs none (same exponent
+    range as fp32), which is why bf16 is this framework's default."""
+
+    def __init__(self, init_scale: float = 65536.0, window: int = 1000,
+                 min_scale: float = 1.0, growth_factor: float = 2.0,
+                 backoff_factor: float = 0.5):
+        self.scale = float(init_scale)
+        self.window = window
+        self.min_scale = min_scale
+        self.growth = growth_factor
+        self.backoff = backoff_factor
+        self._good_steps = 0
+
+    def update(self, found_inf: bool) -> None:
+        if found_inf:
+            self.scale = max(self.scale * self.backoff, self.min_scale)
+            self._good_steps = 0
+        else:
+            self._good_steps += 1
+            if self._good_steps >= self.window:
+                self.scale *= self.growth
+                self._good_steps = 0
+
+    def state_dict(self) -> dict:
+        return {"scale": self.scale, "good_steps": self._good_steps}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.scale = sd["scale"]
+        self._good_steps = sd["good_steps"]
+
+
 class GalvatronOptimizer:
     def __init__(self, blocks: List[FlatParamBlock], lr: float = 1e-4,
                  betas=(0.9, 0.95), eps: float = 1e-8,
                  weight_decay: float = 0.01, clip_grad: float = 1.0,
-                 use_fused: bool = True):
+                 use_fused: bool = True,
+                 loss_scaler: Optional[LossScaler] = None):
         self.blocks = [b for b in blocks if b is not None and b.total > 0]
+        self.loss_scaler = loss_scaler
         self.lr = lr
         self.betas = betas
         self.eps = eps
@@ -86,16 +121,26 @@ class GalvatronOptimizer:
     def step(self) -> float:
         use_native = (self.use_fused and native_available()
                       and self.blocks and self.blocks[0].device.type == "cuda")
-        norm = self._global_grad_norm()
+        inv_ls = 1.0 / self.loss_scaler.scale if self.loss_scaler else 1.0
+        norm = self._global_grad_norm() * inv_ls  # grads carry the loss scale
         self.last_grad_norm = norm
-        scale = 1.0
+        if self.loss_scaler is not None:
+            import math as _math
+            found_inf = not _math.isfinite(norm)
+            self.loss_scaler.update(found_inf)
+            if found_inf:
+                # global decision (the norm was all-reduced): skip the step,
+                # back off the scale (reference fp16 semantics)
+                self.last_grad_norm = float("nan")
+                return float("nan")
+        scale = inv_ls
         if self.clip_grad and self.clip_grad > 0 and norm > self.clip_grad:
-            scale = self.clip_grad / (norm + 1e-6)
-            if not use_native:
-                # fallback path applies the clip as a separate pass; the
-                # native path folds it into the adam kernel (gscale)
-                for b in self.blocks:
-                    b.scale_grads(scale)
+            scale = inv_ls * self.clip_grad / (norm + 1e-6)
+        if scale != 1.0 and not use_native:
+            # fallback path applies unscale+clip as a separate pass; the
+            # native path folds it into the adam kernel (gscale)
+            for b in self.blocks:
+                b.scale_grads(scale)
         self.step_count += 1
         masters, grads, ms, vs, outs = [], [], [], [], []
         for b in self.blocks:
@@ -106,7 +151,7 @@ class GalvatronOptimizer:
             get_ext(False).fused_adamw(
                 masters, grads, ms, vs, outs, self.step_count, self.lr,
                 self.betas[0], self.betas[1], self.eps, self.weight_decay,
-                scale)
+                scale)  # scale = 1/loss_scale * clip factor
         else:
             ref.adamw_step(outs, grads, ms, vs, masters, self.step_count,
                            self.lr, self.betas[0], self.betas[1], self.eps,
@@ -152,10 +197,16 @@ def get_optimizer_and_param_scheduler(stage_model, cfg):
         if blk.flat is not None:
             blocks.append(blk.flat)
     t = cfg.train
+    scaler = None
+    if cfg.parallel.mixed_precision == "fp16":
+        scaler = LossScaler(cfg.parallel.loss_scale_init,
+                            cfg.parallel.loss_scale_window,
+                            cfg.parallel.min_loss_scale)
+        stage_model.loss_scaler = scaler
     opt = GalvatronOptimizer(
         blocks, lr=t.lr, betas=(t.adam_beta1, t.adam_beta2), eps=t.adam_eps,
         weight_decay=t.adam_weight_decay, clip_grad=t.clip_grad,
-        use_fused=t.use_fused_adam)
+        use_fused=t.use_fused_adam, loss_scaler=scaler)
     sched = OptimizerParamScheduler(
         opt, max_lr=t.lr, min_lr=t.min_lr, warmup_steps=t.lr_warmup_iters,
         decay_steps=t.lr_decay_iters or t.train_iters,

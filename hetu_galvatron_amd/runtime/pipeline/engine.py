@@ -78,6 +78,7 @@ class PipelineEngine:
         self.overlap = overlap_grad_reduce
         self.act_dtype = act_dtype
         self.profiler = None  # RuntimeProfiler: fwd_start/fwd_end brackets
+        self.loss_scale = 1.0  # fp16 dynamic loss scaling (LossScaler)
         self._after_fwd_snapped = False
         self._pending_sends = []  # [(reqs, tensor)] in-flight async p2p
         self.device = next(stage_model.parameters()).device \
@@ -113,6 +114,8 @@ class PipelineEngine:
     def _loss_of(self, per_token: torch.Tensor, ctx: Dict, chunks: int):
         denom = float(ctx["seq_len"]) * ctx["global_batch"]
         loss = per_token.float().sum() / denom
+        if self.loss_scale != 1.0:
+            loss = loss * self.loss_scale
         return loss
 
     def _stat_update(self, stats: StepStats, per_token: torch.Tensor) -> None:
