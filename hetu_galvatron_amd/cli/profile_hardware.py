@@ -16,12 +16,17 @@ def main(argv=None):
     scripts_only = "--scripts-only" in args
     if scripts_only:
         args.remove("--scripts-only")
+    algo_sweep = "--nccl-algo-sweep" in args
+    if algo_sweep:
+        args.remove("--nccl-algo-sweep")
     cfg = config_from_cli(args)
     hp = HardwareProfiler(cfg)
     paths = hp.generate_scripts()
     print(f"[profile_hardware] wrote {paths}")
     if not scripts_only:
         hp.profile_bandwidth()
+        if algo_sweep:
+            hp.profile_nccl_algo_sweep()
 
 
 if __name__ == "__main__":

@@ -59,3 +59,21 @@ class HardwareProfiler:
             if env:
                 e.update(env)
             subprocess.run(self._launch_cmd(op), check=True, env=e)
+
+    def profile_nccl_algo_sweep(self, op: str = "allreduce",
+                                algos: Optional[List[str]] = None) -> None:
+        """Re-run a sweep once per RCCL algorithm (NCCL_ALGO) — xGMI is
+        point-to-point (7 links/GPU), so ring vs tree vs direct behave
+        very differently from NVSwitch and the cost model should absorb
+        whichever the profiler finds fastest (SURVEY §5 comm-backend
+        notes).  Each pass writes into its own subdirectory
+        `<out_dir>/algo_<name>/`; compare the JSONs and copy the winner
+        up into hardware_configs/."""
+        base = self.out_dir
+        for algo in algos or ("Ring", "Tree"):
+            self.out_dir = os.path.join(base, f"algo_{algo.lower()}")
+            os.makedirs(self.out_dir, exist_ok=True)
+            try:
+                self.profile_bandwidth([op], env={"NCCL_ALGO": algo})
+            finally:
+                self.out_dir = base
