@@ -195,6 +195,17 @@ def test_dp2_reduce_each_microbatch(dp_type):
              {"parallel": {"reduce_grads_each_microbatch": True}})
 
 
+@pytest.mark.distributed
+def test_world4_zero3_reduce_mb():
+    """Compound: zero3 (+ both-direction prefetch) x per-microbatch shard
+    accumulation on 4 ranks (chunks clamp to the 1 row/rank; the
+    multi-microbatch accumulation case is test_dp2_reduce_each_microbatch)."""
+    plan = HybridParallelPlan.uniform(N_LAYERS, 4, dp_type="zero3",
+                                      global_bsz=4, chunks=3)
+    run_case(4, plan,
+             {"parallel": {"reduce_grads_each_microbatch": True}})
+
+
 def test_uneven_chunks_no_pipeline():
     """1-process path: global batch 4, chunks=3 -> [2,1,1] must reproduce
     the chunks=1 loss exactly (loss normalized by global tokens)."""
