@@ -66,6 +66,27 @@ def test_searched_plan_executes_world2(tmp_path):
 
 
 @pytest.mark.distributed
+def test_searched_pp_plan_executes_world2(tmp_path):
+    """With DP comm priced prohibitively, the search must fall to pp=2 —
+    and the 1F1B plan it saves trains to the baseline."""
+    out = str(tmp_path / "plan_pp.json")
+    lp, op, hw = _tiny_profiles()
+    for k in list(hw.allreduce_latency_per_mb):
+        hw.allreduce_latency_per_mb[k] = 50.0  # make any dp>1 terrible
+    orig = globals()["_tiny_profiles"]
+    globals()["_tiny_profiles"] = lambda: (lp, op, hw)
+    try:
+        best = _search_plan(2, 64, out, settle_chunks=2, disable_tp=1,
+                            disable_sp=1)
+    finally:
+        globals()["_tiny_profiles"] = orig
+    plan = HybridParallelPlan.load(out)
+    plan.validate(2)
+    assert plan.pp_deg == 2, plan.to_config_dict()
+    run_case(2, plan)
+
+
+@pytest.mark.distributed
 def test_searched_tight_memory_plan_executes_world2(tmp_path):
     """A tight budget must force ckpt and/or zero3 into the plan — and the
     heterogeneous result still matches the baseline."""
