@@ -51,9 +51,9 @@ typedef float f32_t;
 DECL_MOE(bf16_t)
 DECL_MOE(f32_t)
 
-void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, bool sbhd = false);
+void flash_fwd_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, bool sbhd = false, int window = 0);
 void attn_di_launch(const __bf16*, const __bf16*, float*, int, int, int, int, hipStream_t, bool sbhd = false);
-void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, float* dbias = nullptr, bool sbhd = false);
+void flash_bwd_launch(const __bf16*, const __bf16*, const __bf16*, const __bf16*, const float*, const float*, __bf16*, __bf16*, __bf16*, int, int, int, int, int, int, float, bool, hipStream_t, const __bf16* bias = nullptr, float* dbias = nullptr, bool sbhd = false, int window = 0);
 void multi_sumsq_launch(const long*, int, float*, hipStream_t);
 void mfma_probe_launch(const __bf16*, const __bf16*, float*, bool, hipStream_t);
 void decode_attn_launch(const __bf16*, const __bf16*, const __bf16*, __bf16*, float*, int, int, int, int, int, int, int, float, const int*, hipStream_t);
@@ -277,7 +277,9 @@ Tensor rope_fwd(const Tensor& x, const Tensor& cos_t, const Tensor& sin_t,
 std::tuple<Tensor, Tensor> flash_attn_fwd(
     const Tensor& q, const Tensor& k, const Tensor& v, bool causal,
     double scale, const c10::optional<Tensor>& bias = c10::nullopt,
-    bool sbhd = false) {
+    bool sbhd = false, int64_t window = 0) {
+  TORCH_CHECK(window == 0 || causal,
+              "flash_attn: sliding window requires causal");
   CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v);
   TORCH_CHECK(is_bf16(q), "flash_attn: bf16 only on the native path");
   TORCH_CHECK(q.dim() == 4, "flash_attn: q must be [b,s,h,d] (or [s,b,h,d] with sbhd)");
@@ -302,14 +304,18 @@ std::tuple<Tensor, Tensor> flash_attn_fwd(
   auto lse = at::empty({b, hq, sq}, q.options().dtype(at::kFloat));
   flash_fwd_launch(bfp(q), bfp(k), bfp(v), bfp_mut(o),
                    lse.data_ptr<float>(), b, sq, skv, hq, hkv, d,
-                   (float)scale, causal, cur_stream(), bp, sbhd);
+                   (float)scale, causal, cur_stream(), bp, sbhd,
+                   (int)window);
   return {o, lse};
 }
 
 std::vector<Tensor> flash_attn_bwd(
     const Tensor& dout, const Tensor& q, const Tensor& k, const Tensor& v,
     const Tensor& o, const Tensor& lse, bool causal, double scale,
-    const c10::optional<Tensor>& bias = c10::nullopt, bool sbhd = false) {
+    const c10::optional<Tensor>& bias = c10::nullopt, bool sbhd = false,
+    int64_t window = 0) {
+  TORCH_CHECK(window == 0 || causal,
+              "flash_attn_bwd: sliding window requires causal");
   CHECK_GPU(dout); CHECK_GPU(q); CHECK_GPU(k); CHECK_GPU(v); CHECK_GPU(o);
   const int b = sbhd ? q.size(1) : q.size(0);
   const int sq = sbhd ? q.size(0) : q.size(1);
@@ -342,7 +348,7 @@ std::vector<Tensor> flash_attn_bwd(
                    lsec.data_ptr<float>(), di.data_ptr<float>(),
                    bfp_mut(dq), bfp_mut(dk_exp), bfp_mut(dv_exp), b, sq, skv,
                    hq, hkv, d, (float)scale, causal, cur_stream(), bp, dbp,
-                   sbhd);
+                   sbhd, (int)window);
   Tensor dk = dk_exp, dv = dv_exp;
   if (hq != hkv) {
     const int rep = hq / hkv;
@@ -687,8 +693,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rope_fwd", &rope_fwd);
-  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false);
-  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false);
+  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false, py::arg("window") = 0);
+  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false, py::arg("window") = 0);
   m.def("decode_attn", &decode_attn);
   m.def("decode_attn_graph", &decode_attn_graph);
   m.def("mfma_probe", &mfma_probe);

@@ -61,7 +61,7 @@ DEV_INLINE void load_bias16(const __bf16* brow, int base, int limit,
 // ---------------------------------------------------------------------------
 // forward (v2)
 // ---------------------------------------------------------------------------
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __device__ __attribute__((noinline))
 void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
                      const __bf16* __restrict__ k,
@@ -71,7 +71,7 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
                      long lse_base, int q_stride, int kv_stride, int off,
                      int sq, int skv, float sl2e, bool causal,
                      const __bf16* __restrict__ bias = nullptr,
-                     long bias_base = 0) {
+                     long bias_base = 0, int window = 0) {
   constexpr int KB = 64;            // kv tile
   constexpr int QBF = 256;          // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;       // padded K row (bf16 elems)
@@ -107,6 +107,14 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
 
   int kv_end = causal ? min(skv, qblk * QBF + QBF + off) : skv;
   const int kv_last_w = causal ? min(skv, q0w + 32 + off) : skv;  // wave's own
+  // sliding window (mistral): keys visible to q are [q+off-window+1, q+off];
+  // whole-workgroup start tile + per-wave lower skip + per-element mask
+  int kv_begin = 0;
+  int kv_first_w = 0;
+  if (WINDOWED) {
+    kv_begin = max(0, (qblk * QBF + off - window + 1) / KB * KB);
+    kv_first_w = max(0, q0w + off - window + 1);
+  }
 
   // staged registers for the next tile
   bf16x8 kst[KPT];
@@ -175,18 +183,18 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
     }
   };
 
-  stage_load(0);
+  stage_load(kv_begin);
   stage_write(0);
   __syncthreads();
   int cur = 0;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KB) {
     const bool have_next = kv0 + KB < kv_end;
     const __bf16* k_lds = smem + cur * BUFSZ;
     const __bf16* vt_lds = k_lds + KB * KROW;
     if (have_next) stage_load(kv0 + KB);  // async: lands at stage_write
 
-    if (kv0 < kv_last_w) {  // per-wave causal skip (wave-uniform)
+    if (kv0 < kv_last_w && kv0 + KB > kv_first_w) {  // per-wave skips
       // ---- S^T = K Q^T (two 32-key tiles) ----
       f32x16 st0 = (f32x16)(0.f), st1 = (f32x16)(0.f);
 #pragma unroll
@@ -206,7 +214,8 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
       if (BIASED)
         brow = bias + bias_base + (long)min(qg, sq - 1) * skv;
       const bool need_mask =
-          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
+          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv) ||
+          (WINDOWED && kv0 < q0w + 32 + off - window + 1);
       constexpr float LOG2E = 1.4426950408889634f;
       float bv[32];
       if (BIASED) {
@@ -222,8 +231,10 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
             x0 += LOG2E * bv[r];
             x1 += LOG2E * bv[16 + r];
           }
-          if (key0 >= skv || (causal && key0 > qg + off)) x0 = neg_big();
-          if (key0 + 32 >= skv || (causal && key0 + 32 > qg + off))
+          if (key0 >= skv || (causal && key0 > qg + off) ||
+              (WINDOWED && key0 <= qg + off - window)) x0 = neg_big();
+          if (key0 + 32 >= skv || (causal && key0 + 32 > qg + off) ||
+              (WINDOWED && key0 + 32 <= qg + off - window))
             x1 = neg_big();
           pv[r] = x0;
           pv[16 + r] = x1;
@@ -320,7 +331,7 @@ void flash_fwd_block(int qblk, const __bf16* __restrict__ q,
 // thin kernel: common indexing + complementary-pair causal load balance
 // (block x runs q blocks {x, nqb-1-x}: constant total KV tiles per block,
 // so the makespan matches the average instead of 2x the deepest block)
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_fwd_kernel(const __bf16* __restrict__ q,
                       const __bf16* __restrict__ k,
@@ -328,7 +339,8 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
                       __bf16* __restrict__ o, float* __restrict__ lse,
                       int b, int sq, int skv, int hq, int hkv, float scale,
                       bool causal, const __bf16* __restrict__ bias = nullptr,
-                      bool paired = true, bool sbhd = false) {
+                      bool paired = true, bool sbhd = false,
+                      int window = 0) {
   constexpr int KB = 64, QBF = 256;
   constexpr int BUFSZ = KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -349,15 +361,15 @@ void flash_fwd_kernel(const __bf16* __restrict__ q,
   const float sl2e = scale * 1.4426950408889634f;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
-  flash_fwd_block<D, BIASED>(blockIdx.x, q, k, v, o, lse, smem, q_base,
+  flash_fwd_block<D, BIASED, WINDOWED>(blockIdx.x, q, k, v, o, lse, smem, q_base,
                              kv_base, lse_base, q_str, kv_str, off, sq,
-                             skv, sl2e, causal, bias, bias_base);
+                             skv, sl2e, causal, bias, bias_base, window);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_fwd_block<D, BIASED>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
+    flash_fwd_block<D, BIASED, WINDOWED>(qb2, q, k, v, o, lse, smem, q_base, kv_base,
                                lse_base, q_str, kv_str, off, sq, skv,
-                               sl2e, causal, bias, bias_base);
+                               sl2e, causal, bias, bias_base, window);
   }
 }
 
@@ -410,7 +422,7 @@ __global__ void attn_di_kernel(const __bf16* __restrict__ dout,
 // the v1 form spilled 34 dwords/lane to scratch).
 // dQ^T[d][q] += K^T dS accumulated in D-layout, epilogue like the forward.
 // ---------------------------------------------------------------------------
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __device__ __attribute__((noinline))
 void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
                         const __bf16* __restrict__ q,
@@ -423,7 +435,7 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
                         int q_stride, int kv_stride, int off, int sq,
                         int skv, float scale, bool causal,
                         const __bf16* __restrict__ bias = nullptr,
-                        long bias_base = 0) {
+                        long bias_base = 0, int window = 0) {
   constexpr int KB = 32;         // kv tile
   constexpr int QBF = 256;       // q rows per workgroup (8 waves x 32)
   constexpr int KROW = D + 8;
@@ -466,6 +478,11 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
 
   const int kv_end = causal ? min(skv, qblk * QBF + QBF + off) : skv;
   const int kv_last_w = causal ? min(skv, q0w + 32 + off) : skv;
+  int kv_begin = 0, kv_first_w = 0;
+  if (WINDOWED) {
+    kv_begin = max(0, (qblk * QBF + off - window + 1) / KB * KB);
+    kv_first_w = max(0, q0w + off - window + 1);
+  }
 
   bf16x8 kst[KPT], vst_r[KPT];
   ushort8 ktst[KPT * 2];
@@ -534,21 +551,21 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
     }
   };
 
-  stage_load(0);
+  stage_load(kv_begin);
   stage_write(0);
   __syncthreads();
   int cur = 0;
 
   const float sl2e = scale;  // bwd stays in natural-log units (lse is ln)
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KB) {
     const bool have_next = kv0 + KB < kv_end;
     const __bf16* k_lds = smem + cur * BUFSZ;
     const __bf16* v_lds = k_lds + KB * KROW;
     const __bf16* kt_lds = v_lds + KB * KROW;
     if (have_next) stage_load(kv0 + KB);
 
-    if (kv0 < kv_last_w) {
+    if (kv0 < kv_last_w && kv0 + KB > kv_first_w) {
       // S^T = K Q^T ; dP^T = V dO^T
       f32x16 st = (f32x16)(0.f), dp = (f32x16)(0.f);
 #pragma unroll
@@ -567,7 +584,8 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
       if (BIASED)
         brow = bias + bias_base + (long)min(qg, sq - 1) * skv;
       const bool need_mask =
-          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv);
+          (causal && kv0 + KB > q0w + off + 1) || (kv0 + KB > skv) ||
+          (WINDOWED && kv0 < q0w + 32 + off - window + 1);
       float dsv[16];
       float bv[16];
       if (BIASED)
@@ -580,7 +598,8 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
         float pr = e;
         if (need_mask) {
           const int key = kv0 + mfma32_d_row(lane, r);
-          const bool ok = key < skv && !(causal && key > qg + off);
+          const bool ok = key < skv && !(causal && key > qg + off) &&
+                          !(WINDOWED && key <= qg + off - window);
           pr = ok ? e : 0.f;
         }
         dsv[r] = pr * (dp[r] - di_c) * scale;
@@ -633,7 +652,7 @@ void flash_bwd_dq_block(int qblk, const __bf16* __restrict__ dout,
   }
 }
 
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
@@ -645,7 +664,8 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
                          int b, int sq, int skv, int hq, int hkv,
                          float scale, bool causal,
                          const __bf16* __restrict__ bias = nullptr,
-                         bool paired = true, bool sbhd = false) {
+                         bool paired = true, bool sbhd = false,
+                         int window = 0) {
   constexpr int KB = 32, QBF = 256;
   constexpr int BUFSZ = 2 * KB * (D + 8) + D * (KB + 8);
   __shared__ __align__(16) __bf16 smem[2 * BUFSZ];
@@ -663,17 +683,17 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
   const long bias_base = (long)h * sq * skv;
   const int off = skv - sq;
   const int nqb = (sq + QBF - 1) / QBF;
-  flash_bwd_dq_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dq,
+  flash_bwd_dq_block<D, BIASED, WINDOWED>(blockIdx.x, dout, q, k, v, lse, di, dq,
                                 smem, q_base, kv_base, lse_base, q_str,
                                 kv_str, off, sq, skv, scale, causal, bias,
-                                bias_base);
+                                bias_base, window);
   const int qb2 = nqb - 1 - (int)blockIdx.x;
   if (causal && paired && qb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_bwd_dq_block<D, BIASED>(qb2, dout, q, k, v, lse, di, dq, smem,
+    flash_bwd_dq_block<D, BIASED, WINDOWED>(qb2, dout, q, k, v, lse, di, dq, smem,
                                   q_base, kv_base, lse_base, q_str,
                                   kv_str, off, sq, skv, scale, causal,
-                                  bias, bias_base);
+                                  bias, bias_base, window);
   }
 }
 
@@ -695,7 +715,7 @@ void flash_bwd_dq_kernel(const __bf16* __restrict__ dout,
 //   phase dK:  S again; dP = mfma(dO_frag[from do_lds rows], V^T[vfr regs])
 //              dK^T[d][key] = mfma(Q^T_frag[from t_lds], dS_frag[permlane])
 // ---------------------------------------------------------------------------
-template <int D, bool DKPH, bool BIASED = false>
+template <int D, bool DKPH, bool BIASED = false, bool WINDOWED = false>
 __device__ __attribute__((noinline))
 void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
                          const __bf16* __restrict__ q,
@@ -711,7 +731,7 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
                          int skv, float scale, bool causal,
                          const __bf16* __restrict__ bias = nullptr,
                          float* __restrict__ dbias = nullptr,
-                         long bias_base = 0) {
+                         long bias_base = 0, int window = 0) {
   constexpr int QT = 64;           // q tile (2 mfma halves per stage)
   constexpr int KBW = 256;         // keys per workgroup (8 waves x 32)
   constexpr int QROW = D + 8;
@@ -754,9 +774,14 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
   for (int dt = 0; dt < ND; ++dt) acc[dt] = (f32x16)(0.f);
 
   // first q tile that can attend any key of this block (per-wave causal
-  // skipping happens per 32-q half inside the loop)
+  // skipping happens per 32-q half inside the loop); with a sliding
+  // window, also the LAST q that can see any of this block's keys:
+  // q <= key - off + window - 1
   int qstart = 0;
   if (causal) qstart = max(0, ((kvblk * KBW - off) / QT) * QT);
+  int qstop = sq;
+  if (WINDOWED)
+    qstop = min(sq, kvblk * KBW + KBW - 1 - off + window);
 
   // staged registers: q rows, transposed image of (dV: dO / dK: Q),
   // dO rows (dK phase only)
@@ -845,8 +870,8 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
   __syncthreads();
   int cur = 0;
 
-  for (int qt0 = qstart; qt0 < sq; qt0 += QT) {
-    const bool have_next = qt0 + QT < sq;
+  for (int qt0 = qstart; qt0 < qstop; qt0 += QT) {
+    const bool have_next = qt0 + QT < qstop;
     const __bf16* q_lds = smem + cur * BUFSZ;
     const __bf16* t_lds = q_lds + QT * QROW;
     const __bf16* do_lds = t_lds + D * TROW;
@@ -856,8 +881,9 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
 #pragma unroll
     for (int qh = 0; qh < QT / 32; ++qh) {
       const int q0h = qt0 + qh * 32;
-      // per-half causal skip (wave-uniform): any key of this wave live?
+      // per-half skips (wave-uniform): any key of this wave live?
       if (causal && q0h + 31 + off < k0w) continue;
+      if (WINDOWED && q0h + off - window + 1 > k0w + 31) continue;
       if (q0h >= sq) continue;
 
       // S = Q K^T (; dP = dO V^T)  D-layout rows=q(crow), cols=key(lane);
@@ -888,7 +914,13 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
         }
         const float e = __expf(sc - lse_r);
         float p = e;
-        if (causal) {
+        if (WINDOWED) {
+          const int keyg = k0w + col;
+          const int qo = q0h + qrow + off;
+          const bool ok = keyg < skv && (!causal || keyg <= qo) &&
+                          keyg > qo - window;
+          p = ok ? e : 0.f;
+        } else if (causal) {
           const int keyg = k0w + col;
           const bool ok = keyg < skv && keyg <= q0h + qrow + off;
           p = ok ? e : 0.f;
@@ -960,7 +992,7 @@ void flash_bwd_dkv_phase(int kvblk, const __bf16* __restrict__ dout,
   }
 }
 
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __device__ void flash_bwd_dkv_block(
     int kvblk, const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
@@ -970,19 +1002,19 @@ __device__ void flash_bwd_dkv_block(
     long dkv_base, long lse_base, int q_stride, int kv_stride,
     int dkv_stride, int off, int sq, int skv, float scale, bool causal,
     const __bf16* bias = nullptr, float* dbias = nullptr,
-    long bias_base = 0) {
-  flash_bwd_dkv_phase<D, false, BIASED>(
+    long bias_base = 0, int window = 0) {
+  flash_bwd_dkv_phase<D, false, BIASED, WINDOWED>(
       kvblk, dout, q, k, v, lse, di, dk_exp, dv_exp, smem_base, lsedi_base,
       q_base, kv_base, dkv_base, lse_base, q_stride, kv_stride, dkv_stride,
-      off, sq, skv, scale, causal, bias, dbias, bias_base);
+      off, sq, skv, scale, causal, bias, dbias, bias_base, window);
   __syncthreads();
-  flash_bwd_dkv_phase<D, true, BIASED>(
+  flash_bwd_dkv_phase<D, true, BIASED, WINDOWED>(
       kvblk, dout, q, k, v, lse, di, dk_exp, dv_exp, smem_base, lsedi_base,
       q_base, kv_base, dkv_base, lse_base, q_stride, kv_stride, dkv_stride,
-      off, sq, skv, scale, causal, bias, dbias, bias_base);
+      off, sq, skv, scale, causal, bias, dbias, bias_base, window);
 }
 
-template <int D, bool BIASED = false>
+template <int D, bool BIASED = false, bool WINDOWED = false>
 __global__ __launch_bounds__(512, 2)
 void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           const __bf16* __restrict__ q,
@@ -996,7 +1028,8 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
                           float scale, bool causal,
                           const __bf16* __restrict__ bias = nullptr,
                           float* __restrict__ dbias = nullptr,
-                          bool paired = true, bool sbhd = false) {
+                          bool paired = true, bool sbhd = false,
+                          int window = 0) {
   constexpr int QT = 64, KBW = 256;  // must match flash_bwd_dkv_phase
   // dK phase is the larger LDS user: q rows + dO rows + transposed image
   constexpr int BUFSZ = 2 * QT * (D + 8) + D * (QT + 8);
@@ -1020,19 +1053,19 @@ void flash_bwd_dkv_kernel(const __bf16* __restrict__ dout,
   const int off = skv - sq;
   const int nkb = (skv + KBW - 1) / KBW;
   const long bias_base = (long)h * sq * skv;
-  flash_bwd_dkv_block<D, BIASED>(blockIdx.x, dout, q, k, v, lse, di, dk_exp,
+  flash_bwd_dkv_block<D, BIASED, WINDOWED>(blockIdx.x, dout, q, k, v, lse, di, dk_exp,
                                  dv_exp, smem, lsedi, q_base, kv_base,
                                  dkv_base, lse_base, q_str, kv_str,
                                  dkv_str, off, sq, skv, scale, causal, bias,
-                                 dbias, bias_base);
+                                 dbias, bias_base, window);
   const int kb2 = nkb - 1 - (int)blockIdx.x;
   if (causal && paired && kb2 > (int)blockIdx.x) {
     __syncthreads();
-    flash_bwd_dkv_block<D, BIASED>(kb2, dout, q, k, v, lse, di, dk_exp,
+    flash_bwd_dkv_block<D, BIASED, WINDOWED>(kb2, dout, q, k, v, lse, di, dk_exp,
                                    dv_exp, smem, lsedi, q_base, kv_base,
                                    dkv_base, lse_base, q_str, kv_str,
                                    dkv_str, off, sq, skv, scale, causal,
-                                   bias, dbias, bias_base);
+                                   bias, dbias, bias_base, window);
   }
 }
 
@@ -1068,7 +1101,7 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
                                const __bf16* v, __bf16* o, float* lse, int b,
                                int sq, int skv, int hq, int hkv, float scale,
                                bool causal, const __bf16* bias,
-                               bool sbhd, hipStream_t st) {
+                               bool sbhd, int window, hipStream_t st) {
   int nqb = (sq + 255) / 256;
   // complementary-pair causal scheduling halves the grid; skip it when
   // the halved grid underfills the 256-CU chip (small-seq shapes)
@@ -1077,23 +1110,28 @@ static void flash_fwd_launch_d(const __bf16* q, const __bf16* k,
   if (bias != nullptr)
     hipLaunchKernelGGL((flash_fwd_kernel<D, true>), grid, dim3(512), 0, st,
                        q, k, v, o, lse, b, sq, skv, hq, hkv, scale, causal,
-                       bias, paired, sbhd);
+                       bias, paired, sbhd, window);
+  else if (window > 0)
+    hipLaunchKernelGGL((flash_fwd_kernel<D, false, true>), grid, dim3(512),
+                       0, st, q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
+                       causal, nullptr, paired, sbhd, window);
   else
     hipLaunchKernelGGL((flash_fwd_kernel<D>), grid, dim3(512), 0, st, q, k,
                        v, o, lse, b, sq, skv, hq, hkv, scale, causal,
-                       nullptr, paired, sbhd);
+                       nullptr, paired, sbhd, 0);
 }
 
 void flash_fwd_launch(const __bf16* q, const __bf16* k, const __bf16* v,
                       __bf16* o, float* lse, int b, int sq, int skv, int hq,
                       int hkv, int d, float scale, bool causal,
-                      hipStream_t st, const __bf16* bias, bool sbhd) {
+                      hipStream_t st, const __bf16* bias, bool sbhd,
+                      int window) {
   if (d == 64)
     flash_fwd_launch_d<64>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                           causal, bias, sbhd, st);
+                           causal, bias, sbhd, window, st);
   else
     flash_fwd_launch_d<128>(q, k, v, o, lse, b, sq, skv, hq, hkv, scale,
-                            causal, bias, sbhd, st);
+                            causal, bias, sbhd, window, st);
 }
 
 void attn_di_launch(const __bf16* dout, const __bf16* o, float* di, int b,
@@ -1116,7 +1154,7 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
                                __bf16* dk_exp, __bf16* dv_exp, int b, int sq,
                                int skv, int hq, int hkv, float scale,
                                bool causal, const __bf16* bias, float* dbias,
-                               bool sbhd, hipStream_t st) {
+                               bool sbhd, int window, hipStream_t st) {
   int nqb = (sq + 255) / 256;
   const bool pq = causal && ((nqb + 1) / 2) * (long)b * hq >= 256;
   dim3 gq((causal && pq) ? (nqb + 1) / 2 : nqb, b * hq);
@@ -1126,17 +1164,27 @@ static void flash_bwd_launch_d(const __bf16* dout, const __bf16* q,
   if (bias != nullptr) {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D, true>), gq, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dq, b, sq, skv, hq, hkv,
-                       scale, causal, bias, pq, sbhd);
+                       scale, causal, bias, pq, sbhd, window);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, true>), gkv, dim3(512), 0,
                        st, dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq,
-                       skv, hq, hkv, scale, causal, bias, dbias, pkv, sbhd);
+                       skv, hq, hkv, scale, causal, bias, dbias, pkv, sbhd,
+                       window);
+  } else if (window > 0) {
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<D, false, true>), gq, dim3(512),
+                       0, st, dout, q, k, v, lse, di, dq, b, sq, skv, hq,
+                       hkv, scale, causal, nullptr, pq, sbhd, window);
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, false, true>), gkv,
+                       dim3(512), 0, st, dout, q, k, v, lse, di, dk_exp,
+                       dv_exp, b, sq, skv, hq, hkv, scale, causal, nullptr,
+                       nullptr, pkv, sbhd, window);
   } else {
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D>), gq, dim3(512), 0, st, dout,
                        q, k, v, lse, di, dq, b, sq, skv, hq, hkv, scale,
-                       causal, nullptr, pq, sbhd);
+                       causal, nullptr, pq, sbhd, 0);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D>), gkv, dim3(512), 0, st,
                        dout, q, k, v, lse, di, dk_exp, dv_exp, b, sq, skv,
-                       hq, hkv, scale, causal, nullptr, nullptr, pkv, sbhd);
+                       hq, hkv, scale, causal, nullptr, nullptr, pkv, sbhd,
+                       0);
   }
 }
 
@@ -1145,15 +1193,15 @@ void flash_bwd_launch(const __bf16* dout, const __bf16* q, const __bf16* k,
                       __bf16* dq, __bf16* dk_exp, __bf16* dv_exp, int b,
                       int sq, int skv, int hq, int hkv, int d, float scale,
                       bool causal, hipStream_t st, const __bf16* bias,
-                      float* dbias, bool sbhd) {
+                      float* dbias, bool sbhd, int window) {
   if (d == 64)
     flash_bwd_launch_d<64>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b, sq,
                            skv, hq, hkv, scale, causal, bias, dbias, sbhd,
-                           st);
+                           window, st);
   else
     flash_bwd_launch_d<128>(dout, q, k, v, lse, di, dq, dk_exp, dv_exp, b,
                             sq, skv, hq, hkv, scale, causal, bias, dbias,
-                            sbhd, st);
+                            sbhd, window, st);
 }
 
 void mfma_probe_launch(const __bf16* A, const __bf16* B, float* Dst, bool alt,

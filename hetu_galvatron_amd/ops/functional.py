@@ -155,10 +155,12 @@ class _FlashAttention(torch.autograd.Function):
         # item); window >= kv length is a no-op -> native
         eff_window = window if (window is not None and
                                 window < skv) else None
-        if eff_window is None and use_native(q) \
-                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        if use_native(q) \
+                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128) \
+                and (eff_window is None or causal):
             o, lse = get_ext(False).flash_attn_fwd(q, k, v, causal, scale,
-                                                   None, sbhd)
+                                                   None, sbhd,
+                                                   eff_window or 0)
         elif sbhd:
             ob, lse = ref.attention_fwd(
                 q.permute(1, 0, 2, 3), k.permute(1, 0, 2, 3),
@@ -178,10 +180,12 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do, dlse):
         q, k, v, o, lse = ctx.saved_tensors
         do = do.contiguous()
-        if ctx.window is None and use_native(q) \
-                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        if use_native(q) \
+                and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128) \
+                and (ctx.window is None or ctx.causal):
             dq, dk, dv = get_ext(False).flash_attn_bwd(
-                do, q, k, v, o, lse, ctx.causal, ctx.scale, None, ctx.sbhd)
+                do, q, k, v, o, lse, ctx.causal, ctx.scale, None, ctx.sbhd,
+                ctx.window or 0)
         elif ctx.sbhd:
             dq, dk, dv = ref.attention_bwd(
                 do.permute(1, 0, 2, 3), q.permute(1, 0, 2, 3),

@@ -602,3 +602,30 @@ def test_chunked_prefill_gpu():
     gen._forward_tokens(ids[:, :5], c2)
     chunked = gen._forward_tokens(ids[:, 5:], c2)
     assert (full - chunked).abs().max() < 0.1  # bf16 path tolerance
+
+
+@pytest.mark.parametrize("window", [64, 200])
+@pytest.mark.parametrize("sq", [256, 512])
+def test_flash_sliding_window(window, sq):
+    """Native mistral-style sliding window vs the fp32 reference
+    (keys visible to q: [q-window+1, q])."""
+    torch.manual_seed(18)
+    b, h, d = 2, 4, 128
+    q = torch.randn(b, sq, h, d, device=dev()).bfloat16()
+    k = torch.randn(b, sq, h, d, device=dev()).bfloat16()
+    v = torch.randn(b, sq, h, d, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext().flash_attn_fwd(q, k, v, True, scale, None, False, window)
+    o_ref, lse_ref = ref.attention_fwd(q.float(), k.float(), v.float(),
+                                       True, scale, window=window)
+    assert_close(o, o_ref, 3e-2, what=f"win{window} o")
+    assert_close(lse, lse_ref, 2e-2, what="win lse")
+    do = torch.randn_like(o)
+    dq, dk, dv = ext().flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                      True, scale, None, False, window)
+    dq_r, dk_r, dv_r = ref.attention_bwd(do.float(), q.float(), k.float(),
+                                         v.float(), None, None, True,
+                                         scale, window=window)
+    assert_close(dq, dq_r, 6e-2, rtol=3e-2, what="win dq")
+    assert_close(dk, dk_r, 6e-2, rtol=3e-2, what="win dk")
+    assert_close(dv, dv_r, 6e-2, rtol=3e-2, what="win dv")
