@@ -368,7 +368,8 @@ std::vector<Tensor> flash_attn_bwd(
 // Single-token decode attention against a KV cache (serving path).
 // q: [b,hq,d]; k_cache/v_cache: [b,max_s,hkv,d]; attends to [0, cur_len).
 Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
-                   const Tensor& v_cache, int64_t cur_len, double scale) {
+                   const Tensor& v_cache, int64_t cur_len, double scale,
+                   int64_t start = 0) {
   CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
   TORCH_CHECK(is_bf16(q), "decode_attn: bf16 only");
   TORCH_CHECK(q.dim() == 3 && k_cache.dim() == 4, "decode_attn shapes");
@@ -376,6 +377,13 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   const int max_s = k_cache.size(1), hkv = k_cache.size(2);
   TORCH_CHECK(d == 64 || d == 128, "decode_attn: head dim must be 64|128");
   TORCH_CHECK(cur_len >= 1 && cur_len <= max_s, "decode_attn: bad cur_len");
+  TORCH_CHECK(start >= 0 && start < cur_len, "decode_attn: bad start");
+  // windowed decode (mistral): positions [start, cur_len).  The kernels
+  // are unchanged — advancing the cache pointers by start rows shifts
+  // every (b, h) address uniformly within its batch row.
+  const __bf16* kp = bfp(k_cache) + (long)start * hkv * d;
+  const __bf16* vp = bfp(v_cache) + (long)start * hkv * d;
+  cur_len -= start;
   auto o = at::empty_like(q);
   // split-KV: fill >=512 workgroups (2/CU) when b*hq alone can't, but keep
   // chunks >=128 positions so the merge stays cheap
@@ -390,7 +398,7 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
                    q.options().dtype(at::kFloat));
     ws_ptr = ws.data_ptr<float>();
   }
-  decode_attn_launch(bfp(q), bfp(k_cache), bfp(v_cache), bfp_mut(o), ws_ptr,
+  decode_attn_launch(bfp(q), kp, vp, bfp_mut(o), ws_ptr,
                      n_chunks, b, hq, hkv, max_s, (int)cur_len, d,
                      (float)scale, nullptr, cur_stream());
   return o;
@@ -695,7 +703,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd);
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false, py::arg("window") = 0);
   m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"), py::arg("bias") = py::none(), py::arg("sbhd") = false, py::arg("window") = 0);
-  m.def("decode_attn", &decode_attn);
+  m.def("decode_attn", &decode_attn, py::arg("q"), py::arg("k_cache"), py::arg("v_cache"), py::arg("cur_len"), py::arg("scale"), py::arg("start") = 0);
   m.def("decode_attn_graph", &decode_attn_graph);
   m.def("mfma_probe", &mfma_probe);
   m.def("ce_max", &ce_max);

@@ -246,19 +246,23 @@ def flash_bias_attention(q, k, v, bias, causal: bool = True,
 
 
 def flash_attention_fwd_only(q, k, v, causal=True, softmax_scale=None,
-                             bias=None):
-    """No-autograd forward returning (o, lse) — building block for ring CP.
-    bias [hq, sq, skv] (t5 relative bias): native CDNA4 kernel path
-    (bf16 bias added to the pre-softmax scores in-register)."""
+                             bias=None, window=None):
+    """No-autograd forward returning (o, lse) — building block for ring CP
+    and the generator prefill.  bias [hq, sq, skv] (t5 relative bias) and
+    mistral sliding `window` both take the native CDNA4 kernel path."""
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    eff_window = window if (window is not None and
+                            window < k.shape[1]) else None
     if use_native(q) and q.dtype == torch.bfloat16 \
             and q.shape[-1] in (64, 128) \
-            and (bias is None or bias.shape[-1] % 4 == 0):
+            and (bias is None or bias.shape[-1] % 4 == 0) \
+            and (eff_window is None or (causal and bias is None)):
         bias_n = None if bias is None else bias.to(torch.bfloat16).contiguous()
         return get_ext(False).flash_attn_fwd(q.contiguous(), k.contiguous(),
                                              v.contiguous(), causal, scale,
-                                             bias_n)
-    return ref.attention_fwd(q, k, v, causal, scale, bias)
+                                             bias_n, False, eff_window or 0)
+    return ref.attention_fwd(q, k, v, causal, scale, bias,
+                             window=eff_window)
 
 
 def flash_attention_bwd_only(do, q, k, v, o, lse, causal=True,
@@ -392,25 +396,27 @@ def moe_unpermute(back: torch.Tensor, probs: torch.Tensor,
 @torch.no_grad()
 def decode_attention(q: torch.Tensor, k_cache: torch.Tensor,
                      v_cache: torch.Tensor, cur_len: int,
-                     softmax_scale=None) -> torch.Tensor:
+                     softmax_scale=None, window=None) -> torch.Tensor:
     """Single-token decode attention against a KV cache (serving path).
 
     q: [b, hq, d]; k_cache/v_cache: [b, max_s, hkv, d]; attends to
-    positions [0, cur_len).  Native CDNA4 kernel on GPU bf16 (memory-bound
-    KV streaming, decode_attn_kernel in elementwise.hip); plain torch
-    reference elsewhere.  Reference role: the optional flash-decode path
-    (nvidia_chunked_flash_attn, attention.py:398-514).
+    positions [0, cur_len), or the last `window` of them (mistral
+    sliding-window decode).  Native CDNA4 kernel on GPU bf16
+    (memory-bound KV streaming, decode_attn_kernel in elementwise.hip);
+    plain torch reference elsewhere.  Reference role: the optional
+    flash-decode path (nvidia_chunked_flash_attn, attention.py:398-514).
     """
     scale = softmax_scale if softmax_scale is not None \
         else q.shape[-1] ** -0.5
+    start = max(0, cur_len - window) if window else 0
     if q.is_cuda and native_available() and q.dtype == torch.bfloat16 \
             and q.shape[-1] in (64, 128):
         return get_ext().decode_attn(q.contiguous(), k_cache, v_cache,
-                                     int(cur_len), float(scale))
+                                     int(cur_len), float(scale), int(start))
     b, hq, d = q.shape
     hkv = k_cache.shape[2]
-    k = k_cache[:, :cur_len].float()
-    v = v_cache[:, :cur_len].float()
+    k = k_cache[:, start:cur_len].float()
+    v = v_cache[:, start:cur_len].float()
     if hq != hkv:
         rep = hq // hkv
         k = k.repeat_interleave(rep, dim=2)

@@ -76,10 +76,9 @@ class GalvatronGenerator:
         self.lm_head = self.blocks[-1].inner
         self.max_batch = max_batch
         self.max_seq = max_seq
-        if getattr(m, "sliding_window", None) is not None:
-            # decode attends the full cache; windowed eviction is a v2 item
-            assert max_seq <= m.sliding_window, \
-                "generator v1: max_seq must fit the sliding window"
+        # mistral sliding window: windowed decode + prefill are native
+        # (graph capture still needs max_seq <= window, asserted there)
+        self.window = getattr(m, "sliding_window", None)
         self.scale = 1.0 / (m.head_dim ** 0.5)
         self._dev = next(self.embedding.parameters()).device
         self._graphs: dict = {}   # batch size -> captured decode step
@@ -114,16 +113,19 @@ class GalvatronGenerator:
         # cache layout [b, s, hkv, d]
         cache.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3))
         s_new = q.shape[0]
+        win = getattr(self, "window", None)
         if s_new == 1:
             o = decode_attention(q[0], cache.k[li], cache.v[li],
-                                 pos + 1, softmax_scale=self.scale)
+                                 pos + 1, softmax_scale=self.scale,
+                                 window=win)
             o = o.unsqueeze(0)  # [1, b, hq, d]
         elif pos == 0:
             qb = q.permute(1, 0, 2, 3).contiguous()
             kb = k.permute(1, 0, 2, 3).contiguous()
             vb = v.permute(1, 0, 2, 3).contiguous()
             ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
-                                             softmax_scale=self.scale)
+                                             softmax_scale=self.scale,
+                                             window=win)
             o = ob.permute(1, 0, 2, 3)
         else:
             # chunked prefill continuation: one cross-length flash call —
@@ -133,7 +135,8 @@ class GalvatronGenerator:
             kb = cache.k[li][:, :pos + s_new].contiguous()
             vb = cache.v[li][:, :pos + s_new].contiguous()
             ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
-                                             softmax_scale=self.scale)
+                                             softmax_scale=self.scale,
+                                             window=win)
             o = ob.permute(1, 0, 2, 3)
         o = o.reshape(s_new, o.shape[1], -1)
         x = attn.linear_proj(o)
@@ -199,6 +202,20 @@ class GalvatronGenerator:
                          max_new_tokens: int = 32,
                          warmup_steps: int = 3) -> torch.Tensor:
         """Greedy generation with the decode step captured in ONE hipGraph.
+        Windowed models: the captured decode attends the full cache, so
+        the capture requires max_seq <= window (eager generate() handles
+        windowed decode natively)."""
+        if self.window is not None:
+            assert self.max_seq <= self.window, \
+                "generate_graphed: max_seq must fit the sliding window " \
+                "(use generate() for windowed decode)"
+        return self._generate_graphed_impl(input_ids, max_new_tokens,
+                                           warmup_steps)
+
+    def _generate_graphed_impl(self, input_ids: torch.Tensor,
+                               max_new_tokens: int = 32,
+                               warmup_steps: int = 3) -> torch.Tensor:
+        """(implementation; see generate_graphed docstring)
 
         The step is fully device-driven — position index, KV-cache
         index_copy, decode kernel length (decode_attn_graph reads it from
@@ -384,11 +401,8 @@ class GalvatronTPGenerator:
         self.tp_group = g0.tp_group.group
         self.max_batch = max_batch
         self.max_seq = max_seq
-        if getattr(m, "sliding_window", None) is not None:
-            # same guard as GalvatronGenerator: decode attends the full
-            # cache, so a window shorter than max_seq would silently diverge
-            assert max_seq <= m.sliding_window, \
-                "generator v1: max_seq must fit the sliding window"
+        # windowed decode is native via decode_attention(window=) below
+        self.window = getattr(m, "sliding_window", None)
         self.scale = 1.0 / (m.head_dim ** 0.5)
         self._dev = next(self.embedding.parameters()).device
 
@@ -425,13 +439,15 @@ class GalvatronTPGenerator:
         s_new = q.shape[0]
         if s_new == 1:
             o = decode_attention(q[0], cache.k[li], cache.v[li], pos + 1,
-                                 softmax_scale=self.scale).unsqueeze(0)
+                                 softmax_scale=self.scale,
+                                 window=self.window).unsqueeze(0)
         else:
             qb = q.permute(1, 0, 2, 3).contiguous()
             kb = cache.k[li][:, :pos + s_new].contiguous()
             vb = cache.v[li][:, :pos + s_new].contiguous()
             ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
-                                             softmax_scale=self.scale)
+                                             softmax_scale=self.scale,
+                                             window=self.window)
             o = ob.permute(1, 0, 2, 3)
         o = o.reshape(s_new, o.shape[1], -1)
         part = F.linear(o, attn.linear_proj.weight)  # row-parallel partial

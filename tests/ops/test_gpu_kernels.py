@@ -629,3 +629,20 @@ def test_flash_sliding_window(window, sq):
     assert_close(dq, dq_r, 6e-2, rtol=3e-2, what="win dq")
     assert_close(dk, dk_r, 6e-2, rtol=3e-2, what="win dk")
     assert_close(dv, dv_r, 6e-2, rtol=3e-2, what="win dv")
+
+
+def test_decode_attn_windowed():
+    """decode_attn(start=): pointer-offset windowing == dense reference
+    over the window slice (incl. the split-KV path at small b*hq)."""
+    torch.manual_seed(19)
+    b, hq, hkv, d, S, W = 1, 4, 2, 128, 4096, 1024
+    q = torch.randn(b, hq, d, device=dev()).bfloat16()
+    kc = torch.randn(b, S, hkv, d, device=dev()).bfloat16()
+    vc = torch.randn(b, S, hkv, d, device=dev()).bfloat16()
+    scale = d ** -0.5
+    o = ext().decode_attn(q, kc, vc, S, scale, S - W)
+    k = kc[:, S - W:S].float().repeat_interleave(hq // hkv, dim=2)
+    v = vc[:, S - W:S].float().repeat_interleave(hq // hkv, dim=2)
+    att = torch.einsum("bhd,bshd->bhs", q.float(), k) * scale
+    want = torch.einsum("bhs,bshd->bhd", att.softmax(-1), v)
+    assert_close(o, want, 3e-2, what="windowed decode")
