@@ -30,6 +30,21 @@ from .attention_impl import (
 from .rope import apply_rope_qk
 
 
+def _dropout_attention(q, k, v, causal, scale, p):
+    """[b,s,h,d] eager sdpa with attention dropout (training only)."""
+    import torch.nn.functional as F
+    hq, hkv = q.shape[2], k.shape[2]
+    qh = q.permute(0, 2, 1, 3)
+    kh = k.permute(0, 2, 1, 3)
+    vh = v.permute(0, 2, 1, 3)
+    if hq != hkv:
+        kh = kh.repeat_interleave(hq // hkv, dim=1)
+        vh = vh.repeat_interleave(hq // hkv, dim=1)
+    o = F.scaled_dot_product_attention(qh, kh, vh, is_causal=causal,
+                                       dropout_p=p, scale=scale)
+    return o.permute(0, 2, 1, 3)
+
+
 class SelfAttention(nn.Module):
     def __init__(self, model_args, tp_group, sp_group, cp_group,
                  use_ulysses: bool = False, sequence_parallel: bool = True,
@@ -72,6 +87,15 @@ class SelfAttention(nn.Module):
         cp = group_size(cp_group) if cp_group is not None else 1
         if self.window is not None:
             assert cp == 1, "sliding window + ring-CP is a v2 item"
+        # attention dropout (reference: flash_attn dropout arg) — eager
+        # sdpa path when active; composes with ulysses (full-seq inner)
+        # but not with ring-CP (dropout inside per-block softmax breaks
+        # the LSE merge) or a sliding window
+        self.attn_dropout = float(getattr(m, "attention_dropout", 0.0) or 0.0)
+        if self.attn_dropout > 0:
+            assert cp == 1, "attention_dropout + ring-CP is unsupported"
+            assert self.window is None, \
+                "attention_dropout + sliding window is unsupported"
         inner = ZigzagRingAttention(cp_group) if cp > 1 else None
         if use_ulysses and group_size(sp_group) > 1:
             self.core_attention = DistributedAttention(
@@ -97,6 +121,32 @@ class SelfAttention(nn.Module):
         k = qkv[:, :, :, self.q_per_group].reshape(s, b, -1, self.head_dim)
         v = qkv[:, :, :, self.q_per_group + 1].reshape(s, b, -1, self.head_dim)
         q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin)
+        if self.attn_dropout > 0 and self.training and attn_bias is None:
+            drop = self.attn_dropout
+            if isinstance(self.core_attention, DistributedAttention):
+                qb = q.permute(1, 0, 2, 3).contiguous()
+                kb = k.permute(1, 0, 2, 3).contiguous()
+                vb = v.permute(1, 0, 2, 3).contiguous()
+                inner = lambda qq, kk, vv, causal=True, softmax_scale=None, \
+                    window=None: _dropout_attention(
+                        qq, kk, vv, causal, softmax_scale, drop)
+                saved = self.core_attention.inner_attention
+                self.core_attention.inner_attention = inner
+                try:
+                    o = self.core_attention(qb, kb, vb, causal=self.causal,
+                                            softmax_scale=self.softmax_scale)
+                finally:
+                    self.core_attention.inner_attention = saved
+                o = o.permute(1, 0, 2, 3).reshape(s, b, -1)
+                return self.linear_proj(o)
+            assert self.core_attention is None, \
+                "attention_dropout composes with ulysses/local only"
+            o = _dropout_attention(
+                q.permute(1, 0, 2, 3).contiguous(),
+                k.permute(1, 0, 2, 3).contiguous(),
+                v.permute(1, 0, 2, 3).contiguous(),
+                self.causal, self.softmax_scale, drop)
+            return self.linear_proj(o.permute(1, 0, 2, 3).reshape(s, b, -1))
         if self.core_attention is None and attn_bias is None:
             # plain local flash takes the native [s,b,h,d] layout directly
             # (sbhd kernels) — no permute+contiguous copies either way

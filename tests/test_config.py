@@ -189,3 +189,35 @@ def test_fp16_loss_scaler_overflow_skip():
     assert math.isnan(norm)
     assert torch.equal(blk.master, before), "skipped step must not update"
     assert opt.loss_scaler.scale == s0 / 2
+
+
+def test_attention_dropout_trains():
+    """attention_dropout was silently ignored; now it routes to the eager
+    sdpa path in training (and is a no-op in eval)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import (
+        GalvatronModel, get_optimizer_and_param_scheduler, get_train_iterator)
+
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama", "attention_dropout": 0.3},
+        "train": {"global_train_batch_size": 4, "train_iters": 2,
+                  "lr": 1e-3, "lr_decay_style": "constant",
+                  "distributed_backend": "gloo"},
+    })
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    dec = model.stage_model.blocks[1].inner
+    assert dec.attention.attn_dropout == 0.3
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"))
+    ctx = next(it)
+    opt.zero_grad()
+    st1 = model.forward_backward(ctx)
+    opt.step()
+    assert st1.loss == st1.loss and st1.loss > 0
+    # dropout actually fires: two forwards of the same batch differ
+    model.stage_model.blocks[1].inner.train()
+    a = model.forward_backward(ctx).loss
+    b = model.forward_backward(ctx).loss
+    assert a != b, "attention dropout should randomize the loss"
