@@ -118,7 +118,10 @@ def search_bench_plan(cfg, world, args):
     if args.chunks:
         cfg.search.settle_chunks = args.chunks
     eng = SearchEngine(cfg, lp, op, hw)
-    best = eng.parallelism_optimization(None)
+    out_path = None
+    if os.path.isdir("gpurun_out") and int(os.environ.get("RANK", "0")) == 0:
+        out_path = f"gpurun_out/bench_searched_plan_n{world}.json"  # provenance
+    best = eng.parallelism_optimization(out_path)
     if best is None:
         return None
     if int(os.environ.get("RANK", "0")) == 0:
