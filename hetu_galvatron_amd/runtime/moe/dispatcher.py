@@ -91,29 +91,40 @@ class AlltoAllDispatcher:
         n, h = x.shape
         k = idx.shape[1]
         flat_idx = idx.reshape(-1)                      # [n*k]
+        counts = torch.bincount(flat_idx, minlength=self.num_experts)
+        self._counts = counts
+        gather_handle = None
+        lst = None
+        recv_counts = None
+        if self.ep > 1:
+            # kick the count exchange FIRST: the permute/sort below runs on
+            # the compute stream while the (tiny) allgather is in flight,
+            # and the unavoidable host sync for the a2a split sizes lands
+            # after that overlap (flex-dispatcher role, RCCL-native)
+            recv_counts = torch.empty(
+                self.ep * self.num_experts, dtype=counts.dtype,
+                device=counts.device)
+            if _is_gloo(self.ep_group.group):
+                lst = [torch.empty_like(counts) for _ in range(self.ep)]
+                gather_handle = dist.all_gather(
+                    lst, counts, group=self.ep_group.group, async_op=True)
+            else:
+                gather_handle = dist.all_gather_into_tensor(
+                    recv_counts, counts, group=self.ep_group.group,
+                    async_op=True)
         order = torch.argsort(flat_idx, stable=True)    # expert-sorted
         self._order = order
         self._probs = probs.reshape(-1)[order]          # [n*k]
         rows = order // k                               # source token row
         permuted = moe_permute(x, rows)                 # [n*k, h]
 
-        counts = torch.bincount(flat_idx, minlength=self.num_experts)
-        self._counts = counts
         if self.ep > 1:
             # tokens grouped by destination rank (experts are contiguous)
             send_splits = counts.reshape(self.ep, self.local_experts) \
                 .sum(-1)
-            recv_counts = torch.empty(
-                self.ep * self.num_experts, dtype=counts.dtype,
-                device=counts.device)
-            # exchange per-expert counts so the receiver can regroup
-            if _is_gloo(self.ep_group.group):
-                lst = [torch.empty_like(counts) for _ in range(self.ep)]
-                dist.all_gather(lst, counts, group=self.ep_group.group)
+            gather_handle.wait()
+            if lst is not None:
                 recv_counts = torch.stack(lst).reshape(-1)
-            else:
-                dist.all_gather_into_tensor(recv_counts, counts,
-                                            group=self.ep_group.group)
             recv_counts = recv_counts.reshape(self.ep, self.num_experts)
             lo = self.ep_rank * self.local_experts
             my_recv = recv_counts[:, lo:lo + self.local_experts]  # [ep, E_l]
