@@ -41,6 +41,12 @@ def main(argv=None):
     from ..search.engine import (SearchEngine, read_computation_profile,
                                  read_hardware_profiles)
 
+    argv = list(argv) if argv is not None else sys.argv[1:]
+    # cost-model introspection mode (reference search_engine.py:788
+    # check_cost_model): print per-strategy time/memory instead of searching
+    check_cm = "--check-cost-model" in argv
+    if check_cm:
+        argv.remove("--check-cost-model")
     cfg = config_from_cli(argv)
     prec = "bf16" if cfg.parallel.mixed_precision == "bf16" else "fp32"
     name = cfg.model.model_name or "model"
@@ -96,6 +102,9 @@ def main(argv=None):
         hw = default_mi355x_hardware(world)
 
     eng = SearchEngine(cfg, lp, op, hw)
+    if check_cm:
+        print(eng.check_cost_model())
+        return
     out = cfg.search.output_config_path or os.path.join(
         pdir, f"galvatron_config_{name}_{cfg.search.num_nodes}nodes_"
         f"{cfg.search.num_gpus_per_node}gpus_per_node_"

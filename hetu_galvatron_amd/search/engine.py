@@ -466,6 +466,43 @@ class SearchEngine:
             default_dp_type=self.cfg.parallel.default_dp_type,
             vtp=vtp, vsp=0, vcp=1)
 
+    def check_cost_model(self, global_bsz: Optional[int] = None,
+                         chunks: int = 1, pp: int = 1) -> str:
+        """Cost-model introspection (reference search_engine.py:788
+        check_cost_model): per-strategy time/memory for each layer type
+        at one task point, as a printable table — the debugging view for
+        why the DP picked what it picked."""
+        gbsz = global_bsz or (self.args.settle_bsz
+                              if self.args.settle_bsz > 0 else 8)
+        strategies = enumerate_strategies(
+            self.world, self.args, pp,
+            default_dp_type=self.cfg.parallel.default_dp_type)
+        lines = [f"cost model @ gbsz={gbsz} chunks={chunks} pp={pp} "
+                 f"world={self.world} (budget {self.args.memory_constraint} GB)",
+                 f"{'strategy':<34}{'t_sync ms':>10}{'t_nosync':>10}"
+                 f"{'states MB':>11}{'act MB':>9}{'total MB':>10}"]
+        for t, lp in enumerate(self.layer_profiles):
+            if len(self.layer_profiles) > 1:
+                lines.append(f"-- layer type {t}")
+            for st in strategies:
+                tsync = layer_time_cost(st, lp, self.hw, gbsz, chunks,
+                                        self.world, self.mixed_precision,
+                                        no_gradient_sync=False)
+                tnos = layer_time_cost(st, lp, self.hw, gbsz, chunks,
+                                       self.world, self.mixed_precision,
+                                       no_gradient_sync=True)
+                mem = layer_memory_cost(
+                    st, lp, gbsz, chunks, 0,
+                    self.cfg.parallel.pipeline_type, self.mixed_precision,
+                    self.cfg.parallel.reduce_grads_each_microbatch)
+                name = (f"tp{st.tp}sp{st.sp}cp{st.cp}dp{st.dp}"
+                        f"-{st.dp_type}{'-ckpt' if st.checkpoint else ''}")
+                lines.append(f"{name:<34}{tsync:>10.3f}{tnos:>10.3f}"
+                             f"{mem['model_states']:>11.0f}"
+                             f"{mem['activation']:>9.0f}"
+                             f"{mem['total']:>10.0f}")
+        return "\n".join(lines)
+
     def parallelism_optimization(self, output_path: Optional[str] = None
                                  ) -> Optional[SearchResult]:
         """Full task-grid search; saves the best plan JSON."""

@@ -383,3 +383,28 @@ def test_plan_report_tool():
     out = subprocess.run([sys.executable, "tools/plan_report.py", p],
                          capture_output=True, text=True, check=True).stdout
     assert "dp8" in out and "pp=1" in out
+
+
+def test_check_cost_model_report(tmp_path):
+    # reference search_engine.py:788 check_cost_model — introspection table
+    eng = make_engine(tmp_path)
+    rep = eng.check_cost_model()
+    # header carries the task point; every world-8 strategy appears with
+    # both time legs and the three memory columns
+    assert "gbsz=64" in rep and "world=8" in rep
+    lines = [ln for ln in rep.splitlines() if ln.startswith("tp")]
+    assert len(lines) >= 8
+    assert any("zero3" in ln for ln in lines)
+    assert any("-ckpt" in ln for ln in lines)
+    for ln in lines:
+        cols = ln.split()
+        assert len(cols) == 6, ln
+        t_sync, t_nosync = float(cols[1]), float(cols[2])
+        assert t_sync >= t_nosync > 0
+        assert float(cols[5]) >= float(cols[4]) > 0  # total >= act
+    # checkpointed rows trade time for activation memory
+    ckpt = next(ln for ln in lines if "-ckpt" in ln)
+    base = next(ln for ln in lines
+                if ln.split()[0] == ckpt.split()[0][:-len("-ckpt")])
+    assert float(ckpt.split()[1]) > float(base.split()[1])
+    assert float(ckpt.split()[4]) < float(base.split()[4])
