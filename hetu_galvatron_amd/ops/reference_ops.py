@@ -94,11 +94,16 @@ def geglu_bwd(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
 
 def rope_freqs(seq_len: int, dim: int, theta: float = 10000.0,
                device=None, dtype=torch.float32,
-               pos_offset: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+               pos_offset: int = 0,
+               interp: float = 1.0) -> Tuple[torch.Tensor, torch.Tensor]:
     """cos/sin tables [seq, dim/2] (host-precomputed per the CDNA4 elementwise
-    guideline: no on-device trig in the hot path)."""
+    guideline: no on-device trig in the hot path).  interp > 1 is linear
+    sequence-length interpolation (reference rotary_pos_embedding.py
+    rotary_seq_len_interpolation_factor: positions divided by the factor)."""
     inv_freq = 1.0 / (theta ** (torch.arange(0, dim, 2, device=device).float() / dim))
     t = torch.arange(pos_offset, pos_offset + seq_len, device=device).float()
+    if interp and interp != 1.0:
+        t = t / interp
     freqs = torch.outer(t, inv_freq)
     return freqs.cos().to(dtype), freqs.sin().to(dtype)
 
@@ -117,6 +122,22 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
     s = sin[:, None, None, :].float()
     if conj:
         s = -s
+    y1 = x1 * c - x2 * s
+    y2 = x2 * c + x1 * s
+    return torch.cat([y1, y2], dim=-1).to(x.dtype)
+
+
+def rope_apply_neox_batched(x: torch.Tensor, cos: torch.Tensor,
+                            sin: torch.Tensor) -> torch.Tensor:
+    """NEOX half-rotation with PER-BATCH tables (multimodal RoPE).
+
+    x: [s, b, h, d]; cos/sin: [s, b, d/2].
+    """
+    d2 = x.shape[-1] // 2
+    x1 = x[..., :d2].float()
+    x2 = x[..., d2:].float()
+    c = cos[:, :, None, :].float()
+    s = sin[:, :, None, :].float()
     y1 = x1 * c - x2 * s
     y2 = x2 * c + x1 * s
     return torch.cat([y1, y2], dim=-1).to(x.dtype)

@@ -124,7 +124,8 @@ class GalvatronDecoderLayer(nn.Module):
                        add_bias=margs.add_bias_linear,
                        sequence_parallel=seq_par, dtype=dtype)
         if margs.position_embedding_type == "rope":
-            self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta)
+            self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta,
+                                          scaling=margs.rope_scaling)
         else:
             self.rotary = None
         if seq_par and group_size(groups.tp_group) > 1:

@@ -150,7 +150,8 @@ class GalvatronMoEDecoderLayer(nn.Module):
                                getattr(self.post_attn_norm, "bias", None),
                                self.attention.linear_proj.bias)
         if margs.position_embedding_type == "rope":
-            self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta)
+            self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta,
+                                          scaling=margs.rope_scaling)
         else:
             self.rotary = None
         self.dropout_p = margs.hidden_dropout

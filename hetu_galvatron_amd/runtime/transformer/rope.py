@@ -31,10 +31,8 @@ class RotaryEmbedding(torch.nn.Module):
     def full_tables(self, seq_len: int, device) -> Tuple[torch.Tensor, torch.Tensor]:
         cached_len, cos, sin = self._cached
         if cos is None or cached_len < seq_len or cos.device != device:
-            cos, sin = rope_freqs(seq_len, self.head_dim, self.theta, device=device)
-            if self.scaling:
-                cos, sin = rope_freqs(int(seq_len), self.head_dim, self.theta,
-                                      device=device)
+            cos, sin = rope_freqs(seq_len, self.head_dim, self.theta,
+                                  device=device, interp=self.scaling or 1.0)
             self._cached = (seq_len, cos, sin)
         return cos[:seq_len], sin[:seq_len]
 
@@ -59,6 +57,51 @@ class RotaryEmbedding(torch.nn.Module):
             cos = cos[sp_rank * sl:(sp_rank + 1) * sl]
             sin = sin[sp_rank * sl:(sp_rank + 1) * sl]
         return cos, sin
+
+
+class MultimodalRotaryEmbedding(torch.nn.Module):
+    """3-section multimodal RoPE (reference rotary_pos_embedding.py:267
+    MultimodalRotaryEmbedding, Qwen2-VL style): the d/2 frequency channels
+    are split into ``mrope_section = [t, h, w]`` groups and each group's
+    rotation angle is driven by the matching row of ``position_ids``
+    [3, b, s] (temporal / height / width positions).  Tables stay
+    host-precomputed; only the per-batch gather runs at step time."""
+
+    def __init__(self, head_dim: int, mrope_section, theta: float = 10000.0):
+        super().__init__()
+        assert sum(mrope_section) == head_dim // 2, \
+            f"mrope_section {mrope_section} must sum to head_dim/2"
+        self.head_dim = head_dim
+        self.theta = theta
+        sect = torch.repeat_interleave(
+            torch.arange(len(mrope_section)),
+            torch.tensor(list(mrope_section)))
+        self.register_buffer("section_of_channel", sect, persistent=False)
+
+    def tables(self, position_ids: torch.Tensor
+               ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """position_ids [3, b, s] int -> cos/sin [s, b, d/2] fp32."""
+        assert position_ids.dim() == 3 and position_ids.shape[0] == 3
+        d2 = self.head_dim // 2
+        dev = position_ids.device
+        inv_freq = 1.0 / (self.theta ** (
+            torch.arange(0, self.head_dim, 2, device=dev).float()
+            / self.head_dim))
+        freqs = position_ids.float()[..., None] * inv_freq  # [3,b,s,d2]
+        sel = self.section_of_channel.to(dev).view(1, 1, 1, d2) \
+            .expand(1, *position_ids.shape[1:], d2)
+        f = freqs.gather(0, sel).squeeze(0)                 # [b,s,d2]
+        f = f.transpose(0, 1).contiguous()                  # [s,b,d2]
+        return f.cos(), f.sin()
+
+
+def apply_mrope_qk(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
+                   sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """q,k: [s, b, h, d]; cos/sin [s, b, d/2] (per-batch mrope tables).
+    NEOX half-rotation, eager (mrope is not in the dense hot path)."""
+    from ...ops.reference_ops import rope_apply_neox_batched
+    return (rope_apply_neox_batched(q, cos, sin),
+            rope_apply_neox_batched(k, cos, sin))
 
 
 def zigzag_slice(x: torch.Tensor, cp_rank: int, cp_size: int) -> torch.Tensor:
