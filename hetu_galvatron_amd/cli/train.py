@@ -43,6 +43,13 @@ def main(argv=None):
         torch.manual_seed(cfg.train.seed)
     rank = dist.get_rank() if dist.is_initialized() else 0
 
+    if rank == 0:
+        # resolved-args dump (reference initialize.py:240 _print_args)
+        import json as _json
+        print("[galvatron args] " +
+              _json.dumps(cfg.model_dump(), default=str, sort_keys=True),
+              flush=True)
+
     model = GalvatronModel(cfg, device=device)
     opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
     # batch ramp-up (reference num_microbatches_calculator): unit =
