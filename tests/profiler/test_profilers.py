@@ -215,3 +215,24 @@ def test_overlap_worker_cpu_default():
     from tests.utils import run_distributed
     res = run_distributed(_overlap_worker, world_size=2)
     assert res[0]["overlap_coe"] >= 1.0
+
+
+def test_profile_hardware_cli_scripts_only(tmp_path, monkeypatch):
+    # CLI entry (reference profile_hardware.py:7): --scripts-only writes the
+    # torchrun sweep scripts without launching anything
+    monkeypatch.chdir(tmp_path)
+    from hetu_galvatron_amd.cli.profile_hardware import main
+    main(["--scripts-only",
+          "profile_hardware.num_nodes=1",
+          "profile_hardware.num_gpus_per_node=8",
+          f"profile_hardware.hardware_config_dir={tmp_path}/hw"])
+    import glob
+    scripts = glob.glob(str(tmp_path / "scripts" / "*.sh"))
+    assert len(scripts) >= 4, scripts
+    names = " ".join(scripts)
+    for op in ("allreduce", "p2p", "sp_time", "overlap"):
+        assert op in names, f"missing sweep script for {op}"
+    # each script must be a single-node torchrun invocation on loopback
+    body = open(scripts[0]).read()
+    assert "torchrun" in body or "torch.distributed.run" in body
+    assert "127.0.0.1" in body
