@@ -408,3 +408,13 @@ def test_check_cost_model_report(tmp_path):
                 if ln.split()[0] == ckpt.split()[0][:-len("-ckpt")])
     assert float(ckpt.split()[1]) > float(base.split()[1])
     assert float(ckpt.split()[4]) < float(base.split()[4])
+
+
+def test_example_yamls_load():
+    import glob
+    from hetu_galvatron_amd.config.loader import load_config
+    yamls = glob.glob("examples/*.yaml")
+    assert len(yamls) >= 5
+    for y in yamls:
+        cfg = load_config(y)
+        assert cfg.model.model_name
