@@ -345,3 +345,32 @@ def test_world8_dp8_zero2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+@pytest.mark.distributed
+def test_dp2_fp16_loss_scaling_matches_single_process():
+    """fp16 + dynamic loss scaling under dp2/zero2: the scale is folded
+    into the fused-optimizer gscale and overflow decisions come off the
+    all-reduced grad norm, so two ranks must track a single fp16 process
+    exactly (same init state, same data)."""
+    from tests.utils import run_distributed
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+
+    _, state_path = get_baseline()
+    extra = {"parallel": {"mixed_precision": "fp16",
+                          "loss_scale_init": 1024.0}}
+    cfg = make_cfg(extra)
+    model = GalvatronModel(cfg)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    base_losses = train_steps(model, cfg)
+
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, dp_type="zero2",
+                                      global_bsz=4)
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path, extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < 5e-3, \
+                f"rank {r} step {s}: dist {a:.4f} vs fp16 baseline {b:.4f}"
