@@ -95,6 +95,7 @@ class ModelArgs(BaseModel):
     moe_aux_loss_free: bool = False
     moe_router_bias_update_rate: float = 1e-3
     moe_aux_loss_type: str = "aux_loss"  # aux_loss | seq_aux_loss
+    moe_router_load_balancing_type: str = "none"  # none | sinkhorn
     # DeepSeek-style group-limited (node-limited) routing
     moe_router_num_groups: Optional[int] = None
     moe_router_group_topk: Optional[int] = None

@@ -59,6 +59,8 @@ class TopKRouter(nn.Module):
         self.bias_update_rate = margs.moe_router_bias_update_rate
         self.capacity_factor = margs.moe_expert_capacity_factor
         self.aux_loss_type = getattr(margs, "moe_aux_loss_type", "aux_loss")
+        self.load_balancing_type = getattr(
+            margs, "moe_router_load_balancing_type", "none")
         self.num_groups = getattr(margs, "moe_router_num_groups", None)
         self.group_topk = getattr(margs, "moe_router_group_topk", None)
         if self.num_groups:
@@ -94,6 +96,12 @@ class TopKRouter(nn.Module):
             scores = logits
 
         sel = scores
+        if self.load_balancing_type == "sinkhorn" and self.training:
+            # sinkhorn routing (reference router.py:140): expert choice
+            # comes from the balanced sinkhorn matrix, probs from the
+            # raw scores at those experts (gradient flows through scores)
+            with torch.no_grad():
+                sel = sinkhorn(logits.detach().float())
         if self.expert_bias is not None:
             sel = scores + self.expert_bias.unsqueeze(0)
         if self.num_groups:

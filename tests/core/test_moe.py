@@ -370,3 +370,50 @@ def test_moe_world4_pp2_etp2():
     for losses in res:
         for a, b in zip(losses, base_losses):
             assert abs(a - b) < TOL, (losses, base_losses)
+
+
+def test_router_sinkhorn_balances_selection():
+    """Sinkhorn routing (reference router.py:140): with training-mode
+    sinkhorn the expert choice is load-balanced even when raw logits all
+    prefer one expert; eval mode routes by raw scores again."""
+    import torch
+    from hetu_galvatron_amd.runtime.moe.router import TopKRouter, sinkhorn
+
+    class M:
+        hidden_size = 16
+        num_experts = 4
+        moe_router_topk = 1
+        moe_aux_loss_coeff = 0.0
+        moe_z_loss_coeff = 0.0
+        moe_router_score_function = "softmax"
+        moe_router_pre_softmax = False
+        moe_aux_loss_free = False
+        moe_router_bias_update_rate = 0.0
+        moe_aux_loss_type = "aux_loss"
+        moe_router_num_groups = None
+        moe_router_group_topk = None
+        moe_expert_capacity_factor = None
+        moe_router_load_balancing_type = "sinkhorn"
+
+    torch.manual_seed(0)
+    r = TopKRouter(M())
+    # bias every logit toward expert 0
+    with torch.no_grad():
+        r.weight.zero_()
+        r.weight[0] += 1.0
+    x = torch.randn(64, 16).abs()  # positive -> expert-0 logit dominates
+    r.train()
+    _, idx, _ = r(x)
+    counts = torch.bincount(idx.flatten(), minlength=4)
+    # sinkhorn spreads the load: no expert takes everything
+    assert counts.max() < 64, counts.tolist()
+    assert (counts > 0).sum() >= 2, counts.tolist()
+    r.eval()
+    _, idx_eval, _ = r(x)
+    # raw routing sends everything to the dominant expert
+    assert torch.bincount(idx_eval.flatten(), minlength=4)[0] == 64
+
+    # sinkhorn output is (approximately) doubly stochastic
+    m = sinkhorn(torch.randn(32, 4))
+    torch.testing.assert_close(m.sum(1), torch.full((32,), 1.0 / 32),
+                               atol=1e-2, rtol=1e-2)
