@@ -86,3 +86,32 @@ def test_batch_calculator_rampup_properties(unit, start_u, incr_u, steps,
     assert start <= g <= target    # monotone between start and target
     calc.update(10 ** 9)
     assert calc.get()[0] == target  # ramp completes
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.lists(st.lists(st.integers(0, 2 ** 15 - 1), min_size=1,
+                         max_size=50), min_size=1, max_size=20),
+       st.sampled_from(["int32", "uint16", "int64"]))
+def test_mmididx_roundtrip_random_docs(docs, dtype_name):
+    """Megatron MMIDIDX writer -> reader identity for arbitrary corpora
+    and every wire dtype (reference indexed_dataset.py format)."""
+    import tempfile
+
+    import numpy as np
+
+    from hetu_galvatron_amd.runtime.datasets.indexed import (
+        MegatronIndexedDataset, MegatronIndexedDatasetBuilder)
+    dt = np.dtype(dtype_name)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = f"{td}/corpus"
+        b = MegatronIndexedDatasetBuilder(prefix, dtype=dt)
+        for d in docs:
+            b.add_document(np.asarray(d, dtype=dt))
+        b.finalize()
+        ds = MegatronIndexedDataset(prefix)
+        assert len(ds) == len(docs)
+        assert ds.doc_lens.tolist() == [len(d) for d in docs]
+        for i, d in enumerate(docs):
+            got = np.asarray(ds.doc(i))
+            assert got.dtype == dt
+            assert got.tolist() == d
