@@ -281,6 +281,15 @@ def test_world4_pp2_cp2():
         vtp=1))
 
 
+@pytest.mark.distributed
+def test_world4_cp4():
+    """cp=4 zigzag ring: 3-hop KV rotation fwd + dkv ring bwd (the cp2
+    tests only exercise a single exchange; multi-hop covers the ring
+    schedule and LSE merges at depth)."""
+    run_case(4, HybridParallelPlan.uniform(
+        N_LAYERS, 4, cp=4, dp_type="ddp", global_bsz=4, chunks=1, vtp=1))
+
+
 def get_baseline_with(cfg_extra):
     """1-process baseline under extra model config (e.g. sliding window)."""
     from hetu_galvatron_amd.runtime import GalvatronModel
