@@ -383,3 +383,25 @@ def test_dp2_fp16_loss_scaling_matches_single_process():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < 5e-3, \
                 f"rank {r} step {s}: dist {a:.4f} vs fp16 baseline {b:.4f}"
+
+
+@pytest.mark.distributed
+def test_world8_pp2_tp2_dp2():
+    """Full 3D composition (pp2 x tp2 x dp2) on 8 ranks: boundary
+    redistribution between tp shards across pipeline stages plus dp
+    gradient reduction — the classic hybrid topology the search emits
+    for larger models."""
+    from tests.utils import run_distributed
+    cfg_extra = {"train": {"global_train_batch_size": 8}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 8, pp=2, tp=2,
+                                      dp_type="zero2", global_bsz=8,
+                                      chunks=2, vtp=2)
+    res = run_distributed(_dist_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
