@@ -405,3 +405,32 @@ def test_world8_pp2_tp2_dp2():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_world4_ulysses2_pp2():
+    """Ulysses SP across pipeline stages (pp2 x sp2): sequence-sharded
+    boundary activations + head-scatter a2a inside each stage."""
+    run_case(4, HybridParallelPlan.uniform(
+        N_LAYERS, 4, pp=2, tp=2, use_sp=True, dp_type="ddp",
+        global_bsz=4, chunks=2, vtp=2, vsp=True))
+
+
+@pytest.mark.distributed
+def test_world8_pp2_cp2_dp2():
+    """pp2 x cp2 x dp2 on 8 ranks: ring attention and dp reduction under
+    a pipeline split."""
+    from tests.utils import run_distributed
+    cfg_extra = {"train": {"global_train_batch_size": 8}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 8, pp=2, cp=2,
+                                      dp_type="ddp", global_bsz=8,
+                                      chunks=2, vtp=1)
+    res = run_distributed(_dist_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"
