@@ -39,3 +39,19 @@ def test_consistency_detects_divergence(perturb):
             assert r["dirty"], "divergence not detected"
         else:
             assert r["dirty"] == []
+
+
+def test_accuracy_alignment_tool(tmp_path):
+    """tools/accuracy_alignment.py (reference scripts/accuracy_alignment/):
+    1-process self-alignment is exact."""
+    import json
+    import subprocess
+    import sys
+    out = tmp_path / "align.json"
+    r = subprocess.run(
+        [sys.executable, "tools/accuracy_alignment.py", "--model",
+         "tiny-llama", "--iters", "2", "--out", str(out)],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = json.load(open(out))
+    assert d["aligned"] and d["max_abs_delta"] == 0.0
