@@ -135,6 +135,14 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         add_qkv_bias=True, tie_word_embeddings=True,
         untie_embeddings_and_output_weights=False,
     ),
+    "tiny-qwen3": dict(
+        model_type="llama", hidden_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, ffn_hidden_size=256,
+        vocab_size=512, max_position_embeddings=256, seq_length=32,
+        hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-6,
+        position_embedding_type="rope", rope_theta=10000.0,
+        qk_layernorm=True,
+    ),
     "tiny-qwen": dict(
         model_type="llama", hidden_size=128, num_hidden_layers=2,
         num_attention_heads=2, num_key_value_heads=1, kv_channels=64,

@@ -65,6 +65,7 @@ class ModelArgs(BaseModel):
     seq_length: int = 4096
     hidden_act: str = "silu"  # silu(swiglu) | gelu | geglu | relu
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
+    qk_layernorm: bool = False  # per-head norm on q/k pre-RoPE (Qwen3/Gemma2)
     norm_epsilon: float = 1e-5
     position_embedding_type: str = "rope"  # rope | learned | relative
     relative_attention_num_buckets: int = 32    # t5 bucketized bias

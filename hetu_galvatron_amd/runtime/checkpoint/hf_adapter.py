@@ -62,6 +62,11 @@ def hf_llama_to_canonical(hf: Dict[str, torch.Tensor],
                 hf[p + "self_attn.q_proj.bias"],
                 hf[p + "self_attn.k_proj.bias"],
                 hf[p + "self_attn.v_proj.bias"], margs)
+        if p + "self_attn.q_norm.weight" in hf:  # qwen3/gemma2 qk norm
+            out[c + "attention.q_layernorm.weight"] = \
+                hf[p + "self_attn.q_norm.weight"]
+            out[c + "attention.k_layernorm.weight"] = \
+                hf[p + "self_attn.k_norm.weight"]
         out[c + "attention.linear_proj.weight"] = \
             hf[p + "self_attn.o_proj.weight"]
         out[c + "mlp.fc1.weight"] = torch.cat(
@@ -95,6 +100,11 @@ def canonical_to_hf_llama(can: Dict[str, torch.Tensor],
             out[p + "self_attn.q_proj.bias"] = qb
             out[p + "self_attn.k_proj.bias"] = kb
             out[p + "self_attn.v_proj.bias"] = vb
+        if c + "attention.q_layernorm.weight" in can:
+            out[p + "self_attn.q_norm.weight"] = \
+                can[c + "attention.q_layernorm.weight"]
+            out[p + "self_attn.k_norm.weight"] = \
+                can[c + "attention.k_layernorm.weight"]
         out[p + "self_attn.o_proj.weight"] = \
             can[c + "attention.linear_proj.weight"]
         fc1 = can[c + "mlp.fc1.weight"]

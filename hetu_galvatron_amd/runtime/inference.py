@@ -105,6 +105,9 @@ class GalvatronGenerator:
         x = layer.input_norm(hidden)
         qkv = attn.linear_qkv(x)
         q, k, v = self._split_qkv(attn, qkv)
+        if getattr(attn, "q_layernorm", None) is not None:
+            q = attn.q_layernorm(q)
+            k = attn.k_layernorm(k)
         if layer.rotary is not None:
             cos, sin = layer.rotary.full_tables(pos + q.shape[0], hidden.device)
             from .transformer.rope import apply_rope_qk
@@ -277,6 +280,9 @@ class GalvatronGenerator:
             x = layer.input_norm(h)
             qkv = attn.linear_qkv(x)
             q, k, v = self._split_qkv(attn, qkv)
+            if getattr(attn, "q_layernorm", None) is not None:
+                q = attn.q_layernorm(q)
+                k = attn.k_layernorm(k)
             if layer.rotary is not None:
                 cos_t, sin_t = layer.rotary.full_tables(self.max_seq,
                                                         h.device)
@@ -430,6 +436,9 @@ class GalvatronTPGenerator:
         qkv = F.linear(x, attn.linear_qkv.weight,
                        getattr(attn.linear_qkv, "bias", None))
         q, k, v = GalvatronGenerator._split_qkv(attn, qkv)
+        if getattr(attn, "q_layernorm", None) is not None:
+            q = attn.q_layernorm(q)
+            k = attn.k_layernorm(k)
         if layer.rotary is not None:
             cos, sin = layer.rotary.full_tables(pos + q.shape[0], x.device)
             from .transformer.rope import apply_rope_qk

@@ -84,6 +84,20 @@ class SelfAttention(nn.Module):
             if getattr(m, "attention_softmax_scale", None) is not None \
             else 1.0 / math.sqrt(self.head_dim)
         self.window = getattr(m, "sliding_window", None)
+        # qk_layernorm (reference attention.py:917-921, Qwen3/Llama4/
+        # Gemma2): per-head-dim norm on q and k after the QKV split,
+        # before RoPE; params are tp-replicated (every tp rank normalizes
+        # its own heads with the same [head_dim] weight)
+        self.q_layernorm = self.k_layernorm = None
+        if getattr(m, "qk_layernorm", False):
+            from .norm import LayerNorm, RMSNorm
+            ncls = RMSNorm if m.normalization == "rmsnorm" else LayerNorm
+            self.q_layernorm = ncls(self.head_dim, eps=m.norm_epsilon,
+                                    dtype=dtype)
+            self.k_layernorm = ncls(self.head_dim, eps=m.norm_epsilon,
+                                    dtype=dtype)
+            for p_ in list(self.q_layernorm.parameters()) +                     list(self.k_layernorm.parameters()):
+                p_.tp_replicated = True
         cp = group_size(cp_group) if cp_group is not None else 1
         if self.window is not None:
             assert cp == 1, "sliding window + ring-CP is a v2 item"
@@ -120,6 +134,9 @@ class SelfAttention(nn.Module):
         q = qkv[:, :, :, : self.q_per_group].reshape(s, b, -1, self.head_dim)
         k = qkv[:, :, :, self.q_per_group].reshape(s, b, -1, self.head_dim)
         v = qkv[:, :, :, self.q_per_group + 1].reshape(s, b, -1, self.head_dim)
+        if self.q_layernorm is not None:
+            q = self.q_layernorm(q)
+            k = self.k_layernorm(k)
         q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin)
         if self.attn_dropout > 0 and self.training and attn_bias is None:
             drop = self.attn_dropout

@@ -434,3 +434,22 @@ def test_world8_pp2_cp2_dp2():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_tp2_qk_layernorm():
+    """qk_layernorm under Megatron-TP: the [head_dim] norm weights are
+    tp-replicated (each rank normalizes its own heads), so their grads
+    must be summed over the tp group — the tp_replicated tag path."""
+    extra = {"model": {"qk_layernorm": True}}
+    base_losses, state_path = get_baseline_with(extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, tp=2, vtp=2,
+                                      global_bsz=4)
+    from tests.utils import run_distributed
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path, extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
+                f"({losses} vs {base_losses})"

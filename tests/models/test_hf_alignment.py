@@ -255,3 +255,33 @@ def test_t5_logits_match_hf():
             x = blk(x, ctx)
         got = blk.inner.lm_head(x).permute(1, 0, 2)
     assert (got - want).abs().max() < 1e-4, (got - want).abs().max()
+
+
+def test_qwen3_qk_layernorm_logits_match_hf():
+    """qk_layernorm semantics (reference attention.py:917-921): per-head
+    RMSNorm on q/k after the QKV split, before RoPE — validated end to
+    end against HF Qwen3."""
+    from transformers import Qwen3Config, Qwen3ForCausalLM
+    from hetu_galvatron_amd.runtime.checkpoint.hf_adapter import (
+        canonical_to_hf_llama)
+    cfg, model = build("tiny-qwen3")
+    m = cfg.model
+    hf_sd = canonical_to_hf_llama(
+        canonical_state_from_stage(model.stage_model), m)
+    hf = Qwen3ForCausalLM(Qwen3Config(
+        vocab_size=m.vocab_size, hidden_size=m.hidden_size,
+        intermediate_size=m.ffn_hidden_size,
+        num_hidden_layers=m.num_hidden_layers,
+        num_attention_heads=m.num_attention_heads,
+        num_key_value_heads=m.kv_heads, head_dim=m.head_dim,
+        max_position_embeddings=m.max_position_embeddings,
+        rms_norm_eps=m.norm_epsilon, rope_theta=m.rope_theta,
+        attention_bias=False, tie_word_embeddings=False))
+    missing, unexpected = hf.load_state_dict(hf_sd, strict=False)
+    assert not missing and not unexpected, (missing, unexpected)
+    hf.eval()
+    ids = torch.randint(0, m.vocab_size, (2, 16))
+    with torch.no_grad():
+        want = hf(ids).logits
+    got = our_logits(model, ids)
+    assert (got - want).abs().max() < TOL, (got - want).abs().max()
