@@ -66,6 +66,7 @@ class ModelArgs(BaseModel):
     hidden_act: str = "silu"  # silu(swiglu) | gelu | geglu | relu
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
     qk_layernorm: bool = False  # per-head norm on q/k pre-RoPE (Qwen3/Gemma2)
+    rotary_interleaved: bool = False  # GPT-J pairwise RoPE (default: NEOX halves)
     norm_epsilon: float = 1e-5
     position_embedding_type: str = "rope"  # rope | learned | relative
     relative_attention_num_buckets: int = 32    # t5 bucketized bias

@@ -127,6 +127,21 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
     return torch.cat([y1, y2], dim=-1).to(x.dtype)
 
 
+def rope_apply_interleaved(x: torch.Tensor, cos: torch.Tensor,
+                           sin: torch.Tensor) -> torch.Tensor:
+    """Interleaved (GPT-J style) rotary embedding: channel pairs
+    (2i, 2i+1) rotate together (reference rotary_interleaved arg).
+    x: [s, b, h, d]; cos/sin: [s, d/2]."""
+    x1 = x[..., 0::2].float()
+    x2 = x[..., 1::2].float()
+    c = cos[:, None, None, :].float()
+    s_ = sin[:, None, None, :].float()
+    y = torch.empty_like(x, dtype=torch.float32)
+    y[..., 0::2] = x1 * c - x2 * s_
+    y[..., 1::2] = x2 * c + x1 * s_
+    return y.to(x.dtype)
+
+
 def rope_apply_neox_batched(x: torch.Tensor, cos: torch.Tensor,
                             sin: torch.Tensor) -> torch.Tensor:
     """NEOX half-rotation with PER-BATCH tables (multimodal RoPE).

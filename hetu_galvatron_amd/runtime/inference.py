@@ -112,7 +112,9 @@ class GalvatronGenerator:
             cos, sin = layer.rotary.full_tables(pos + q.shape[0], hidden.device)
             from .transformer.rope import apply_rope_qk
             q, k = apply_rope_qk(q.contiguous(), k.contiguous(),
-                                 cos[pos:], sin[pos:])
+                                 cos[pos:], sin[pos:],
+                                 interleaved=getattr(
+                                     attn, "rope_interleaved", False))
         # cache layout [b, s, hkv, d]
         cache.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3))
         s_new = q.shape[0]
@@ -272,6 +274,10 @@ class GalvatronGenerator:
         """One decode step on static buffers (capture-safe)."""
         from ..ops._ext import get_ext
         from ..ops import apply_rope
+        assert not getattr(self.layers[0].attention, "rope_interleaved",
+                           False), \
+            "graphed decode uses the fused NEOX RoPE kernel; " \
+            "rotary_interleaved models must use generate()"
         ext = get_ext()
         h = self.embedding.word_embeddings(ids_buf)  # [1, b, h]
         for li, layer in enumerate(self.layers):
@@ -443,7 +449,9 @@ class GalvatronTPGenerator:
             cos, sin = layer.rotary.full_tables(pos + q.shape[0], x.device)
             from .transformer.rope import apply_rope_qk
             q, k = apply_rope_qk(q.contiguous(), k.contiguous(),
-                                 cos[pos:], sin[pos:])
+                                 cos[pos:], sin[pos:],
+                                 interleaved=getattr(
+                                     attn, "rope_interleaved", False))
         cache.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3))
         s_new = q.shape[0]
         if s_new == 1:

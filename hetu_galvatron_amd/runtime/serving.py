@@ -107,11 +107,16 @@ class ContinuousBatchingEngine:
             residual = h
             x = layer.input_norm(h)
             q, k, v = gen._split_qkv(attn, attn.linear_qkv(x))
+            if getattr(attn, "q_layernorm", None) is not None:
+                q = attn.q_layernorm(q)
+                k = attn.k_layernorm(k)
             if layer.rotary is not None:
                 cos, sin = layer.rotary.full_tables(q.shape[0], h.device)
                 from .transformer.rope import apply_rope_qk
                 q, k = apply_rope_qk(q.contiguous(), k.contiguous(),
-                                     cos, sin)
+                                     cos, sin,
+                                     interleaved=getattr(
+                                         attn, "rope_interleaved", False))
             view.write(li, k.permute(1, 0, 2, 3), v.permute(1, 0, 2, 3), 0)
             qb = q.permute(1, 0, 2, 3).contiguous()
             kb = k.permute(1, 0, 2, 3).contiguous()

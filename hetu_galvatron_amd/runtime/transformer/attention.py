@@ -84,6 +84,7 @@ class SelfAttention(nn.Module):
             if getattr(m, "attention_softmax_scale", None) is not None \
             else 1.0 / math.sqrt(self.head_dim)
         self.window = getattr(m, "sliding_window", None)
+        self.rope_interleaved = bool(getattr(m, "rotary_interleaved", False))
         # qk_layernorm (reference attention.py:917-921, Qwen3/Llama4/
         # Gemma2): per-head-dim norm on q and k after the QKV split,
         # before RoPE; params are tp-replicated (every tp rank normalizes
@@ -137,7 +138,8 @@ class SelfAttention(nn.Module):
         if self.q_layernorm is not None:
             q = self.q_layernorm(q)
             k = self.k_layernorm(k)
-        q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin)
+        q, k = apply_rope_qk(q.contiguous(), k.contiguous(), cos, sin,
+                             interleaved=self.rope_interleaved)
         if self.attn_dropout > 0 and self.training and attn_bias is None:
             drop = self.attn_dropout
             if isinstance(self.core_attention, DistributedAttention):

@@ -125,9 +125,16 @@ def zigzag_unslice_index(cp_size: int) -> list:
 
 
 def apply_rope_qk(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
-                  sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+                  sin: torch.Tensor, interleaved: bool = False
+                  ) -> Tuple[torch.Tensor, torch.Tensor]:
     """q,k: [s, b, h, d] SBH-heads layout. cos=None => no RoPE (learned
-    positions, gpt family)."""
+    positions, gpt family).  interleaved: GPT-J pairwise rotation
+    (eager — the HIP kernel implements the NEOX-half layout; interleaved
+    models are rare enough that the fused path stays NEOX-only)."""
     if cos is None:
         return q, k
+    if interleaved:
+        from ...ops.reference_ops import rope_apply_interleaved
+        return (rope_apply_interleaved(q, cos, sin),
+                rope_apply_interleaved(k, cos, sin))
     return apply_rope(q, cos, sin), apply_rope(k, cos, sin)

@@ -66,3 +66,46 @@ def test_mrope_section_must_cover_half_dim():
     import pytest
     with pytest.raises(AssertionError):
         MultimodalRotaryEmbedding(64, [8, 8, 8])
+
+
+def test_rope_interleaved_semantics():
+    """GPT-J pairwise rotation: pairs (2i, 2i+1) rotate by theta_i — the
+    interleaved table equals NEOX applied to the de-interleaved tensor."""
+    import torch
+    from hetu_galvatron_amd.ops.reference_ops import (rope_apply,
+                                                      rope_apply_interleaved,
+                                                      rope_freqs)
+    torch.manual_seed(0)
+    s, b, h, d = 6, 2, 3, 16
+    x = torch.randn(s, b, h, d)
+    cos, sin = rope_freqs(s, d)
+    got = rope_apply_interleaved(x, cos, sin)
+    # de-interleave -> NEOX halves -> re-interleave
+    perm = torch.cat([torch.arange(0, d, 2), torch.arange(1, d, 2)])
+    inv = torch.argsort(perm)
+    want = rope_apply(x[..., perm], cos, sin)[..., inv]
+    torch.testing.assert_close(got, want, atol=1e-6, rtol=1e-6)
+    # norm-preserving (pure rotation)
+    torch.testing.assert_close(got.norm(), x.norm(), atol=1e-4, rtol=1e-5)
+
+
+def test_rope_interleaved_end_to_end_flag():
+    """rotary_interleaved must change the model's output (the flag is
+    consumed, not dropped)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+
+    losses = {}
+    for flag in (False, True):
+        cfg = load_config(base={
+            "model": {"model_name": "tiny-llama",
+                      "rotary_interleaved": flag},
+            "train": {"global_train_batch_size": 2, "train_iters": 1,
+                      "lr": 1e-3}})
+        torch.manual_seed(0)
+        m = GalvatronModel(cfg)
+        torch.manual_seed(1)
+        it = get_train_iterator(cfg, torch.device("cpu"))
+        losses[flag] = m.forward_backward(next(it)).loss
+    assert losses[False] != losses[True]
