@@ -67,6 +67,7 @@ class ModelArgs(BaseModel):
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
     qk_layernorm: bool = False  # per-head norm on q/k pre-RoPE (Qwen3/Gemma2)
     rotary_interleaved: bool = False  # GPT-J pairwise RoPE (default: NEOX halves)
+    init_method_std: float = 0.02  # weight-init normal std (output projs scaled /sqrt(2L))
     norm_epsilon: float = 1e-5
     position_embedding_type: str = "rope"  # rope | learned | relative
     relative_attention_num_buckets: int = 32    # t5 bucketized bias

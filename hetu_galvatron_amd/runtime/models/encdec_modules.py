@@ -30,6 +30,7 @@ import torch.nn.functional as F
 from ...config.schema import ModelArgs
 from ..redistribute import natural_rows
 from ..tensor_parallel import group_size
+from ..tensor_parallel.layers import normal_init, scaled_init
 from ..tensor_parallel.mappings import gather_from_sequence_parallel_region
 from ..transformer import MLP, SelfAttention, build_norm
 from ..transformer.attention import CrossAttention
@@ -113,7 +114,10 @@ class GalvatronEncoderLayer(nn.Module):
                        if margs.hidden_act in ("silu", "swiglu")
                        else margs.hidden_act,
                        add_bias=margs.add_bias_linear,
-                       sequence_parallel=seq_par, dtype=dtype)
+                       sequence_parallel=seq_par, dtype=dtype,
+                       init_method=normal_init(margs.init_method_std),
+                       output_init_method=scaled_init(
+                           margs.init_method_std, margs.num_hidden_layers))
         if seq_par and group_size(groups.tp_group) > 1:
             _tag_tp_replicated(self.input_norm.weight,
                                getattr(self.input_norm, "bias", None),
@@ -214,7 +218,10 @@ class GalvatronDecoderLayerX(nn.Module):
                        if margs.hidden_act in ("silu", "swiglu")
                        else margs.hidden_act,
                        add_bias=margs.add_bias_linear,
-                       sequence_parallel=seq_par, dtype=dtype)
+                       sequence_parallel=seq_par, dtype=dtype,
+                       init_method=normal_init(margs.init_method_std),
+                       output_init_method=scaled_init(
+                           margs.init_method_std, margs.num_hidden_layers))
         if seq_par and group_size(groups.tp_group) > 1:
             _tag_tp_replicated(self.input_norm.weight,
                                getattr(self.input_norm, "bias", None),

@@ -25,6 +25,7 @@ from ..tensor_parallel import (
     ColumnParallelLinear, VocabParallelEmbedding, group_rank, group_size,
     vocab_parallel_cross_entropy,
 )
+from ..tensor_parallel.layers import normal_init, scaled_init
 from ..transformer import MLP, RotaryEmbedding, SelfAttention, build_norm
 
 
@@ -56,16 +57,19 @@ class GalvatronEmbedding(nn.Module):
         if s.use_ulysses:
             # vocab-sp: weight replicated over the sp group, seq sharded
             self.word_embeddings = VocabParallelEmbedding(
-                margs.vocab_size, margs.hidden_size, None, dtype=dtype)
+                margs.vocab_size, margs.hidden_size, None, dtype=dtype,
+                init_method=normal_init(margs.init_method_std))
         else:
             self.word_embeddings = VocabParallelEmbedding(
                 margs.vocab_size, margs.hidden_size, groups.tp_group,
-                sequence_parallel=True, dtype=dtype)
+                sequence_parallel=True, dtype=dtype,
+                init_method=normal_init(margs.init_method_std))
         if margs.position_embedding_type == "learned":
             self.position_embeddings = nn.Embedding(
                 margs.max_position_embeddings, margs.hidden_size,
                 **({"dtype": dtype} if dtype else {}))
-            nn.init.normal_(self.position_embeddings.weight, 0.0, 0.02)
+            nn.init.normal_(self.position_embeddings.weight, 0.0,
+                            margs.init_method_std)
         else:
             self.position_embeddings = None
         self.dropout_p = margs.hidden_dropout
@@ -122,7 +126,10 @@ class GalvatronDecoderLayer(nn.Module):
                        hidden_act="silu" if margs.hidden_act in ("silu", "swiglu")
                        else margs.hidden_act,
                        add_bias=margs.add_bias_linear,
-                       sequence_parallel=seq_par, dtype=dtype)
+                       sequence_parallel=seq_par, dtype=dtype,
+                       init_method=normal_init(margs.init_method_std),
+                       output_init_method=scaled_init(
+                           margs.init_method_std, margs.num_hidden_layers))
         if margs.position_embedding_type == "rope":
             self.rotary = RotaryEmbedding(margs.head_dim, margs.rope_theta,
                                           scaling=margs.rope_scaling)
@@ -204,11 +211,13 @@ class GalvatronCausalLMHead(nn.Module):
         if s.use_ulysses:
             self.lm_head = ColumnParallelLinear(
                 margs.hidden_size, margs.vocab_size, None, bias=False,
-                dtype=dtype)
+                dtype=dtype,
+                init_method=normal_init(margs.init_method_std))
         else:
             self.lm_head = ColumnParallelLinear(
                 margs.hidden_size, margs.vocab_size, groups.tp_group,
-                bias=False, sequence_parallel=True, dtype=dtype)
+                bias=False, sequence_parallel=True, dtype=dtype,
+                init_method=normal_init(margs.init_method_std))
 
     def tie_to(self, embedding: GalvatronEmbedding) -> None:
         """Share the vocab-sharded weight with the embedding (same layout)."""

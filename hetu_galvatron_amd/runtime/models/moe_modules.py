@@ -71,15 +71,17 @@ class GalvatronMoEMLP(nn.Module):
         if margs.moe_grouped_gemm:
             self.experts = GroupedMLP(n_local, margs.hidden_size, ffn_local,
                                       dtype=dtype, gated=gated,
-                                      act=margs.hidden_act)
+                                      act=margs.hidden_act,
+                                      init_std=margs.init_method_std)
         else:
             self.experts = SequentialMLP(n_local, margs.hidden_size, ffn_local,
                                          dtype=dtype, gated=gated,
-                                         act=margs.hidden_act)
+                                         act=margs.hidden_act,
+                                         init_std=margs.init_method_std)
         if margs.moe_shared_expert_intermediate_size:
             self.shared = SharedExpertMLP(
                 margs.hidden_size, margs.moe_shared_expert_intermediate_size,
-                dtype=dtype)
+                dtype=dtype, init_std=margs.init_method_std)
             if self.etp > 1:
                 for prm in self.shared.parameters():
                     _tag_tp_replicated(prm)

@@ -32,7 +32,7 @@ class GroupedMLP(nn.Module):
 
     def __init__(self, num_local_experts: int, hidden_size: int,
                  ffn_hidden: int, dtype=None, gated: bool = True,
-                 act: str = "silu"):
+                 act: str = "silu", init_std: float = 0.02):
         super().__init__()
         self.num_local_experts = num_local_experts
         self.gated = gated
@@ -43,8 +43,8 @@ class GroupedMLP(nn.Module):
             torch.empty(num_local_experts, hidden_size, out1, **kw)))
         self.w2 = _mark_expert(nn.Parameter(
             torch.empty(num_local_experts, ffn_hidden, hidden_size, **kw)))
-        nn.init.normal_(self.w1, 0.0, 0.02)
-        nn.init.normal_(self.w2, 0.0, 0.02)
+        nn.init.normal_(self.w1, 0.0, init_std)
+        nn.init.normal_(self.w2, 0.0, init_std)
 
     def _act(self, h: torch.Tensor) -> torch.Tensor:
         import torch.nn.functional as F
@@ -83,7 +83,7 @@ class SequentialMLP(nn.Module):
 
     def __init__(self, num_local_experts: int, hidden_size: int,
                  ffn_hidden: int, dtype=None, gated: bool = True,
-                 act: str = "silu"):
+                 act: str = "silu", init_std: float = 0.02):
         super().__init__()
         self.gated = gated
         self.act = act
@@ -96,7 +96,7 @@ class SequentialMLP(nn.Module):
             nn.Linear(ffn_hidden, hidden_size, bias=False, **kw)
             for _ in range(num_local_experts)])
         for m in list(self.fc1) + list(self.fc2):
-            nn.init.normal_(m.weight, 0.0, 0.02)
+            nn.init.normal_(m.weight, 0.0, init_std)
             _mark_expert(m.weight)
 
     def forward(self, x, tokens_per_expert):
@@ -117,13 +117,14 @@ class SharedExpertMLP(nn.Module):
     """Always-on shared expert added to the routed output
     (reference moe/mlp.py:215)."""
 
-    def __init__(self, hidden_size: int, inter: int, dtype=None):
+    def __init__(self, hidden_size: int, inter: int, dtype=None,
+                 init_std: float = 0.02):
         super().__init__()
         kw = {"dtype": dtype} if dtype else {}
         self.fc1 = nn.Linear(hidden_size, 2 * inter, bias=False, **kw)
         self.fc2 = nn.Linear(inter, hidden_size, bias=False, **kw)
-        nn.init.normal_(self.fc1.weight, 0.0, 0.02)
-        nn.init.normal_(self.fc2.weight, 0.0, 0.02)
+        nn.init.normal_(self.fc1.weight, 0.0, init_std)
+        nn.init.normal_(self.fc2.weight, 0.0, init_std)
 
     def forward(self, x):
         return self.fc2(swiglu(self.fc1(x)))

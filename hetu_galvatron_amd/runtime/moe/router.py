@@ -70,7 +70,8 @@ class TopKRouter(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(self.num_experts, margs.hidden_size,
                         dtype=torch.float32))
-        nn.init.normal_(self.weight, 0.0, 0.02)
+        nn.init.normal_(self.weight, 0.0,
+                        getattr(margs, "init_method_std", 0.02))
         if self.aux_loss_free:
             self.register_buffer(
                 "expert_bias", torch.zeros(self.num_experts,

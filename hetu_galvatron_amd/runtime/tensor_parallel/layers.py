@@ -35,6 +35,18 @@ from .mappings import (
 )
 
 
+def normal_init(std: float):
+    """Megatron init_method_normal (reference train.init_method_std)."""
+    return lambda w: nn.init.normal_(w, 0.0, std)
+
+
+def scaled_init(std: float, num_layers: int):
+    """Megatron scaled_init_method_normal for output projections:
+    std / sqrt(2 * num_layers)."""
+    import math
+    return normal_init(std / math.sqrt(2.0 * max(num_layers, 1)))
+
+
 def divide(a: int, b: int) -> int:
     assert a % b == 0, f"{a} not divisible by {b}"
     return a // b

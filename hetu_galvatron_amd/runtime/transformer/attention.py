@@ -23,6 +23,7 @@ from ..tensor_parallel import (
     RowParallelLinear,
     group_size,
 )
+from ..tensor_parallel.layers import normal_init, scaled_init
 from .attention_impl import (
     DistributedAttention, ZigzagRingAttention, eager_bias_attention,
     flash_bias_attention, local_attention,
@@ -72,14 +73,18 @@ class SelfAttention(nn.Module):
         self.heads_local = self.num_heads // tp
         qkv_out = (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
         lin_group = tp_group if not use_ulysses else None
+        std = getattr(m, "init_method_std", 0.02)
+        L = m.num_hidden_layers
         self.linear_qkv = ColumnParallelLinear(
             self.hidden_size, qkv_out, lin_group,
             bias=m.add_qkv_bias or m.add_bias_linear,
-            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
+            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype,
+            init_method=normal_init(std))
         self.linear_proj = RowParallelLinear(
             self.num_heads * self.head_dim, self.hidden_size, lin_group,
             bias=m.add_bias_linear,
-            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype)
+            sequence_parallel=sequence_parallel and not use_ulysses, dtype=dtype,
+            init_method=scaled_init(std, L))
         self.softmax_scale = m.attention_softmax_scale \
             if getattr(m, "attention_softmax_scale", None) is not None \
             else 1.0 / math.sqrt(self.head_dim)

@@ -18,17 +18,20 @@ from ..tensor_parallel import ColumnParallelLinear, RowParallelLinear
 class MLP(nn.Module):
     def __init__(self, hidden_size: int, ffn_hidden_size: int, group,
                  hidden_act: str = "silu", add_bias: bool = False,
-                 sequence_parallel: bool = False, dtype=None):
+                 sequence_parallel: bool = False, dtype=None,
+                 init_method=None, output_init_method=None):
         super().__init__()
         self.hidden_act = hidden_act
         self.gated = hidden_act in ("silu", "swiglu", "geglu")
         fc1_out = ffn_hidden_size * (2 if self.gated else 1)
         self.fc1 = ColumnParallelLinear(
             hidden_size, fc1_out, group, bias=add_bias,
-            sequence_parallel=sequence_parallel, dtype=dtype)
+            sequence_parallel=sequence_parallel, dtype=dtype,
+            init_method=init_method)
         self.fc2 = RowParallelLinear(
             ffn_hidden_size, hidden_size, group, bias=add_bias,
-            sequence_parallel=sequence_parallel, dtype=dtype)
+            sequence_parallel=sequence_parallel, dtype=dtype,
+            init_method=output_init_method)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h = self.fc1(x)

@@ -221,3 +221,28 @@ def test_attention_dropout_trains():
     a = model.forward_backward(ctx).loss
     b = model.forward_backward(ctx).loss
     assert a != b, "attention dropout should randomize the loss"
+
+
+def test_init_method_std_is_consumed():
+    """train-quality init controls (reference init_method_std +
+    Megatron scaled output init): std flows to every weight; output
+    projections are scaled by 1/sqrt(2L)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel
+
+    stds = {}
+    for std in (0.02, 0.1):
+        torch.manual_seed(0)
+        cfg = load_config(base={"model": {"model_name": "tiny-llama",
+                                          "init_method_std": std}})
+        m = GalvatronModel(cfg)
+        blk = m.stage_model.blocks[1].inner  # first decoder layer
+        stds[std] = (blk.attention.linear_qkv.weight.std().item(),
+                     blk.attention.linear_proj.weight.std().item())
+    import math
+    L = 2  # tiny-llama layers
+    for std, (qkv_std, proj_std) in stds.items():
+        assert abs(qkv_std - std) < std * 0.1, (std, qkv_std)
+        want = std / math.sqrt(2 * L)
+        assert abs(proj_std - want) < want * 0.1, (std, proj_std)
