@@ -304,6 +304,11 @@ def build_hybrid_parallel_model(cfg: GalvatronConfig, plan: HybridParallelPlan,
     if my_stage > 0:
         recv_layout = gen_layer_comm_groups([strategies[lo - 1]], world, rank,
                                             cache)[0][0]
+    if cfg.train.deterministic_mode:
+        for blk in blocks:
+            r = getattr(getattr(blk.inner, "mlp", None), "router", None)
+            if r is not None:
+                r.deterministic = True
     sm = StageModel(
         blocks=blocks, stage=my_stage, pp_deg=pp, world_size=world, rank=rank,
         plan=plan, embed_comm_group=embed_group, cache=cache,

@@ -61,6 +61,10 @@ class TopKRouter(nn.Module):
         self.aux_loss_type = getattr(margs, "moe_aux_loss_type", "aux_loss")
         self.load_balancing_type = getattr(
             margs, "moe_router_load_balancing_type", "none")
+        # train.deterministic_mode (reference: deterministic router topk):
+        # stable argsort instead of torch.topk so expert ties break by
+        # index on every backend/run; set by the builder
+        self.deterministic = False
         self.num_groups = getattr(margs, "moe_router_num_groups", None)
         self.group_topk = getattr(margs, "moe_router_group_topk", None)
         if self.num_groups:
@@ -118,7 +122,13 @@ class TopKRouter(nn.Module):
                                 dtype=torch.bool).scatter_(1, keep, True)
             sel = gs.masked_fill(~gmask.unsqueeze(-1),
                                  float("-inf")).view(n, E)
-        top_vals, idx = torch.topk(sel, self.topk, dim=-1)
+        if self.deterministic:
+            order = torch.argsort(sel, dim=-1, descending=True,
+                                  stable=True)
+            idx = order[:, : self.topk]
+            top_vals = torch.gather(sel, 1, idx)
+        else:
+            top_vals, idx = torch.topk(sel, self.topk, dim=-1)
         gathered = torch.gather(scores, 1, idx)
 
         if self.score_function == "sigmoid":

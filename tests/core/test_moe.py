@@ -417,3 +417,22 @@ def test_router_sinkhorn_balances_selection():
     m = sinkhorn(torch.randn(32, 4))
     torch.testing.assert_close(m.sum(1), torch.full((32,), 1.0 / 32),
                                atol=1e-2, rtol=1e-2)
+
+
+def test_router_deterministic_mode():
+    """train.deterministic_mode: router topk breaks ties by expert index
+    (stable argsort path, reference deterministic_mode arg)."""
+    cfg = make_cfg({"train": {"deterministic_mode": True}})
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    routers = [b.inner.mlp.router for b in model.stage_model.blocks
+               if hasattr(getattr(b.inner, "mlp", None), "router")]
+    assert routers and all(r.deterministic for r in routers)
+    r = routers[0]
+    # tied scores: deterministic mode must pick the LOWEST expert indices
+    x = torch.zeros(4, cfg.model.hidden_size)
+    with torch.no_grad():
+        r.weight.zero_()
+    _, idx, _ = r(x)
+    assert idx.tolist() == [[0, 1]] * 4, idx.tolist()
