@@ -160,6 +160,8 @@ class DataArgs(BaseModel):
     tokenizer_model: Optional[str] = None
     num_workers: int = 0
     synthetic_dataset_size: int = 1024
+    eod_mask_loss: bool = False  # zero the loss on end-of-document tokens
+    eod_token_id: Optional[int] = None  # required when eod_mask_loss
 
 
 class CkptArgs(BaseModel):

@@ -39,15 +39,26 @@ class SyntheticCausalLMDataset(torch.utils.data.Dataset):
                              generator=g)
 
 
-def build_batch_context(tokens: torch.Tensor, device) -> Dict:
-    """tokens: [B, S+1] -> ctx with input_ids/labels [B, S]."""
+def build_batch_context(tokens: torch.Tensor, device,
+                        eod_token: int = None,
+                        eod_mask_loss: bool = False) -> Dict:
+    """tokens: [B, S+1] -> ctx with input_ids/labels [B, S].
+    eod_mask_loss (reference dataloader get_batch loss_mask): zero the
+    loss on end-of-document labels; loss_denom is the GLOBAL unmasked
+    token count (identical on every rank -> grads normalize exactly)."""
     tokens = tokens.to(device)
-    return {
+    ctx = {
         "input_ids": tokens[:, :-1].contiguous(),
         "labels": tokens[:, 1:].contiguous(),
         "batch_size": tokens.shape[0],
         "seq_len": tokens.shape[1] - 1,
     }
+    if eod_mask_loss:
+        assert eod_token is not None, "eod_mask_loss needs data.eod_token_id"
+        mask = (ctx["labels"] != eod_token).to(torch.float32)
+        ctx["loss_mask"] = mask
+        ctx["loss_denom"] = float(mask.sum())
+    return ctx
 
 
 def build_enc_dec_batch_context(enc_ids: torch.Tensor,
@@ -102,4 +113,6 @@ def get_train_iterator(cfg: GalvatronConfig, device,
     while True:
         batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])
         idx = (idx + B) % len(ds)
-        yield build_batch_context(batch, device)
+        yield build_batch_context(batch, device,
+                                  eod_token=cfg.data.eod_token_id,
+                                  eod_mask_loss=cfg.data.eod_mask_loss)

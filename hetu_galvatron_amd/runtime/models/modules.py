@@ -243,4 +243,9 @@ class GalvatronCausalLMHead(nn.Module):
         loss = vocab_parallel_cross_entropy(
             logits.float() if logits.dtype != torch.bfloat16 else logits,
             tgt, group)
+        if "loss_mask" in ctx:  # eod_mask_loss (reference get_batch)
+            mask = ctx["loss_mask"][c.dp_idx * b_loc:(c.dp_idx + 1) * b_loc]
+            mask = mask[:, rows].transpose(0, 1)
+            loss = loss * mask.to(loss.dtype)
+            ctx["_loss_count_local"] = float(mask.sum())
         return loss  # [rows, b_loc] fp32

@@ -49,6 +49,8 @@ def chunk_batch(ctx: Dict, chunks: int, dp: int = 1) -> List[Dict]:
         sub["labels"] = ctx["labels"][lo:lo + b_mb]
         if "enc_input_ids" in ctx:
             sub["enc_input_ids"] = ctx["enc_input_ids"][lo:lo + b_mb]
+        if "loss_mask" in ctx:
+            sub["loss_mask"] = ctx["loss_mask"][lo:lo + b_mb]
         sub["batch_size"] = b_mb
         out.append(sub)
         lo += b_mb
@@ -112,16 +114,19 @@ class PipelineEngine:
         return h
 
     def _loss_of(self, per_token: torch.Tensor, ctx: Dict, chunks: int):
-        denom = float(ctx["seq_len"]) * ctx["global_batch"]
+        denom = ctx.get("loss_denom") or \
+            float(ctx["seq_len"]) * ctx["global_batch"]
         loss = per_token.float().sum() / denom
         if self.loss_scale != 1.0:
             loss = loss * self.loss_scale
         return loss
 
-    def _stat_update(self, stats: StepStats, per_token: torch.Tensor) -> None:
+    def _stat_update(self, stats: StepStats, per_token: torch.Tensor,
+                     ctx: Dict = None) -> None:
         with torch.no_grad():
             stats.loss_sum += per_token.float().sum().item()
-            stats.token_count += per_token.numel()
+            n = (ctx or {}).pop("_loss_count_local", None)
+            stats.token_count += n if n is not None else per_token.numel()
 
     def _set_auto_sync(self, flag: bool) -> None:
         for blk in self.sm.blocks:
@@ -201,7 +206,7 @@ class PipelineEngine:
             if m == n - 1:
                 self._set_auto_sync(True)
             per_token = self._forward_chunk(ctx, None)
-            self._stat_update(stats, per_token)
+            self._stat_update(stats, per_token, ctx)
             loss = self._loss_of(per_token, ctx, n)
             self._snap_after_fwd()
             loss.backward()
@@ -235,7 +240,7 @@ class PipelineEngine:
                 feed = recv_act[:-s_enc]
         out = self._forward_chunk(ctx, feed)
         if self.sm.is_last:
-            self._stat_update(stats, out)
+            self._stat_update(stats, out, ctx)
         elif self.sm.send_carries_memory:
             # ride the memory along the boundary; autograd routes the
             # receiver's grad back into both the stage output and the

@@ -246,3 +246,41 @@ def test_init_method_std_is_consumed():
         assert abs(qkv_std - std) < std * 0.1, (std, qkv_std)
         want = std / math.sqrt(2 * L)
         assert abs(proj_std - want) < want * 0.1, (std, proj_std)
+
+
+def test_eod_mask_loss():
+    """eod_mask_loss (reference get_batch loss_mask): EOD labels carry no
+    loss and no gradient; normalization is over unmasked tokens."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.dataloader import build_batch_context
+
+    eod = 7
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "data": {"eod_mask_loss": True, "eod_token_id": eod},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "lr": 1e-3}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    g = torch.Generator().manual_seed(3)
+    toks = torch.randint(0, cfg.model.vocab_size, (2, 65), generator=g)
+    toks[0, 10] = eod
+    toks[1, 20] = eod
+    ctx = build_batch_context(toks, torch.device("cpu"), eod_token=eod,
+                              eod_mask_loss=True)
+    n_masked = int((toks[:, 1:] == eod).sum())
+    assert ctx["loss_denom"] == 2 * 64 - n_masked
+    stats = model.forward_backward(dict(ctx))
+    assert stats.token_count == ctx["loss_denom"]
+    # unmasked run over the same data differs (mask is consumed)
+    torch.manual_seed(0)
+    model2 = GalvatronModel(load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "lr": 1e-3}}))
+    ctx2 = build_batch_context(toks, torch.device("cpu"))
+    stats2 = model2.forward_backward(ctx2)
+    assert stats2.token_count == 2 * 64
+    assert abs(stats.loss - stats2.loss) > 1e-6
