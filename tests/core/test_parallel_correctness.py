@@ -453,3 +453,21 @@ def test_tp2_qk_layernorm():
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f} " \
                 f"({losses} vs {base_losses})"
+
+
+@pytest.mark.distributed
+def test_dp2_eod_mask_loss():
+    """eod_mask_loss under dp2: identical masked normalization on every
+    rank (loss_denom is computed from the full global batch)."""
+    eod = 5
+    extra = {"data": {"eod_mask_loss": True, "eod_token_id": eod}}
+    base_losses, state_path = get_baseline_with(extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, dp_type="ddp",
+                                      global_bsz=4)
+    from tests.utils import run_distributed
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path, extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
