@@ -108,6 +108,24 @@ def main(argv=None):
             print(f"[rerun] persistent fault at iter {i}: exiting "
                   f"{rsm.exit_code}", file=sys.stderr)
             sys.exit(rsm.exit_code)
+        ei = cfg.train.eval_interval
+        if ei and (i + 1) % ei == 0:
+            # validation pass (forward-only) over the valid split
+            vit = get_train_iterator(cfg, device, global_batch=cur_gbs,
+                                     split="valid")
+            vstats = None
+            for _ in range(max(cfg.train.eval_iters, 1)):
+                st = model.evaluate(next(vit))
+                if vstats is None:
+                    vstats = st
+                else:
+                    vstats.loss_sum += st.loss_sum
+                    vstats.token_count += st.token_count
+            vloss = model.global_loss(vstats)
+            if rank == 0:
+                print(f"[eval] iter {i + 1}: valid loss {vloss:.4f}",
+                      flush=True)
+            mlog.log({"valid_loss": vloss}, i)
         ci = cfg.train.check_weight_consistency_interval
         if ci and (i + 1) % ci == 0:
             from ..utils.consistency import check_param_consistency

@@ -143,6 +143,8 @@ class TrainArgs(BaseModel):
     local_rank: int = 0
     use_fused_adam: bool = True
     deterministic_mode: bool = False
+    eval_interval: int = 0  # run validation every N iters (0 = off)
+    eval_iters: int = 10
     # every N iters, verify replicated params are identical across their
     # groups (reference test_mode realtime checks); 0 = off
     check_weight_consistency_interval: int = 0

@@ -70,10 +70,29 @@ def build_enc_dec_batch_context(enc_ids: torch.Tensor,
     return ctx
 
 
+def split_range(n: int, split: str, which: str) -> tuple:
+    """[lo, hi) index range of the train/valid/test partition under the
+    reference's weight string (data.split, e.g. "969,30,1")."""
+    w = [float(x) for x in split.split(",")]
+    while len(w) < 3:
+        w.append(0.0)
+    tot = sum(w) or 1.0
+    n_train = int(n * w[0] / tot)
+    n_valid = int(n * w[1] / tot)
+    lo, hi = {"train": (0, n_train),
+              "valid": (n_train, n_train + n_valid),
+              "test": (n_train + n_valid, n)}[which]
+    if hi <= lo:  # degenerate split: fall back to the whole set
+        return 0, n
+    return lo, hi
+
+
 def get_train_iterator(cfg: GalvatronConfig, device,
-                       global_batch: Optional[int] = None) -> Iterator[Dict]:
+                       global_batch: Optional[int] = None,
+                       split: str = "train") -> Iterator[Dict]:
     """Yield batch contexts of the global batch size, cycling the dataset
-    (reference: dataloader.py:462 get_train_valid_test_data_iterators)."""
+    (reference: dataloader.py:462 get_train_valid_test_data_iterators).
+    split selects the train/valid/test partition per data.split."""
     B = global_batch or cfg.train.global_train_batch_size
     if cfg.data.dataset == "megatron" and cfg.data.data_path:
         from .datasets import build_pretraining_dataset
@@ -109,10 +128,12 @@ def get_train_iterator(cfg: GalvatronConfig, device,
             enc = torch.randint(0, cfg.model.vocab_size, (B, s_enc),
                                 generator=g)
             yield build_enc_dec_batch_context(enc, batch, device)
+    lo, hi = split_range(len(ds), cfg.data.split, split)
+    n = hi - lo
     idx = 0
     while True:
-        batch = torch.stack([ds[(idx + i) % len(ds)] for i in range(B)])
-        idx = (idx + B) % len(ds)
+        batch = torch.stack([ds[lo + (idx + i) % n] for i in range(B)])
+        idx = (idx + B) % n
         yield build_batch_context(batch, device,
                                   eod_token=cfg.data.eod_token_id,
                                   eod_mask_loss=cfg.data.eod_mask_loss)

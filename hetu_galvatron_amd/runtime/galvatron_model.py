@@ -82,6 +82,10 @@ class GalvatronModel(nn.Module):
             self.engine.loss_scale = scaler.scale
         return self.engine.forward_backward(ctx, chunks or self.chunks)
 
+    def evaluate(self, ctx):
+        """Forward-only validation pass -> StepStats."""
+        return self.engine.evaluate(ctx, self.plan.chunks)
+
     def global_loss(self, stats: StepStats, device=None) -> float:
         """Token-weighted mean loss across all ranks (handles vtp row
         duplication by weighting with per-rank token counts)."""

@@ -199,6 +199,31 @@ class PipelineEngine:
         self._finalize_grads()
         return stats
 
+    def evaluate(self, ctx: Dict, chunks: int) -> StepStats:
+        """Forward-only pass (validation): microbatched, p2p forwards
+        between stages, no autograd graph.  Loss statistics accumulate on
+        the last stage exactly as in training."""
+        ctx = dict(ctx)
+        ctx["global_batch"] = ctx["batch_size"]
+        dp = max((blk.groups.strategy.dp for blk in self.sm.blocks
+                  if getattr(blk, "groups", None) is not None), default=1)
+        mb = chunk_batch(ctx, chunks, dp)
+        for m in mb:
+            m["global_batch"] = ctx["global_batch"]
+        stats = StepStats()
+        with torch.no_grad():
+            for m in mb:
+                recv = None
+                if not self.sm.is_first:
+                    recv = p2p.recv_tensor(self._recv_shape(m),
+                                           self.act_dtype, self.prev_rank,
+                                           self.device)
+                _inp, out = self._fwd_step(m, stats, recv)
+                if not self.sm.is_last:
+                    self._send_async(out, self.next_rank)
+            self._drain_sends()
+        return stats
+
     def _no_pipeline(self, mb: List[Dict], stats: StepStats) -> None:
         """reference: pipeline.py:306 no_pipeline_forward_backward."""
         n = len(mb)

@@ -471,3 +471,40 @@ def test_dp2_eod_mask_loss():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
+
+
+def _eval_worker(rank, world, plan_dict, state_path):
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    cfg = make_cfg()
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.from_config_dict(plan_dict)
+    model = GalvatronModel(cfg, plan)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    it = get_train_iterator(cfg, torch.device("cpu"), split="valid")
+    stats = model.evaluate(next(it))
+    return model.global_loss(stats)
+
+
+@pytest.mark.distributed
+def test_pp2_evaluate_matches_single_process():
+    """Forward-only validation across a pipeline split equals the
+    1-process eval loss on the same weights/valid batch."""
+    from tests.utils import run_distributed
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    _, state_path = get_baseline()
+    cfg = make_cfg()
+    model = GalvatronModel(cfg)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    it = get_train_iterator(cfg, torch.device("cpu"), split="valid")
+    want = model.global_loss(model.evaluate(next(it)))
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, pp=2, chunks=2,
+                                      global_bsz=4)
+    res = run_distributed(_eval_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for r, got in enumerate(res):
+        assert abs(got - want) < 1e-3, (r, got, want)
