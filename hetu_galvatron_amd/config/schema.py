@@ -145,6 +145,7 @@ class TrainArgs(BaseModel):
     deterministic_mode: bool = False
     eval_interval: int = 0  # run validation every N iters (0 = off)
     eval_iters: int = 10
+    lr_warmup_init: float = 0.0  # warmup ramps from this LR (reference)
     # every N iters, verify replicated params are identical across their
     # groups (reference test_mode realtime checks); 0 = off
     check_weight_consistency_interval: int = 0
@@ -175,6 +176,7 @@ class CkptArgs(BaseModel):
     load_format: str = "auto"  # auto | hf | dist
     no_load_optim: bool = False
     no_save_optim: bool = False
+    load_iteration: int = 0  # 0 = latest (reference load_iteration)
     distributed_checkpoint: bool = False
 
 

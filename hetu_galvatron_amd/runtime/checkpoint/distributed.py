@@ -97,8 +97,9 @@ def load_distributed_checkpoint(model, opt, sched, cfg,
                                 rerun_state_machine=None) -> int:
     load_dir = load_dir or cfg.ckpt.load
     if iteration is None:
-        iteration = latest_iteration(load_dir)
-        assert iteration is not None, f"no checkpoint in {load_dir}"
+        iteration = (cfg.ckpt.load_iteration
+                     or latest_iteration(load_dir))
+        assert iteration, f"no checkpoint in {load_dir}"
     rank, world = _rank_world()
     it_dir = os.path.join(load_dir, f"iter_{iteration:07d}")
 

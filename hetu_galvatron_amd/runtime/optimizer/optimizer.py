@@ -210,5 +210,6 @@ def get_optimizer_and_param_scheduler(stage_model, cfg):
     sched = OptimizerParamScheduler(
         opt, max_lr=t.lr, min_lr=t.min_lr, warmup_steps=t.lr_warmup_iters,
         decay_steps=t.lr_decay_iters or t.train_iters,
-        decay_style=t.lr_decay_style, wsd_decay_steps=t.lr_wsd_decay_iters)
+        decay_style=t.lr_decay_style, wsd_decay_steps=t.lr_wsd_decay_iters,
+        warmup_init_lr=t.lr_warmup_init)
     return opt, sched

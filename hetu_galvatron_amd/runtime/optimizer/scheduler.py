@@ -11,6 +11,7 @@ import math
 class OptimizerParamScheduler:
     def __init__(self, optimizer, max_lr: float, min_lr: float = 0.0,
                  warmup_steps: int = 0, decay_steps: int = 1,
+                 warmup_init_lr: float = 0.0,
                  decay_style: str = "cosine", wsd_decay_steps: int = 0,
                  start_wd: float = None, end_wd: float = None,
                  wd_incr_steps: int = 0):
@@ -18,6 +19,7 @@ class OptimizerParamScheduler:
         self.max_lr = max_lr
         self.min_lr = min_lr
         self.warmup_steps = warmup_steps
+        self.warmup_init_lr = warmup_init_lr
         self.decay_steps = max(decay_steps, 1)
         self.decay_style = decay_style
         self.wsd_decay_steps = wsd_decay_steps
@@ -30,7 +32,8 @@ class OptimizerParamScheduler:
     def get_lr(self) -> float:
         s = self.num_steps
         if self.warmup_steps > 0 and s <= self.warmup_steps:
-            return self.max_lr * s / self.warmup_steps
+            return self.warmup_init_lr + \
+                (self.max_lr - self.warmup_init_lr) * s / self.warmup_steps
         if self.decay_style == "constant":
             return self.max_lr
         if s >= self.decay_steps:

@@ -311,3 +311,27 @@ def test_eval_loop_and_split():
     out = buf.getvalue()
     assert "[eval] iter 1: valid loss" in out
     assert "[eval] iter 2: valid loss" in out
+
+
+def test_lr_warmup_init_and_load_iteration_fields():
+    """lr_warmup_init ramps warmup from a nonzero LR; ckpt.load_iteration
+    selects a specific saved iteration (reference args)."""
+    from hetu_galvatron_amd.runtime.optimizer.scheduler import (
+        OptimizerParamScheduler)
+
+    class FakeOpt:
+        param_groups = [{"lr": 0.0}]
+        weight_decay = 0.0
+    sch = OptimizerParamScheduler(FakeOpt(), max_lr=1.0, warmup_steps=10,
+                                  decay_steps=20, decay_style="constant",
+                                  warmup_init_lr=0.5)
+    sch.step()
+    assert abs(sch.get_lr() - (0.5 + 0.5 * 1 / 10)) < 1e-9
+    for _ in range(9):
+        sch.step()
+    assert abs(sch.get_lr() - 1.0) < 1e-9
+    from hetu_galvatron_amd.config import load_config
+    cfg = load_config(base={"ckpt": {"load_iteration": 3},
+                            "train": {"lr_warmup_init": 0.5}})
+    assert cfg.ckpt.load_iteration == 3
+    assert cfg.train.lr_warmup_init == 0.5
