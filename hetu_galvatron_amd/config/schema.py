@@ -91,6 +91,7 @@ class ModelArgs(BaseModel):
     moe_router_score_function: str = "softmax"  # softmax | sigmoid
     moe_token_dispatcher_type: str = "alltoall"  # allgather | alltoall
     moe_expert_capacity_factor: Optional[float] = None
+    moe_pad_expert_input_to_capacity: bool = False  # static expert shapes
     moe_router_pre_softmax: bool = False
     moe_grouped_gemm: bool = True
     moe_shared_expert_intermediate_size: int = 0

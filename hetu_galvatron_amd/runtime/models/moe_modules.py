@@ -51,7 +51,10 @@ class GalvatronMoEMLP(nn.Module):
         if margs.moe_token_dispatcher_type == "allgather":
             self.dispatcher = AllGatherDispatcher(ep_group, margs.num_experts)
         else:
-            self.dispatcher = AlltoAllDispatcher(ep_group, margs.num_experts)
+            self.dispatcher = AlltoAllDispatcher(
+                ep_group, margs.num_experts,
+                capacity_factor=margs.moe_expert_capacity_factor,
+                pad_to_capacity=margs.moe_pad_expert_input_to_capacity)
         n_local = margs.num_experts // max(ep, 1)
         ffn = margs.moe_ffn_hidden_size or margs.ffn_hidden_size
         gated = margs.hidden_act in ("silu", "swiglu", "geglu")

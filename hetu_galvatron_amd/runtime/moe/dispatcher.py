@@ -75,7 +75,8 @@ class AlltoAllDispatcher:
     combine(expert_out) -> [n, h] merged with routing probs.
     """
 
-    def __init__(self, ep_group, num_experts: int):
+    def __init__(self, ep_group, num_experts: int,
+                 capacity_factor=None, pad_to_capacity: bool = False):
         self.ep_group = ep_group
         self.num_experts = num_experts
         self.ep = ep_group.size if ep_group is not None else 1
@@ -84,6 +85,14 @@ class AlltoAllDispatcher:
                         else 0)
         assert num_experts % max(self.ep, 1) == 0
         self.local_experts = num_experts // max(self.ep, 1)
+        # moe_pad_expert_input_to_capacity (reference token_dispatcher):
+        # every local expert's input is padded/dropped to a STATIC
+        # capacity -> fixed GEMM shapes step to step (graph-capture
+        # friendly); dropped tokens contribute zero in combine (their
+        # scattered-back output rows stay zero)
+        self.capacity_factor = capacity_factor
+        self.pad_to_capacity = bool(pad_to_capacity and capacity_factor)
+        self._pad_state = None
 
     def dispatch(self, x: torch.Tensor, probs: torch.Tensor,
                  idx: torch.Tensor):
@@ -143,11 +152,61 @@ class AlltoAllDispatcher:
             out = torch.cat(by_expert) if by_expert else buf
             self._seg_sizes = [int(v) for v in seg_sizes]
             tokens_per_expert = my_recv.sum(0)
+            if self.pad_to_capacity:
+                cap = self._capacity(n * k * self.ep)
+                out, tokens_per_expert = self._pad_grouped(
+                    out, tokens_per_expert, cap)
             return out, tokens_per_expert
+        if self.pad_to_capacity:
+            cap = self._capacity(n * k)
+            permuted, counts = self._pad_grouped(permuted, counts, cap)
         return permuted, counts
+
+    def _capacity(self, routed_tokens: int) -> int:
+        import math
+        return int(math.ceil(routed_tokens / self.num_experts *
+                             float(self.capacity_factor)))
+
+    def _pad_grouped(self, grouped: torch.Tensor,
+                     tokens_per_expert: torch.Tensor, cap: int):
+        """grouped: rows sorted by local expert; keep the first `cap`
+        rows per expert (token order), zero-pad the rest."""
+        sizes = [int(v) for v in tokens_per_expert]
+        segs = torch.split(grouped, sizes)
+        pieces = []
+        kept_dst = []   # row in padded buffer for each kept grouped row
+        kept_src = []   # original grouped row index
+        base = 0
+        for e, m in enumerate(sizes):
+            keep = min(m, cap)
+            pieces.append(segs[e][:keep])
+            if keep < cap:
+                pieces.append(grouped.new_zeros(
+                    (cap - keep,) + grouped.shape[1:]))
+            kept_dst.append(torch.arange(e * cap, e * cap + keep))
+            kept_src.append(torch.arange(base, base + keep))
+            base += m
+        dev = grouped.device
+        self._pad_state = (base,
+                           torch.cat(kept_dst).to(dev),
+                           torch.cat(kept_src).to(dev))
+        out = torch.cat(pieces) if pieces else grouped
+        tpe = torch.full((len(sizes),), cap,
+                         dtype=tokens_per_expert.dtype,
+                         device=tokens_per_expert.device)
+        return out, tpe
+
+    def _unpad_grouped(self, expert_out: torch.Tensor) -> torch.Tensor:
+        """padded [E_l*cap, h] -> grouped-size buffer; dropped rows 0."""
+        total, kept_dst, kept_src = self._pad_state
+        buf = expert_out.new_zeros((total,) + expert_out.shape[1:])
+        buf[kept_src] = expert_out[kept_dst]
+        return buf
 
     def combine(self, expert_out: torch.Tensor, n_tokens: int,
                 topk: int) -> torch.Tensor:
+        if self.pad_to_capacity:
+            expert_out = self._unpad_grouped(expert_out)
         if self.ep > 1:
             # undo the per-expert regrouping back to src-major order
             sizes_srcmajor = self._seg_sizes
