@@ -218,6 +218,9 @@ class SearchArgs(BaseModel):
     settle_bsz: int = -1
     settle_chunks: int = -1
     max_tp_deg: int = 8
+    max_sp_deg: int = 8
+    recommend_min_bsz: int = 0  # 1: start the bsz sweep at world size
+    debug_costmodel_coe: float = 1.0  # cost-model calibration multiplier
     max_pp_deg: int = 8
     max_cp_deg: int = 1
     disable_dp: int = 0

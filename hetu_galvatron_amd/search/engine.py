@@ -187,6 +187,10 @@ class SearchEngine:
         self.layer_profile = layer_profile
         self.other_profile = other_profile or OtherProfile()
         self.hw = hardware
+        coe = float(getattr(self.args, "debug_costmodel_coe", 1.0))
+        if self.hw is not None and coe != 1.0:
+            from dataclasses import replace
+            self.hw = replace(self.hw, costmodel_coe=self.hw.costmodel_coe * coe)
         self.mem_unit = mem_unit_mb
         self.mixed_precision = cfg.parallel.mixed_precision == "bf16"
         self.results: List[SearchResult] = []
@@ -243,7 +247,11 @@ class SearchEngine:
         a = self.args
         if a.settle_bsz and a.settle_bsz > 0:
             return [a.settle_bsz]
-        out, b = [], a.min_bsz
+        lo = a.min_bsz
+        if getattr(a, "recommend_min_bsz", 0):
+            # start where every GPU has at least one sample per chunk
+            lo = max(lo, self.world)
+        out, b = [], lo
         while b <= a.max_bsz:
             out.append(b)
             b += a.bsz_scale

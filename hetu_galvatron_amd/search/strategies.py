@@ -48,7 +48,8 @@ def enumerate_strategies(world_size: int, args: SearchArgs,
             else:
                 if args.sp_space in ("tp", "tp+sp"):
                     modes.append(("tp", tsp))
-                if args.sp_space in ("sp", "tp+sp") and not args.disable_sp:
+                if args.sp_space in ("sp", "tp+sp") and not args.disable_sp \
+                        and tsp <= getattr(args, "max_sp_deg", 8):
                     modes.append(("sp", tsp))
             for mode, deg in modes:
                 tp = deg if mode == "tp" else 1

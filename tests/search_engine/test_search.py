@@ -418,3 +418,23 @@ def test_example_yamls_load():
     for y in yamls:
         cfg = load_config(y)
         assert cfg.model.model_name
+
+
+def test_search_knobs_recommend_bsz_sp_cap_coe(tmp_path):
+    # recommend_min_bsz starts the sweep at world size
+    eng = make_engine(tmp_path, settle_bsz=-1, min_bsz=2, max_bsz=18,
+                      bsz_scale=8, recommend_min_bsz=1)
+    assert eng._bsz_candidates()[0] >= 8
+    # max_sp_deg caps ONLY the ulysses leg
+    from hetu_galvatron_amd.search.strategies import enumerate_strategies
+    cfg = load_config(base={"model": {"model_name": "tiny-llama"},
+                            "search": {"max_sp_deg": 2}})
+    strats = enumerate_strategies(8, cfg.search, pp_deg=1)
+    assert all(s.sp <= 2 for s in strats if s.use_ulysses)
+    assert any(s.tp == 8 for s in strats)  # tp leg unaffected
+    # debug_costmodel_coe scales predicted time (throughput /2)
+    e1 = make_engine(tmp_path)
+    e2 = make_engine(tmp_path, debug_costmodel_coe=2.0)
+    r1 = e1.parallelism_optimization(None)
+    r2 = e2.parallelism_optimization(None)
+    assert abs(r2.time_ms / r1.time_ms - 2.0) < 0.05
