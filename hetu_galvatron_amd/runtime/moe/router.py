@@ -16,7 +16,14 @@ import torch.nn.functional as F
 
 class MoEAuxLossAutoScaler(torch.autograd.Function):
     """Pass the activation through; inject the aux-loss gradient on backward
-    (reference moe_utils.py:166)."""
+    (reference moe_utils.py:166 MoEAuxLossAutoScaler).
+
+    main_loss_backward_scale mirrors the main loss's backward scaling —
+    the engine sets it to loss_scale/num_microbatches each step so the
+    accumulated aux gradient represents coeff * mean-over-microbatches
+    (and survives fp16 scale-unscale)."""
+
+    main_loss_backward_scale: float = 1.0
 
     @staticmethod
     def forward(ctx, output: torch.Tensor, aux_loss: torch.Tensor):
@@ -26,7 +33,8 @@ class MoEAuxLossAutoScaler(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
         (aux_loss,) = ctx.saved_tensors
-        return grad_output, torch.ones_like(aux_loss)
+        scale = MoEAuxLossAutoScaler.main_loss_backward_scale
+        return grad_output, torch.full_like(aux_loss, scale)
 
 
 def attach_aux_loss(x: torch.Tensor, aux_loss: torch.Tensor) -> torch.Tensor:

@@ -187,6 +187,9 @@ class PipelineEngine:
         for m in mb:
             m["global_batch"] = ctx["global_batch"]
         stats = StepStats()
+        from ..moe.router import MoEAuxLossAutoScaler
+        MoEAuxLossAutoScaler.main_loss_backward_scale = \
+            self.loss_scale / max(len(mb), 1)
         self._set_auto_sync(False)
         self._after_fwd_snapped = False
         if self.sm.pp_deg == 1:
