@@ -214,9 +214,14 @@ def adapt_hf_config(path: str) -> Dict[str, Any]:
         if hf_key in hf:
             out[our_key] = hf[hf_key]
     mt = hf.get("model_type", "llama")
-    if mt in ("llama", "mistral", "qwen2"):
+    if mt in ("llama", "mistral", "qwen2", "qwen3"):
         out.update(model_type="llama", hidden_act="silu", normalization="rmsnorm",
                    position_embedding_type="rope")
+        if mt == "qwen3":
+            out["qk_layernorm"] = True
+        if hf.get("rope_scaling") and isinstance(hf["rope_scaling"], dict) \
+                and hf["rope_scaling"].get("factor"):
+            out["rope_scaling"] = float(hf["rope_scaling"]["factor"])
     elif mt == "mixtral":
         out.update(model_type="moe-llama", hidden_act="silu", normalization="rmsnorm",
                    position_embedding_type="rope",

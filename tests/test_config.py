@@ -335,3 +335,28 @@ def test_lr_warmup_init_and_load_iteration_fields():
                             "train": {"lr_warmup_init": 0.5}})
     assert cfg.ckpt.load_iteration == 3
     assert cfg.train.lr_warmup_init == 0.5
+
+
+def test_hf_config_adapter_roundtrip(tmp_path):
+    """HF config.json -> our model args (reference hf_config_adapter
+    resolve_model_config:285) and back (create_hf_config:333)."""
+    import json
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.config.model_configs import create_hf_config
+
+    hf = {"model_type": "qwen3", "hidden_size": 256, "num_hidden_layers": 4,
+          "num_attention_heads": 8, "num_key_value_heads": 2,
+          "intermediate_size": 512, "vocab_size": 1000,
+          "max_position_embeddings": 2048, "rms_norm_eps": 1e-6,
+          "rope_theta": 1e6, "rope_scaling": {"factor": 4.0},
+          "sliding_window": None}
+    (tmp_path / "config.json").write_text(json.dumps(hf))
+    cfg = load_config(base={"model": {"hf_config_path": str(tmp_path)}})
+    m = cfg.model
+    assert (m.hidden_size, m.num_hidden_layers, m.kv_heads) == (256, 4, 2)
+    assert m.qk_layernorm and m.rope_scaling == 4.0
+    assert m.normalization == "rmsnorm" and m.model_type == "llama"
+    back = create_hf_config(m)
+    for k in ("hidden_size", "num_hidden_layers", "num_attention_heads",
+              "vocab_size", "max_position_embeddings"):
+        assert back[k] == hf[k], k
