@@ -84,6 +84,15 @@ MODEL_PRESETS: Dict[str, Dict[str, Any]] = {
         hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-6,
         position_embedding_type="rope", rope_theta=1000000.0, add_qkv_bias=True,
     ),
+    "qwen3-8b": dict(
+        model_type="llama", hidden_size=4096, num_hidden_layers=36,
+        num_attention_heads=32, num_key_value_heads=8, head_dim=128,
+        ffn_hidden_size=12288, vocab_size=151936,
+        max_position_embeddings=40960, seq_length=4096,
+        hidden_act="silu", normalization="rmsnorm", norm_epsilon=1e-6,
+        position_embedding_type="rope", rope_theta=1000000.0,
+        qk_layernorm=True, tie_word_embeddings=False,
+    ),
     "mixtral-8x7b": dict(
         model_type="moe-llama", hidden_size=4096, num_hidden_layers=32,
         num_attention_heads=32, num_key_value_heads=8, ffn_hidden_size=14336,
