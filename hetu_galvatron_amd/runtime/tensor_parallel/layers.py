@@ -23,6 +23,7 @@ import torch.nn.functional as F
 
 from .mappings import (
     _gather_along_first_dim,
+    unwrap_group,
     _reduce_scatter_along_first_dim,
     _is_gloo,
     all_reduce_sync,
@@ -79,12 +80,28 @@ class LinearWithAsyncCommunication(torch.autograd.Function):
         group = ctx.group
         ws = group_size(group)
         handle = None
+        gather_handle = None
         if ctx.sequence_parallel and ws > 1:
-            total_input = _gather_along_first_dim(inp, group)
+            if _is_gloo(group):
+                total_input = _gather_along_first_dim(inp, group)
+            else:
+                # async input re-gather for wgrad, overlapped with the
+                # dgrad GEMM below (reference layers.py:321-376)
+                x = inp.contiguous()
+                shape = list(x.shape)
+                shape[0] *= ws
+                total_input = torch.empty(shape, dtype=x.dtype,
+                                          device=x.device)
+                gather_handle = dist.all_gather_into_tensor(
+                    total_input, x, group=unwrap_group(group),
+                    async_op=True)
         else:
             total_input = inp
         dy = dy.contiguous()
         grad_input = dy.matmul(weight)
+        if gather_handle is not None:
+            gather_handle.wait()  # total_input needed by wgrad (and by
+            # the reduce-scatter's producer ordering below)
         sub_grad_input = None
         if ws > 1 and not _is_gloo(group):
             if ctx.sequence_parallel:
