@@ -246,19 +246,23 @@ class CrossAttention(nn.Module):
         if use_ulysses and sp_group is not None:
             sp = group_size(sp_group)
             assert self.num_heads % sp == 0, "sp must divide heads"
+        std = getattr(m, "init_method_std", 0.02)
         self.linear_q = ColumnParallelLinear(
             m.hidden_size, self.num_heads * self.head_dim, lin_group,
             bias=m.add_qkv_bias or m.add_bias_linear,
-            sequence_parallel=seq_par, dtype=dtype)
+            sequence_parallel=seq_par, dtype=dtype,
+            init_method=normal_init(std))
         # kv over the full-seq memory: no sequence-parallel gather
         self.linear_kv = ColumnParallelLinear(
             m.hidden_size, 2 * self.num_kv_heads * self.head_dim, lin_group,
             bias=m.add_qkv_bias or m.add_bias_linear,
-            sequence_parallel=False, dtype=dtype)
+            sequence_parallel=False, dtype=dtype,
+            init_method=normal_init(std))
         self.linear_proj = RowParallelLinear(
             self.num_heads * self.head_dim, m.hidden_size, lin_group,
             bias=m.add_bias_linear, sequence_parallel=seq_par,
-            dtype=dtype)
+            dtype=dtype,
+            init_method=scaled_init(std, m.num_hidden_layers))
         self.softmax_scale = m.attention_softmax_scale \
             if getattr(m, "attention_softmax_scale", None) is not None \
             else 1.0 / math.sqrt(self.head_dim)
