@@ -267,3 +267,45 @@ def test_windowed_generator_incremental_matches_recompute():
         logits = gen._forward_tokens(seq, cache)
         seq = torch.cat([seq, logits.argmax(-1, keepdim=True)], dim=1)
     assert torch.equal(out, seq)
+
+
+def test_qwen3_decode_matches_recompute():
+    """qk_layernorm in the generator's re-implemented layer step: cached
+    decode equals full-prefix recompute on a qk-norm model."""
+    model = make_model("tiny-qwen3")
+    gen = GalvatronGenerator(model, max_batch=2, max_seq=64)
+    assert gen.layers[0].attention.q_layernorm is not None
+    torch.manual_seed(5)
+    ids = torch.randint(0, model.cfg.model.vocab_size, (2, 7))
+    out = gen.generate(ids, max_new_tokens=4, temperature=0.0)
+    seq = ids
+    for _ in range(4):
+        cache = KVCache(len(gen.layers), 2, 64, model.cfg.model.kv_heads,
+                        model.cfg.model.head_dim, seq.device,
+                        dtype=torch.float32)
+        logits = gen._forward_tokens(seq, cache)
+        seq = torch.cat([seq, logits.argmax(-1, keepdim=True)], dim=1)
+    assert torch.equal(out, seq)
+
+
+def test_rotary_interleaved_decode_matches_recompute():
+    """GPT-J interleaved RoPE through the decode path."""
+    from hetu_galvatron_amd.config import load_config
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama", "rotary_interleaved": True},
+        "parallel": {"mixed_precision": "fp32"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "distributed_backend": "gloo"}})
+    torch.manual_seed(0)
+    model = GalvatronModel(cfg)
+    gen = GalvatronGenerator(model, max_batch=1, max_seq=32)
+    torch.manual_seed(6)
+    ids = torch.randint(0, cfg.model.vocab_size, (1, 5))
+    out = gen.generate(ids, max_new_tokens=3, temperature=0.0)
+    seq = ids
+    for _ in range(3):
+        cache = KVCache(len(gen.layers), 1, 32, cfg.model.kv_heads,
+                        cfg.model.head_dim, seq.device, dtype=torch.float32)
+        logits = gen._forward_tokens(seq, cache)
+        seq = torch.cat([seq, logits.argmax(-1, keepdim=True)], dim=1)
+    assert torch.equal(out, seq)
