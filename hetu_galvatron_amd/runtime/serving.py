@@ -24,12 +24,19 @@ from .inference import GalvatronGenerator, KVCache
 __all__ = ["ContinuousBatchingEngine"]
 
 
-def _rope_rows(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor):
+def _rope_rows(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+               interleaved: bool = False):
     """x: [1, b, h, d]; cos/sin: [b, d/2] (one table row per slot)."""
     d2 = x.shape[-1] // 2
-    x1, x2 = x[..., :d2].float(), x[..., d2:].float()
     c = cos[None, :, None, :].float()
     s = sin[None, :, None, :].float()
+    if interleaved:
+        x1, x2 = x[..., 0::2].float(), x[..., 1::2].float()
+        y = torch.empty_like(x, dtype=torch.float32)
+        y[..., 0::2] = x1 * c - x2 * s
+        y[..., 1::2] = x2 * c + x1 * s
+        return y.to(x.dtype)
+    x1, x2 = x[..., :d2].float(), x[..., d2:].float()
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1).to(x.dtype)
 
 
@@ -121,8 +128,9 @@ class ContinuousBatchingEngine:
             qb = q.permute(1, 0, 2, 3).contiguous()
             kb = k.permute(1, 0, 2, 3).contiguous()
             vb = v.permute(1, 0, 2, 3).contiguous()
-            ob, _ = flash_attention_fwd_only(qb, kb, vb, causal=True,
-                                             softmax_scale=gen.scale)
+            ob, _ = flash_attention_fwd_only(
+                qb, kb, vb, causal=True, softmax_scale=gen.scale,
+                window=getattr(gen, "window", None))
             o = ob.permute(1, 0, 2, 3).reshape(q.shape[0], 1, -1)
             h = residual + attn.linear_proj(o)
             residual = h
@@ -149,10 +157,14 @@ class ContinuousBatchingEngine:
             residual = h
             x = layer.input_norm(h)
             q, k, v = gen._split_qkv(attn, attn.linear_qkv(x))
+            if getattr(attn, "q_layernorm", None) is not None:
+                q = attn.q_layernorm(q)
+                k = attn.k_layernorm(k)
             if layer.rotary is not None:
                 cos_t, sin_t = layer.rotary.full_tables(self.max_seq, dev)
-                q = _rope_rows(q, cos_t[pos], sin_t[pos])
-                k = _rope_rows(k, cos_t[pos], sin_t[pos])
+                il = getattr(attn, "rope_interleaved", False)
+                q = _rope_rows(q, cos_t[pos], sin_t[pos], il)
+                k = _rope_rows(k, cos_t[pos], sin_t[pos], il)
             outs = []
             for i, s in enumerate(active):
                 L = self.lengths[s]
@@ -160,7 +172,8 @@ class ContinuousBatchingEngine:
                 self.cache.v[li][s, L] = v[0, i]
                 o = decode_attention(q[:, i], self.cache.k[li][s:s + 1],
                                      self.cache.v[li][s:s + 1], L + 1,
-                                     softmax_scale=gen.scale)
+                                     softmax_scale=gen.scale,
+                                     window=getattr(gen, "window", None))
                 outs.append(o)
             o = torch.cat(outs, 0).unsqueeze(0).reshape(1, len(active), -1)
             h = residual + attn.linear_proj(o)

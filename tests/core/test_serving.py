@@ -147,3 +147,35 @@ def test_eos_early_release():
         steps += 1
         assert steps < 10
     assert eng.outputs[rid] == toks[:toks.index(eos) + 1]
+
+
+def test_engine_matches_generator_qwen3_and_windowed():
+    """The engine's hand-rolled prefill/step must track GalvatronGenerator
+    for qk-layernorm models and sliding-window models (the three serving
+    gaps: step-side qk norm, window in prefill/decode, interleaved rows)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel
+    from hetu_galvatron_amd.runtime.inference import GalvatronGenerator
+    from hetu_galvatron_amd.runtime.serving import ContinuousBatchingEngine
+
+    for name, extra in (("tiny-qwen3", {}),
+                        ("tiny-llama", {"sliding_window": 8}),
+                        ("tiny-llama", {"rotary_interleaved": True})):
+        cfg = load_config(base={
+            "model": dict({"model_name": name}, **extra),
+            "parallel": {"mixed_precision": "fp32"},
+            "train": {"global_train_batch_size": 2, "train_iters": 1,
+                      "distributed_backend": "gloo"}})
+        torch.manual_seed(0)
+        model = GalvatronModel(cfg)
+        gen = GalvatronGenerator(model, max_batch=1, max_seq=32)
+        torch.manual_seed(9)
+        ids = torch.randint(0, cfg.model.vocab_size, (12,))
+        want = gen.generate(ids.unsqueeze(0), max_new_tokens=4,
+                            temperature=0.0)[0, 12:].tolist()
+        eng = ContinuousBatchingEngine(model, max_slots=2, max_seq=32)
+        rid = eng.add_request(ids, max_new_tokens=4)
+        while eng.n_active:
+            eng.step()
+        assert eng.collect(rid) == want, (name, extra)
