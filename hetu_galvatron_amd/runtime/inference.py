@@ -399,6 +399,8 @@ class GalvatronTPGenerator:
         m = self.cfg.model
         assert m.position_embedding_type == "rope"
         self.margs = m
+        assert m.hidden_act in ("silu", "swiglu"), \
+            "tp decode implements the silu/swiglu MLP only"
         self.blocks = sm.blocks
         self.layers = [b.inner for b in self.blocks if b.kind == "decoder"]
         self.embedding = self.blocks[0].inner
