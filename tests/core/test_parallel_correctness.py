@@ -508,3 +508,46 @@ def test_pp2_evaluate_matches_single_process():
                           args=(plan.to_config_dict(), state_path))
     for r, got in enumerate(res):
         assert abs(got - want) < 1e-3, (r, got, want)
+
+
+def _eval_then_train_worker(rank, world, plan_dict, state_path):
+    from hetu_galvatron_amd.core.initialize import initialize_galvatron
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+    from hetu_galvatron_amd.runtime.checkpoint.state import load_full_state
+    cfg = make_cfg()
+    initialize_galvatron(cfg, backend="gloo")
+    plan = HybridParallelPlan.from_config_dict(plan_dict)
+    model = GalvatronModel(cfg, plan)
+    state = torch.load(state_path, weights_only=True)
+    load_full_state(model.stage_model, state, cfg.model)
+    # interleave an eval pass before every train step: the forward-only
+    # path must leave zero3 param/gather state ready for training
+    from hetu_galvatron_amd.runtime import get_optimizer_and_param_scheduler
+    opt, sched = get_optimizer_and_param_scheduler(model.stage_model, cfg)
+    it = get_train_iterator(model.cfg, torch.device("cpu"))
+    vit = get_train_iterator(model.cfg, torch.device("cpu"), split="valid")
+    losses = []
+    for _ in range(STEPS):
+        model.evaluate(next(vit))
+        opt.zero_grad()
+        stats = model.forward_backward(next(it))
+        opt.step()
+        sched.step()
+        losses.append(model.global_loss(stats))
+    return losses
+
+
+@pytest.mark.distributed
+def test_zero3_eval_interleaved_with_training():
+    """evaluate() between zero3 train steps must not disturb gather /
+    reshard bookkeeping: losses still match the plain baseline."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, dp_type="zero3",
+                                      global_bsz=4)
+    res = run_distributed(_eval_then_train_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
