@@ -551,3 +551,21 @@ def test_zero3_eval_interleaved_with_training():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
+
+
+@pytest.mark.distributed
+def test_pp2_eod_mask_loss():
+    """eod masking with the head on the last pipeline stage (loss_mask
+    rides the per-rank batch context, microbatch-sliced by chunk_batch)."""
+    eod = 5
+    extra = {"data": {"eod_mask_loss": True, "eod_token_id": eod}}
+    base_losses, state_path = get_baseline_with(extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 2, pp=2, chunks=2,
+                                      global_bsz=4)
+    from tests.utils import run_distributed
+    res = run_distributed(_dist_worker, world_size=2,
+                          args=(plan.to_config_dict(), state_path, extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
