@@ -46,7 +46,8 @@ def _build_rel_bias(margs: ModelArgs, groups, bidirectional: bool, dtype):
     rb = RelativePositionBias(
         margs.relative_attention_num_buckets,
         margs.relative_attention_max_distance,
-        margs.num_attention_heads, bidirectional=bidirectional, dtype=dtype)
+        margs.num_attention_heads, bidirectional=bidirectional, dtype=dtype,
+        init_std=margs.init_method_std)
     if group_size(groups.tp_group) > 1:
         # full table on every tp rank; per-rank grads cover disjoint head
         # columns, the tp grad all-reduce assembles the full gradient

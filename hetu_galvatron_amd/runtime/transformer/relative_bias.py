@@ -64,7 +64,7 @@ def _onehot_buckets(sq: int, skv: int, bidirectional: bool, num_buckets: int,
 
 class RelativePositionBias(nn.Module):
     def __init__(self, num_buckets: int, max_distance: int, num_heads: int,
-                 bidirectional: bool, dtype=None):
+                 bidirectional: bool, dtype=None, init_std: float = 0.02):
         super().__init__()
         self.num_buckets = num_buckets
         self.max_distance = max_distance
@@ -72,7 +72,7 @@ class RelativePositionBias(nn.Module):
         self.bidirectional = bidirectional
         self.weight = nn.Parameter(torch.empty(
             num_buckets, num_heads, **({"dtype": dtype} if dtype else {})))
-        nn.init.normal_(self.weight, 0.0, 0.02)
+        nn.init.normal_(self.weight, 0.0, init_std)
 
     def forward(self, sq: int, skv: int, device,
                 head_start: int = 0, head_end: int | None = None
