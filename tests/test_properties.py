@@ -115,3 +115,21 @@ def test_mmididx_roundtrip_random_docs(docs, dtype_name):
             got = np.asarray(ds.doc(i))
             assert got.dtype == dt
             assert got.tolist() == d
+
+
+@settings(max_examples=100, deadline=None)
+@given(cp=st.sampled_from([1, 2, 4, 8]), tsp=st.sampled_from([1, 2, 4, 8]),
+       unit=st.integers(1, 3))
+def test_natural_rows_partition_sequence(cp, tsp, unit):
+    """Every (cp_idx, tp_idx) owns a disjoint row set and the union is
+    exactly [0, S) — the invariant redistribute() and the vocab-CE row
+    slicing both build on."""
+    from hetu_galvatron_amd.runtime.redistribute import natural_rows
+    S = 2 * cp * tsp * unit
+    seen = torch.zeros(S, dtype=torch.int32)
+    for ci in range(cp):
+        for ti in range(tsp):
+            rows = natural_rows(S, cp, tsp, ci, ti, torch.device("cpu"))
+            assert rows.shape[0] == S // (cp * tsp)
+            seen[rows] += 1
+    assert (seen == 1).all()
