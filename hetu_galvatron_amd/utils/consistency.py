@@ -6,7 +6,8 @@ runtime to catch desynchronization (missed grad sync, nondeterministic
 kernels).  Here a framework-wide sweep: every parameter that is
 REPLICATED across a group must be bit-identical on all its ranks —
   * tp_replicated-tagged params across the layer's tp group,
-  * every param across its ZeRO sdp group replicas (ddp mode),
+  * every param across its ZeRO sdp group replicas (ddp and zero2 —
+    zero2's post-step allgather re-replicates the bf16 params),
   * expert params across the edp group.
 Enable per-interval via train.check_weight_consistency_interval.
 """
@@ -46,8 +47,12 @@ def check_param_consistency(stage_model) -> List[str]:
             if expert:
                 grp = edp.group if (edp is not None and edp.size > 1) \
                     else None
-            elif blk.flat is not None and blk.flat.mode == "ddp" \
+            elif blk.flat is not None \
+                    and blk.flat.mode in ("ddp", "zero2") \
                     and sdp is not None and sdp.size > 1:
+                # ddp: fully replicated; zero2: bf16 params re-replicated
+                # by the post-step allgather — either way replicas must
+                # be identical between steps
                 grp = sdp.group
             else:
                 grp = None
