@@ -118,6 +118,9 @@ class SelfAttention(nn.Module):
                 "attention_dropout + sliding window is unsupported"
         inner = ZigzagRingAttention(cp_group) if cp > 1 else None
         if use_ulysses and group_size(sp_group) > 1:
+            sp = group_size(sp_group)
+            assert self.num_heads % sp == 0, \
+                f"ulysses sp={sp} must divide num_heads={self.num_heads}"
             self.core_attention = DistributedAttention(
                 sp_group, inner_attention=inner)
         elif inner is not None:

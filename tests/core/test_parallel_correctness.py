@@ -569,3 +569,22 @@ def test_pp2_eod_mask_loss():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
+
+
+@pytest.mark.distributed
+def test_world8_ulysses2_dp4():
+    """Ulysses sp2 composed with dp4 on 8 ranks (tiny-llama has 2 heads,
+    capping the ulysses degree at 2)."""
+    from tests.utils import run_distributed
+    cfg_extra = {"train": {"global_train_batch_size": 8}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 8, tp=2, use_sp=True,
+                                      dp_type="ddp", global_bsz=8,
+                                      vtp=2, vsp=True)
+    res = run_distributed(_dist_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
