@@ -588,3 +588,22 @@ def test_world8_ulysses2_dp4():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
+
+
+@pytest.mark.distributed
+def test_world4_ulysses4_deep_a2a():
+    """Ulysses at its full degree (sp=4 over 4 heads): every head lands
+    on a different rank — the deepest head-scatter the a2a supports."""
+    extra = {"model": {"num_attention_heads": 4,
+                       "num_key_value_heads": 2}}
+    base_losses, state_path = get_baseline_with(extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 4, tp=4, use_sp=True,
+                                      dp_type="ddp", global_bsz=4,
+                                      vtp=4, vsp=True)
+    from tests.utils import run_distributed
+    res = run_distributed(_dist_worker, world_size=4,
+                          args=(plan.to_config_dict(), state_path, extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
