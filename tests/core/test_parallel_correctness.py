@@ -607,3 +607,12 @@ def test_world4_ulysses4_deep_a2a():
         for s, (a, b) in enumerate(zip(losses, base_losses)):
             assert abs(a - b) < TOL, \
                 f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
+
+
+@pytest.mark.distributed
+def test_world4_tp2_cp2():
+    """Megatron-TP x ring-CP on the same layers (tp2 x cp2, world 4):
+    sequence sharded by BOTH the zigzag pairs and the SP split, kv rings
+    inside each tp shard."""
+    run_case(4, HybridParallelPlan.uniform(
+        N_LAYERS, 4, tp=2, cp=2, dp_type="ddp", global_bsz=4, vtp=2))
