@@ -616,3 +616,23 @@ def test_world4_tp2_cp2():
     inside each tp shard."""
     run_case(4, HybridParallelPlan.uniform(
         N_LAYERS, 4, tp=2, cp=2, dp_type="ddp", global_bsz=4, vtp=2))
+
+
+@pytest.mark.distributed
+def test_world8_pp2_tp2_cp2():
+    """4-way composition pp2 x tp2 x cp2 (dp=1) on 8 ranks: pipeline
+    boundaries carry tp-and-zigzag-sharded activations, rings and SP
+    collectives nest inside each stage."""
+    from tests.utils import run_distributed
+    cfg_extra = {"train": {"global_train_batch_size": 8}}
+    base_losses, state_path = get_baseline_with(cfg_extra)
+    plan = HybridParallelPlan.uniform(N_LAYERS, 8, pp=2, tp=2, cp=2,
+                                      dp_type="ddp", global_bsz=8,
+                                      chunks=2, vtp=2)
+    res = run_distributed(_dist_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path,
+                                cfg_extra))
+    for r, losses in enumerate(res):
+        for s, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s}: {a:.4f} vs {b:.4f}"
