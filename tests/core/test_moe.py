@@ -531,3 +531,20 @@ def test_aux_loss_grad_invariant_to_chunks():
         assert abs(float(aux.grad) - 0.25) < 1e-7
     finally:
         MoEAuxLossAutoScaler.main_loss_backward_scale = old
+
+
+@pytest.mark.distributed
+def test_moe_world8_ep4_dp2():
+    """ep=4 (one expert per ep rank) x dp=2 on 8 ranks — the widest
+    expert sharding tiny-moe supports, expert grads reduced over edp=2."""
+    from tests.utils import run_distributed
+    base_losses, state_path = get_baseline()
+    plan = HybridParallelPlan.uniform(
+        num_layers=2, world_size=8, pp=1, tp=1, dp_type="ddp",
+        global_bsz=4, chunks=1, ep=4)
+    res = run_distributed(_moe_worker, world_size=8,
+                          args=(plan.to_config_dict(), state_path, {}))
+    for r, losses in enumerate(res):
+        for s_, (a, b) in enumerate(zip(losses, base_losses)):
+            assert abs(a - b) < TOL, \
+                f"rank {r} step {s_}: {a:.4f} vs {b:.4f}"
