@@ -534,9 +534,9 @@ def test_aux_loss_grad_invariant_to_chunks():
 
 
 @pytest.mark.distributed
-def test_moe_world8_ep4_dp2():
-    """ep=4 (one expert per ep rank) x dp=2 on 8 ranks — the widest
-    expert sharding tiny-moe supports, expert grads reduced over edp=2."""
+def test_moe_world8_ep4():
+    """ep=4 (one expert per ep rank) on 8 ranks — the widest expert
+    sharding tiny-moe supports; dense dp=8, expert grads over edp=2."""
     from tests.utils import run_distributed
     base_losses, state_path = get_baseline()
     # dense dp = 8 (ep shards experts inside the dp domain), so the
