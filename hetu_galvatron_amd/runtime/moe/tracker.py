@@ -21,15 +21,22 @@ def save_aux_loss(name: str, layer_idx: int, value: torch.Tensor) -> None:
 
 
 def reduce_and_get(group=None) -> Dict[str, float]:
-    """Mean per key over recorded steps, then mean over ranks."""
+    """Mean per key over recorded steps, then mean over the ranks that
+    RECORDED the key.  Key sets differ across ranks under pipeline
+    parallelism (each stage owns different MoE layers), so this gathers
+    dicts (collective: every rank must call) instead of all-reducing a
+    fixed-shape tensor — a shape-mismatched all_reduce would deadlock."""
     out = {k: sum(v) / max(len(v), 1) for k, v in _TRACKER.items()}
-    if dist.is_initialized() and dist.get_world_size() > 1 and out:
-        keys = sorted(out)
-        t = torch.tensor([out[k] for k in keys], dtype=torch.float64)
-        dist.all_reduce(t, group=group)
+    if dist.is_initialized() and dist.get_world_size() > 1:
         ws = dist.get_world_size(group) if group is not None \
             else dist.get_world_size()
-        out = {k: float(t[i] / ws) for i, k in enumerate(keys)}
+        gathered: list = [None] * ws
+        dist.all_gather_object(gathered, out, group=group)
+        merged: Dict[str, list] = {}
+        for d in gathered:
+            for k, v in (d or {}).items():
+                merged.setdefault(k, []).append(v)
+        out = {k: sum(v) / len(v) for k, v in merged.items()}
     return out
 
 

@@ -552,3 +552,25 @@ def test_moe_world8_ep4():
     for r, losses in enumerate(res):
         assert all(l == l for l in losses)
         assert losses[-1] < losses[0] + 0.5
+
+
+def _tracker_pp_worker(rank, world):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from hetu_galvatron_amd.runtime.moe import tracker
+    tracker.clear()
+    # each "stage" records a DIFFERENT layer's aux (pp key divergence)
+    tracker.save_aux_loss("moe_aux", rank, 1.0 + rank)
+    out = tracker.reduce_and_get()
+    tracker.clear()
+    return out
+
+
+@pytest.mark.distributed
+def test_aux_tracker_handles_divergent_keys():
+    """reduce_and_get must not deadlock when ranks recorded different
+    layer keys (pipeline-split MoE); every key surfaces on every rank."""
+    from tests.utils import run_distributed
+    res = run_distributed(_tracker_pp_worker, world_size=2)
+    for out in res:
+        assert out == {"moe_aux/layer_0": 1.0, "moe_aux/layer_1": 2.0}, out
