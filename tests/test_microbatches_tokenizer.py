@@ -96,3 +96,35 @@ def test_train_cli_t5(tmp_path):
           "train.global_train_batch_size=2", "train.train_iters=2",
           "train.lr=1e-4", "train.lr_decay_style=constant",
           "train.distributed_backend=gloo"])
+
+
+def test_torchrun_train_moe_pp2(tmp_path):
+    """cli.train end-to-end with MoE across a pipeline split: the aux
+    tracker's metric reduction must handle per-stage key divergence
+    (regression for the dict-gather fix)."""
+    import socket
+    import subprocess
+    import sys
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes", "1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", str(port),
+           "-m", "hetu_galvatron_amd.cli.train",
+           "model.model_name=tiny-moe",
+           "parallel.pp_deg=2", "parallel.chunks=2",
+           "parallel.mixed_precision=fp32",
+           "train.global_train_batch_size=4", "train.train_iters=2",
+           "train.lr=1e-3", "train.lr_decay_style=constant",
+           "train.distributed_backend=gloo",
+           f"logging.tensorboard_dir={tmp_path}"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "iter     2" in r.stdout
+    import json
+    lines = [json.loads(l) for l in
+             open(tmp_path / "metrics.jsonl").read().splitlines()]
+    moe_keys = [k for l in lines for k in l if k.startswith("moe/")]
+    assert moe_keys, "aux-loss metrics missing from the pp2 MoE run"
