@@ -438,3 +438,14 @@ def test_search_knobs_recommend_bsz_sp_cap_coe(tmp_path):
     r1 = e1.parallelism_optimization(None)
     r2 = e2.parallelism_optimization(None)
     assert abs(r2.time_ms / r1.time_ms - 2.0) < 0.05
+
+
+def test_plan_report_70b_example():
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "tools/plan_report.py",
+         "examples/plan_llama3_70b_8gpu_seq32k.json"],
+        capture_output=True, text=True, check=True).stdout
+    assert "pp=1" in out and "sp8" in out and "tp8" in out
+    assert "+ckpt" in out  # heterogeneous per-layer plan renders
