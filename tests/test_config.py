@@ -381,3 +381,21 @@ def test_training_is_deterministic_run_to_run(tmp_path):
     a = run("a")
     b = run("b")
     assert a and a == b
+
+
+def test_eval_under_fp16():
+    """Forward-only validation under fp16 (no scaler interaction; raw
+    per-token sums in the stats)."""
+    import torch
+    from hetu_galvatron_amd.config import load_config
+    from hetu_galvatron_amd.runtime import GalvatronModel, get_train_iterator
+    cfg = load_config(base={
+        "model": {"model_name": "tiny-llama"},
+        "parallel": {"mixed_precision": "fp16"},
+        "train": {"global_train_batch_size": 2, "train_iters": 1,
+                  "lr": 1e-3}})
+    torch.manual_seed(0)
+    m = GalvatronModel(cfg)
+    it = get_train_iterator(cfg, torch.device("cpu"), split="valid")
+    st = m.evaluate(next(it))
+    assert st.loss == st.loss and st.loss > 0
